@@ -1,0 +1,291 @@
+import asyncio
+import time
+
+import numpy as np
+import pytest
+
+from clearml_serving_amd.schemas import (
+    CanaryEP,
+    EndpointMetricLogging,
+    MetricType,
+    ModelEndpoint,
+    ModelMonitoring,
+)
+from clearml_serving_amd.serving.processor import (
+    EndpointNotFoundError,
+    FastWriteCounter,
+    ModelRequestProcessor,
+)
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def _sklearn_model(tmp_path, slope=2.0):
+    import joblib
+    from sklearn.linear_model import LinearRegression
+
+    X = np.array([[0.0], [1.0], [2.0]])
+    y = slope * X[:, 0]
+    model = LinearRegression().fit(X, y)
+    p = tmp_path / "model-{}.pkl".format(slope)
+    joblib.dump(model, str(p))
+    return str(p)
+
+
+def test_fast_write_counter():
+    c = FastWriteCounter()
+    assert c.value() == 0
+    c.inc(); c.inc(); c.dec()
+    assert c.value() == 1
+    assert c.value() == 1  # reads don't drift
+
+
+def test_normalize_endpoint_url(processor):
+    n = processor._normalize_endpoint_url
+    assert n("model", "1") == "model/1"
+    assert n("model", None) == "model"
+    assert n("model/", "") == "model"
+
+
+def test_add_endpoint_and_serialize_roundtrip(processor, store, tmp_path):
+    path = _sklearn_model(tmp_path)
+    rec = store.register_model(name="lin", project="p", path=path)
+    url = processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="lin_model", version="1",
+        model_id=rec.model_id,
+    ))
+    assert url == "lin_model/1"
+    processor.serialize()
+
+    p2 = ModelRequestProcessor(task_id=processor.get_id(), store=store)
+    p2.deserialize(skip_sync=True)
+    assert "lin_model/1" in p2.get_endpoints()
+    assert p2.get_endpoints()["lin_model/1"].model_id == rec.model_id
+
+
+def test_add_endpoint_by_model_query(processor, store, tmp_path):
+    path = _sklearn_model(tmp_path)
+    store.register_model(name="query-me", project="p", path=path, tags=["prod"])
+    url = processor.add_endpoint(
+        ModelEndpoint(engine_type="sklearn", serving_url="q"),
+        model_name="^query-me$", model_project="p", model_tags=["prod"],
+    )
+    assert processor.get_endpoints()[url].model_id
+
+
+def test_process_request_sklearn(processor, store, tmp_path):
+    path = _sklearn_model(tmp_path, slope=3.0)
+    rec = store.register_model(name="lin", project="p", path=path)
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="lin", model_id=rec.model_id,
+    ))
+    out = run(processor.process_request("lin", "", [[2.0]]))
+    assert abs(float(np.asarray(out)[0]) - 6.0) < 1e-6
+
+
+def test_endpoint_not_found(processor):
+    with pytest.raises(EndpointNotFoundError):
+        run(processor.process_request("nope", "", {}))
+
+
+def test_canary_fixed_weights(processor, store, tmp_path):
+    for ver, slope in (("1", 1.0), ("2", 10.0)):
+        rec = store.register_model(
+            name="m" + ver, project="p", path=_sklearn_model(tmp_path, slope))
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="sklearn", serving_url="mod", version=ver,
+            model_id=rec.model_id))
+    processor.add_canary_endpoint(CanaryEP(
+        endpoint="mod_canary", weights=[0.5, 0.5],
+        load_endpoints=["mod/1", "mod/2"]))
+    processor._update_canary_lookup()
+    seen = set()
+    for _ in range(100):
+        out = run(processor.process_request("mod_canary", "", [[1.0]]))
+        seen.add(round(float(np.asarray(out)[0]), 3))
+    assert seen == {1.0, 10.0}
+
+
+def test_canary_prefix_newest_first(processor, store, tmp_path):
+    # three versions; prefix canary with 2 weights must pick the two newest,
+    # first weight -> newest version (reference :795-808)
+    for ver in ("1", "2", "10"):
+        rec = store.register_model(
+            name="m" + ver, project="p",
+            path=_sklearn_model(tmp_path, float(ver)))
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="sklearn", serving_url="pfx", version=ver,
+            model_id=rec.model_id))
+    processor.add_canary_endpoint(CanaryEP(
+        endpoint="pfx_canary", weights=[1.0, 0.0],
+        load_endpoint_prefix="pfx"))
+    processor._update_canary_lookup()
+    route = processor._canary_route["pfx_canary"]
+    # zero-padded version sort: 10 > 2 > 1
+    assert route["endpoints"] == ["pfx/10", "pfx/2"]
+    out = run(processor.process_request("pfx_canary", "", [[1.0]]))
+    assert abs(float(np.asarray(out)[0]) - 10.0) < 1e-6
+
+
+def test_canary_weight_normalization(processor, store, tmp_path):
+    rec = store.register_model(name="m", project="p",
+                               path=_sklearn_model(tmp_path))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="w", version="1",
+        model_id=rec.model_id))
+    processor.add_canary_endpoint(CanaryEP(
+        endpoint="wc", weights=[2.0, 6.0], load_endpoints=["w/1", "w/1"]))
+    processor._update_canary_lookup()
+    assert processor._canary_route["wc"]["weights"] == [0.25, 0.75]
+
+
+def test_canary_both_modes_rejected(processor):
+    with pytest.raises(ValueError):
+        processor.add_canary_endpoint(CanaryEP(
+            endpoint="c", weights=[1.0], load_endpoints=["a/1"],
+            load_endpoint_prefix="a"))
+
+
+def test_auto_update_version_assignment(processor, store, tmp_path):
+    processor.add_model_monitoring(ModelMonitoring(
+        base_serving_url="auto", engine_type="sklearn",
+        monitor_name="^auto-model$", max_versions=2))
+    m1 = store.register_model(name="auto-model", project="p",
+                              path=_sklearn_model(tmp_path, 1.0))
+    processor._update_monitored_models()
+    assert processor._model_monitoring_versions["auto"] == {m1.model_id: 1}
+    assert "auto/1" in processor.get_synced_endpoints()
+
+    time.sleep(0.01)
+    m2 = store.register_model(name="auto-model", project="p",
+                              path=_sklearn_model(tmp_path, 2.0))
+    processor._update_monitored_models()
+    versions = processor._model_monitoring_versions["auto"]
+    assert versions == {m1.model_id: 1, m2.model_id: 2}
+
+    # a third model pushes out the oldest (max_versions=2)
+    time.sleep(0.01)
+    m3 = store.register_model(name="auto-model", project="p",
+                              path=_sklearn_model(tmp_path, 3.0))
+    processor._update_monitored_models()
+    versions = processor._model_monitoring_versions["auto"]
+    assert versions == {m2.model_id: 2, m3.model_id: 3}
+    eps = processor.get_synced_endpoints()
+    assert "auto/2" in eps and "auto/3" in eps and "auto/1" not in eps
+    # serving the latest version works end to end
+    out = run(processor.process_request("auto", "3", [[1.0]]))
+    assert abs(float(np.asarray(out)[0]) - 3.0) < 1e-6
+
+
+def test_auto_update_persisted(processor, store, tmp_path):
+    processor.add_model_monitoring(ModelMonitoring(
+        base_serving_url="auto2", engine_type="sklearn",
+        monitor_name="^auto2$", max_versions=1))
+    m1 = store.register_model(name="auto2", project="p",
+                              path=_sklearn_model(tmp_path, 1.0))
+    processor._update_monitored_models()
+    processor.serialize()
+
+    p2 = ModelRequestProcessor(task_id=processor.get_id(), store=store)
+    p2.deserialize(skip_sync=True)
+    assert p2._model_monitoring_versions["auto2"] == {m1.model_id: 1}
+    assert "auto2/1" in p2.get_synced_endpoints()
+
+
+def test_metric_logging_validation(processor, store, tmp_path):
+    with pytest.raises(ValueError):
+        processor.add_metric_logging(EndpointMetricLogging(endpoint="ghost/1"))
+    # prefix metrics don't need an existing endpoint
+    processor.add_metric_logging(EndpointMetricLogging(
+        endpoint="anything/*",
+        metrics={"x": MetricType(type="value")}))
+    assert "anything/*" in processor.list_endpoint_logging()
+
+
+def test_stats_emission(processor, store, tmp_path):
+    rec = store.register_model(name="m", project="p",
+                               path=_sklearn_model(tmp_path, 5.0))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="stats_ep", model_id=rec.model_id))
+    processor._metric_log_freq = 1.0
+    run(processor.process_request("stats_ep", "", [[1.0]]))
+    stat = processor._stats_queue.get(timeout=0)
+    assert stat["_url"] == "stats_ep"
+    assert stat["_count"] == 1
+    assert stat["_latency"] >= 0
+
+
+def test_hot_reload_swaps_engine_cache(processor, store, tmp_path):
+    rec = store.register_model(name="m", project="p",
+                               path=_sklearn_model(tmp_path, 2.0))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="swap", model_id=rec.model_id))
+    processor.serialize()
+    processor.deserialize(skip_sync=True)
+    out = run(processor.process_request("swap", "", [[1.0]]))
+    assert abs(float(np.asarray(out)[0]) - 2.0) < 1e-6
+    assert "swap" in processor._engine_processor_lookup
+
+    # second processor instance changes the model; first one hot-reloads
+    p2 = ModelRequestProcessor(task_id=processor.get_id(), store=store)
+    p2.deserialize(skip_sync=True)
+    rec2 = store.register_model(name="m2", project="p",
+                                path=_sklearn_model(tmp_path, 7.0))
+    p2.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="swap", model_id=rec2.model_id))
+    p2.serialize()
+
+    changed = processor.deserialize()  # full stall-swap path
+    assert changed
+    assert "swap" not in processor._engine_processor_lookup  # cache flushed
+    out = run(processor.process_request("swap", "", [[1.0]]))
+    assert abs(float(np.asarray(out)[0]) - 7.0) < 1e-6
+
+
+def test_deserialize_noop_when_unchanged(processor):
+    processor.serialize()
+    assert processor.deserialize() is True
+    assert processor.deserialize() is False  # revision unchanged -> no-op
+
+
+def test_custom_engine_with_preprocess_artifact(processor, store, tmp_path):
+    code = tmp_path / "preprocess.py"
+    code.write_text(
+        "class Preprocess(object):\n"
+        "    def preprocess(self, body, state, collect_custom_statistics_fn=None):\n"
+        "        return body['x']\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        return data * 2\n"
+        "    def postprocess(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        return {'y': data}\n"
+    )
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom", serving_url="dbl"),
+        preprocess_code=str(code),
+    )
+    out = run(processor.process_request("dbl", "", {"x": 21}))
+    assert out == {"y": 42}
+
+
+def test_custom_async_pipeline(processor, store, tmp_path):
+    # an async ensemble endpoint that fans out to a sklearn endpoint via
+    # the injected send_request hook
+    rec = store.register_model(name="m", project="p",
+                               path=_sklearn_model(tmp_path, 2.0))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="base", model_id=rec.model_id))
+    code = tmp_path / "pipeline.py"
+    code.write_text(
+        "class Preprocess(object):\n"
+        "    async def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        r = await self.send_request(endpoint='base', data=data)\n"
+        "        return {'pipelined': r}\n"
+    )
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom_async", serving_url="pipe"),
+        preprocess_code=str(code),
+    )
+    out = run(processor.process_request("pipe", "", [[5.0]]))
+    assert abs(float(np.asarray(out["pipelined"])[0]) - 10.0) < 1e-6

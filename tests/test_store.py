@@ -1,0 +1,71 @@
+import os
+import time
+
+
+def test_session_lifecycle(store):
+    sid = store.create_session(name="svc", project="proj", tags=["a"])
+    assert store.get_session(sid)["name"] == "svc"
+    assert store.resolve_session(None) == sid
+    assert store.revision(sid) == 0
+    store.set_config_object(sid, "endpoints", {"a": 1})
+    assert store.revision(sid) == 1
+    assert store.get_config_object(sid, "endpoints") == {"a": 1}
+    assert store.get_config_object(sid, "missing", {"d": 1}) == {"d": 1}
+
+
+def test_params(store):
+    sid = store.create_session()
+    store.set_params(sid, {"serving_base_url": "http://x", "freq": 0.5})
+    assert store.get_params(sid) == {"serving_base_url": "http://x", "freq": 0.5}
+
+
+def test_model_registry_query(store):
+    m1 = store.register_model(name="alpha", project="p1", tags=["t1"])
+    time.sleep(0.01)
+    m2 = store.register_model(name="alpha-v2", project="p1", tags=["t1", "t2"],
+                              published=True)
+    time.sleep(0.01)
+    m3 = store.register_model(name="beta", project="p2")
+
+    # newest first
+    ids = [m.model_id for m in store.query_models()]
+    assert ids == [m3.model_id, m2.model_id, m1.model_id]
+    assert [m.model_id for m in store.query_models(project="p1")] == \
+        [m2.model_id, m1.model_id]
+    # name is a regex
+    assert len(store.query_models(name="^alpha")) == 2
+    assert len(store.query_models(name="^alpha$")) == 1
+    assert [m.model_id for m in store.query_models(tags=["t2"])] == [m2.model_id]
+    assert [m.model_id for m in store.query_models(only_published=True)] == \
+        [m2.model_id]
+
+
+def test_model_file_copy(store, tmp_path):
+    f = tmp_path / "model.bin"
+    f.write_bytes(b"weights")
+    rec = store.register_model(name="m", path=str(f))
+    local = store.get_model_local_path(rec.model_id)
+    assert os.path.isfile(local)
+    assert open(local, "rb").read() == b"weights"
+
+
+def test_artifact_store(store, tmp_path):
+    sid = store.create_session()
+    f = tmp_path / "preprocess.py"
+    f.write_text("class Preprocess:\n    pass\n")
+    digest = store.upload_artifact(sid, "py_code_x", str(f))
+    art = store.get_artifact(sid, "py_code_x")
+    assert art["sha256"] == digest
+    assert os.path.isfile(art["path"])
+
+
+def test_cross_instance_visibility(tmp_path):
+    from clearml_serving_amd.store import ServingStore
+
+    s1 = ServingStore(str(tmp_path / "shared"))
+    s2 = ServingStore(str(tmp_path / "shared"))
+    sid = s1.create_session(name="visible")
+    assert s2.get_session(sid)["name"] == "visible"
+    s1.set_config_object(sid, "endpoints", {"x": 1})
+    assert s2.revision(sid) == 1
+    assert s2.get_config_object(sid, "endpoints") == {"x": 1}
