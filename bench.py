@@ -1,0 +1,205 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark (driver contract).
+
+Measures the BASELINE.json headline metric: requests/sec (+ p50 latency) for
+ResNet-50 and BERT-base endpoints with dynamic auto-batching, bf16, synthetic
+data, random-init weights, on N GPUs of one node.
+
+One rank per GPU (torch.distributed over RCCL); each rank runs an independent
+serving replica (data-parallel serving = the reference's N-workers topology,
+entrypoint.sh:56-72) -- weak scaling. Each *step* pushes a fixed burst of
+requests through the full serving path (ModelRequestProcessor dispatch ->
+dynamic batcher -> hipGraph replay on the GPU) and awaits all responses.
+
+  python bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import tempfile
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+REQ_PER_STEP = int(os.environ.get("CMLS_BENCH_REQ", 512))
+# per rank per step: REQ/2 ResNet-50 + REQ/2 BERT-base requests
+
+
+def log(msg):
+    if int(os.environ.get("RANK", 0)) == 0:
+        print("[bench] {}".format(msg), file=sys.stderr, flush=True)
+
+
+def build_processor(device_idx, tmpdir, use_graphs=True):
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.store import ServingStore
+
+    store = ServingStore(os.path.join(tmpdir, "store"))
+    processor = ModelRequestProcessor(store=store, name="bench",
+                                      force_create=True)
+    processor._metric_log_freq = 0.0  # stats sampling off in the bench
+
+    cards = {
+        "resnet50_bench": {"arch": "resnet50", "num_classes": 1000,
+                           "dtype": "bfloat16"},
+        "bert_bench": {"arch": "bert-base", "num_labels": 2,
+                       "dtype": "bfloat16"},
+    }
+    for name, card in cards.items():
+        p = os.path.join(tmpdir, name + ".json")
+        with open(p, "wt") as f:
+            json.dump(card, f)
+        rec = store.register_model(name=name, project="bench", path=p)
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="hip", serving_url=name, model_id=rec.model_id,
+            auxiliary_cfg={
+                "max_batch_size": 64,
+                "max_queue_delay_us": 4000,
+                "dtype": "bfloat16" if torch.cuda.is_available() else "float32",
+                "use_graphs": use_graphs,
+                "gpu": device_idx,
+            },
+        ))
+    return processor
+
+
+async def run_step(processor, resnet_inputs, bert_inputs, latencies=None):
+    """One bench step: REQ_PER_STEP requests through the serving path."""
+
+    async def one(endpoint, body):
+        t0 = time.perf_counter()
+        await processor.process_request(endpoint, "", body)
+        if latencies is not None:
+            latencies.append(time.perf_counter() - t0)
+
+    tasks = []
+    n = REQ_PER_STEP // 2
+    for i in range(n):
+        tasks.append(one("resnet50_bench", resnet_inputs[i % len(resnet_inputs)]))
+        tasks.append(one("bert_bench", bert_inputs[i % len(bert_inputs)]))
+    await asyncio.gather(*tasks)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--no-graphs", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    has_gpu = torch.cuda.is_available()
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if has_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+        if has_gpu:
+            torch.cuda.set_device(local_rank)
+
+    device_idx = local_rank if has_gpu else 0
+
+    # synthetic requests of the headline shapes (BASELINE.json configs 2+3):
+    # ResNet-50: [3, 224, 224] bf16; BERT-base: seq_len 128 token ids
+    torch.manual_seed(1234 + rank)
+    n_distinct = 8
+    dtype = torch.bfloat16 if has_gpu else torch.float32
+    resnet_inputs = [torch.randn(3, 224, 224, dtype=dtype)
+                     for _ in range(n_distinct)]
+    bert_inputs = [
+        {"input_ids": torch.randint(0, 30000, (128,), dtype=torch.int64),
+         "attention_mask": torch.ones(128, dtype=torch.int32)}
+        for _ in range(n_distinct)
+    ]
+
+    with tempfile.TemporaryDirectory() as tmpdir:
+        processor = build_processor(device_idx, tmpdir,
+                                    use_graphs=not args.no_graphs and has_gpu)
+
+        loop = asyncio.new_event_loop()
+        asyncio.set_event_loop(loop)
+
+        log("warmup: {} steps (includes hipGraph capture per bucket)".format(
+            args.warmup))
+        for _ in range(max(args.warmup, 1)):
+            loop.run_until_complete(
+                run_step(processor, resnet_inputs, bert_inputs))
+
+        # ---- timed region: barrier + sync on both sides ---- #
+        if dist:
+            dist.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        latencies = []
+        for _ in range(args.steps):
+            loop.run_until_complete(
+                run_step(processor, resnet_inputs, bert_inputs, latencies))
+        if has_gpu:
+            torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+        elapsed = time.perf_counter() - t0
+
+        # slowest rank defines job time
+        if dist:
+            t = torch.tensor([elapsed], dtype=torch.float64,
+                             device="cuda" if has_gpu else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+
+        total_requests = world * args.steps * REQ_PER_STEP
+        rps = total_requests / elapsed
+        p50 = statistics.median(latencies) * 1000 if latencies else None
+        p99 = (sorted(latencies)[int(len(latencies) * 0.99)] * 1000
+               if latencies else None)
+
+        if rank == 0:
+            result = {
+                "metric": "requests/sec + p50 latency, ResNet-50 & BERT-base "
+                          "dyn-batch, 1/2/4/8 MI355X",
+                "value": round(rps, 2),
+                "unit": "requests/s",
+                "n_gpus": world,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(elapsed / args.steps * 1000, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16" if has_gpu else "fp32",
+                "data": "synthetic",
+                "config": {
+                    "model": "resnet50+bert-base",
+                    "global_batch": REQ_PER_STEP * world,
+                    "seq_len": 128,
+                    "image": "3x224x224",
+                    "parallelism": "dp{}".format(world),
+                    "max_batch_size": 64,
+                    "p50_latency_ms": round(p50, 2) if p50 else None,
+                    "p99_latency_ms": round(p99, 2) if p99 else None,
+                    "requests_per_step": REQ_PER_STEP,
+                },
+            }
+            print(json.dumps(result), flush=True)
+
+        processor.stop()
+        if dist:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,77 @@
+"""Serving-registry adapter for the native LLM engine.
+
+Maps the reference's vLLM serve types (reference: preprocess_service.py:
+1097-1336 -- ``v1_chat_completions``, ``v1_completions``, ``v1_models``,
+``v1_embeddings``) onto the in-process engine: one engine singleton per
+process (the reference also keeps a process-wide vLLM singleton,
+preprocess_service.py:619-631).
+"""
+
+from typing import Any, Optional
+
+from ...schemas import ModelEndpoint
+from ...serving.preprocess import BasePreprocessRequest
+
+
+@BasePreprocessRequest.register_engine("llm", modules=["torch"])
+class LlmPreprocessRequest(BasePreprocessRequest):
+    is_preprocess_async = True
+    is_process_async = True
+    is_postprocess_async = True
+
+    _engine_singleton = None
+
+    def __init__(self, model_endpoint: ModelEndpoint, task=None):
+        super().__init__(model_endpoint, task)
+        from .engine import LlmEngine, LlmEngineConfig
+
+        aux = dict(model_endpoint.auxiliary_cfg or {})
+        # user Preprocess.load() may return a config dict (reference lets the
+        # preprocess override vllm_model_config, examples/vllm/preprocess.py)
+        if isinstance(self._model, dict):
+            aux.update(self._model)
+        if LlmPreprocessRequest._engine_singleton is None:
+            model_path = self._get_local_model_file()
+            cfg = LlmEngineConfig.from_aux(model_path, aux)
+            LlmPreprocessRequest._engine_singleton = LlmEngine(cfg)
+            LlmPreprocessRequest._engine_singleton.start()
+        self._engine = LlmPreprocessRequest._engine_singleton
+        self._served_name = model_endpoint.serving_url
+
+    async def preprocess(self, request, state, collect_custom_statistics_fn=None):
+        if self._preprocess is not None and hasattr(self._preprocess, "preprocess"):
+            import asyncio
+
+            fn = self._preprocess.preprocess
+            if asyncio.iscoroutinefunction(fn):
+                return await fn(request, state, collect_custom_statistics_fn)
+            return fn(request, state, collect_custom_statistics_fn)
+        return request
+
+    async def postprocess(self, data, state, collect_custom_statistics_fn=None):
+        if self._preprocess is not None and hasattr(self._preprocess, "postprocess"):
+            import asyncio
+
+            fn = self._preprocess.postprocess
+            if asyncio.iscoroutinefunction(fn):
+                return await fn(data, state, collect_custom_statistics_fn)
+            return fn(data, state, collect_custom_statistics_fn)
+        return data
+
+    async def process(self, data, state, collect_custom_statistics_fn=None):
+        """Non-OpenAI route: {"prompt": ..., "max_tokens": ..} -> completion."""
+        return await self._engine.generate_simple(data)
+
+    # ---- OpenAI serve types (route /serve/openai/v1/...) -------------- #
+    async def v1_chat_completions(self, body, state, collect_fn=None):
+        return await self._engine.openai_chat_completions(body, self._served_name)
+
+    async def v1_completions(self, body, state, collect_fn=None):
+        return await self._engine.openai_completions(body, self._served_name)
+
+    async def v1_models(self, body, state, collect_fn=None):
+        return self._engine.openai_models(self._served_name)
+
+
+# reference-CLI compatibility: `--engine vllm` runs the native LLM engine
+BasePreprocessRequest.register_engine("vllm")(LlmPreprocessRequest)

@@ -1,0 +1,145 @@
+"""In-process HIP DL engine: the native replacement for the Triton sidecar.
+
+Where the reference serializes numpy tensors into gRPC ``ModelInferRequest``
+messages to a tritonserver container (reference: preprocess_service.py:
+267-446) and configures batching via config.pbtxt pass-through
+(triton_helper.py:291-409), this engine runs the model in the serving
+process: ``process()`` awaits the per-endpoint dynamic batcher, which
+coalesces concurrent requests into bucketed bf16 batches, executes on a
+dedicated HIP stream (hipGraph-captured per bucket) and scatters outputs
+back to the awaiting requests.
+
+Model resolution, per endpoint:
+1. user ``Preprocess.load()`` returned a callable -> use it directly
+2. model file is a JSON model card -> built from the native model library
+   (clearml_serving_amd.models: resnet50, bert-base, ...)
+3. model file is TorchScript (.pt/.pth/.ts) -> torch.jit.load
+
+``auxiliary_cfg`` knobs (the native equivalent of the reference's
+``--aux-config`` pbtxt pass-through):
+    max_batch_size (default 64), max_queue_delay_us (default 2000),
+    preferred_batch_size (list), dtype ("bfloat16"|"float16"|"float32"),
+    use_graphs (default true), gpu (device index), input_format ("dict")
+"""
+
+import json
+import os
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from ..schemas import ModelEndpoint
+from ..serving.batcher import DEFAULT_BUCKETS, DynamicBatcher
+from ..serving.preprocess import BasePreprocessRequest
+
+_DTYPES = {
+    "bfloat16": torch.bfloat16, "bf16": torch.bfloat16,
+    "float16": torch.float16, "fp16": torch.float16,
+    "float32": torch.float32, "fp32": torch.float32,
+}
+
+
+def _pick_device(aux: Dict[str, Any]) -> torch.device:
+    if torch.cuda.is_available():
+        idx = int(aux.get("gpu", 0))
+        return torch.device("cuda", idx)
+    return torch.device("cpu")
+
+
+@BasePreprocessRequest.register_engine("hip", modules=["torch"])
+class HipPreprocessRequest(BasePreprocessRequest):
+    is_preprocess_async = False
+    is_process_async = True      # process() awaits the dynamic batcher
+    is_postprocess_async = False
+
+    def __init__(self, model_endpoint: ModelEndpoint, task=None):
+        super().__init__(model_endpoint, task)
+        aux = dict(model_endpoint.auxiliary_cfg or {})
+        self.device = _pick_device(aux)
+        self.dtype = _DTYPES.get(str(aux.get("dtype", "bfloat16")).lower(),
+                                 torch.bfloat16)
+        if self.device.type == "cpu" and self.dtype is not torch.float32:
+            # CPU path (tests / no-GPU dev): fp32 keeps MIOpen-free numerics
+            self.dtype = torch.float32
+
+        model = self._model if callable(self._model) else None
+        if model is None:
+            model = self._load_model()
+        self._torch_model = model
+
+        buckets = aux.get("preferred_batch_size") or list(DEFAULT_BUCKETS)
+        self._batcher = DynamicBatcher(
+            model_fn=self._forward,
+            device=self.device,
+            max_batch_size=int(aux.get("max_batch_size", 64)),
+            max_queue_delay_us=int(aux.get("max_queue_delay_us", 2000)),
+            preferred_batch_sizes=buckets,
+            use_graphs=bool(aux.get("use_graphs", True)),
+            dtype=self.dtype,
+            name=model_endpoint.serving_url,
+        )
+
+    # ------------------------------------------------------------------ #
+    def _load_model(self):
+        local = self._get_local_model_file()
+        if not local:
+            raise ValueError(
+                "hip endpoint '{}' has no model (register a model card or "
+                "TorchScript file)".format(self.model_endpoint.serving_url))
+        if os.path.isdir(local):
+            for name in ("model_card.json", "card.json"):
+                p = os.path.join(local, name)
+                if os.path.exists(p):
+                    local = p
+                    break
+        if local.endswith(".json"):
+            from .. import models
+
+            model = models.build_model(local, device=str(self.device))
+            if self.device.type == "cuda":
+                model = model.to(self.dtype)
+            else:
+                model = model.float()
+            return model
+        model = torch.jit.load(local, map_location=str(self.device))
+        model.eval()
+        if self.device.type == "cuda":
+            model = model.to(self.dtype)
+        return model
+
+    def _forward(self, batch):
+        with torch.inference_mode():
+            out = self._torch_model(batch)
+        return out
+
+    # ------------------------------------------------------------------ #
+    @staticmethod
+    def _to_tensor(value) -> torch.Tensor:
+        if isinstance(value, torch.Tensor):
+            return value
+        arr = np.asarray(value)
+        if arr.dtype == np.float64:
+            arr = arr.astype(np.float32)
+        return torch.from_numpy(arr)
+
+    async def process(self, data: Any, state: dict,
+                      collect_custom_statistics_fn=None) -> Any:
+        """data: one request's input -- np array/list, or {name: array}."""
+        if isinstance(data, dict):
+            inputs = {k: self._to_tensor(v) for k, v in data.items()}
+        else:
+            inputs = self._to_tensor(data)
+        out = await self._batcher.submit(inputs)
+        if isinstance(out, dict):
+            return {k: v.float().numpy() if v.is_floating_point() else v.numpy()
+                    for k, v in out.items()}
+        if isinstance(out, (tuple, list)):
+            return [v.float().numpy() for v in out]
+        return out.float().numpy() if out.is_floating_point() else out.numpy()
+
+
+# alias for reference-CLI compatibility: `model add --engine triton` serves
+# through the in-process HIP engine (there is no Triton sidecar to talk to)
+BasePreprocessRequest.register_engine("triton")(HipPreprocessRequest)
+BasePreprocessRequest.register_engine("pytorch")(HipPreprocessRequest)

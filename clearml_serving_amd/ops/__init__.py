@@ -1,0 +1,327 @@
+"""HIP/CDNA4 kernel library: python dispatch layer.
+
+The compute tier the reference outsources to Triton/vLLM
+lives here as hand-written gfx950 kernels (``csrc/*.hip``), compiled in-tree
+to ``_hip_ops.so`` by ``__graft_entry__.build()`` (hipcc --offload-arch=gfx950
+via torch.utils.cpp_extension).
+
+Dispatch policy:
+- tensor on GPU  -> the HIP extension is REQUIRED; a missing .so raises
+  instead of silently falling back to eager PyTorch (so a GPU run can never
+  "pass" on the fallback path).
+- tensor on CPU  -> plain PyTorch reference implementations (used by the CPU
+  test suite and as the fp32 numerics reference for the GPU kernels).
+
+Set CLEARML_SERVING_AMD_FORCE_EAGER=1 to bypass the extension on GPU
+(debug/ablation only).
+"""
+
+import importlib.machinery
+import importlib.util
+import math
+import os
+from typing import Optional, Tuple
+
+import torch
+
+_ext = None
+_ext_error: Optional[str] = None
+_FORCE_EAGER = os.environ.get("CLEARML_SERVING_AMD_FORCE_EAGER") == "1"
+
+
+def _try_load_ext() -> None:
+    global _ext, _ext_error
+    if _ext is not None or _ext_error is not None:
+        return
+    so_path = os.path.join(os.path.dirname(__file__), "_hip_ops.so")
+    if not os.path.exists(so_path):
+        _ext_error = "extension not built ({} missing)".format(so_path)
+        return
+    try:
+        loader = importlib.machinery.ExtensionFileLoader("_hip_ops", so_path)
+        spec = importlib.util.spec_from_loader("_hip_ops", loader)
+        mod = importlib.util.module_from_spec(spec)
+        loader.exec_module(mod)
+        _ext = mod
+    except Exception as ex:  # loud, but deferred to first GPU use
+        _ext_error = "failed loading {}: {}".format(so_path, ex)
+
+
+_try_load_ext()
+
+
+def has_extension() -> bool:
+    return _ext is not None
+
+
+def _require_ext(opname: str):
+    if _FORCE_EAGER:
+        return None
+    if _ext is None:
+        raise RuntimeError(
+            "clearml_serving_amd HIP extension required for GPU op '{}' but "
+            "not loaded: {}. Build it with __graft_entry__.build() "
+            "(hipcc --offload-arch=gfx950).".format(opname, _ext_error)
+        )
+    return _ext
+
+
+def extension_or_none():
+    return _ext
+
+
+# --------------------------------------------------------------------- #
+# LayerNorm (+ fused residual add)
+# --------------------------------------------------------------------- #
+def layernorm(
+    x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+    eps: float = 1e-5, residual: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """out = LN(x + residual) over the last dim. bf16/fp32 in, same dtype out."""
+    if x.is_cuda:
+        ext = _require_ext("layernorm")
+        if ext is not None:
+            return ext.layernorm(x, weight, bias, eps, residual)
+    if residual is not None:
+        x = x + residual
+    return torch.nn.functional.layer_norm(
+        x.float(), (x.shape[-1],), weight.float(), bias.float(), eps
+    ).to(x.dtype)
+
+
+def rmsnorm(
+    x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6,
+    residual: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """out = x / rms(x) * weight (llama-style), optional fused residual add.
+
+    When ``residual`` is given the op ALSO writes x+residual back into
+    ``residual`` in-place on the GPU path (the running residual stream),
+    matching the CPU reference below.
+    """
+    if x.is_cuda:
+        ext = _require_ext("rmsnorm")
+        if ext is not None:
+            return ext.rmsnorm(x, weight, eps, residual)
+    xf = x.float()
+    if residual is not None:
+        xf = xf + residual.float()
+        residual.copy_(xf.to(residual.dtype))
+    out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (out * weight.float()).to(x.dtype)
+
+
+# --------------------------------------------------------------------- #
+# Activations / epilogues
+# --------------------------------------------------------------------- #
+def bias_gelu(x: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = gelu(x + bias), erf formulation (BERT)."""
+    if x.is_cuda:
+        ext = _require_ext("bias_gelu")
+        if ext is not None:
+            return ext.bias_gelu(x, bias)
+    if bias is not None:
+        x = x + bias
+    return torch.nn.functional.gelu(x.float()).to(x.dtype)
+
+
+def bias_relu_add(
+    x: torch.Tensor, bias: Optional[torch.Tensor] = None,
+    residual: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """out = relu(x + bias + residual) -- ResNet bottleneck epilogue."""
+    if x.is_cuda:
+        ext = _require_ext("bias_relu_add")
+        if ext is not None:
+            return ext.bias_relu_add(x, bias, residual)
+    if bias is not None:
+        x = x + bias
+    if residual is not None:
+        x = x + residual
+    return torch.relu(x)
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    """out = silu(gate) * up -- llama MLP."""
+    if gate.is_cuda:
+        ext = _require_ext("silu_mul")
+        if ext is not None:
+            return ext.silu_mul(gate, up)
+    return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate.dtype)
+
+
+def softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    """Numerically-stable softmax over the last dim."""
+    if x.is_cuda and dim in (-1, x.dim() - 1):
+        ext = _require_ext("softmax")
+        if ext is not None:
+            return ext.softmax_lastdim(x)
+    return torch.softmax(x.float(), dim=dim).to(x.dtype)
+
+
+# --------------------------------------------------------------------- #
+# Attention
+# --------------------------------------------------------------------- #
+def attention(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    causal: bool = False, scale: Optional[float] = None,
+    seq_lens: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Fused scaled-dot-product attention (prefill / encoder).
+
+    q,k,v: [B, H, S, D] (same S for q and k/v here; GQA via H_kv divides H).
+    ``seq_lens`` (int32 [B]) masks keys >= len (padding). Output [B, H, S, D].
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attention")
+        if ext is not None:
+            return ext.attention_prefill(q, k, v, bool(causal), float(scale),
+                                         seq_lens)
+    # reference path (fp32 math)
+    qf, kf, vf = q.float(), k.float(), v.float()
+    hq, hkv = q.shape[1], k.shape[1]
+    if hkv != hq:
+        rep = hq // hkv
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    sq, sk = q.shape[2], k.shape[2]
+    if causal:
+        mask = torch.ones(sq, sk, dtype=torch.bool, device=q.device).tril(
+            diagonal=sk - sq)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    if seq_lens is not None:
+        key_idx = torch.arange(sk, device=q.device)
+        pad = key_idx[None, None, None, :] >= seq_lens[:, None, None, None]
+        scores = scores.masked_fill(pad, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    return torch.matmul(probs, vf).to(q.dtype)
+
+
+def attention_decode(
+    q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+    block_table: torch.Tensor, seq_lens: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Paged-KV decode attention: one new token per sequence.
+
+    q:        [B, H, D]         (current-step queries)
+    k_cache:  [num_blocks, H_kv, block_size, D]
+    v_cache:  [num_blocks, H_kv, block_size, D]
+    block_table: int32 [B, max_blocks] physical block ids per sequence
+    seq_lens: int32 [B] total keys per sequence (including current token)
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attention_decode")
+        if ext is not None:
+            return ext.attention_decode(q, k_cache, v_cache, block_table,
+                                        seq_lens, float(scale))
+    # reference path: gather pages then dense attention per sequence
+    bsz, hq, d = q.shape
+    hkv = k_cache.shape[1]
+    block_size = k_cache.shape[2]
+    rep = hq // hkv
+    out = torch.empty_like(q, dtype=q.dtype)
+    qf = q.float()
+    for b in range(bsz):
+        n = int(seq_lens[b])
+        nblocks = (n + block_size - 1) // block_size
+        blocks = block_table[b, :nblocks].long()
+        k = k_cache[blocks].float()  # [nb, hkv, bs, d]
+        v = v_cache[blocks].float()
+        k = k.permute(1, 0, 2, 3).reshape(hkv, nblocks * block_size, d)[:, :n]
+        v = v.permute(1, 0, 2, 3).reshape(hkv, nblocks * block_size, d)[:, :n]
+        if rep > 1:
+            k = k.repeat_interleave(rep, dim=0)
+            v = v.repeat_interleave(rep, dim=0)
+        scores = torch.einsum("hd,hnd->hn", qf[b], k) * scale
+        probs = torch.softmax(scores, dim=-1)
+        out[b] = torch.einsum("hn,hnd->hd", probs, v).to(q.dtype)
+    return out
+
+
+# --------------------------------------------------------------------- #
+# Rotary embedding
+# --------------------------------------------------------------------- #
+def rope_inplace(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+    theta: float = 10000.0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Apply rotary position embedding in place (NeoX interleaving: rotate
+    pairs (i, i+D/2)). q: [T, H, D], k: [T, H_kv, D], positions: int32 [T]."""
+    if q.is_cuda:
+        ext = _require_ext("rope")
+        if ext is not None:
+            ext.rope_inplace(q, k, positions, float(theta))
+            return q, k
+    d = q.shape[-1]
+    half = d // 2
+    inv_freq = 1.0 / (theta ** (torch.arange(half, dtype=torch.float32,
+                                             device=q.device) / half))
+    angles = positions.float()[:, None] * inv_freq[None, :]  # [T, half]
+    cos = angles.cos()[:, None, :]
+    sin = angles.sin()[:, None, :]
+    for t in (q, k):
+        tf = t.float()
+        lo, hi = tf[..., :half], tf[..., half:]
+        t[..., :half] = (lo * cos - hi * sin).to(t.dtype)
+        t[..., half:] = (hi * cos + lo * sin).to(t.dtype)
+    return q, k
+
+
+# --------------------------------------------------------------------- #
+# Sampling
+# --------------------------------------------------------------------- #
+def sample_top_k_top_p(
+    logits: torch.Tensor, temperature: float = 1.0, top_k: int = 0,
+    top_p: float = 1.0, generator: Optional[torch.Generator] = None,
+) -> torch.Tensor:
+    """Sample token ids from [B, V] logits with temperature/top-k/top-p.
+
+    Greedy when temperature == 0. Returns int64 [B].
+    """
+    if temperature == 0.0:
+        return logits.argmax(dim=-1)
+    if logits.is_cuda:
+        ext = _require_ext("sample")
+        if ext is not None:
+            seed = int(torch.randint(0, 2**31 - 1, (1,),
+                                     generator=generator).item())
+            return ext.sample_top_k_top_p(
+                logits, float(temperature), int(top_k), float(top_p), seed)
+    lf = logits.float() / temperature
+    if top_k and top_k < lf.shape[-1]:
+        kth = torch.topk(lf, top_k, dim=-1).values[..., -1:]
+        lf = lf.masked_fill(lf < kth, float("-inf"))
+    if top_p < 1.0:
+        sorted_logits, idx = torch.sort(lf, descending=True, dim=-1)
+        probs = torch.softmax(sorted_logits, dim=-1)
+        # drop tokens whose inclusive cumulative prob exceeds p; always keep
+        # the top token
+        cutoff = probs.cumsum(dim=-1) > top_p
+        cutoff[..., 0] = False
+        sorted_logits = sorted_logits.masked_fill(cutoff, float("-inf"))
+        lf = torch.full_like(lf, float("-inf")).scatter(-1, idx, sorted_logits)
+    probs = torch.softmax(lf, dim=-1)
+    return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+
+# --------------------------------------------------------------------- #
+# GEMM (bf16 MFMA)
+# --------------------------------------------------------------------- #
+def gemm_bf16(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ B[K,N] in bf16 with fp32 accumulation on MFMA.
+
+    Showcase/hot-path GEMM; plain library GEMMs go through torch.matmul
+    (hipBLASLt) -- this kernel exists for fused epilogues and to anchor the
+    MFMA performance path.
+    """
+    if a.is_cuda:
+        ext = _require_ext("gemm_bf16")
+        if ext is not None:
+            return ext.gemm_bf16(a, b)
+    return (a.float() @ b.float()).to(a.dtype)

@@ -1,0 +1,292 @@
+// Fused multi-head attention (prefill / encoder) for gfx950.
+//
+// One kernel: S = QK^T -> online softmax -> O = P V, never materializing
+// S/P in HBM (the reference's Triton/vLLM engines own this step; SURVEY.md
+// §2.6). Flash-attention-2 style tiling on MFMA:
+//
+//   grid  = (ceil(Sq/64), B*H);  block = 4 waves (256 threads)
+//   each wave owns 16 query rows; the workgroup shares K/V tiles of 64 keys
+//   staged in LDS (K row-major, V transposed at stage time so PV B-fragments
+//   are contiguous ds_read_b128).
+//
+// MFMA: v_mfma_f32_16x16x32_bf16.  Fragment maps (cdna4 §3):
+//   A[i][k]: lane l holds i = l%16, k = (l/16)*8 + e   (e = 0..7)
+//   B[k][j]: lane l holds j = l%16, k = (l/16)*8 + e
+//   C/D    : lane l holds col = l%16, row = (l/16)*4 + reg
+//
+// LDS rows are padded by 8 bf16 (stride 144 B): the 16-lane groups of
+// ds_read_b128 then hit 16 distinct 16-B slots (bank-conflict-free; the
+// unpadded 128 B stride is 8-way).
+//
+// Online softmax is the standard running (m, l) update; row statistics
+// reduce over the 16 lanes of each C-layout row group via shfl_xor 1/2/4/8.
+//
+// Supports: head_dim 64/128, GQA (H_kv | H), causal masking, per-batch
+// kv sequence lengths (padding masks). bf16 in/out, fp32 accumulate.
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_t;   // 4 VGPRs
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+#define MFMA_16x16x32(A, B, C) \
+  __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
+
+constexpr int BLOCK_M = 64;   // query rows per workgroup
+constexpr int BLOCK_N = 64;   // keys per tile
+constexpr int NWAVES = 4;     // one 16-row M-slice per wave
+constexpr int PAD = 8;        // bf16 elements of row padding (16 B)
+
+// row-group reduction: combine over the 16 lanes holding one C-layout row
+__device__ __forceinline__ float rowgroup_max(float v) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+__device__ __forceinline__ float rowgroup_sum(float v) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS>
+__global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
+    const __hip_bfloat16* __restrict__ q,   // [B, H, Sq, D]
+    const __hip_bfloat16* __restrict__ k,   // [B, Hkv, Sk, D]
+    const __hip_bfloat16* __restrict__ v,   // [B, Hkv, Sk, D]
+    __hip_bfloat16* __restrict__ out,       // [B, H, Sq, D]
+    const int* __restrict__ seq_lens,       // [B] or null
+    int B, int H, int Hkv, int Sq, int Sk, float scale) {
+  constexpr int D = HEAD_DIM;
+  constexpr int KSTRIDE = D + PAD;          // LDS K row stride (bf16)
+  constexpr int VSTRIDE = BLOCK_N + PAD;    // LDS V^T row stride
+  constexpr int PSTRIDE = BLOCK_N + PAD;
+
+  __shared__ short lds_k[BLOCK_N * KSTRIDE];
+  __shared__ short lds_vt[D * VSTRIDE];
+  __shared__ short lds_p[NWAVES * 16 * PSTRIDE];
+
+  const int m_tile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int hkv = h / (H / Hkv);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const int m0 = m_tile * BLOCK_M;          // first query row of the block
+  const int wm0 = m0 + wave * 16;           // this wave's first query row
+
+  const long q_base = (((long)b * H + h) * Sq) * D;
+  const long kv_base = (((long)b * Hkv + hkv) * Sk) * D;
+  const int kv_len = HAS_SEQLENS ? min(seq_lens[b], Sk) : Sk;
+  // causal: queries attend to keys <= q_idx (Sq == Sk alignment)
+  const int kv_hi = CAUSAL ? min(kv_len, m0 + BLOCK_M + (Sk - Sq)) : kv_len;
+
+  // ---- load this wave's Q fragments (scaled once; fp32->bf16 later in S) --
+  const int frag_row = lane & 15;           // i
+  const int frag_ko = (lane >> 4) * 8;      // k offset
+  bf16x8_t q_frag[D / 32];
+#pragma unroll
+  for (int kc = 0; kc < D / 32; ++kc) {
+    const int qrow = wm0 + frag_row;
+    if (qrow < Sq) {
+      const __hip_bfloat16* src = q + q_base + (long)qrow * D + kc * 32 + frag_ko;
+      q_frag[kc] = *reinterpret_cast<const bf16x8_t*>(src);
+    } else {
+      q_frag[kc] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  // ---- running softmax state: 4 rows per lane (reg r -> row base+r) ------ //
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+  f32x4_t acc_o[D / 16];
+#pragma unroll
+  for (int dt = 0; dt < D / 16; ++dt) acc_o[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  // ================= KV tile loop ================= //
+  for (int n0 = 0; n0 < kv_hi; n0 += BLOCK_N) {
+    // ---- stage K [64][D] row-major and V^T [D][64] into LDS ---- //
+    // 256 threads; each loads 16-B pieces. K: row = key, col = d.
+    {
+      constexpr int pieces = BLOCK_N * D / 8;       // 8 bf16 per piece
+      for (int p = tid; p < pieces; p += 256) {
+        const int key = p / (D / 8);
+        const int d8 = (p % (D / 8)) * 8;
+        const int gkey = n0 + key;
+        bf16x8_t val{0, 0, 0, 0, 0, 0, 0, 0};
+        if (gkey < kv_len) {
+          val = *reinterpret_cast<const bf16x8_t*>(
+              k + kv_base + (long)gkey * D + d8);
+        }
+        *reinterpret_cast<bf16x8_t*>(&lds_k[key * KSTRIDE + d8]) = val;
+        // V: same piece indexing, scatter-transposed (8 ds_write_b16)
+        bf16x8_t vv{0, 0, 0, 0, 0, 0, 0, 0};
+        if (gkey < kv_len) {
+          vv = *reinterpret_cast<const bf16x8_t*>(
+              v + kv_base + (long)gkey * D + d8);
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e) lds_vt[(d8 + e) * VSTRIDE + key] = vv[e];
+      }
+    }
+    __syncthreads();
+
+    // ---- S = Q K^T for this wave's 16 rows x 64 keys ---- //
+    f32x4_t acc_s[BLOCK_N / 16];
+#pragma unroll
+    for (int t = 0; t < BLOCK_N / 16; ++t) acc_s[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kc = 0; kc < D / 32; ++kc) {
+#pragma unroll
+      for (int t = 0; t < BLOCK_N / 16; ++t) {
+        // B[k][j] = K[j + 16t][k]: contiguous 8 bf16 at row j, col k-offset
+        const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+            &lds_k[(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
+        acc_s[t] = MFMA_16x16x32(q_frag[kc], kf, acc_s[t]);
+      }
+    }
+
+    // ---- mask + online softmax ---- //
+    const int row_base = wm0 + (lane >> 4) * 4;  // C layout: row = base + reg
+    const int col_base = n0 + (lane & 15);       // col = base + 16t
+    float pvals[4][BLOCK_N / 16];
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = row_base + r;
+      float rmax = -INFINITY;
+#pragma unroll
+      for (int t = 0; t < BLOCK_N / 16; ++t) {
+        float s = acc_s[t][r] * scale;
+        const int key = col_base + t * 16;
+        bool valid = key < kv_len;
+        if (CAUSAL) valid = valid && (key <= qrow + (Sk - Sq));
+        s = valid ? s : -INFINITY;
+        pvals[r][t] = s;
+        rmax = fmaxf(rmax, s);
+      }
+      rmax = rowgroup_max(rmax);
+      const float m_new = fmaxf(m_run[r], rmax);
+      // all-masked rows keep m = -inf; exp() below yields 0 contributions
+      alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+      float rsum = 0.f;
+#pragma unroll
+      for (int t = 0; t < BLOCK_N / 16; ++t) {
+        const float p = (pvals[r][t] == -INFINITY || m_new == -INFINITY)
+                            ? 0.f
+                            : __expf(pvals[r][t] - m_new);
+        pvals[r][t] = p;
+        rsum += p;
+      }
+      rsum = rowgroup_sum(rsum);
+      l_run[r] = l_run[r] * alpha[r] + rsum;
+      m_run[r] = m_new;
+    }
+    // rescale O by alpha (per C-layout row)
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[dt][r] *= alpha[r];
+    }
+
+    // ---- write P (bf16) to this wave's LDS slab, re-read as A fragments --
+    short* pslab = &lds_p[wave * 16 * PSTRIDE];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int t = 0; t < BLOCK_N / 16; ++t) {
+        __hip_bfloat16 pb = __float2bfloat16(pvals[r][t]);
+        pslab[prow * PSTRIDE + (lane & 15) + t * 16] =
+            *reinterpret_cast<short*>(&pb);
+      }
+    }
+    // cross-LANE hand-off through LDS within the wave: the compiler's
+    // per-thread alias analysis cannot see it, so drain the wave's ds queue
+    // before any lane reads another lane's P ("memory" orders the compiler's
+    // ds ops around the asm; the MFMA consumes the loads by data dependency)
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V ---- //
+#pragma unroll
+    for (int kc = 0; kc < BLOCK_N / 32; ++kc) {
+      const bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
+          &pslab[frag_row * PSTRIDE + kc * 32 + frag_ko]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        // B[k][j] = V[k + 32kc][j + 16dt] = vt[j + 16dt][k]: contiguous in k
+        const bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            &lds_vt[(dt * 16 + frag_row) * VSTRIDE + kc * 32 + frag_ko]);
+        acc_o[dt] = MFMA_16x16x32(pf, vf, acc_o[dt]);
+      }
+    }
+    __syncthreads();  // K/V tiles reused next iteration
+  }
+
+  // ---- epilogue: O / l, bf16 store ---- //
+  const int row_base = wm0 + (lane >> 4) * 4;
+  const int col = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = row_base + r;
+    if (qrow >= Sq) continue;
+    const float inv_l = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
+    __hip_bfloat16* dst = out + q_base + (long)qrow * D;
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+      dst[dt * 16 + col] = __float2bfloat16(acc_o[dt][r] * inv_l);
+    }
+  }
+}
+
+}  // namespace
+
+torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, bool causal, double scale,
+                                c10::optional<torch::Tensor> seq_lens) {
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4,
+              "q/k/v must be [B, H, S, D]");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1), Sk = k.size(2);
+  TORCH_CHECK(H % Hkv == 0, "H must be a multiple of H_kv");
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  TORCH_CHECK(v.size(1) == Hkv && v.size(2) == Sk && v.size(3) == D);
+
+  auto out = torch::empty_like(q);
+  const int* sl = nullptr;
+  torch::Tensor sl_t;
+  if (seq_lens.has_value() && seq_lens->defined()) {
+    sl_t = seq_lens->to(q.device(), at::kInt).contiguous();
+    TORCH_CHECK(sl_t.numel() == B, "seq_lens must be [B]");
+    sl = sl_t.data_ptr<int>();
+  }
+  dim3 grid((Sq + BLOCK_M - 1) / BLOCK_M, B * H);
+  dim3 block(256);
+  hipStream_t stream_ = cmls::current_stream();
+
+#define LAUNCH_ATTN(DD, CC, SS)                                              \
+  hipLaunchKernelGGL((attn_prefill_kernel<DD, CC, SS>), grid, block, 0,      \
+                     stream_,                                        \
+                     (const __hip_bfloat16*)q.data_ptr(),                    \
+                     (const __hip_bfloat16*)k.data_ptr(),                    \
+                     (const __hip_bfloat16*)v.data_ptr(),                    \
+                     (__hip_bfloat16*)out.data_ptr(), sl, B, H, Hkv, Sq, Sk, \
+                     (float)scale)
+  if (D == 64) {
+    if (causal) { if (sl) LAUNCH_ATTN(64, true, true); else LAUNCH_ATTN(64, true, false); }
+    else        { if (sl) LAUNCH_ATTN(64, false, true); else LAUNCH_ATTN(64, false, false); }
+  } else {
+    if (causal) { if (sl) LAUNCH_ATTN(128, true, true); else LAUNCH_ATTN(128, true, false); }
+    else        { if (sl) LAUNCH_ATTN(128, false, true); else LAUNCH_ATTN(128, false, false); }
+  }
+#undef LAUNCH_ATTN
+  return out;
+}

@@ -1,0 +1,39 @@
+// pybind11 bindings for the gfx950 kernel library (_hip_ops.so).
+#include <torch/extension.h>
+
+torch::Tensor bias_relu_add(torch::Tensor x, c10::optional<torch::Tensor> bias,
+                            c10::optional<torch::Tensor> residual);
+torch::Tensor bias_gelu(torch::Tensor x, c10::optional<torch::Tensor> bias);
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up);
+torch::Tensor layernorm(torch::Tensor x, torch::Tensor weight,
+                        torch::Tensor bias, double eps,
+                        c10::optional<torch::Tensor> residual);
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor weight, double eps,
+                      c10::optional<torch::Tensor> residual);
+torch::Tensor softmax_lastdim(torch::Tensor x);
+torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, bool causal, double scale,
+                                c10::optional<torch::Tensor> seq_lens);
+torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
+                                 long top_k, double top_p, long seed);
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  double theta);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "clearml-serving-amd gfx950 kernel library";
+  m.def("bias_relu_add", &bias_relu_add, py::arg("x"),
+        py::arg("bias") = py::none(), py::arg("residual") = py::none());
+  m.def("bias_gelu", &bias_gelu, py::arg("x"), py::arg("bias") = py::none());
+  m.def("silu_mul", &silu_mul);
+  m.def("layernorm", &layernorm, py::arg("x"), py::arg("weight"),
+        py::arg("bias"), py::arg("eps") = 1e-5,
+        py::arg("residual") = py::none());
+  m.def("rmsnorm", &rmsnorm, py::arg("x"), py::arg("weight"),
+        py::arg("eps") = 1e-6, py::arg("residual") = py::none());
+  m.def("softmax_lastdim", &softmax_lastdim);
+  m.def("attention_prefill", &attention_prefill, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("causal"), py::arg("scale"),
+        py::arg("seq_lens") = py::none());
+  m.def("sample_top_k_top_p", &sample_top_k_top_p);
+  m.def("rope_inplace", &rope_inplace);
+}

@@ -18,6 +18,13 @@ def _validate_engine(value: str) -> None:
     from .serving.preprocess import BasePreprocessRequest
 
     if not BasePreprocessRequest.validate_engine_type(value):
+        # GPU engines register from the engines package; import it lazily so
+        # plain schema use does not pull in torch
+        try:
+            from . import engines  # noqa: F401
+        except Exception:
+            pass
+    if not BasePreprocessRequest.validate_engine_type(value):
         raise TypeError("{} not supported engine type".format(value))
 
 

@@ -30,6 +30,9 @@ import torch
 
 DEFAULT_BUCKETS = (1, 2, 4, 8, 16, 32, 64)
 
+# hipGraph capture is process-global state: serialize captures across batchers
+_capture_lock = threading.Lock()
+
 TensorOrDict = Union[torch.Tensor, Dict[str, torch.Tensor]]
 
 
@@ -161,7 +164,8 @@ class DynamicBatcher:
 
         entry = self._graphs.get(bucket)
         if entry is None:
-            entry = self._capture(stacked, bucket)
+            with _capture_lock:
+                entry = self._capture(stacked, bucket)
             self._graphs[bucket] = entry
         # copy inputs into the graph's static buffers, replay, read outputs
         _copy_into(entry["in"], stacked)

@@ -1,0 +1,189 @@
+"""Native model library + hip engine + dynamic batcher (CPU execution path)."""
+
+import asyncio
+import json
+
+import numpy as np
+import pytest
+import torch
+
+from clearml_serving_amd.models import build_model, list_archs
+from clearml_serving_amd.schemas import ModelEndpoint
+from clearml_serving_amd.serving.batcher import DynamicBatcher
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_arch_registry():
+    for a in ("resnet50", "bert-base"):
+        assert a in list_archs()
+
+
+def test_resnet50_builds_and_runs():
+    m = build_model({"arch": "resnet50", "num_classes": 10, "dtype": "float32"})
+    x = torch.randn(2, 3, 64, 64)
+    with torch.inference_mode():
+        y = m(x)
+    assert y.shape == (2, 10)
+    assert torch.isfinite(y).all()
+
+
+def test_bert_base_builds_and_runs():
+    m = build_model({"arch": "bert-base", "num_labels": 3, "dtype": "float32",
+                     "vocab_size": 1000})
+    ids = torch.randint(0, 1000, (2, 16))
+    mask = torch.ones(2, 16, dtype=torch.int32)
+    with torch.inference_mode():
+        y = m({"input_ids": ids, "attention_mask": mask})
+    assert y.shape == (2, 3)
+    assert torch.isfinite(y).all()
+
+
+def test_bert_padding_mask_matters():
+    torch.manual_seed(0)
+    m = build_model({"arch": "bert-base", "num_labels": 2, "dtype": "float32",
+                     "vocab_size": 100})
+    ids = torch.randint(0, 100, (1, 16))
+    full = torch.ones(1, 16, dtype=torch.int32)
+    half = torch.cat([torch.ones(1, 8, dtype=torch.int32),
+                      torch.zeros(1, 8, dtype=torch.int32)], dim=1)
+    with torch.inference_mode():
+        y_full = m({"input_ids": ids, "attention_mask": full})
+        y_half = m({"input_ids": ids, "attention_mask": half})
+        # padding-masked forward must equal the truncated forward
+        y_trunc = m({"input_ids": ids[:, :8],
+                     "attention_mask": torch.ones(1, 8, dtype=torch.int32)})
+    assert not torch.allclose(y_full, y_half, atol=1e-4)
+    torch.testing.assert_close(y_half, y_trunc, atol=1e-4, rtol=1e-4)
+
+
+# ------------------------------------------------------------------ #
+# dynamic batcher
+# ------------------------------------------------------------------ #
+def test_batcher_coalesces_and_slices():
+    calls = []
+
+    def model_fn(x):
+        calls.append(x.shape[0])
+        return x * 2.0
+
+    batcher = DynamicBatcher(model_fn, device="cpu", max_batch_size=8,
+                             max_queue_delay_us=50_000, use_graphs=False)
+
+    async def main():
+        outs = await asyncio.gather(
+            *[batcher.submit(torch.full((3,), float(i))) for i in range(6)])
+        return outs
+
+    outs = run(main())
+    for i, o in enumerate(outs):
+        torch.testing.assert_close(o, torch.full((3,), 2.0 * i))
+    # everything coalesced into few batches, padded to a bucket
+    assert sum(calls) >= 6
+    assert len(calls) <= 3
+
+
+def test_batcher_bucket_padding():
+    shapes = []
+
+    def model_fn(x):
+        shapes.append(tuple(x.shape))
+        return x
+
+    batcher = DynamicBatcher(model_fn, device="cpu", max_batch_size=16,
+                             max_queue_delay_us=30_000, use_graphs=False)
+
+    async def main():
+        return await asyncio.gather(
+            *[batcher.submit(torch.zeros(2)) for _ in range(3)])
+
+    run(main())
+    # 3 requests pad up to bucket size 4
+    assert all(s[0] in (1, 2, 4) for s in shapes)
+
+
+def test_batcher_dict_inputs():
+    def model_fn(d):
+        return d["a"] + d["b"]
+
+    batcher = DynamicBatcher(model_fn, device="cpu", max_batch_size=4,
+                             max_queue_delay_us=10_000, use_graphs=False)
+
+    async def main():
+        return await batcher.submit(
+            {"a": torch.ones(2), "b": torch.full((2,), 3.0)})
+
+    out = run(main())
+    torch.testing.assert_close(out, torch.full((2,), 4.0))
+
+
+def test_batcher_error_propagates():
+    def model_fn(x):
+        raise RuntimeError("boom")
+
+    batcher = DynamicBatcher(model_fn, device="cpu", use_graphs=False,
+                             max_queue_delay_us=1000)
+
+    async def main():
+        with pytest.raises(RuntimeError, match="boom"):
+            await batcher.submit(torch.zeros(1))
+
+    run(main())
+
+
+# ------------------------------------------------------------------ #
+# hip engine through the full serving path (CPU device)
+# ------------------------------------------------------------------ #
+@pytest.fixture()
+def model_card_registered(store, tmp_path):
+    card = tmp_path / "model_card.json"
+    card.write_text(json.dumps({
+        "arch": "bert-base", "num_labels": 2, "dtype": "float32",
+        "vocab_size": 500,
+    }))
+    return store.register_model(name="bert-tiny-card", project="p",
+                                path=str(card))
+
+
+def test_hip_engine_serves_model_card(processor, store, model_card_registered):
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="bert_ep",
+        model_id=model_card_registered.model_id,
+        auxiliary_cfg={"max_queue_delay_us": 1000, "use_graphs": False},
+    ))
+    body = {
+        "input_ids": np.random.randint(0, 500, (16,)).tolist(),
+        "attention_mask": [1] * 16,
+    }
+    out = run(processor.process_request("bert_ep", "", body))
+    assert np.asarray(out).shape == (2,)
+
+
+def test_triton_alias_serves_same_engine(processor, store, model_card_registered):
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="triton", serving_url="bert_ep2",
+        model_id=model_card_registered.model_id,
+        auxiliary_cfg={"max_queue_delay_us": 1000, "use_graphs": False},
+    ))
+    out = run(processor.process_request("bert_ep2", "", {
+        "input_ids": [1, 2, 3, 4], "attention_mask": [1, 1, 1, 1]}))
+    assert np.asarray(out).shape == (2,)
+
+
+def test_torchscript_model_serving(processor, store, tmp_path):
+    class TinyNet(torch.nn.Module):
+        def forward(self, x):
+            return x.sum(dim=-1, keepdim=True) * 2.0
+
+    path = tmp_path / "tiny.pt"
+    torch.jit.script(TinyNet()).save(str(path))
+    rec = store.register_model(name="ts-model", project="p", path=str(path))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="ts_ep", model_id=rec.model_id,
+        auxiliary_cfg={"max_queue_delay_us": 1000, "use_graphs": False,
+                       "dtype": "float32"},
+    ))
+    out = run(processor.process_request("ts_ep", "", [1.0, 2.0, 3.0]))
+    assert abs(float(np.asarray(out).ravel()[0]) - 12.0) < 1e-5

@@ -1,0 +1,111 @@
+"""CPU-path correctness of the op dispatch layer (reference implementations)."""
+
+import math
+
+import pytest
+import torch
+
+import clearml_serving_amd.ops as ops
+
+
+def test_layernorm_cpu():
+    x = torch.randn(4, 768)
+    w, b = torch.randn(768), torch.randn(768)
+    ref = torch.nn.functional.layer_norm(x, (768,), w, b, 1e-5)
+    torch.testing.assert_close(ops.layernorm(x, w, b), ref, atol=1e-5,
+                               rtol=1e-5)
+
+
+def test_layernorm_residual_cpu():
+    x, r = torch.randn(4, 64), torch.randn(4, 64)
+    w, b = torch.ones(64), torch.zeros(64)
+    ref = torch.nn.functional.layer_norm(x + r, (64,), w, b, 1e-5)
+    torch.testing.assert_close(ops.layernorm(x, w, b, residual=r), ref,
+                               atol=1e-5, rtol=1e-5)
+
+
+def test_attention_cpu_matches_sdpa():
+    q = torch.randn(2, 4, 32, 64)
+    k = torch.randn(2, 4, 32, 64)
+    v = torch.randn(2, 4, 32, 64)
+    ref = torch.nn.functional.scaled_dot_product_attention(q, k, v)
+    torch.testing.assert_close(ops.attention(q, k, v), ref, atol=1e-4,
+                               rtol=1e-4)
+
+
+def test_attention_cpu_causal_matches_sdpa():
+    q = torch.randn(1, 2, 16, 64)
+    k = torch.randn(1, 2, 16, 64)
+    v = torch.randn(1, 2, 16, 64)
+    ref = torch.nn.functional.scaled_dot_product_attention(q, k, v,
+                                                           is_causal=True)
+    torch.testing.assert_close(ops.attention(q, k, v, causal=True), ref,
+                               atol=1e-4, rtol=1e-4)
+
+
+def test_attention_cpu_gqa():
+    q = torch.randn(1, 8, 16, 64)
+    k = torch.randn(1, 2, 16, 64)
+    v = torch.randn(1, 2, 16, 64)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q, k.repeat_interleave(4, 1), v.repeat_interleave(4, 1))
+    torch.testing.assert_close(ops.attention(q, k, v), ref, atol=1e-4,
+                               rtol=1e-4)
+
+
+def test_attention_decode_cpu():
+    # paged reference vs dense attention on the gathered cache
+    torch.manual_seed(0)
+    b, h, hkv, d, bs = 2, 8, 2, 64, 16
+    nblocks = 8
+    k_cache = torch.randn(nblocks, hkv, bs, d)
+    v_cache = torch.randn(nblocks, hkv, bs, d)
+    block_table = torch.tensor([[0, 2, 4, 6], [1, 3, 5, 7]], dtype=torch.int32)
+    seq_lens = torch.tensor([50, 33], dtype=torch.int32)
+    q = torch.randn(b, h, d)
+    out = ops.attention_decode(q, k_cache, v_cache, block_table, seq_lens)
+    # dense check for sequence 0
+    ks = torch.cat([k_cache[0], k_cache[2], k_cache[4], k_cache[6]], dim=1)[:, :50]
+    vs = torch.cat([v_cache[0], v_cache[2], v_cache[4], v_cache[6]], dim=1)[:, :50]
+    ks = ks.repeat_interleave(4, 0)
+    vs = vs.repeat_interleave(4, 0)
+    scores = torch.einsum("hd,hnd->hn", q[0], ks) / math.sqrt(d)
+    ref0 = torch.einsum("hn,hnd->hd", torch.softmax(scores, -1), vs)
+    torch.testing.assert_close(out[0], ref0, atol=1e-4, rtol=1e-4)
+
+
+def test_rope_cpu_rotation_property():
+    # positions 0 must be identity
+    q = torch.randn(1, 2, 64)
+    k = torch.randn(1, 1, 64)
+    q0, k0 = q.clone(), k.clone()
+    ops.rope_inplace(q, k, torch.zeros(1, dtype=torch.int32))
+    torch.testing.assert_close(q, q0, atol=1e-6, rtol=1e-6)
+    torch.testing.assert_close(k, k0, atol=1e-6, rtol=1e-6)
+
+
+def test_sample_cpu_top_p():
+    logits = torch.tensor([[math.log(0.5), math.log(0.3), math.log(0.15),
+                            math.log(0.05)]])
+    g = torch.Generator().manual_seed(0)
+    for _ in range(50):
+        s = ops.sample_top_k_top_p(logits, temperature=1.0, top_p=0.5,
+                                   generator=g)
+        assert s.item() == 0  # only token 0 is inside the 0.5 nucleus
+
+
+def test_sample_cpu_top_k():
+    torch.manual_seed(0)
+    logits = torch.randn(4, 100)
+    topk = logits.topk(5, dim=-1).indices
+    g = torch.Generator().manual_seed(1)
+    for _ in range(20):
+        s = ops.sample_top_k_top_p(logits, temperature=1.0, top_k=5,
+                                   generator=g)
+        for b in range(4):
+            assert s[b].item() in topk[b].tolist()
+
+
+def test_gpu_dispatch_requires_extension_policy():
+    # documents the loud-failure contract (actual raise exercised on GPU)
+    assert hasattr(ops, "_require_ext")

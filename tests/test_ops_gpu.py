@@ -1,0 +1,255 @@
+"""GPU numerics: every HIP kernel vs a plain PyTorch fp32 reference.
+
+The reference implementations live in ops/__init__.py (CPU dispatch path);
+each test computes the op in fp32 on the same data and compares within
+bf16-accumulation tolerances.
+"""
+
+import math
+
+import pytest
+import torch
+
+import clearml_serving_amd.ops as ops
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def ref_f32(fn, *tensors, **kw):
+    cpu = [t.detach().float().cpu() if isinstance(t, torch.Tensor) else t
+           for t in tensors]
+    return fn(*cpu, **kw)
+
+
+def assert_close_bf16(got, ref, atol=2e-2, rtol=2e-2):
+    got = got.detach().float().cpu()
+    ref = ref.detach().float().cpu()
+    torch.testing.assert_close(got, ref, atol=atol, rtol=rtol)
+
+
+def test_extension_is_loaded():
+    # on a GPU box the extension must be present -- no silent eager fallback
+    assert ops.has_extension(), "HIP extension missing on GPU box"
+
+
+@pytest.mark.parametrize("shape", [(8, 64, 56, 56), (2, 2048, 7, 7)])
+def test_relu_add(shape):
+    x = torch.randn(shape, device=DEV, dtype=torch.bfloat16)
+    r = torch.randn(shape, device=DEV, dtype=torch.bfloat16)
+    got = ops.bias_relu_add(x, residual=r)
+    ref = torch.relu(x.float() + r.float())
+    assert_close_bf16(got, ref)
+
+
+def test_bias_relu_add_channel_bias():
+    x = torch.randn(4, 64, 14, 14, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(64, device=DEV, dtype=torch.bfloat16)
+    got = ops.bias_relu_add(x, bias=b)
+    ref = torch.relu(x.float() + b.float()[None, :, None, None])
+    assert_close_bf16(got, ref)
+
+
+@pytest.mark.parametrize("rows,cols", [(128, 3072), (1000, 768)])
+def test_bias_gelu(rows, cols):
+    x = torch.randn(rows, cols, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(cols, device=DEV, dtype=torch.bfloat16)
+    got = ops.bias_gelu(x, b)
+    ref = torch.nn.functional.gelu(x.float() + b.float())
+    assert_close_bf16(got, ref)
+
+
+def test_silu_mul():
+    g = torch.randn(64, 11008, device=DEV, dtype=torch.bfloat16)
+    u = torch.randn(64, 11008, device=DEV, dtype=torch.bfloat16)
+    got = ops.silu_mul(g, u)
+    ref = torch.nn.functional.silu(g.float()) * u.float()
+    assert_close_bf16(got, ref)
+
+
+@pytest.mark.parametrize("h", [768, 1024, 4096, 8192])
+def test_layernorm(h):
+    x = torch.randn(64, h, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(h, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(h, device=DEV, dtype=torch.bfloat16)
+    got = ops.layernorm(x, w, b)
+    ref = torch.nn.functional.layer_norm(
+        x.float(), (h,), w.float(), b.float(), 1e-5)
+    assert_close_bf16(got, ref)
+
+
+def test_layernorm_fused_residual():
+    x = torch.randn(32, 768, device=DEV, dtype=torch.bfloat16)
+    r = torch.randn(32, 768, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(768, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(768, device=DEV, dtype=torch.bfloat16)
+    got = ops.layernorm(x, w, b, residual=r)
+    ref = torch.nn.functional.layer_norm(
+        (x.float() + r.float()), (768,), w.float(), b.float(), 1e-5)
+    assert_close_bf16(got, ref)
+
+
+@pytest.mark.parametrize("h", [4096, 8192])
+def test_rmsnorm(h):
+    x = torch.randn(16, h, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(h, device=DEV, dtype=torch.bfloat16)
+    got = ops.rmsnorm(x, w)
+    xf = x.float()
+    ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    assert_close_bf16(got, ref)
+
+
+def test_rmsnorm_residual_update():
+    x = torch.randn(8, 4096, device=DEV, dtype=torch.bfloat16)
+    r = torch.randn(8, 4096, device=DEV, dtype=torch.bfloat16)
+    r_ref = (x.float() + r.float())
+    w = torch.ones(4096, device=DEV, dtype=torch.bfloat16)
+    got = ops.rmsnorm(x, w, residual=r)
+    ref = r_ref * torch.rsqrt(r_ref.pow(2).mean(-1, keepdim=True) + 1e-6)
+    assert_close_bf16(got, ref)
+    # residual stream updated in place to x + r
+    assert_close_bf16(r, r_ref)
+
+
+@pytest.mark.parametrize("shape", [(32, 1000), (12, 384, 384)])
+def test_softmax(shape):
+    x = torch.randn(shape, device=DEV, dtype=torch.bfloat16) * 4
+    got = ops.softmax(x)
+    ref = torch.softmax(x.float(), dim=-1)
+    assert_close_bf16(got, ref, atol=1e-2)
+
+
+# ------------------------------------------------------------------ #
+# attention
+# ------------------------------------------------------------------ #
+def _attn_ref(q, k, v, causal=False, seq_lens=None):
+    return ops.attention(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                         causal=causal,
+                         seq_lens=None if seq_lens is None else seq_lens.cpu())
+
+
+@pytest.mark.parametrize("b,h,s,d", [(2, 12, 128, 64), (1, 8, 384, 64),
+                                     (2, 4, 512, 128), (1, 2, 77, 64)])
+def test_attention_prefill(b, h, s, d):
+    torch.manual_seed(0)
+    q = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    got = ops.attention(q, k, v)
+    ref = _attn_ref(q, k, v)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_transpose_detecting():
+    # asymmetric, non-random pattern that catches swapped row/col in the
+    # C-write or operand maps (guide §5.4 rule 16)
+    b, h, s, d = 1, 1, 64, 64
+    q = torch.zeros(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.zeros(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.zeros(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    for i in range(s):
+        q[0, 0, i, i % d] = 1.0
+        k[0, 0, i, (i * 7) % d] = 1.0
+        v[0, 0, i, :] = (i % 13) * 0.25 - 1.0
+    got = ops.attention(q, k, v)
+    ref = _attn_ref(q, k, v)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_causal():
+    b, h, s, d = 2, 4, 256, 128
+    torch.manual_seed(1)
+    q = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    got = ops.attention(q, k, v, causal=True)
+    ref = _attn_ref(q, k, v, causal=True)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_seq_lens_padding():
+    b, h, s, d = 3, 2, 128, 64
+    torch.manual_seed(2)
+    q = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    seq_lens = torch.tensor([128, 60, 7], device=DEV, dtype=torch.int32)
+    got = ops.attention(q, k, v, seq_lens=seq_lens)
+    ref = _attn_ref(q, k, v, seq_lens=seq_lens)
+    # only compare valid query rows (padded-query outputs are don't-care)
+    for i, n in enumerate([128, 60, 7]):
+        assert_close_bf16(got[i, :, :n], ref[i, :, :n], atol=3e-2, rtol=3e-2)
+
+
+def test_attention_gqa():
+    b, hq, hkv, s, d = 2, 8, 2, 128, 128
+    torch.manual_seed(3)
+    q = torch.randn(b, hq, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(b, hkv, s, d, device=DEV, dtype=torch.bfloat16)
+    got = ops.attention(q, k, v)
+    ref = _attn_ref(q, k, v)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_outlier_key_forces_rescale():
+    # spike one K row so the running max jumps mid-sequence (forces the
+    # online-softmax rescale branch, guide §5.4 rule 26)
+    b, h, s, d = 1, 1, 256, 64
+    torch.manual_seed(4)
+    q = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(b, h, s, d, device=DEV, dtype=torch.bfloat16)
+    k[0, 0, 200] = q[0, 0, 10] * 8  # huge score at tile 4 for query 10
+    got = ops.attention(q, k, v)
+    ref = _attn_ref(q, k, v)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+# ------------------------------------------------------------------ #
+# rope / sampling
+# ------------------------------------------------------------------ #
+def test_rope():
+    t, h, hkv, d = 64, 8, 2, 128
+    torch.manual_seed(5)
+    q = torch.randn(t, h, d, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(t, hkv, d, device=DEV, dtype=torch.bfloat16)
+    pos = torch.arange(100, 100 + t, device=DEV, dtype=torch.int32)
+    q_ref = q.float().cpu().clone()
+    k_ref = k.float().cpu().clone()
+    ops.rope_inplace(q, k, pos)
+    ops.rope_inplace(q_ref, k_ref, pos.cpu())
+    assert_close_bf16(q, q_ref, atol=3e-2, rtol=3e-2)
+    assert_close_bf16(k, k_ref, atol=3e-2, rtol=3e-2)
+
+
+def test_sample_greedy_matches_argmax():
+    torch.manual_seed(6)
+    logits = torch.randn(8, 32000, device=DEV, dtype=torch.bfloat16)
+    got = ops.sample_top_k_top_p(logits, temperature=0.0)
+    assert torch.equal(got, logits.argmax(dim=-1))
+
+
+def test_sample_respects_top_k():
+    torch.manual_seed(7)
+    logits = torch.randn(16, 1000, device=DEV, dtype=torch.float32)
+    topk_sets = logits.topk(10, dim=-1).indices
+    for _ in range(5):
+        s = ops.sample_top_k_top_p(logits, temperature=1.0, top_k=10)
+        for b in range(16):
+            assert s[b].item() in topk_sets[b].tolist()
+
+
+def test_sample_distribution_sanity():
+    # two tokens with 9:1 odds: frequency must track the distribution
+    logits = torch.full((512, 4), -1e9, device=DEV, dtype=torch.float32)
+    logits[:, 0] = math.log(0.9)
+    logits[:, 1] = math.log(0.1)
+    counts = torch.zeros(4)
+    for trial in range(20):
+        s = ops.sample_top_k_top_p(logits, temperature=1.0)
+        counts += torch.bincount(s.cpu(), minlength=4).float()
+    frac = counts / counts.sum()
+    assert abs(frac[0].item() - 0.9) < 0.03
+    assert counts[2] == 0 and counts[3] == 0
