@@ -66,6 +66,7 @@ class DynamicBatcher:
         self._worker_task: Optional[asyncio.Task] = None
         self._stream = torch.cuda.Stream() if self.is_cuda else None
         self._graphs: Dict[int, dict] = {}  # bucket -> {graph, in, out}
+        self._pinned: Dict = {}             # (key,bucket,shape,dtype) -> slab
         self._exec_lock = threading.Lock()
         self._closed = False
         # telemetry for the Prometheus exporter
@@ -100,6 +101,13 @@ class DynamicBatcher:
             batch: List = [first]
             deadline = time.monotonic() + self.max_queue_delay_s
             while len(batch) < self.max_batch_size:
+                # drain whatever is already queued without timer machinery
+                try:
+                    while len(batch) < self.max_batch_size:
+                        batch.append(self._queue.get_nowait())
+                    break
+                except asyncio.QueueEmpty:
+                    pass
                 timeout = deadline - time.monotonic()
                 if timeout <= 0:
                     break
@@ -130,32 +138,58 @@ class DynamicBatcher:
         self.stats["occupancy_sum"] += n / float(bucket)
 
         is_dict = isinstance(inputs[0], dict)
-        if is_dict:
-            keys = list(inputs[0].keys())
-            stacked = {
-                k: self._stack_pad([x[k] for x in inputs], bucket) for k in keys
-            }
-        else:
-            stacked = self._stack_pad(inputs, bucket)
+
+        def _stage():
+            # must run with the batcher's stream current: the async H2D copy
+            # out of the pinned slab orders against the model launches below
+            if is_dict:
+                keys = list(inputs[0].keys())
+                return {
+                    k: self._stack_pad([x[k] for x in inputs], bucket, key=k)
+                    for k in keys
+                }
+            return self._stack_pad(inputs, bucket)
 
         with self._exec_lock:
             if self.is_cuda:
                 with torch.cuda.stream(self._stream):
+                    stacked = _stage()
                     out = self._run_model(stacked, bucket)
                     out_cpu = _to_cpu(out)
                 self._stream.synchronize()
             else:
-                out_cpu = _to_cpu(self._run_model(stacked, bucket))
+                out_cpu = _to_cpu(self._run_model(_stage(), bucket))
         return [_slice(out_cpu, i) for i in range(n)]
 
-    def _stack_pad(self, tensors: List[torch.Tensor], bucket: int) -> torch.Tensor:
-        t = torch.stack([torch.as_tensor(x) for x in tensors], dim=0)
+    def _stack_pad(self, tensors: List[torch.Tensor], bucket: int,
+                   key: str = "") -> torch.Tensor:
+        ts = [torch.as_tensor(x) for x in tensors]
+        want = self.dtype if (self.dtype is not None
+                              and ts[0].is_floating_point()) else ts[0].dtype
+        if ts[0].dtype != want:
+            ts = [t.to(want) for t in ts]
+        if self.is_cuda:
+            # stage through a reusable pinned host slab -> async H2D
+            buf = self._pinned_slab(key, bucket, ts[0].shape, want)
+            n = len(ts)
+            torch.stack(ts, dim=0, out=buf[:n])
+            if n < bucket:
+                buf[n:bucket] = buf[0]
+            return buf[:bucket].to(self.device, non_blocking=True)
+        t = torch.stack(ts, dim=0)
         if t.shape[0] < bucket:
             pad = t[:1].expand(bucket - t.shape[0], *t.shape[1:])
             t = torch.cat([t, pad], dim=0)
-        if self.dtype is not None and t.is_floating_point():
-            t = t.to(self.dtype)
-        return t.to(self.device, non_blocking=True)
+        return t
+
+    def _pinned_slab(self, key: str, bucket: int, shape, dtype) -> torch.Tensor:
+        slab_key = (key, bucket, tuple(shape), dtype)
+        slab = self._pinned.get(slab_key)
+        if slab is None:
+            slab = torch.empty((bucket, *shape), dtype=dtype,
+                               pin_memory=True)
+            self._pinned[slab_key] = slab
+        return slab
 
     @torch.inference_mode()
     def _run_model(self, stacked: TensorOrDict, bucket: int) -> TensorOrDict:
