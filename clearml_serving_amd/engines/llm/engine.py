@@ -1,0 +1,535 @@
+"""Native LLM engine: paged KV cache + continuous batching scheduler.
+
+The in-process replacement for the reference's vLLM delegation
+(preprocess_service.py:619-1095). One engine per process; requests stream
+tokens through asyncio queues.
+
+Engine step (continuous batching):
+1. admit waiting prompts while the prefill-token budget and free KV blocks
+   allow (prompts padded to one batch; per-seq lengths mask attention)
+2. run ONE prefill forward for the admitted batch, scatter K/V into pages,
+   sample each sequence's first token
+3. run ONE decode forward for every running sequence (paged decode
+   attention), sample next tokens
+4. emit tokens to per-request streams; finished sequences free their pages
+
+KV sizing: pages are allocated once at startup from a fraction of free HBM
+(288 GB/GPU -> tens of thousands of 16-token pages for an 8B model).
+"""
+
+import asyncio
+import json
+import os
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, AsyncGenerator, Dict, List, Optional
+
+import torch
+
+from ... import ops
+from ...models.llama import PRESETS, LlamaConfig, LlamaForCausalLM
+
+
+# --------------------------------------------------------------------- #
+# tokenizer
+# --------------------------------------------------------------------- #
+class SimpleTokenizer:
+    """Byte-level fallback tokenizer (synthetic serving / tests: the
+    environment has no model hub, so real checkpoints supply their own
+    tokenizer.json)."""
+
+    eos_id = 0
+    vocab_size = 257
+
+    def encode(self, text: str) -> List[int]:
+        return [b + 1 for b in text.encode("utf-8")]
+
+    def decode(self, ids: List[int]) -> str:
+        # ids above the byte range (random-init models) fold back into it
+        return bytes(min(max(0, i - 1) % 256, 255) for i in ids
+                     if i > 0).decode("utf-8", errors="replace")
+
+
+class HfTokenizer:
+    def __init__(self, path: str):
+        from tokenizers import Tokenizer
+
+        self._tok = Tokenizer.from_file(path)
+        self.vocab_size = self._tok.get_vocab_size()
+        eos = None
+        for cand in ("</s>", "<|eot_id|>", "<|end_of_text|>", "<|endoftext|>"):
+            tid = self._tok.token_to_id(cand)
+            if tid is not None:
+                eos = tid
+                break
+        self.eos_id = eos if eos is not None else 0
+
+    def encode(self, text: str) -> List[int]:
+        return self._tok.encode(text).ids
+
+    def decode(self, ids: List[int]) -> str:
+        return self._tok.decode(ids)
+
+
+# --------------------------------------------------------------------- #
+@dataclass
+class SamplingParams:
+    temperature: float = 1.0
+    top_k: int = 0
+    top_p: float = 1.0
+    max_tokens: int = 128
+    stop_token_ids: List[int] = field(default_factory=list)
+    ignore_eos: bool = False
+
+    @classmethod
+    def from_request(cls, body: Dict[str, Any], default_max: int = 128):
+        return cls(
+            temperature=float(body.get("temperature", 1.0)),
+            top_k=int(body.get("top_k", 0) or 0),
+            top_p=float(body.get("top_p", 1.0)),
+            max_tokens=int(body.get("max_tokens", default_max)),
+            ignore_eos=bool(body.get("ignore_eos", False)),
+        )
+
+
+@dataclass
+class LlmEngineConfig:
+    arch: str = "llama"
+    preset: str = "llama-3-8b"
+    overrides: Dict[str, Any] = field(default_factory=dict)
+    dtype: str = "bfloat16"
+    block_size: int = 16
+    max_num_seqs: int = 64
+    max_model_len: int = 4096
+    max_prefill_tokens: int = 8192
+    gpu_memory_fraction: float = 0.85
+    num_kv_blocks: Optional[int] = None  # explicit override (CPU tests)
+    weights: Optional[str] = None
+    tokenizer_path: Optional[str] = None
+    device: Optional[str] = None
+
+    @classmethod
+    def from_aux(cls, model_path: Optional[str], aux: Dict[str, Any]):
+        cfg = cls()
+        card: Dict[str, Any] = {}
+        if model_path:
+            if os.path.isdir(model_path):
+                for name in ("model_card.json", "card.json"):
+                    p = os.path.join(model_path, name)
+                    if os.path.exists(p):
+                        with open(p) as f:
+                            card = json.load(f)
+                        break
+                tok = os.path.join(model_path, "tokenizer.json")
+                if os.path.exists(tok):
+                    cfg.tokenizer_path = tok
+                w = os.path.join(model_path, "model.safetensors")
+                if os.path.exists(w):
+                    cfg.weights = w
+            elif model_path.endswith(".json"):
+                with open(model_path) as f:
+                    card = json.load(f)
+        card.pop("arch", None)
+        for key in ("preset", "dtype", "block_size", "max_num_seqs",
+                    "max_model_len", "max_prefill_tokens",
+                    "gpu_memory_fraction", "num_kv_blocks", "weights",
+                    "device"):
+            for src in (card, aux):
+                if key in src and src[key] is not None:
+                    setattr(cfg, key, src[key])
+        cfg.overrides = {**card.get("overrides", {}),
+                         **(aux.get("overrides") or {})}
+        return cfg
+
+
+class BlockAllocator:
+    def __init__(self, num_blocks: int):
+        self.num_blocks = num_blocks
+        self._free = list(range(num_blocks - 1, -1, -1))
+
+    def alloc(self, n: int) -> List[int]:
+        if n > len(self._free):
+            raise RuntimeError("KV cache exhausted")
+        return [self._free.pop() for _ in range(n)]
+
+    def free(self, blocks: List[int]) -> None:
+        self._free.extend(blocks)
+
+    @property
+    def available(self) -> int:
+        return len(self._free)
+
+
+class Sequence:
+    def __init__(self, req_id: str, prompt_ids: List[int],
+                 params: SamplingParams):
+        self.req_id = req_id
+        self.prompt_ids = prompt_ids
+        self.output_ids: List[int] = []
+        self.params = params
+        self.blocks: List[int] = []
+        self.stream: "asyncio.Queue" = asyncio.Queue()
+        self.finished = False
+        self.finish_reason: Optional[str] = None
+        self.created = time.time()
+        self.first_token_time: Optional[float] = None
+
+    def __len__(self):
+        return len(self.prompt_ids) + len(self.output_ids)
+
+
+class LlmEngine:
+    def __init__(self, cfg: LlmEngineConfig):
+        self.cfg = cfg
+        self.device = torch.device(
+            cfg.device or ("cuda" if torch.cuda.is_available() else "cpu"))
+        self.dtype = (torch.bfloat16 if self.device.type == "cuda"
+                      else torch.float32)
+        self.model: Optional[LlamaForCausalLM] = None
+        self.kv_caches: List = []
+        self.allocator: Optional[BlockAllocator] = None
+        self.tokenizer = None
+        self.waiting: List[Sequence] = []
+        self.running: List[Sequence] = []
+        self._loop_task: Optional[asyncio.Task] = None
+        self._wake: Optional[asyncio.Event] = None
+        self._started = False
+        self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
+                      "prefill_batches": 0, "decode_batches": 0}
+
+    # ------------------------------------------------------------------ #
+    def start(self) -> None:
+        if self._started:
+            return
+        cfg = self.cfg
+        mcfg = LlamaConfig(**{**PRESETS[cfg.preset].__dict__,
+                              **cfg.overrides})
+        self.model_config = mcfg
+        model = LlamaForCausalLM(mcfg)
+        if cfg.weights:
+            from ...models import load_weights
+
+            load_weights(model, cfg.weights)
+        self.model = model.eval().to(self.device).to(self.dtype)
+
+        self.tokenizer = (HfTokenizer(cfg.tokenizer_path)
+                          if cfg.tokenizer_path else SimpleTokenizer())
+
+        # KV cache sizing from free HBM
+        bs = cfg.block_size
+        per_block_bytes = (2 * mcfg.kv_heads * bs * mcfg.head_dim *
+                           self.dtype.itemsize * mcfg.layers)
+        if cfg.num_kv_blocks:
+            num_blocks = int(cfg.num_kv_blocks)
+        elif self.device.type == "cuda":
+            free, _ = torch.cuda.mem_get_info(self.device)
+            budget = int(free * cfg.gpu_memory_fraction)
+            num_blocks = max(budget // per_block_bytes, 64)
+        else:
+            num_blocks = 512
+        self.allocator = BlockAllocator(num_blocks)
+        self.kv_caches = [
+            (torch.zeros(num_blocks, mcfg.kv_heads, bs, mcfg.head_dim,
+                         dtype=self.dtype, device=self.device),
+             torch.zeros(num_blocks, mcfg.kv_heads, bs, mcfg.head_dim,
+                         dtype=self.dtype, device=self.device))
+            for _ in range(mcfg.layers)
+        ]
+        self._started = True
+
+    # ------------------------------------------------------------------ #
+    # request API
+    # ------------------------------------------------------------------ #
+    async def add_request(self, prompt_ids: List[int],
+                          params: SamplingParams) -> Sequence:
+        if len(prompt_ids) >= self.cfg.max_model_len:
+            raise ValueError(
+                "prompt length {} exceeds max_model_len {}".format(
+                    len(prompt_ids), self.cfg.max_model_len))
+        seq = Sequence(uuid.uuid4().hex, prompt_ids, params)
+        self.waiting.append(seq)
+        self._ensure_loop()
+        self._wake.set()
+        return seq
+
+    async def generate(self, prompt: str, params: SamplingParams
+                       ) -> AsyncGenerator[Dict[str, Any], None]:
+        ids = self.tokenizer.encode(prompt)
+        seq = await self.add_request(ids, params)
+        while True:
+            item = await seq.stream.get()
+            yield item
+            if item.get("finished"):
+                return
+
+    async def generate_simple(self, body: Dict[str, Any]) -> Dict[str, Any]:
+        """Non-OpenAI route: {"prompt": str, "max_tokens": ...}."""
+        prompt = body.get("prompt") or body.get("text") or ""
+        params = SamplingParams.from_request(body)
+        tokens: List[int] = []
+        async for item in self.generate(prompt, params):
+            tokens.extend(item.get("token_ids", []))
+        return {
+            "text": self.tokenizer.decode(tokens),
+            "tokens": len(tokens),
+        }
+
+    # ------------------------------------------------------------------ #
+    # engine loop
+    # ------------------------------------------------------------------ #
+    def _ensure_loop(self) -> None:
+        loop = asyncio.get_running_loop()
+        if self._loop_task is None or getattr(self, "_loop_ref", None) is not loop:
+            self._loop_ref = loop
+            self._wake = asyncio.Event()
+            self._loop_task = loop.create_task(self._engine_loop())
+
+    async def _engine_loop(self) -> None:
+        while True:
+            if not self.waiting and not self.running:
+                self._wake.clear()
+                await self._wake.wait()
+            try:
+                await asyncio.to_thread(self.step)
+            except Exception as ex:
+                # poison every active sequence rather than hanging clients
+                for seq in self.waiting + self.running:
+                    seq.stream.put_nowait(
+                        {"error": str(ex), "finished": True, "token_ids": []})
+                self.waiting.clear()
+                self.running.clear()
+                raise
+            # hand emitted tokens to the event loop promptly
+            await asyncio.sleep(0)
+
+    # ------------------------------------------------------------------ #
+    def step(self) -> None:
+        """One scheduler iteration: admit + prefill, then decode."""
+        self.stats["steps"] += 1
+        admitted = self._admit()
+        if admitted:
+            self._prefill(admitted)
+        decoding = [s for s in self.running if not s.finished
+                    and s not in admitted]
+        if decoding:
+            self._decode(decoding)
+        for s in list(self.running):
+            if s.finished:
+                self.running.remove(s)
+                self.allocator.free(s.blocks)
+                s.blocks = []
+
+    def _admit(self) -> List[Sequence]:
+        admitted: List[Sequence] = []
+        tokens = 0
+        bs = self.cfg.block_size
+        while self.waiting and len(self.running) < self.cfg.max_num_seqs:
+            seq = self.waiting[0]
+            need = len(seq.prompt_ids)
+            if admitted and tokens + need > self.cfg.max_prefill_tokens:
+                break
+            need_blocks = (need + bs - 1) // bs
+            # keep one spare block per running seq for decode growth
+            if need_blocks + len(self.running) + 1 > self.allocator.available:
+                break
+            seq.blocks = self.allocator.alloc(need_blocks)
+            self.waiting.pop(0)
+            self.running.append(seq)
+            admitted.append(seq)
+            tokens += need
+        return admitted
+
+    def _slot(self, seq: Sequence, pos: int) -> int:
+        bs = self.cfg.block_size
+        return seq.blocks[pos // bs] * bs + pos % bs
+
+    @torch.inference_mode()
+    def _prefill(self, seqs: List[Sequence]) -> None:
+        self.stats["prefill_batches"] += 1
+        bs_cfg = self.cfg.block_size
+        b = len(seqs)
+        lens = [len(s.prompt_ids) for s in seqs]
+        smax = max(lens)
+        dev = self.device
+
+        tokens = torch.zeros(b, smax, dtype=torch.long)
+        positions = torch.zeros(b, smax, dtype=torch.int32)
+        slot_map = torch.full((b, smax), -1, dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            n = lens[i]
+            tokens[i, :n] = torch.tensor(s.prompt_ids, dtype=torch.long)
+            positions[i, :n] = torch.arange(n, dtype=torch.int32)
+            slot_map[i, :n] = torch.tensor(
+                [self._slot(s, p) for p in range(n)], dtype=torch.int32)
+        seq_lens = torch.tensor(lens, dtype=torch.int32, device=dev)
+        attn_ctx = {
+            "mode": "prefill", "batch": b, "seq": smax,
+            "seq_lens": seq_lens,
+            "slot_mapping": slot_map.view(-1).to(dev),
+        }
+        last_idx = torch.tensor(
+            [i * smax + lens[i] - 1 for i in range(b)], dtype=torch.long,
+            device=dev)
+        logits = self.model(
+            tokens.view(-1).to(dev), positions.view(-1).to(dev),
+            kv_caches=self.kv_caches, attn_ctx=attn_ctx,
+            last_token_idx=last_idx)
+        self.stats["prompt_tokens"] += sum(lens)
+        self._sample_and_emit(seqs, logits)
+
+    @torch.inference_mode()
+    def _decode(self, seqs: List[Sequence]) -> None:
+        self.stats["decode_batches"] += 1
+        bs_cfg = self.cfg.block_size
+        dev = self.device
+        b = len(seqs)
+        # grow block tables for the token being generated
+        for s in seqs:
+            pos = len(s) - 1  # position of the last generated token
+            if pos // bs_cfg >= len(s.blocks):
+                s.blocks.extend(self.allocator.alloc(1))
+        tokens = torch.tensor([s.output_ids[-1] for s in seqs],
+                              dtype=torch.long, device=dev)
+        positions = torch.tensor([len(s) - 1 for s in seqs],
+                                 dtype=torch.int32, device=dev)
+        slot_map = torch.tensor([self._slot(s, len(s) - 1) for s in seqs],
+                                dtype=torch.int32, device=dev)
+        max_blocks = max(len(s.blocks) for s in seqs)
+        btab = torch.zeros(b, max_blocks, dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            btab[i, :len(s.blocks)] = torch.tensor(s.blocks, dtype=torch.int32)
+        attn_ctx = {
+            "mode": "decode",
+            "seq_lens": torch.tensor([len(s) for s in seqs],
+                                     dtype=torch.int32, device=dev),
+            "block_table": btab.to(dev),
+            "slot_mapping": slot_map,
+        }
+        logits = self.model(tokens, positions, kv_caches=self.kv_caches,
+                            attn_ctx=attn_ctx, last_token_idx=None)
+        self._sample_and_emit(seqs, logits)
+
+    def _sample_and_emit(self, seqs: List[Sequence], logits: torch.Tensor) -> None:
+        # group rows by identical sampling params for batched kernels
+        groups: Dict[tuple, List[int]] = {}
+        for i, s in enumerate(seqs):
+            key = (s.params.temperature, s.params.top_k, s.params.top_p)
+            groups.setdefault(key, []).append(i)
+        next_ids = torch.empty(len(seqs), dtype=torch.long)
+        for (temp, top_k, top_p), idxs in groups.items():
+            rows = logits[idxs] if len(idxs) < len(seqs) else logits
+            sampled = ops.sample_top_k_top_p(
+                rows, temperature=temp, top_k=top_k, top_p=top_p)
+            next_ids[idxs] = sampled.cpu()
+        eos = self.tokenizer.eos_id
+        now = time.time()
+        for i, s in enumerate(seqs):
+            tok = int(next_ids[i])
+            s.output_ids.append(tok)
+            if s.first_token_time is None:
+                s.first_token_time = now
+            self.stats["generated_tokens"] += 1
+            finished = False
+            reason = None
+            if not s.params.ignore_eos and (
+                    tok == eos or tok in s.params.stop_token_ids):
+                finished, reason = True, "stop"
+            elif len(s.output_ids) >= s.params.max_tokens:
+                finished, reason = True, "length"
+            elif len(s) >= self.cfg.max_model_len:
+                finished, reason = True, "length"
+            s.finished = finished
+            s.finish_reason = reason
+            s.stream.put_nowait({
+                "token_ids": [tok],
+                "text": self.tokenizer.decode([tok]),
+                "finished": finished,
+                "finish_reason": reason,
+            })
+
+    # ------------------------------------------------------------------ #
+    # OpenAI-compatible handlers (route /serve/openai/v1/*)
+    # ------------------------------------------------------------------ #
+    def _chat_prompt(self, messages: List[Dict[str, str]]) -> str:
+        parts = []
+        for m in messages:
+            parts.append("<|{}|>\n{}".format(m.get("role", "user"),
+                                             m.get("content", "")))
+        parts.append("<|assistant|>\n")
+        return "\n".join(parts)
+
+    async def openai_chat_completions(self, body: Dict[str, Any],
+                                      model_name: str):
+        messages = body.get("messages") or []
+        prompt = self._chat_prompt(messages)
+        params = SamplingParams.from_request(body)
+        rid = "chatcmpl-" + uuid.uuid4().hex[:24]
+        if body.get("stream"):
+            return self._sse_stream(prompt, params, rid, model_name,
+                                    chat=True)
+        text, reason, ntok, nprompt = await self._collect(prompt, params)
+        return {
+            "id": rid, "object": "chat.completion", "created": int(time.time()),
+            "model": model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant", "content": text},
+                         "finish_reason": reason}],
+            "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
+                      "total_tokens": nprompt + ntok},
+        }
+
+    async def openai_completions(self, body: Dict[str, Any], model_name: str):
+        prompt = body.get("prompt") or ""
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        params = SamplingParams.from_request(body)
+        rid = "cmpl-" + uuid.uuid4().hex[:24]
+        if body.get("stream"):
+            return self._sse_stream(prompt, params, rid, model_name,
+                                    chat=False)
+        text, reason, ntok, nprompt = await self._collect(prompt, params)
+        return {
+            "id": rid, "object": "text_completion", "created": int(time.time()),
+            "model": model_name,
+            "choices": [{"index": 0, "text": text, "finish_reason": reason}],
+            "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
+                      "total_tokens": nprompt + ntok},
+        }
+
+    def openai_models(self, model_name: str):
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "clearml-serving-amd"}]}
+
+    async def _collect(self, prompt: str, params: SamplingParams):
+        ids = self.tokenizer.encode(prompt)
+        tokens: List[int] = []
+        reason = None
+        async for item in self.generate(prompt, params):
+            tokens.extend(item.get("token_ids", []))
+            reason = item.get("finish_reason") or reason
+        return self.tokenizer.decode(tokens), reason, len(tokens), len(ids)
+
+    def _sse_stream(self, prompt: str, params: SamplingParams, rid: str,
+                    model_name: str, chat: bool):
+        from fastapi.responses import StreamingResponse
+
+        async def gen():
+            async for item in self.generate(prompt, params):
+                if chat:
+                    delta = {"content": item.get("text", "")}
+                    choice = {"index": 0, "delta": delta,
+                              "finish_reason": item.get("finish_reason")}
+                    obj = "chat.completion.chunk"
+                else:
+                    choice = {"index": 0, "text": item.get("text", ""),
+                              "finish_reason": item.get("finish_reason")}
+                    obj = "text_completion"
+                chunk = {"id": rid, "object": obj,
+                         "created": int(time.time()), "model": model_name,
+                         "choices": [choice]}
+                yield "data: {}\n\n".format(json.dumps(chunk))
+            yield "data: [DONE]\n\n"
+
+        return StreamingResponse(gen(), media_type="text/event-stream")

@@ -68,4 +68,4 @@ def load_weights(model: torch.nn.Module, path: str) -> None:
     model.load_state_dict(state)
 
 
-from . import bert, resnet  # noqa: E402,F401  (register architectures)
+from . import bert, llama, resnet  # noqa: E402,F401  (register architectures)

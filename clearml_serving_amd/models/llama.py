@@ -1,0 +1,193 @@
+"""Llama-family decoder for the native LLM engine.
+
+Llama-3 architecture (RMSNorm, SwiGLU MLP, RoPE, GQA) built directly on the
+HIP op library with a paged KV cache (the reference delegates all of this to
+vLLM, preprocess_service.py:619-1095):
+
+- prefill: padded [B, S] token batch; fused causal flash attention with
+  per-sequence lengths; computed K/V scattered into cache pages
+  (ops.kv_cache_write) so decode continues where prefill left off
+- decode:  [B] one-token batch; paged decode attention (ops.attention_decode)
+- RMSNorm keeps the residual stream in place (one HBM pass per norm);
+  SwiGLU uses the fused silu_mul kernel; QKV and gate+up are single merged
+  GEMMs (hipBLASLt)
+
+Presets: llama-3-8b (the BASELINE.json TP=8 target), plus scaled-down
+variants for tests.
+"""
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from . import register_arch
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden: int = 4096
+    layers: int = 32
+    heads: int = 32
+    kv_heads: int = 8
+    intermediate: int = 14336
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    max_position: int = 8192
+    tie_embeddings: bool = False
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden // self.heads
+
+
+class LlamaLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_size: int = 1):
+        super().__init__()
+        self.cfg = cfg
+        assert cfg.heads % tp_size == 0 and cfg.kv_heads % tp_size == 0, \
+            "TP degree must divide head counts"
+        self.heads = cfg.heads // tp_size
+        self.kv_heads = cfg.kv_heads // tp_size
+        self.head_dim = cfg.head_dim
+        self.tp_size = tp_size
+        h, hd = cfg.hidden, cfg.head_dim
+        q_out = self.heads * hd
+        kv_out = self.kv_heads * hd
+        inter = cfg.intermediate // tp_size
+
+        self.attn_norm = nn.Parameter(torch.ones(h))
+        self.qkv = nn.Linear(h, q_out + 2 * kv_out, bias=False)
+        self.o_proj = nn.Linear(q_out, h, bias=False)
+        self.mlp_norm = nn.Parameter(torch.ones(h))
+        self.gate_up = nn.Linear(h, 2 * inter, bias=False)
+        self.down = nn.Linear(inter, h, bias=False)
+        self.inter = inter
+
+    def forward(self, x, residual, positions, kv_cache, attn_ctx):
+        cfg = self.cfg
+        # fused: residual += x_prev; x = rmsnorm(residual)
+        x = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps, residual=residual)
+        t = x.shape[0]
+
+        qkv = self.qkv(x)
+        q, k, v = qkv.split(
+            [self.heads * self.head_dim, self.kv_heads * self.head_dim,
+             self.kv_heads * self.head_dim], dim=-1)
+        q = q.view(t, self.heads, self.head_dim)
+        k = k.contiguous().view(t, self.kv_heads, self.head_dim)
+        v = v.contiguous().view(t, self.kv_heads, self.head_dim)
+        q, k = ops.rope_inplace(q.contiguous(), k, positions,
+                                theta=cfg.rope_theta)
+
+        if kv_cache is not None:
+            k_cache, v_cache = kv_cache
+            ops.kv_cache_write(k, v, k_cache, v_cache, attn_ctx["slot_mapping"])
+
+        scale = 1.0 / math.sqrt(self.head_dim)
+        if attn_ctx["mode"] == "prefill":
+            b, s = attn_ctx["batch"], attn_ctx["seq"]
+            qb = q.view(b, s, self.heads, self.head_dim).transpose(1, 2) \
+                .contiguous()
+            kb = k.view(b, s, self.kv_heads, self.head_dim).transpose(1, 2) \
+                .contiguous()
+            vb = v.view(b, s, self.kv_heads, self.head_dim).transpose(1, 2) \
+                .contiguous()
+            ctx = ops.attention(qb, kb, vb, causal=True, scale=scale,
+                                seq_lens=attn_ctx["seq_lens"])
+            ctx = ctx.transpose(1, 2).reshape(t, self.heads * self.head_dim)
+        else:  # decode: one token per sequence
+            k_cache, v_cache = kv_cache
+            ctx = ops.attention_decode(
+                q, k_cache, v_cache, attn_ctx["block_table"],
+                attn_ctx["seq_lens"], scale=scale)
+            ctx = ctx.view(t, self.heads * self.head_dim)
+        attn_out = self.o_proj(ctx)
+        if self.tp_size > 1:
+            from ..parallel import tp as tp_mod
+
+            tp_mod.maybe_all_reduce(attn_out)  # row-parallel o_proj
+
+        x = ops.rmsnorm(attn_out, self.mlp_norm, cfg.rms_eps,
+                        residual=residual)
+        gate_up = self.gate_up(x)
+        gate, up = gate_up.split([self.inter, self.inter], dim=-1)
+        mlp_out = self.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
+        if self.tp_size > 1:
+            from ..parallel import tp as tp_mod
+
+            tp_mod.maybe_all_reduce(mlp_out)  # row-parallel down projection
+        return mlp_out  # residual carries the stream
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_size: int = 1):
+        super().__init__()
+        self.cfg = cfg
+        self.tp_rank, self.tp_size = tp_rank, tp_size
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.layers = nn.ModuleList(
+            [LlamaLayer(cfg, tp_rank, tp_size) for _ in range(cfg.layers)])
+        self.final_norm = nn.Parameter(torch.ones(cfg.hidden))
+        # TP: lm_head column-parallel over vocab (shards all-gathered)
+        assert cfg.vocab_size % tp_size == 0
+        self.lm_head = nn.Linear(cfg.hidden, cfg.vocab_size // tp_size,
+                                 bias=False)
+        if cfg.tie_embeddings and tp_size == 1:
+            self.lm_head.weight = self.embed.weight
+        self._init_weights()
+
+    def _init_weights(self):
+        std = 0.02
+        for m in self.modules():
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, std=std)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, std=std)
+
+    def forward(
+        self,
+        tokens: torch.Tensor,      # [T] flat token ids
+        positions: torch.Tensor,   # [T] int32
+        kv_caches: Optional[List] = None,  # per layer (k_cache, v_cache)
+        attn_ctx: Optional[Dict] = None,
+        last_token_idx: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Returns logits [n_seqs, vocab] at the selected token positions."""
+        x = self.embed(tokens.long())
+        residual = torch.zeros_like(x)
+        for i, layer in enumerate(self.layers):
+            cache = kv_caches[i] if kv_caches is not None else None
+            x = layer(x, residual, positions, cache, attn_ctx)
+        x = ops.rmsnorm(x, self.final_norm, self.cfg.rms_eps,
+                        residual=residual)
+        if last_token_idx is not None:
+            x = x[last_token_idx.long()]
+        logits = self.lm_head(x)
+        if self.tp_size > 1:
+            # TP: lm_head is column-sharded -> all-gather the vocab shards
+            from ..parallel import tp as tp_mod
+
+            logits = tp_mod.gather_logits(logits)
+        return logits
+
+
+PRESETS = {
+    "llama-3-8b": LlamaConfig(),
+    "llama-3-1b": LlamaConfig(hidden=2048, layers=16, heads=32, kv_heads=8,
+                              intermediate=8192, tie_embeddings=True),
+    "llama-tiny": LlamaConfig(vocab_size=512, hidden=256, layers=2, heads=4,
+                              kv_heads=2, intermediate=512, rope_theta=10000.0,
+                              max_position=512),
+}
+
+
+@register_arch("llama")
+def llama(preset: str = "llama-3-8b", **overrides) -> LlamaForCausalLM:
+    cfg_base = PRESETS[preset]
+    cfg = LlamaConfig(**{**cfg_base.__dict__, **overrides})
+    return LlamaForCausalLM(cfg)

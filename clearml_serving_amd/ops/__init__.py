@@ -244,6 +244,27 @@ def attention_decode(
     return out
 
 
+def kv_cache_write(
+    k_new: torch.Tensor, v_new: torch.Tensor, k_cache: torch.Tensor,
+    v_cache: torch.Tensor, slot_mapping: torch.Tensor,
+) -> None:
+    """Scatter [T, H_kv, D] new keys/values into the paged caches at flat
+    slots (block_id * block_size + offset); slot -1 skips the token."""
+    if k_new.is_cuda:
+        ext = _require_ext("kv_cache_write")
+        if ext is not None:
+            ext.kv_cache_write(k_new, v_new, k_cache, v_cache, slot_mapping)
+            return
+    block_size = k_cache.shape[2]
+    for t in range(k_new.shape[0]):
+        slot = int(slot_mapping[t])
+        if slot < 0:
+            continue
+        blk, off = slot // block_size, slot % block_size
+        k_cache[blk, :, off] = k_new[t]
+        v_cache[blk, :, off] = v_new[t]
+
+
 # --------------------------------------------------------------------- #
 # Rotary embedding
 # --------------------------------------------------------------------- #

@@ -1,0 +1,293 @@
+#include "hip/hip_runtime.h"
+// Paged-KV decode attention + KV-cache scatter for gfx950.
+//
+// The decode step of the native LLM engine (replaces the work the reference
+// delegates to vLLM): one query token per sequence against a
+// paged KV cache.
+//
+// attention_decode:
+//   grid = (B, Hkv); block = 4 waves.
+//   Each workgroup serves ALL GQ = H/Hkv query heads of one kv head, so K/V
+//   rows stream from HBM once per kv head (not once per q head).
+//   Within a wave, keys are processed 4 at a time by 16-lane groups: each
+//   lane holds an 8-element slice of the 128-wide row (one bf16x8 = 16 B
+//   load, coalesced across the group). Scores reduce over the group with 4
+//   shfl_xor steps; the online-softmax state (m, l) is wave-uniform per
+//   q head; per-lane O accumulators merge across groups (shfl) and waves
+//   (LDS) at the end -- flash-decoding style combine.
+//
+// kv_cache_write: scatter [T, Hkv, D] new keys/values into
+//   [num_blocks, Hkv, block_size, D] caches at slot_mapping[t].
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
+
+__device__ __forceinline__ float wave_max_all(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+__device__ __forceinline__ float wave_sum_all(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <int D, int GQ>
+__global__ __launch_bounds__(256, 2) void attn_decode_kernel(
+    const __hip_bfloat16* __restrict__ q,        // [B, H, D]
+    const __hip_bfloat16* __restrict__ k_cache,  // [NB, Hkv, BS, D]
+    const __hip_bfloat16* __restrict__ v_cache,  // [NB, Hkv, BS, D]
+    const int* __restrict__ block_table,         // [B, max_blocks]
+    const int* __restrict__ seq_lens,            // [B]
+    __hip_bfloat16* __restrict__ out,            // [B, H, D]
+    int H, int Hkv, int block_size, int max_blocks, float scale) {
+  constexpr int NW = 4;        // waves
+  constexpr int GROUPS = 4;    // 16-lane key groups per wave
+  constexpr int EPL = D / 16;  // elements per lane (8 for D=128)
+
+  const int b = blockIdx.x;
+  const int hkv = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int group = lane >> 4;        // 0..3
+  const int sub = lane & 15;          // d-slice owner
+  const int n_keys = seq_lens[b];
+
+  // q fragments for the GQ query heads sharing this kv head
+  float qf[GQ][EPL];
+#pragma unroll
+  for (int g = 0; g < GQ; ++g) {
+    const int h = hkv * GQ + g;
+    const __hip_bfloat16* qrow = q + ((long)b * H + h) * D + sub * EPL;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) qf[g][e] = to_f32(qrow[e]) * scale;
+  }
+
+  float m_run[GQ], l_run[GQ], o_acc[GQ][EPL];
+#pragma unroll
+  for (int g = 0; g < GQ; ++g) {
+    m_run[g] = -INFINITY;
+    l_run[g] = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) o_acc[g][e] = 0.f;
+  }
+
+  const int* btab = block_table + (long)b * max_blocks;
+  const long hk_off = (long)hkv * block_size * D;
+
+  // keys processed 16 per workgroup iteration: wave w group g -> key
+  // it*16 + w*4 + g
+  const int per_iter = NW * GROUPS;
+  const int n_iters = (n_keys + per_iter - 1) / per_iter;
+  for (int it = 0; it < n_iters; ++it) {
+    const int key = it * per_iter + wave * GROUPS + group;
+    const bool valid = key < n_keys;
+    float score[GQ];
+    const __hip_bfloat16* krow = nullptr;
+    const __hip_bfloat16* vrow = nullptr;
+    if (valid) {
+      const int blk = btab[key / block_size];
+      const long base = ((long)blk * Hkv) * block_size * D + hk_off +
+                        (long)(key % block_size) * D;
+      krow = k_cache + base;
+      vrow = v_cache + base;  // same layout
+      bf16x8 kv;
+      kv.u = *reinterpret_cast<const uint32x4*>(krow + sub * EPL);
+#pragma unroll
+      for (int g = 0; g < GQ; ++g) {
+        float acc = 0.f;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e) acc += qf[g][e] * to_f32(kv.h[e]);
+        score[g] = acc;
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < GQ; ++g) score[g] = -INFINITY;
+    }
+    // group-level dot reduction (16 lanes hold partial sums)
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) {
+        float other = __shfl_xor(score[g], off, 64);
+        score[g] = valid ? score[g] + other : -INFINITY;
+      }
+    }
+    // online update per q head: wave-wide max over this iteration's 4 keys
+    bf16x8 vv;
+    if (valid) vv.u = *reinterpret_cast<const uint32x4*>(vrow + sub * EPL);
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+      const float it_max = wave_max_all(valid ? score[g] : -INFINITY);
+      if (it_max == -INFINITY) continue;
+      const float m_new = fmaxf(m_run[g], it_max);
+      const float alpha =
+          (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+      const float p = valid ? __expf(score[g] - m_new) : 0.f;
+      // each group's 16 lanes hold the same p -> sum/16 = sum over keys
+      l_run[g] = l_run[g] * alpha + wave_sum_all(p) * (1.0f / 16.0f);
+      m_run[g] = m_new;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) {
+        const float vval = valid ? to_f32(vv.h[e]) : 0.f;
+        o_acc[g][e] = o_acc[g][e] * alpha + p * vval;
+      }
+    }
+  }
+
+  // merge the 4 key groups within each wave: lanes with equal `sub` but
+  // different `group` hold partial O for the same d-slice (m/l are already
+  // wave-uniform)
+#pragma unroll
+  for (int g = 0; g < GQ; ++g) {
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+      o_acc[g][e] += __shfl_xor(o_acc[g][e], 16, 64);
+      o_acc[g][e] += __shfl_xor(o_acc[g][e], 32, 64);
+    }
+  }
+
+  // merge the 4 waves through LDS (flash-decoding combine)
+  __shared__ float lds_m[NW][GQ], lds_l[NW][GQ];
+  __shared__ float lds_o[NW][GQ][D];
+  if (group == 0) {  // one group per wave writes its state
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+      if (sub == 0) { lds_m[wave][g] = m_run[g]; lds_l[wave][g] = l_run[g]; }
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) lds_o[wave][g][sub * EPL + e] = o_acc[g][e];
+    }
+  }
+  __syncthreads();
+  if (wave == 0 && group == 0) {
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+      float m_tot = -INFINITY;
+      for (int w = 0; w < NW; ++w) m_tot = fmaxf(m_tot, lds_m[w][g]);
+      float l_tot = 0.f;
+      float o_tot[EPL];
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) o_tot[e] = 0.f;
+      for (int w = 0; w < NW; ++w) {
+        if (lds_m[w][g] == -INFINITY) continue;
+        const float f = __expf(lds_m[w][g] - m_tot);
+        l_tot += lds_l[w][g] * f;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e)
+          o_tot[e] += lds_o[w][g][sub * EPL + e] * f;
+      }
+      const int h = hkv * GQ + g;
+      __hip_bfloat16* dst = out + ((long)b * H + h) * D + sub * EPL;
+      const float inv_l = l_tot > 0.f ? 1.0f / l_tot : 0.f;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) dst[e] = __float2bfloat16(o_tot[e] * inv_l);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------- //
+template <typename T>
+__global__ void kv_cache_write_kernel(const T* __restrict__ knew,  // [T,Hkv,D]
+                                      const T* __restrict__ vnew,
+                                      T* __restrict__ k_cache,  // [NB,Hkv,BS,D]
+                                      T* __restrict__ v_cache,
+                                      const int* __restrict__ slots,  // [T]
+                                      int hkv, int d, int block_size) {
+  const int t = blockIdx.x;
+  const int slot = slots[t];
+  if (slot < 0) return;
+  const long blk = slot / block_size;
+  const int off = slot % block_size;
+  for (int i = threadIdx.x; i < hkv * d; i += blockDim.x) {
+    const int h = i / d;
+    const int dd = i % d;
+    const long dst =
+        ((blk * hkv + h) * block_size + off) * (long)d + dd;
+    k_cache[dst] = knew[((long)t * hkv + h) * d + dd];
+    v_cache[dst] = vnew[((long)t * hkv + h) * d + dd];
+  }
+}
+
+}  // namespace
+
+torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
+                               torch::Tensor v_cache,
+                               torch::Tensor block_table,
+                               torch::Tensor seq_lens, double scale) {
+  TORCH_CHECK(q.dim() == 3, "q must be [B, H, D]");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attention: bf16 only");
+  TORCH_CHECK(q.is_contiguous() && k_cache.is_contiguous() &&
+              v_cache.is_contiguous());
+  const int B = q.size(0), H = q.size(1), D = q.size(2);
+  const int Hkv = k_cache.size(1), BS = k_cache.size(2);
+  const int max_blocks = block_table.size(1);
+  const int GQ = H / Hkv;
+  TORCH_CHECK(H % Hkv == 0 && (D == 64 || D == 128));
+  TORCH_CHECK(GQ == 1 || GQ == 2 || GQ == 4 || GQ == 8,
+              "GQA group must be 1/2/4/8, got ", GQ);
+  auto bt = block_table.to(q.device(), at::kInt).contiguous();
+  auto sl = seq_lens.to(q.device(), at::kInt).contiguous();
+  auto out = torch::empty_like(q);
+  dim3 grid(B, Hkv);
+  hipStream_t stream_ = cmls::current_stream();
+
+#define LAUNCH_DEC(DD, GG)                                                   \
+  hipLaunchKernelGGL((attn_decode_kernel<DD, GG>), grid, dim3(256), 0,       \
+                     stream_, (const __hip_bfloat16*)q.data_ptr(),           \
+                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
+                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
+                     bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
+                     (__hip_bfloat16*)out.data_ptr(), H, Hkv, BS,            \
+                     max_blocks, (float)scale)
+  if (D == 128) {
+    if (GQ == 1) LAUNCH_DEC(128, 1);
+    else if (GQ == 2) LAUNCH_DEC(128, 2);
+    else if (GQ == 4) LAUNCH_DEC(128, 4);
+    else LAUNCH_DEC(128, 8);
+  } else {
+    if (GQ == 1) LAUNCH_DEC(64, 1);
+    else if (GQ == 2) LAUNCH_DEC(64, 2);
+    else if (GQ == 4) LAUNCH_DEC(64, 4);
+    else LAUNCH_DEC(64, 8);
+  }
+#undef LAUNCH_DEC
+  return out;
+}
+
+void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
+                    torch::Tensor k_cache, torch::Tensor v_cache,
+                    torch::Tensor slot_mapping) {
+  TORCH_CHECK(knew.dim() == 3, "knew must be [T, Hkv, D]");
+  TORCH_CHECK(knew.is_contiguous() && vnew.is_contiguous() &&
+              k_cache.is_contiguous() && v_cache.is_contiguous());
+  const int T = knew.size(0), Hkv = knew.size(1), D = knew.size(2);
+  const int BS = k_cache.size(2);
+  auto slots = slot_mapping.to(knew.device(), at::kInt).contiguous();
+  TORCH_CHECK(slots.numel() == T);
+  if (T == 0) return;
+  hipStream_t stream_ = cmls::current_stream();
+  const auto st = knew.scalar_type();
+  const int block = std::min(256, Hkv * D);
+  if (st == at::kBFloat16) {
+    hipLaunchKernelGGL(kv_cache_write_kernel<__hip_bfloat16>, dim3(T),
+                       dim3(block), 0, stream_,
+                       (const __hip_bfloat16*)knew.data_ptr(),
+                       (const __hip_bfloat16*)vnew.data_ptr(),
+                       (__hip_bfloat16*)k_cache.data_ptr(),
+                       (__hip_bfloat16*)v_cache.data_ptr(),
+                       slots.data_ptr<int>(), Hkv, D, BS);
+  } else if (st == at::kHalf) {
+    hipLaunchKernelGGL(kv_cache_write_kernel<__half>, dim3(T), dim3(block), 0,
+                       stream_, (const __half*)knew.data_ptr(),
+                       (const __half*)vnew.data_ptr(),
+                       (__half*)k_cache.data_ptr(),
+                       (__half*)v_cache.data_ptr(), slots.data_ptr<int>(),
+                       Hkv, D, BS);
+  } else {
+    TORCH_CHECK(false, "kv_cache_write: unsupported dtype ", st);
+  }
+}

@@ -16,6 +16,13 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
                                 c10::optional<torch::Tensor> seq_lens);
 torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
                                  long top_k, double top_p, long seed);
+torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
+                               torch::Tensor v_cache,
+                               torch::Tensor block_table,
+                               torch::Tensor seq_lens, double scale);
+void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
+                    torch::Tensor k_cache, torch::Tensor v_cache,
+                    torch::Tensor slot_mapping);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   double theta);
 
@@ -36,4 +43,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("seq_lens") = py::none());
   m.def("sample_top_k_top_p", &sample_top_k_top_p);
   m.def("rope_inplace", &rope_inplace);
+  m.def("attention_decode", &attention_decode);
+  m.def("kv_cache_write", &kv_cache_write);
 }

@@ -20,8 +20,10 @@ __global__ void rope_kernel(T* __restrict__ q, T* __restrict__ k,
     for (int i = threadIdx.x; i < half; i += blockDim.x) {
       const float freq = __powf(theta, -(float)i / (float)half);
       const float angle = pos * freq;
+      // precise sincos: fast-math __sincosf loses range reduction past a few
+      // pi and breaks large-position rotations (measured: 30% mismatch)
       float c, s;
-      __sincosf(angle, &s, &c);
+      sincosf(angle, &s, &c);
       const float lo = to_f32(base[i]);
       const float hi_v = to_f32(base[i + half]);
       base[i] = from_f32<T>(lo * c - hi_v * s);

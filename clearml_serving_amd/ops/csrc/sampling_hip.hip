@@ -90,8 +90,10 @@ torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
     auto sl = std::get<0>(sorted);
     auto si = std::get<1>(sorted);
     auto probs = (sl.to(at::kFloat) / temperature).softmax(-1);
-    auto cum = probs.cumsum(-1) - probs;
-    auto cut = cum > top_p;
+    // drop tokens whose inclusive cumulative prob exceeds p; keep the top
+    // token unconditionally (matches the python reference in ops/__init__.py)
+    auto cut = probs.cumsum(-1) > top_p;
+    cut.index_put_({torch::indexing::Slice(), 0}, false);
     auto masked = sl.where(~cut, torch::full({}, -INFINITY, sl.options()));
     filtered = torch::full_like(filtered, -INFINITY)
                    .scatter(-1, si, masked);

@@ -168,6 +168,14 @@ def create_app(
         combined = dict(body) if isinstance(body, dict) else {"body": body}
         combined["request"] = request
         model = combined.get("model") or ""
+        if endpoint_type.rstrip("/") == "v1/models" and not model:
+            # modelless listing: enumerate LLM endpoints
+            proc = state["processor"]
+            eps = proc.get_synced_endpoints() if proc else {}
+            return {"object": "list", "data": [
+                {"id": url, "object": "model", "owned_by": "clearml-serving-amd"}
+                for url, ep in eps.items() if ep.engine_type in ("llm", "vllm")
+            ]}
         out = await process_with_exceptions(
             base_url=model, version=None,
             request_body=combined, serve_type=endpoint_type,

@@ -1,0 +1,139 @@
+"""GPU tests: paged decode attention numerics + end-to-end LLM generation."""
+
+import asyncio
+
+import pytest
+import torch
+
+import clearml_serving_amd.ops as ops
+from clearml_serving_amd.engines.llm.engine import (
+    LlmEngine,
+    LlmEngineConfig,
+    SamplingParams,
+)
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def run(coro):
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(coro)
+    finally:
+        loop.close()
+
+
+@pytest.mark.parametrize("b,h,hkv,d,seqs", [
+    (4, 32, 8, 128, [1, 17, 200, 1000]),
+    (2, 8, 8, 64, [33, 64]),
+    (3, 16, 2, 128, [5, 130, 257]),
+])
+def test_attention_decode_numerics(b, h, hkv, d, seqs):
+    torch.manual_seed(0)
+    block_size = 16
+    max_blocks = (max(seqs) + block_size - 1) // block_size
+    nblocks = b * max_blocks + 1
+    k_cache = torch.randn(nblocks, hkv, block_size, d, device=DEV,
+                          dtype=torch.bfloat16)
+    v_cache = torch.randn(nblocks, hkv, block_size, d, device=DEV,
+                          dtype=torch.bfloat16)
+    # shuffled physical block assignment
+    perm = torch.randperm(nblocks - 1) + 1
+    block_table = perm[:b * max_blocks].reshape(b, max_blocks).to(
+        torch.int32).to(DEV)
+    seq_lens = torch.tensor(seqs, dtype=torch.int32, device=DEV)
+    q = torch.randn(b, h, d, device=DEV, dtype=torch.bfloat16)
+
+    got = ops.attention_decode(q, k_cache, v_cache, block_table, seq_lens)
+    ref = ops.attention_decode(
+        q.float().cpu(), k_cache.float().cpu(), v_cache.float().cpu(),
+        block_table.cpu(), seq_lens.cpu())
+    torch.testing.assert_close(got.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_kv_cache_write_scatter():
+    t, hkv, d, bs = 7, 4, 128, 16
+    k_new = torch.randn(t, hkv, d, device=DEV, dtype=torch.bfloat16)
+    v_new = torch.randn(t, hkv, d, device=DEV, dtype=torch.bfloat16)
+    k_cache = torch.zeros(4, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    v_cache = torch.zeros(4, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    slots = torch.tensor([0, 1, 15, 16, 33, 63, -1], dtype=torch.int32,
+                         device=DEV)
+    ops.kv_cache_write(k_new, v_new, k_cache, v_cache, slots)
+    torch.testing.assert_close(k_cache[0, :, 0], k_new[0])
+    torch.testing.assert_close(k_cache[0, :, 1], k_new[1])
+    torch.testing.assert_close(k_cache[0, :, 15], k_new[2])
+    torch.testing.assert_close(k_cache[1, :, 0], k_new[3])
+    torch.testing.assert_close(v_cache[2, :, 1], v_new[4])
+    torch.testing.assert_close(v_cache[3, :, 15], v_new[5])
+    assert k_cache[3, :, 14].abs().sum() == 0  # slot -1 skipped
+
+
+def test_llm_engine_gpu_generation():
+    cfg = LlmEngineConfig(preset="llama-3-1b", num_kv_blocks=2048,
+                          block_size=16, max_model_len=1024, device=DEV)
+    eng = LlmEngine(cfg)
+    eng.start()
+
+    async def main():
+        params = SamplingParams(temperature=0.0, max_tokens=16,
+                                ignore_eos=True)
+
+        async def one(text):
+            toks = []
+            async for item in eng.generate(text, params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return await asyncio.gather(*[one("prompt %d" % i) for i in range(4)])
+
+    outs = run(main())
+    assert all(len(o) == 16 for o in outs)
+    # determinism under greedy on GPU
+    outs2 = run(main())
+    assert outs == outs2
+    assert eng.allocator.available == eng.allocator.num_blocks
+
+
+def test_llm_engine_gpu_decode_matches_prefill():
+    """GPU paged decode vs teacher-forced prefill on the same weights."""
+    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=256,
+                          block_size=16, max_model_len=256, device=DEV)
+    eng = LlmEngine(cfg)
+    eng.start()
+    prompt = [3, 7, 11, 19, 23, 29, 31, 37]
+
+    async def gen():
+        seq = await eng.add_request(
+            list(prompt), SamplingParams(temperature=0.0, max_tokens=8,
+                                         ignore_eos=True))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    generated = run(gen())
+    full = prompt + generated
+    t = len(full)
+    tokens = torch.tensor(full, dtype=torch.long, device=DEV)
+    positions = torch.arange(t, dtype=torch.int32, device=DEV)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": t,
+                "seq_lens": torch.tensor([t], dtype=torch.int32, device=DEV),
+                "slot_mapping": torch.full((t,), -1, dtype=torch.int32,
+                                           device=DEV)}
+    with torch.inference_mode():
+        logits = eng.model(tokens, positions, kv_caches=None,
+                           attn_ctx=attn_ctx)
+    # bf16 nondeterminism tolerance: compare argmax agreement, allowing ties
+    agree = 0
+    for step in range(8):
+        pos = len(prompt) + step - 1
+        top2 = logits[pos].float().topk(2)
+        if generated[step] == int(top2.indices[0]) or \
+                (top2.values[0] - top2.values[1]) < 0.05:
+            agree += 1
+    assert agree >= 7, "paged decode diverges from teacher-forced prefill"
