@@ -52,6 +52,12 @@ def build_processor(device_idx, tmpdir, use_graphs=True):
         "bert_bench": {"arch": "bert-base", "num_labels": 2,
                        "dtype": "bfloat16"},
     }
+    specs = {
+        "resnet50_bench": dict(input_size=[3, 224, 224], input_type="float32"),
+        "bert_bench": dict(input_size=[[128], [128]],
+                           input_type=["int64", "int32"],
+                           input_name=["input_ids", "attention_mask"]),
+    }
     for name, card in cards.items():
         p = os.path.join(tmpdir, name + ".json")
         with open(p, "wt") as f:
@@ -66,6 +72,7 @@ def build_processor(device_idx, tmpdir, use_graphs=True):
                 "use_graphs": use_graphs,
                 "gpu": device_idx,
             },
+            **specs[name],
         ))
     return processor
 

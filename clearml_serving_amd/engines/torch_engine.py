@@ -22,7 +22,6 @@ Model resolution, per endpoint:
     use_graphs (default true), gpu (device index), input_format ("dict")
 """
 
-import json
 import os
 from typing import Any, Dict, Optional
 
@@ -79,6 +78,39 @@ class HipPreprocessRequest(BasePreprocessRequest):
             dtype=self.dtype,
             name=model_endpoint.serving_url,
         )
+        if aux.get("warmup", True):
+            sample = self._sample_from_spec()
+            if sample is not None:
+                self._batcher.warmup(sample)
+
+    def _sample_from_spec(self) -> Optional[Any]:
+        """Build a zero request from the endpoint I/O spec so every batch
+        bucket's hipGraph is captured before traffic arrives. Dynamic (-1)
+        dims skip warmup (capture then happens lazily per shape)."""
+        ep = self.model_endpoint
+        if not ep.input_size or not ep.input_type:
+            return None
+        if any(d is None or int(d) < 0 for shape in ep.input_size
+               for d in shape):
+            return None
+        tensors = []
+        for shape, t in zip(ep.input_size, ep.input_type):
+            np_dtype = np.dtype(t or "float32")
+            tensors.append(torch.zeros([int(d) for d in shape],
+                                       dtype=self._np_to_torch(np_dtype)))
+        if ep.input_name and len(ep.input_name) == len(tensors):
+            return dict(zip(ep.input_name, tensors))
+        return tensors[0] if len(tensors) == 1 else None
+
+    @staticmethod
+    def _np_to_torch(np_dtype) -> torch.dtype:
+        return {
+            np.dtype("float32"): torch.float32,
+            np.dtype("float16"): torch.float16,
+            np.dtype("int64"): torch.int64,
+            np.dtype("int32"): torch.int32,
+            np.dtype("uint8"): torch.uint8,
+        }.get(np_dtype, torch.float32)
 
     # ------------------------------------------------------------------ #
     def _load_model(self):

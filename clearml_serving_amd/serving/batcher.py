@@ -128,6 +128,14 @@ class DynamicBatcher:
                     if not fut.done():
                         fut.set_exception(ex)
 
+    def warmup(self, sample: TensorOrDict) -> None:
+        """Pre-capture every bucket's hipGraph from one sample request so no
+        capture happens on the serving path (call at endpoint creation)."""
+        if not self.use_graphs:
+            return
+        for bucket in self.buckets:
+            self._execute([sample] * bucket)
+
     # ------------------------------------------------------------------ #
     def _execute(self, inputs: List[TensorOrDict]) -> List[TensorOrDict]:
         """Assemble the batch, run on the batcher's HIP stream, slice back."""
