@@ -36,10 +36,15 @@ def main():
         # seq_lens control: mask to half length
         sl = torch.full((b,), s // 2, dtype=torch.int32, device="cuda")
         masked = t(lambda: ops.attention(q, k, v, seq_lens=sl))
+        # per-block fixed-cost control: 64-key kv (1 tile per block)
+        k1 = k[:, :, :64].contiguous()
+        v1 = v[:, :, :64].contiguous()
+        kv64 = t(lambda: ops.attention(q, k1, v1))
         print("s={:5d}: full {:8.1f}us  causal {:8.1f}us ({:.2f}x)  "
-              "half-kv {:8.1f}us ({:.2f}x)  seqlen-half {:8.1f}us ({:.2f}x)"
+              "half-kv {:8.1f}us ({:.2f}x)  seqlen-half {:8.1f}us ({:.2f}x)  "
+              "kv64 {:8.1f}us ({:.3f}x)"
               .format(s, full, causal, causal / full, half, half / full,
-                      masked, masked / full))
+                      masked, masked / full, kv64, kv64 / full))
 
 
 if __name__ == "__main__":
