@@ -287,7 +287,9 @@ def rope_inplace(
     cos = angles.cos()[:, None, :]
     sin = angles.sin()[:, None, :]
     for t in (q, k):
-        tf = t.float()
+        # clone: .float() on an fp32 tensor returns SELF, and lo/hi must not
+        # alias the destination (in-place write of :half corrupted hi's input)
+        tf = t.float().clone()
         lo, hi = tf[..., :half], tf[..., half:]
         t[..., :half] = (lo * cos - hi * sin).to(t.dtype)
         t[..., half:] = (hi * cos + lo * sin).to(t.dtype)

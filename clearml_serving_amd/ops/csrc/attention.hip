@@ -67,8 +67,12 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   __shared__ short lds_vt[D * VSTRIDE];
   __shared__ short lds_p[NWAVES * 16 * PSTRIDE];
 
-  const int m_tile = blockIdx.x;
-  const int bh = blockIdx.y;
+  // grid = (B*H, n_m_tiles): bh on x so the dispatcher's XCD round-robin
+  // (XCD = linear_id % 8) decorrelates from the causal work skew, and heavy
+  // (high-m) tiles dispatch FIRST so light blocks backfill the tail --
+  // with m_tile on x, causal ran at full-attention wall time (measured 2x).
+  const int bh = blockIdx.x;
+  const int m_tile = (int)gridDim.y - 1 - (int)blockIdx.y;
   const int b = bh / H;
   const int h = bh % H;
   const int hkv = h / (H / Hkv);
@@ -125,14 +129,19 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
               k + kv_base + (long)gkey * D + d8);
         }
         *reinterpret_cast<bf16x8_t*>(&lds_k[key * KSTRIDE + d8]) = val;
-        // V: same piece indexing, scatter-transposed (8 ds_write_b16)
+        // V: same piece indexing, scatter-transposed (8 ds_write_b16).
+        // key-XOR swizzle: unswizzled, the 16 lanes sharing a key write
+        // addresses 8*VSTRIDE*2 = 1152 B apart = all on one bank (16-way
+        // conflict, measured 64% of kernel wall); XOR key bits 3-5 with the
+        // d-row block spreads them over 8 banks (2-way).
         bf16x8_t vv{0, 0, 0, 0, 0, 0, 0, 0};
         if (gkey < kv_len) {
           vv = *reinterpret_cast<const bf16x8_t*>(
               v + kv_base + (long)gkey * D + d8);
         }
+        const int kswz = key ^ (((d8 >> 3) & 7) << 3);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) lds_vt[(d8 + e) * VSTRIDE + key] = vv[e];
+        for (int e = 0; e < 8; ++e) lds_vt[(d8 + e) * VSTRIDE + kswz] = vv[e];
       }
     }
     __syncthreads();
@@ -220,9 +229,13 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
           &pslab[frag_row * PSTRIDE + kc * 32 + frag_ko]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
-        // B[k][j] = V[k + 32kc][j + 16dt] = vt[j + 16dt][k]: contiguous in k
+        // B[k][j] = V[k + 32kc][j + 16dt] = vt[j + 16dt][k]: contiguous in k;
+        // apply the staging key-XOR (8-aligned base ^ bits 3-5 stays
+        // 16-B aligned)
+        const int vrow = dt * 16 + frag_row;
+        const int vkey = (kc * 32 + frag_ko) ^ (((vrow >> 3) & 7) << 3);
         const bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-            &lds_vt[(dt * 16 + frag_row) * VSTRIDE + kc * 32 + frag_ko]);
+            &lds_vt[vrow * VSTRIDE + vkey]);
         acc_o[dt] = MFMA_16x16x32(pf, vf, acc_o[dt]);
       }
     }
@@ -268,7 +281,7 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
     TORCH_CHECK(sl_t.numel() == B, "seq_lens must be [B]");
     sl = sl_t.data_ptr<int>();
   }
-  dim3 grid((Sq + BLOCK_M - 1) / BLOCK_M, B * H);
+  dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
   dim3 block(256);
   hipStream_t stream_ = cmls::current_stream();
 
