@@ -23,14 +23,16 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 
-__device__ __forceinline__ float wave_max_all(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+// values uniform within each 16-lane group: combining the 4 groups of a
+// wave needs only xor 16 and 32
+__device__ __forceinline__ float group_max4(float v) {
+  v = fmaxf(v, __shfl_xor(v, 16, 64));
+  v = fmaxf(v, __shfl_xor(v, 32, 64));
   return v;
 }
-__device__ __forceinline__ float wave_sum_all(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+__device__ __forceinline__ float group_sum4(float v) {
+  v += __shfl_xor(v, 16, 64);
+  v += __shfl_xor(v, 32, 64);
   return v;
 }
 
@@ -42,6 +44,9 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     const int* __restrict__ block_table,         // [B, max_blocks]
     const int* __restrict__ seq_lens,            // [B]
     __hip_bfloat16* __restrict__ out,            // [B, H, D]
+    float* __restrict__ partial_o,               // [B, H, SPLITS, D] f32
+    float* __restrict__ partial_ml,              // [B, H, SPLITS, 2] f32
+    int splits,
     int H, int Hkv, int block_size, int max_blocks, float scale) {
   constexpr int NW = 4;        // waves
   constexpr int GROUPS = 4;    // 16-lane key groups per wave
@@ -54,7 +59,22 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   const int wave = tid >> 6;
   const int group = lane >> 4;        // 0..3
   const int sub = lane & 15;          // d-slice owner
-  const int n_keys = seq_lens[b];
+  const int split = blockIdx.z;
+  const int total_keys = seq_lens[b];
+  // chunk rounded to the 16-key workgroup iteration so splits don't overlap
+  const int chunk = ((total_keys + splits - 1) / splits + 15) & ~15;
+  const int key_lo = split * chunk;
+  const int n_keys = min(total_keys, key_lo + chunk);
+  if (key_lo >= total_keys && split > 0) {
+    // empty split: publish a neutral partial (m=-inf skips it in combine)
+    if (threadIdx.x < 2 * GQ && partial_ml) {
+      const int g = threadIdx.x / 2;
+      const int h = hkv * GQ + g;
+      float* ml = partial_ml + (((long)b * H + h) * splits + split) * 2;
+      ml[threadIdx.x & 1] = (threadIdx.x & 1) ? 0.f : -INFINITY;
+    }
+    return;
+  }
 
   // q fragments for the GQ query heads sharing this kv head
   float qf[GQ][EPL];
@@ -81,9 +101,9 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   // keys processed 16 per workgroup iteration: wave w group g -> key
   // it*16 + w*4 + g
   const int per_iter = NW * GROUPS;
-  const int n_iters = (n_keys + per_iter - 1) / per_iter;
+  const int n_iters = (n_keys - key_lo + per_iter - 1) / per_iter;
   for (int it = 0; it < n_iters; ++it) {
-    const int key = it * per_iter + wave * GROUPS + group;
+    const int key = key_lo + it * per_iter + wave * GROUPS + group;
     const bool valid = key < n_keys;
     float score[GQ];
     const __hip_bfloat16* krow = nullptr;
@@ -121,14 +141,13 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     if (valid) vv.u = *reinterpret_cast<const uint32x4*>(vrow + sub * EPL);
 #pragma unroll
     for (int g = 0; g < GQ; ++g) {
-      const float it_max = wave_max_all(valid ? score[g] : -INFINITY);
+      const float it_max = group_max4(valid ? score[g] : -INFINITY);
       if (it_max == -INFINITY) continue;
       const float m_new = fmaxf(m_run[g], it_max);
       const float alpha =
           (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
       const float p = valid ? __expf(score[g] - m_new) : 0.f;
-      // each group's 16 lanes hold the same p -> sum/16 = sum over keys
-      l_run[g] = l_run[g] * alpha + wave_sum_all(p) * (1.0f / 16.0f);
+      l_run[g] = l_run[g] * alpha + group_sum4(p);
       m_run[g] = m_new;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) {
@@ -180,11 +199,63 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
           o_tot[e] += lds_o[w][g][sub * EPL + e] * f;
       }
       const int h = hkv * GQ + g;
-      __hip_bfloat16* dst = out + ((long)b * H + h) * D + sub * EPL;
-      const float inv_l = l_tot > 0.f ? 1.0f / l_tot : 0.f;
+      if (splits == 1) {
+        __hip_bfloat16* dst = out + ((long)b * H + h) * D + sub * EPL;
+        const float inv_l = l_tot > 0.f ? 1.0f / l_tot : 0.f;
 #pragma unroll
-      for (int e = 0; e < EPL; ++e) dst[e] = __float2bfloat16(o_tot[e] * inv_l);
+        for (int e = 0; e < EPL; ++e)
+          dst[e] = __float2bfloat16(o_tot[e] * inv_l);
+      } else {
+        // flash-decoding: publish unnormalized partial + (m, l)
+        float* po = partial_o + (((long)b * H + h) * splits + split) * D +
+                    sub * EPL;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e) po[e] = o_tot[e];
+        if (sub == 0) {
+          float* ml = partial_ml + (((long)b * H + h) * splits + split) * 2;
+          ml[0] = m_tot;
+          ml[1] = l_tot;
+        }
+      }
     }
+  }
+}
+
+// combine the per-split partials: grid (B, H), one wave per block
+template <int D>
+__global__ void decode_combine_kernel(
+    const float* __restrict__ partial_o,   // [B, H, SPLITS, D]
+    const float* __restrict__ partial_ml,  // [B, H, SPLITS, 2]
+    __hip_bfloat16* __restrict__ out,      // [B, H, D]
+    int splits, int H) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  const long base = ((long)b * H + h) * splits;
+  float m_tot = -INFINITY;
+  for (int s = 0; s < splits; ++s)
+    m_tot = fmaxf(m_tot, partial_ml[(base + s) * 2]);
+  float l_tot = 0.f;
+  float acc[(D + 63) / 64];
+#pragma unroll
+  for (int i = 0; i < (D + 63) / 64; ++i) acc[i] = 0.f;
+  for (int s = 0; s < splits; ++s) {
+    const float m = partial_ml[(base + s) * 2];
+    if (m == -INFINITY) continue;
+    const float f = __expf(m - m_tot);
+    l_tot += partial_ml[(base + s) * 2 + 1] * f;
+    const float* po = partial_o + (base + s) * D;
+#pragma unroll
+    for (int i = 0; i < (D + 63) / 64; ++i) {
+      const int d = i * 64 + threadIdx.x;
+      if (d < D) acc[i] += po[d] * f;
+    }
+  }
+  const float inv_l = l_tot > 0.f ? 1.0f / l_tot : 0.f;
+  __hip_bfloat16* dst = out + ((long)b * H + h) * D;
+#pragma unroll
+  for (int i = 0; i < (D + 63) / 64; ++i) {
+    const int d = i * 64 + threadIdx.x;
+    if (d < D) dst[d] = __float2bfloat16(acc[i] * inv_l);
   }
 }
 
@@ -231,8 +302,26 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
   auto bt = block_table.to(q.device(), at::kInt).contiguous();
   auto sl = seq_lens.to(q.device(), at::kInt).contiguous();
   auto out = torch::empty_like(q);
-  dim3 grid(B, Hkv);
   hipStream_t stream_ = cmls::current_stream();
+
+  // flash-decoding split count: B*Hkv workgroups alone under-fill 256 CUs at
+  // serving batch sizes; split the key range so ~4 workgroups land per CU
+  int splits = 1;
+  const long base_wg = (long)B * Hkv;
+  if (base_wg < 1024) {
+    splits = (int)std::min<long>((1024 + base_wg - 1) / base_wg, 16);
+  }
+  torch::Tensor partial_o, partial_ml;
+  float* po = nullptr;
+  float* pml = nullptr;
+  if (splits > 1) {
+    auto opts = q.options().dtype(at::kFloat);
+    partial_o = torch::empty({B, H, splits, D}, opts);
+    partial_ml = torch::empty({B, H, splits, 2}, opts);
+    po = partial_o.data_ptr<float>();
+    pml = partial_ml.data_ptr<float>();
+  }
+  dim3 grid(B, Hkv, splits);
 
 #define LAUNCH_DEC(DD, GG)                                                   \
   hipLaunchKernelGGL((attn_decode_kernel<DD, GG>), grid, dim3(256), 0,       \
@@ -240,8 +329,8 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                      (const __hip_bfloat16*)k_cache.data_ptr(),              \
                      (const __hip_bfloat16*)v_cache.data_ptr(),              \
                      bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
-                     (__hip_bfloat16*)out.data_ptr(), H, Hkv, BS,            \
-                     max_blocks, (float)scale)
+                     (__hip_bfloat16*)out.data_ptr(), po, pml, splits, H,    \
+                     Hkv, BS, max_blocks, (float)scale)
   if (D == 128) {
     if (GQ == 1) LAUNCH_DEC(128, 1);
     else if (GQ == 2) LAUNCH_DEC(128, 2);
@@ -254,6 +343,18 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
     else LAUNCH_DEC(64, 8);
   }
 #undef LAUNCH_DEC
+  if (splits > 1) {
+    dim3 cgrid(B, H);
+    if (D == 128) {
+      hipLaunchKernelGGL(decode_combine_kernel<128>, cgrid, dim3(64), 0,
+                         stream_, po, pml, (__hip_bfloat16*)out.data_ptr(),
+                         splits, H);
+    } else {
+      hipLaunchKernelGGL(decode_combine_kernel<64>, cgrid, dim3(64), 0,
+                         stream_, po, pml, (__hip_bfloat16*)out.data_ptr(),
+                         splits, H);
+    }
+  }
   return out;
 }
 

@@ -7,62 +7,85 @@
 // the C++ wrapper); the hot path (plain sampling, the serving default) is a
 // single kernel.
 //
+// Occupancy: decode batches are small (B <= 64 rows), so the vocabulary is
+// split across gridDim.y workgroups (B workgroups alone leave 3/4 of the 256
+// CUs idle); per-split winners combine through a packed (value, index)
+// 64-bit atomicMax -- deterministic, one extra 8-B atomic per workgroup.
+//
 // Replaces the sampling step the reference delegates to vLLM
 // ("token-sampling kernel (top-k/top-p)").
 #include "common.h"
 
 namespace {
 
-// counter-based RNG: one 32-bit hash per (seed, row, col) -- statistically
-// adequate for sampling draws (not for cryptography).
 __device__ __forceinline__ unsigned int pcg_hash(unsigned int x) {
   x = x * 747796405u + 2891336453u;
   unsigned int w = ((x >> ((x >> 28u) + 4u)) ^ x) * 277803737u;
   return (w >> 22u) ^ w;
 }
 
+// order-preserving float -> uint32 (works for +/-, inf)
+__device__ __forceinline__ unsigned int float_orderable(float f) {
+  unsigned int u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
 template <typename T>
 __global__ void gumbel_argmax_kernel(const T* __restrict__ logits,
-                                     long* __restrict__ out, int rows,
-                                     int vocab, float inv_temp,
+                                     unsigned long long* __restrict__ best,
+                                     int rows, int vocab, float inv_temp,
                                      unsigned int seed) {
-  extern __shared__ float tmp[];
   const int row = blockIdx.x;
   if (row >= rows) return;
   const T* lrow = logits + (long)row * vocab;
+  const int chunk = (vocab + gridDim.y - 1) / gridDim.y;
+  const int lo = blockIdx.y * chunk;
+  const int hi = min(vocab, lo + chunk);
+  const unsigned int row_seed = pcg_hash(seed ^ ((unsigned)row * 9781u));
 
-  float best = -INFINITY;
-  int best_idx = 0;
-  for (int idx = threadIdx.x; idx < vocab; idx += blockDim.x) {
+  float bv = -INFINITY;
+  int bi = lo;
+  for (int idx = lo + threadIdx.x; idx < hi; idx += blockDim.x) {
     const float lg = to_f32(lrow[idx]);
     if (lg == -INFINITY) continue;  // filtered token
-    unsigned int h = pcg_hash(seed ^ pcg_hash((unsigned)row * 9781u + idx));
-    // u in (0, 1): avoid exactly 0
+    const unsigned int h = pcg_hash(row_seed + (unsigned)idx);
     const float u = ((float)h + 1.0f) * (1.0f / 4294967296.0f);
     const float g = -__logf(-__logf(u));
     const float s = lg * inv_temp + g;
-    if (s > best) { best = s; best_idx = idx; }
+    if (s > bv) { bv = s; bi = idx; }
   }
-  // block argmax: pack value+lane, reduce via shfl then LDS
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int nwaves = blockDim.x >> 6;
+  // wave reduce
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
-    float ov = __shfl_xor(best, off, 64);
-    int oi = __shfl_xor(best_idx, off, 64);
-    if (ov > best) { best = ov; best_idx = oi; }
+    const float ov = __shfl_xor(bv, off, 64);
+    const int oi = __shfl_xor(bi, off, 64);
+    if (ov > bv || (ov == bv && oi < bi)) { bv = ov; bi = oi; }
   }
-  float* vals = tmp;
-  int* idxs = (int*)(tmp + 16);
-  if (lane == 0) { vals[wave] = best; idxs[wave] = best_idx; }
+  __shared__ float svals[16];
+  __shared__ int sidx[16];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) { svals[wave] = bv; sidx[wave] = bi; }
   __syncthreads();
   if (threadIdx.x == 0) {
+    const int nwaves = blockDim.x >> 6;
     for (int w = 1; w < nwaves; ++w) {
-      if (vals[w] > best) { best = vals[w]; best_idx = idxs[w]; }
+      if (svals[w] > bv || (svals[w] == bv && sidx[w] < bi)) {
+        bv = svals[w];
+        bi = sidx[w];
+      }
     }
-    out[row] = best_idx;
+    const unsigned long long packed =
+        ((unsigned long long)float_orderable(bv) << 32) |
+        (unsigned int)(vocab - 1 - bi);  // tie -> lowest index wins
+    atomicMax(&best[row], packed);
   }
+}
+
+__global__ void unpack_kernel(const unsigned long long* __restrict__ best,
+                              long* __restrict__ out, int rows, int vocab) {
+  const int row = blockIdx.x * blockDim.x + threadIdx.x;
+  if (row < rows) out[row] = vocab - 1 - (int)(best[row] & 0xFFFFFFFFu);
 }
 
 }  // namespace
@@ -99,30 +122,33 @@ torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
   }
   filtered = filtered.contiguous();
 
+  auto best = torch::zeros(
+      {rows}, logits.options().dtype(at::kLong));  // packed (val, idx)
   auto out = torch::empty({rows}, logits.options().dtype(at::kLong));
+  const int vsplits =
+      std::max(1, std::min(16, 1024 / std::max(rows, 1)));
+  dim3 grid(rows, vsplits);
   const int block = 256;
-  const int smem = 16 * sizeof(float) + 16 * sizeof(int);
   hipStream_t stream_ = cmls::current_stream();
   const auto st = filtered.scalar_type();
   const float inv_t = 1.0f / (float)temperature;
+  auto* bptr = reinterpret_cast<unsigned long long*>(best.data_ptr<long>());
   if (st == at::kBFloat16) {
-    hipLaunchKernelGGL(gumbel_argmax_kernel<__hip_bfloat16>, dim3(rows),
-                       dim3(block), smem, stream_,
-                       (const __hip_bfloat16*)filtered.data_ptr(),
-                       out.data_ptr<long>(), rows, vocab, inv_t,
-                       (unsigned int)seed);
+    hipLaunchKernelGGL(gumbel_argmax_kernel<__hip_bfloat16>, grid,
+                       dim3(block), 0, stream_,
+                       (const __hip_bfloat16*)filtered.data_ptr(), bptr,
+                       rows, vocab, inv_t, (unsigned int)seed);
   } else if (st == at::kHalf) {
-    hipLaunchKernelGGL(gumbel_argmax_kernel<__half>, dim3(rows), dim3(block),
-                       smem, stream_,
-                       (const __half*)filtered.data_ptr(),
-                       out.data_ptr<long>(), rows, vocab, inv_t,
-                       (unsigned int)seed);
+    hipLaunchKernelGGL(gumbel_argmax_kernel<__half>, grid, dim3(block), 0,
+                       stream_, (const __half*)filtered.data_ptr(), bptr,
+                       rows, vocab, inv_t, (unsigned int)seed);
   } else {
     auto f = filtered.to(at::kFloat).contiguous();
-    hipLaunchKernelGGL(gumbel_argmax_kernel<float>, dim3(rows), dim3(block),
-                       smem, stream_, (const float*)f.data_ptr(),
-                       out.data_ptr<long>(), rows, vocab, inv_t,
-                       (unsigned int)seed);
+    hipLaunchKernelGGL(gumbel_argmax_kernel<float>, grid, dim3(block), 0,
+                       stream_, (const float*)f.data_ptr(), bptr, rows,
+                       vocab, inv_t, (unsigned int)seed);
   }
+  hipLaunchKernelGGL(unpack_kernel, dim3((rows + 63) / 64), dim3(64), 0,
+                     stream_, bptr, out.data_ptr<long>(), rows, vocab);
   return out;
 }
