@@ -206,7 +206,15 @@ class LlmEngine:
         mcfg = LlamaConfig(**{**PRESETS[cfg.preset].__dict__,
                               **cfg.overrides})
         self.model_config = mcfg
-        model = LlamaForCausalLM(mcfg)
+        # TP over RCCL/xGMI: one process per GPU (torchrun); every rank
+        # builds its shard, rank 0 owns scheduling + the HTTP front
+        from ...parallel import tp as tp_mod
+
+        self.tp_size = tp_mod.world_size()
+        self.tp_rank = tp_mod.rank()
+        torch.manual_seed(1234)  # identical replicated params across ranks
+        model = LlamaForCausalLM(mcfg, tp_rank=self.tp_rank,
+                                 tp_size=self.tp_size)
         if cfg.weights:
             from ...models import load_weights
 
@@ -216,9 +224,10 @@ class LlmEngine:
         self.tokenizer = (HfTokenizer(cfg.tokenizer_path)
                           if cfg.tokenizer_path else SimpleTokenizer())
 
-        # KV cache sizing from free HBM
+        # KV cache sizing from free HBM (heads sharded under TP)
         bs = cfg.block_size
-        per_block_bytes = (2 * mcfg.kv_heads * bs * mcfg.head_dim *
+        kv_heads = mcfg.kv_heads // self.tp_size
+        per_block_bytes = (2 * kv_heads * bs * mcfg.head_dim *
                            self.dtype.itemsize * mcfg.layers)
         if cfg.num_kv_blocks:
             num_blocks = int(cfg.num_kv_blocks)
@@ -228,11 +237,20 @@ class LlmEngine:
             num_blocks = max(budget // per_block_bytes, 64)
         else:
             num_blocks = 512
+        if self.tp_size > 1:
+            # all ranks must agree on capacity: take the global minimum
+            import torch.distributed as dist
+
+            t = torch.tensor([num_blocks], dtype=torch.long,
+                             device=self.device if self.device.type == "cuda"
+                             else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            num_blocks = int(t.item())
         self.allocator = BlockAllocator(num_blocks)
         self.kv_caches = [
-            (torch.zeros(num_blocks, mcfg.kv_heads, bs, mcfg.head_dim,
+            (torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
                          dtype=self.dtype, device=self.device),
-             torch.zeros(num_blocks, mcfg.kv_heads, bs, mcfg.head_dim,
+             torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
                          dtype=self.dtype, device=self.device))
             for _ in range(mcfg.layers)
         ]
@@ -344,24 +362,36 @@ class LlmEngine:
         bs = self.cfg.block_size
         return seq.blocks[pos // bs] * bs + pos % bs
 
-    @torch.inference_mode()
     def _prefill(self, seqs: List[Sequence]) -> None:
         self.stats["prefill_batches"] += 1
-        bs_cfg = self.cfg.block_size
-        b = len(seqs)
-        lens = [len(s.prompt_ids) for s in seqs]
+        plan = {
+            "mode": "prefill",
+            "prompts": [list(s.prompt_ids) for s in seqs],
+            "slots": [[self._slot(s, p) for p in range(len(s.prompt_ids))]
+                      for s in seqs],
+        }
+        self._tp_broadcast(plan)
+        logits = self._exec_prefill(plan)
+        self.stats["prompt_tokens"] += sum(len(p) for p in plan["prompts"])
+        self._sample_and_emit(seqs, logits)
+
+    @torch.inference_mode()
+    def _exec_prefill(self, plan: Dict[str, Any]) -> torch.Tensor:
+        prompts = plan["prompts"]
+        slots = plan["slots"]
+        b = len(prompts)
+        lens = [len(p) for p in prompts]
         smax = max(lens)
         dev = self.device
 
         tokens = torch.zeros(b, smax, dtype=torch.long)
         positions = torch.zeros(b, smax, dtype=torch.int32)
         slot_map = torch.full((b, smax), -1, dtype=torch.int32)
-        for i, s in enumerate(seqs):
+        for i in range(b):
             n = lens[i]
-            tokens[i, :n] = torch.tensor(s.prompt_ids, dtype=torch.long)
+            tokens[i, :n] = torch.tensor(prompts[i], dtype=torch.long)
             positions[i, :n] = torch.arange(n, dtype=torch.int32)
-            slot_map[i, :n] = torch.tensor(
-                [self._slot(s, p) for p in range(n)], dtype=torch.int32)
+            slot_map[i, :n] = torch.tensor(slots[i], dtype=torch.int32)
         seq_lens = torch.tensor(lens, dtype=torch.int32, device=dev)
         attn_ctx = {
             "mode": "prefill", "batch": b, "seq": smax,
@@ -371,44 +401,86 @@ class LlmEngine:
         last_idx = torch.tensor(
             [i * smax + lens[i] - 1 for i in range(b)], dtype=torch.long,
             device=dev)
-        logits = self.model(
+        return self.model(
             tokens.view(-1).to(dev), positions.view(-1).to(dev),
             kv_caches=self.kv_caches, attn_ctx=attn_ctx,
             last_token_idx=last_idx)
-        self.stats["prompt_tokens"] += sum(lens)
-        self._sample_and_emit(seqs, logits)
 
-    @torch.inference_mode()
     def _decode(self, seqs: List[Sequence]) -> None:
         self.stats["decode_batches"] += 1
         bs_cfg = self.cfg.block_size
-        dev = self.device
-        b = len(seqs)
         # grow block tables for the token being generated
         for s in seqs:
             pos = len(s) - 1  # position of the last generated token
             if pos // bs_cfg >= len(s.blocks):
                 s.blocks.extend(self.allocator.alloc(1))
-        tokens = torch.tensor([s.output_ids[-1] for s in seqs],
-                              dtype=torch.long, device=dev)
-        positions = torch.tensor([len(s) - 1 for s in seqs],
-                                 dtype=torch.int32, device=dev)
-        slot_map = torch.tensor([self._slot(s, len(s) - 1) for s in seqs],
-                                dtype=torch.int32, device=dev)
-        max_blocks = max(len(s.blocks) for s in seqs)
+        plan = {
+            "mode": "decode",
+            "tokens": [s.output_ids[-1] for s in seqs],
+            "positions": [len(s) - 1 for s in seqs],
+            "slots": [self._slot(s, len(s) - 1) for s in seqs],
+            "seq_lens": [len(s) for s in seqs],
+            "blocks": [list(s.blocks) for s in seqs],
+        }
+        self._tp_broadcast(plan)
+        logits = self._exec_decode(plan)
+        self._sample_and_emit(seqs, logits)
+
+    @torch.inference_mode()
+    def _exec_decode(self, plan: Dict[str, Any]) -> torch.Tensor:
+        dev = self.device
+        b = len(plan["tokens"])
+        tokens = torch.tensor(plan["tokens"], dtype=torch.long, device=dev)
+        positions = torch.tensor(plan["positions"], dtype=torch.int32,
+                                 device=dev)
+        slot_map = torch.tensor(plan["slots"], dtype=torch.int32, device=dev)
+        max_blocks = max(len(bl) for bl in plan["blocks"])
         btab = torch.zeros(b, max_blocks, dtype=torch.int32)
-        for i, s in enumerate(seqs):
-            btab[i, :len(s.blocks)] = torch.tensor(s.blocks, dtype=torch.int32)
+        for i, bl in enumerate(plan["blocks"]):
+            btab[i, :len(bl)] = torch.tensor(bl, dtype=torch.int32)
         attn_ctx = {
             "mode": "decode",
-            "seq_lens": torch.tensor([len(s) for s in seqs],
-                                     dtype=torch.int32, device=dev),
+            "seq_lens": torch.tensor(plan["seq_lens"], dtype=torch.int32,
+                                     device=dev),
             "block_table": btab.to(dev),
             "slot_mapping": slot_map,
         }
-        logits = self.model(tokens, positions, kv_caches=self.kv_caches,
-                            attn_ctx=attn_ctx, last_token_idx=None)
-        self._sample_and_emit(seqs, logits)
+        return self.model(tokens, positions, kv_caches=self.kv_caches,
+                          attn_ctx=attn_ctx, last_token_idx=None)
+
+    # ------------------------------------------------------------------ #
+    # tensor-parallel coordination (rank 0 schedules, workers follow)
+    # ------------------------------------------------------------------ #
+    def _tp_broadcast(self, plan: Optional[Dict[str, Any]]) -> None:
+        if getattr(self, "tp_size", 1) <= 1 or self.tp_rank != 0:
+            return
+        import torch.distributed as dist
+
+        dist.broadcast_object_list([plan], src=0)
+
+    def run_tp_worker(self) -> None:
+        """Worker-rank loop: execute rank 0's step plans until shutdown.
+
+        The model's row-parallel all-reduces and the logits all-gather are
+        the synchronization points; workers discard logits (sampling happens
+        on rank 0 only)."""
+        import torch.distributed as dist
+
+        assert self.tp_size > 1 and self.tp_rank != 0
+        while True:
+            box = [None]
+            dist.broadcast_object_list(box, src=0)
+            plan = box[0]
+            if plan is None or plan.get("mode") == "stop":
+                return
+            if plan["mode"] == "prefill":
+                self._exec_prefill(plan)
+            else:
+                self._exec_decode(plan)
+
+    def tp_shutdown(self) -> None:
+        if getattr(self, "tp_size", 1) > 1 and self.tp_rank == 0:
+            self._tp_broadcast({"mode": "stop"})
 
     def _sample_and_emit(self, seqs: List[Sequence], logits: torch.Tensor) -> None:
         # group rows by identical sampling params for batched kernels

@@ -40,10 +40,21 @@ _DTYPES = {
 
 
 def _pick_device(aux: Dict[str, Any]) -> torch.device:
-    if torch.cuda.is_available():
-        idx = int(aux.get("gpu", 0))
-        return torch.device("cuda", idx)
-    return torch.device("cpu")
+    """Endpoint -> GPU placement. Explicit via auxiliary_cfg {"gpu": N};
+    otherwise the device with the most free HBM takes the model, so
+    multi-model sessions spread across the node's 8 GPUs automatically
+    (the reference delegates placement to Triton instance groups;
+    SURVEY.md §2.6)."""
+    if not torch.cuda.is_available():
+        return torch.device("cpu")
+    if "gpu" in aux:
+        return torch.device("cuda", int(aux["gpu"]))
+    best, best_free = 0, -1
+    for i in range(torch.cuda.device_count()):
+        free, _ = torch.cuda.mem_get_info(i)
+        if free > best_free:
+            best, best_free = i, free
+    return torch.device("cuda", best)
 
 
 @BasePreprocessRequest.register_engine("hip", modules=["torch"])

@@ -54,6 +54,44 @@ def broadcast_tokens(t: torch.Tensor, src: int = 0, group=None) -> torch.Tensor:
     return t
 
 
+def shard_llama_weights(full_state: dict, cfg, rank: int, world: int) -> dict:
+    """Slice a full (tp=1) llama state dict into rank's TP shard.
+
+    Column-parallel qkv / gate_up / lm_head take row slices of the weight
+    matrices; row-parallel o_proj / down take column slices; embeddings and
+    norms replicate. Used by the weight loader and the TP correctness tests.
+    """
+    hd = cfg.head_dim
+    hpr = cfg.heads // world          # q heads per rank
+    kvpr = cfg.kv_heads // world
+    ipr = cfg.intermediate // world
+    vpr = cfg.vocab_size // world
+    q_out, kv_out = cfg.heads * hd, cfg.kv_heads * hd
+
+    out = {}
+    for k, v in full_state.items():
+        if ".qkv.weight" in k:
+            q = v[rank * hpr * hd:(rank + 1) * hpr * hd]
+            kk = v[q_out + rank * kvpr * hd: q_out + (rank + 1) * kvpr * hd]
+            vv = v[q_out + kv_out + rank * kvpr * hd:
+                   q_out + kv_out + (rank + 1) * kvpr * hd]
+            out[k] = __import__("torch").cat([q, kk, vv], dim=0)
+        elif ".o_proj.weight" in k:
+            out[k] = v[:, rank * hpr * hd:(rank + 1) * hpr * hd]
+        elif ".gate_up.weight" in k:
+            gate = v[rank * ipr:(rank + 1) * ipr]
+            up = v[cfg.intermediate + rank * ipr:
+                   cfg.intermediate + (rank + 1) * ipr]
+            out[k] = __import__("torch").cat([gate, up], dim=0)
+        elif ".down.weight" in k:
+            out[k] = v[:, rank * ipr:(rank + 1) * ipr]
+        elif k.startswith("lm_head."):
+            out[k] = v[rank * vpr:(rank + 1) * vpr]
+        else:
+            out[k] = v
+    return out
+
+
 def init_from_env(device_type: Optional[str] = None) -> int:
     """Initialize the process group from torchrun env (RANK/WORLD_SIZE);
     returns local rank. Safe to call when already initialized."""

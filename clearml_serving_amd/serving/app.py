@@ -195,6 +195,48 @@ def create_app(
             "endpoints": sorted(proc.get_synced_endpoints().keys()) if proc else [],
         }
 
+    @app.get("/status")
+    async def status():
+        """Routing/state overview -- the single-node equivalent of the
+        reference's endpoint table + Sankey routing plot posted to the
+        ClearML UI (model_request_processor.py:1141-1278)."""
+        proc = state["processor"]
+        if proc is None:
+            return {"status": "starting"}
+        endpoints = {}
+        for url, ep in proc.get_synced_endpoints().items():
+            entry = ep.as_dict(remove_null_entries=True)
+            engine = proc._engine_processor_lookup.get(url)
+            batcher = getattr(engine, "_batcher", None)
+            if batcher is not None:
+                s = dict(batcher.stats)
+                if s.get("batches"):
+                    s["mean_batch_occupancy"] = round(
+                        s.pop("occupancy_sum") / s["batches"], 3)
+                entry["batcher"] = s
+                entry["device"] = str(batcher.device)
+            llm = getattr(engine, "_engine", None)
+            if llm is not None and hasattr(llm, "stats"):
+                entry["llm"] = dict(llm.stats)
+            endpoints[url] = entry
+        routes = {
+            url: route for url, route in proc._canary_route.items()
+        }
+        return {
+            "session": proc.get_id(),
+            "revision": proc._last_revision,
+            "endpoints": endpoints,
+            "canary_routes": routes,
+            "monitoring": {
+                k: v.as_dict(remove_null_entries=True)
+                for k, v in proc.get_model_monitoring().items()
+            },
+            "metric_logging": {
+                k: v.as_dict()
+                for k, v in proc.list_endpoint_logging().items()
+            },
+        }
+
     return app
 
 

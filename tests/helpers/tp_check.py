@@ -1,0 +1,82 @@
+"""TP=2 correctness worker (launched by test_tp_cpu via torchrun, gloo).
+
+Builds a full tp=1 llama-tiny, shards its weights into this rank's tp=2
+model, and checks TP logits == full-model logits; then runs the engine's
+plan-broadcast protocol end to end (rank 0 schedules, rank 1 follows)."""
+
+import asyncio
+import os
+import sys
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+from clearml_serving_amd.engines.llm.engine import (  # noqa: E402
+    LlmEngine, LlmEngineConfig, SamplingParams)
+from clearml_serving_amd.models.llama import (  # noqa: E402
+    PRESETS, LlamaForCausalLM)
+from clearml_serving_amd.parallel import tp  # noqa: E402
+
+
+def check_tp_math(rank, world):
+    cfg = PRESETS["llama-tiny"]
+    torch.manual_seed(7)
+    full = LlamaForCausalLM(cfg).eval()
+    shard_state = tp.shard_llama_weights(full.state_dict(), cfg, rank, world)
+    tp_model = LlamaForCausalLM(cfg, tp_rank=rank, tp_size=world).eval()
+    tp_model.load_state_dict(shard_state)
+
+    tokens = torch.arange(10, 26, dtype=torch.long)
+    positions = torch.arange(16, dtype=torch.int32)
+    ctx = {"mode": "prefill", "batch": 1, "seq": 16,
+           "seq_lens": torch.tensor([16], dtype=torch.int32),
+           "slot_mapping": torch.full((16,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        ref = full(tokens, positions, kv_caches=None, attn_ctx=ctx)
+        got = tp_model(tokens, positions, kv_caches=None, attn_ctx=ctx)
+    torch.testing.assert_close(got, ref, atol=2e-4, rtol=2e-4)
+    if rank == 0:
+        print("TP-MATH-OK", flush=True)
+
+
+def check_engine_protocol(rank, world):
+    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
+                          block_size=16, max_model_len=128, device="cpu")
+    eng = LlmEngine(cfg)
+    eng.start()
+    assert eng.tp_size == world
+    if rank == 0:
+        async def gen():
+            params = SamplingParams(temperature=0.0, max_tokens=6,
+                                    ignore_eos=True)
+
+            async def one(i):
+                toks = []
+                async for item in eng.generate("req %d" % i, params):
+                    toks.extend(item["token_ids"])
+                return toks
+
+            return await asyncio.gather(one(0), one(1), one(2))
+
+        outs = asyncio.new_event_loop().run_until_complete(gen())
+        assert all(len(o) == 6 for o in outs), outs
+        eng.tp_shutdown()
+        print("TP-ENGINE-OK", flush=True)
+    else:
+        eng.run_tp_worker()
+
+
+def main():
+    dist.init_process_group(backend="gloo")
+    rank, world = dist.get_rank(), dist.get_world_size()
+    check_tp_math(rank, world)
+    check_engine_protocol(rank, world)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -92,3 +92,12 @@ def test_stats_flow_to_prometheus(client, processor):
     lat_count = reg.get_sample_value(
         "test_model_sklearn__latency_count")
     assert lat_count == 3.0
+
+
+def test_status_endpoint(client):
+    client.post("/serve/test_model_sklearn", json={"x0": 1.0, "x1": 0.0})
+    r = client.get("/status")
+    assert r.status_code == 200
+    body = r.json()
+    assert "test_model_sklearn" in body["endpoints"]
+    assert body["revision"] is not None
