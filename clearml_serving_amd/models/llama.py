@@ -75,14 +75,16 @@ class LlamaLayer(nn.Module):
         t = x.shape[0]
 
         qkv = self.qkv(x)
+        # token-strided views into the merged projection -- the rope /
+        # kv_cache_write / decode-attention kernels are stride-aware, so no
+        # .contiguous() copies on the decode hot path
         q, k, v = qkv.split(
             [self.heads * self.head_dim, self.kv_heads * self.head_dim,
              self.kv_heads * self.head_dim], dim=-1)
-        q = q.view(t, self.heads, self.head_dim)
-        k = k.contiguous().view(t, self.kv_heads, self.head_dim)
-        v = v.contiguous().view(t, self.kv_heads, self.head_dim)
-        q, k = ops.rope_inplace(q.contiguous(), k, positions,
-                                theta=cfg.rope_theta)
+        q = q.unflatten(-1, (self.heads, self.head_dim))
+        k = k.unflatten(-1, (self.kv_heads, self.head_dim))
+        v = v.unflatten(-1, (self.kv_heads, self.head_dim))
+        q, k = ops.rope_inplace(q, k, positions, theta=cfg.rope_theta)
 
         if kv_cache is not None:
             k_cache, v_cache = kv_cache
@@ -91,12 +93,9 @@ class LlamaLayer(nn.Module):
         scale = 1.0 / math.sqrt(self.head_dim)
         if attn_ctx["mode"] == "prefill":
             b, s = attn_ctx["batch"], attn_ctx["seq"]
-            qb = q.view(b, s, self.heads, self.head_dim).transpose(1, 2) \
-                .contiguous()
-            kb = k.view(b, s, self.kv_heads, self.head_dim).transpose(1, 2) \
-                .contiguous()
-            vb = v.view(b, s, self.kv_heads, self.head_dim).transpose(1, 2) \
-                .contiguous()
+            qb = q.unflatten(0, (b, s)).transpose(1, 2).contiguous()
+            kb = k.unflatten(0, (b, s)).transpose(1, 2).contiguous()
+            vb = v.unflatten(0, (b, s)).transpose(1, 2).contiguous()
             ctx = ops.attention(qb, kb, vb, causal=True, scale=scale,
                                 seq_lens=attn_ctx["seq_lens"])
             ctx = ctx.transpose(1, 2).reshape(t, self.heads * self.head_dim)
@@ -116,7 +115,7 @@ class LlamaLayer(nn.Module):
                         residual=residual)
         gate_up = self.gate_up(x)
         gate, up = gate_up.split([self.inter, self.inter], dim=-1)
-        mlp_out = self.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
+        mlp_out = self.down(ops.silu_mul(gate, up))  # stride-aware, no copy
         if self.tp_size > 1:
             from ..parallel import tp as tp_mod
 

@@ -46,7 +46,7 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     __hip_bfloat16* __restrict__ out,            // [B, H, D]
     float* __restrict__ partial_o,               // [B, H, SPLITS, D] f32
     float* __restrict__ partial_ml,              // [B, H, SPLITS, 2] f32
-    int splits,
+    int splits, long q_bstride,
     int H, int Hkv, int block_size, int max_blocks, float scale) {
   constexpr int NW = 4;        // waves
   constexpr int GROUPS = 4;    // 16-lane key groups per wave
@@ -81,7 +81,8 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
 #pragma unroll
   for (int g = 0; g < GQ; ++g) {
     const int h = hkv * GQ + g;
-    const __hip_bfloat16* qrow = q + ((long)b * H + h) * D + sub * EPL;
+    const __hip_bfloat16* qrow = q + (long)b * q_bstride + (long)h * D +
+                                 sub * EPL;
 #pragma unroll
     for (int e = 0; e < EPL; ++e) qf[g][e] = to_f32(qrow[e]) * scale;
   }
@@ -266,7 +267,8 @@ __global__ void kv_cache_write_kernel(const T* __restrict__ knew,  // [T,Hkv,D]
                                       T* __restrict__ k_cache,  // [NB,Hkv,BS,D]
                                       T* __restrict__ v_cache,
                                       const int* __restrict__ slots,  // [T]
-                                      int hkv, int d, int block_size) {
+                                      int hkv, int d, int block_size,
+                                      long k_tstride, long v_tstride) {
   const int t = blockIdx.x;
   const int slot = slots[t];
   if (slot < 0) return;
@@ -277,8 +279,8 @@ __global__ void kv_cache_write_kernel(const T* __restrict__ knew,  // [T,Hkv,D]
     const int dd = i % d;
     const long dst =
         ((blk * hkv + h) * block_size + off) * (long)d + dd;
-    k_cache[dst] = knew[((long)t * hkv + h) * d + dd];
-    v_cache[dst] = vnew[((long)t * hkv + h) * d + dd];
+    k_cache[dst] = knew[(long)t * k_tstride + (long)h * d + dd];
+    v_cache[dst] = vnew[(long)t * v_tstride + (long)h * d + dd];
   }
 }
 
@@ -290,8 +292,9 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                                torch::Tensor seq_lens, double scale) {
   TORCH_CHECK(q.dim() == 3, "q must be [B, H, D]");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attention: bf16 only");
-  TORCH_CHECK(q.is_contiguous() && k_cache.is_contiguous() &&
-              v_cache.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "q heads must be dense");
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int Hkv = k_cache.size(1), BS = k_cache.size(2);
   const int max_blocks = block_table.size(1);
@@ -329,8 +332,8 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                      (const __hip_bfloat16*)k_cache.data_ptr(),              \
                      (const __hip_bfloat16*)v_cache.data_ptr(),              \
                      bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
-                     (__hip_bfloat16*)out.data_ptr(), po, pml, splits, H,    \
-                     Hkv, BS, max_blocks, (float)scale)
+                     (__hip_bfloat16*)out.data_ptr(), po, pml, splits,       \
+                     q.stride(0), H, Hkv, BS, max_blocks, (float)scale)
   if (D == 128) {
     if (GQ == 1) LAUNCH_DEC(128, 1);
     else if (GQ == 2) LAUNCH_DEC(128, 2);
@@ -362,8 +365,11 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
                     torch::Tensor slot_mapping) {
   TORCH_CHECK(knew.dim() == 3, "knew must be [T, Hkv, D]");
-  TORCH_CHECK(knew.is_contiguous() && vnew.is_contiguous() &&
-              k_cache.is_contiguous() && v_cache.is_contiguous());
+  // token-strided views of a merged QKV projection are accepted
+  TORCH_CHECK(knew.stride(2) == 1 && knew.stride(1) == knew.size(2));
+  TORCH_CHECK(vnew.stride(2) == 1 && vnew.stride(1) == vnew.size(2));
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  const long kts = knew.stride(0), vts = vnew.stride(0);
   const int T = knew.size(0), Hkv = knew.size(1), D = knew.size(2);
   const int BS = k_cache.size(2);
   auto slots = slot_mapping.to(knew.device(), at::kInt).contiguous();
@@ -379,14 +385,14 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                        (const __hip_bfloat16*)vnew.data_ptr(),
                        (__hip_bfloat16*)k_cache.data_ptr(),
                        (__hip_bfloat16*)v_cache.data_ptr(),
-                       slots.data_ptr<int>(), Hkv, D, BS);
+                       slots.data_ptr<int>(), Hkv, D, BS, kts, vts);
   } else if (st == at::kHalf) {
     hipLaunchKernelGGL(kv_cache_write_kernel<__half>, dim3(T), dim3(block), 0,
                        stream_, (const __half*)knew.data_ptr(),
                        (const __half*)vnew.data_ptr(),
                        (__half*)k_cache.data_ptr(),
                        (__half*)v_cache.data_ptr(), slots.data_ptr<int>(),
-                       Hkv, D, BS);
+                       Hkv, D, BS, kts, vts);
   } else {
     TORCH_CHECK(false, "kv_cache_write: unsupported dtype ", st);
   }

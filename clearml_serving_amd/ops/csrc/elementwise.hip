@@ -86,7 +86,10 @@ __global__ void bias_gelu_vec8_kernel(const T* __restrict__ x,
 template <typename T>
 __global__ void silu_mul_vec8_kernel(const T* __restrict__ gate,
                                      const T* __restrict__ up,
-                                     T* __restrict__ out, long n8) {
+                                     T* __restrict__ out, long n8, long c8,
+                                     long gs8, long us8) {
+  // rows of length c8 vectors; gate/up row strides gs8/us8 (in vectors) --
+  // accepts views into a merged gate_up projection without a copy
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   constexpr int VE = 16 / sizeof(T);
@@ -95,9 +98,10 @@ __global__ void silu_mul_vec8_kernel(const T* __restrict__ gate,
   const V* uv = reinterpret_cast<const V*>(up);
   V* ov = reinterpret_cast<V*>(out);
   for (; i < n8; i += stride) {
+    const long r = i / c8, c = i % c8;
     union { V v; T e[VE]; } g, u;
-    g.v = gv[i];
-    u.v = uv[i];
+    g.v = gv[r * gs8 + c];
+    u.v = uv[r * us8 + c];
 #pragma unroll
     for (int j = 0; j < VE; ++j) {
       float x = to_f32(g.e[j]);
@@ -135,14 +139,25 @@ inline int grid_for(long work_items, int block = 256) {
 torch::Tensor bias_relu_add(torch::Tensor x,
                             c10::optional<torch::Tensor> bias,
                             c10::optional<torch::Tensor> residual) {
-  CHECK_LASTDIM_CONTIG(x);
+  const bool has_bias = bias.has_value() && bias->defined();
+  const bool has_res = residual.has_value() && residual->defined();
+  // the no-bias path is pure elementwise over dense memory -- any dense
+  // layout (incl. channels_last NHWC) is fine as long as x and residual
+  // share it; the channel-bias path needs standard NCHW strides
+  if (has_bias) {
+    TORCH_CHECK(x.is_contiguous(), "bias path needs standard contiguous x");
+  } else {
+    TORCH_CHECK(x.is_non_overlapping_and_dense(), "x must be dense");
+  }
   auto out = torch::empty_like(x);
   const long n = x.numel();
   if (n == 0) return out;
   hipStream_t stream_ = cmls::current_stream();
-  const bool has_bias = bias.has_value() && bias->defined();
-  const bool has_res = residual.has_value() && residual->defined();
-  if (has_res) TORCH_CHECK(residual->is_contiguous() && residual->sizes() == x.sizes());
+  if (has_res) {
+    TORCH_CHECK(residual->sizes() == x.sizes() &&
+                    residual->strides() == x.strides(),
+                "residual must match x layout");
+  }
 
   DISPATCH_FLOAT_TYPES(x, "bias_relu_add", [&] {
     constexpr int VE = 16 / sizeof(scalar_t);
@@ -198,19 +213,28 @@ torch::Tensor bias_gelu(torch::Tensor x, c10::optional<torch::Tensor> bias) {
 
 torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
   CHECK_LASTDIM_CONTIG(gate);
-  TORCH_CHECK(gate.sizes() == up.sizes() && up.is_contiguous());
-  auto out = torch::empty_like(gate);
+  CHECK_LASTDIM_CONTIG(up);
+  TORCH_CHECK(gate.sizes() == up.sizes());
+  TORCH_CHECK(gate.dim() <= 2 || gate.is_contiguous(),
+              "silu_mul: >2D inputs must be contiguous");
+  const long rows = gate.dim() == 2 ? gate.size(0) : 1;
+  const long cols = gate.numel() / rows;
+  auto out = torch::empty({rows, cols}, gate.options()).view(gate.sizes());
   const long n = gate.numel();
   if (n == 0) return out;
   hipStream_t stream_ = cmls::current_stream();
+  const long gs = gate.dim() == 2 ? gate.stride(0) : cols;
+  const long us = up.dim() == 2 ? up.stride(0) : cols;
   DISPATCH_FLOAT_TYPES(gate, "silu_mul", [&] {
     constexpr int VE = 16 / sizeof(scalar_t);
-    TORCH_CHECK(n % VE == 0, "numel must be a multiple of ", VE);
+    TORCH_CHECK(cols % VE == 0 && gs % VE == 0 && us % VE == 0,
+                "row length/strides must be multiples of ", VE);
     hipLaunchKernelGGL(silu_mul_vec8_kernel<scalar_t>, dim3(grid_for(n / VE)),
                        dim3(256), 0, stream_,
                        (const scalar_t*)gate.data_ptr(),
                        (const scalar_t*)up.data_ptr(),
-                       (scalar_t*)out.data_ptr(), n / VE);
+                       (scalar_t*)out.data_ptr(), n / VE, cols / VE,
+                       gs / VE, us / VE);
   });
   return out;
 }
