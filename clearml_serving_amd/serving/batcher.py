@@ -22,6 +22,7 @@ by latency targets, not memory -- raise max_batch_size freely via
 """
 
 import asyncio
+import queue as _queue
 import threading
 import time
 from typing import Callable, Dict, List, Optional, Sequence, Union
@@ -66,8 +67,17 @@ class DynamicBatcher:
         self._worker_task: Optional[asyncio.Task] = None
         self._stream = torch.cuda.Stream() if self.is_cuda else None
         self._graphs: Dict[int, dict] = {}  # bucket -> {graph, in, out}
-        self._pinned: Dict = {}             # (key,bucket,shape,dtype) -> slab
-        self._exec_lock = threading.Lock()
+        self._pinned: Dict = {}             # (slot,key,bucket,...) -> slab
+        self._exec_lock = threading.Lock()  # CPU path only
+        # pipelining: up to 2 batches in flight (CPU staging of batch n+1
+        # overlaps GPU execution of batch n); the enqueue lock makes each
+        # batch's GPU ops contiguous on the stream, so the in-order stream
+        # keeps static graph buffers race-free
+        self._slots: "_queue.Queue" = _queue.Queue()
+        self._slots.put(0)
+        self._slots.put(1)
+        self._enqueue_lock = threading.Lock()
+        self._inflight = set()
         self._closed = False
         # telemetry for the Prometheus exporter
         self.stats = {"batches": 0, "requests": 0, "occupancy_sum": 0.0}
@@ -118,15 +128,24 @@ class DynamicBatcher:
                     break
             inputs = [b[0] for b in batch]
             futures = [b[1] for b in batch]
-            try:
-                outputs = await asyncio.to_thread(self._execute, inputs)
-                for fut, out in zip(futures, outputs):
-                    if not fut.done():
-                        fut.set_result(out)
-            except Exception as ex:
-                for fut in futures:
-                    if not fut.done():
-                        fut.set_exception(ex)
+            # fire-and-continue: the next batch forms while this one stages
+            # and executes (in-flight count bounded by the slot queue);
+            # hold a reference so the task isn't garbage-collected mid-run
+            task = asyncio.get_running_loop().create_task(
+                self._dispatch(inputs, futures))
+            self._inflight.add(task)
+            task.add_done_callback(self._inflight.discard)
+
+    async def _dispatch(self, inputs, futures) -> None:
+        try:
+            outputs = await asyncio.to_thread(self._execute, inputs)
+            for fut, out in zip(futures, outputs):
+                if not fut.done():
+                    fut.set_result(out)
+        except Exception as ex:
+            for fut in futures:
+                if not fut.done():
+                    fut.set_exception(ex)
 
     def warmup(self, sample: TensorOrDict) -> None:
         """Pre-capture every bucket's hipGraph from one sample request so no
@@ -147,50 +166,66 @@ class DynamicBatcher:
 
         is_dict = isinstance(inputs[0], dict)
 
-        def _stage():
-            # must run with the batcher's stream current: the async H2D copy
-            # out of the pinned slab orders against the model launches below
-            if is_dict:
-                keys = list(inputs[0].keys())
-                return {
-                    k: self._stack_pad([x[k] for x in inputs], bucket, key=k)
-                    for k in keys
-                }
-            return self._stack_pad(inputs, bucket)
+        if not self.is_cuda:
+            def _stage_cpu():
+                if is_dict:
+                    keys = list(inputs[0].keys())
+                    return {k: self._stack_pad([x[k] for x in inputs],
+                                               bucket, key=k) for k in keys}
+                return self._stack_pad(inputs, bucket)
 
-        with self._exec_lock:
-            if self.is_cuda:
-                with torch.cuda.stream(self._stream):
-                    stacked = _stage()
-                    out = self._run_model(stacked, bucket)
-                    out_cpu = _to_cpu(out)
-                self._stream.synchronize()
+            with self._exec_lock:
+                out_cpu = _to_cpu(self._run_model(_stage_cpu(), bucket))
+            return [_slice(out_cpu, i) for i in range(n)]
+
+        slot = self._slots.get()  # bounds in-flight batches (2)
+        try:
+            # CPU staging into this slot's pinned slab overlaps the other
+            # slot's GPU execution
+            if is_dict:
+                staged = {k: self._stack_pad([x[k] for x in inputs], bucket,
+                                             key=k, slot=slot)
+                          for k in inputs[0].keys()}
             else:
-                out_cpu = _to_cpu(self._run_model(_stage(), bucket))
-        return [_slice(out_cpu, i) for i in range(n)]
+                staged = self._stack_pad(inputs, bucket, slot=slot)
+
+            with self._enqueue_lock:
+                with torch.cuda.stream(self._stream):
+                    dev = _to_device(staged, self.device)
+                    out = self._run_model(dev, bucket)
+                    out_cpu = self._pinned_out(out, slot)
+                done = torch.cuda.Event()
+                done.record(self._stream)
+            done.synchronize()
+            # slices are views into the reusable pinned slab: clone them
+            return [_slice_clone(out_cpu, i) for i in range(n)]
+        finally:
+            self._slots.put(slot)
 
     def _stack_pad(self, tensors: List[torch.Tensor], bucket: int,
-                   key: str = "") -> torch.Tensor:
+                   key: str = "", slot: int = 0) -> torch.Tensor:
         ts = [torch.as_tensor(x) for x in tensors]
         want = self.dtype if (self.dtype is not None
                               and ts[0].is_floating_point()) else ts[0].dtype
         if ts[0].dtype != want:
             ts = [t.to(want) for t in ts]
         if self.is_cuda:
-            # stage through a reusable pinned host slab -> async H2D
-            buf = self._pinned_slab(key, bucket, ts[0].shape, want)
+            # fill this slot's reusable pinned host slab (the H2D copy is
+            # enqueued later on the stream, inside the enqueue lock)
+            buf = self._pinned_slab(("in", slot, key), bucket, ts[0].shape,
+                                    want)
             n = len(ts)
             torch.stack(ts, dim=0, out=buf[:n])
             if n < bucket:
                 buf[n:bucket] = buf[0]
-            return buf[:bucket].to(self.device, non_blocking=True)
+            return buf[:bucket]
         t = torch.stack(ts, dim=0)
         if t.shape[0] < bucket:
             pad = t[:1].expand(bucket - t.shape[0], *t.shape[1:])
             t = torch.cat([t, pad], dim=0)
         return t
 
-    def _pinned_slab(self, key: str, bucket: int, shape, dtype) -> torch.Tensor:
+    def _pinned_slab(self, key, bucket: int, shape, dtype) -> torch.Tensor:
         slab_key = (key, bucket, tuple(shape), dtype)
         slab = self._pinned.get(slab_key)
         if slab is None:
@@ -198,6 +233,18 @@ class DynamicBatcher:
                                pin_memory=True)
             self._pinned[slab_key] = slab
         return slab
+
+    def _pinned_out(self, out, slot):
+        # async D2H into this slot's pinned output slab(s)
+        if isinstance(out, dict):
+            return {k: self._pinned_out(v, (slot, k)) for k, v in out.items()}
+        if isinstance(out, (tuple, list)):
+            return type(out)(self._pinned_out(v, (slot, i))
+                             for i, v in enumerate(out))
+        buf = self._pinned_slab(("out", slot), out.shape[0], out.shape[1:],
+                                out.dtype)
+        buf.copy_(out, non_blocking=True)
+        return buf
 
     @torch.inference_mode()
     def _run_model(self, stacked: TensorOrDict, bucket: int) -> TensorOrDict:
@@ -257,3 +304,19 @@ def _slice(x, i: int):
     if isinstance(x, (tuple, list)):
         return type(x)(_slice(v, i) for v in x)
     return x[i]
+
+
+def _slice_clone(x, i: int):
+    if isinstance(x, dict):
+        return {k: _slice_clone(v, i) for k, v in x.items()}
+    if isinstance(x, (tuple, list)):
+        return type(x)(_slice_clone(v, i) for v in x)
+    return x[i].clone()
+
+
+def _to_device(x, device):
+    if isinstance(x, dict):
+        return {k: _to_device(v, device) for k, v in x.items()}
+    if isinstance(x, (tuple, list)):
+        return type(x)(_to_device(v, device) for v in x)
+    return x.to(device, non_blocking=True)
