@@ -99,6 +99,21 @@ def bench_memops(results):
                         gbps=x.numel() * 2 * 3 / dt / 1e9))
 
 
+def bench_gemm(results):
+    for (m, n, k) in [(4096, 4096, 4096), (8192, 8192, 8192),
+                      (8192, 3072, 768)]:
+        a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+        b = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+        dt = timeit(lambda: ops.gemm_bf16(a, b), iters=20)
+        results.append(dict(op="gemm_bf16(ours)",
+                            shape="{}x{}x{}".format(m, n, k),
+                            us=dt * 1e6, tflops=2.0 * m * n * k / dt / 1e12))
+        dtl = timeit(lambda: a @ b.T, iters=20)
+        results.append(dict(op="gemm_hipblaslt",
+                            shape="{}x{}x{}".format(m, n, k),
+                            us=dtl * 1e6, tflops=2.0 * m * n * k / dtl / 1e12))
+
+
 def bench_sampling(results):
     logits = torch.randn(64, 128256, device="cuda", dtype=torch.bfloat16)
     dt = timeit(lambda: ops.sample_top_k_top_p(logits, temperature=0.8))
@@ -115,6 +130,7 @@ def main():
     bench_attention(results)
     bench_decode(results)
     bench_memops(results)
+    bench_gemm(results)
     bench_sampling(results)
     for r in results:
         perf = ("{:8.1f} TF".format(r["tflops"]) if "tflops" in r

@@ -337,14 +337,23 @@ def sample_top_k_top_p(
 # GEMM (bf16 MFMA)
 # --------------------------------------------------------------------- #
 def gemm_bf16(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    """C[M,N] = A[M,K] @ B[K,N] in bf16 with fp32 accumulation on MFMA.
+    """C[M,N] = A[M,K] @ B[N,K]^T in bf16 with fp32 accumulation on MFMA
+    (nn.Linear orientation: b is a [out, in] weight matrix).
 
-    Showcase/hot-path GEMM; plain library GEMMs go through torch.matmul
-    (hipBLASLt) -- this kernel exists for fused epilogues and to anchor the
-    MFMA performance path.
+    The in-tree MFMA GEMM of the kernel library; library GEMMs in the
+    serving models go through torch.matmul (hipBLASLt) where that wins.
+    Shapes off the 128/128/32 tile grid are zero-padded here.
     """
     if a.is_cuda:
         ext = _require_ext("gemm_bf16")
         if ext is not None:
-            return ext.gemm_bf16(a, b)
-    return (a.float() @ b.float()).to(a.dtype)
+            m, k = a.shape
+            n = b.shape[0]
+            pm, pn, pk = (-m) % 128, (-n) % 128, (-k) % 32
+            if pm or pn or pk:
+                a2 = torch.nn.functional.pad(a, (0, pk, 0, pm))
+                b2 = torch.nn.functional.pad(b, (0, pk, 0, pn))
+                return ext.gemm_bf16(a2.contiguous(),
+                                     b2.contiguous())[:m, :n].contiguous()
+            return ext.gemm_bf16(a.contiguous(), b.contiguous())
+    return (a.float() @ b.float().T).to(a.dtype)

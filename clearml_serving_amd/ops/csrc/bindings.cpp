@@ -23,6 +23,7 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
 void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
                     torch::Tensor slot_mapping);
+torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   double theta);
 
@@ -45,4 +46,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_inplace", &rope_inplace);
   m.def("attention_decode", &attention_decode);
   m.def("kv_cache_write", &kv_cache_write);
+  m.def("gemm_bf16", &gemm_bf16);
 }

@@ -253,3 +253,31 @@ def test_sample_distribution_sanity():
     frac = counts / counts.sum()
     assert abs(frac[0].item() - 0.9) < 0.03
     assert counts[2] == 0 and counts[3] == 0
+
+
+# ------------------------------------------------------------------ #
+# MFMA GEMM
+# ------------------------------------------------------------------ #
+@pytest.mark.parametrize("m,n,k", [(128, 128, 32), (256, 512, 1024),
+                                   (100, 200, 300)])
+def test_gemm_bf16(m, n, k):
+    torch.manual_seed(0)
+    a = torch.randn(m, k, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(n, k, device=DEV, dtype=torch.bfloat16)
+    got = ops.gemm_bf16(a, b)
+    ref = a.float() @ b.float().T
+    assert_close_bf16(got, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gemm_bf16_transpose_detecting():
+    # asymmetric B catches swapped C-write / operand maps (guide §3)
+    m = n = k = 128
+    a = torch.zeros(m, k, device=DEV, dtype=torch.bfloat16)
+    b = torch.zeros(n, k, device=DEV, dtype=torch.bfloat16)
+    for i in range(m):
+        a[i, (i * 3) % k] = 1.0
+    for j in range(n):
+        b[j, (j * 7 + 1) % k] = float(j % 5 + 1) * 0.25
+    got = ops.gemm_bf16(a, b)
+    ref = a.float() @ b.float().T
+    assert_close_bf16(got, ref, atol=1e-2, rtol=1e-2)
