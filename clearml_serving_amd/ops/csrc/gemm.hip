@@ -43,8 +43,15 @@ __global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
   __shared__ short lds_b[2][BN * BK];
 
   const int tiles_n = N / BN;
-  // XCD-friendly: consecutive blocks walk N first (panel reuse within XCD)
-  const int bid = blockIdx.x;
+  // bijective XCD remap (guide T1): the dispatcher places block b on XCD
+  // b%8, so give each XCD a CONTIGUOUS chunk of the N-first walk -- then
+  // neighbor tiles sharing A/B panels hit the same XCD's L2
+  const int nwg = gridDim.x;
+  const int xcd = blockIdx.x % 8;
+  const int idx = blockIdx.x / 8;
+  const int q8 = nwg / 8, r8 = nwg % 8;
+  const int bid = (xcd < r8 ? xcd * (q8 + 1)
+                            : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
   const int bm = bid / tiles_n;
   const int bn = bid % tiles_n;
 
