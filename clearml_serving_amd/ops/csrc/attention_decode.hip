@@ -99,61 +99,62 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   const int* btab = block_table + (long)b * max_blocks;
   const long hk_off = (long)hkv * block_size * D;
 
-  // keys processed 16 per workgroup iteration: wave w group g -> key
-  // it*16 + w*4 + g
+  // keys processed 16 per workgroup iteration (wave w, group g -> key
+  // it*16 + w*4 + g), UNROLLED 4 iterations deep: 8 x 16-B loads per lane
+  // in flight (one K + one V per unrolled key) -- a single load in flight
+  // was latency-bound at ~1.7 TB/s
   const int per_iter = NW * GROUPS;
   const int n_iters = (n_keys - key_lo + per_iter - 1) / per_iter;
-  for (int it = 0; it < n_iters; ++it) {
-    const int key = key_lo + it * per_iter + wave * GROUPS + group;
-    const bool valid = key < n_keys;
-    float score[GQ];
-    const __hip_bfloat16* krow = nullptr;
-    const __hip_bfloat16* vrow = nullptr;
-    if (valid) {
-      const int blk = btab[key / block_size];
+  constexpr int UNROLL = 4;
+  for (int it0 = 0; it0 < n_iters; it0 += UNROLL) {
+    bf16x8 kvec[UNROLL], vvec[UNROLL];
+    bool valid[UNROLL];
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      const int key = key_lo + (it0 + u) * per_iter + wave * GROUPS + group;
+      valid[u] = (it0 + u) < n_iters && key < n_keys;
+      const int kc = min(key, n_keys - 1);  // clamped: always a mapped page
+      const int blk = btab[kc / block_size];
       const long base = ((long)blk * Hkv) * block_size * D + hk_off +
-                        (long)(key % block_size) * D;
-      krow = k_cache + base;
-      vrow = v_cache + base;  // same layout
-      bf16x8 kv;
-      kv.u = *reinterpret_cast<const uint32x4*>(krow + sub * EPL);
+                        (long)(kc % block_size) * D;
+      kvec[u].u = *reinterpret_cast<const uint32x4*>(
+          k_cache + base + sub * EPL);
+      vvec[u].u = *reinterpret_cast<const uint32x4*>(
+          v_cache + base + sub * EPL);
+    }
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      float score[GQ];
 #pragma unroll
       for (int g = 0; g < GQ; ++g) {
         float acc = 0.f;
 #pragma unroll
-        for (int e = 0; e < EPL; ++e) acc += qf[g][e] * to_f32(kv.h[e]);
+        for (int e = 0; e < EPL; ++e) acc += qf[g][e] * to_f32(kvec[u].h[e]);
         score[g] = acc;
       }
-    } else {
+      // group-level dot reduction (16 lanes hold partial sums)
 #pragma unroll
-      for (int g = 0; g < GQ; ++g) score[g] = -INFINITY;
-    }
-    // group-level dot reduction (16 lanes hold partial sums)
+      for (int g = 0; g < GQ; ++g) {
 #pragma unroll
-    for (int g = 0; g < GQ; ++g) {
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1) {
-        float other = __shfl_xor(score[g], off, 64);
-        score[g] = valid ? score[g] + other : -INFINITY;
+        for (int off = 1; off < 16; off <<= 1)
+          score[g] += __shfl_xor(score[g], off, 64);
+        if (!valid[u]) score[g] = -INFINITY;
       }
-    }
-    // online update per q head: wave-wide max over this iteration's 4 keys
-    bf16x8 vv;
-    if (valid) vv.u = *reinterpret_cast<const uint32x4*>(vrow + sub * EPL);
+      // online update per q head: wave max over this iteration's 4 keys
 #pragma unroll
-    for (int g = 0; g < GQ; ++g) {
-      const float it_max = group_max4(valid ? score[g] : -INFINITY);
-      if (it_max == -INFINITY) continue;
-      const float m_new = fmaxf(m_run[g], it_max);
-      const float alpha =
-          (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
-      const float p = valid ? __expf(score[g] - m_new) : 0.f;
-      l_run[g] = l_run[g] * alpha + group_sum4(p);
-      m_run[g] = m_new;
+      for (int g = 0; g < GQ; ++g) {
+        const float it_max = group_max4(score[g]);
+        if (it_max == -INFINITY) continue;
+        const float m_new = fmaxf(m_run[g], it_max);
+        const float alpha =
+            (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+        const float p = valid[u] ? __expf(score[g] - m_new) : 0.f;
+        l_run[g] = l_run[g] * alpha + group_sum4(p);
+        m_run[g] = m_new;
 #pragma unroll
-      for (int e = 0; e < EPL; ++e) {
-        const float vval = valid ? to_f32(vv.h[e]) : 0.f;
-        o_acc[g][e] = o_acc[g][e] * alpha + p * vval;
+        for (int e = 0; e < EPL; ++e) {
+          o_acc[g][e] = o_acc[g][e] * alpha + p * to_f32(vvec[u].h[e]);
+        }
       }
     }
   }

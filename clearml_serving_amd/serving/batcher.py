@@ -104,6 +104,18 @@ class DynamicBatcher:
             self._worker_task.cancel()
             self._worker_task = None
 
+    def shutdown(self) -> None:
+        """Thread-safe teardown (called when the endpoint is removed on a
+        config reload): stop the worker and drop graphs/slabs so HBM frees
+        with the engine instance."""
+        self._closed = True
+        task, loop = self._worker_task, self._loop
+        self._worker_task = None
+        if task is not None and loop is not None and not loop.is_closed():
+            loop.call_soon_threadsafe(task.cancel)
+        self._graphs.clear()
+        self._pinned.clear()
+
     # ------------------------------------------------------------------ #
     async def _worker(self) -> None:
         while not self._closed:

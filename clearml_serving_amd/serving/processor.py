@@ -446,7 +446,10 @@ class ModelRequestProcessor:
                 ep = synced.get(url)
                 if ep is None or ep.as_dict() != \
                         self._engine_processor_lookup[url].model_endpoint.as_dict():
-                    self._engine_processor_lookup.pop(url, None)
+                    engine = self._engine_processor_lookup.pop(url, None)
+                    batcher = getattr(engine, "_batcher", None)
+                    if batcher is not None:
+                        batcher.shutdown()  # free graphs/streams/HBM
             self._last_revision = revision
             self._update_lock_flag = False
         import gc

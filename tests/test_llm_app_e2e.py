@@ -1,0 +1,82 @@
+"""OpenAI-compatible routes through the full HTTP app with the native LLM
+engine (llama-tiny on CPU)."""
+
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from clearml_serving_amd.schemas import ModelEndpoint
+from clearml_serving_amd.serving.app import create_app
+
+
+@pytest.fixture()
+def llm_client(processor, store, tmp_path):
+    # reset the per-process engine singleton so each test run is isolated
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+
+    LlmPreprocessRequest._engine_singleton = None
+
+    card = tmp_path / "card.json"
+    card.write_text(json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 128, "device": "cpu",
+    }))
+    rec = store.register_model(name="llama-tiny", project="p", path=str(card))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="llm", serving_url="test_llm", model_id=rec.model_id))
+    processor.serialize()
+    app = create_app(processor=processor, poll_frequency_sec=3600)
+    with TestClient(app) as c:
+        yield c
+    LlmPreprocessRequest._engine_singleton = None
+
+
+def test_chat_completion_route(llm_client):
+    r = llm_client.post("/serve/openai/v1/chat/completions", json={
+        "model": "test_llm", "max_tokens": 4, "temperature": 0.0,
+        "ignore_eos": True,
+        "messages": [{"role": "user", "content": "hello"}],
+    })
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["usage"]["completion_tokens"] == 4
+
+
+def test_completion_route(llm_client):
+    r = llm_client.post("/serve/openai/v1/completions", json={
+        "model": "test_llm", "prompt": "abc", "max_tokens": 3,
+        "temperature": 0.0, "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["object"] == "text_completion"
+
+
+def test_streaming_sse(llm_client):
+    with llm_client.stream("POST", "/serve/openai/v1/chat/completions", json={
+        "model": "test_llm", "max_tokens": 3, "temperature": 0.0,
+        "ignore_eos": True, "stream": True,
+        "messages": [{"role": "user", "content": "hi"}],
+    }) as r:
+        assert r.status_code == 200
+        chunks = list(r.iter_lines())
+    data_lines = [c for c in chunks if c.startswith("data: ")]
+    assert data_lines[-1] == "data: [DONE]"
+    assert len(data_lines) >= 4  # 3 tokens + DONE
+    first = json.loads(data_lines[0][len("data: "):])
+    assert first["object"] == "chat.completion.chunk"
+
+
+def test_models_listing(llm_client):
+    r = llm_client.get("/serve/openai/v1/models")
+    assert r.status_code == 200
+    assert any(m["id"] == "test_llm" for m in r.json()["data"])
+
+
+def test_generic_route_prompt(llm_client):
+    r = llm_client.post("/serve/test_llm", json={
+        "prompt": "xyz", "max_tokens": 2, "temperature": 0.0,
+        "ignore_eos": True})
+    assert r.status_code == 200, r.text
+    assert r.json()["tokens"] == 2

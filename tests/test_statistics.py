@@ -1,0 +1,65 @@
+"""Statistics registry: metric types, naming compatibility, prefix configs."""
+
+from prometheus_client import CollectorRegistry
+
+from clearml_serving_amd.schemas import EndpointMetricLogging, MetricType
+from clearml_serving_amd.statistics.collector import StatsRegistry, _prom_name
+
+
+def test_prom_name_folding():
+    # reference naming: {url}:{var} with non-alnum -> "_"
+    assert _prom_name("model/1", "_latency") == "model_1__latency"
+    assert _prom_name("a-b", "x.y") == "a_b_x_y"
+
+
+class FakeProcessor:
+    def __init__(self, cfg):
+        self._cfg = cfg
+
+    def list_endpoint_logging(self):
+        return self._cfg
+
+
+def test_all_metric_types():
+    cfg = {
+        "ep/1": EndpointMetricLogging(endpoint="ep/1", metrics={
+            "x1": MetricType(type="scalar", buckets=[0, 1, 2]),
+            "detect": MetricType(type="enum", buckets=["cat", "dog"]),
+            "gauge_v": MetricType(type="value"),
+            "cnt": MetricType(type="counter"),
+        }),
+    }
+    reg = CollectorRegistry()
+    sr = StatsRegistry(processor=FakeProcessor(cfg), registry=reg)
+    sr.report_batch([
+        {"_url": "ep/1", "_latency": 0.02, "_count": 2, "x1": 1.5,
+         "detect": "cat", "gauge_v": 42.0, "cnt": 3},
+        {"_url": "ep/1", "_latency": 0.2, "_count": 2, "detect": "dog"},
+    ])
+    assert reg.get_sample_value("ep_1__count_total") == 4.0
+    assert reg.get_sample_value("ep_1__latency_count") == 2.0
+    # scalar histogram with custom buckets
+    assert reg.get_sample_value("ep_1_x1_bucket", {"le": "2.0"}) == 1.0
+    # enum counters per value
+    assert reg.get_sample_value("ep_1_detect_total", {"value": "cat"}) == 1.0
+    assert reg.get_sample_value("ep_1_detect_total", {"value": "dog"}) == 1.0
+    assert reg.get_sample_value("ep_1_gauge_v") == 42.0
+    assert reg.get_sample_value("ep_1_cnt_total") == 3.0
+
+
+def test_prefix_metric_config_applies():
+    cfg = {"models/*": EndpointMetricLogging(endpoint="models/*", metrics={
+        "score": MetricType(type="scalar", buckets=[0, 0.5, 1.0])})}
+    reg = CollectorRegistry()
+    sr = StatsRegistry(processor=FakeProcessor(cfg), registry=reg)
+    sr.report_batch([{"_url": "models/alpha/1", "score": 0.7}])
+    assert reg.get_sample_value(
+        "models_alpha_1_score_bucket", {"le": "1.0"}) == 1.0
+
+
+def test_unconfigured_endpoint_gets_reserved_metrics():
+    reg = CollectorRegistry()
+    sr = StatsRegistry(processor=FakeProcessor({}), registry=reg)
+    sr.report_batch([{"_url": "ghost", "_latency": 0.01, "_count": 1}])
+    assert reg.get_sample_value("ghost__count_total") == 1.0
+    assert reg.get_sample_value("ghost__latency_count") == 1.0
