@@ -41,10 +41,16 @@ class BertSelfAttention(nn.Module):
     def forward(self, x: torch.Tensor, seq_lens: Optional[torch.Tensor]):
         b, s, h = x.shape
         qkv = self.qkv(x)  # [B, S, 3H]
-        qkv = qkv.view(b, s, 3, self.heads, self.head_dim)
-        q, k, v = (qkv[:, :, i].transpose(1, 2).contiguous() for i in range(3))
-        ctx = ops.attention(q, k, v, causal=False, seq_lens=seq_lens)
-        ctx = ctx.transpose(1, 2).reshape(b, s, h)
+        # [B, S, H, D] strided views: the attention kernel is
+        # layout-agnostic, so no transpose copies
+        hd = self.head_dim
+        q, k, v = qkv.split([h, h, h], dim=-1)
+        q = q.unflatten(-1, (self.heads, hd))
+        k = k.unflatten(-1, (self.heads, hd))
+        v = v.unflatten(-1, (self.heads, hd))
+        ctx = ops.attention(q, k, v, causal=False, seq_lens=seq_lens,
+                            layout="bshd")
+        ctx = ctx.reshape(b, s, h)
         return self.out(ctx), self.out_bias
 
 

@@ -93,12 +93,12 @@ class LlamaLayer(nn.Module):
         scale = 1.0 / math.sqrt(self.head_dim)
         if attn_ctx["mode"] == "prefill":
             b, s = attn_ctx["batch"], attn_ctx["seq"]
-            qb = q.unflatten(0, (b, s)).transpose(1, 2).contiguous()
-            kb = k.unflatten(0, (b, s)).transpose(1, 2).contiguous()
-            vb = v.unflatten(0, (b, s)).transpose(1, 2).contiguous()
+            qb = q.unflatten(0, (b, s))  # [B, S, H, D] strided views
+            kb = k.unflatten(0, (b, s))
+            vb = v.unflatten(0, (b, s))
             ctx = ops.attention(qb, kb, vb, causal=True, scale=scale,
-                                seq_lens=attn_ctx["seq_lens"])
-            ctx = ctx.transpose(1, 2).reshape(t, self.heads * self.head_dim)
+                                seq_lens=attn_ctx["seq_lens"], layout="bshd")
+            ctx = ctx.reshape(t, self.heads * self.head_dim)
         else:  # decode: one token per sequence
             k_cache, v_cache = kv_cache
             ctx = ops.attention_decode(

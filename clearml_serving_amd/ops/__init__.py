@@ -166,19 +166,28 @@ def attention(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     causal: bool = False, scale: Optional[float] = None,
     seq_lens: Optional[torch.Tensor] = None,
+    layout: str = "bhsd",
 ) -> torch.Tensor:
     """Fused scaled-dot-product attention (prefill / encoder).
 
-    q,k,v: [B, H, S, D] (same S for q and k/v here; GQA via H_kv divides H).
-    ``seq_lens`` (int32 [B]) masks keys >= len (padding). Output [B, H, S, D].
+    layout "bhsd": q,k,v = [B, H, S, D]; layout "bshd": [B, S, H, D]
+    (strided views into a merged QKV projection are accepted -- no
+    transpose copies). GQA via H_kv dividing H; ``seq_lens`` (int32 [B])
+    masks keys >= len. Output has the input layout.
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
+    bshd = layout == "bshd"
     if q.is_cuda:
         ext = _require_ext("attention")
         if ext is not None:
             return ext.attention_prefill(q, k, v, bool(causal), float(scale),
-                                         seq_lens)
+                                         seq_lens, bshd)
+    if bshd:
+        out = attention(q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3),
+                        v.permute(0, 2, 1, 3), causal=causal, scale=scale,
+                        seq_lens=seq_lens)
+        return out.permute(0, 2, 1, 3).contiguous()
     # reference path (fp32 math)
     qf, kf, vf = q.float(), k.float(), v.float()
     hq, hkv = q.shape[1], k.shape[1]

@@ -50,13 +50,20 @@ __device__ __forceinline__ float rowgroup_sum(float v) {
   return v;
 }
 
+struct AttnStrides {
+  long qb, qh, qs;   // q/out: batch, head, seq strides (elements)
+  long kb, kh, ks;   // k/v share layout
+  long ob, oh, os;
+};
+
 template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS>
 __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
-    const __hip_bfloat16* __restrict__ q,   // [B, H, Sq, D]
-    const __hip_bfloat16* __restrict__ k,   // [B, Hkv, Sk, D]
-    const __hip_bfloat16* __restrict__ v,   // [B, Hkv, Sk, D]
-    __hip_bfloat16* __restrict__ out,       // [B, H, Sq, D]
+    const __hip_bfloat16* __restrict__ q,   // [B, H, Sq, D] via strides
+    const __hip_bfloat16* __restrict__ k,   // [B, Hkv, Sk, D] via strides
+    const __hip_bfloat16* __restrict__ v,
+    __hip_bfloat16* __restrict__ out,
     const int* __restrict__ seq_lens,       // [B] or null
+    AttnStrides st,
     int B, int H, int Hkv, int Sq, int Sk, float scale) {
   constexpr int D = HEAD_DIM;
   constexpr int KSTRIDE = D + PAD;          // LDS K row stride (bf16)
@@ -84,8 +91,9 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   const int m0 = m_tile * BLOCK_M;          // first query row of the block
   const int wm0 = m0 + wave * 16;           // this wave's first query row
 
-  const long q_base = (((long)b * H + h) * Sq) * D;
-  const long kv_base = (((long)b * Hkv + hkv) * Sk) * D;
+  const long q_base = (long)b * st.qb + (long)h * st.qh;
+  const long kv_base = (long)b * st.kb + (long)hkv * st.kh;
+  const long o_base = (long)b * st.ob + (long)h * st.oh;
   const int kv_len = HAS_SEQLENS ? min(seq_lens[b], Sk) : Sk;
   // causal: queries attend to keys <= q_idx (Sq == Sk alignment)
   const int kv_hi = CAUSAL ? min(kv_len, m0 + BLOCK_M + (Sk - Sq)) : kv_len;
@@ -98,7 +106,8 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   for (int kc = 0; kc < D / 32; ++kc) {
     const int qrow = wm0 + frag_row;
     if (qrow < Sq) {
-      const __hip_bfloat16* src = q + q_base + (long)qrow * D + kc * 32 + frag_ko;
+      const __hip_bfloat16* src =
+          q + q_base + (long)qrow * st.qs + kc * 32 + frag_ko;
       q_frag[kc] = *reinterpret_cast<const bf16x8_t*>(src);
     } else {
       q_frag[kc] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
@@ -126,7 +135,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
         bf16x8_t val{0, 0, 0, 0, 0, 0, 0, 0};
         if (gkey < kv_len) {
           val = *reinterpret_cast<const bf16x8_t*>(
-              k + kv_base + (long)gkey * D + d8);
+              k + kv_base + (long)gkey * st.ks + d8);
         }
         *reinterpret_cast<bf16x8_t*>(&lds_k[key * KSTRIDE + d8]) = val;
         // V: same piece indexing, scatter-transposed (8 ds_write_b16).
@@ -137,7 +146,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
         bf16x8_t vv{0, 0, 0, 0, 0, 0, 0, 0};
         if (gkey < kv_len) {
           vv = *reinterpret_cast<const bf16x8_t*>(
-              v + kv_base + (long)gkey * D + d8);
+              v + kv_base + (long)gkey * st.ks + d8);
         }
         const int kswz = key ^ (((d8 >> 3) & 7) << 3);
 #pragma unroll
@@ -250,7 +259,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     const int qrow = row_base + r;
     if (qrow >= Sq) continue;
     const float inv_l = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
-    __hip_bfloat16* dst = out + q_base + (long)qrow * D;
+    __hip_bfloat16* dst = out + o_base + (long)qrow * st.os;
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
       dst[dt * 16 + col] = __float2bfloat16(acc_o[dt][r] * inv_l);
@@ -262,18 +271,31 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 
 torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
                                 torch::Tensor v, bool causal, double scale,
-                                c10::optional<torch::Tensor> seq_lens) {
+                                c10::optional<torch::Tensor> seq_lens,
+                                bool bshd) {
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4,
-              "q/k/v must be [B, H, S, D]");
+              "q/k/v must be 4-D");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
-  const int B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
-  const int Hkv = k.size(1), Sk = k.size(2);
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "head_dim must be dense");
+  // bshd: tensors are [B, S, H, D] (e.g. views into a merged QKV
+  // projection) -- no transpose copies; default is [B, H, S, D]
+  const int hdim = bshd ? 2 : 1, sdim = bshd ? 1 : 2;
+  const int B = q.size(0), H = q.size(hdim), Sq = q.size(sdim),
+            D = q.size(3);
+  const int Hkv = k.size(hdim), Sk = k.size(sdim);
   TORCH_CHECK(H % Hkv == 0, "H must be a multiple of H_kv");
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
-  TORCH_CHECK(v.size(1) == Hkv && v.size(2) == Sk && v.size(3) == D);
+  TORCH_CHECK(v.size(hdim) == Hkv && v.size(sdim) == Sk && v.size(3) == D);
+  TORCH_CHECK(k.stride(0) == v.stride(0) && k.stride(1) == v.stride(1) &&
+              k.stride(2) == v.stride(2), "k/v must share layout");
 
-  auto out = torch::empty_like(q);
+  auto out = bshd ? torch::empty({B, Sq, H, D}, q.options())
+                  : torch::empty({B, H, Sq, D}, q.options());
+  AttnStrides st;
+  st.qb = q.stride(0); st.qh = q.stride(hdim); st.qs = q.stride(sdim);
+  st.kb = k.stride(0); st.kh = k.stride(hdim); st.ks = k.stride(sdim);
+  st.ob = out.stride(0); st.oh = out.stride(hdim); st.os = out.stride(sdim);
   const int* sl = nullptr;
   torch::Tensor sl_t;
   if (seq_lens.has_value() && seq_lens->defined()) {
@@ -291,8 +313,8 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
                      (const __hip_bfloat16*)q.data_ptr(),                    \
                      (const __hip_bfloat16*)k.data_ptr(),                    \
                      (const __hip_bfloat16*)v.data_ptr(),                    \
-                     (__hip_bfloat16*)out.data_ptr(), sl, B, H, Hkv, Sq, Sk, \
-                     (float)scale)
+                     (__hip_bfloat16*)out.data_ptr(), sl, st, B, H, Hkv,     \
+                     Sq, Sk, (float)scale)
   if (D == 64) {
     if (causal) { if (sl) LAUNCH_ATTN(64, true, true); else LAUNCH_ATTN(64, true, false); }
     else        { if (sl) LAUNCH_ATTN(64, false, true); else LAUNCH_ATTN(64, false, false); }

@@ -13,7 +13,8 @@ torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor weight, double eps,
 torch::Tensor softmax_lastdim(torch::Tensor x);
 torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
                                 torch::Tensor v, bool causal, double scale,
-                                c10::optional<torch::Tensor> seq_lens);
+                                c10::optional<torch::Tensor> seq_lens,
+                                bool bshd);
 torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
                                  long top_k, double top_p, long seed);
 torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
@@ -41,7 +42,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_lastdim", &softmax_lastdim);
   m.def("attention_prefill", &attention_prefill, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal"), py::arg("scale"),
-        py::arg("seq_lens") = py::none());
+        py::arg("seq_lens") = py::none(), py::arg("bshd") = false);
   m.def("sample_top_k_top_p", &sample_top_k_top_p);
   m.def("rope_inplace", &rope_inplace);
   m.def("attention_decode", &attention_decode);

@@ -281,3 +281,20 @@ def test_gemm_bf16_transpose_detecting():
     got = ops.gemm_bf16(a, b)
     ref = a.float() @ b.float().T
     assert_close_bf16(got, ref, atol=1e-2, rtol=1e-2)
+
+
+def test_attention_bshd_layout_strided_views():
+    # [B, S, H, D] strided views into a merged projection (the serving path)
+    b, s, h, d = 2, 128, 8, 64
+    torch.manual_seed(9)
+    qkv = torch.randn(b, s, 3 * h * d, device=DEV, dtype=torch.bfloat16)
+    q, k, v = qkv.split([h * d, h * d, h * d], dim=-1)
+    q = q.unflatten(-1, (h, d))
+    k = k.unflatten(-1, (h, d))
+    v = v.unflatten(-1, (h, d))
+    got = ops.attention(q, k, v, layout="bshd")
+    ref = ops.attention(
+        q.permute(0, 2, 1, 3).float().cpu(),
+        k.permute(0, 2, 1, 3).float().cpu(),
+        v.permute(0, 2, 1, 3).float().cpu()).permute(0, 2, 1, 3)
+    assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
