@@ -167,6 +167,7 @@ class Sequence:
         self.req_id = req_id
         self.prompt_ids = prompt_ids
         self.output_ids: List[int] = []
+        self.generated = 0  # survives preemption (output folds into prompt)
         self.params = params
         self.blocks: List[int] = []
         self.stream: "asyncio.Queue" = asyncio.Queue()
@@ -196,7 +197,8 @@ class LlmEngine:
         self._wake: Optional[asyncio.Event] = None
         self._started = False
         self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
-                      "prefill_batches": 0, "decode_batches": 0}
+                      "prefill_batches": 0, "decode_batches": 0,
+                      "preemptions": 0}
 
     # ------------------------------------------------------------------ #
     def start(self) -> None:
@@ -406,14 +408,44 @@ class LlmEngine:
             kv_caches=self.kv_caches, attn_ctx=attn_ctx,
             last_token_idx=last_idx)
 
+    def _preempt_one(self) -> bool:
+        """KV pressure relief: evict the newest running sequence back to the
+        waiting queue (vLLM-style recompute preemption -- its prompt plus
+        generated tokens re-prefill when pages free up)."""
+        candidates = [s for s in self.running if not s.finished]
+        if len(candidates) <= 1:
+            return False
+        victim = max(candidates, key=lambda s: s.created)
+        self.running.remove(victim)
+        self.allocator.free(victim.blocks)
+        victim.blocks = []
+        victim.prompt_ids = victim.prompt_ids + victim.output_ids
+        victim.output_ids = []
+        self.waiting.insert(0, victim)
+        self.stats["preemptions"] = self.stats.get("preemptions", 0) + 1
+        return True
+
     def _decode(self, seqs: List[Sequence]) -> None:
         self.stats["decode_batches"] += 1
         bs_cfg = self.cfg.block_size
-        # grow block tables for the token being generated
-        for s in seqs:
+        # grow block tables for the token being generated; under KV pressure
+        # preempt the newest sequences instead of failing the whole engine
+        for s in list(seqs):
             pos = len(s) - 1  # position of the last generated token
             if pos // bs_cfg >= len(s.blocks):
+                while self.allocator.available < 1:
+                    if not self._preempt_one():
+                        raise RuntimeError(
+                            "KV cache exhausted with a single sequence -- "
+                            "raise num_kv_blocks / gpu_memory_fraction")
+                    if s not in self.running:  # we evicted s itself
+                        break
+                if s not in self.running:
+                    seqs.remove(s)
+                    continue
                 s.blocks.extend(self.allocator.alloc(1))
+        if not seqs:
+            return
         plan = {
             "mode": "decode",
             "tokens": [s.output_ids[-1] for s in seqs],
@@ -499,6 +531,7 @@ class LlmEngine:
         for i, s in enumerate(seqs):
             tok = int(next_ids[i])
             s.output_ids.append(tok)
+            s.generated += 1
             if s.first_token_time is None:
                 s.first_token_time = now
             self.stats["generated_tokens"] += 1
@@ -507,7 +540,7 @@ class LlmEngine:
             if not s.params.ignore_eos and (
                     tok == eos or tok in s.params.stop_token_ids):
                 finished, reason = True, "stop"
-            elif len(s.output_ids) >= s.params.max_tokens:
+            elif s.generated >= s.params.max_tokens:
                 finished, reason = True, "length"
             elif len(s) >= self.cfg.max_model_len:
                 finished, reason = True, "length"

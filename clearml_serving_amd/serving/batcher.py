@@ -79,8 +79,9 @@ class DynamicBatcher:
         self._enqueue_lock = threading.Lock()
         self._inflight = set()
         self._closed = False
-        # telemetry for the Prometheus exporter
-        self.stats = {"batches": 0, "requests": 0, "occupancy_sum": 0.0}
+        # telemetry for the Prometheus exporter / GET /status
+        self.stats = {"batches": 0, "requests": 0, "occupancy_sum": 0.0,
+                      "stage_ms_sum": 0.0, "gpu_wait_ms_sum": 0.0}
 
     # ------------------------------------------------------------------ #
     async def submit(self, inputs: TensorOrDict) -> TensorOrDict:
@@ -192,6 +193,7 @@ class DynamicBatcher:
 
         slot = self._slots.get()  # bounds in-flight batches (2)
         try:
+            t_stage = time.monotonic()
             # CPU staging into this slot's pinned slab overlaps the other
             # slot's GPU execution
             if is_dict:
@@ -208,7 +210,11 @@ class DynamicBatcher:
                     out_cpu = self._pinned_out(out, slot)
                 done = torch.cuda.Event()
                 done.record(self._stream)
+            t_enq = time.monotonic()
             done.synchronize()
+            t_done = time.monotonic()
+            self.stats["stage_ms_sum"] += (t_enq - t_stage) * 1000
+            self.stats["gpu_wait_ms_sum"] += (t_done - t_enq) * 1000
             # slices are views into the reusable pinned slab: clone them
             return [_slice_clone(out_cpu, i) for i in range(n)]
         finally:

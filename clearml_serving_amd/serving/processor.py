@@ -118,8 +118,21 @@ class ModelRequestProcessor:
                 name=name or "Serving-Service", project=project or "DevOps",
                 tags=tags,
             )
+            from .. import __version__
+
+            self._store.set_params(self._session_id,
+                                   {"serving_version": __version__})
         else:
             self._session_id = self._store.resolve_session(task_id)
+            # version guard (reference: __main__.py:24-40): warn when the
+            # session was created by a different package version
+            from .. import __version__
+
+            created_by = self._store.get_params(self._session_id).get(
+                "serving_version")
+            if created_by and created_by.split(".")[0] !=                     __version__.split(".")[0]:
+                print("Warning: serving session created by version {} but "
+                      "this package is {}".format(created_by, __version__))
         # give engine instances access to store + session (preprocess.py)
         self._store._session_id = self._session_id
 

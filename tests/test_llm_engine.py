@@ -172,3 +172,34 @@ def test_config_from_aux(tmp_path):
     assert cfg.block_size == 32
     assert cfg.max_num_seqs == 8
     assert cfg.max_model_len == 512
+
+
+def test_kv_pressure_preemption():
+    """With too few KV blocks for all sequences, the engine preempts the
+    newest sequence (recompute) instead of failing -- all requests still
+    complete."""
+    eng = tiny_engine()
+    # shrink capacity: 6 blocks = 96 token slots total
+    from clearml_serving_amd.engines.llm.engine import BlockAllocator
+
+    eng.allocator = BlockAllocator(6)
+
+    async def main():
+        # 30-token prompts + 40 generated = 5 blocks per finished sequence;
+        # two admitted sequences need 10 blocks > 6 -> decode growth must
+        # preempt (admission control alone cannot absorb it)
+        params = SamplingParams(temperature=0.0, max_tokens=40,
+                                ignore_eos=True)
+
+        async def one(i):
+            toks = []
+            async for item in eng.generate("p%d" % i + "x" * 28, params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return await asyncio.gather(*[one(i) for i in range(4)])
+
+    outs = run(main())
+    assert all(len(o) == 40 for o in outs)
+    assert eng.allocator.available == 6
+    assert eng.stats["preemptions"] >= 1
