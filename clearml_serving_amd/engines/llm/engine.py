@@ -218,9 +218,20 @@ class LlmEngine:
         model = LlamaForCausalLM(mcfg, tp_rank=self.tp_rank,
                                  tp_size=self.tp_size)
         if cfg.weights:
-            from ...models import load_weights
+            if self.tp_size > 1:
+                # full checkpoint -> this rank's shard
+                from safetensors.torch import load_file
 
-            load_weights(model, cfg.weights)
+                full = (load_file(cfg.weights)
+                        if cfg.weights.endswith(".safetensors")
+                        else torch.load(cfg.weights, map_location="cpu",
+                                        weights_only=True))
+                model.load_state_dict(tp_mod.shard_llama_weights(
+                    full, mcfg, self.tp_rank, self.tp_size))
+            else:
+                from ...models import load_weights
+
+                load_weights(model, cfg.weights)
         self.model = model.eval().to(self.device).to(self.dtype)
 
         self.tokenizer = (HfTokenizer(cfg.tokenizer_path)
