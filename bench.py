@@ -27,7 +27,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-REQ_PER_STEP = int(os.environ.get("CMLS_BENCH_REQ", 1024))
+REQ_PER_STEP = int(os.environ.get("CMLS_BENCH_REQ", 512))
 # per rank per step: REQ/2 ResNet-50 + REQ/2 BERT-base requests
 
 

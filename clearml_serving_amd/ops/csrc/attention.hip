@@ -70,8 +70,13 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   constexpr int VSTRIDE = BLOCK_N + PAD;    // LDS V^T row stride
   constexpr int PSTRIDE = BLOCK_N + PAD;
 
-  __shared__ short lds_k[BLOCK_N * KSTRIDE];
-  __shared__ short lds_vt[D * VSTRIDE];
+  // double-buffered K/V tiles: tile n+1's global loads are issued into
+  // registers BEFORE computing on tile n (T14 issue-early / write-late) --
+  // HBM latency hides under the MFMA/softmax work, and the ds_write of the
+  // next tile lands in the other buffer (no barrier between compute and
+  // write, only before the swapped buffer is read)
+  __shared__ short lds_k[2][BLOCK_N * KSTRIDE];
+  __shared__ short lds_vt[2][D * VSTRIDE];
   __shared__ short lds_p[NWAVES * 16 * PSTRIDE];
 
   // grid = (B*H, n_m_tiles): bh on x so the dispatcher's XCD round-robin
@@ -122,38 +127,53 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
   for (int dt = 0; dt < D / 16; ++dt) acc_o[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-  // ================= KV tile loop ================= //
-  for (int n0 = 0; n0 < kv_hi; n0 += BLOCK_N) {
-    // ---- stage K [64][D] row-major and V^T [D][64] into LDS ---- //
-    // 256 threads; each loads 16-B pieces. K: row = key, col = d.
-    {
-      constexpr int pieces = BLOCK_N * D / 8;       // 8 bf16 per piece
-      for (int p = tid; p < pieces; p += 256) {
-        const int key = p / (D / 8);
-        const int d8 = (p % (D / 8)) * 8;
-        const int gkey = n0 + key;
-        bf16x8_t val{0, 0, 0, 0, 0, 0, 0, 0};
-        if (gkey < kv_len) {
-          val = *reinterpret_cast<const bf16x8_t*>(
-              k + kv_base + (long)gkey * st.ks + d8);
-        }
-        *reinterpret_cast<bf16x8_t*>(&lds_k[key * KSTRIDE + d8]) = val;
-        // V: same piece indexing, scatter-transposed (8 ds_write_b16).
-        // key-XOR swizzle: unswizzled, the 16 lanes sharing a key write
-        // addresses 8*VSTRIDE*2 = 1152 B apart = all on one bank (16-way
-        // conflict, measured 64% of kernel wall); XOR key bits 3-5 with the
-        // d-row block spreads them over 8 banks (2-way).
-        bf16x8_t vv{0, 0, 0, 0, 0, 0, 0, 0};
-        if (gkey < kv_len) {
-          vv = *reinterpret_cast<const bf16x8_t*>(
-              v + kv_base + (long)gkey * st.ks + d8);
-        }
-        const int kswz = key ^ (((d8 >> 3) & 7) << 3);
+  // staging helpers: each thread owns PIECES 16-B pieces of the K and V
+  // tiles (piece p -> key p/(D/8), d-offset (p%(D/8))*8)
+  constexpr int PIECES = BLOCK_N * D / 8 / 256;
+  bf16x8_t kreg[PIECES], vreg[PIECES];
+
+  auto stage_load = [&](int n0) {
 #pragma unroll
-        for (int e = 0; e < 8; ++e) lds_vt[(d8 + e) * VSTRIDE + kswz] = vv[e];
+    for (int i = 0; i < PIECES; ++i) {
+      const int p = tid + i * 256;
+      const int gkey = n0 + p / (D / 8);
+      const int d8 = (p % (D / 8)) * 8;
+      if (gkey < kv_len) {
+        kreg[i] = *reinterpret_cast<const bf16x8_t*>(
+            k + kv_base + (long)gkey * st.ks + d8);
+        vreg[i] = *reinterpret_cast<const bf16x8_t*>(
+            v + kv_base + (long)gkey * st.ks + d8);
+      } else {
+        kreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        vreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
-    __syncthreads();
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < PIECES; ++i) {
+      const int p = tid + i * 256;
+      const int key = p / (D / 8);
+      const int d8 = (p % (D / 8)) * 8;
+      *reinterpret_cast<bf16x8_t*>(&lds_k[buf][key * KSTRIDE + d8]) = kreg[i];
+      // V scatter-transposed with the key-XOR bank swizzle (see PV reads)
+      const int kswz = key ^ (((d8 >> 3) & 7) << 3);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        lds_vt[buf][(d8 + e) * VSTRIDE + kswz] = vreg[i][e];
+    }
+  };
+
+  // ================= KV tile loop ================= //
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  int cur = 0;
+  for (int n0 = 0; n0 < kv_hi; n0 += BLOCK_N) {
+    // issue next tile's global loads now; they complete under this tile's
+    // MFMA + softmax work
+    const bool has_next = n0 + BLOCK_N < kv_hi;
+    if (has_next) stage_load(n0 + BLOCK_N);
 
     // ---- S = Q K^T for this wave's 16 rows x 64 keys ---- //
     f32x4_t acc_s[BLOCK_N / 16];
@@ -165,7 +185,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
       for (int t = 0; t < BLOCK_N / 16; ++t) {
         // B[k][j] = K[j + 16t][k]: contiguous 8 bf16 at row j, col k-offset
         const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
-            &lds_k[(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
+            &lds_k[cur][(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
         acc_s[t] = MFMA_16x16x32(q_frag[kc], kf, acc_s[t]);
       }
     }
@@ -244,11 +264,15 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
         const int vrow = dt * 16 + frag_row;
         const int vkey = (kc * 32 + frag_ko) ^ (((vrow >> 3) & 7) << 3);
         const bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-            &lds_vt[vrow * VSTRIDE + vkey]);
+            &lds_vt[cur][vrow * VSTRIDE + vkey]);
         acc_o[dt] = MFMA_16x16x32(pf, vf, acc_o[dt]);
       }
     }
-    __syncthreads();  // K/V tiles reused next iteration
+    if (has_next) {
+      stage_write(cur ^ 1);   // other buffer: no read/write hazard with cur
+      __syncthreads();        // next iteration reads the freshly written buf
+      cur ^= 1;
+    }
   }
 
   // ---- epilogue: O / l, bf16 store ---- //
