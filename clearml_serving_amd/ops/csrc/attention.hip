@@ -33,9 +33,11 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 #define MFMA_16x16x32(A, B, C) \
   __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
 
-constexpr int BLOCK_M = 64;   // query rows per workgroup
+constexpr int BLOCK_M = 128;  // query rows per workgroup
 constexpr int BLOCK_N = 64;   // keys per tile
-constexpr int NWAVES = 4;     // one 16-row M-slice per wave
+constexpr int NWAVES = 8;     // one 16-row M-slice per wave; 8 waves =
+                              // 2 waves/SIMD so MFMA + staging latency hide
+                              // across waves even at 1 block/CU
 constexpr int PAD = 8;        // bf16 elements of row padding (16 B)
 
 // row-group reduction: combine over the 16 lanes holding one C-layout row
@@ -57,7 +59,7 @@ struct AttnStrides {
 };
 
 template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS>
-__global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
+__global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
     const __hip_bfloat16* __restrict__ q,   // [B, H, Sq, D] via strides
     const __hip_bfloat16* __restrict__ k,   // [B, Hkv, Sk, D] via strides
     const __hip_bfloat16* __restrict__ v,
@@ -129,13 +131,13 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 
   // staging helpers: each thread owns PIECES 16-B pieces of the K and V
   // tiles (piece p -> key p/(D/8), d-offset (p%(D/8))*8)
-  constexpr int PIECES = BLOCK_N * D / 8 / 256;
+  constexpr int PIECES = BLOCK_N * D / 8 / (NWAVES * 64);
   bf16x8_t kreg[PIECES], vreg[PIECES];
 
   auto stage_load = [&](int n0) {
 #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
-      const int p = tid + i * 256;
+      const int p = tid + i * NWAVES * 64;
       const int gkey = n0 + p / (D / 8);
       const int d8 = (p % (D / 8)) * 8;
       if (gkey < kv_len) {
@@ -152,7 +154,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   auto stage_write = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
-      const int p = tid + i * 256;
+      const int p = tid + i * NWAVES * 64;
       const int key = p / (D / 8);
       const int d8 = (p % (D / 8)) * 8;
       *reinterpret_cast<bf16x8_t*>(&lds_k[buf][key * KSTRIDE + d8]) = kreg[i];
@@ -328,7 +330,7 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
     sl = sl_t.data_ptr<int>();
   }
   dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
-  dim3 block(256);
+  dim3 block(NWAVES * 64);
   hipStream_t stream_ = cmls::current_stream();
 
 #define LAUNCH_ATTN(DD, CC, SS)                                              \
