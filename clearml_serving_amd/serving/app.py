@@ -85,8 +85,13 @@ def create_app(
         # single-node-native is a direct in-process registry -- §5.5 SURVEY)
         try:
             from ..statistics.collector import install_stats_sink
+            from ..statistics.kafka_forwarder import maybe_install
 
-            install_stats_sink(proc)
+            kafka = (proc.store.get_params(proc.get_id())
+                     .get("kafka_service_server")
+                     or os.environ.get("CLEARML_DEFAULT_KAFKA_SERVE_URL"))
+            if not maybe_install(proc, kafka):
+                install_stats_sink(proc)  # in-process Prometheus (default)
         except Exception as ex:
             proc._report_text("statistics sink unavailable: {}".format(ex))
         app.state.processor = proc
