@@ -203,6 +203,17 @@ def main():
             }
             print(json.dumps(result), flush=True)
 
+        # per-endpoint batcher telemetry to stderr (optimization guidance)
+        for url, eng in processor._engine_processor_lookup.items():
+            b = getattr(eng, "_batcher", None)
+            if b is not None and b.stats["batches"]:
+                s = b.stats
+                log("{}: batches={} occupancy={:.2f} stage_ms/b={:.2f} "
+                    "gpu_wait_ms/b={:.2f}".format(
+                        url, s["batches"],
+                        s["occupancy_sum"] / s["batches"],
+                        s["stage_ms_sum"] / s["batches"],
+                        s["gpu_wait_ms_sum"] / s["batches"]))
         processor.stop()
         if dist:
             dist.destroy_process_group()
