@@ -120,7 +120,10 @@ def main():
 
     with tempfile.TemporaryDirectory() as tmpdir:
         store_root, session_id = setup_store(tmpdir)
-        proc = multiprocessing.Process(
+        # spawn: the parent touched the HIP runtime; a forked child cannot
+        # re-initialize it
+        ctx = multiprocessing.get_context("spawn")
+        proc = ctx.Process(
             target=server_proc, args=(store_root, session_id, PORT),
             daemon=True)
         proc.start()

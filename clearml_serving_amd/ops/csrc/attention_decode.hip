@@ -105,7 +105,7 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   // was latency-bound at ~1.7 TB/s
   const int per_iter = NW * GROUPS;
   const int n_iters = (n_keys - key_lo + per_iter - 1) / per_iter;
-  constexpr int UNROLL = 8;
+  constexpr int UNROLL = 4;
   for (int it0 = 0; it0 < n_iters; it0 += UNROLL) {
     bf16x8 kvec[UNROLL], vvec[UNROLL];
     bool valid[UNROLL];
