@@ -25,14 +25,23 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 PORT = int(os.environ.get("CMLS_HTTP_BENCH_PORT", 18080))
 
 
-def server_proc(store_root, session_id, port):
+def server_proc(store_root, session_id, port, workers):
     import uvicorn
 
-    from clearml_serving_amd.serving.app import create_app
+    if workers > 1:
+        # multi-worker: app built per worker process from env (the
+        # reference's gunicorn topology -- N workers, N engine copies)
+        os.environ["CLEARML_SERVING_TASK_ID"] = session_id
+        os.environ["CLEARML_SERVING_AMD_STORE"] = store_root
+        os.environ["CLEARML_SERVING_POLL_FREQ"] = "60"
+        uvicorn.run("clearml_serving_amd.serving.app:app", host="127.0.0.1",
+                    port=port, log_level="warning", workers=workers)
+    else:
+        from clearml_serving_amd.serving.app import create_app
 
-    app = create_app(session_id=session_id, store_root=store_root,
-                     poll_frequency_sec=3600)
-    uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
+        app = create_app(session_id=session_id, store_root=store_root,
+                         poll_frequency_sec=3600)
+        uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
 
 
 def setup_store(tmpdir):
@@ -116,6 +125,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("-n", type=int, default=8000)
     ap.add_argument("-c", type=int, default=128)
+    ap.add_argument("--workers", type=int, default=1)
     args = ap.parse_args()
 
     with tempfile.TemporaryDirectory() as tmpdir:
@@ -124,8 +134,8 @@ def main():
         # re-initialize it
         ctx = multiprocessing.get_context("spawn")
         proc = ctx.Process(
-            target=server_proc, args=(store_root, session_id, PORT),
-            daemon=True)
+            target=server_proc,
+            args=(store_root, session_id, PORT, args.workers), daemon=True)
         proc.start()
         payload = {"input_ids": list(range(1, 129)),
                    "attention_mask": [1] * 128}
