@@ -133,9 +133,10 @@ def main():
         # spawn: the parent touched the HIP runtime; a forked child cannot
         # re-initialize it
         ctx = multiprocessing.get_context("spawn")
+        # not daemonic: uvicorn's multi-worker mode spawns child processes
         proc = ctx.Process(
             target=server_proc,
-            args=(store_root, session_id, PORT, args.workers), daemon=True)
+            args=(store_root, session_id, PORT, args.workers), daemon=False)
         proc.start()
         payload = {"input_ids": list(range(1, 129)),
                    "attention_mask": [1] * 128}
