@@ -30,7 +30,14 @@ def server_proc(store_root, session_id, port, workers):
 
     if workers > 1:
         # multi-worker: app built per worker process from env (the
-        # reference's gunicorn topology -- N workers, N engine copies)
+        # reference's gunicorn topology -- N workers, N engine copies).
+        # uvicorn's worker bootstrap fdopens stdin, which the spawned
+        # context closed -- give it /dev/null
+        import sys
+
+        devnull = os.open(os.devnull, os.O_RDONLY)
+        os.dup2(devnull, 0)
+        sys.stdin = os.fdopen(0)
         os.environ["CLEARML_SERVING_TASK_ID"] = session_id
         os.environ["CLEARML_SERVING_AMD_STORE"] = store_root
         os.environ["CLEARML_SERVING_POLL_FREQ"] = "60"
