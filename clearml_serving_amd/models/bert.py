@@ -55,8 +55,10 @@ class BertSelfAttention(nn.Module):
 
 
 class BertLayer(nn.Module):
-    def __init__(self, hidden: int, heads: int, intermediate: int):
+    def __init__(self, hidden: int, heads: int, intermediate: int,
+                 ln_eps: float = 1e-12):
         super().__init__()
+        self.ln_eps = ln_eps
         self.attn = BertSelfAttention(hidden, heads)
         self.ln1_w = nn.Parameter(torch.ones(hidden))
         self.ln1_b = nn.Parameter(torch.zeros(hidden))
@@ -71,26 +73,28 @@ class BertLayer(nn.Module):
         attn_out, attn_bias = self.attn(x, seq_lens)
         # fused: LN(x + attn_out + bias)
         x = ops.layernorm(attn_out + attn_bias, self.ln1_w, self.ln1_b,
-                          residual=x)
+                          eps=self.ln_eps, residual=x)
         mlp = self.fc2(ops.bias_gelu(self.fc1(x), self.fc1_bias))
         return ops.layernorm(mlp + self.fc2_bias, self.ln2_w, self.ln2_b,
-                             residual=x)
+                             eps=self.ln_eps, residual=x)
 
 
 class BertEncoder(nn.Module):
     def __init__(
         self, vocab_size: int = 30522, hidden: int = 768, layers: int = 12,
         heads: int = 12, intermediate: int = 3072, max_pos: int = 512,
-        type_vocab: int = 2, num_labels: int = 2,
+        type_vocab: int = 2, num_labels: int = 2, ln_eps: float = 1e-12,
     ):
         super().__init__()
+        self.ln_eps = ln_eps  # BERT's canonical LayerNorm eps
         self.word_emb = nn.Embedding(vocab_size, hidden)
         self.pos_emb = nn.Embedding(max_pos, hidden)
         self.type_emb = nn.Embedding(type_vocab, hidden)
         self.emb_ln_w = nn.Parameter(torch.ones(hidden))
         self.emb_ln_b = nn.Parameter(torch.zeros(hidden))
         self.layers = nn.ModuleList(
-            [BertLayer(hidden, heads, intermediate) for _ in range(layers)]
+            [BertLayer(hidden, heads, intermediate, ln_eps)
+             for _ in range(layers)]
         )
         self.pooler = nn.Linear(hidden, hidden)
         self.classifier = nn.Linear(hidden, num_labels)
@@ -113,11 +117,15 @@ class BertEncoder(nn.Module):
         mask = inputs.get("attention_mask")
         seq_lens = mask.to(torch.int32).sum(dim=-1) if mask is not None else None
         positions = torch.arange(s, device=ids.device)[None, :].expand(b, s)
+        # token_type defaults to segment 0 (BERT always adds the segment
+        # embedding; omitting it diverges from the canonical model)
         types = inputs.get("token_type_ids")
         emb = self.word_emb(ids) + self.pos_emb(positions)
         if types is not None:
             emb = emb + self.type_emb(types.long())
-        x = ops.layernorm(emb, self.emb_ln_w, self.emb_ln_b)
+        else:
+            emb = emb + self.type_emb.weight[0]
+        x = ops.layernorm(emb, self.emb_ln_w, self.emb_ln_b, eps=self.ln_eps)
 
         for layer in self.layers:
             x = layer(x, seq_lens)

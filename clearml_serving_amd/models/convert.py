@@ -1,0 +1,80 @@
+"""HuggingFace checkpoint conversion into the native model layouts.
+
+The environment has no model hub, but the converters are validated against
+random-init ``transformers`` models (same architecture, same numerics) in
+tests/test_hf_convert.py. Use for real deployments:
+
+    state = convert_hf_bert(load_file("model.safetensors"), num_layers=12)
+    save_file(state, "bert_native.safetensors")   # -> model card "weights"
+"""
+
+from typing import Dict
+
+import torch
+
+
+def convert_hf_bert(hf: Dict[str, torch.Tensor],
+                    num_layers: int) -> Dict[str, torch.Tensor]:
+    """transformers BertForSequenceClassification -> models.bert.BertEncoder."""
+
+    def g(key):
+        # accept both "bert.<...>" and bare "<...>" prefixes
+        return hf.get("bert." + key, hf.get(key))
+
+    out = {
+        "word_emb.weight": g("embeddings.word_embeddings.weight"),
+        "pos_emb.weight": g("embeddings.position_embeddings.weight"),
+        "type_emb.weight": g("embeddings.token_type_embeddings.weight"),
+        "emb_ln_w": g("embeddings.LayerNorm.weight"),
+        "emb_ln_b": g("embeddings.LayerNorm.bias"),
+        "pooler.weight": g("pooler.dense.weight"),
+        "pooler.bias": g("pooler.dense.bias"),
+        "classifier.weight": hf["classifier.weight"],
+        "classifier.bias": hf["classifier.bias"],
+    }
+    for i in range(num_layers):
+        p = "encoder.layer.{}.".format(i)
+        q_w, k_w, v_w = (g(p + "attention.self.{}.weight".format(n))
+                         for n in ("query", "key", "value"))
+        q_b, k_b, v_b = (g(p + "attention.self.{}.bias".format(n))
+                         for n in ("query", "key", "value"))
+        o = "layers.{}.".format(i)
+        out[o + "attn.qkv.weight"] = torch.cat([q_w, k_w, v_w], dim=0)
+        out[o + "attn.qkv.bias"] = torch.cat([q_b, k_b, v_b], dim=0)
+        out[o + "attn.out.weight"] = g(p + "attention.output.dense.weight")
+        out[o + "attn.out_bias"] = g(p + "attention.output.dense.bias")
+        out[o + "ln1_w"] = g(p + "attention.output.LayerNorm.weight")
+        out[o + "ln1_b"] = g(p + "attention.output.LayerNorm.bias")
+        out[o + "fc1.weight"] = g(p + "intermediate.dense.weight")
+        out[o + "fc1_bias"] = g(p + "intermediate.dense.bias")
+        out[o + "fc2.weight"] = g(p + "output.dense.weight")
+        out[o + "fc2_bias"] = g(p + "output.dense.bias")
+        out[o + "ln2_w"] = g(p + "output.LayerNorm.weight")
+        out[o + "ln2_b"] = g(p + "output.LayerNorm.bias")
+    return out
+
+
+def convert_hf_llama(hf: Dict[str, torch.Tensor],
+                     num_layers: int) -> Dict[str, torch.Tensor]:
+    """transformers LlamaForCausalLM -> models.llama.LlamaForCausalLM."""
+    out = {
+        "embed.weight": hf["model.embed_tokens.weight"],
+        "final_norm": hf["model.norm.weight"],
+        "lm_head.weight": hf.get("lm_head.weight",
+                                 hf["model.embed_tokens.weight"]),
+    }
+    for i in range(num_layers):
+        p = "model.layers.{}.".format(i)
+        o = "layers.{}.".format(i)
+        out[o + "qkv.weight"] = torch.cat(
+            [hf[p + "self_attn.q_proj.weight"],
+             hf[p + "self_attn.k_proj.weight"],
+             hf[p + "self_attn.v_proj.weight"]], dim=0)
+        out[o + "o_proj.weight"] = hf[p + "self_attn.o_proj.weight"]
+        out[o + "attn_norm"] = hf[p + "input_layernorm.weight"]
+        out[o + "mlp_norm"] = hf[p + "post_attention_layernorm.weight"]
+        out[o + "gate_up.weight"] = torch.cat(
+            [hf[p + "mlp.gate_proj.weight"], hf[p + "mlp.up_proj.weight"]],
+            dim=0)
+        out[o + "down.weight"] = hf[p + "mlp.down_proj.weight"]
+    return out

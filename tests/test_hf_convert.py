@@ -1,0 +1,98 @@
+"""Architecture correctness vs the canonical HuggingFace implementations.
+
+Random-init transformers models (no downloads) are converted into the
+native layouts; forward outputs must match to fp32 tolerance -- this pins
+the native BERT/Llama architectures to the reference semantics, not just
+to themselves.
+"""
+
+import pytest
+import torch
+
+transformers = pytest.importorskip("transformers")
+
+from clearml_serving_amd.models.bert import BertEncoder  # noqa: E402
+from clearml_serving_amd.models.convert import (  # noqa: E402
+    convert_hf_bert,
+    convert_hf_llama,
+)
+from clearml_serving_amd.models.llama import (  # noqa: E402
+    LlamaConfig,
+    LlamaForCausalLM,
+)
+
+
+def test_bert_matches_transformers():
+    from transformers import BertConfig, BertForSequenceClassification
+
+    torch.manual_seed(0)
+    cfg = BertConfig(vocab_size=211, hidden_size=64, num_hidden_layers=2,
+                     num_attention_heads=4, intermediate_size=128,
+                     max_position_embeddings=64, num_labels=3,
+                     hidden_dropout_prob=0.0,
+                     attention_probs_dropout_prob=0.0)
+    hf = BertForSequenceClassification(cfg).eval()
+
+    ours = BertEncoder(vocab_size=211, hidden=64, layers=2, heads=4,
+                       intermediate=128, max_pos=64, num_labels=3).eval()
+    ours.load_state_dict(convert_hf_bert(hf.state_dict(), num_layers=2))
+
+    ids = torch.randint(0, 211, (2, 16))
+    mask = torch.ones(2, 16, dtype=torch.int64)
+    with torch.inference_mode():
+        ref = hf(input_ids=ids, attention_mask=mask).logits
+        got = ours({"input_ids": ids, "attention_mask": mask})
+    torch.testing.assert_close(got, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_bert_matches_transformers_with_padding():
+    from transformers import BertConfig, BertForSequenceClassification
+
+    torch.manual_seed(1)
+    cfg = BertConfig(vocab_size=100, hidden_size=64, num_hidden_layers=1,
+                     num_attention_heads=2, intermediate_size=96,
+                     max_position_embeddings=32, num_labels=2,
+                     hidden_dropout_prob=0.0,
+                     attention_probs_dropout_prob=0.0)
+    hf = BertForSequenceClassification(cfg).eval()
+    ours = BertEncoder(vocab_size=100, hidden=64, layers=1, heads=2,
+                       intermediate=96, max_pos=32, num_labels=2).eval()
+    ours.load_state_dict(convert_hf_bert(hf.state_dict(), num_layers=1))
+
+    ids = torch.randint(0, 100, (2, 12))
+    mask = torch.tensor([[1] * 12, [1] * 5 + [0] * 7], dtype=torch.int64)
+    with torch.inference_mode():
+        ref = hf(input_ids=ids, attention_mask=mask).logits
+        got = ours({"input_ids": ids, "attention_mask": mask})
+    torch.testing.assert_close(got, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_llama_matches_transformers():
+    from transformers import LlamaConfig as HfLlamaConfig
+    from transformers import LlamaForCausalLM as HfLlama
+
+    torch.manual_seed(2)
+    hf_cfg = HfLlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=112,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=10000.0, rms_norm_eps=1e-5,
+        attention_dropout=0.0, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = HfLlama(hf_cfg).eval()
+
+    cfg = LlamaConfig(vocab_size=128, hidden=64, layers=2, heads=4,
+                      kv_heads=2, intermediate=112, rope_theta=10000.0,
+                      rms_eps=1e-5, max_position=64)
+    ours = LlamaForCausalLM(cfg).eval()
+    ours.load_state_dict(convert_hf_llama(hf.state_dict(), num_layers=2))
+
+    t = 10
+    ids = torch.randint(0, 128, (t,))
+    positions = torch.arange(t, dtype=torch.int32)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": t,
+                "seq_lens": torch.tensor([t], dtype=torch.int32),
+                "slot_mapping": torch.full((t,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        ref = hf(input_ids=ids[None]).logits[0]
+        got = ours(ids, positions, kv_caches=None, attn_ctx=attn_ctx)
+    torch.testing.assert_close(got, ref, atol=3e-4, rtol=3e-4)
