@@ -101,3 +101,42 @@ def test_status_endpoint(client):
     body = r.json()
     assert "test_model_sklearn" in body["endpoints"]
     assert body["revision"] is not None
+
+
+def test_engine_exception_maps_to_422(processor, store, tmp_path):
+    code = tmp_path / "bad.py"
+    code.write_text(
+        "class Preprocess(object):\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        raise ValueError('bad input shape')\n")
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.app import create_app
+
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom", serving_url="bad_ep"),
+        preprocess_code=str(code))
+    processor.serialize()  # app startup re-deserializes from the store
+    app = create_app(processor=processor, poll_frequency_sec=3600)
+    with TestClient(app) as c:
+        r = c.post("/serve/bad_ep", json={})
+        assert r.status_code == 422
+        assert "bad input shape" in r.json()["detail"]
+
+
+def test_engine_crash_maps_to_500(processor, store, tmp_path):
+    code = tmp_path / "crash.py"
+    code.write_text(
+        "class Preprocess(object):\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        raise RuntimeError('boom')\n")
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.app import create_app
+
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom", serving_url="crash_ep"),
+        preprocess_code=str(code))
+    processor.serialize()
+    app = create_app(processor=processor, poll_frequency_sec=3600)
+    with TestClient(app) as c:
+        r = c.post("/serve/crash_ep", json={})
+        assert r.status_code == 500

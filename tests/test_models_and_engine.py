@@ -187,3 +187,35 @@ def test_torchscript_model_serving(processor, store, tmp_path):
     ))
     out = run(processor.process_request("ts_ep", "", [1.0, 2.0, 3.0]))
     assert abs(float(np.asarray(out).ravel()[0]) - 12.0) < 1e-5
+
+
+def test_hip_engine_auto_update_version_swap(processor, store, tmp_path):
+    """Auto-update materializes hip-engine versions; newest serves after a
+    registry update (the Triton-sidecar repo-sync equivalent)."""
+    from clearml_serving_amd.schemas import ModelMonitoring
+
+    def card(labels):
+        p = tmp_path / ("c%d.json" % labels)
+        p.write_text(json.dumps({"arch": "bert-base", "num_labels": labels,
+                                 "dtype": "float32", "vocab_size": 100}))
+        return p
+
+    processor.add_model_monitoring(ModelMonitoring(
+        base_serving_url="hip_auto", engine_type="hip",
+        monitor_name="^bert-auto$", max_versions=1,
+        auxiliary_cfg={"max_queue_delay_us": 500, "use_graphs": False}))
+    store.register_model(name="bert-auto", project="p", path=str(card(2)))
+    processor._update_monitored_models()
+    body = {"input_ids": [1, 2, 3], "attention_mask": [1, 1, 1]}
+    out = run(processor.process_request("hip_auto", "1", body))
+    assert np.asarray(out).shape == (2,)
+
+    import time as _t
+
+    _t.sleep(0.01)
+    store.register_model(name="bert-auto", project="p", path=str(card(3)))
+    processor._update_monitored_models()
+    assert "hip_auto/2" in processor.get_synced_endpoints()
+    assert "hip_auto/1" not in processor.get_synced_endpoints()
+    out = run(processor.process_request("hip_auto", "2", body))
+    assert np.asarray(out).shape == (3,)
