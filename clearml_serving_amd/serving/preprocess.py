@@ -60,6 +60,7 @@ class BasePreprocessRequest:
         self._store = task
         self._preprocess = None
         self._model = None
+        self._artifact_sha = None
         if self.model_endpoint.preprocess_artifact:
             try:
                 self._instantiate_custom_preprocess_cls()
@@ -89,6 +90,7 @@ class BasePreprocessRequest:
         with open(path, "rb") as f:
             if sha256(f.read()).hexdigest() != artifact["sha256"]:
                 raise ValueError("artifact hash mismatch for {}".format(path))
+        self._artifact_sha = artifact["sha256"]  # staleness check on reload
 
         if path.endswith(".zip"):
             # package artifact: unpack beside the archive, import its

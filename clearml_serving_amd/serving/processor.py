@@ -453,12 +453,21 @@ class ModelRequestProcessor:
             self._model_monitoring_versions = mon_versions
             self._apply_params(params)
             self._update_canary_lookup()
-            # flush engine instances whose endpoint config changed/disappeared
+            # flush engine instances whose endpoint config OR preprocess
+            # artifact content changed (the reference hashes artifacts into
+            # its config hash, model_request_processor.py:636-654)
             synced = self.get_synced_endpoints()
             for url in list(self._engine_processor_lookup.keys()):
+                cached = self._engine_processor_lookup[url]
                 ep = synced.get(url)
-                if ep is None or ep.as_dict() != \
-                        self._engine_processor_lookup[url].model_endpoint.as_dict():
+                stale = ep is None or \
+                    ep.as_dict() != cached.model_endpoint.as_dict()
+                if not stale and ep.preprocess_artifact:
+                    art = self._store.get_artifact(self._session_id,
+                                                   ep.preprocess_artifact)
+                    stale = art is not None and \
+                        art["sha256"] != cached._artifact_sha
+                if stale:
                     engine = self._engine_processor_lookup.pop(url, None)
                     batcher = getattr(engine, "_batcher", None)
                     if batcher is not None:

@@ -289,3 +289,53 @@ def test_custom_async_pipeline(processor, store, tmp_path):
     )
     out = run(processor.process_request("pipe", "", [[5.0]]))
     assert abs(float(np.asarray(out["pipelined"])[0]) - 10.0) < 1e-6
+
+
+def test_hot_reload_waits_for_inflight_requests(processor, store, tmp_path):
+    """The stall-swap protocol must drain in-flight requests before swapping
+    endpoint tables (reference :700-720) -- slow request + concurrent
+    reload, everything completes and the new config serves afterwards."""
+    import threading
+
+    code = tmp_path / "slow.py"
+    code.write_text(
+        "import time\n"
+        "class Preprocess(object):\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        time.sleep(0.4)\n"
+        "        return {'v': 1}\n")
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom", serving_url="slow"),
+        preprocess_code=str(code))
+    processor.serialize()
+    processor.deserialize(skip_sync=True)
+
+    results = []
+
+    def call():
+        results.append(run(processor.process_request("slow", "", {})))
+
+    t = threading.Thread(target=call)
+    t.start()
+    time.sleep(0.1)  # request is now inside process()
+
+    # concurrent config change -> full stall-swap deserialize
+    p2 = ModelRequestProcessor(task_id=processor.get_id(), store=store)
+    p2.deserialize(skip_sync=True)
+    code2 = tmp_path / "fast.py"
+    code2.write_text(
+        "class Preprocess(object):\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        return {'v': 2}\n")
+    p2.add_endpoint(ModelEndpoint(engine_type="custom", serving_url="slow"),
+                    preprocess_code=str(code2))
+    p2.serialize()
+
+    t0 = time.time()
+    assert processor.deserialize() is True  # blocks until drain
+    drain_time = time.time() - t0
+    t.join(timeout=5)
+    assert results == [{"v": 1}]  # in-flight request finished on old code
+    assert drain_time >= 0.15    # the swap actually waited
+    out = run(processor.process_request("slow", "", {}))
+    assert out == {"v": 2}       # new code serves after the swap
