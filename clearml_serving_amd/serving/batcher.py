@@ -74,8 +74,10 @@ class DynamicBatcher:
         # batch's GPU ops contiguous on the stream, so the in-order stream
         # keeps static graph buffers race-free
         self._slots: "_queue.Queue" = _queue.Queue()
-        self._slots.put(0)
-        self._slots.put(1)
+        import os as _os
+
+        for s in range(int(_os.environ.get("CMLS_BATCH_SLOTS", 2))):
+            self._slots.put(s)
         self._enqueue_lock = threading.Lock()
         self._inflight = set()
         self._closed = False
