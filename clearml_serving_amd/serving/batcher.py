@@ -76,7 +76,7 @@ class DynamicBatcher:
         self._slots: "_queue.Queue" = _queue.Queue()
         import os as _os
 
-        for s in range(int(_os.environ.get("CMLS_BATCH_SLOTS", 2))):
+        for s in range(int(_os.environ.get("CMLS_BATCH_SLOTS", 3))):
             self._slots.put(s)
         self._enqueue_lock = threading.Lock()
         self._inflight = set()
@@ -231,9 +231,10 @@ class DynamicBatcher:
             ts = [t.to(want) for t in ts]
         if self.is_cuda:
             # fill this slot's reusable pinned host slab (the H2D copy is
-            # enqueued later on the stream, inside the enqueue lock)
-            buf = self._pinned_slab(("in", slot, key), bucket, ts[0].shape,
-                                    want)
+            # enqueued later on the stream, inside the enqueue lock); one
+            # max-batch slab per (slot, input) serves every bucket
+            buf = self._pinned_slab(("in", slot, key), self.buckets[-1],
+                                    ts[0].shape, want)
             n = len(ts)
             torch.stack(ts, dim=0, out=buf[:n])
             if n < bucket:
