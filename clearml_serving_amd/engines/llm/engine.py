@@ -311,7 +311,9 @@ class LlmEngine:
     # ------------------------------------------------------------------ #
     def _ensure_loop(self) -> None:
         loop = asyncio.get_running_loop()
-        if self._loop_task is None or getattr(self, "_loop_ref", None) is not loop:
+        if (self._loop_task is None
+                or getattr(self, "_loop_ref", None) is not loop
+                or self._loop_task.done()):  # restart after a step crash
             self._loop_ref = loop
             self._wake = asyncio.Event()
             self._loop_task = loop.create_task(self._engine_loop())
@@ -324,10 +326,14 @@ class LlmEngine:
             try:
                 await asyncio.to_thread(self.step)
             except Exception as ex:
-                # poison every active sequence rather than hanging clients
+                # poison every active sequence rather than hanging clients;
+                # return their KV pages before dropping them
                 for seq in self.waiting + self.running:
                     seq.stream.put_nowait(
                         {"error": str(ex), "finished": True, "token_ids": []})
+                    if seq.blocks:
+                        self.allocator.free(seq.blocks)
+                        seq.blocks = []
                 self.waiting.clear()
                 self.running.clear()
                 raise

@@ -203,3 +203,37 @@ def test_kv_pressure_preemption():
     assert all(len(o) == 40 for o in outs)
     assert eng.allocator.available == 6
     assert eng.stats["preemptions"] >= 1
+
+
+def test_engine_loop_restarts_after_step_crash():
+    """A crashed step poisons in-flight requests but the engine serves new
+    requests afterwards (loop restarts on next add_request)."""
+    eng = tiny_engine()
+    original = eng.step
+    calls = {"n": 0}
+
+    def crashing_step():
+        calls["n"] += 1
+        if calls["n"] == 1:
+            raise RuntimeError("injected step failure")
+        return original()
+
+    eng.step = crashing_step
+
+    async def first():
+        seq = await eng.add_request([1, 2, 3], SamplingParams(
+            temperature=0.0, max_tokens=4, ignore_eos=True))
+        item = await seq.stream.get()
+        return item
+
+    item = run(first())
+    assert "error" in item and item["finished"]
+
+    async def second():
+        toks = []
+        async for it in eng.generate("ok", SamplingParams(
+                temperature=0.0, max_tokens=4, ignore_eos=True)):
+            toks.extend(it["token_ids"])
+        return toks
+
+    assert len(run(second())) == 4
