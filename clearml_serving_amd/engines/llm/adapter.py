@@ -7,8 +7,6 @@ process (the reference also keeps a process-wide vLLM singleton,
 preprocess_service.py:619-631).
 """
 
-from typing import Any, Optional
-
 from ...schemas import ModelEndpoint
 from ...serving.preprocess import BasePreprocessRequest
 

@@ -15,7 +15,7 @@ control flow in forward).
 
 import json
 import os
-from typing import Any, Dict, Optional, Union
+from typing import Optional, Union
 
 import torch
 

@@ -19,7 +19,6 @@ classification logits [B, num_labels] (sequence output available via
 so every batch bucket is hipGraph-capturable.
 """
 
-import math
 from typing import Dict, Optional
 
 import torch

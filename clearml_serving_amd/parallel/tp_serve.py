@@ -9,7 +9,6 @@ construction cannot work for workers -- they never see requests).
 """
 
 import argparse
-import asyncio
 import os
 
 import torch

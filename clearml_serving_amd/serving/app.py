@@ -15,7 +15,6 @@ import traceback
 from typing import Optional, Union
 
 from fastapi import FastAPI, HTTPException, Request, Response
-from fastapi.responses import JSONResponse
 from fastapi.routing import APIRoute, APIRouter
 
 from .processor import EndpointNotFoundError, ModelRequestProcessor

@@ -22,10 +22,9 @@ import asyncio
 import importlib.util
 import os
 import sys
-import threading
 import zipfile
 from hashlib import sha256
-from typing import Any, Callable, Dict, Optional
+from typing import Dict, Optional
 
 import numpy as np
 

@@ -21,7 +21,6 @@ local ``ServingStore``. Semantics preserved:
 
 import asyncio
 import itertools
-import json
 import os
 import random
 import threading
