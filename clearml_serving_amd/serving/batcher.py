@@ -83,7 +83,8 @@ class DynamicBatcher:
         self._closed = False
         # telemetry for the Prometheus exporter / GET /status
         self.stats = {"batches": 0, "requests": 0, "occupancy_sum": 0.0,
-                      "stage_ms_sum": 0.0, "gpu_wait_ms_sum": 0.0}
+                      "stage_ms_sum": 0.0, "gpu_wait_ms_sum": 0.0,
+                      "queue_wait_ms_sum": 0.0}
 
     # ------------------------------------------------------------------ #
     async def submit(self, inputs: TensorOrDict) -> TensorOrDict:
@@ -91,7 +92,7 @@ class DynamicBatcher:
         request's slice of the model output."""
         self._ensure_worker()
         fut = asyncio.get_running_loop().create_future()
-        await self._queue.put((inputs, fut))
+        await self._queue.put((inputs, fut, time.monotonic()))
         return await fut
 
     def _ensure_worker(self) -> None:
@@ -141,6 +142,9 @@ class DynamicBatcher:
                     batch.append(item)
                 except asyncio.TimeoutError:
                     break
+            now = time.monotonic()
+            self.stats["queue_wait_ms_sum"] += sum(
+                (now - b[2]) * 1000 for b in batch)
             inputs = [b[0] for b in batch]
             futures = [b[1] for b in batch]
             # fire-and-continue: the next batch forms while this one stages

@@ -181,6 +181,49 @@ class GpuStatsExporter:
         self._stop = True
 
 
+class BatcherStagesCollector:
+    """Per-stage serving telemetry: queue-wait, batch
+    staging and GPU time per endpoint, exported as _sum/_count pairs
+    (rate(sum)/rate(count) gives the live mean in Prometheus)."""
+
+    def __init__(self, processor):
+        self._processor = processor
+
+    def collect(self):
+        from prometheus_client.core import (
+            CounterMetricFamily,
+            GaugeMetricFamily,
+        )
+
+        occ = GaugeMetricFamily(
+            "serving_batch_occupancy", "mean batch fill of its bucket",
+            labels=["endpoint"])
+        batches = CounterMetricFamily(
+            "serving_batches_total", "executed batches", labels=["endpoint"])
+        fams = {
+            s: CounterMetricFamily(
+                "serving_{}_ms_total".format(s),
+                "cumulative {} milliseconds".format(s), labels=["endpoint"])
+            for s in ("queue_wait", "stage", "gpu_wait")
+        }
+        for url, engine in list(
+                self._processor._engine_processor_lookup.items()):
+            b = getattr(engine, "_batcher", None)
+            if b is None or not b.stats["batches"]:
+                continue
+            s = b.stats
+            label = _prom_name(url, "")[:-1] or url
+            batches.add_metric([label], s["batches"])
+            occ.add_metric([label], s["occupancy_sum"] / s["batches"])
+            fams["queue_wait"].add_metric([label], s["queue_wait_ms_sum"])
+            fams["stage"].add_metric([label], s["stage_ms_sum"])
+            fams["gpu_wait"].add_metric([label], s["gpu_wait_ms_sum"])
+        yield batches
+        yield occ
+        for f in fams.values():
+            yield f
+
+
 _http_started = False
 
 
@@ -192,6 +235,10 @@ def install_stats_sink(processor, port: Optional[int] = None) -> StatsRegistry:
     global _http_started
     registry = StatsRegistry(processor=processor)
     processor.set_stats_sink(registry.report_batch)
+    try:
+        REGISTRY.register(BatcherStagesCollector(processor))
+    except Exception:
+        pass  # already registered in this process
     port = port if port is not None else int(
         os.environ.get("CLEARML_SERVING_STATS_PORT", 9999))
     if port > 0 and not _http_started:

@@ -63,3 +63,24 @@ def test_unconfigured_endpoint_gets_reserved_metrics():
     sr.report_batch([{"_url": "ghost", "_latency": 0.01, "_count": 1}])
     assert reg.get_sample_value("ghost__count_total") == 1.0
     assert reg.get_sample_value("ghost__latency_count") == 1.0
+
+
+def test_batcher_stages_collector():
+    from clearml_serving_amd.statistics.collector import BatcherStagesCollector
+
+    class FakeBatcher:
+        stats = {"batches": 4, "requests": 10, "occupancy_sum": 3.0,
+                 "stage_ms_sum": 8.0, "gpu_wait_ms_sum": 12.0,
+                 "queue_wait_ms_sum": 20.0}
+
+    class FakeEngine:
+        _batcher = FakeBatcher()
+
+    class FakeProc:
+        _engine_processor_lookup = {"ep/1": FakeEngine()}
+
+    fams = {f.name: f for f in BatcherStagesCollector(FakeProc()).collect()}
+    assert fams["serving_batches"].samples[0].value == 4
+    assert abs(fams["serving_batch_occupancy"].samples[0].value - 0.75) < 1e-9
+    assert fams["serving_queue_wait_ms"].samples[0].value == 20.0
+    assert fams["serving_gpu_wait_ms"].samples[0].value == 12.0
