@@ -103,6 +103,8 @@ class LlmEngineConfig:
     max_num_seqs: int = 64
     max_model_len: int = 4096
     max_prefill_tokens: int = 8192
+    prefill_chunk: int = 2048  # max prefill tokens per step (decode
+                               # interleaves between chunks of long prompts)
     gpu_memory_fraction: float = 0.85
     num_kv_blocks: Optional[int] = None  # explicit override (CPU tests)
     weights: Optional[str] = None
@@ -132,7 +134,7 @@ class LlmEngineConfig:
                     card = json.load(f)
         card.pop("arch", None)
         for key in ("preset", "dtype", "block_size", "max_num_seqs",
-                    "max_model_len", "max_prefill_tokens",
+                    "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "weights",
                     "device"):
             for src in (card, aux):
@@ -168,6 +170,7 @@ class Sequence:
         self.prompt_ids = prompt_ids
         self.output_ids: List[int] = []
         self.generated = 0  # survives preemption (output folds into prompt)
+        self.prefilled = 0  # prompt tokens already in the KV cache
         self.params = params
         self.blocks: List[int] = []
         self.stream: "asyncio.Queue" = asyncio.Queue()
@@ -342,13 +345,32 @@ class LlmEngine:
 
     # ------------------------------------------------------------------ #
     def step(self) -> None:
-        """One scheduler iteration: admit + prefill, then decode."""
+        """One scheduler iteration: admit, prefill (chunked -- at most
+        ``prefill_chunk`` prompt tokens per step so decode latency stays
+        bounded under long-prompt load), then decode everything else."""
         self.stats["steps"] += 1
-        admitted = self._admit()
-        if admitted:
-            self._prefill(admitted)
+        self._admit()
+        budget = self.cfg.prefill_chunk
+        pending = [s for s in self.running
+                   if s.prefilled < len(s.prompt_ids)]
+        # fresh short prompts batch through the dense prefill path
+        fresh = []
+        for s in pending:
+            need = len(s.prompt_ids)
+            if s.prefilled == 0 and need <= budget:
+                fresh.append(s)
+                budget -= need
+        if fresh:
+            self._prefill(fresh)
+            for s in fresh:
+                s.prefilled = len(s.prompt_ids)
+        # one long/continuing prompt advances by a chunk (paged attention
+        # against its cached history)
+        cont = [s for s in pending if s not in fresh]
+        if cont and budget > 0:
+            self._prefill_chunk(cont[0], budget)
         decoding = [s for s in self.running if not s.finished
-                    and s not in admitted]
+                    and s.prefilled >= len(s.prompt_ids) and s.output_ids]
         if decoding:
             self._decode(decoding)
         for s in list(self.running):
@@ -393,6 +415,53 @@ class LlmEngine:
         logits = self._exec_prefill(plan)
         self.stats["prompt_tokens"] += sum(len(p) for p in plan["prompts"])
         self._sample_and_emit(seqs, logits)
+
+    def _prefill_chunk(self, seq: "Sequence", budget: int) -> None:
+        """Advance one sequence's prefill by up to ``budget`` tokens using
+        paged attention over its cached history; samples the first token
+        when the chunk completes the prompt."""
+        self.stats["prefill_batches"] += 1
+        start = seq.prefilled
+        chunk = min(budget, len(seq.prompt_ids) - start)
+        plan = {
+            "mode": "chunk",
+            "tokens": seq.prompt_ids[start:start + chunk],
+            "start": start,
+            "kv_len": start + chunk,
+            "slots": [self._slot(seq, p) for p in range(start, start + chunk)],
+            "blocks": list(seq.blocks),
+            "complete": start + chunk >= len(seq.prompt_ids),
+        }
+        self._tp_broadcast(plan)
+        logits = self._exec_chunk(plan)
+        seq.prefilled = start + chunk
+        self.stats["prompt_tokens"] += chunk
+        if plan["complete"]:
+            self._sample_and_emit([seq], logits)
+
+    @torch.inference_mode()
+    def _exec_chunk(self, plan: Dict[str, Any]) -> torch.Tensor:
+        dev = self.device
+        chunk = len(plan["tokens"])
+        start = plan["start"]
+        tokens = torch.tensor(plan["tokens"], dtype=torch.long, device=dev)
+        positions = torch.arange(start, start + chunk, dtype=torch.int32,
+                                 device=dev)
+        btab = torch.tensor([plan["blocks"]], dtype=torch.int32, device=dev)
+        attn_ctx = {
+            "mode": "prefill_paged", "batch": 1, "seq": chunk,
+            "kv_lens": torch.tensor([plan["kv_len"]], dtype=torch.int32,
+                                    device=dev),
+            "q_lens": torch.tensor([chunk], dtype=torch.int32, device=dev),
+            "block_table": btab,
+            "slot_mapping": torch.tensor(plan["slots"], dtype=torch.int32,
+                                         device=dev),
+        }
+        last_idx = torch.tensor([chunk - 1], dtype=torch.long, device=dev) \
+            if plan["complete"] else torch.empty(0, dtype=torch.long,
+                                                 device=dev)
+        return self.model(tokens, positions, kv_caches=self.kv_caches,
+                          attn_ctx=attn_ctx, last_token_idx=last_idx)
 
     @torch.inference_mode()
     def _exec_prefill(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -524,6 +593,8 @@ class LlmEngine:
                 return
             if plan["mode"] == "prefill":
                 self._exec_prefill(plan)
+            elif plan["mode"] == "chunk":
+                self._exec_chunk(plan)
             else:
                 self._exec_decode(plan)
 

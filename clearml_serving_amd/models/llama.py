@@ -99,6 +99,16 @@ class LlamaLayer(nn.Module):
             ctx = ops.attention(qb, kb, vb, causal=True, scale=scale,
                                 seq_lens=attn_ctx["seq_lens"], layout="bshd")
             ctx = ctx.reshape(t, self.heads * self.head_dim)
+        elif attn_ctx["mode"] == "prefill_paged":
+            # chunked prefill: this chunk's queries attend to the full paged
+            # history (K/V of the chunk were just scattered into the cache)
+            k_cache, v_cache = kv_cache
+            b, s = attn_ctx["batch"], attn_ctx["seq"]
+            ctx = ops.attention_prefill_paged(
+                q.unflatten(0, (b, s)), k_cache, v_cache,
+                attn_ctx["block_table"], attn_ctx["kv_lens"],
+                attn_ctx["q_lens"], scale=scale)
+            ctx = ctx.reshape(t, self.heads * self.head_dim)
         else:  # decode: one token per sequence
             k_cache, v_cache = kv_cache
             ctx = ops.attention_decode(

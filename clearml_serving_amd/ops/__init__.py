@@ -209,6 +209,43 @@ def attention(
     return torch.matmul(probs, vf).to(q.dtype)
 
 
+def attention_prefill_paged(
+    q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+    block_table: torch.Tensor, kv_lens: torch.Tensor, q_lens: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Chunked-prefill attention: chunk queries [B, Sq, H, D] (bshd) attend
+    causally to the full PAGED history (kv_lens keys per seq, already in the
+    cache); q_lens masks per-seq chunk padding. Output [B, Sq, H, D]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attention_prefill_paged")
+        if ext is not None:
+            return ext.attention_prefill_paged(
+                q, k_cache, v_cache, block_table, kv_lens, q_lens,
+                float(scale))
+    # reference path: gather pages -> dense causal attention with history
+    b, sq, h, d = q.shape
+    hkv = k_cache.shape[1]
+    bs = k_cache.shape[2]
+    out = torch.zeros_like(q)
+    for i in range(b):
+        n = int(kv_lens[i])
+        ql = int(q_lens[i])
+        nb = (n + bs - 1) // bs
+        blocks = block_table[i, :nb].long()
+        k = k_cache[blocks].float().permute(1, 0, 2, 3).reshape(
+            hkv, nb * bs, d)[:, :n]
+        v = v_cache[blocks].float().permute(1, 0, 2, 3).reshape(
+            hkv, nb * bs, d)[:, :n]
+        o = attention(
+            q[i, :ql].permute(1, 0, 2)[None].float(),
+            k[None], v[None], causal=True, scale=scale)
+        out[i, :ql] = o[0].permute(1, 0, 2).to(q.dtype)
+    return out
+
+
 def attention_decode(
     q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
     block_table: torch.Tensor, seq_lens: torch.Tensor,

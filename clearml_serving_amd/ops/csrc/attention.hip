@@ -58,13 +58,18 @@ struct AttnStrides {
   long ob, oh, os;
 };
 
-template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS>
+template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false>
 __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
     const __hip_bfloat16* __restrict__ q,   // [B, H, Sq, D] via strides
-    const __hip_bfloat16* __restrict__ k,   // [B, Hkv, Sk, D] via strides
-    const __hip_bfloat16* __restrict__ v,
+    const __hip_bfloat16* __restrict__ k,   // dense via strides, or paged
+    const __hip_bfloat16* __restrict__ v,   //   cache [NB, Hkv, BS, D]
     __hip_bfloat16* __restrict__ out,
-    const int* __restrict__ seq_lens,       // [B] or null
+    const int* __restrict__ seq_lens,       // [B] kv lengths (or null)
+    const int* __restrict__ q_lens,         // [B] query lengths (or null):
+                                            // chunked prefill -- per-seq
+                                            // valid q rows + causal offset
+    const int* __restrict__ block_table,    // [B, max_blocks] when PAGED
+    int block_size, int max_blocks,
     AttnStrides st,
     int B, int H, int Hkv, int Sq, int Sk, float scale) {
   constexpr int D = HEAD_DIM;
@@ -99,11 +104,17 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
   const int wm0 = m0 + wave * 16;           // this wave's first query row
 
   const long q_base = (long)b * st.qb + (long)h * st.qh;
-  const long kv_base = (long)b * st.kb + (long)hkv * st.kh;
+  const long kv_base = PAGED ? (long)hkv * block_size * D
+                             : (long)b * st.kb + (long)hkv * st.kh;
   const long o_base = (long)b * st.ob + (long)h * st.oh;
   const int kv_len = HAS_SEQLENS ? min(seq_lens[b], Sk) : Sk;
-  // causal: queries attend to keys <= q_idx (Sq == Sk alignment)
-  const int kv_hi = CAUSAL ? min(kv_len, m0 + BLOCK_M + (Sk - Sq)) : kv_len;
+  // per-seq query length (chunked prefill pads chunks to Sq)
+  const int q_len = q_lens ? q_lens[b] : Sq;
+  // causal: the chunk's LAST query row attends up to the LAST key
+  // (history offset = kv_len - q_len)
+  const int causal_off = kv_len - q_len;
+  const int kv_hi = CAUSAL ? min(kv_len, m0 + BLOCK_M + causal_off) : kv_len;
+  const int* btab = PAGED ? block_table + (long)b * max_blocks : nullptr;
 
   // ---- load this wave's Q fragments (scaled once; fp32->bf16 later in S) --
   const int frag_row = lane & 15;           // i
@@ -141,10 +152,16 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
       const int gkey = n0 + p / (D / 8);
       const int d8 = (p % (D / 8)) * 8;
       if (gkey < kv_len) {
-        kreg[i] = *reinterpret_cast<const bf16x8_t*>(
-            k + kv_base + (long)gkey * st.ks + d8);
-        vreg[i] = *reinterpret_cast<const bf16x8_t*>(
-            v + kv_base + (long)gkey * st.ks + d8);
+        long off;
+        if (PAGED) {
+          const long blk = btab[gkey / block_size];
+          off = (blk * Hkv) * (long)block_size * D + kv_base +
+                (long)(gkey % block_size) * D + d8;
+        } else {
+          off = kv_base + (long)gkey * st.ks + d8;
+        }
+        kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);
+        vreg[i] = *reinterpret_cast<const bf16x8_t*>(v + off);
       } else {
         kreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
         vreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
@@ -206,7 +223,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
         float s = acc_s[t][r] * scale;
         const int key = col_base + t * 16;
         bool valid = key < kv_len;
-        if (CAUSAL) valid = valid && (key <= qrow + (Sk - Sq));
+        if (CAUSAL) valid = valid && (key <= qrow + causal_off);
         s = valid ? s : -INFINITY;
         pvals[r][t] = s;
         rmax = fmaxf(rmax, s);
@@ -283,7 +300,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow = row_base + r;
-    if (qrow >= Sq) continue;
+    if (qrow >= q_len) continue;
     const float inv_l = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
     __hip_bfloat16* dst = out + o_base + (long)qrow * st.os;
 #pragma unroll
@@ -335,12 +352,12 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
 
 #define LAUNCH_ATTN(DD, CC, SS)                                              \
   hipLaunchKernelGGL((attn_prefill_kernel<DD, CC, SS>), grid, block, 0,      \
-                     stream_,                                        \
+                     stream_,                                                \
                      (const __hip_bfloat16*)q.data_ptr(),                    \
                      (const __hip_bfloat16*)k.data_ptr(),                    \
                      (const __hip_bfloat16*)v.data_ptr(),                    \
-                     (__hip_bfloat16*)out.data_ptr(), sl, st, B, H, Hkv,     \
-                     Sq, Sk, (float)scale)
+                     (__hip_bfloat16*)out.data_ptr(), sl, nullptr, nullptr,  \
+                     0, 0, st, B, H, Hkv, Sq, Sk, (float)scale)
   if (D == 64) {
     if (causal) { if (sl) LAUNCH_ATTN(64, true, true); else LAUNCH_ATTN(64, true, false); }
     else        { if (sl) LAUNCH_ATTN(64, false, true); else LAUNCH_ATTN(64, false, false); }
@@ -349,5 +366,55 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k,
     else        { if (sl) LAUNCH_ATTN(128, false, true); else LAUNCH_ATTN(128, false, false); }
   }
 #undef LAUNCH_ATTN
+  return out;
+}
+
+
+torch::Tensor attention_prefill_paged(
+    torch::Tensor q,            // [B, Sq, H, D] (bshd; strided views ok)
+    torch::Tensor k_cache,      // [NB, Hkv, BS, D]
+    torch::Tensor v_cache,
+    torch::Tensor block_table,  // int32 [B, max_blocks]
+    torch::Tensor kv_lens,      // int32 [B] total keys (history + chunk)
+    torch::Tensor q_lens,       // int32 [B] valid query rows per sequence
+    double scale) {
+  // chunked-prefill attention: the chunk's queries attend causally to the
+  // full paged history + the chunk itself (K/V already scattered into the
+  // cache by kv_cache_write before this call).
+  TORCH_CHECK(q.dim() == 4 && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.stride(3) == 1 && q.stride(2) == q.size(3),
+              "q heads must be dense");
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  const int B = q.size(0), Sq = q.size(1), H = q.size(2), D = q.size(3);
+  const int Hkv = k_cache.size(1), BS = k_cache.size(2);
+  const int max_blocks = block_table.size(1);
+  TORCH_CHECK(D == 64 || D == 128);
+  TORCH_CHECK(H % Hkv == 0);
+  auto bt = block_table.to(q.device(), at::kInt).contiguous();
+  auto kl = kv_lens.to(q.device(), at::kInt).contiguous();
+  auto ql = q_lens.to(q.device(), at::kInt).contiguous();
+  auto out = torch::empty({B, Sq, H, D}, q.options());
+  AttnStrides st;
+  st.qb = q.stride(0); st.qh = q.stride(2); st.qs = q.stride(1);
+  st.kb = 0; st.kh = 0; st.ks = 0;  // unused on the paged path
+  st.ob = out.stride(0); st.oh = out.stride(2); st.os = out.stride(1);
+  // Sk bound for the kv loop = max kv_len; kernel clamps per sequence
+  const int Sk = kl.max().item<int>();
+  dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
+  dim3 block(NWAVES * 64);
+  hipStream_t stream_ = cmls::current_stream();
+
+#define LAUNCH_PAGED(DD)                                                     \
+  hipLaunchKernelGGL((attn_prefill_kernel<DD, true, true, true>), grid,      \
+                     block, 0, stream_,                                      \
+                     (const __hip_bfloat16*)q.data_ptr(),                    \
+                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
+                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
+                     (__hip_bfloat16*)out.data_ptr(), kl.data_ptr<int>(),    \
+                     ql.data_ptr<int>(), bt.data_ptr<int>(), BS, max_blocks, \
+                     st, B, H, Hkv, Sq, Sk, (float)scale)
+  if (D == 64) LAUNCH_PAGED(64);
+  else LAUNCH_PAGED(128);
+#undef LAUNCH_PAGED
   return out;
 }

@@ -25,6 +25,11 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
                     torch::Tensor slot_mapping);
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b);
+torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor k_cache,
+                                      torch::Tensor v_cache,
+                                      torch::Tensor block_table,
+                                      torch::Tensor kv_lens,
+                                      torch::Tensor q_lens, double scale);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   double theta);
 
@@ -48,4 +53,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_decode", &attention_decode);
   m.def("kv_cache_write", &kv_cache_write);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("attention_prefill_paged", &attention_prefill_paged);
 }
