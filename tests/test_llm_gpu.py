@@ -137,3 +137,58 @@ def test_llm_engine_gpu_decode_matches_prefill():
                 (top2.values[0] - top2.values[1]) < 0.05:
             agree += 1
     assert agree >= 7, "paged decode diverges from teacher-forced prefill"
+
+
+def test_attention_prefill_paged_numerics():
+    torch.manual_seed(11)
+    b, h, hkv, d, bs = 2, 8, 2, 128, 16
+    hist = [40, 7]          # cached history lengths
+    chunk = [24, 16]        # new chunk lengths (padded batch to 24)
+    sq = max(chunk)
+    nb = 8
+    k_cache = torch.randn(nb, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    v_cache = torch.randn_like(k_cache)
+    block_table = torch.tensor([[0, 2, 4, 6], [1, 3, 5, 7]],
+                               dtype=torch.int32, device=DEV)
+    kv_lens = torch.tensor([hist[0] + chunk[0], hist[1] + chunk[1]],
+                           dtype=torch.int32, device=DEV)
+    q_lens = torch.tensor(chunk, dtype=torch.int32, device=DEV)
+    q = torch.randn(b, sq, h, d, device=DEV, dtype=torch.bfloat16)
+
+    got = ops.attention_prefill_paged(q, k_cache, v_cache, block_table,
+                                      kv_lens, q_lens)
+    ref = ops.attention_prefill_paged(
+        q.float().cpu(), k_cache.float().cpu(), v_cache.float().cpu(),
+        block_table.cpu(), kv_lens.cpu(), q_lens.cpu())
+    for i in range(b):
+        torch.testing.assert_close(got[i, :chunk[i]].float().cpu(),
+                                   ref[i, :chunk[i]].float(),
+                                   atol=3e-2, rtol=3e-2)
+
+
+def test_chunked_prefill_gpu_matches_unchunked():
+    prompt = [(7 * i + 3) % 400 for i in range(200)]
+
+    def gen(chunk):
+        torch.manual_seed(77)
+        cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=256,
+                              block_size=16, max_model_len=512,
+                              prefill_chunk=chunk, device=DEV)
+        eng = LlmEngine(cfg)
+        eng.start()
+
+        async def go():
+            seq = await eng.add_request(list(prompt), SamplingParams(
+                temperature=0.0, max_tokens=8, ignore_eos=True))
+            toks = []
+            while True:
+                item = await seq.stream.get()
+                toks.extend(item["token_ids"])
+                if item["finished"]:
+                    return toks
+
+        return run(go())
+
+    full = gen(1024)
+    chunked = gen(48)
+    assert full == chunked
