@@ -17,7 +17,13 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     is_process_async = True
     is_postprocess_async = True
 
-    _engine_singleton = None
+    # one engine per MODEL (keyed by model id): unlike the reference, which
+    # is limited to a single process-wide vLLM engine (preprocess_service.py
+    # :816-834 marks multi-model "TODO"), several LLM endpoints can serve
+    # different models from one process -- placement spreads them over the
+    # node's GPUs via auxiliary_cfg {"device": "cuda:N"}
+    _engines = {}
+    _engine_singleton = None  # kept for tests/back-compat (first engine)
 
     def __init__(self, model_endpoint: ModelEndpoint, task=None):
         super().__init__(model_endpoint, task)
@@ -28,12 +34,16 @@ class LlmPreprocessRequest(BasePreprocessRequest):
         # preprocess override vllm_model_config, examples/vllm/preprocess.py)
         if isinstance(self._model, dict):
             aux.update(self._model)
-        if LlmPreprocessRequest._engine_singleton is None:
+        key = model_endpoint.model_id or "__default__"
+        if key not in LlmPreprocessRequest._engines:
             model_path = self._get_local_model_file()
             cfg = LlmEngineConfig.from_aux(model_path, aux)
-            LlmPreprocessRequest._engine_singleton = LlmEngine(cfg)
-            LlmPreprocessRequest._engine_singleton.start()
-        self._engine = LlmPreprocessRequest._engine_singleton
+            engine = LlmEngine(cfg)
+            engine.start()
+            LlmPreprocessRequest._engines[key] = engine
+            if LlmPreprocessRequest._engine_singleton is None:
+                LlmPreprocessRequest._engine_singleton = engine
+        self._engine = LlmPreprocessRequest._engines[key]
         self._served_name = model_endpoint.serving_url
 
     async def preprocess(self, request, state, collect_custom_statistics_fn=None):

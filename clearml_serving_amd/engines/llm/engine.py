@@ -186,8 +186,15 @@ class Sequence:
 class LlmEngine:
     def __init__(self, cfg: LlmEngineConfig):
         self.cfg = cfg
-        self.device = torch.device(
-            cfg.device or ("cuda" if torch.cuda.is_available() else "cpu"))
+        if cfg.device:
+            self.device = torch.device(cfg.device)
+        elif torch.cuda.is_available():
+            from ..torch_engine import _pick_device
+
+            # most-free-HBM placement (multi-model nodes spread engines)
+            self.device = _pick_device({})
+        else:
+            self.device = torch.device("cpu")
         self.dtype = (torch.bfloat16 if self.device.type == "cuda"
                       else torch.float32)
         self.model: Optional[LlamaForCausalLM] = None

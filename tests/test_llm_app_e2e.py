@@ -16,6 +16,7 @@ def llm_client(processor, store, tmp_path):
     from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
 
     LlmPreprocessRequest._engine_singleton = None
+    LlmPreprocessRequest._engines = {}
 
     card = tmp_path / "card.json"
     card.write_text(json.dumps({
@@ -30,6 +31,7 @@ def llm_client(processor, store, tmp_path):
     with TestClient(app) as c:
         yield c
     LlmPreprocessRequest._engine_singleton = None
+    LlmPreprocessRequest._engines = {}
 
 
 def test_chat_completion_route(llm_client):
@@ -80,3 +82,40 @@ def test_generic_route_prompt(llm_client):
         "ignore_eos": True})
     assert r.status_code == 200, r.text
     assert r.json()["tokens"] == 2
+
+
+def test_two_llm_models_one_process(processor, store, tmp_path):
+    """Two DIFFERENT llama models serve from one process (the reference's
+    vLLM integration is limited to one engine per container)."""
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+
+    LlmPreprocessRequest._engines = {}
+    LlmPreprocessRequest._engine_singleton = None
+    import asyncio
+
+    for name, vocab in (("llm_a", 300), ("llm_b", 500)):
+        card = tmp_path / (name + ".json")
+        card.write_text(json.dumps({
+            "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 32,
+            "block_size": 16, "max_model_len": 64, "device": "cpu",
+            "overrides": {"vocab_size": vocab}}))
+        rec = store.register_model(name=name, project="p", path=str(card))
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="llm", serving_url=name, model_id=rec.model_id))
+
+    async def ask(name):
+        return await processor.process_request(name, "", {
+            "prompt": "x", "max_tokens": 2, "temperature": 0.0,
+            "ignore_eos": True})
+
+    loop = asyncio.new_event_loop()
+    a = loop.run_until_complete(ask("llm_a"))
+    b = loop.run_until_complete(ask("llm_b"))
+    loop.close()
+    assert a["tokens"] == 2 and b["tokens"] == 2
+    assert len(LlmPreprocessRequest._engines) == 2
+    vocabs = {e.model_config.vocab_size
+              for e in LlmPreprocessRequest._engines.values()}
+    assert vocabs == {300, 500}
+    LlmPreprocessRequest._engines = {}
+    LlmPreprocessRequest._engine_singleton = None
