@@ -32,6 +32,17 @@ from ..schemas import ModelEndpoint
 from ..serving.batcher import DEFAULT_BUCKETS, DynamicBatcher
 from ..serving.preprocess import BasePreprocessRequest
 
+def _to_numpy(out):
+    if isinstance(out, dict):
+        return {k: _to_numpy(v) for k, v in out.items()}
+    if isinstance(out, (tuple, list)):
+        return [_to_numpy(v) for v in out]
+    # bf16/fp16 need an fp32 hop; fp32/int tensors convert directly
+    if out.dtype in (torch.bfloat16, torch.float16):
+        out = out.float()
+    return out.numpy()
+
+
 _DTYPES = {
     "bfloat16": torch.bfloat16, "bf16": torch.bfloat16,
     "float16": torch.float16, "fp16": torch.float16,
@@ -179,12 +190,7 @@ class HipPreprocessRequest(BasePreprocessRequest):
         else:
             inputs = self._to_tensor(data)
         out = await self._batcher.submit(inputs)
-        if isinstance(out, dict):
-            return {k: v.float().numpy() if v.is_floating_point() else v.numpy()
-                    for k, v in out.items()}
-        if isinstance(out, (tuple, list)):
-            return [v.float().numpy() for v in out]
-        return out.float().numpy() if out.is_floating_point() else out.numpy()
+        return _to_numpy(out)
 
 
 # alias for reference-CLI compatibility: `model add --engine triton` serves

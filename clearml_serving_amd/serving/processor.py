@@ -142,6 +142,7 @@ class ModelRequestProcessor:
         self._canary_endpoints: Dict[str, CanaryEP] = {}
         self._canary_route: Dict[str, dict] = {}
         self._metric_logging: Dict[str, EndpointMetricLogging] = {}
+        self._metric_cfg_cache: Dict[str, Any] = {}
         self._engine_processor_lookup: Dict[str, BasePreprocessRequest] = {}
 
         self._last_revision: Optional[int] = None
@@ -471,6 +472,7 @@ class ModelRequestProcessor:
                     batcher = getattr(engine, "_batcher", None)
                     if batcher is not None:
                         batcher.shutdown()  # free graphs/streams/HBM
+            self._metric_cfg_cache.clear()
             self._last_revision = revision
             self._update_lock_flag = False
         import gc
@@ -668,19 +670,29 @@ class ModelRequestProcessor:
 
         processor._preprocess.send_request = _send
 
+    def _resolve_metric_cfg(self, url: str):
+        # per-request prefix scans are hot-path cost: resolve once per url
+        # (cache invalidated on config reload)
+        try:
+            return self._metric_cfg_cache[url]
+        except KeyError:
+            cfg = self._metric_logging.get(url) or self._prefix_metric(url)
+            self._metric_cfg_cache[url] = cfg
+            return cfg
+
     async def _do_process_request(
         self, processor: BasePreprocessRequest, url: str, body: Any,
         serve_type: str,
     ) -> Any:
-        tic = time.time()
         state: Dict[str, Any] = {}
 
         # metric sampling decision (:1316-1323)
         freq = self._metric_log_freq
-        metric_cfg = self._metric_logging.get(url) or self._prefix_metric(url)
+        metric_cfg = self._resolve_metric_cfg(url)
         if metric_cfg and metric_cfg.log_frequency is not None:
             freq = metric_cfg.log_frequency
-        collect = freq and (freq >= 1.0 or random.random() <= freq)
+        collect = bool(freq) and (freq >= 1.0 or random.random() <= freq)
+        tic = time.time() if collect else 0.0
         stats: Dict[str, Any] = {}
         collect_fn = stats.update if collect else None
 
