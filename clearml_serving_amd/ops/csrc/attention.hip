@@ -210,40 +210,72 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
     }
 
     // ---- mask + online softmax ---- //
+    // interior tiles (every key valid for every query of the block) skip
+    // masking entirely: the per-element compare+select chain was ~20% of
+    // the kernel's VALU work (PMC: VALU instrs 10.5x MFMA instrs)
     const int row_base = wm0 + (lane >> 4) * 4;  // C layout: row = base + reg
     const int col_base = n0 + (lane & 15);       // col = base + 16t
     float pvals[4][BLOCK_N / 16];
     float alpha[4];
+    const bool tile_full =
+        (n0 + BLOCK_N <= kv_len) &&
+        (!CAUSAL || (n0 + BLOCK_N - 1 <= m0 + causal_off));
+    if (tile_full) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow = row_base + r;
-      float rmax = -INFINITY;
+      for (int r = 0; r < 4; ++r) {
+        float rmax = -INFINITY;
 #pragma unroll
-      for (int t = 0; t < BLOCK_N / 16; ++t) {
-        float s = acc_s[t][r] * scale;
-        const int key = col_base + t * 16;
-        bool valid = key < kv_len;
-        if (CAUSAL) valid = valid && (key <= qrow + causal_off);
-        s = valid ? s : -INFINITY;
-        pvals[r][t] = s;
-        rmax = fmaxf(rmax, s);
+        for (int t = 0; t < BLOCK_N / 16; ++t) {
+          const float s = acc_s[t][r] * scale;
+          pvals[r][t] = s;
+          rmax = fmaxf(rmax, s);
+        }
+        rmax = rowgroup_max(rmax);
+        const float m_new = fmaxf(m_run[r], rmax);
+        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+        float rsum = 0.f;
+#pragma unroll
+        for (int t = 0; t < BLOCK_N / 16; ++t) {
+          const float p = __expf(pvals[r][t] - m_new);
+          pvals[r][t] = p;
+          rsum += p;
+        }
+        rsum = rowgroup_sum(rsum);
+        l_run[r] = l_run[r] * alpha[r] + rsum;
+        m_run[r] = m_new;
       }
-      rmax = rowgroup_max(rmax);
-      const float m_new = fmaxf(m_run[r], rmax);
-      // all-masked rows keep m = -inf; exp() below yields 0 contributions
-      alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
-      float rsum = 0.f;
+    } else {
 #pragma unroll
-      for (int t = 0; t < BLOCK_N / 16; ++t) {
-        const float p = (pvals[r][t] == -INFINITY || m_new == -INFINITY)
-                            ? 0.f
-                            : __expf(pvals[r][t] - m_new);
-        pvals[r][t] = p;
-        rsum += p;
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = row_base + r;
+        float rmax = -INFINITY;
+#pragma unroll
+        for (int t = 0; t < BLOCK_N / 16; ++t) {
+          float s = acc_s[t][r] * scale;
+          const int key = col_base + t * 16;
+          bool valid = key < kv_len;
+          if (CAUSAL) valid = valid && (key <= qrow + causal_off);
+          s = valid ? s : -INFINITY;
+          pvals[r][t] = s;
+          rmax = fmaxf(rmax, s);
+        }
+        rmax = rowgroup_max(rmax);
+        const float m_new = fmaxf(m_run[r], rmax);
+        // all-masked rows keep m = -inf; exp() yields 0 contributions
+        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+        float rsum = 0.f;
+#pragma unroll
+        for (int t = 0; t < BLOCK_N / 16; ++t) {
+          const float p = (pvals[r][t] == -INFINITY || m_new == -INFINITY)
+                              ? 0.f
+                              : __expf(pvals[r][t] - m_new);
+          pvals[r][t] = p;
+          rsum += p;
+        }
+        rsum = rowgroup_sum(rsum);
+        l_run[r] = l_run[r] * alpha[r] + rsum;
+        m_run[r] = m_new;
       }
-      rsum = rowgroup_sum(rsum);
-      l_run[r] = l_run[r] * alpha[r] + rsum;
-      m_run[r] = m_new;
     }
     // rescale O by alpha (per C-layout row)
 #pragma unroll
