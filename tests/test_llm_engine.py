@@ -20,6 +20,13 @@ def run(coro):
     try:
         return loop.run_until_complete(coro)
     finally:
+        # cancel lingering workers (batcher/engine loops) before closing
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
         loop.close()
 
 

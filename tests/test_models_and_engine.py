@@ -13,7 +13,18 @@ from clearml_serving_amd.serving.batcher import DynamicBatcher
 
 
 def run(coro):
-    return asyncio.new_event_loop().run_until_complete(coro)
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(coro)
+    finally:
+        # cancel lingering workers (batcher/engine loops) before closing
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
 
 
 def test_arch_registry():

@@ -19,7 +19,18 @@ from clearml_serving_amd.serving.processor import (
 
 
 def run(coro):
-    return asyncio.new_event_loop().run_until_complete(coro)
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(coro)
+    finally:
+        # cancel lingering workers (batcher/engine loops) before closing
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
 
 
 def _sklearn_model(tmp_path, slope=2.0):
@@ -339,3 +350,23 @@ def test_hot_reload_waits_for_inflight_requests(processor, store, tmp_path):
     assert drain_time >= 0.15    # the swap actually waited
     out = run(processor.process_request("slow", "", {}))
     assert out == {"v": 2}       # new code serves after the swap
+
+
+def test_preprocess_package_folder(processor, store, tmp_path):
+    """A preprocess FOLDER uploads as a zip package and hot-loads its
+    preprocess.py (reference package-artifact path)."""
+    pkg = tmp_path / "pkg"
+    pkg.mkdir()
+    (pkg / "helper.py").write_text("SCALE = 3\n")
+    (pkg / "preprocess.py").write_text(
+        "import os, sys\n"
+        "sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))\n"
+        "from helper import SCALE\n"
+        "class Preprocess(object):\n"
+        "    def process(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        return {'y': data['x'] * SCALE}\n")
+    processor.add_endpoint(
+        ModelEndpoint(engine_type="custom", serving_url="pkg_ep"),
+        preprocess_code=str(pkg))
+    out = run(processor.process_request("pkg_ep", "", {"x": 7}))
+    assert out == {"y": 21}
