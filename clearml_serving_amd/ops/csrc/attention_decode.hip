@@ -122,40 +122,51 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
       vvec[u].u = *reinterpret_cast<const uint32x4*>(
           v_cache + base + sub * EPL);
     }
+    // scores for the whole 4-iteration chunk (4x16 = 64 wave keys)
+    float score[UNROLL][GQ];
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
-      float score[GQ];
 #pragma unroll
       for (int g = 0; g < GQ; ++g) {
         float acc = 0.f;
 #pragma unroll
         for (int e = 0; e < EPL; ++e) acc += qf[g][e] * to_f32(kvec[u].h[e]);
-        score[g] = acc;
+        score[u][g] = acc;
       }
       // group-level dot reduction (16 lanes hold partial sums)
 #pragma unroll
       for (int g = 0; g < GQ; ++g) {
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
-          score[g] += __shfl_xor(score[g], off, 64);
-        if (!valid[u]) score[g] = -INFINITY;
+          score[u][g] += __shfl_xor(score[u][g], off, 64);
+        if (!valid[u]) score[u][g] = -INFINITY;
       }
-      // online update per q head: wave max over this iteration's 4 keys
+    }
+    // ONE online-softmax update per chunk per head (rescaling O per
+    // 4-key iteration was the serial VALU chain limiting bandwidth)
 #pragma unroll
-      for (int g = 0; g < GQ; ++g) {
-        const float it_max = group_max4(score[g]);
-        if (it_max == -INFINITY) continue;
-        const float m_new = fmaxf(m_run[g], it_max);
-        const float alpha =
-            (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
-        const float p = valid[u] ? __expf(score[g] - m_new) : 0.f;
-        l_run[g] = l_run[g] * alpha + group_sum4(p);
-        m_run[g] = m_new;
+    for (int g = 0; g < GQ; ++g) {
+      float cmax = score[0][g];
 #pragma unroll
-        for (int e = 0; e < EPL; ++e) {
-          o_acc[g][e] = o_acc[g][e] * alpha + p * to_f32(vvec[u].h[e]);
-        }
+      for (int u = 1; u < UNROLL; ++u) cmax = fmaxf(cmax, score[u][g]);
+      cmax = group_max4(cmax);
+      if (cmax == -INFINITY) continue;
+      const float m_new = fmaxf(m_run[g], cmax);
+      const float alpha =
+          (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+      float psum = 0.f;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) o_acc[g][e] *= alpha;
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u) {
+        const float p = valid[u] ? __expf(score[u][g] - m_new) : 0.f;
+        psum += p;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e)
+          o_acc[g][e] += p * to_f32(vvec[u].h[e]);
       }
+      l_run[g] = l_run[g] * alpha + group_sum4(psum);
+      m_run[g] = m_new;
     }
   }
 
