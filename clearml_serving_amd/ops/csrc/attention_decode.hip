@@ -22,6 +22,7 @@
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
+typedef __attribute__((ext_vector_type(2))) short bf16x2_t;
 
 // values uniform within each 16-lane group: combining the 4 groups of a
 // wave needs only xor 16 and 32
@@ -76,15 +77,17 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     return;
   }
 
-  // q fragments for the GQ query heads sharing this kv head
-  float qf[GQ][EPL];
+  // q fragments: packed bf16 pairs feed v_dot2_f32_bf16 (2 MACs per
+  // instruction in the K dot products; scale folds into the score later)
+  bf16x2_t qp[GQ][EPL / 2];
 #pragma unroll
   for (int g = 0; g < GQ; ++g) {
     const int h = hkv * GQ + g;
     const __hip_bfloat16* qrow = q + (long)b * q_bstride + (long)h * D +
                                  sub * EPL;
 #pragma unroll
-    for (int e = 0; e < EPL; ++e) qf[g][e] = to_f32(qrow[e]) * scale;
+    for (int e = 0; e < EPL / 2; ++e)
+      qp[g][e] = *reinterpret_cast<const bf16x2_t*>(qrow + 2 * e);
   }
 
   float m_run[GQ], l_run[GQ], o_acc[GQ][EPL];
@@ -126,11 +129,13 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     float score[UNROLL][GQ];
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
+      const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kvec[u]);
 #pragma unroll
       for (int g = 0; g < GQ; ++g) {
         float acc = 0.f;
 #pragma unroll
-        for (int e = 0; e < EPL; ++e) acc += qf[g][e] * to_f32(kvec[u].h[e]);
+        for (int e = 0; e < EPL / 2; ++e)
+          acc = __builtin_amdgcn_fdot2_f32_bf16(qp[g][e], kp[e], acc, false);
         score[u][g] = acc;
       }
       // group-level dot reduction (16 lanes hold partial sums)
@@ -139,7 +144,7 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
           score[u][g] += __shfl_xor(score[u][g], off, 64);
-        if (!valid[u]) score[u][g] = -INFINITY;
+        score[u][g] = valid[u] ? score[u][g] * scale : -INFINITY;
       }
     }
     // ONE online-softmax update per chunk per head (rescaling O per
