@@ -208,7 +208,7 @@ class LlmEngine:
         self._started = False
         self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
                       "prefill_batches": 0, "decode_batches": 0,
-                      "preemptions": 0}
+                      "preemptions": 0, "aborts": 0}
 
     # ------------------------------------------------------------------ #
     def start(self) -> None:
@@ -294,15 +294,32 @@ class LlmEngine:
         self._wake.set()
         return seq
 
+    def abort(self, seq: "Sequence") -> None:
+        """Stop generating for a sequence (client gone): drop it from the
+        queues; its pages free at the end of the current step."""
+        seq.finished = True
+        seq.finish_reason = seq.finish_reason or "abort"
+        if seq in self.waiting:
+            self.waiting.remove(seq)
+            if seq.blocks:
+                self.allocator.free(seq.blocks)
+                seq.blocks = []
+        self.stats["aborts"] = self.stats.get("aborts", 0) + 1
+
     async def generate(self, prompt: str, params: SamplingParams
                        ) -> AsyncGenerator[Dict[str, Any], None]:
         ids = self.tokenizer.encode(prompt)
         seq = await self.add_request(ids, params)
-        while True:
-            item = await seq.stream.get()
-            yield item
-            if item.get("finished"):
-                return
+        try:
+            while True:
+                item = await seq.stream.get()
+                yield item
+                if item.get("finished"):
+                    return
+        finally:
+            # consumer cancelled/disconnected mid-stream: stop generating
+            if not seq.finished:
+                self.abort(seq)
 
     async def generate_simple(self, body: Dict[str, Any]) -> Dict[str, Any]:
         """Non-OpenAI route: {"prompt": str, "max_tokens": ...}."""

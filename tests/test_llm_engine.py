@@ -303,3 +303,33 @@ def test_chunked_prefill_interleaves_decode():
     # chunked prefill happened AND decode steps ran during the same window
     assert eng.stats["prefill_batches"] >= 5
     assert eng.stats["decode_batches"] >= 11  # 12 tokens = prefill sample + 11 decodes
+
+
+def test_client_disconnect_aborts_generation():
+    """Cancelling the consumer mid-stream frees the sequence's KV pages and
+    stops further generation for it."""
+    eng = tiny_engine()
+
+    async def main():
+        params = SamplingParams(temperature=0.0, max_tokens=200,
+                                ignore_eos=True)
+        agen = eng.generate("long request", params)
+        got = 0
+        async for item in agen:
+            got += len(item["token_ids"])
+            if got >= 3:
+                await agen.aclose()  # simulates client disconnect
+                break
+        # let the engine run a few more steps
+        for _ in range(20):
+            await asyncio.sleep(0.01)
+            if not eng.running and not eng.waiting:
+                break
+        return got
+
+    got = run(main())
+    assert got >= 3
+    assert eng.stats["aborts"] == 1
+    assert not eng.running and not eng.waiting
+    assert eng.allocator.available == eng.allocator.num_blocks
+    assert eng.stats["generated_tokens"] < 50  # didn't run to max_tokens
