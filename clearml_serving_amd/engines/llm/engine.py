@@ -388,11 +388,11 @@ class LlmEngine:
             self._prefill(fresh)
             for s in fresh:
                 s.prefilled = len(s.prompt_ids)
-        # one long/continuing prompt advances by a chunk (paged attention
-        # against its cached history)
+        # long/continuing prompts advance by chunks (batched paged
+        # attention against their cached histories)
         cont = [s for s in pending if s not in fresh]
         if cont and budget > 0:
-            self._prefill_chunk(cont[0], budget)
+            self._prefill_chunk(cont, budget)
         decoding = [s for s in self.running if not s.finished
                     and s.prefilled >= len(s.prompt_ids) and s.output_ids]
         if decoding:
@@ -440,52 +440,79 @@ class LlmEngine:
         self.stats["prompt_tokens"] += sum(len(p) for p in plan["prompts"])
         self._sample_and_emit(seqs, logits)
 
-    def _prefill_chunk(self, seq: "Sequence", budget: int) -> None:
-        """Advance one sequence's prefill by up to ``budget`` tokens using
-        paged attention over its cached history; samples the first token
-        when the chunk completes the prompt."""
+    def _prefill_chunk(self, seqs: List["Sequence"], budget: int) -> None:
+        """Advance several sequences' prefill by up to ``budget`` tokens
+        total in ONE batched paged-attention forward; samples first tokens
+        for the sequences whose chunk completes the prompt."""
+        batch = []
+        for s in seqs:
+            if budget <= 0:
+                break
+            chunk = min(budget, len(s.prompt_ids) - s.prefilled)
+            batch.append((s, chunk))
+            budget -= chunk
+        if not batch:
+            return
         self.stats["prefill_batches"] += 1
-        start = seq.prefilled
-        chunk = min(budget, len(seq.prompt_ids) - start)
         plan = {
             "mode": "chunk",
-            "tokens": seq.prompt_ids[start:start + chunk],
-            "start": start,
-            "kv_len": start + chunk,
-            "slots": [self._slot(seq, p) for p in range(start, start + chunk)],
-            "blocks": list(seq.blocks),
-            "complete": start + chunk >= len(seq.prompt_ids),
+            "tokens": [s.prompt_ids[s.prefilled:s.prefilled + c]
+                       for s, c in batch],
+            "starts": [s.prefilled for s, c in batch],
+            "kv_lens": [s.prefilled + c for s, c in batch],
+            "slots": [[self._slot(s, p)
+                       for p in range(s.prefilled, s.prefilled + c)]
+                      for s, c in batch],
+            "blocks": [list(s.blocks) for s, c in batch],
+            "complete": [s.prefilled + c >= len(s.prompt_ids)
+                         for s, c in batch],
         }
         self._tp_broadcast(plan)
         logits = self._exec_chunk(plan)
-        seq.prefilled = start + chunk
-        self.stats["prompt_tokens"] += chunk
-        if plan["complete"]:
-            self._sample_and_emit([seq], logits)
+        done = []
+        for (s, c), complete in zip(batch, plan["complete"]):
+            s.prefilled += c
+            self.stats["prompt_tokens"] += c
+            if complete:
+                done.append(s)
+        if done:
+            self._sample_and_emit(done, logits)
 
     @torch.inference_mode()
     def _exec_chunk(self, plan: Dict[str, Any]) -> torch.Tensor:
         dev = self.device
-        chunk = len(plan["tokens"])
-        start = plan["start"]
-        tokens = torch.tensor(plan["tokens"], dtype=torch.long, device=dev)
-        positions = torch.arange(start, start + chunk, dtype=torch.int32,
-                                 device=dev)
-        btab = torch.tensor([plan["blocks"]], dtype=torch.int32, device=dev)
+        b = len(plan["tokens"])
+        lens = [len(t) for t in plan["tokens"]]
+        smax = max(lens)
+        tokens = torch.zeros(b, smax, dtype=torch.long)
+        positions = torch.zeros(b, smax, dtype=torch.int32)
+        slot_map = torch.full((b, smax), -1, dtype=torch.int32)
+        for i in range(b):
+            n = lens[i]
+            tokens[i, :n] = torch.tensor(plan["tokens"][i], dtype=torch.long)
+            positions[i, :n] = torch.arange(
+                plan["starts"][i], plan["starts"][i] + n, dtype=torch.int32)
+            slot_map[i, :n] = torch.tensor(plan["slots"][i],
+                                           dtype=torch.int32)
+        max_blocks = max(len(bl) for bl in plan["blocks"])
+        btab = torch.zeros(b, max_blocks, dtype=torch.int32)
+        for i, bl in enumerate(plan["blocks"]):
+            btab[i, :len(bl)] = torch.tensor(bl, dtype=torch.int32)
         attn_ctx = {
-            "mode": "prefill_paged", "batch": 1, "seq": chunk,
-            "kv_lens": torch.tensor([plan["kv_len"]], dtype=torch.int32,
+            "mode": "prefill_paged", "batch": b, "seq": smax,
+            "kv_lens": torch.tensor(plan["kv_lens"], dtype=torch.int32,
                                     device=dev),
-            "q_lens": torch.tensor([chunk], dtype=torch.int32, device=dev),
-            "block_table": btab,
-            "slot_mapping": torch.tensor(plan["slots"], dtype=torch.int32,
-                                         device=dev),
+            "q_lens": torch.tensor(lens, dtype=torch.int32, device=dev),
+            "block_table": btab.to(dev),
+            "slot_mapping": slot_map.view(-1).to(dev),
         }
-        last_idx = torch.tensor([chunk - 1], dtype=torch.long, device=dev) \
-            if plan["complete"] else torch.empty(0, dtype=torch.long,
-                                                 device=dev)
-        return self.model(tokens, positions, kv_caches=self.kv_caches,
-                          attn_ctx=attn_ctx, last_token_idx=last_idx)
+        rows = [i * smax + lens[i] - 1
+                for i in range(b) if plan["complete"][i]]
+        last_idx = torch.tensor(rows, dtype=torch.long, device=dev)
+        return self.model(tokens.view(-1).to(dev),
+                          positions.view(-1).to(dev),
+                          kv_caches=self.kv_caches, attn_ctx=attn_ctx,
+                          last_token_idx=last_idx)
 
     @torch.inference_mode()
     def _exec_prefill(self, plan: Dict[str, Any]) -> torch.Tensor:
