@@ -28,7 +28,6 @@
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;   // 4 VGPRs
-typedef __attribute__((ext_vector_type(4))) short bf16x4_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define MFMA_16x16x32(A, B, C) \
@@ -53,11 +52,6 @@ __device__ __forceinline__ float rowgroup_sum(float v) {
   return v;
 }
 
-__device__ __forceinline__ int v4_tile_elem(int kt, int dt) {
-  // tile index -> element offset (64 elems per 4x16 tile)
-  return (((dt << 1) | (kt & 1)) * 8 + (kt >> 1)) * 64;
-}
-
 struct AttnStrides {
   long qb, qh, qs;   // q/out: batch, head, seq strides (elements)
   long kb, kh, ks;   // k/v share layout
@@ -80,6 +74,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
     int B, int H, int Hkv, int Sq, int Sk, float scale) {
   constexpr int D = HEAD_DIM;
   constexpr int KSTRIDE = D + PAD;          // LDS K row stride (bf16)
+  constexpr int VSTRIDE = BLOCK_N + PAD;    // LDS V^T row stride
   constexpr int PSTRIDE = BLOCK_N + PAD;
 
   // double-buffered K/V tiles: tile n+1's global loads are issued into
@@ -88,12 +83,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
   // next tile lands in the other buffer (no barrier between compute and
   // write, only before the swapped buffer is read)
   __shared__ short lds_k[2][BLOCK_N * KSTRIDE];
-  // V as [4-key][16-col] subtiles for the gfx950 LDS transpose read
-  // (ds_read_b64_tr_b16: each 16-lane group transposes one 4x16 tile);
-  // tile order (dt, kt&1, kt>>1) puts one tr read's 4 group-tiles at
-  // consecutive 64-element strides. Staging writes ONE ds_write_b128 per
-  // piece (the previous scatter-transpose was 8 b16 writes per piece).
-  __shared__ short lds_v4[2][BLOCK_N * D];
+  __shared__ short lds_vt[2][D * VSTRIDE];
   __shared__ short lds_p[NWAVES * 16 * PSTRIDE];
 
   // grid = (B*H, n_m_tiles): bh on x so the dispatcher's XCD round-robin
@@ -185,10 +175,11 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
       const int key = p / (D / 8);
       const int d8 = (p % (D / 8)) * 8;
       *reinterpret_cast<bf16x8_t*>(&lds_k[buf][key * KSTRIDE + d8]) = kreg[i];
-      // V piece -> its 4x16 subtile: one contiguous b128 write
-      const int vaddr = v4_tile_elem(key >> 2, d8 >> 4) + (key & 3) * 16 +
-                        (d8 & 15);
-      *reinterpret_cast<bf16x8_t*>(&lds_v4[buf][vaddr]) = vreg[i];
+      // V scatter-transposed with the key-XOR bank swizzle (see PV reads)
+      const int kswz = key ^ (((d8 >> 3) & 7) << 3);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        lds_vt[buf][(d8 + e) * VSTRIDE + kswz] = vreg[i][e];
     }
   };
 
@@ -318,24 +309,13 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
           &pslab[frag_row * PSTRIDE + kc * 32 + frag_ko]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
-        // B[k][j] = V[32kc + (l>>4)*8 + e][16dt + (l&15)] via two LDS
-        // transpose reads: group g covers keys 8g..8g+7 = subtiles
-        // kt = 8kc+2g (lo, e 0-3) and 8kc+2g+1 (hi, e 4-7)
-        const int lo_base = v4_tile_elem(8 * kc, dt) + (lane >> 4) * 64 +
-                            (lane & 15) * 4;
-        const int hi_base = v4_tile_elem(8 * kc + 1, dt) + (lane >> 4) * 64 +
-                            (lane & 15) * 4;
-        unsigned alo = (unsigned)(unsigned long long)&lds_v4[cur][lo_base];
-        unsigned ahi = (unsigned)(unsigned long long)&lds_v4[cur][hi_base];
-        bf16x4_t vlo, vhi;
-        asm volatile(
-            "ds_read_b64_tr_b16 %0, %2\n\t"
-            "ds_read_b64_tr_b16 %1, %3\n\t"
-            "s_waitcnt lgkmcnt(0)"
-            : "=&v"(vlo), "=&v"(vhi) : "v"(alo), "v"(ahi) : "memory");
-        bf16x8_t vf;
-#pragma unroll
-        for (int e = 0; e < 4; ++e) { vf[e] = vlo[e]; vf[e + 4] = vhi[e]; }
+        // B[k][j] = V[k + 32kc][j + 16dt] = vt[j + 16dt][k]: contiguous in k;
+        // apply the staging key-XOR (8-aligned base ^ bits 3-5 stays
+        // 16-B aligned)
+        const int vrow = dt * 16 + frag_row;
+        const int vkey = (kc * 32 + frag_ko) ^ (((vrow >> 3) & 7) << 3);
+        const bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            &lds_vt[cur][vrow * VSTRIDE + vkey]);
         acc_o[dt] = MFMA_16x16x32(pf, vf, acc_o[dt]);
       }
     }
