@@ -107,6 +107,7 @@ class LlmEngineConfig:
                                # interleaves between chunks of long prompts)
     gpu_memory_fraction: float = 0.85
     num_kv_blocks: Optional[int] = None  # explicit override (CPU tests)
+    quantization: Optional[str] = None   # "fp8": fp8 weights via hipBLASLt
     weights: Optional[str] = None
     tokenizer_path: Optional[str] = None
     device: Optional[str] = None
@@ -135,8 +136,8 @@ class LlmEngineConfig:
         card.pop("arch", None)
         for key in ("preset", "dtype", "block_size", "max_num_seqs",
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
-                    "gpu_memory_fraction", "num_kv_blocks", "weights",
-                    "device"):
+                    "gpu_memory_fraction", "num_kv_blocks", "quantization",
+                    "weights", "device"):
             for src in (card, aux):
                 if key in src and src[key] is not None:
                     setattr(cfg, key, src[key])
@@ -243,6 +244,13 @@ class LlmEngine:
 
                 load_weights(model, cfg.weights)
         self.model = model.eval().to(self.device).to(self.dtype)
+        if cfg.quantization == "fp8":
+            if self.device.type != "cuda":
+                raise ValueError("fp8 quantization needs a GPU (hipBLASLt)")
+            from ...models.quant import quantize_llama_fp8
+
+            n = quantize_llama_fp8(self.model)
+            print("[llm] fp8-quantized {} projection layers".format(n))
 
         self.tokenizer = (HfTokenizer(cfg.tokenizer_path)
                           if cfg.tokenizer_path else SimpleTokenizer())

@@ -1,0 +1,59 @@
+"""fp8 (OCP e4m3) weight quantization for LLM serving.
+
+gfx950 runs fp8 MFMA at 2x the bf16 rate, and decode is weight-streaming
+bound -- fp8 weights halve the bytes per step. Measured with hipBLASLt
+(torch._scaled_mm) on MI355X: 1.69x on prefill-shaped GEMMs (M=2048),
+1.15x at decode M=64.
+
+Scheme: per-output-row weight scales (absmax/448), dynamic per-tensor
+activation scales. Enable per endpoint via model card / aux
+``{"quantization": "fp8"}``.
+"""
+
+import torch
+import torch.nn as nn
+
+F8 = torch.float8_e4m3fn
+F8_MAX = 448.0
+
+
+class Fp8Linear(nn.Module):
+    """Drop-in replacement for a bias-free nn.Linear: fp8 weights with
+    per-row scales, dynamic per-tensor activation quantization."""
+
+    def __init__(self, linear: nn.Linear):
+        super().__init__()
+        w = linear.weight.data.float()
+        row_max = w.abs().amax(dim=1, keepdim=True).clamp(min=1e-8)
+        scale = row_max / F8_MAX
+        self.register_buffer("weight_fp8",
+                             (w / scale).to(F8).contiguous())
+        # _scaled_mm rowwise wants scale_b shaped [1, N] for b = w.T
+        self.register_buffer("weight_scale",
+                             scale.to(torch.float32).reshape(1, -1))
+        self.out_features = linear.out_features
+        self.in_features = linear.in_features
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        orig_shape = x.shape
+        x2 = x.reshape(-1, orig_shape[-1])
+        amax = x2.abs().amax().clamp(min=1e-8)
+        x_scale = (amax / F8_MAX).to(torch.float32).reshape(1, 1)
+        x8 = (x2.float() / x_scale).to(F8)
+        out = torch._scaled_mm(
+            x8, self.weight_fp8.t(), scale_a=x_scale,
+            scale_b=self.weight_scale, out_dtype=x.dtype)
+        return out.reshape(*orig_shape[:-1], self.out_features)
+
+
+def quantize_llama_fp8(model: nn.Module) -> int:
+    """Swap the projection Linears of a LlamaForCausalLM to Fp8Linear
+    (lm_head stays bf16 for logit fidelity). Returns layers converted."""
+    n = 0
+    for layer in getattr(model, "layers", []):
+        for name in ("qkv", "o_proj", "gate_up", "down"):
+            lin = getattr(layer, name, None)
+            if isinstance(lin, nn.Linear):
+                setattr(layer, name, Fp8Linear(lin))
+                n += 1
+    return n

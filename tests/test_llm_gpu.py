@@ -199,3 +199,35 @@ def test_chunked_prefill_gpu_matches_unchunked():
     full = gen(1024)
     chunked = gen(48)
     assert full == chunked
+
+
+def test_fp8_quantized_engine():
+    """fp8-weight llama serves and tracks the bf16 engine's greedy output
+    (same seed) closely enough for serving (majority token agreement)."""
+    def gen(quant):
+        torch.manual_seed(31)
+        cfg = LlmEngineConfig(preset="llama-3-1b", num_kv_blocks=1024,
+                              block_size=16, max_model_len=512, device=DEV,
+                              quantization=quant)
+        eng = LlmEngine(cfg)
+        eng.start()
+
+        async def go():
+            seq = await eng.add_request(
+                [(i * 13 + 7) % 1000 for i in range(64)],
+                SamplingParams(temperature=0.0, max_tokens=12,
+                               ignore_eos=True))
+            toks = []
+            while True:
+                item = await seq.stream.get()
+                toks.extend(item["token_ids"])
+                if item["finished"]:
+                    return toks
+
+        return run(go())
+
+    bf16 = gen(None)
+    fp8 = gen("fp8")
+    assert len(fp8) == 12
+    agree = sum(a == b for a, b in zip(bf16, fp8))
+    assert agree >= 8, (bf16, fp8)
