@@ -37,12 +37,21 @@ class Fp8Linear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         orig_shape = x.shape
         x2 = x.reshape(-1, orig_shape[-1])
+        m = x2.shape[0]
+        # hipBLASLt scaled_mm wants M a multiple of 16 and matching scale
+        # granularity on both operands (rowwise here)
+        pad = (-m) % 16
+        if pad:
+            x2 = torch.nn.functional.pad(x2, (0, 0, 0, pad))
         amax = x2.abs().amax().clamp(min=1e-8)
-        x_scale = (amax / F8_MAX).to(torch.float32).reshape(1, 1)
-        x8 = (x2.float() / x_scale).to(F8)
+        s = (amax / F8_MAX).to(torch.float32)
+        x_scale = s.expand(x2.shape[0], 1).contiguous()
+        x8 = (x2.float() / s).to(F8)
         out = torch._scaled_mm(
             x8, self.weight_fp8.t(), scale_a=x_scale,
             scale_b=self.weight_scale, out_dtype=x.dtype)
+        if pad:
+            out = out[:m]
         return out.reshape(*orig_shape[:-1], self.out_features)
 
 
