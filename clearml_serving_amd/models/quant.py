@@ -8,6 +8,13 @@ bound -- fp8 weights halve the bytes per step. Measured with hipBLASLt
 Scheme: per-output-row weight scales (absmax/448), dynamic per-tensor
 activation scales. Enable per endpoint via model card / aux
 ``{"quantization": "fp8"}``.
+
+STATUS: experimental. Measured on llama-3-8B decode (random weights): the
+un-fused dynamic activation quantization (amax + scale + cast per
+projection) currently outweighs the GEMM gain -- 3.7k vs 6.0k out-tok/s
+against bf16. The win requires fusing the activation quant into the
+producing kernels (rmsnorm/silu_mul emitting fp8 + scale) and an
+fp8-native decode attention; see docs/ROADMAP.md.
 """
 
 import torch
