@@ -230,3 +230,22 @@ def test_hip_engine_auto_update_version_swap(processor, store, tmp_path):
     assert "hip_auto/1" not in processor.get_synced_endpoints()
     out = run(processor.process_request("hip_auto", "2", body))
     assert np.asarray(out).shape == (3,)
+
+
+def test_batcher_respects_queue_delay():
+    """A lone request waits at most ~max_queue_delay before executing."""
+    import time as _t
+
+    def model_fn(x):
+        return x
+
+    batcher = DynamicBatcher(model_fn, device="cpu", max_batch_size=64,
+                             max_queue_delay_us=30_000, use_graphs=False)
+
+    async def main():
+        t0 = _t.monotonic()
+        await batcher.submit(torch.zeros(2))
+        return _t.monotonic() - t0
+
+    elapsed = run(main())
+    assert 0.02 <= elapsed < 0.5  # waited for the window, not forever

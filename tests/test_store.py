@@ -69,3 +69,26 @@ def test_cross_instance_visibility(tmp_path):
     s1.set_config_object(sid, "endpoints", {"x": 1})
     assert s2.revision(sid) == 1
     assert s2.get_config_object(sid, "endpoints") == {"x": 1}
+
+
+def test_concurrent_writers(tmp_path):
+    """Two processes' worth of writers against one store (WAL): no lost
+    updates, monotonic revision."""
+    import threading
+
+    from clearml_serving_amd.store import ServingStore
+
+    s1 = ServingStore(str(tmp_path / "c"))
+    s2 = ServingStore(str(tmp_path / "c"))
+    sid = s1.create_session(name="c")
+
+    def writer(store, key):
+        for i in range(20):
+            store.set_config_object(sid, key, {"i": i})
+
+    t1 = threading.Thread(target=writer, args=(s1, "a"))
+    t2 = threading.Thread(target=writer, args=(s2, "b"))
+    t1.start(); t2.start(); t1.join(); t2.join()
+    assert s1.get_config_object(sid, "a") == {"i": 19}
+    assert s1.get_config_object(sid, "b") == {"i": 19}
+    assert s1.revision(sid) == 40
