@@ -37,8 +37,12 @@ __device__ __forceinline__ float group_sum4(float v) {
   return v;
 }
 
-template <int D, int GQ>
-__global__ __launch_bounds__(256, 2) void attn_decode_kernel(
+// UNROLL/MINB default to the measured-best production point (4-deep key
+// unroll, 148 VGPR -> 3 waves/SIMD for GQ=4).  <.., 2, 4> is the
+// occupancy-first alternative (116 VGPR -> 4 waves/SIMD at half the loads
+// in flight per wave); scripts/decode_ab.hip A/Bs them.
+template <int D, int GQ, int UNROLL = 4, int MINB = 2>
+__global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
     const __hip_bfloat16* __restrict__ q,        // [B, H, D]
     const __hip_bfloat16* __restrict__ k_cache,  // [NB, Hkv, BS, D]
     const __hip_bfloat16* __restrict__ v_cache,  // [NB, Hkv, BS, D]
@@ -108,7 +112,6 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   // was latency-bound at ~1.7 TB/s
   const int per_iter = NW * GROUPS;
   const int n_iters = (n_keys - key_lo + per_iter - 1) / per_iter;
-  constexpr int UNROLL = 4;
   for (int it0 = 0; it0 < n_iters; it0 += UNROLL) {
     bf16x8 kvec[UNROLL], vvec[UNROLL];
     bool valid[UNROLL];
@@ -303,6 +306,7 @@ __global__ void kv_cache_write_kernel(const T* __restrict__ knew,  // [T,Hkv,D]
 
 }  // namespace
 
+#ifndef CMLS_KERNEL_ONLY
 torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                                torch::Tensor v_cache,
                                torch::Tensor block_table,
@@ -414,3 +418,4 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
     TORCH_CHECK(false, "kv_cache_write: unsupported dtype ", st);
   }
 }
+#endif  // CMLS_KERNEL_ONLY

@@ -6,6 +6,9 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
 
+// CMLS_KERNEL_ONLY: torch-free build for standalone kernel harnesses
+// (scripts/*.hip) that #include a kernel source directly.
+#ifndef CMLS_KERNEL_ONLY
 #include <c10/hip/HIPStream.h>
 #include <torch/extension.h>
 
@@ -19,8 +22,6 @@ inline hipStream_t current_stream() {
 }
 }  // namespace cmls
 
-#define WAVE_SIZE 64
-
 #define HIP_CHECK(expr)                                                     \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \
@@ -29,6 +30,9 @@ inline hipStream_t current_stream() {
 
 #define CHECK_LASTDIM_CONTIG(x)                                             \
   TORCH_CHECK((x).stride(-1) == 1, #x " must be contiguous in last dim")
+#endif  // CMLS_KERNEL_ONLY
+
+#define WAVE_SIZE 64
 
 // ---------------------------------------------------------------------- //
 // dtype conversion
