@@ -355,10 +355,20 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                      bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
                      (__hip_bfloat16*)out.data_ptr(), po, pml, splits,       \
                      q.stride(0), H, Hkv, BS, max_blocks, (float)scale)
+#define LAUNCH_DEC4(DD, GG, UU, MM)                                          \
+  hipLaunchKernelGGL((attn_decode_kernel<DD, GG, UU, MM>), grid, dim3(256),  \
+                     0, stream_, (const __hip_bfloat16*)q.data_ptr(),        \
+                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
+                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
+                     bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
+                     (__hip_bfloat16*)out.data_ptr(), po, pml, splits,       \
+                     q.stride(0), H, Hkv, BS, max_blocks, (float)scale)
   if (D == 128) {
     if (GQ == 1) LAUNCH_DEC(128, 1);
     else if (GQ == 2) LAUNCH_DEC(128, 2);
-    else if (GQ == 4) LAUNCH_DEC(128, 4);
+    // GQ=4 (llama-3 shapes): occupancy-first variant measured 3-6% faster
+    // at B=32-64, S=1-4k (scripts/decode_ab.hip, profiles/decode_ab.txt)
+    else if (GQ == 4) LAUNCH_DEC4(128, 4, 2, 4);
     else LAUNCH_DEC(128, 8);
   } else {
     if (GQ == 1) LAUNCH_DEC(64, 1);
@@ -367,6 +377,7 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
     else LAUNCH_DEC(64, 8);
   }
 #undef LAUNCH_DEC
+#undef LAUNCH_DEC4
   if (splits > 1) {
     dim3 cgrid(B, H);
     if (D == 128) {
