@@ -80,6 +80,16 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     async def v1_models(self, body, state, collect_fn=None):
         return self._engine.openai_models(self._served_name)
 
+    async def v1_embeddings(self, body, state, collect_fn=None):
+        return await self._engine.openai_embeddings(body, self._served_name)
+
+    # vLLM-compatible token utility routes (/serve/openai/tokenize)
+    async def tokenize(self, body, state, collect_fn=None):
+        return self._engine.openai_tokenize(body)
+
+    async def detokenize(self, body, state, collect_fn=None):
+        return self._engine.openai_detokenize(body)
+
 
 # reference-CLI compatibility: `--engine vllm` runs the native LLM engine
 BasePreprocessRequest.register_engine("vllm")(LlmPreprocessRequest)

@@ -553,6 +553,45 @@ class LlmEngine:
             kv_caches=self.kv_caches, attn_ctx=attn_ctx,
             last_token_idx=last_idx)
 
+    @torch.inference_mode()
+    def _exec_embed(self, plan: Dict[str, Any]) -> torch.Tensor:
+        """Dense forward WITHOUT touching the KV cache: returns mean-pooled,
+        L2-normalized final hidden states [b, hidden] (the v1/embeddings
+        serve path; reference delegates this to vLLM's pooling stack,
+        preprocess_service.py:632-1095)."""
+        prompts = plan["prompts"]
+        b = len(prompts)
+        lens = [len(p) for p in prompts]
+        smax = max(lens)
+        dev = self.device
+        tokens = torch.zeros(b, smax, dtype=torch.long)
+        positions = torch.zeros(b, smax, dtype=torch.int32)
+        for i in range(b):
+            tokens[i, :lens[i]] = torch.tensor(prompts[i], dtype=torch.long)
+            positions[i, :lens[i]] = torch.arange(lens[i], dtype=torch.int32)
+        attn_ctx = {
+            "mode": "prefill", "batch": b, "seq": smax,
+            "seq_lens": torch.tensor(lens, dtype=torch.int32, device=dev),
+        }
+        hidden = self.model(
+            tokens.view(-1).to(dev), positions.view(-1).to(dev),
+            kv_caches=None, attn_ctx=attn_ctx,
+            return_hidden=True).view(b, smax, -1).float()
+        mask = (torch.arange(smax, device=dev)[None, :]
+                < torch.tensor(lens, device=dev)[:, None]).unsqueeze(-1)
+        pooled = (hidden * mask).sum(1) / mask.sum(1).clamp(min=1)
+        return torch.nn.functional.normalize(pooled, dim=-1)
+
+    async def embed_batch(self, texts: List[str]) -> List[List[float]]:
+        if self.model is None:
+            raise RuntimeError("LLM engine not started")
+        prompts = [self.tokenizer.encode(t)[:self.cfg.max_model_len]
+                   for t in texts]
+        plan = {"mode": "embed", "prompts": prompts}
+        self._tp_broadcast(plan)
+        out = self._exec_embed(plan)
+        return out.cpu().tolist()
+
     def _preempt_one(self) -> bool:
         """KV pressure relief: evict the newest running sequence back to the
         waiting queue (vLLM-style recompute preemption -- its prompt plus
@@ -654,6 +693,8 @@ class LlmEngine:
                 self._exec_prefill(plan)
             elif plan["mode"] == "chunk":
                 self._exec_chunk(plan)
+            elif plan["mode"] == "embed":
+                self._exec_embed(plan)
             else:
                 self._exec_decode(plan)
 
@@ -753,6 +794,31 @@ class LlmEngine:
         return {"object": "list",
                 "data": [{"id": model_name, "object": "model",
                           "owned_by": "clearml-serving-amd"}]}
+
+    async def openai_embeddings(self, body: Dict[str, Any], model_name: str):
+        inp = body.get("input")
+        if inp is None:
+            raise ValueError("embeddings request requires 'input'")
+        texts = inp if isinstance(inp, list) else [inp]
+        vecs = await self.embed_batch([str(t) for t in texts])
+        nprompt = sum(len(self.tokenizer.encode(str(t))) for t in texts)
+        return {
+            "object": "list", "model": model_name,
+            "data": [{"object": "embedding", "index": i, "embedding": v}
+                     for i, v in enumerate(vecs)],
+            "usage": {"prompt_tokens": nprompt, "total_tokens": nprompt},
+        }
+
+    def openai_tokenize(self, body: Dict[str, Any]):
+        prompt = body.get("prompt")
+        if prompt is None and body.get("messages"):
+            prompt = self._chat_prompt(body["messages"])
+        ids = self.tokenizer.encode(prompt or "")
+        return {"tokens": ids, "count": len(ids),
+                "max_model_len": self.cfg.max_model_len}
+
+    def openai_detokenize(self, body: Dict[str, Any]):
+        return {"prompt": self.tokenizer.decode(body.get("tokens") or [])}
 
     async def _collect(self, prompt: str, params: SamplingParams):
         ids = self.tokenizer.encode(prompt)

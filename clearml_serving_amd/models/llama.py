@@ -165,8 +165,11 @@ class LlamaForCausalLM(nn.Module):
         kv_caches: Optional[List] = None,  # per layer (k_cache, v_cache)
         attn_ctx: Optional[Dict] = None,
         last_token_idx: Optional[torch.Tensor] = None,
+        return_hidden: bool = False,
     ) -> torch.Tensor:
-        """Returns logits [n_seqs, vocab] at the selected token positions."""
+        """Returns logits [n_seqs, vocab] at the selected token positions
+        (or the final-norm hidden states [T, hidden] when ``return_hidden``
+        -- the embeddings serve path)."""
         x = self.embed(tokens.long())
         residual = torch.zeros_like(x)
         for i, layer in enumerate(self.layers):
@@ -176,6 +179,8 @@ class LlamaForCausalLM(nn.Module):
                         residual=residual)
         if last_token_idx is not None:
             x = x[last_token_idx.long()]
+        if return_hidden:
+            return x
         logits = self.lm_head(x)
         if self.tp_size > 1:
             # TP: lm_head is column-sharded -> all-gather the vocab shards

@@ -151,20 +151,9 @@ def create_app(
                 return body
         return body
 
-    @router.post("/{model_id}/{version}")
-    @router.post("/{model_id}")
-    async def base_serve_model(
-        model_id: str, request: Request, version: Optional[str] = None
-    ):
-        body = await _read_body(request)
-        out = await process_with_exceptions(
-            base_url=model_id, version=version,
-            request_body=body, serve_type="process",
-        )
-        if isinstance(out, Response):
-            return out
-        return _jsonable(out)
-
+    # registered BEFORE the generic /{model_id}/{version} route: two-segment
+    # OpenAI paths like /serve/openai/tokenize would otherwise match the
+    # generic route (model_id="openai", version="tokenize")
     @router.post("/openai/{endpoint_type:path}")
     @router.get("/openai/{endpoint_type:path}")
     async def openai_serve_model(endpoint_type: str, request: Request):
@@ -183,6 +172,20 @@ def create_app(
         out = await process_with_exceptions(
             base_url=model, version=None,
             request_body=combined, serve_type=endpoint_type,
+        )
+        if isinstance(out, Response):
+            return out
+        return _jsonable(out)
+
+    @router.post("/{model_id}/{version}")
+    @router.post("/{model_id}")
+    async def base_serve_model(
+        model_id: str, request: Request, version: Optional[str] = None
+    ):
+        body = await _read_body(request)
+        out = await process_with_exceptions(
+            base_url=model_id, version=version,
+            request_body=body, serve_type="process",
         )
         if isinstance(out, Response):
             return out

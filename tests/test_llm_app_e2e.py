@@ -119,3 +119,23 @@ def test_two_llm_models_one_process(processor, store, tmp_path):
     assert vocabs == {300, 500}
     LlmPreprocessRequest._engines = {}
     LlmPreprocessRequest._engine_singleton = None
+
+
+def test_embeddings_route(llm_client):
+    r = llm_client.post("/serve/openai/v1/embeddings", json={
+        "model": "test_llm", "input": ["alpha", "beta"]})
+    assert r.status_code == 200, r.text
+    out = r.json()
+    assert out["object"] == "list" and len(out["data"]) == 2
+    assert len(out["data"][0]["embedding"]) > 0
+    assert out["usage"]["prompt_tokens"] > 0
+
+
+def test_tokenize_route(llm_client):
+    r = llm_client.post("/serve/openai/tokenize", json={
+        "model": "test_llm", "prompt": "hi there"})
+    assert r.status_code == 200, r.text
+    toks = r.json()["tokens"]
+    r2 = llm_client.post("/serve/openai/detokenize", json={
+        "model": "test_llm", "tokens": toks})
+    assert r2.json()["prompt"] == "hi there"

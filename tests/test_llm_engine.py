@@ -427,3 +427,35 @@ def test_two_chunked_prompts_batch_and_match_solo():
     # budget packs across prompt boundaries: 140 tokens / 64-token chunks
     # = 3 batched paged-prefill calls (serial, unpacked would take 4)
     assert eng.stats["prefill_batches"] == 3
+
+
+def test_embeddings_normalized_and_batch_invariant():
+    """v1/embeddings path: unit-norm vectors, identical whether texts are
+    embedded together or one at a time (padding must not leak)."""
+    torch.manual_seed(3)
+    eng = tiny_engine()
+    texts = ["hello world", "a much longer sentence for the second row"]
+
+    async def both():
+        return await eng.embed_batch(texts)
+
+    async def solo(t):
+        return (await eng.embed_batch([t]))[0]
+
+    vb = run(both())
+    v0 = run(solo(texts[0]))
+    v1 = run(solo(texts[1]))
+    for v in (vb[0], vb[1], v0, v1):
+        assert abs(sum(x * x for x in v) - 1.0) < 1e-4
+    assert max(abs(a - b) for a, b in zip(vb[0], v0)) < 1e-4
+    assert max(abs(a - b) for a, b in zip(vb[1], v1)) < 1e-4
+    # different texts -> different embeddings
+    assert max(abs(a - b) for a, b in zip(vb[0], vb[1])) > 1e-3
+
+
+def test_tokenize_detokenize_roundtrip():
+    eng = tiny_engine()
+    out = eng.openai_tokenize({"prompt": "round trip!"})
+    assert out["count"] == len(out["tokens"]) > 0
+    back = eng.openai_detokenize({"tokens": out["tokens"]})
+    assert back["prompt"] == "round trip!"
