@@ -83,6 +83,19 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     async def v1_embeddings(self, body, state, collect_fn=None):
         return await self._engine.openai_embeddings(body, self._served_name)
 
+    async def pooling(self, body, state, collect_fn=None):
+        return await self._engine.openai_pooling(body, self._served_name)
+
+    async def v1_score(self, body, state, collect_fn=None):
+        return await self._engine.openai_score(body, self._served_name)
+
+    async def v1_rerank(self, body, state, collect_fn=None):
+        return await self._engine.openai_rerank(body, self._served_name)
+
+    # vLLM also mounts rerank at /v2/rerank for cohere-client compat
+    async def v2_rerank(self, body, state, collect_fn=None):
+        return await self._engine.openai_rerank(body, self._served_name)
+
     # vLLM-compatible token utility routes (/serve/openai/tokenize)
     async def tokenize(self, body, state, collect_fn=None):
         return self._engine.openai_tokenize(body)

@@ -580,7 +580,9 @@ class LlmEngine:
         mask = (torch.arange(smax, device=dev)[None, :]
                 < torch.tensor(lens, device=dev)[:, None]).unsqueeze(-1)
         pooled = (hidden * mask).sum(1) / mask.sum(1).clamp(min=1)
-        return torch.nn.functional.normalize(pooled, dim=-1)
+        if plan.get("normalize", True):
+            pooled = torch.nn.functional.normalize(pooled, dim=-1)
+        return pooled
 
     async def embed_batch(self, texts: List[str]) -> List[List[float]]:
         if self.model is None:
@@ -808,6 +810,54 @@ class LlmEngine:
                      for i, v in enumerate(vecs)],
             "usage": {"prompt_tokens": nprompt, "total_tokens": nprompt},
         }
+
+    async def openai_pooling(self, body: Dict[str, Any], model_name: str):
+        """vLLM-style /pooling: raw mean-pooled hidden states (no L2
+        normalization, unlike /v1/embeddings)."""
+        inp = body.get("input")
+        if inp is None:
+            raise ValueError("pooling request requires 'input'")
+        texts = inp if isinstance(inp, list) else [inp]
+        prompts = [self.tokenizer.encode(str(t))[:self.cfg.max_model_len]
+                   for t in texts]
+        plan = {"mode": "embed", "prompts": prompts, "normalize": False}
+        self._tp_broadcast(plan)
+        out = self._exec_embed(plan).cpu().tolist()
+        return {"object": "list", "model": model_name,
+                "data": [{"object": "pooling", "index": i, "data": v}
+                         for i, v in enumerate(out)]}
+
+    async def openai_score(self, body: Dict[str, Any], model_name: str):
+        """Bi-encoder similarity score between text_1 and text_2 (cosine of
+        pooled embeddings). The reference delegates to vLLM cross-encoder
+        models (preprocess_service.py:632-1095); with a decoder-only model
+        the bi-encoder cosine is the native equivalent."""
+        t1 = body.get("text_1")
+        t2 = body.get("text_2")
+        if t1 is None or t2 is None:
+            raise ValueError("score request requires 'text_1' and 'text_2'")
+        rights = t2 if isinstance(t2, list) else [t2]
+        vecs = await self.embed_batch([str(t1)] + [str(r) for r in rights])
+        q = vecs[0]
+        scores = [sum(a * b for a, b in zip(q, v)) for v in vecs[1:]]
+        return {"object": "list", "model": model_name,
+                "data": [{"object": "score", "index": i, "score": s}
+                         for i, s in enumerate(scores)]}
+
+    async def openai_rerank(self, body: Dict[str, Any], model_name: str):
+        query = body.get("query")
+        docs = body.get("documents") or []
+        if query is None or not docs:
+            raise ValueError("rerank request requires 'query' and 'documents'")
+        vecs = await self.embed_batch([str(query)] + [str(d) for d in docs])
+        q = vecs[0]
+        scored = sorted(
+            ((sum(a * b for a, b in zip(q, v)), i)
+             for i, v in enumerate(vecs[1:])), reverse=True)
+        top_n = body.get("top_n") or len(docs)
+        return {"model": model_name, "results": [
+            {"index": i, "document": {"text": str(docs[i])},
+             "relevance_score": s} for s, i in scored[:top_n]]}
 
     def openai_tokenize(self, body: Dict[str, Any]):
         prompt = body.get("prompt")

@@ -139,3 +139,30 @@ def test_tokenize_route(llm_client):
     r2 = llm_client.post("/serve/openai/detokenize", json={
         "model": "test_llm", "tokens": toks})
     assert r2.json()["prompt"] == "hi there"
+
+
+def test_score_and_rerank_routes(llm_client):
+    r = llm_client.post("/serve/openai/v1/score", json={
+        "model": "test_llm", "text_1": "query text",
+        "text_2": ["query text", "unrelated words entirely"]})
+    assert r.status_code == 200, r.text
+    scores = [d["score"] for d in r.json()["data"]]
+    assert len(scores) == 2
+    # identical text scores (cosine) ~1.0 and above the unrelated text
+    assert scores[0] > 0.99 and scores[0] >= scores[1]
+
+    r = llm_client.post("/serve/openai/v1/rerank", json={
+        "model": "test_llm", "query": "query text",
+        "documents": ["unrelated words entirely", "query text"], "top_n": 1})
+    assert r.status_code == 200, r.text
+    res = r.json()["results"]
+    assert len(res) == 1 and res[0]["index"] == 1
+
+
+def test_pooling_route_unnormalized(llm_client):
+    r = llm_client.post("/serve/openai/pooling", json={
+        "model": "test_llm", "input": "some text"})
+    assert r.status_code == 200, r.text
+    vec = r.json()["data"][0]["data"]
+    norm = sum(x * x for x in vec) ** 0.5
+    assert norm > 0 and abs(norm - 1.0) > 1e-3  # raw, not unit-norm
