@@ -230,3 +230,21 @@ def test_fp8_quantized_engine():
     fp8 = gen("fp8")
     assert len(fp8) == 12
     assert all(isinstance(t, int) for t in fp8)
+
+
+def test_embeddings_gpu_matches_cpu():
+    """embed_batch on GPU (HIP bshd attention + rmsnorm) agrees with the
+    same engine run on CPU (fp32 torch reference ops)."""
+    def emb(device):
+        torch.manual_seed(17)
+        cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
+                              block_size=16, max_model_len=128, device=device)
+        eng = LlmEngine(cfg)
+        eng.start()
+        out = run(eng.embed_batch(["embedding parity probe",
+                                   "second, longer text for the batch"]))
+        return torch.tensor(out)
+
+    g = emb(DEV)
+    c = emb("cpu")
+    assert torch.allclose(g, c, atol=2e-2), (g - c).abs().max()
