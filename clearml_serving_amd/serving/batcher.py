@@ -97,7 +97,11 @@ class DynamicBatcher:
 
     def _ensure_worker(self) -> None:
         loop = asyncio.get_running_loop()
-        if self._worker_task is None or self._loop is not loop:
+        if (self._worker_task is None or self._loop is not loop
+                or self._worker_task.done()):
+            # .done() guard: restart a worker that died on an unexpected
+            # exception (mirrors the LLM engine-loop restart) -- otherwise
+            # every future queued after the crash would hang forever
             self._loop = loop
             self._queue = asyncio.Queue()
             self._worker_task = loop.create_task(self._worker())

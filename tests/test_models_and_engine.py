@@ -249,3 +249,27 @@ def test_batcher_respects_queue_delay():
 
     elapsed = run(main())
     assert 0.02 <= elapsed < 0.5  # waited for the window, not forever
+
+
+def test_batcher_worker_restarts_after_crash():
+    """A worker killed by an unexpected exception is replaced on the next
+    submit instead of hanging every future."""
+    def model_fn(x):
+        return x * 2
+
+    batcher = DynamicBatcher(model_fn, device="cpu", max_batch_size=4,
+                             max_queue_delay_us=1000, use_graphs=False)
+
+    async def main():
+        out = await batcher.submit(torch.ones(2))
+        # kill the worker the way a code bug would
+        batcher._worker_task.cancel()
+        try:
+            await batcher._worker_task
+        except asyncio.CancelledError:
+            pass
+        out2 = await batcher.submit(torch.ones(2))
+        return out, out2
+
+    out, out2 = run(main())
+    assert torch.equal(out, out2)
