@@ -23,6 +23,7 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     # different models from one process -- placement spreads them over the
     # node's GPUs via auxiliary_cfg {"device": "cuda:N"}
     _engines = {}
+    _engine_refs = {}  # model key -> live adapter count
     _engine_singleton = None  # kept for tests/back-compat (first engine)
 
     def __init__(self, model_endpoint: ModelEndpoint, task=None):
@@ -43,8 +44,28 @@ class LlmPreprocessRequest(BasePreprocessRequest):
             LlmPreprocessRequest._engines[key] = engine
             if LlmPreprocessRequest._engine_singleton is None:
                 LlmPreprocessRequest._engine_singleton = engine
+        self._engine_key = key
+        cls = LlmPreprocessRequest
+        cls._engine_refs[key] = cls._engine_refs.get(key, 0) + 1
         self._engine = LlmPreprocessRequest._engines[key]
         self._served_name = model_endpoint.serving_url
+
+    def shutdown(self) -> None:
+        """Called by the processor when this endpoint is flushed on a config
+        reload. Auto-update endpoints cycle model versions -- without this,
+        every superseded version's engine (weights + KV cache HBM) would
+        stay cached in ``_engines`` for the life of the process."""
+        cls = LlmPreprocessRequest
+        key = getattr(self, "_engine_key", None)
+        if key is None or key not in cls._engines:
+            return
+        cls._engine_refs[key] = cls._engine_refs.get(key, 1) - 1
+        if cls._engine_refs[key] <= 0:
+            engine = cls._engines.pop(key)
+            cls._engine_refs.pop(key, None)
+            engine.stop()
+            if cls._engine_singleton is engine:
+                cls._engine_singleton = None
 
     async def preprocess(self, request, state, collect_custom_statistics_fn=None):
         if self._preprocess is not None and hasattr(self._preprocess, "preprocess"):

@@ -353,6 +353,29 @@ class LlmEngine:
             self._wake = asyncio.Event()
             self._loop_task = loop.create_task(self._engine_loop())
 
+    def stop(self) -> None:
+        """Thread-safe teardown (endpoint removed on a config reload):
+        cancel the scheduler loop, poison active sequences so no client
+        hangs, and drop the model + KV cache references so their HBM frees
+        with the processor's post-flush gc."""
+        task, loop = self._loop_task, getattr(self, "_loop_ref", None)
+        self._loop_task = None
+        waiting, running = self.waiting, self.running
+        self.waiting, self.running = [], []
+
+        def _teardown():
+            if task is not None and not task.done():
+                task.cancel()
+            for seq in waiting + running:
+                seq.stream.put_nowait({"error": "engine stopped",
+                                       "finished": True, "token_ids": []})
+
+        if loop is not None and not loop.is_closed():
+            loop.call_soon_threadsafe(_teardown)
+        self.tp_shutdown()
+        self.kv_caches = []
+        self.model = None
+
     async def _engine_loop(self) -> None:
         while True:
             if not self.waiting and not self.running:

@@ -472,6 +472,9 @@ class ModelRequestProcessor:
                     batcher = getattr(engine, "_batcher", None)
                     if batcher is not None:
                         batcher.shutdown()  # free graphs/streams/HBM
+                    shutdown = getattr(engine, "shutdown", None)
+                    if callable(shutdown):
+                        shutdown()  # e.g. LLM engine refcount/teardown
             self._metric_cfg_cache.clear()
             self._last_revision = revision
             self._update_lock_flag = False

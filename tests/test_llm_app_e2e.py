@@ -166,3 +166,58 @@ def test_pooling_route_unnormalized(llm_client):
     vec = r.json()["data"][0]["data"]
     norm = sum(x * x for x in vec) ** 0.5
     assert norm > 0 and abs(norm - 1.0) > 1e-3  # raw, not unit-norm
+
+
+def test_removed_llm_endpoint_frees_engine(processor, store, tmp_path):
+    """Swapping an LLM endpoint to a new model tears down the superseded
+    model's engine (auto-update would otherwise leak one engine's HBM per
+    version)."""
+    import asyncio
+
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+
+    LlmPreprocessRequest._engine_singleton = None
+    LlmPreprocessRequest._engines = {}
+    LlmPreprocessRequest._engine_refs = {}
+
+    def card(name):
+        p = tmp_path / (name + ".json")
+        p.write_text(json.dumps({
+            "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 32,
+            "block_size": 16, "max_model_len": 64, "device": "cpu"}))
+        return store.register_model(name=name, project="p", path=str(p))
+
+    loop = asyncio.new_event_loop()
+    try:
+        r1 = card("v1")
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="llm", serving_url="swap_llm", model_id=r1.model_id))
+        processor.serialize()
+        out = loop.run_until_complete(processor.process_request(
+            "swap_llm", None, {"prompt": "x", "max_tokens": 2,
+                               "ignore_eos": True, "temperature": 0.0}))
+        assert out["tokens"] == 2
+        assert len(LlmPreprocessRequest._engines) == 1
+        old_engine = LlmPreprocessRequest._engines[r1.model_id]
+
+        r2 = card("v2")
+        processor.add_endpoint(ModelEndpoint(
+            engine_type="llm", serving_url="swap_llm", model_id=r2.model_id))
+        processor.serialize()
+        processor.deserialize()  # what the sync daemon does on config change
+        out = loop.run_until_complete(processor.process_request(
+            "swap_llm", None, {"prompt": "x", "max_tokens": 2,
+                               "ignore_eos": True, "temperature": 0.0}))
+        assert out["tokens"] == 2
+        # old engine evicted and its model/KV references dropped
+        assert r1.model_id not in LlmPreprocessRequest._engines
+        assert old_engine.model is None and old_engine.kv_caches == []
+        assert len(LlmPreprocessRequest._engines) == 1
+    finally:
+        for t in asyncio.all_tasks(loop):
+            t.cancel()
+        loop.run_until_complete(asyncio.sleep(0))
+        loop.close()
+        LlmPreprocessRequest._engine_singleton = None
+        LlmPreprocessRequest._engines = {}
+        LlmPreprocessRequest._engine_refs = {}
