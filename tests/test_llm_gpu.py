@@ -247,4 +247,7 @@ def test_embeddings_gpu_matches_cpu():
 
     g = emb(DEV)
     c = emb("cpu")
-    assert torch.allclose(g, c, atol=2e-2), (g - c).abs().max()
+    # bf16 GPU vs fp32 CPU: compare direction (the semantic content of a
+    # normalized embedding), not elementwise values
+    cos = (g * c).sum(-1)
+    assert (cos > 0.99).all(), cos
