@@ -370,3 +370,41 @@ def test_preprocess_package_folder(processor, store, tmp_path):
         preprocess_code=str(pkg))
     out = run(processor.process_request("pkg_ep", "", {"x": 7}))
     assert out == {"y": 21}
+
+
+def test_xgboost_engine_without_library_fails_loudly(processor, store, tmp_path):
+    """xgboost isn't installed in this image: the endpoint registers and
+    validates, and first use raises a clear ModuleNotFoundError instead of
+    silently serving nothing."""
+    f = tmp_path / "booster.json"
+    f.write_text("{}")
+    rec = store.register_model(name="xgb", project="p", path=str(f))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="xgboost", serving_url="xgb_ep", model_id=rec.model_id))
+    with pytest.raises(Exception) as ei:
+        run(processor.process_request("xgb_ep", "", [[1.0, 2.0]]))
+    assert "xgboost" in str(ei.value)
+
+
+def test_xgboost_engine_predict():
+    xgboost = pytest.importorskip("xgboost")  # covered where installed
+    import numpy as np
+
+    from clearml_serving_amd.serving.preprocess import XGBoostPreprocessRequest
+
+    X = np.random.rand(64, 4)
+    y = (X[:, 0] > 0.5).astype(int)
+    bst = xgboost.train({"max_depth": 2}, xgboost.DMatrix(X, label=y), 5)
+    out = bst.predict(xgboost.DMatrix(X[:4]))
+    assert out.shape[0] == 4
+
+
+def test_lightgbm_engine_predict():
+    lightgbm = pytest.importorskip("lightgbm")
+    import numpy as np
+
+    X = np.random.rand(64, 4)
+    y = (X[:, 0] > 0.5).astype(int)
+    bst = lightgbm.train({"objective": "binary", "min_data_in_leaf": 4},
+                         lightgbm.Dataset(X, label=y), num_boost_round=5)
+    assert bst.predict(X[:4]).shape[0] == 4
