@@ -111,6 +111,10 @@ class LlmEngineConfig:
     weights: Optional[str] = None
     tokenizer_path: Optional[str] = None
     device: Optional[str] = None
+    decode_graphs: bool = True  # capture decode steps into hipGraphs per
+                                # batch bucket (GPU, TP=1): collapses the
+                                # ~350 Python-dispatched launches of a
+                                # llama-8B decode step into one replay
 
     @classmethod
     def from_aux(cls, model_path: Optional[str], aux: Dict[str, Any]):
@@ -137,7 +141,7 @@ class LlmEngineConfig:
         for key in ("preset", "dtype", "block_size", "max_num_seqs",
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "quantization",
-                    "weights", "device"):
+                    "weights", "device", "decode_graphs"):
             for src in (card, aux):
                 if key in src and src[key] is not None:
                     setattr(cfg, key, src[key])
@@ -207,6 +211,15 @@ class LlmEngine:
         self._loop_task: Optional[asyncio.Task] = None
         self._wake: Optional[asyncio.Event] = None
         self._started = False
+        # decode hipGraphs: one per batch bucket, shared memory pool
+        self._decode_graphs: Dict[int, Dict[str, Any]] = {}
+        self._decode_pin: Dict[int, Dict[str, Any]] = {}
+        self._graph_pool = None
+        self._decode_buckets = [n for n in
+                                (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
+                                 192, 256)
+                                if n < self.cfg.max_num_seqs]
+        self._decode_buckets.append(self.cfg.max_num_seqs)
         self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
                       "prefill_batches": 0, "decode_batches": 0,
                       "preemptions": 0, "aborts": 0}
@@ -373,6 +386,9 @@ class LlmEngine:
         if loop is not None and not loop.is_closed():
             loop.call_soon_threadsafe(_teardown)
         self.tp_shutdown()
+        self._decode_graphs.clear()
+        self._decode_pin.clear()
+        self._graph_pool = None
         self.kv_caches = []
         self.model = None
 
@@ -669,6 +685,8 @@ class LlmEngine:
 
     @torch.inference_mode()
     def _exec_decode(self, plan: Dict[str, Any]) -> torch.Tensor:
+        if self._use_decode_graphs():
+            return self._exec_decode_graph(plan)
         dev = self.device
         b = len(plan["tokens"])
         tokens = torch.tensor(plan["tokens"], dtype=torch.long, device=dev)
@@ -688,6 +706,110 @@ class LlmEngine:
         }
         return self.model(tokens, positions, kv_caches=self.kv_caches,
                           attn_ctx=attn_ctx, last_token_idx=None)
+
+    # -------------------- decode hipGraph capture --------------------- #
+    # A llama-8B decode step dispatches ~350 kernels from Python; at B=64
+    # the weight streaming itself is only ~2.5 ms, so the step is
+    # launch/dispatch bound. Capturing one hipGraph per batch bucket
+    # (padded rows: slot=-1 so the KV scatter skips them, block_table=0 /
+    # seq_len=1 so the attention read is bounded) replays the whole step
+    # as a single launch. vLLM uses the same strategy on its decode path.
+    def _use_decode_graphs(self) -> bool:
+        return (self.cfg.decode_graphs
+                and getattr(self, "tp_size", 1) <= 1
+                and self.device.type == "cuda"
+                and os.environ.get("CMLS_LLM_GRAPHS", "1") != "0")
+
+    def _decode_bucket(self, b: int) -> int:
+        for cand in self._decode_buckets:
+            if cand >= b:
+                return cand
+        return self._decode_buckets[-1]
+
+    def _graph_entry(self, bucket: int) -> Dict[str, Any]:
+        entry = self._decode_graphs.get(bucket)
+        if entry is not None:
+            return entry
+        dev = self.device
+        mb = (self.cfg.max_model_len + self.cfg.block_size - 1) \
+            // self.cfg.block_size
+        static = {
+            "tokens": torch.zeros(bucket, dtype=torch.long, device=dev),
+            "positions": torch.zeros(bucket, dtype=torch.int32, device=dev),
+            "slots": torch.full((bucket,), -1, dtype=torch.int32, device=dev),
+            "seq_lens": torch.ones(bucket, dtype=torch.int32, device=dev),
+            "block_table": torch.zeros(bucket, mb, dtype=torch.int32,
+                                       device=dev),
+        }
+        attn_ctx = {"mode": "decode", "seq_lens": static["seq_lens"],
+                    "block_table": static["block_table"],
+                    "slot_mapping": static["slots"]}
+
+        def fwd():
+            return self.model(static["tokens"], static["positions"],
+                              kv_caches=self.kv_caches, attn_ctx=attn_ctx,
+                              last_token_idx=None)
+
+        # warm up on a side stream (allocator + kernels settle), capture on
+        # the current stream; all graphs share one memory pool
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                fwd()
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        if self._graph_pool is None:
+            with torch.cuda.graph(graph):
+                static["logits"] = fwd()
+            self._graph_pool = graph.pool()
+        else:
+            with torch.cuda.graph(graph, pool=self._graph_pool):
+                static["logits"] = fwd()
+        entry = {"graph": graph, **static}
+        self._decode_graphs[bucket] = entry
+        return entry
+
+    def _exec_decode_graph(self, plan: Dict[str, Any]) -> torch.Tensor:
+        b = len(plan["tokens"])
+        bucket = self._decode_bucket(b)
+        entry = self._graph_entry(bucket)
+        # stage this step's inputs in pinned host memory, one async H2D per
+        # buffer into the graph's static tensors
+        pin = self._decode_pin.get(bucket)
+        mb = entry["block_table"].shape[1]
+        if pin is None:
+            pin = {
+                "tokens": torch.zeros(bucket, dtype=torch.long,
+                                      pin_memory=True),
+                "positions": torch.zeros(bucket, dtype=torch.int32,
+                                         pin_memory=True),
+                "slots": torch.empty(bucket, dtype=torch.int32,
+                                     pin_memory=True),
+                "seq_lens": torch.ones(bucket, dtype=torch.int32,
+                                       pin_memory=True),
+                "block_table": torch.zeros(bucket, mb, dtype=torch.int32,
+                                           pin_memory=True),
+            }
+            self._decode_pin[bucket] = pin
+        pin["tokens"][:b] = torch.tensor(plan["tokens"], dtype=torch.long)
+        pin["tokens"][b:] = 0
+        pin["positions"][:b] = torch.tensor(plan["positions"],
+                                            dtype=torch.int32)
+        pin["positions"][b:] = 0
+        pin["slots"][:b] = torch.tensor(plan["slots"], dtype=torch.int32)
+        pin["slots"][b:] = -1  # padded rows: KV scatter kernel skips slot<0
+        pin["seq_lens"][:b] = torch.tensor(plan["seq_lens"],
+                                           dtype=torch.int32)
+        pin["seq_lens"][b:] = 1
+        bt = pin["block_table"]
+        bt.zero_()
+        for i, bl in enumerate(plan["blocks"]):
+            bt[i, :len(bl)] = torch.tensor(bl, dtype=torch.int32)
+        for k in ("tokens", "positions", "slots", "seq_lens", "block_table"):
+            entry[k].copy_(pin[k], non_blocking=True)
+        entry["graph"].replay()
+        return entry["logits"][:b]
 
     # ------------------------------------------------------------------ #
     # tensor-parallel coordination (rank 0 schedules, workers follow)
