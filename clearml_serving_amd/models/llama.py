@@ -68,13 +68,21 @@ class LlamaLayer(nn.Module):
         self.down = nn.Linear(inter, h, bias=False)
         self.inter = inter
 
+    @staticmethod
+    def _proj(mod, x):
+        # decode-shaped rows route to the in-tree skinny MFMA kernel;
+        # prefill rows and non-plain modules (e.g. Fp8Linear) fall through
+        if type(mod) is nn.Linear:
+            return ops.skinny_linear(x, mod.weight)
+        return mod(x)
+
     def forward(self, x, residual, positions, kv_cache, attn_ctx):
         cfg = self.cfg
         # fused: residual += x_prev; x = rmsnorm(residual)
         x = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps, residual=residual)
         t = x.shape[0]
 
-        qkv = self.qkv(x)
+        qkv = self._proj(self.qkv, x)
         # token-strided views into the merged projection -- the rope /
         # kv_cache_write / decode-attention kernels are stride-aware, so no
         # .contiguous() copies on the decode hot path
@@ -115,7 +123,7 @@ class LlamaLayer(nn.Module):
                 q, k_cache, v_cache, attn_ctx["block_table"],
                 attn_ctx["seq_lens"], scale=scale)
             ctx = ctx.view(t, self.heads * self.head_dim)
-        attn_out = self.o_proj(ctx)
+        attn_out = self._proj(self.o_proj, ctx)
         if self.tp_size > 1:
             from ..parallel import tp as tp_mod
 
@@ -125,7 +133,7 @@ class LlamaLayer(nn.Module):
                         residual=residual)
         gate_up = self.gate_up(x)
         gate, up = gate_up.split([self.inter, self.inter], dim=-1)
-        mlp_out = self.down(ops.silu_mul(gate, up))  # stride-aware, no copy
+        mlp_out = self._proj(self.down, ops.silu_mul(gate, up))
         if self.tp_size > 1:
             from ..parallel import tp as tp_mod
 

@@ -403,3 +403,26 @@ def gemm_bf16(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
                                      b2.contiguous())[:m, :n].contiguous()
             return ext.gemm_bf16(a.contiguous(), b.contiguous())
     return (a.float() @ b.float().T).to(a.dtype)
+
+
+def skinny_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """y = x @ weight.T for decode-shaped activations (rows <= 128).
+
+    Routes to the in-tree skinny MFMA kernel on GPU exactly where it
+    MEASURES faster than hipBLASLt (profiles/skinny_bench.txt): the
+    small-N llama projections at M <= 16 (qkv 1.2x, o_proj 2.0-2.8x --
+    hipBLASLt's tiles under-fill the 256-CU chip there) and the large-K
+    down projection at M <= 2. Everything else falls back to
+    torch.nn.functional.linear (hipBLASLt).
+    """
+    if (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 2
+            and x.shape[1] % 32 == 0 and weight.shape[0] % 16 == 0
+            and os.environ.get("CMLS_SKINNY", "1") != "0"):
+        m, k = x.shape
+        n = weight.shape[0]
+        if ((1 <= m <= 16 and n <= 8192 and k <= 8192)
+                or (m <= 2 and n <= 8192)):
+            ext = _require_ext("skinny_gemm")
+            if ext is not None:
+                return ext.skinny_gemm(x.contiguous(), weight, 1)
+    return torch.nn.functional.linear(x, weight)

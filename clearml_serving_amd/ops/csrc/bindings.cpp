@@ -25,6 +25,7 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
                     torch::Tensor slot_mapping);
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b);
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, int64_t variant);
 torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor k_cache,
                                       torch::Tensor v_cache,
                                       torch::Tensor block_table,
@@ -53,5 +54,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_decode", &attention_decode);
   m.def("kv_cache_write", &kv_cache_write);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("skinny_gemm", &skinny_gemm, py::arg("a"), py::arg("w"),
+        py::arg("variant") = 0);
   m.def("attention_prefill_paged", &attention_prefill_paged);
 }

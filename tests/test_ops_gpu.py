@@ -298,3 +298,30 @@ def test_attention_bshd_layout_strided_views():
         k.permute(0, 2, 1, 3).float().cpu(),
         v.permute(0, 2, 1, 3).float().cpu()).permute(0, 2, 1, 3)
     assert_close_bf16(got, ref, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("variant", [1, 2])
+@pytest.mark.parametrize("m", [1, 7, 16, 64, 100, 128])
+def test_skinny_gemm_matches_fp32(m, variant):
+    torch.manual_seed(m)
+    k, n = 256, 320
+    x = (torch.randn(m, k, device=DEV) / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=DEV) / 8).to(torch.bfloat16)
+    ext = ops._require_ext("skinny_gemm")
+    got = ext.skinny_gemm(x, w, variant).float().cpu()
+    ref = x.float().cpu() @ w.float().cpu().T
+    assert torch.allclose(got, ref, atol=5e-2, rtol=5e-2), \
+        (got - ref).abs().max()
+
+
+def test_skinny_linear_llama_shapes():
+    """The production shapes route through the kernel and agree with
+    hipBLASLt."""
+    torch.manual_seed(0)
+    for k, n in ((4096, 6144), (4096, 4096), (14336, 4096)):
+        x = (torch.randn(64, k, device=DEV) / 16).to(torch.bfloat16)
+        w = (torch.randn(n, k, device=DEV) / 16).to(torch.bfloat16)
+        got = ops.skinny_linear(x, w).float()
+        ref = torch.nn.functional.linear(x, w).float()
+        assert torch.allclose(got, ref, atol=8e-2, rtol=8e-2), \
+            (got - ref).abs().max()
