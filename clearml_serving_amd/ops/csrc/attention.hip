@@ -108,8 +108,11 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
                              : (long)b * st.kb + (long)hkv * st.kh;
   const long o_base = (long)b * st.ob + (long)h * st.oh;
   const int kv_len = HAS_SEQLENS ? min(seq_lens[b], Sk) : Sk;
-  // per-seq query length (chunked prefill pads chunks to Sq)
-  const int q_len = q_lens ? q_lens[b] : Sq;
+  // per-seq query length: explicit q_lens (chunked prefill), else kv_len
+  // when seq_lens is given (padded dense batch: rows beyond kv_len are
+  // pad -- q_len = Sq here made causal_off NEGATIVE for short sequences
+  // and truncated their history by the pad amount)
+  const int q_len = q_lens ? q_lens[b] : (HAS_SEQLENS ? kv_len : Sq);
   // causal: the chunk's LAST query row attends up to the LAST key
   // (history offset = kv_len - q_len)
   const int causal_off = kv_len - q_len;

@@ -325,3 +325,25 @@ def test_skinny_linear_llama_shapes():
         ref = torch.nn.functional.linear(x, w).float()
         assert torch.allclose(got, ref, atol=8e-2, rtol=8e-2), \
             (got - ref).abs().max()
+
+
+def test_attention_causal_unequal_seq_lens():
+    """Padded causal batch with per-sequence lengths: every valid query of a
+    SHORT sequence must attend its full history (regression: the kernel's
+    causal offset went negative for padded rows and silently truncated
+    short sequences' attention -- caught by the embeddings parity test)."""
+    torch.manual_seed(0)
+    b, s, h, d = 3, 33, 4, 64
+    q = (torch.randn(b, s, h, d, device=DEV) / 8).to(torch.bfloat16)
+    k = (torch.randn(b, s, h, d, device=DEV) / 8).to(torch.bfloat16)
+    v = (torch.randn(b, s, h, d, device=DEV) / 8).to(torch.bfloat16)
+    lens = torch.tensor([22, 33, 5], dtype=torch.int32, device=DEV)
+    got = ops.attention(q, k, v, causal=True, seq_lens=lens, layout="bshd")
+    ref = ops.attention(q.cpu().float(), k.cpu().float(), v.cpu().float(),
+                        causal=True, seq_lens=lens.cpu(), layout="bshd")
+    for i in range(b):
+        n = int(lens[i])
+        gi = got[i, :n].float().cpu()
+        ri = ref[i, :n].float()
+        assert torch.allclose(gi, ri, atol=2e-2, rtol=2e-2), \
+            (i, (gi - ri).abs().max())
