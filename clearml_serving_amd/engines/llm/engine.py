@@ -618,7 +618,10 @@ class LlmEngine:
             return_hidden=True).view(b, smax, -1).float()
         mask = (torch.arange(smax, device=dev)[None, :]
                 < torch.tensor(lens, device=dev)[:, None]).unsqueeze(-1)
-        pooled = (hidden * mask).sum(1) / mask.sum(1).clamp(min=1)
+        # where(), not multiply: pad rows hold garbage (possibly NaN/inf
+        # from fully-masked attention) and NaN * 0 is NaN
+        pooled = torch.where(mask, hidden, 0.0).sum(1) \
+            / mask.sum(1).clamp(min=1)
         if plan.get("normalize", True):
             pooled = torch.nn.functional.normalize(pooled, dim=-1)
         return pooled
