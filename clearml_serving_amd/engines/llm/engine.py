@@ -211,6 +211,7 @@ class LlmEngine:
         self._loop_task: Optional[asyncio.Task] = None
         self._wake: Optional[asyncio.Event] = None
         self._started = False
+        self._stopped = False
         # decode hipGraphs: one per batch bucket, shared memory pool
         self._decode_graphs: Dict[int, Dict[str, Any]] = {}
         self._decode_pin: Dict[int, Dict[str, Any]] = {}
@@ -305,6 +306,8 @@ class LlmEngine:
     # ------------------------------------------------------------------ #
     async def add_request(self, prompt_ids: List[int],
                           params: SamplingParams) -> Sequence:
+        if self._stopped:
+            raise RuntimeError("LLM engine stopped (endpoint was removed)")
         if len(prompt_ids) >= self.cfg.max_model_len:
             raise ValueError(
                 "prompt length {} exceeds max_model_len {}".format(
@@ -371,6 +374,7 @@ class LlmEngine:
         cancel the scheduler loop, poison active sequences so no client
         hangs, and drop the model + KV cache references so their HBM frees
         with the processor's post-flush gc."""
+        self._stopped = True
         task, loop = self._loop_task, getattr(self, "_loop_ref", None)
         self._loop_task = None
         waiting, running = self.waiting, self.running

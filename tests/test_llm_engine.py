@@ -459,3 +459,21 @@ def test_tokenize_detokenize_roundtrip():
     assert out["count"] == len(out["tokens"]) > 0
     back = eng.openai_detokenize({"tokens": out["tokens"]})
     assert back["prompt"] == "round trip!"
+
+
+def test_stopped_engine_rejects_requests():
+    eng = tiny_engine()
+
+    async def go():
+        seq = await eng.add_request([1, 2, 3], SamplingParams(
+            temperature=0.0, max_tokens=2, ignore_eos=True))
+        while True:
+            item = await seq.stream.get()
+            if item["finished"]:
+                break
+        eng.stop()
+        with pytest.raises(RuntimeError, match="stopped"):
+            await eng.add_request([4, 5], SamplingParams(max_tokens=2))
+
+    run(go())
+    assert eng.model is None
