@@ -72,6 +72,9 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
     int block_size, int max_blocks,
     AttnStrides st,
     int B, int H, int Hkv, int Sq, int Sk, float scale) {
+  // softmax in the exp2 domain: v_exp_f32 IS 2^x, so folding log2(e) into
+  // the score scale deletes one full-rate VALU mul per exp call
+  const float scale2 = scale * 1.4426950408889634f;
   constexpr int D = HEAD_DIM;
   constexpr int KSTRIDE = D + PAD;          // LDS K row stride (bf16)
   constexpr int VSTRIDE = BLOCK_N + PAD;    // LDS V^T row stride
@@ -229,17 +232,17 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
         float rmax = -INFINITY;
 #pragma unroll
         for (int t = 0; t < BLOCK_N / 16; ++t) {
-          const float s = acc_s[t][r] * scale;
+          const float s = acc_s[t][r] * scale2;
           pvals[r][t] = s;
           rmax = fmaxf(rmax, s);
         }
         rmax = rowgroup_max(rmax);
         const float m_new = fmaxf(m_run[r], rmax);
-        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m_run[r] - m_new);
         float rsum = 0.f;
 #pragma unroll
         for (int t = 0; t < BLOCK_N / 16; ++t) {
-          const float p = __expf(pvals[r][t] - m_new);
+          const float p = __builtin_amdgcn_exp2f(pvals[r][t] - m_new);
           pvals[r][t] = p;
           rsum += p;
         }
@@ -254,7 +257,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
         float rmax = -INFINITY;
 #pragma unroll
         for (int t = 0; t < BLOCK_N / 16; ++t) {
-          float s = acc_s[t][r] * scale;
+          float s = acc_s[t][r] * scale2;
           const int key = col_base + t * 16;
           bool valid = key < kv_len;
           if (CAUSAL) valid = valid && (key <= qrow + causal_off);
@@ -265,13 +268,13 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_kernel(
         rmax = rowgroup_max(rmax);
         const float m_new = fmaxf(m_run[r], rmax);
         // all-masked rows keep m = -inf; exp() yields 0 contributions
-        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+        alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m_run[r] - m_new);
         float rsum = 0.f;
 #pragma unroll
         for (int t = 0; t < BLOCK_N / 16; ++t) {
           const float p = (pvals[r][t] == -INFINITY || m_new == -INFINITY)
                               ? 0.f
-                              : __expf(pvals[r][t] - m_new);
+                              : __builtin_amdgcn_exp2f(pvals[r][t] - m_new);
           pvals[r][t] = p;
           rsum += p;
         }
