@@ -53,6 +53,8 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
     float* __restrict__ partial_ml,              // [B, H, SPLITS, 2] f32
     int splits, long q_bstride,
     int H, int Hkv, int block_size, int max_blocks, float scale) {
+  // softmax in the exp2 domain (v_exp_f32 is 2^x; log2e folds into scale)
+  const float scale2 = scale * 1.4426950408889634f;
   constexpr int NW = 4;        // waves
   constexpr int GROUPS = 4;    // 16-lane key groups per wave
   constexpr int EPL = D / 16;  // elements per lane (8 for D=128)
@@ -147,7 +149,7 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
           score[u][g] += __shfl_xor(score[u][g], off, 64);
-        score[u][g] = valid[u] ? score[u][g] * scale : -INFINITY;
+        score[u][g] = valid[u] ? score[u][g] * scale2 : -INFINITY;
       }
     }
     // ONE online-softmax update per chunk per head (rescaling O per
@@ -161,13 +163,14 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
       if (cmax == -INFINITY) continue;
       const float m_new = fmaxf(m_run[g], cmax);
       const float alpha =
-          (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+          (m_run[g] == -INFINITY) ? 0.f
+                                  : __builtin_amdgcn_exp2f(m_run[g] - m_new);
       float psum = 0.f;
 #pragma unroll
       for (int e = 0; e < EPL; ++e) o_acc[g][e] *= alpha;
 #pragma unroll
       for (int u = 0; u < UNROLL; ++u) {
-        const float p = valid[u] ? __expf(score[u][g] - m_new) : 0.f;
+        const float p = valid[u] ? __builtin_amdgcn_exp2f(score[u][g] - m_new) : 0.f;
         psum += p;
 #pragma unroll
         for (int e = 0; e < EPL; ++e)
@@ -213,7 +216,7 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
       for (int e = 0; e < EPL; ++e) o_tot[e] = 0.f;
       for (int w = 0; w < NW; ++w) {
         if (lds_m[w][g] == -INFINITY) continue;
-        const float f = __expf(lds_m[w][g] - m_tot);
+        const float f = __builtin_amdgcn_exp2f(lds_m[w][g] - m_tot);
         l_tot += lds_l[w][g] * f;
 #pragma unroll
         for (int e = 0; e < EPL; ++e)
@@ -262,7 +265,8 @@ __global__ void decode_combine_kernel(
   for (int s = 0; s < splits; ++s) {
     const float m = partial_ml[(base + s) * 2];
     if (m == -INFINITY) continue;
-    const float f = __expf(m - m_tot);
+    // m values are in the exp2 (log2-probability) domain
+    const float f = __builtin_amdgcn_exp2f(m - m_tot);
     l_tot += partial_ml[(base + s) * 2 + 1] * f;
     const float* po = partial_o + (base + s) * D;
 #pragma unroll
