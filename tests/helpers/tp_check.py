@@ -61,8 +61,14 @@ def check_engine_protocol(rank, world):
 
             return await asyncio.gather(one(0), one(1), one(2))
 
-        outs = asyncio.new_event_loop().run_until_complete(gen())
+        loop = asyncio.new_event_loop()
+        outs = loop.run_until_complete(gen())
         assert all(len(o) == 6 for o in outs), outs
+        # embeddings path through the same plan-broadcast protocol
+        vecs = loop.run_until_complete(
+            eng.embed_batch(["tp embed a", "a longer tp embed text b"]))
+        assert len(vecs) == 2
+        assert abs(sum(x * x for x in vecs[0]) - 1.0) < 1e-4
         eng.tp_shutdown()
         print("TP-ENGINE-OK", flush=True)
     else:
