@@ -477,3 +477,52 @@ def test_stopped_engine_rejects_requests():
 
     run(go())
     assert eng.model is None
+
+
+def test_engine_loads_safetensors_weights(tmp_path):
+    """cfg.weights: a real safetensors checkpoint loads (and TP-shards)
+    into the engine's model instead of random init."""
+    safetensors = pytest.importorskip("safetensors.torch")
+
+    from clearml_serving_amd.models.llama import PRESETS, LlamaForCausalLM
+
+    torch.manual_seed(99)
+    src_model = LlamaForCausalLM(PRESETS["llama-tiny"])
+    path = tmp_path / "model.safetensors"
+    safetensors.save_file(
+        {k: v.clone() for k, v in src_model.state_dict().items()}, str(path))
+
+    eng = tiny_engine(weights=str(path))
+    got = eng.model.state_dict()
+    for key in ("embed.weight", "layers.0.qkv.weight", "lm_head.weight"):
+        assert torch.allclose(got[key].float(),
+                              src_model.state_dict()[key].float(),
+                              atol=1e-2), key
+
+    async def go():
+        toks = []
+        async for it in eng.generate("weights ok", SamplingParams(
+                temperature=0.0, max_tokens=3, ignore_eos=True)):
+            toks.extend(it["token_ids"])
+        return toks
+
+    assert len(run(go())) == 3
+
+
+def test_config_from_model_folder(tmp_path):
+    """A model FOLDER (card.json + model.safetensors [+ tokenizer.json])
+    resolves weights and settings like the reference's model-dir layout."""
+    import json as _json
+
+    from clearml_serving_amd.engines.llm.engine import LlmEngineConfig
+
+    d = tmp_path / "model_dir"
+    d.mkdir()
+    (d / "card.json").write_text(_json.dumps(
+        {"preset": "llama-tiny", "max_model_len": 77}))
+    (d / "model.safetensors").write_bytes(b"\0" * 8)
+    cfg = LlmEngineConfig.from_aux(str(d), {"block_size": 8})
+    assert cfg.preset == "llama-tiny"
+    assert cfg.max_model_len == 77
+    assert cfg.block_size == 8          # aux overrides layer on top
+    assert cfg.weights == str(d / "model.safetensors")
