@@ -408,3 +408,18 @@ def test_lightgbm_engine_predict():
     bst = lightgbm.train({"objective": "binary", "min_data_in_leaf": 4},
                          lightgbm.Dataset(X, label=y), num_boost_round=5)
     assert bst.predict(X[:4]).shape[0] == 4
+
+
+def test_version_guard_warns_on_major_mismatch(store, capsys):
+    """Sessions record the package version at creation; opening one created
+    by a different MAJOR version prints a warning (reference parity:
+    __main__.py:24-40)."""
+    from clearml_serving_amd import __version__
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+
+    p = ModelRequestProcessor(store=store, name="vguard", force_create=True)
+    assert store.get_params(p.get_id())["serving_version"] == __version__
+
+    store.set_params(p.get_id(), {"serving_version": "99.0.0"})
+    ModelRequestProcessor(task_id=p.get_id(), store=store)
+    assert "created by version 99.0.0" in capsys.readouterr().out
