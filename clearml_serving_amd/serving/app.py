@@ -232,6 +232,8 @@ def create_app(
         return {
             "session": proc.get_id(),
             "revision": proc._last_revision,
+            "instances": proc.store.list_instances(proc.get_id(),
+                                                   max_age_sec=600),
             "endpoints": endpoints,
             "canary_routes": routes,
             "monitoring": {

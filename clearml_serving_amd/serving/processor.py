@@ -23,6 +23,7 @@ import asyncio
 import itertools
 import os
 import random
+import socket
 import threading
 import time
 from collections import deque
@@ -163,6 +164,10 @@ class ModelRequestProcessor:
     # ------------------------------------------------------------------ #
     # identity / config
     # ------------------------------------------------------------------ #
+    @property
+    def instance_id(self) -> str:
+        return "{}:{}".format(socket.gethostname(), os.getpid())
+
     def get_id(self) -> str:
         return self._session_id
 
@@ -772,6 +777,12 @@ class ModelRequestProcessor:
     def _sync_daemon(self, poll_frequency_sec: float) -> None:
         while not self._stop:
             try:
+                # keep-alive: this serving instance announces itself (the
+                # reference pings the control Task, :999-1007); no revision
+                # bump, so other instances don't see it as a config change
+                self._store.ping_instance(
+                    self._session_id, self.instance_id,
+                    {"pid": os.getpid(), "host": socket.gethostname()})
                 changed = self.deserialize(skip_sync=False)
                 changed |= self._update_monitored_models()
                 if changed:

@@ -58,6 +58,13 @@ CREATE TABLE IF NOT EXISTS params (
     value TEXT,
     PRIMARY KEY (session_id, name)
 );
+CREATE TABLE IF NOT EXISTS instances (
+    session_id TEXT NOT NULL,
+    instance_id TEXT NOT NULL,
+    info TEXT DEFAULT '{}',
+    last_ping REAL,
+    PRIMARY KEY (session_id, instance_id)
+);
 CREATE TABLE IF NOT EXISTS models (
     model_id TEXT PRIMARY KEY,
     name TEXT NOT NULL,
@@ -202,6 +209,37 @@ class ServingStore:
                 )
             self._bump(session_id)
             self._conn.commit()
+
+    def ping_instance(self, session_id: str, instance_id: str,
+                      info: Optional[Dict[str, Any]] = None) -> None:
+        """Serving-container keep-alive (reference parity: the sync daemon
+        pings the control Task, model_request_processor.py:999-1007).
+        Deliberately does NOT bump the session revision -- pings must not
+        look like config changes to other instances."""
+        with self._lock:
+            self._conn.execute(
+                "INSERT INTO instances (session_id, instance_id, info,"
+                " last_ping) VALUES (?,?,?,?)"
+                " ON CONFLICT(session_id, instance_id)"
+                " DO UPDATE SET last_ping=excluded.last_ping,"
+                " info=excluded.info",
+                (session_id, instance_id, json.dumps(info or {}),
+                 time.time()),
+            )
+            self._conn.commit()
+
+    def list_instances(self, session_id: str,
+                       max_age_sec: Optional[float] = None
+                       ) -> List[Dict[str, Any]]:
+        cur = self._conn.execute(
+            "SELECT instance_id, info, last_ping FROM instances"
+            " WHERE session_id=? ORDER BY last_ping DESC", (session_id,))
+        out = [{"instance_id": r[0], "info": json.loads(r[1] or "{}"),
+                "last_ping": r[2]} for r in cur.fetchall()]
+        if max_age_sec is not None:
+            now = time.time()
+            out = [i for i in out if now - (i["last_ping"] or 0) <= max_age_sec]
+        return out
 
     def get_params(self, session_id: str) -> Dict[str, Any]:
         cur = self._conn.execute(

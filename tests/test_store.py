@@ -92,3 +92,30 @@ def test_concurrent_writers(tmp_path):
     assert s1.get_config_object(sid, "a") == {"i": 19}
     assert s1.get_config_object(sid, "b") == {"i": 19}
     assert s1.revision(sid) == 40
+
+
+def test_instance_keepalive_no_revision_bump(tmp_path):
+    """Serving-container pings register liveness WITHOUT bumping the
+    session revision (a ping must not look like a config change)."""
+    import time as _t
+
+    from clearml_serving_amd.store import ServingStore
+
+    s = ServingStore(str(tmp_path / "k"))
+    sid = s.create_session(name="k")
+    rev = s.revision(sid)
+    s.ping_instance(sid, "hostA:1", {"pid": 1})
+    s.ping_instance(sid, "hostB:2", {"pid": 2})
+    assert s.revision(sid) == rev
+    inst = s.list_instances(sid)
+    assert {i["instance_id"] for i in inst} == {"hostA:1", "hostB:2"}
+    # re-ping updates, not duplicates
+    s.ping_instance(sid, "hostA:1", {"pid": 1})
+    assert len(s.list_instances(sid)) == 2
+    # stale filtering
+    old = _t.time() - 10_000
+    s._conn.execute("UPDATE instances SET last_ping=? WHERE instance_id=?",
+                    (old, "hostB:2"))
+    s._conn.commit()
+    assert [i["instance_id"] for i in s.list_instances(sid, max_age_sec=600)] \
+        == ["hostA:1"]
