@@ -1,0 +1,78 @@
+"""The examples are executable fixtures (the reference's substitute for
+tests): run the sklearn example exactly as its readme does --
+train script, CLI session + endpoint, HTTP serve -- in an isolated store."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(180)
+def test_sklearn_example_readme_flow(tmp_path, monkeypatch):
+    pytest.importorskip("sklearn")
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    # 1. train + register (writes sklearn-model.pkl into cwd)
+    run_py([os.path.join(ROOT, "examples/sklearn/train_model.py")],
+           cwd=str(tmp_path))
+    # 2. CLI: create session + add endpoint (readme commands)
+    run_py(["-m", "clearml_serving_amd", "create", "--name",
+            "serving example"], cwd=ROOT)
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "sklearn", "--endpoint", "test_model_sklearn", "--preprocess",
+            "examples/sklearn/preprocess.py", "--name",
+            "train sklearn model", "--project", "serving examples"],
+           cwd=ROOT)
+    # 3. HTTP serve (readme curl)
+    monkeypatch.setenv("CLEARML_SERVING_AMD_STORE", str(tmp_path / "store"))
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/test_model_sklearn",
+                        json={"x0": 1, "x1": 2})
+        assert r.status_code == 200, r.text
+        assert r.json() == {"y": [1]}
+
+
+@pytest.mark.timeout(120)
+def test_custom_example_readme_flow(tmp_path):
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=90)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "custom ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "custom", "--endpoint", "test_model_custom", "--preprocess",
+            "examples/custom/preprocess.py"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/test_model_custom", json={"x0": 1, "x1": 2})
+        assert r.status_code == 200, r.text
+        assert r.json() == {"y": 3.0}
