@@ -76,3 +76,40 @@ def test_custom_example_readme_flow(tmp_path):
         r = client.post("/serve/test_model_custom", json={"x0": 1, "x1": 2})
         assert r.status_code == 200, r.text
         assert r.json() == {"y": 3.0}
+
+
+@pytest.mark.timeout(180)
+def test_pipeline_example_readme_flow(tmp_path):
+    """Ensemble example: custom_async endpoint fanning out to the sklearn
+    endpoint via the injected in-process send_request hook."""
+    pytest.importorskip("sklearn")
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    run_py([os.path.join(ROOT, "examples/sklearn/train_model.py")],
+           cwd=str(tmp_path))
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "pipe ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "sklearn", "--endpoint", "test_model_sklearn", "--preprocess",
+            "examples/sklearn/preprocess.py", "--name",
+            "train sklearn model", "--project", "serving examples"])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "custom_async", "--endpoint", "ensemble", "--preprocess",
+            "examples/pipeline/async_preprocess.py"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/ensemble", json={"x0": 1, "x1": 2})
+        assert r.status_code == 200, r.text
+        assert r.json() == {"y": [1, 1]}
