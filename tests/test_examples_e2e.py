@@ -113,3 +113,55 @@ def test_pipeline_example_readme_flow(tmp_path):
         r = client.post("/serve/ensemble", json={"x0": 1, "x1": 2})
         assert r.status_code == 200, r.text
         assert r.json() == {"y": [1, 1]}
+
+
+@pytest.mark.timeout(240)
+def test_canary_autoupdate_example_readme_flow(tmp_path):
+    """Canary + auto-update example: registering a SECOND model version
+    materializes /2 on sync and the prefix canary routes to both."""
+    pytest.importorskip("sklearn")
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    run_py([os.path.join(ROOT, "examples/sklearn/train_model.py")],
+           cwd=str(tmp_path))
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "canary ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "auto-update",
+            "--engine", "sklearn", "--endpoint", "test_model_sklearn_auto",
+            "--preprocess", "examples/sklearn/preprocess.py", "--name",
+            "train sklearn model", "--project", "serving examples",
+            "--max-versions", "2"])
+    run_py(["-m", "clearml_serving_amd", "model", "canary", "--endpoint",
+            "test_model_sklearn_canary", "--weights", "0.9", "0.1",
+            "--input-endpoint-prefix", "test_model_sklearn_auto"])
+    # second registration = new model version
+    run_py([os.path.join(ROOT, "examples/sklearn/train_model.py")],
+           cwd=str(tmp_path))
+
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        # force one sync (the daemon would do this within a poll interval)
+        proc = app.state.processor
+        proc.deserialize()
+        proc._update_monitored_models()
+        proc._update_canary_lookup()
+        eps = proc.get_synced_endpoints()
+        assert "test_model_sklearn_auto/1" in eps
+        assert "test_model_sklearn_auto/2" in eps
+        for _ in range(10):
+            r = client.post("/serve/test_model_sklearn_canary",
+                            json={"x0": 1, "x1": 2})
+            assert r.status_code == 200, r.text
+            assert r.json() == {"y": [1]}
