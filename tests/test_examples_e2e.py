@@ -165,3 +165,57 @@ def test_canary_autoupdate_example_readme_flow(tmp_path):
                             json={"x0": 1, "x1": 2})
             assert r.status_code == 200, r.text
             assert r.json() == {"y": [1]}
+
+
+@pytest.mark.timeout(240)
+def test_llm_example_readme_flow(tmp_path):
+    """LLM example flow with a scaled-down model card: upload the card via
+    the CLI, add the llm endpoint, chat through the OpenAI route."""
+    import json as _json
+
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+
+    LlmPreprocessRequest._engine_singleton = None
+    LlmPreprocessRequest._engines = {}
+    LlmPreprocessRequest._engine_refs = {}
+
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    card = tmp_path / "llama_card.json"
+    card.write_text(_json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 128, "device": "cpu"}))
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "llm ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "upload", "--name",
+            "llama card", "--project", "examples", "--path", str(card)])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine", "llm",
+            "--endpoint", "test_llm", "--name", "llama card", "--project",
+            "examples"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    try:
+        with TestClient(app) as client:
+            r = client.post("/serve/openai/v1/chat/completions", json={
+                "model": "test_llm", "max_tokens": 4, "temperature": 0.0,
+                "ignore_eos": True,
+                "messages": [{"role": "user", "content": "hi"}]})
+            assert r.status_code == 200, r.text
+            out = r.json()
+            assert out["choices"][0]["finish_reason"] == "length"
+            assert out["usage"]["completion_tokens"] == 4
+    finally:
+        LlmPreprocessRequest._engine_singleton = None
+        LlmPreprocessRequest._engines = {}
+        LlmPreprocessRequest._engine_refs = {}
