@@ -84,3 +84,65 @@ def test_batcher_stages_collector():
     assert abs(fams["serving_batch_occupancy"].samples[0].value - 0.75) < 1e-9
     assert fams["serving_queue_wait_ms"].samples[0].value == 20.0
     assert fams["serving_gpu_wait_ms"].samples[0].value == 12.0
+
+
+def test_kafka_forwarder_with_fake_kafka(monkeypatch):
+    """Kafka topology parity exercised with a stub kafka module: payloads
+    are JSON batches on the reference topic; oversized batches split
+    recursively in half (reference: model_request_processor.py:1097-1102)."""
+    import json
+    import sys
+    import types
+
+    sent = []
+
+    class _Future:
+        def get(self, timeout=None):
+            return None
+
+    class FakeProducer:
+        max_bytes = 10_000
+
+        def __init__(self, **kwargs):
+            self.kwargs = kwargs
+
+        def send(self, topic, payload):
+            if len(payload) > self.max_bytes:
+                raise fake_errors.MessageSizeTooLargeError()
+            sent.append((topic, payload))
+            return _Future()
+
+    fake_kafka = types.ModuleType("kafka")
+    fake_kafka.KafkaProducer = FakeProducer
+    fake_errors = types.ModuleType("kafka.errors")
+
+    class MessageSizeTooLargeError(Exception):
+        pass
+
+    fake_errors.MessageSizeTooLargeError = MessageSizeTooLargeError
+    fake_kafka.errors = fake_errors
+    monkeypatch.setitem(sys.modules, "kafka", fake_kafka)
+    monkeypatch.setitem(sys.modules, "kafka.errors", fake_errors)
+
+    from clearml_serving_amd.statistics.kafka_forwarder import (
+        KafkaStatsForwarder,
+    )
+
+    fwd = KafkaStatsForwarder("kafka:9092")
+    small = [{"_url": "ep", "_latency": 0.01, "_count": 1}] * 3
+    fwd(small)
+    assert len(sent) == 1
+    topic, payload = sent[0]
+    assert topic == "clearml_inference_stats"
+    assert json.loads(payload) == small
+
+    # oversized batch splits recursively until each chunk fits
+    sent.clear()
+    big = [{"_url": "ep", "pad": "x" * 200, "i": i} for i in range(100)]
+    fwd(big)
+    assert len(sent) >= 2
+    recovered = []
+    for _, p in sent:
+        assert len(p) <= FakeProducer.max_bytes
+        recovered.extend(json.loads(p))
+    assert recovered == big
