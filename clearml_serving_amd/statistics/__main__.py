@@ -38,21 +38,34 @@ def main():
     print("statistics service: kafka={} prometheus=:{}".format(
         kafka_server, port))
 
-    from kafka import KafkaConsumer
-
     while True:
         try:
-            consumer = KafkaConsumer(
-                "clearml_inference_stats", bootstrap_servers=kafka_server)
-            for msg in consumer:
-                try:
-                    registry.report_batch(json.loads(msg.value))
-                except Exception as ex:
-                    print("bad stats batch: {}".format(ex))
+            consume(registry, kafka_server)
         except Exception as ex:
             # reference retries broker connection forever (:233-240)
             print("kafka consumer error ({}), retrying in 30s".format(ex))
             time.sleep(30)
+
+
+def consume(registry: StatsRegistry, kafka_server: str,
+            max_batches: int = None) -> int:
+    """One consumer pass: report every batch from the topic into the
+    registry. Factored out of main() so the path is unit-testable with a
+    stub kafka module; returns batches reported."""
+    from kafka import KafkaConsumer
+
+    consumer = KafkaConsumer(
+        "clearml_inference_stats", bootstrap_servers=kafka_server)
+    n = 0
+    for msg in consumer:
+        try:
+            registry.report_batch(json.loads(msg.value))
+            n += 1
+        except Exception as ex:
+            print("bad stats batch: {}".format(ex))
+        if max_batches is not None and n >= max_batches:
+            break
+    return n
 
 
 if __name__ == "__main__":

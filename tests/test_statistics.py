@@ -146,3 +146,29 @@ def test_kafka_forwarder_with_fake_kafka(monkeypatch):
         assert len(p) <= FakeProducer.max_bytes
         recovered.extend(json.loads(p))
     assert recovered == big
+
+
+def test_standalone_service_consume_loop(monkeypatch, processor):
+    """statistics/__main__.py consume(): Kafka batches land as Prometheus
+    metrics through the same registry the in-process sink uses."""
+    import json
+    import sys
+    import types
+
+    msgs = [types.SimpleNamespace(value=json.dumps(
+        [{"_url": "svc_ep", "_latency": 0.02, "_count": 2}]).encode())]
+
+    fake_kafka = types.ModuleType("kafka")
+    fake_kafka.KafkaConsumer = lambda *a, **k: iter(msgs)
+    monkeypatch.setitem(sys.modules, "kafka", fake_kafka)
+
+    from prometheus_client import CollectorRegistry
+
+    from clearml_serving_amd.statistics.__main__ import consume
+    from clearml_serving_amd.statistics.collector import StatsRegistry
+
+    reg = CollectorRegistry()
+    sr = StatsRegistry(processor=processor, registry=reg)
+    n = consume(sr, "kafka:9092", max_batches=1)
+    assert n == 1
+    assert reg.get_sample_value("svc_ep__count_total") == 2.0
