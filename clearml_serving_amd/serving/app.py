@@ -70,8 +70,18 @@ def create_app(
 
     state = {"processor": processor}
 
-    @app.on_event("startup")
-    async def startup():
+    import contextlib
+
+    @contextlib.asynccontextmanager
+    async def lifespan(_app):
+        await _startup()
+        yield
+        if state["processor"] is not None:
+            state["processor"].stop()
+
+    app.router.lifespan_context = lifespan
+
+    async def _startup():
         if state["processor"] is None:
             store = ServingStore(store_root)
             sid = session_id or os.environ.get("CLEARML_SERVING_TASK_ID")
@@ -94,11 +104,6 @@ def create_app(
         except Exception as ex:
             proc._report_text("statistics sink unavailable: {}".format(ex))
         app.state.processor = proc
-
-    @app.on_event("shutdown")
-    async def shutdown():
-        if state["processor"] is not None:
-            state["processor"].stop()
 
     async def process_with_exceptions(
         base_url: str, version: Optional[str], request_body, serve_type: str
