@@ -62,6 +62,9 @@ def func_config_service(args):
     if args.metric_log_freq is not None:
         print("Configuring serving service [{}] metric log freq: {}".format(
             processor.get_id(), args.metric_log_freq))
+    if getattr(args, "triton_grpc_server", None):
+        print("Note: --triton-grpc-server ignored -- DL models are served "
+              "in-process by the native HIP engine (no Triton sidecar)")
     processor.configure(
         external_serving_base_url=args.base_serving_url,
         external_kafka_service_server=args.kafka_metric_server,
@@ -291,6 +294,11 @@ def cli():
     parser_config = subparsers.add_parser("config", help="Configure the serving session")
     parser_config.add_argument("--base-serving-url", type=str)
     parser_config.add_argument("--kafka-metric-server", type=str)
+    # accepted for reference-script portability: there is no Triton sidecar
+    # in this stack (the native HIP engine serves DL models in-process)
+    parser_config.add_argument("--triton-grpc-server", type=str,
+                               help="ignored (no Triton sidecar; kept for "
+                                    "reference CLI compatibility)")
     parser_config.add_argument("--metric-log-freq", type=float)
     parser_config.set_defaults(func=func_config_service)
 
