@@ -1,4 +1,6 @@
 // Rotary position embedding, in place, NeoX pairing (i, i + D/2).
+// LLM-engine hot op (the reference delegates rope to vLLM internals,
+// preprocess_service.py:619-1095).
 // q: [T, H, D], k: [T, Hkv, D], positions: int32 [T].
 #include "common.h"
 
