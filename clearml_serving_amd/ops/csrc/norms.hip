@@ -1,6 +1,6 @@
 // Fused normalization kernels: LayerNorm(+residual) and RMSNorm(+residual).
 // Part of the in-process kernel tier that replaces the reference's Triton
-// delegation (SURVEY.md Â§2.6 'LayerNorm/Softmax' row; the reference has no
+// delegation (SURVEY.md §2.6 'LayerNorm/Softmax' row; the reference has no
 // kernels of its own -- Triton/vLLM supply them).
 //
 // One workgroup per row; 16-byte vector loads (8 bf16 / 4 fp32 per lane per

@@ -1,5 +1,5 @@
 // Row softmax over the last dim (numerically stable, one workgroup per row).
-// SURVEY.md Â§2.6 kernel-library row ('LayerNorm/Softmax').
+// SURVEY.md §2.6 kernel-library row ('LayerNorm/Softmax').
 // Used standalone (parity op); attention uses its own fused online softmax.
 #include "common.h"
 
