@@ -273,3 +273,30 @@ def test_batcher_worker_restarts_after_crash():
 
     out, out2 = run(main())
     assert torch.equal(out, out2)
+
+
+def test_hip_engine_honors_triton_style_batch_config(processor, store,
+                                                     tmp_path):
+    """The reference passes Triton dynamic_batching via aux config
+    (preferred_batch_size, max_batch_size, max_queue_delay -- huggingface
+    example readme:113); the hip engine maps the same keys onto the
+    DynamicBatcher."""
+    import json
+
+    card = tmp_path / "card.json"
+    card.write_text(json.dumps({"arch": "resnet50"}))
+    rec = store.register_model(name="r50", project="p", path=str(card))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="batchcfg", model_id=rec.model_id,
+        input_size=[3, 32, 32], input_type="float32",
+        auxiliary_cfg={"preferred_batch_size": [1, 2, 4, 8],
+                       "max_batch_size": 8, "max_queue_delay_us": 1234,
+                       "use_graphs": False, "device": "cpu"}))
+    out = run(processor.process_request(
+        "batchcfg", "", np.zeros((3, 32, 32), dtype=np.float32)))
+    assert out is not None
+    engine = processor._engine_processor_lookup["batchcfg"]
+    b = engine._batcher
+    assert b.max_batch_size == 8
+    assert b.buckets == [1, 2, 4, 8]
+    assert abs(b.max_queue_delay_s - 0.001234) < 1e-9
