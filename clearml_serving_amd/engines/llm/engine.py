@@ -223,7 +223,8 @@ class LlmEngine:
         self._decode_buckets.append(self.cfg.max_num_seqs)
         self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
                       "prefill_batches": 0, "decode_batches": 0,
-                      "preemptions": 0, "aborts": 0}
+                      "preemptions": 0, "aborts": 0,
+                      "graph_captures": 0, "graph_replays": 0}
 
     # ------------------------------------------------------------------ #
     def start(self) -> None:
@@ -775,6 +776,7 @@ class LlmEngine:
                 static["logits"] = fwd()
         entry = {"graph": graph, **static}
         self._decode_graphs[bucket] = entry
+        self.stats["graph_captures"] += 1
         return entry
 
     def _exec_decode_graph(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -816,6 +818,7 @@ class LlmEngine:
         for k in ("tokens", "positions", "slots", "seq_lens", "block_table"):
             entry[k].copy_(pin[k], non_blocking=True)
         entry["graph"].replay()
+        self.stats["graph_replays"] += 1
         return entry["logits"][:b]
 
     # ------------------------------------------------------------------ #
