@@ -37,6 +37,9 @@ def main():
                     default=int(os.environ.get("CLEARML_SERVING_PORT", 8080)))
     ap.add_argument("--session-id", type=str,
                     default=os.environ.get("CLEARML_SERVING_TASK_ID"))
+    ap.add_argument("--smoke", action="store_true",
+                    help="serve ONE in-process chat completion and exit "
+                         "(launcher self-test; no HTTP server)")
     args = ap.parse_args()
 
     local_rank = tp.init_from_env()
@@ -57,6 +60,21 @@ def main():
 
     if rank == 0:
         processor._engine_processor_lookup[url] = adapter
+        if args.smoke:
+            import asyncio
+
+            async def one():
+                return await processor.process_request(
+                    base_url=url, version=None, serve_type="v1/chat/completions",
+                    request_body={"messages": [{"role": "user",
+                                                "content": "smoke"}],
+                                  "max_tokens": 4, "temperature": 0.0,
+                                  "ignore_eos": True})
+            out = asyncio.new_event_loop().run_until_complete(one())
+            assert out["usage"]["completion_tokens"] == 4, out
+            engine.tp_shutdown()
+            print("TP-SERVE-OK", flush=True)
+            return
         processor.launch(poll_frequency_sec=300)
         import uvicorn
 
