@@ -300,3 +300,16 @@ def test_hip_engine_honors_triton_style_batch_config(processor, store,
     assert b.max_batch_size == 8
     assert b.buckets == [1, 2, 4, 8]
     assert abs(b.max_queue_delay_s - 0.001234) < 1e-9
+
+
+def test_pick_device_most_free_hbm(monkeypatch):
+    """Free-HBM placement: multi-model sessions spread across GPUs."""
+    from clearml_serving_amd.engines import torch_engine as te
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "device_count", lambda: 4)
+    free = {0: 10, 1: 40, 2: 25, 3: 40}
+    monkeypatch.setattr(torch.cuda, "mem_get_info",
+                        lambda i: (free[i], 100))
+    assert te._pick_device({}) == torch.device("cuda", 1)  # first max wins
+    assert te._pick_device({"gpu": 3}) == torch.device("cuda", 3)
