@@ -482,3 +482,29 @@ def test_chaos_reloads_under_load(processor, store, tmp_path):
     assert ok >= 50 and ok + miss >= 175
     # state is consistent at the end
     processor.deserialize()
+
+
+def test_user_unload_called_on_engine_flush(processor, store, tmp_path):
+    """Removing an endpoint lets the user Preprocess.unload() run when the
+    cached engine instance is flushed (reference parity:
+    preprocess_service.py:100-111)."""
+    import gc
+
+    flag = tmp_path / "unloaded"
+    code = tmp_path / "pp.py"
+    code.write_text(
+        "class Preprocess(object):\n"
+        "    def process(self, d, s, c=None):\n"
+        "        return d\n"
+        "    def unload(self):\n"
+        "        open(r'%s', 'w').write('1')\n" % str(flag))
+    processor.add_endpoint(ModelEndpoint(engine_type="custom",
+                                         serving_url="unl"),
+                           preprocess_code=str(code))
+    processor.serialize()
+    assert run(processor.process_request("unl", "", {"v": 1})) == {"v": 1}
+    processor.remove_endpoint("unl")
+    processor.serialize()
+    processor.deserialize()
+    gc.collect()
+    assert flag.exists()
