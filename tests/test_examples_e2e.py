@@ -219,3 +219,49 @@ def test_llm_example_readme_flow(tmp_path):
         LlmPreprocessRequest._engine_singleton = None
         LlmPreprocessRequest._engines = {}
         LlmPreprocessRequest._engine_refs = {}
+
+
+@pytest.mark.timeout(240)
+def test_bert_example_readme_flow(tmp_path):
+    """BERT example flow (the reference's huggingface/Triton recipe) on
+    CPU fp32: card upload, hip-engine endpoint with I/O spec + aux batch
+    config via CLI, pre-tokenized payload through HTTP."""
+    import json as _json
+
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=150)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    card = tmp_path / "bert_card.json"
+    card.write_text(_json.dumps({"arch": "bert-base", "num_labels": 2,
+                                 "dtype": "float32", "vocab_size": 30522}))
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "bert ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "upload", "--name",
+            "bert base card", "--project", "examples", "--path", str(card)])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine", "hip",
+            "--endpoint", "transformer_model", "--name", "bert base card",
+            "--project", "examples", "--preprocess",
+            "examples/bert/preprocess.py",
+            "--input-size", "[8]", "[8]", "--input-type", "int64", "int32",
+            "--input-name", "input_ids", "attention_mask",
+            "--aux-config", "max_batch_size=8", "max_queue_delay_us=1000",
+            "use_graphs=false"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    payload = _json.load(open(os.path.join(ROOT,
+                                           "examples/bert/example_payload.json")))
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/transformer_model", json=payload)
+        assert r.status_code == 200, r.text
+        out = r.json()
+        assert out["label"] in (0, 1) and len(out["logits"]) == 2
