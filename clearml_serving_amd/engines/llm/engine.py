@@ -1030,6 +1030,11 @@ class LlmEngine:
         tokens: List[int] = []
         reason = None
         async for item in self.generate(prompt, params):
+            if item.get("error"):
+                # engine-side failure: surface it (the route maps to 500)
+                # instead of returning truncated text as if successful
+                raise RuntimeError("generation failed: {}".format(
+                    item["error"]))
             tokens.extend(item.get("token_ids", []))
             reason = item.get("finish_reason") or reason
         return self.tokenizer.decode(tokens), reason, len(tokens), len(ids)
@@ -1040,6 +1045,11 @@ class LlmEngine:
 
         async def gen():
             async for item in self.generate(prompt, params):
+                if item.get("error"):
+                    yield "data: {}\n\n".format(json.dumps(
+                        {"error": {"message": str(item["error"]),
+                                   "type": "engine_error"}}))
+                    break
                 if chat:
                     delta = {"content": item.get("text", "")}
                     choice = {"index": 0, "delta": delta,

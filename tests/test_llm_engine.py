@@ -526,3 +526,37 @@ def test_config_from_model_folder(tmp_path):
     assert cfg.max_model_len == 77
     assert cfg.block_size == 8          # aux overrides layer on top
     assert cfg.weights == str(d / "model.safetensors")
+
+
+def test_engine_error_surfaces_in_responses():
+    """A step crash must surface as an error to BOTH response styles --
+    a raised RuntimeError for non-streaming, an error SSE chunk for
+    streaming -- never as silently truncated output."""
+    eng = tiny_engine()
+
+    def boom():
+        raise ValueError("injected kernel failure")
+
+    eng.step = boom
+
+    async def collect():
+        with pytest.raises(RuntimeError, match="injected kernel failure"):
+            await eng._collect("x", SamplingParams(max_tokens=4,
+                                                   ignore_eos=True))
+
+    run(collect())
+
+    eng2 = tiny_engine()
+    eng2.step = boom
+
+    async def stream():
+        resp = eng2._sse_stream("x", SamplingParams(max_tokens=4,
+                                                    ignore_eos=True),
+                                "rid", "m", chat=True)
+        chunks = []
+        async for c in resp.body_iterator:
+            chunks.append(c)
+        return chunks
+
+    chunks = run(stream())
+    assert any("engine_error" in c for c in chunks), chunks
