@@ -352,6 +352,9 @@ class LlmEngine:
         params = SamplingParams.from_request(body)
         tokens: List[int] = []
         async for item in self.generate(prompt, params):
+            if item.get("error"):
+                raise RuntimeError("generation failed: {}".format(
+                    item["error"]))
             tokens.extend(item.get("token_ids", []))
         return {
             "text": self.tokenizer.decode(tokens),
