@@ -248,7 +248,9 @@ def test_batcher_respects_queue_delay():
         return _t.monotonic() - t0
 
     elapsed = run(main())
-    assert 0.02 <= elapsed < 0.5  # waited for the window, not forever
+    # generous upper bound: loaded CI boxes schedule slowly; the point
+    # is ONLY that the lone request didn't wait forever
+    assert 0.02 <= elapsed < 3.0
 
 
 def test_batcher_worker_restarts_after_crash():
