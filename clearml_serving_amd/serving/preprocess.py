@@ -52,6 +52,10 @@ class BasePreprocessRequest:
     is_preprocess_async = False
     is_process_async = False
     is_postprocess_async = False
+    # sync process() that may BLOCK (user code, threaded pipeline fan-out):
+    # the dispatcher runs it in a worker thread so the event loop keeps
+    # serving (the sync send_request path depends on a live loop)
+    run_process_off_loop = False
 
     def __init__(self, model_endpoint: ModelEndpoint, task=None):
         """``task`` kept for signature parity; here it is the ServingStore."""
@@ -280,7 +284,12 @@ class LightGBMPreprocessRequest(BasePreprocessRequest):
 @BasePreprocessRequest.register_engine("custom")
 class CustomPreprocessRequest(BasePreprocessRequest):
     """Fully user-managed model: Preprocess.process() runs synchronously
-    (reference: preprocess_service.py:504-517)."""
+    (reference: preprocess_service.py:504-517) -- in a worker thread, so user
+    code may block (e.g. the reference's ThreadPoolExecutor pipeline,
+    examples/pipeline/preprocess.py:18-32, whose sync send_request needs the
+    serving loop to stay live)."""
+
+    run_process_off_loop = True
 
     def process(self, data, state: dict, collect_custom_statistics_fn=None):
         if self._preprocess is not None and hasattr(self._preprocess, "process"):

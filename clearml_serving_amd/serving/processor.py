@@ -21,6 +21,7 @@ local ``ServingStore``. Semantics preserved:
 
 import asyncio
 import itertools
+import logging
 import os
 import random
 import socket
@@ -39,6 +40,8 @@ from ..schemas import (
 )
 from ..store import ServingStore
 from .preprocess import BasePreprocessRequest
+
+logger = logging.getLogger("clearml_serving_amd.processor")
 
 
 class FastWriteCounter:
@@ -652,6 +655,14 @@ class ModelRequestProcessor:
         if processor._preprocess is None:
             return
 
+        # the loop serving requests: sync send_request (user threads, e.g.
+        # the reference's ThreadPoolExecutor pipeline,
+        # examples/pipeline/preprocess.py:18-32) schedules onto it
+        try:
+            serving_loop = asyncio.get_running_loop()
+        except RuntimeError:
+            serving_loop = None
+
         if processor.is_process_async:
             async def _send(endpoint, version=None, data=None):
                 try:
@@ -659,21 +670,45 @@ class ModelRequestProcessor:
                         base_url=endpoint, version=version, request_body=data
                     )
                 except Exception:
+                    # reference parity: pipeline fan-out failures return None
+                    # (the reference's HTTP hop does the same,
+                    # preprocess_service.py:255-264) -- but log them loudly
+                    logger.exception(
+                        "pipeline send_request to '%s' failed", endpoint)
                     return None
         else:
             def _send(endpoint, version=None, data=None):
+                coro = self.process_request(
+                    base_url=endpoint, version=version, request_body=data)
                 try:
+                    asyncio.get_running_loop()
+                except RuntimeError:
+                    pass  # off-loop (worker thread): the supported sync path
+                else:
+                    # called inline ON the event loop: blocking here would
+                    # deadlock (the loop can't progress the inner request).
+                    # Fail loudly instead of silently returning None.
+                    coro.close()
+                    raise RuntimeError(
+                        "sync send_request called on the serving event loop; "
+                        "call it from a worker thread (ThreadPoolExecutor) or "
+                        "use engine_type=custom_async with "
+                        "'await self.send_request(...)'")
+                try:
+                    if serving_loop is not None and not serving_loop.is_closed():
+                        fut = asyncio.run_coroutine_threadsafe(
+                            coro, serving_loop)
+                        return fut.result()
+                    # no serving loop captured (offline/unit-test use): run
+                    # the coroutine on a private loop in this thread
                     loop = asyncio.new_event_loop()
                     try:
-                        return loop.run_until_complete(
-                            self.process_request(
-                                base_url=endpoint, version=version,
-                                request_body=data,
-                            )
-                        )
+                        return loop.run_until_complete(coro)
                     finally:
                         loop.close()
                 except Exception:
+                    logger.exception(
+                        "pipeline send_request to '%s' failed", endpoint)
                     return None
 
         processor._preprocess.send_request = _send
@@ -712,6 +747,12 @@ class ModelRequestProcessor:
         if serve_type == "process":
             if processor.is_process_async:
                 processed = await processor.process(preprocessed, state, collect_fn)
+            elif processor.run_process_off_loop:
+                # blocking user code (custom engine / threaded pipelines):
+                # keep the event loop live so sync send_request can schedule
+                # nested requests onto it
+                processed = await asyncio.to_thread(
+                    processor.process, preprocessed, state, collect_fn)
             else:
                 processed = processor.process(preprocessed, state, collect_fn)
         else:

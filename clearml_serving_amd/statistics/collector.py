@@ -39,8 +39,9 @@ RESERVED = ("_latency", "_count", "_url")
 
 
 def _prom_name(endpoint: str, variable: str) -> str:
-    # "{url}:{var}" with non-alphanumerics -> "_" (metrics.py:322-324)
-    return re.sub(r"[^a-zA-Z0-9_]", "_", "{}:{}".format(endpoint, variable))
+    # "{url}:{var}" -- the reference's regex [^a-zA-Z0-9_:] KEEPS colons
+    # (metrics.py:322-324), so reference-authored Grafana dashboards match
+    return re.sub(r"[^a-zA-Z0-9_:]", "_", "{}:{}".format(endpoint, variable))
 
 
 class StatsRegistry:

@@ -8,8 +8,8 @@ from clearml_serving_amd.statistics.collector import StatsRegistry, _prom_name
 
 def test_prom_name_folding():
     # reference naming: {url}:{var} with non-alnum -> "_"
-    assert _prom_name("model/1", "_latency") == "model_1__latency"
-    assert _prom_name("a-b", "x.y") == "a_b_x_y"
+    assert _prom_name("model/1", "_latency") == "model_1:_latency"
+    assert _prom_name("a-b", "x.y") == "a_b:x_y"
 
 
 class FakeProcessor:
@@ -36,15 +36,15 @@ def test_all_metric_types():
          "detect": "cat", "gauge_v": 42.0, "cnt": 3},
         {"_url": "ep/1", "_latency": 0.2, "_count": 2, "detect": "dog"},
     ])
-    assert reg.get_sample_value("ep_1__count_total") == 4.0
-    assert reg.get_sample_value("ep_1__latency_count") == 2.0
+    assert reg.get_sample_value("ep_1:_count_total") == 4.0
+    assert reg.get_sample_value("ep_1:_latency_count") == 2.0
     # scalar histogram with custom buckets
-    assert reg.get_sample_value("ep_1_x1_bucket", {"le": "2.0"}) == 1.0
+    assert reg.get_sample_value("ep_1:x1_bucket", {"le": "2.0"}) == 1.0
     # enum counters per value
-    assert reg.get_sample_value("ep_1_detect_total", {"value": "cat"}) == 1.0
-    assert reg.get_sample_value("ep_1_detect_total", {"value": "dog"}) == 1.0
-    assert reg.get_sample_value("ep_1_gauge_v") == 42.0
-    assert reg.get_sample_value("ep_1_cnt_total") == 3.0
+    assert reg.get_sample_value("ep_1:detect_total", {"value": "cat"}) == 1.0
+    assert reg.get_sample_value("ep_1:detect_total", {"value": "dog"}) == 1.0
+    assert reg.get_sample_value("ep_1:gauge_v") == 42.0
+    assert reg.get_sample_value("ep_1:cnt_total") == 3.0
 
 
 def test_prefix_metric_config_applies():
@@ -54,15 +54,15 @@ def test_prefix_metric_config_applies():
     sr = StatsRegistry(processor=FakeProcessor(cfg), registry=reg)
     sr.report_batch([{"_url": "models/alpha/1", "score": 0.7}])
     assert reg.get_sample_value(
-        "models_alpha_1_score_bucket", {"le": "1.0"}) == 1.0
+        "models_alpha_1:score_bucket", {"le": "1.0"}) == 1.0
 
 
 def test_unconfigured_endpoint_gets_reserved_metrics():
     reg = CollectorRegistry()
     sr = StatsRegistry(processor=FakeProcessor({}), registry=reg)
     sr.report_batch([{"_url": "ghost", "_latency": 0.01, "_count": 1}])
-    assert reg.get_sample_value("ghost__count_total") == 1.0
-    assert reg.get_sample_value("ghost__latency_count") == 1.0
+    assert reg.get_sample_value("ghost:_count_total") == 1.0
+    assert reg.get_sample_value("ghost:_latency_count") == 1.0
 
 
 def test_batcher_stages_collector():
@@ -171,4 +171,4 @@ def test_standalone_service_consume_loop(monkeypatch, processor):
     sr = StatsRegistry(processor=processor, registry=reg)
     n = consume(sr, "kafka:9092", max_batches=1)
     assert n == 1
-    assert reg.get_sample_value("svc_ep__count_total") == 2.0
+    assert reg.get_sample_value("svc_ep:_count_total") == 2.0
