@@ -35,10 +35,21 @@ class LlmPreprocessRequest(BasePreprocessRequest):
         # preprocess override vllm_model_config, examples/vllm/preprocess.py)
         if isinstance(self._model, dict):
             aux.update(self._model)
-        key = model_endpoint.model_id or "__default__"
+        model_path = self._get_local_model_file()
+        cfg = LlmEngineConfig.from_aux(model_path, aux)
+        if model_endpoint.model_id:
+            key = model_endpoint.model_id
+        else:
+            # aux-config-only endpoints: key on the RESOLVED engine config so
+            # two endpoints with different presets/overrides never collide on
+            # one cached engine (they share it only when the config matches)
+            import hashlib
+            import json as _json
+
+            key = "cfg:" + hashlib.sha256(_json.dumps(
+                cfg.__dict__, sort_keys=True, default=str
+            ).encode()).hexdigest()[:16]
         if key not in LlmPreprocessRequest._engines:
-            model_path = self._get_local_model_file()
-            cfg = LlmEngineConfig.from_aux(model_path, aux)
             engine = LlmEngine(cfg)
             engine.start()
             LlmPreprocessRequest._engines[key] = engine

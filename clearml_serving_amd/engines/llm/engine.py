@@ -22,6 +22,7 @@ import json
 import os
 import time
 import uuid
+from collections import deque
 from dataclasses import dataclass, field
 from typing import Any, AsyncGenerator, Dict, List, Optional
 
@@ -64,6 +65,12 @@ class HfTokenizer:
                 eos = tid
                 break
         self.eos_id = eos if eos is not None else 0
+        # llama-3 checkpoints carry header-id special tokens: chat requests
+        # then format with the real llama-3 template instead of the
+        # synthetic fallback (_chat_prompt)
+        self.is_llama3 = (
+            self._tok.token_to_id("<|start_header_id|>") is not None
+            and self._tok.token_to_id("<|eot_id|>") is not None)
 
     def encode(self, text: str) -> List[int]:
         return self._tok.encode(text).ids
@@ -80,15 +87,52 @@ class SamplingParams:
     top_p: float = 1.0
     max_tokens: int = 128
     stop_token_ids: List[int] = field(default_factory=list)
+    stop: List[str] = field(default_factory=list)
     ignore_eos: bool = False
 
     @classmethod
     def from_request(cls, body: Dict[str, Any], default_max: int = 128):
+        """Validate at add-request time so one malformed request 422s on its
+        own instead of crashing the shared step() for every in-flight
+        sequence (a bad temperature would assert inside the batched
+        sampling kernel)."""
+        try:
+            temperature = float(body.get("temperature", 1.0))
+            top_k = int(body.get("top_k", 0) or 0)
+            top_p = float(body.get("top_p", 1.0))
+            max_tokens = int(body.get("max_tokens", default_max))
+        except (TypeError, ValueError):
+            raise ValueError("sampling parameters must be numeric "
+                             "(temperature/top_k/top_p/max_tokens)")
+        if temperature < 0.0:
+            raise ValueError(
+                "temperature must be >= 0 (0 = greedy), got {}".format(
+                    temperature))
+        if top_k < 0:
+            raise ValueError("top_k must be >= 0 (0 = disabled), got {}"
+                             .format(top_k))
+        if not (0.0 < top_p <= 1.0):
+            raise ValueError("top_p must be in (0, 1], got {}".format(top_p))
+        if max_tokens < 1:
+            raise ValueError("max_tokens must be >= 1, got {}".format(
+                max_tokens))
+        stop = body.get("stop") or []
+        if isinstance(stop, str):
+            stop = [stop]
+        if not all(isinstance(s, str) for s in stop):
+            raise ValueError("'stop' must be a string or list of strings")
+        stop_ids = body.get("stop_token_ids") or []
+        try:
+            stop_ids = [int(t) for t in stop_ids]
+        except (TypeError, ValueError):
+            raise ValueError("'stop_token_ids' must be a list of ints")
         return cls(
-            temperature=float(body.get("temperature", 1.0)),
-            top_k=int(body.get("top_k", 0) or 0),
-            top_p=float(body.get("top_p", 1.0)),
-            max_tokens=int(body.get("max_tokens", default_max)),
+            temperature=temperature,
+            top_k=top_k,
+            top_p=top_p,
+            max_tokens=max_tokens,
+            stop_token_ids=stop_ids,
+            stop=[s for s in stop if s],
             ignore_eos=bool(body.get("ignore_eos", False)),
         )
 
@@ -208,6 +252,9 @@ class LlmEngine:
         self.tokenizer = None
         self.waiting: List[Sequence] = []
         self.running: List[Sequence] = []
+        # aborts cross threads (event loop -> step worker): appended on the
+        # loop, drained at the top of step() -- deque ops are GIL-atomic
+        self._aborted: "deque[Sequence]" = deque()
         self._loop_task: Optional[asyncio.Task] = None
         self._wake: Optional[asyncio.Event] = None
         self._started = False
@@ -320,16 +367,20 @@ class LlmEngine:
         return seq
 
     def abort(self, seq: "Sequence") -> None:
-        """Stop generating for a sequence (client gone): drop it from the
-        queues; its pages free at the end of the current step."""
+        """Stop generating for a sequence (client gone).
+
+        Runs on the event loop while step() runs in a worker thread
+        (asyncio.to_thread), so scheduler state is NOT mutated here: the
+        sequence goes onto a thread-safe queue that step() drains at its
+        top, keeping all waiting/running/allocator mutation on the step
+        thread (a mid-_admit list removal here could make _admit pop and
+        drop a different queued sequence, hanging its client forever)."""
         seq.finished = True
         seq.finish_reason = seq.finish_reason or "abort"
-        if seq in self.waiting:
-            self.waiting.remove(seq)
-            if seq.blocks:
-                self.allocator.free(seq.blocks)
-                seq.blocks = []
+        self._aborted.append(seq)
         self.stats["aborts"] = self.stats.get("aborts", 0) + 1
+        if self._wake is not None:
+            self._wake.set()  # free the pages promptly even when idle
 
     async def generate(self, prompt: str, params: SamplingParams
                        ) -> AsyncGenerator[Dict[str, Any], None]:
@@ -351,13 +402,16 @@ class LlmEngine:
         prompt = body.get("prompt") or body.get("text") or ""
         params = SamplingParams.from_request(body)
         tokens: List[int] = []
+        reason = None
         async for item in self.generate(prompt, params):
             if item.get("error"):
                 raise RuntimeError("generation failed: {}".format(
                     item["error"]))
             tokens.extend(item.get("token_ids", []))
+            reason = item.get("finish_reason") or reason
         return {
-            "text": self.tokenizer.decode(tokens),
+            "text": self._truncate_at_stop(
+                self.tokenizer.decode(tokens), params, reason),
             "tokens": len(tokens),
         }
 
@@ -428,6 +482,7 @@ class LlmEngine:
         ``prefill_chunk`` prompt tokens per step so decode latency stays
         bounded under long-prompt load), then decode everything else."""
         self.stats["steps"] += 1
+        self._drain_aborts()
         self._admit()
         budget = self.cfg.prefill_chunk
         pending = [s for s in self.running
@@ -458,12 +513,31 @@ class LlmEngine:
                 self.allocator.free(s.blocks)
                 s.blocks = []
 
+    def _drain_aborts(self) -> None:
+        """Apply aborts queued by abort() -- all scheduler-state mutation
+        happens here on the step thread."""
+        while True:
+            try:
+                seq = self._aborted.popleft()
+            except IndexError:
+                return
+            if seq in self.waiting:
+                self.waiting.remove(seq)
+            if seq in self.running:
+                self.running.remove(seq)
+            if seq.blocks:
+                self.allocator.free(seq.blocks)
+                seq.blocks = []
+
     def _admit(self) -> List[Sequence]:
         admitted: List[Sequence] = []
         tokens = 0
         bs = self.cfg.block_size
         while self.waiting and len(self.running) < self.cfg.max_num_seqs:
             seq = self.waiting[0]
+            if seq.finished:  # aborted after the drain at step() top
+                self.waiting.pop(0)
+                continue
             need = len(seq.prompt_ids)
             if admitted and tokens + need > self.cfg.max_prefill_tokens:
                 break
@@ -862,6 +936,27 @@ class LlmEngine:
         if getattr(self, "tp_size", 1) > 1 and self.tp_rank == 0:
             self._tp_broadcast({"mode": "stop"})
 
+    def _hit_stop_string(self, s: Sequence) -> bool:
+        """OpenAI 'stop' strings: decode a bounded tail window of the output
+        after each token and look for any stop sequence (the reference's
+        vLLM path matches decoded text the same way). The final text is
+        truncated at the stop by _truncate_at_stop()."""
+        window = min(len(s.output_ids), 16)
+        tail = self.tokenizer.decode(s.output_ids[-window:])
+        return any(st in tail for st in s.params.stop)
+
+    @staticmethod
+    def _truncate_at_stop(text: str, params: SamplingParams,
+                          reason: Optional[str]) -> str:
+        if reason != "stop" or not params.stop:
+            return text
+        cut = len(text)
+        for st in params.stop:
+            idx = text.find(st)
+            if idx >= 0:
+                cut = min(cut, idx)
+        return text[:cut]
+
     def _sample_and_emit(self, seqs: List[Sequence], logits: torch.Tensor) -> None:
         # group rows by identical sampling params for batched kernels
         groups: Dict[tuple, List[int]] = {}
@@ -885,8 +980,10 @@ class LlmEngine:
             self.stats["generated_tokens"] += 1
             finished = False
             reason = None
-            if not s.params.ignore_eos and (
-                    tok == eos or tok in s.params.stop_token_ids):
+            if (not s.params.ignore_eos and tok == eos) \
+                    or tok in s.params.stop_token_ids:
+                finished, reason = True, "stop"
+            elif s.params.stop and self._hit_stop_string(s):
                 finished, reason = True, "stop"
             elif s.generated >= s.params.max_tokens:
                 finished, reason = True, "length"
@@ -905,6 +1002,17 @@ class LlmEngine:
     # OpenAI-compatible handlers (route /serve/openai/v1/*)
     # ------------------------------------------------------------------ #
     def _chat_prompt(self, messages: List[Dict[str, str]]) -> str:
+        if getattr(self.tokenizer, "is_llama3", False):
+            # real llama-3 chat template (the tokenizer carries the
+            # header-id special tokens, so they encode to their single ids)
+            parts = ["<|begin_of_text|>"]
+            for m in messages:
+                parts.append("<|start_header_id|>{}<|end_header_id|>\n\n{}"
+                             "<|eot_id|>".format(m.get("role", "user"),
+                                                 m.get("content", "")))
+            parts.append("<|start_header_id|>assistant<|end_header_id|>\n\n")
+            return "".join(parts)
+        # synthetic fallback (byte tokenizer / random-init serving)
         parts = []
         for m in messages:
             parts.append("<|{}|>\n{}".format(m.get("role", "user"),
@@ -1040,7 +1148,9 @@ class LlmEngine:
                     item["error"]))
             tokens.extend(item.get("token_ids", []))
             reason = item.get("finish_reason") or reason
-        return self.tokenizer.decode(tokens), reason, len(tokens), len(ids)
+        text = self._truncate_at_stop(
+            self.tokenizer.decode(tokens), params, reason)
+        return text, reason, len(tokens), len(ids)
 
     def _sse_stream(self, prompt: str, params: SamplingParams, rid: str,
                     model_name: str, chat: bool):

@@ -85,6 +85,7 @@ class HipPreprocessRequest(BasePreprocessRequest):
             torch.backends.cudnn.benchmark = True
         self.dtype = _DTYPES.get(str(aux.get("dtype", "bfloat16")).lower(),
                                  torch.bfloat16)
+        self._requested_dtype = self.dtype  # pre-CPU-override, for validation
         if self.device.type == "cpu" and self.dtype is not torch.float32:
             # CPU path (tests / no-GPU dev): fp32 keeps MIOpen-free numerics
             self.dtype = torch.float32
@@ -153,6 +154,15 @@ class HipPreprocessRequest(BasePreprocessRequest):
                     local = p
                     break
         if local.endswith(".json"):
+            if self._requested_dtype is torch.float16:
+                # native model-library transformers run on the in-tree
+                # attention kernels, which are bf16-only (attention.hip);
+                # fail at endpoint construction with a clear message instead
+                # of a TORCH_CHECK crash on the first GPU request
+                raise ValueError(
+                    "dtype=float16 is not supported for model-card endpoints "
+                    "(the native attention kernels are bf16); set "
+                    "dtype=bfloat16, or serve a TorchScript model for fp16")
             from .. import models
 
             model = models.build_model(local, device=str(self.device))

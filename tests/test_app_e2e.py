@@ -87,10 +87,10 @@ def test_stats_flow_to_prometheus(client, processor):
         item = processor._stats_queue.get(timeout=0)
     stats_registry.report_batch(batch)
     # reserved metrics present under reference naming: {url}:{var} -> _
-    val = reg.get_sample_value("test_model_sklearn__count_total")
+    val = reg.get_sample_value("test_model_sklearn:_count_total")
     assert val == 3.0
     lat_count = reg.get_sample_value(
-        "test_model_sklearn__latency_count")
+        "test_model_sklearn:_latency_count")
     assert lat_count == 3.0
 
 

@@ -560,3 +560,168 @@ def test_engine_error_surfaces_in_responses():
 
     chunks = run(stream())
     assert any("engine_error" in c for c in chunks), chunks
+
+
+def test_sampling_params_validation_rejects_bad_values():
+    """Malformed sampling params 422 at request time instead of crashing the
+    shared step() for every in-flight sequence."""
+    for bad in ({"temperature": -1}, {"top_k": -1}, {"top_p": 0.0},
+                {"top_p": 1.5}, {"max_tokens": 0},
+                {"temperature": "hot"}, {"stop": [1, 2]},
+                {"stop_token_ids": ["x"]}):
+        with pytest.raises(ValueError):
+            SamplingParams.from_request(bad)
+    # valid edge values pass
+    p = SamplingParams.from_request(
+        {"temperature": 0, "top_k": 0, "top_p": 1.0, "max_tokens": 1,
+         "stop": "END", "stop_token_ids": [7]})
+    assert p.temperature == 0.0 and p.stop == ["END"] \
+        and p.stop_token_ids == [7]
+
+
+def test_stop_token_ids_finish_generation():
+    eng = tiny_engine()
+
+    async def gen():
+        # greedy output is deterministic: find its 3rd token, then re-run
+        # with that token as a stop id
+        params = SamplingParams(temperature=0.0, max_tokens=8,
+                                ignore_eos=True)
+        toks = []
+        async for item in eng.generate("stop test", params):
+            toks.extend(item["token_ids"])
+        stop_at = toks[2]
+        params2 = SamplingParams(temperature=0.0, max_tokens=8,
+                                 ignore_eos=True, stop_token_ids=[stop_at])
+        out, reason = [], None
+        async for item in eng.generate("stop test", params2):
+            out.extend(item["token_ids"])
+            reason = item.get("finish_reason") or reason
+        return toks, out, reason
+
+    toks, out, reason = run(gen())
+    assert out == toks[:3]
+    assert reason == "stop"
+
+
+def test_stop_strings_finish_and_truncate():
+    eng = tiny_engine()
+
+    async def main():
+        params = SamplingParams(temperature=0.0, max_tokens=10,
+                                ignore_eos=True)
+        full = await eng.generate_simple(
+            {"prompt": "abc", "max_tokens": 10, "temperature": 0,
+             "ignore_eos": True})
+        # pick a substring from the middle of the deterministic output
+        text = full["text"]
+        assert len(text) >= 3
+        stop = text[2]
+        res = await eng.generate_simple(
+            {"prompt": "abc", "max_tokens": 10, "temperature": 0,
+             "ignore_eos": True, "stop": stop})
+        return text, stop, res["text"]
+
+    text, stop, stopped = run(main())
+    assert stop not in stopped
+    assert stopped == text[:text.find(stop)]
+
+
+def test_abort_of_waiting_seq_does_not_drop_neighbors():
+    """Regression (advisor): abort() used to mutate waiting[] from the event
+    loop while _admit() popped it from the step thread, which could admit-
+    then-drop a DIFFERENT queued sequence. Aborts now drain at the top of
+    step() on the step thread."""
+    eng = tiny_engine()
+
+    async def main():
+        paramsA = SamplingParams(temperature=0.0, max_tokens=4,
+                                 ignore_eos=True)
+        paramsB = SamplingParams(temperature=0.0, max_tokens=4,
+                                 ignore_eos=True)
+        # enqueue two, abort the first before the engine runs a step
+        seq_a = await eng.add_request([1, 2, 3], paramsA)
+        seq_b = await eng.add_request([4, 5, 6], paramsB)
+        eng.abort(seq_a)
+        out_b = []
+        while True:
+            item = await asyncio.wait_for(seq_b.stream.get(), timeout=10)
+            out_b.extend(item["token_ids"])
+            if item.get("finished"):
+                break
+        return seq_a, out_b
+
+    seq_a, out_b = run(main())
+    assert len(out_b) == 4          # b generated fully
+    assert seq_a.finished and not seq_a.blocks
+    assert not eng.waiting and not eng.running
+    assert eng.allocator.available == eng.allocator.num_blocks
+
+
+def test_llama3_chat_template_detected(tmp_path):
+    """A tokenizer carrying llama-3 header-id tokens switches chat formatting
+    to the real llama-3 template (VERDICT: _chat_prompt was synthetic)."""
+    eng = tiny_engine()
+
+    class FakeL3Tok:
+        is_llama3 = True
+        eos_id = 0
+        vocab_size = 300
+
+        def encode(self, t):
+            return [1]
+
+        def decode(self, ids):
+            return ""
+
+    eng.tokenizer = FakeL3Tok()
+    p = eng._chat_prompt([
+        {"role": "system", "content": "be brief"},
+        {"role": "user", "content": "hi"},
+    ])
+    assert p.startswith("<|begin_of_text|>")
+    assert "<|start_header_id|>system<|end_header_id|>\n\nbe brief<|eot_id|>" in p
+    assert p.endswith("<|start_header_id|>assistant<|end_header_id|>\n\n")
+
+    eng.tokenizer = SimpleTokenizer()
+    p2 = eng._chat_prompt([{"role": "user", "content": "hi"}])
+    assert p2.endswith("<|assistant|>\n")
+
+
+def test_engine_cache_key_distinguishes_aux_only_endpoints(tmp_path):
+    """Two aux-config-only LLM endpoints (no model_id) with different configs
+    must NOT share one cached engine (advisor: '__default__' collision)."""
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+    from clearml_serving_amd.schemas import ModelEndpoint
+
+    saved_engines = dict(LlmPreprocessRequest._engines)
+    saved_refs = dict(LlmPreprocessRequest._engine_refs)
+    LlmPreprocessRequest._engines.clear()
+    LlmPreprocessRequest._engine_refs.clear()
+    try:
+        ep1 = ModelEndpoint(
+            engine_type="llm", serving_url="llm_a",
+            auxiliary_cfg={"preset": "llama-tiny", "num_kv_blocks": 64,
+                           "device": "cpu", "max_model_len": 128})
+        ep2 = ModelEndpoint(
+            engine_type="llm", serving_url="llm_b",
+            auxiliary_cfg={"preset": "llama-tiny", "num_kv_blocks": 32,
+                           "device": "cpu", "max_model_len": 64})
+        r1 = LlmPreprocessRequest(ep1)
+        r2 = LlmPreprocessRequest(ep2)
+        assert r1._engine is not r2._engine
+        assert r1._engine.cfg.num_kv_blocks == 64
+        assert r2._engine.cfg.num_kv_blocks == 32
+        # identical config DOES share the engine
+        ep3 = ModelEndpoint(
+            engine_type="llm", serving_url="llm_c",
+            auxiliary_cfg={"preset": "llama-tiny", "num_kv_blocks": 64,
+                           "device": "cpu", "max_model_len": 128})
+        r3 = LlmPreprocessRequest(ep3)
+        assert r3._engine is r1._engine
+        r1.shutdown(); r2.shutdown(); r3.shutdown()
+    finally:
+        LlmPreprocessRequest._engines.clear()
+        LlmPreprocessRequest._engines.update(saved_engines)
+        LlmPreprocessRequest._engine_refs.clear()
+        LlmPreprocessRequest._engine_refs.update(saved_refs)

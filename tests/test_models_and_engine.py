@@ -315,3 +315,38 @@ def test_pick_device_most_free_hbm(monkeypatch):
                         lambda i: (free[i], 100))
     assert te._pick_device({}) == torch.device("cuda", 1)  # first max wins
     assert te._pick_device({"gpu": 3}) == torch.device("cuda", 3)
+
+
+def test_fp16_model_card_endpoint_rejected(processor, store,
+                                           model_card_registered):
+    """dtype=float16 on a model-card endpoint fails at construction with a
+    clear error (native attention kernels are bf16-only) instead of a
+    TORCH_CHECK crash on the first GPU request."""
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="bert_fp16",
+        model_id=model_card_registered.model_id,
+        auxiliary_cfg={"dtype": "float16", "use_graphs": False},
+    ))
+    body = {"input_ids": [1, 2, 3], "attention_mask": [1] * 3}
+    with pytest.raises(ValueError, match="float16"):
+        run(processor.process_request("bert_fp16", "", body))
+
+
+def test_fp16_torchscript_endpoint_allowed(processor, store, tmp_path):
+    """TorchScript fp16 endpoints stay supported (torch-native ops handle
+    fp16; only the in-tree attention path is bf16-only)."""
+
+    class M(torch.nn.Module):
+        def forward(self, x):
+            return x * 2.0
+
+    p = tmp_path / "m.pt"
+    torch.jit.script(M()).save(str(p))
+    rec = store.register_model(name="dbl", project="p", path=str(p))
+    processor.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="dbl_ep", model_id=rec.model_id,
+        auxiliary_cfg={"dtype": "float16", "use_graphs": False,
+                       "max_queue_delay_us": 500},
+    ))
+    out = run(processor.process_request("dbl_ep", "", [1.0, 2.0]))
+    assert np.allclose(np.asarray(out), [2.0, 4.0])
