@@ -182,8 +182,21 @@ class BasePreprocessRequest:
             except Exception:
                 pass
 
+    # multi-process front mode: GPU engine types resolve to an SHM proxy
+    # that ships tensors to the engine-owner process (serving/front.py)
+    _engine_overrides: Dict[str, type] = {}
+
+    @classmethod
+    def override_engine(cls, engine_name: str, engine_cls: Optional[type]):
+        if engine_cls is None:
+            cls._engine_overrides.pop(engine_name, None)
+        else:
+            cls._engine_overrides[engine_name] = engine_cls
+
     @classmethod
     def get_engine_cls(cls, engine: str) -> type:
+        if engine in cls._engine_overrides:
+            return cls._engine_overrides[engine]
         if engine not in cls.__preprocessing_lookup:
             raise ValueError("Engine '{}' not recognized".format(engine))
         return cls.__preprocessing_lookup[engine]

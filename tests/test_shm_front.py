@@ -1,0 +1,261 @@
+"""Multi-process serving topology: SHM rings, framing, front<->owner e2e.
+
+The topology under test (serving/launch.py): N HTTP front workers sharing
+one port via SO_REUSEPORT, one engine-owner process per GPU, connected by
+SPSC shared-memory rings -- GPU dynamic batches stay whole regardless of
+HTTP worker count (replaces the reference's N-workers-N-model-copies
+gunicorn mode, entrypoint.sh:56-72)."""
+
+import asyncio
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+import uuid
+
+import numpy as np
+import pytest
+
+from clearml_serving_amd.serving import shm_transport as st
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+# --------------------------------------------------------------------- #
+# ring semantics (both implementations)
+# --------------------------------------------------------------------- #
+def _ring_roundtrip(make, unlink):
+    name = "/cmls_t_{}".format(uuid.uuid4().hex[:8])
+    w = make(name, 1 << 14, True)
+    r = make(name, 0, False)
+    try:
+        assert w.push(b"a")
+        assert w.push(b"bb" * 100)
+        assert r.drain(10) == [b"a", b"bb" * 100]
+        # wrap across the end many times
+        for i in range(500):
+            payload = bytes([i % 251]) * (i % 900 + 1)
+            assert w.push(payload)
+            assert r.drain(4) == [payload]
+        # backpressure: fill to capacity, then drain everything
+        n = 0
+        while w.push(b"z" * 1000):
+            n += 1
+        assert n >= 10
+        got = []
+        while True:
+            batch = r.drain(64)
+            if not batch:
+                break
+            got.extend(batch)
+        assert len(got) == n
+        # ring usable again after full drain
+        assert w.push(b"after")
+        assert r.drain(4) == [b"after"]
+    finally:
+        w.close()
+        r.close()
+        unlink(name)
+
+
+def test_python_ring_roundtrip():
+    _ring_roundtrip(st.PyShmRing, st.PyShmRing.unlink)
+
+
+@pytest.mark.skipif(not st.HAVE_NATIVE_RING,
+                    reason="_shmring extension not built")
+def test_native_ring_roundtrip():
+    _ring_roundtrip(st.ShmRing, st.ShmRing.unlink)
+
+
+@pytest.mark.skipif(not st.HAVE_NATIVE_RING,
+                    reason="_shmring extension not built")
+def test_native_and_python_rings_interoperate():
+    """Both implementations share the record format: a producer of one kind
+    must be drainable by the other (fronts may run without the extension)."""
+    name = "/cmls_x_{}".format(uuid.uuid4().hex[:8])
+    w = st.ShmRing(name, 1 << 14, True)
+    r = st.PyShmRing(name, 0, False)
+    try:
+        for p in (b"one", b"two" * 50, b"\x00" * 999):
+            assert w.push(p)
+        assert r.drain(10) == [b"one", b"two" * 50, b"\x00" * 999]
+        # and the reverse direction on the same ring
+        assert r.push(b"back")
+        assert w.drain(10) == [b"back"]
+    finally:
+        w.close()
+        r.close()
+        st.ShmRing.unlink(name)
+
+
+# --------------------------------------------------------------------- #
+# framing
+# --------------------------------------------------------------------- #
+def test_request_framing_tensor_dict():
+    data = {"input_ids": np.arange(128, dtype=np.int64),
+            "attention_mask": np.ones(128, dtype=np.int32)}
+    buf = st.pack_request(42, "bert_ep/1", data)
+    rid, url, out = st.unpack_request(buf)
+    assert rid == 42 and url == "bert_ep/1"
+    assert set(out) == {"input_ids", "attention_mask"}
+    np.testing.assert_array_equal(out["input_ids"], data["input_ids"])
+    assert out["attention_mask"].dtype == np.int32
+
+
+def test_request_framing_single_array_and_lists():
+    a = np.random.rand(3, 4).astype(np.float32)
+    rid, url, out = st.unpack_request(st.pack_request(1, "m", a))
+    np.testing.assert_array_equal(out, a)
+    # plain python list converts (float64 -> float32)
+    rid, url, out = st.unpack_request(st.pack_request(2, "m", [1.0, 2.0]))
+    assert out.dtype == np.float32
+    np.testing.assert_allclose(out, [1.0, 2.0])
+
+
+def test_request_framing_pickle_fallback():
+    body = {"prompt": "hi", "max_tokens": 4, "nested": {"a": [1, "x"]}}
+    rid, url, out = st.unpack_request(st.pack_request(3, "llm", body))
+    assert out == body
+
+
+def test_response_framing():
+    rid, status, out = st.unpack_response(
+        st.pack_response(7, np.float32([1, 2, 3])))
+    assert rid == 7 and status == st.STATUS_OK_TENSORS
+    np.testing.assert_allclose(out, [1, 2, 3])
+    rid, status, out = st.unpack_response(
+        st.pack_response(8, {"text": "ok", "tokens": 3}))
+    assert status == st.STATUS_OK_PICKLE and out["text"] == "ok"
+    rid, status, out = st.unpack_response(st.pack_response(9, error="boom"))
+    assert status == st.STATUS_ERROR and out == "boom"
+
+
+def test_placement_deterministic():
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.front import _placement
+
+    ep = ModelEndpoint(engine_type="hip", serving_url="m1",
+                       auxiliary_cfg={"gpu": 3})
+    assert _placement(ep, 8) == 3
+    assert _placement(ep, 2) == 1
+    ep2 = ModelEndpoint(engine_type="hip", serving_url="m2")
+    assert _placement(ep2, 4) == _placement(ep2, 4)  # stable
+
+
+# --------------------------------------------------------------------- #
+# full multi-process e2e over real HTTP (CPU engines + SHM proxy)
+# --------------------------------------------------------------------- #
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.fixture()
+def serving_session(tmp_path):
+    """Store with a hip (CPU bert) endpoint + an sklearn endpoint."""
+    import joblib
+    from sklearn.linear_model import LinearRegression
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="shm-e2e",
+                                 force_create=True)
+    card = tmp_path / "bert.json"
+    card.write_text(json.dumps({"arch": "bert-base", "num_labels": 2,
+                                "dtype": "float32", "vocab_size": 500}))
+    rec = store.register_model(name="bert", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="transformer_model",
+        model_id=rec.model_id,
+        auxiliary_cfg={"max_queue_delay_us": 2000, "use_graphs": False,
+                       "warmup": False}))
+
+    X = np.array([[0.0, 0.0], [1.0, 1.0], [2.0, 2.0]])
+    m = LinearRegression().fit(X, np.array([0.0, 2.0, 4.0]))
+    mp = tmp_path / "lin.pkl"
+    joblib.dump(m, mp)
+    rec2 = store.register_model(name="lin", project="p", path=str(mp))
+    code = tmp_path / "sk.py"
+    code.write_text(
+        "import numpy as np\n"
+        "class Preprocess(object):\n"
+        "    def preprocess(self, body, state, collect_custom_statistics_fn=None):\n"
+        "        return np.array([[body['x0'], body['x1']]])\n"
+        "    def postprocess(self, data, state, collect_custom_statistics_fn=None):\n"
+        "        return dict(y=data.tolist()[0])\n")
+    proc.add_endpoint(ModelEndpoint(
+        engine_type="sklearn", serving_url="test_model_sklearn",
+        model_id=rec2.model_id), preprocess_code=str(code))
+    proc.serialize()
+    return store_root, proc.get_id()
+
+
+@pytest.mark.timeout(180)
+def test_multiproc_front_owner_http_e2e(serving_session):
+    """launch.py with 2 fronts + 1 owner: hip endpoint served through the
+    SHM proxy (batching in the owner), sklearn served front-local; both
+    answer over real HTTP on a shared SO_REUSEPORT port."""
+    store_root, session_id = serving_session
+    port = _free_port()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    launcher = subprocess.Popen(
+        [sys.executable, "-m", "clearml_serving_amd.serving.launch",
+         "--store", store_root, "--session", session_id,
+         "--host", "127.0.0.1", "--port", str(port),
+         "--workers", "2", "--owners", "1", "--ring-mb", "8",
+         "--no-restart"],
+        env=env, cwd=REPO)
+    try:
+        import httpx
+
+        base = "http://127.0.0.1:{}".format(port)
+        with httpx.Client(base_url=base, timeout=30.0) as client:
+            for _ in range(240):
+                try:
+                    if client.get("/health").status_code == 200:
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.5)
+                assert launcher.poll() is None, "launcher died"
+            else:
+                pytest.fail("service did not come up")
+
+            # sklearn (front-local CPU engine)
+            r = client.post("/serve/test_model_sklearn",
+                            json={"x0": 1.0, "x1": 1.0})
+            assert r.status_code == 200, r.text
+            assert abs(float(np.asarray(r.json()["y"]).ravel()[0]) - 2.0) < 1e-5
+
+            # hip endpoint through the SHM proxy (dict tensor payload)
+            payload = {"input_ids": list(range(1, 65)),
+                       "attention_mask": [1] * 64}
+            outs = []
+            for _ in range(12):
+                r = client.post("/serve/transformer_model", json=payload)
+                assert r.status_code == 200, r.text
+                outs.append(r.json())
+            assert all(len(o) == 2 for o in outs)
+            # deterministic model: identical inputs -> identical logits
+            assert all(np.allclose(o, outs[0], atol=1e-4) for o in outs)
+
+            # 404 still routes correctly through the front
+            r = client.post("/serve/nope", json={})
+            assert r.status_code == 404
+    finally:
+        launcher.terminate()
+        try:
+            launcher.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            launcher.kill()
