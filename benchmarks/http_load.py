@@ -1,13 +1,20 @@
 #!/usr/bin/env python3
 """HTTP-level load test: the reference's ApacheBench recipe
 (`ab -l -n 8000 -c 128 -p payload.json .../serve/transformer_model`,
-examples/huggingface/readme.md:141-144) as a self-contained async client.
+examples/huggingface/readme.md:141-144) as a self-contained multi-process
+load client against the serving stack.
 
-Starts uvicorn in-process against a temp store with a BERT-base endpoint,
-fires N requests at concurrency C through real HTTP, reports rps + latency
-percentiles.
+Topologies under test (serving/launch.py):
+  --workers 0   single process owns HTTP + GPU (round-1 topology)
+  --workers N   N SO_REUSEPORT HTTP fronts + 1 engine owner per GPU,
+                shared-memory tensor handoff (batches stay whole)
 
-    python benchmarks/http_load.py [-n 8000] [-c 128]
+The client is multi-process aiohttp (a single Python event loop cannot
+saturate a multi-process server); per ab semantics connections are NOT
+reused unless --keepalive is given.
+
+    python benchmarks/http_load.py [-n 8000] [-c 128] [--workers 8]
+        [--client-procs 4] [--keepalive] [--out FILE]
 """
 
 import argparse
@@ -16,6 +23,7 @@ import json
 import multiprocessing
 import os
 import statistics
+import subprocess
 import sys
 import tempfile
 import time
@@ -23,32 +31,7 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 PORT = int(os.environ.get("CMLS_HTTP_BENCH_PORT", 18080))
-
-
-def server_proc(store_root, session_id, port, workers):
-    import uvicorn
-
-    if workers > 1:
-        # multi-worker: app built per worker process from env (the
-        # reference's gunicorn topology -- N workers, N engine copies).
-        # uvicorn's worker bootstrap fdopens stdin, which the spawned
-        # context closed -- give it /dev/null
-        import sys
-
-        devnull = os.open(os.devnull, os.O_RDONLY)
-        os.dup2(devnull, 0)
-        sys.stdin = os.fdopen(0)
-        os.environ["CLEARML_SERVING_TASK_ID"] = session_id
-        os.environ["CLEARML_SERVING_AMD_STORE"] = store_root
-        os.environ["CLEARML_SERVING_POLL_FREQ"] = "60"
-        uvicorn.run("clearml_serving_amd.serving.app:app", host="127.0.0.1",
-                    port=port, log_level="warning", workers=workers)
-    else:
-        from clearml_serving_amd.serving.app import create_app
-
-        app = create_app(session_id=session_id, store_root=store_root,
-                         poll_frequency_sec=3600)
-        uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def setup_store(tmpdir):
@@ -62,11 +45,11 @@ def setup_store(tmpdir):
     proc = ModelRequestProcessor(store=store, name="http-bench",
                                  force_create=True)
     card = os.path.join(tmpdir, "bert.json")
+    gpu = torch.cuda.is_available()
     with open(card, "wt") as f:
         json.dump({"arch": "bert-base", "num_labels": 2,
-                   "dtype": "bfloat16"}, f)
+                   "dtype": "bfloat16" if gpu else "float32"}, f)
     rec = store.register_model(name="bert", project="bench", path=card)
-    gpu = torch.cuda.is_available()
     proc.add_endpoint(ModelEndpoint(
         engine_type="hip", serving_url="transformer_model",
         model_id=rec.model_id,
@@ -80,81 +63,139 @@ def setup_store(tmpdir):
     return os.path.join(tmpdir, "store"), proc.get_id()
 
 
-async def load(n, c, payload, url):
-    import httpx
+def client_proc(proc_idx, n, c, payload, url, keepalive, warmup, conn):
+    """One load-client process: n requests at concurrency c (aiohttp)."""
 
-    latencies = []
-    errors = [0]
-    sem = asyncio.Semaphore(c)
+    async def run():
+        import aiohttp
 
-    async with httpx.AsyncClient(base_url=url, timeout=60.0) as client:
-        # wait for readiness
-        for _ in range(120):
-            try:
-                r = await client.get("/health")
-                if r.status_code == 200:
-                    break
-            except Exception:
-                pass
-            await asyncio.sleep(1.0)
+        connector = aiohttp.TCPConnector(limit=c * 2,
+                                         force_close=not keepalive)
+        timeout = aiohttp.ClientTimeout(total=120)
+        latencies = []
+        errors = [0]
+        sem = asyncio.Semaphore(c)
+        body = json.dumps(payload).encode()
+        headers = {"Content-Type": "application/json"}
 
-        async def one():
-            async with sem:
-                t0 = time.perf_counter()
-                try:
-                    r = await client.post("/serve/transformer_model",
-                                          json=payload)
-                    if r.status_code != 200:
+        async with aiohttp.ClientSession(connector=connector,
+                                         timeout=timeout) as session:
+            async def one():
+                async with sem:
+                    t0 = time.perf_counter()
+                    try:
+                        async with session.post(url, data=body,
+                                                headers=headers) as r:
+                            await r.read()
+                            if r.status != 200:
+                                errors[0] += 1
+                    except Exception:
                         errors[0] += 1
-                except Exception:
-                    errors[0] += 1
-                latencies.append(time.perf_counter() - t0)
+                    latencies.append(time.perf_counter() - t0)
 
-        # warmup (graph capture)
-        await asyncio.gather(*[one() for _ in range(min(256, n))])
-        latencies.clear()
-        errors[0] = 0
-        t0 = time.perf_counter()
-        await asyncio.gather(*[one() for _ in range(n)])
-        dt = time.perf_counter() - t0
-    lat = sorted(latencies)
-    return {
-        "metric": "HTTP requests/s, BERT-base, ab-recipe (-n {} -c {})".format(n, c),
-        "value": round(n / dt, 2),
-        "errors": errors[0],
-        "p50_ms": round(statistics.median(lat) * 1000, 2),
-        "p95_ms": round(lat[int(len(lat) * 0.95)] * 1000, 2),
-        "p99_ms": round(lat[int(len(lat) * 0.99)] * 1000, 2),
-    }
+            await asyncio.gather(*[one() for _ in range(warmup)])
+            latencies.clear()
+            errors[0] = 0
+            t0 = time.perf_counter()
+            await asyncio.gather(*[one() for _ in range(n)])
+            dt = time.perf_counter() - t0
+        return {"n": n, "dt": dt, "errors": errors[0],
+                "latencies": latencies}
+
+    res = asyncio.new_event_loop().run_until_complete(run())
+    conn.send(res)
+    conn.close()
+
+
+def wait_healthy(url, tries=240):
+    import urllib.request
+
+    for _ in range(tries):
+        try:
+            with urllib.request.urlopen(url + "/health", timeout=2) as r:
+                if r.status == 200:
+                    return True
+        except Exception:
+            pass
+        time.sleep(0.5)
+    return False
 
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("-n", type=int, default=8000)
     ap.add_argument("-c", type=int, default=128)
-    ap.add_argument("--workers", type=int, default=1)
+    ap.add_argument("--workers", type=int, default=0,
+                    help="HTTP front workers (0 = single-process topology)")
+    ap.add_argument("--client-procs", type=int, default=4)
+    ap.add_argument("--keepalive", action="store_true",
+                    help="reuse connections (ab default is one per request)")
+    ap.add_argument("--warmup", type=int, default=256)
+    ap.add_argument("--out", default=None)
     args = ap.parse_args()
 
     with tempfile.TemporaryDirectory() as tmpdir:
         store_root, session_id = setup_store(tmpdir)
-        # spawn: the parent touched the HIP runtime; a forked child cannot
-        # re-initialize it
-        ctx = multiprocessing.get_context("spawn")
-        # not daemonic: uvicorn's multi-worker mode spawns child processes
-        proc = ctx.Process(
-            target=server_proc,
-            args=(store_root, session_id, PORT, args.workers), daemon=False)
-        proc.start()
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        server = subprocess.Popen(
+            [sys.executable, "-m", "clearml_serving_amd.serving.launch",
+             "--store", store_root, "--session", session_id,
+             "--host", "127.0.0.1", "--port", str(PORT),
+             "--workers", str(args.workers)],
+            env=env, cwd=REPO)
+        base = "http://127.0.0.1:{}".format(PORT)
         payload = {"input_ids": list(range(1, 129)),
                    "attention_mask": [1] * 128}
         try:
-            result = asyncio.new_event_loop().run_until_complete(
-                load(args.n, args.c, payload,
-                     "http://127.0.0.1:{}".format(PORT)))
-            print(json.dumps(result))
+            if not wait_healthy(base):
+                raise RuntimeError("service did not come up")
+            P = max(args.client_procs, 1)
+            ctx = multiprocessing.get_context("spawn")
+            pipes, procs = [], []
+            per_n = args.n // P
+            per_c = max(args.c // P, 1)
+            per_warm = max(args.warmup // P, 1)
+            url = base + "/serve/transformer_model"
+            for i in range(P):
+                parent, child = ctx.Pipe()
+                p = ctx.Process(target=client_proc,
+                                args=(i, per_n, per_c, payload, url,
+                                      args.keepalive, per_warm, child))
+                p.start()
+                pipes.append(parent)
+                procs.append(p)
+            results = [pipe.recv() for pipe in pipes]
+            for p in procs:
+                p.join(timeout=30)
+            total_n = sum(r["n"] for r in results)
+            # wall time = max over client procs (they start ~together)
+            dt = max(r["dt"] for r in results)
+            lat = sorted(x for r in results for x in r["latencies"])
+            errors = sum(r["errors"] for r in results)
+            out = {
+                "metric": "HTTP requests/s, BERT-base, ab-recipe "
+                          "(-n {} -c {})".format(args.n, args.c),
+                "value": round(total_n / dt, 2),
+                "errors": errors,
+                "p50_ms": round(statistics.median(lat) * 1000, 2),
+                "p95_ms": round(lat[int(len(lat) * 0.95)] * 1000, 2),
+                "p99_ms": round(lat[int(len(lat) * 0.99)] * 1000, 2),
+                "topology": ("{} fronts + shm engine-owner".format(
+                    args.workers) if args.workers else "single process"),
+                "keepalive": bool(args.keepalive),
+                "client_procs": P,
+            }
+            print(json.dumps(out))
+            if args.out:
+                with open(args.out, "wt") as f:
+                    json.dump(out, f)
         finally:
-            proc.terminate()
-            proc.join(timeout=10)
+            server.terminate()
+            try:
+                server.wait(timeout=15)
+            except subprocess.TimeoutExpired:
+                server.kill()
 
 
 if __name__ == "__main__":
