@@ -77,28 +77,43 @@ def build_processor(device_idx, tmpdir, use_graphs=True):
     return processor
 
 
-async def run_step(processor, resnet_inputs, bert_inputs, latencies=None):
-    """One bench step: REQ_PER_STEP requests through the serving path."""
+async def run_step(processor, resnet_inputs, bert_inputs, latencies=None,
+                   checksum=None, offset=0):
+    """One bench step: REQ_PER_STEP requests through the serving path.
+
+    ``offset`` rotates through the input pools so consecutive steps see
+    different request tensors (no caching artifacts); ``checksum``
+    accumulates a scalar over every response (outputs are consumed, and the
+    total is printed with the result for cross-run comparison)."""
 
     async def one(endpoint, body):
         t0 = time.perf_counter()
-        await processor.process_request(endpoint, "", body)
+        out = await processor.process_request(endpoint, "", body)
         if latencies is not None:
             latencies.append(time.perf_counter() - t0)
+        if checksum is not None:
+            import numpy as np
+
+            arr = np.asarray(out, dtype=np.float64)
+            checksum[0] += float(arr.sum())
+            checksum[1] += arr.size
 
     tasks = []
     n = REQ_PER_STEP // 2
     for i in range(n):
-        tasks.append(one("resnet50_bench", resnet_inputs[i % len(resnet_inputs)]))
-        tasks.append(one("bert_bench", bert_inputs[i % len(bert_inputs)]))
+        j = (offset + i) % len(resnet_inputs)
+        tasks.append(one("resnet50_bench", resnet_inputs[j]))
+        tasks.append(one("bert_bench", bert_inputs[j]))
     await asyncio.gather(*tasks)
 
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=8)
-    ap.add_argument("--warmup", type=int, default=3)
+    # defaults sized for a >=60 s steady-state timed region on one MI355X
+    # (~50 ms/step measured); still finishes within minutes
+    ap.add_argument("--steps", type=int, default=1200)
+    ap.add_argument("--warmup", type=int, default=50)
     ap.add_argument("--no-graphs", action="store_true")
     args = ap.parse_args()
 
@@ -122,7 +137,9 @@ def main():
     # synthetic requests of the headline shapes (BASELINE.json configs 2+3):
     # ResNet-50: [3, 224, 224] bf16; BERT-base: seq_len 128 token ids
     torch.manual_seed(1234 + rank)
-    n_distinct = 8
+    # per-request unique tensors: the pool is larger than any batch window,
+    # and run_step rotates its offset so steps never repeat a request lineup
+    n_distinct = int(os.environ.get("CMLS_BENCH_DISTINCT", 256))
     dtype = torch.bfloat16 if has_gpu else torch.float32
     resnet_inputs = [torch.randn(3, 224, 224, dtype=dtype)
                      for _ in range(n_distinct)]
@@ -141,9 +158,10 @@ def main():
 
         log("warmup: {} steps (includes hipGraph capture per bucket)".format(
             args.warmup))
-        for _ in range(max(args.warmup, 1)):
+        for w in range(max(args.warmup, 1)):
             loop.run_until_complete(
-                run_step(processor, resnet_inputs, bert_inputs))
+                run_step(processor, resnet_inputs, bert_inputs,
+                         offset=w * 13))
 
         # ---- timed region: barrier + sync on both sides ---- #
         if dist:
@@ -152,9 +170,11 @@ def main():
             torch.cuda.synchronize()
         t0 = time.perf_counter()
         latencies = []
-        for _ in range(args.steps):
+        checksum = [0.0, 0]
+        for k in range(args.steps):
             loop.run_until_complete(
-                run_step(processor, resnet_inputs, bert_inputs, latencies))
+                run_step(processor, resnet_inputs, bert_inputs, latencies,
+                         checksum=checksum, offset=k * 13))
         if has_gpu:
             torch.cuda.synchronize()
         if dist:
@@ -199,6 +219,9 @@ def main():
                     "p50_latency_ms": round(p50, 2) if p50 else None,
                     "p99_latency_ms": round(p99, 2) if p99 else None,
                     "requests_per_step": REQ_PER_STEP,
+                    "timed_region_s": round(elapsed, 2),
+                    "output_checksum": round(checksum[0], 3),
+                    "outputs_consumed": checksum[1],
                 },
             }
             print(json.dumps(result), flush=True)
