@@ -20,6 +20,7 @@ KV sizing: pages are allocated once at startup from a fraction of free HBM
 import asyncio
 import json
 import os
+import random
 import time
 import uuid
 from collections import deque
@@ -347,6 +348,10 @@ class LlmEngine:
                          dtype=self.dtype, device=self.device))
             for _ in range(mcfg.layers)
         ]
+        if self.tp_size > 1:
+            from .plan_codec import PlanCodec
+
+            self._plan_codec = PlanCodec(cfg, self.device)
         self._started = True
 
     # ------------------------------------------------------------------ #
@@ -556,6 +561,11 @@ class LlmEngine:
         bs = self.cfg.block_size
         return seq.blocks[pos // bs] * bs + pos % bs
 
+    def _sample_spec(self, seqs: List[Sequence]) -> List[tuple]:
+        seed = random.getrandbits(31)
+        return [(s.params.temperature, s.params.top_k, s.params.top_p, seed)
+                for s in seqs]
+
     def _prefill(self, seqs: List[Sequence]) -> None:
         self.stats["prefill_batches"] += 1
         plan = {
@@ -563,11 +573,12 @@ class LlmEngine:
             "prompts": [list(s.prompt_ids) for s in seqs],
             "slots": [[self._slot(s, p) for p in range(len(s.prompt_ids))]
                       for s in seqs],
+            "sample": self._sample_spec(seqs),
         }
         self._tp_broadcast(plan)
         logits = self._exec_prefill(plan)
         self.stats["prompt_tokens"] += sum(len(p) for p in plan["prompts"])
-        self._sample_and_emit(seqs, logits)
+        self._sample_and_emit(seqs, logits, sample=plan["sample"])
 
     def _prefill_chunk(self, seqs: List["Sequence"], budget: int) -> None:
         """Advance several sequences' prefill by up to ``budget`` tokens
@@ -596,6 +607,10 @@ class LlmEngine:
             "complete": [s.prefilled + c >= len(s.prompt_ids)
                          for s, c in batch],
         }
+        # sampling spec only for rows whose chunk completes the prompt
+        done_seqs = [s for (s, c), comp in zip(batch, plan["complete"])
+                     if comp]
+        plan["sample"] = self._sample_spec(done_seqs)
         self._tp_broadcast(plan)
         logits = self._exec_chunk(plan)
         done = []
@@ -605,7 +620,7 @@ class LlmEngine:
             if complete:
                 done.append(s)
         if done:
-            self._sample_and_emit(done, logits)
+            self._sample_and_emit(done, logits, sample=plan["sample"])
 
     @torch.inference_mode()
     def _exec_chunk(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -641,7 +656,7 @@ class LlmEngine:
         return self.model(tokens.view(-1).to(dev),
                           positions.view(-1).to(dev),
                           kv_caches=self.kv_caches, attn_ctx=attn_ctx,
-                          last_token_idx=last_idx)
+                          last_token_idx=last_idx, gather_logits=False)
 
     @torch.inference_mode()
     def _exec_prefill(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -672,7 +687,7 @@ class LlmEngine:
         return self.model(
             tokens.view(-1).to(dev), positions.view(-1).to(dev),
             kv_caches=self.kv_caches, attn_ctx=attn_ctx,
-            last_token_idx=last_idx)
+            last_token_idx=last_idx, gather_logits=False)
 
     @torch.inference_mode()
     def _exec_embed(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -763,10 +778,11 @@ class LlmEngine:
             "slots": [self._slot(s, len(s) - 1) for s in seqs],
             "seq_lens": [len(s) for s in seqs],
             "blocks": [list(s.blocks) for s in seqs],
+            "sample": self._sample_spec(seqs),
         }
         self._tp_broadcast(plan)
         logits = self._exec_decode(plan)
-        self._sample_and_emit(seqs, logits)
+        self._sample_and_emit(seqs, logits, sample=plan["sample"])
 
     @torch.inference_mode()
     def _exec_decode(self, plan: Dict[str, Any]) -> torch.Tensor:
@@ -790,7 +806,8 @@ class LlmEngine:
             "slot_mapping": slot_map,
         }
         return self.model(tokens, positions, kv_caches=self.kv_caches,
-                          attn_ctx=attn_ctx, last_token_idx=None)
+                          attn_ctx=attn_ctx, last_token_idx=None,
+                          gather_logits=False)
 
     # -------------------- decode hipGraph capture --------------------- #
     # A llama-8B decode step dispatches ~350 kernels from Python; at B=64
@@ -800,8 +817,13 @@ class LlmEngine:
     # seq_len=1 so the attention read is bounded) replays the whole step
     # as a single launch. vLLM uses the same strategy on its decode path.
     def _use_decode_graphs(self) -> bool:
+        # TP: RCCL all-reduces ARE capturable in hipGraphs (the tp_serve
+        # launcher sets TORCH_NCCL_ASYNC_ERROR_HANDLING=0, required for
+        # collective capture); CMLS_TP_GRAPHS=0 is the kill switch
+        if (getattr(self, "tp_size", 1) > 1
+                and os.environ.get("CMLS_TP_GRAPHS", "1") == "0"):
+            return False
         return (self.cfg.decode_graphs
-                and getattr(self, "tp_size", 1) <= 1
                 and self.device.type == "cuda"
                 and os.environ.get("CMLS_LLM_GRAPHS", "1") != "0")
 
@@ -833,7 +855,7 @@ class LlmEngine:
         def fwd():
             return self.model(static["tokens"], static["positions"],
                               kv_caches=self.kv_caches, attn_ctx=attn_ctx,
-                              last_token_idx=None)
+                              last_token_idx=None, gather_logits=False)
 
         # warm up on a side stream (allocator + kernels settle), capture on
         # the current stream; all graphs share one memory pool
@@ -902,35 +924,89 @@ class LlmEngine:
     # tensor-parallel coordination (rank 0 schedules, workers follow)
     # ------------------------------------------------------------------ #
     def _tp_broadcast(self, plan: Optional[Dict[str, Any]]) -> None:
+        """ONE fixed-layout int32 tensor broadcast per step (RCCL on GPU).
+        Round 1 used broadcast_object_list -- pickle + CPU sync per decode
+        step, fatal for a TP=8 latency budget; plans that overflow the fixed
+        buffer (rare giant embed batches) fall back to the object path."""
         if getattr(self, "tp_size", 1) <= 1 or self.tp_rank != 0:
             return
         import torch.distributed as dist
 
-        dist.broadcast_object_list([plan], src=0)
+        buf = self._plan_codec.encode(plan)
+        if buf is None:
+            dist.broadcast(self._plan_codec.mark_object(), src=0)
+            dist.broadcast_object_list([plan], src=0)
+        else:
+            dist.broadcast(buf, src=0)
+
+    def _tp_receive(self) -> Dict[str, Any]:
+        import torch.distributed as dist
+
+        dist.broadcast(self._plan_codec.buf, src=0)
+        plan = self._plan_codec.decode(self._plan_codec.buf)
+        if plan is None:  # object fallback
+            box = [None]
+            dist.broadcast_object_list(box, src=0)
+            plan = box[0] or {"mode": "stop"}
+        return plan
 
     def run_tp_worker(self) -> None:
         """Worker-rank loop: execute rank 0's step plans until shutdown.
 
-        The model's row-parallel all-reduces and the logits all-gather are
-        the synchronization points; workers discard logits (sampling happens
-        on rank 0 only)."""
-        import torch.distributed as dist
-
+        The model's row-parallel all-reduces are per-layer sync points; the
+        sampling collectives (packed-argmax all-reduce / rank-0 gather) make
+        every rank participate in token selection, so no rank ever
+        materializes full-vocab logits."""
         assert self.tp_size > 1 and self.tp_rank != 0
         while True:
-            box = [None]
-            dist.broadcast_object_list(box, src=0)
-            plan = box[0]
-            if plan is None or plan.get("mode") == "stop":
+            plan = self._tp_receive()
+            mode = plan.get("mode")
+            if mode == "stop":
                 return
-            if plan["mode"] == "prefill":
-                self._exec_prefill(plan)
-            elif plan["mode"] == "chunk":
-                self._exec_chunk(plan)
-            elif plan["mode"] == "embed":
+            if mode == "prefill":
+                logits = self._exec_prefill(plan)
+            elif mode == "chunk":
+                logits = self._exec_chunk(plan)
+            elif mode == "embed":
                 self._exec_embed(plan)
+                continue
             else:
-                self._exec_decode(plan)
+                logits = self._exec_decode(plan)
+            if plan.get("sample"):
+                self._tp_sample_rows(logits, plan["sample"])
+
+    def _tp_sample_rows(self, logits: torch.Tensor, sample) -> torch.Tensor:
+        """Collective sampling over vocab-sharded logits; every rank calls
+        this with the identical ``sample`` row list (from the plan), so the
+        collectives line up. Returns next token ids (valid on rank 0; the
+        argmax/gumbel paths are valid on every rank)."""
+        from ...parallel import tp as tp_mod
+
+        groups: Dict[tuple, List[int]] = {}
+        for i, row in enumerate(sample):
+            groups.setdefault(tuple(row), []).append(i)
+        next_ids = torch.zeros(len(sample), dtype=torch.long)
+        voff = self.tp_rank * logits.shape[-1]
+        for (temp, top_k, top_p, seed), idxs in groups.items():
+            rows = logits[idxs]
+            if top_k > 0 or top_p < 1.0:
+                # full distribution needed: gather these rows to rank 0
+                full = tp_mod.gather_rows_to_rank0(rows)
+                if self.tp_rank == 0:
+                    gen = torch.Generator(device=full.device)
+                    gen.manual_seed(int(seed) & 0x7FFFFFFF)
+                    sampled = ops.sample_top_k_top_p(
+                        full, temperature=temp, top_k=top_k, top_p=top_p,
+                        generator=gen)
+                else:
+                    sampled = torch.zeros(len(idxs), dtype=torch.long)
+            elif temp == 0.0:
+                sampled = tp_mod.argmax_sharded(rows, voff)
+            else:
+                sampled = tp_mod.sample_gumbel_sharded(
+                    rows, voff, temp, [seed])
+            next_ids[idxs] = sampled.to("cpu", dtype=torch.long)
+        return next_ids
 
     def tp_shutdown(self) -> None:
         if getattr(self, "tp_size", 1) > 1 and self.tp_rank == 0:
@@ -957,18 +1033,25 @@ class LlmEngine:
                 cut = min(cut, idx)
         return text[:cut]
 
-    def _sample_and_emit(self, seqs: List[Sequence], logits: torch.Tensor) -> None:
-        # group rows by identical sampling params for batched kernels
-        groups: Dict[tuple, List[int]] = {}
-        for i, s in enumerate(seqs):
-            key = (s.params.temperature, s.params.top_k, s.params.top_p)
-            groups.setdefault(key, []).append(i)
-        next_ids = torch.empty(len(seqs), dtype=torch.long)
-        for (temp, top_k, top_p), idxs in groups.items():
-            rows = logits[idxs] if len(idxs) < len(seqs) else logits
-            sampled = ops.sample_top_k_top_p(
-                rows, temperature=temp, top_k=top_k, top_p=top_p)
-            next_ids[idxs] = sampled.cpu()
+    def _sample_and_emit(self, seqs: List[Sequence], logits: torch.Tensor,
+                         sample: Optional[List[tuple]] = None) -> None:
+        if getattr(self, "tp_size", 1) > 1:
+            # collective sampling over vocab shards (workers run the same
+            # call from run_tp_worker with the identical plan spec)
+            next_ids = self._tp_sample_rows(
+                logits, sample or self._sample_spec(seqs))
+        else:
+            # group rows by identical sampling params for batched kernels
+            groups: Dict[tuple, List[int]] = {}
+            for i, s in enumerate(seqs):
+                key = (s.params.temperature, s.params.top_k, s.params.top_p)
+                groups.setdefault(key, []).append(i)
+            next_ids = torch.empty(len(seqs), dtype=torch.long)
+            for (temp, top_k, top_p), idxs in groups.items():
+                rows = logits[idxs] if len(idxs) < len(seqs) else logits
+                sampled = ops.sample_top_k_top_p(
+                    rows, temperature=temp, top_k=top_k, top_p=top_p)
+                next_ids[idxs] = sampled.cpu()
         eos = self.tokenizer.eos_id
         now = time.time()
         for i, s in enumerate(seqs):

@@ -174,6 +174,7 @@ class LlamaForCausalLM(nn.Module):
         attn_ctx: Optional[Dict] = None,
         last_token_idx: Optional[torch.Tensor] = None,
         return_hidden: bool = False,
+        gather_logits: bool = True,
     ) -> torch.Tensor:
         """Returns logits [n_seqs, vocab] at the selected token positions
         (or the final-norm hidden states [T, hidden] when ``return_hidden``
@@ -190,8 +191,12 @@ class LlamaForCausalLM(nn.Module):
         if return_hidden:
             return x
         logits = self.lm_head(x)
-        if self.tp_size > 1:
-            # TP: lm_head is column-sharded -> all-gather the vocab shards
+        if self.tp_size > 1 and gather_logits:
+            # lm_head is column-sharded. The serving engine passes
+            # gather_logits=False and samples ON the shards (packed-argmax /
+            # Gumbel-max all-reduce of O(batch) words, parallel/tp.py) --
+            # the full-vocab all-gather (b x 128k x 2B per decode step) is
+            # only for offline full-logits consumers like the TP math tests.
             from ..parallel import tp as tp_mod
 
             logits = tp_mod.gather_logits(logits)

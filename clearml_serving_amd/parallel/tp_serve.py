@@ -42,6 +42,12 @@ def main():
                          "(launcher self-test; no HTTP server)")
     args = ap.parse_args()
 
+    # decode hipGraphs under TP capture the RCCL all-reduces; NCCL work
+    # objects are only capturable with async error handling off (the
+    # watchdog would poll events owned by the graph)
+    os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
+    os.environ.setdefault("NCCL_ASYNC_ERROR_HANDLING", "0")
+
     local_rank = tp.init_from_env()
     rank, world = tp.rank(), tp.world_size()
     if torch.cuda.is_available():

@@ -42,6 +42,36 @@ def check_tp_math(rank, world):
         print("TP-MATH-OK", flush=True)
 
 
+def check_sharded_sampling(rank, world):
+    """Distributed sampling primitives at world=2: packed-argmax all-reduce
+    and rank-0 row gather must equal their single-process counterparts."""
+    torch.manual_seed(3)
+    full_logits = torch.randn(6, 64)
+    # same full matrix on every rank (same seed); take this rank's shard
+    shard = full_logits[:, rank * 32:(rank + 1) * 32].contiguous()
+    got = tp.argmax_sharded(shard, rank * 32)
+    assert torch.equal(got, full_logits.argmax(dim=-1)), (got,)
+
+    gathered = tp.gather_rows_to_rank0(shard)
+    if rank == 0:
+        assert torch.equal(gathered, full_logits)
+    else:
+        assert gathered is None
+
+    # gumbel-max across shards: empirical dist ~ softmax on a 4-token vocab
+    logits = torch.tensor([[1.5, 0.5, -0.5, -1.5]])
+    shard2 = logits[:, rank * 2:(rank + 1) * 2].contiguous()
+    counts = torch.zeros(4)
+    n = 3000
+    for i in range(n):
+        idx = tp.sample_gumbel_sharded(shard2, rank * 2, 1.0, [i])
+        counts[idx.item()] += 1
+    probs = torch.softmax(logits[0], dim=-1)
+    assert torch.allclose(counts / n, probs, atol=0.04), (counts / n, probs)
+    if rank == 0:
+        print("TP-SAMPLE-OK", flush=True)
+
+
 def check_engine_protocol(rank, world):
     cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
                           block_size=16, max_model_len=128, device="cpu")
@@ -64,6 +94,24 @@ def check_engine_protocol(rank, world):
         loop = asyncio.new_event_loop()
         outs = loop.run_until_complete(gen())
         assert all(len(o) == 6 for o in outs), outs
+
+        # stochastic + top-k sampling exercise the gumbel and rank-0-gather
+        # collective paths through the full engine protocol
+        async def gen2():
+            toks = []
+            params2 = SamplingParams(temperature=0.8, top_k=8, top_p=0.95,
+                                     max_tokens=5, ignore_eos=True)
+            async for item in eng.generate("topk req", params2):
+                toks.extend(item["token_ids"])
+            toks3 = []
+            params3 = SamplingParams(temperature=0.8, max_tokens=5,
+                                     ignore_eos=True)
+            async for item in eng.generate("gumbel req", params3):
+                toks3.extend(item["token_ids"])
+            return toks, toks3
+
+        t2, t3 = loop.run_until_complete(gen2())
+        assert len(t2) == 5 and len(t3) == 5
         # embeddings path through the same plan-broadcast protocol
         vecs = loop.run_until_complete(
             eng.embed_batch(["tp embed a", "a longer tp embed text b"]))
@@ -79,6 +127,7 @@ def main():
     dist.init_process_group(backend="gloo")
     rank, world = dist.get_rank(), dist.get_world_size()
     check_tp_math(rank, world)
+    check_sharded_sampling(rank, world)
     check_engine_protocol(rank, world)
     dist.barrier()
     dist.destroy_process_group()
