@@ -78,11 +78,20 @@ class LlamaLayer(nn.Module):
 
     def forward(self, x, residual, positions, kv_cache, attn_ctx):
         cfg = self.cfg
-        # fused: residual += x_prev; x = rmsnorm(residual)
-        x = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps, residual=residual)
-        t = x.shape[0]
-
-        qkv = self._proj(self.qkv, x)
+        # fp8 serving path (models/quant.py): activation quantization is
+        # FUSED into the producing kernels -- rmsnorm/silu_mul emit fp8
+        # bytes + per-row scales in the same HBM pass
+        fp8 = not isinstance(self.qkv, nn.Linear)
+        if fp8:
+            x8, xs = ops.rmsnorm_fp8(x, self.attn_norm, cfg.rms_eps,
+                                     residual=residual)
+            t = x8.shape[0]
+            qkv = self.qkv.forward_q(x8, xs)
+        else:
+            # fused: residual += x_prev; x = rmsnorm(residual)
+            x = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps, residual=residual)
+            t = x.shape[0]
+            qkv = self._proj(self.qkv, x)
         # token-strided views into the merged projection -- the rope /
         # kv_cache_write / decode-attention kernels are stride-aware, so no
         # .contiguous() copies on the decode hot path
@@ -123,17 +132,30 @@ class LlamaLayer(nn.Module):
                 q, k_cache, v_cache, attn_ctx["block_table"],
                 attn_ctx["seq_lens"], scale=scale)
             ctx = ctx.view(t, self.heads * self.head_dim)
-        attn_out = self._proj(self.o_proj, ctx)
+        if fp8:
+            # attention context has no fused producer: one-pass dynamic quant
+            c8, cs = ops.quant_fp8(ctx)
+            attn_out = self.o_proj.forward_q(c8, cs)
+        else:
+            attn_out = self._proj(self.o_proj, ctx)
         if self.tp_size > 1:
             from ..parallel import tp as tp_mod
 
             tp_mod.maybe_all_reduce(attn_out)  # row-parallel o_proj
 
-        x = ops.rmsnorm(attn_out, self.mlp_norm, cfg.rms_eps,
-                        residual=residual)
-        gate_up = self.gate_up(x)
-        gate, up = gate_up.split([self.inter, self.inter], dim=-1)
-        mlp_out = self._proj(self.down, ops.silu_mul(gate, up))
+        if fp8:
+            x8, xs = ops.rmsnorm_fp8(attn_out, self.mlp_norm, cfg.rms_eps,
+                                     residual=residual)
+            gate_up = self.gate_up.forward_q(x8, xs)
+            gate, up = gate_up.split([self.inter, self.inter], dim=-1)
+            g8, gs = ops.silu_mul_fp8(gate, up)
+            mlp_out = self.down.forward_q(g8, gs)
+        else:
+            x = ops.rmsnorm(attn_out, self.mlp_norm, cfg.rms_eps,
+                            residual=residual)
+            gate_up = self.gate_up(x)
+            gate, up = gate_up.split([self.inter, self.inter], dim=-1)
+            mlp_out = self._proj(self.down, ops.silu_mul(gate, up))
         if self.tp_size > 1:
             from ..parallel import tp as tp_mod
 

@@ -1,32 +1,41 @@
 """fp8 (OCP e4m3) weight quantization for LLM serving.
 
-gfx950 runs fp8 MFMA at 2x the bf16 rate, and decode is weight-streaming
-bound -- fp8 weights halve the bytes per step. Measured with hipBLASLt
-(torch._scaled_mm) on MI355X: 1.69x on prefill-shaped GEMMs (M=2048),
-1.15x at decode M=64.
+Decode is weight-streaming bound: fp8 weights halve the HBM bytes per step.
+Round 1 ran fp8 through hipBLASLt with FOUR un-fused elementwise passes of
+dynamic activation quantization per projection and measured SLOWER than
+bf16 end to end (3.7k vs 6.0k out-tok/s). Round 2 fuses the quantization
+into the producing kernels -- ops.rmsnorm_fp8 / ops.silu_mul_fp8 emit fp8
+bytes + per-row scales directly (one HBM pass), and decode-shaped GEMMs run
+on the in-tree fp8 MFMA skinny kernel (ops.skinny_gemm_fp8) that fills the
+256-CU chip where hipBLASLt's tiles cannot. Prefill-shaped GEMMs stay on
+hipBLASLt scaled_mm (measured 1.69x vs bf16 at M=2048).
 
-Scheme: per-output-row weight scales (absmax/448), dynamic per-tensor
+Scheme: per-output-row weight scales (absmax/448), dynamic per-row (token)
 activation scales. Enable per endpoint via model card / aux
-``{"quantization": "fp8"}``.
-
-STATUS: experimental. Measured on llama-3-8B decode (random weights): the
-un-fused dynamic activation quantization (amax + scale + cast per
-projection) currently outweighs the GEMM gain -- 3.7k vs 6.0k out-tok/s
-against bf16. The win requires fusing the activation quant into the
-producing kernels (rmsnorm/silu_mul emitting fp8 + scale) and an
-fp8-native decode attention; see docs/ROADMAP.md.
+``{"quantization": "fp8"}``. The reference delegates all quantization to
+vLLM (preprocess_service.py:619-1095).
 """
 
 import torch
 import torch.nn as nn
+
+from .. import ops
 
 F8 = torch.float8_e4m3fn
 F8_MAX = 448.0
 
 
 class Fp8Linear(nn.Module):
-    """Drop-in replacement for a bias-free nn.Linear: fp8 weights with
-    per-row scales, dynamic per-tensor activation quantization."""
+    """Bias-free nn.Linear replacement: fp8 weights with per-row scales.
+
+    Two entry points:
+      forward(x)           -- bf16 in: quantize (one fused kernel) then GEMM
+      forward_q(x8, xs)    -- pre-quantized activations from a fused
+                              producer (rmsnorm_fp8 / silu_mul_fp8)
+    """
+
+    # decode-shaped rows route to the in-tree fp8 MFMA kernel
+    SKINNY_MAX_M = 64
 
     def __init__(self, linear: nn.Linear):
         super().__init__()
@@ -35,30 +44,44 @@ class Fp8Linear(nn.Module):
         scale = row_max / F8_MAX
         self.register_buffer("weight_fp8",
                              (w / scale).to(F8).contiguous())
-        # _scaled_mm rowwise wants scale_b shaped [1, N] for b = w.T
         self.register_buffer("weight_scale",
-                             scale.to(torch.float32).reshape(1, -1))
+                             scale.to(torch.float32).reshape(-1)
+                             .contiguous())  # [N]
         self.out_features = linear.out_features
         self.in_features = linear.in_features
+        self.out_dtype = linear.weight.dtype  # model compute dtype
+
+    def forward_q(self, x8: torch.Tensor, x_scale: torch.Tensor,
+                  out_dtype=None) -> torch.Tensor:
+        out_dtype = out_dtype or self.out_dtype
+        m = x8.shape[0]
+        if (m <= self.SKINNY_MAX_M and x8.is_cuda
+                and self.in_features % 64 == 0
+                and self.out_features % 16 == 0):
+            out = ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8,
+                                      self.weight_scale)
+            return out if out.dtype == out_dtype else out.to(out_dtype)
+        if not x8.is_cuda:
+            return ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8,
+                                       self.weight_scale).to(out_dtype)
+        # prefill-shaped: hipBLASLt scaled_mm (rowwise scales); M % 16 pad
+        pad = (-m) % 16
+        xq = x8.view(F8)
+        xs = x_scale.reshape(-1, 1)
+        if pad:
+            xq = torch.nn.functional.pad(xq.view(torch.uint8),
+                                         (0, 0, 0, pad)).view(F8)
+            xs = torch.nn.functional.pad(xs, (0, 0, 0, pad), value=1.0)
+        out = torch._scaled_mm(
+            xq, self.weight_fp8.t(), scale_a=xs.contiguous(),
+            scale_b=self.weight_scale.reshape(1, -1), out_dtype=out_dtype)
+        return out[:m] if pad else out
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         orig_shape = x.shape
         x2 = x.reshape(-1, orig_shape[-1])
-        m = x2.shape[0]
-        # hipBLASLt scaled_mm wants M a multiple of 16 and matching scale
-        # granularity on both operands (rowwise here)
-        pad = (-m) % 16
-        if pad:
-            x2 = torch.nn.functional.pad(x2, (0, 0, 0, pad))
-        amax = x2.abs().amax().clamp(min=1e-8)
-        s = (amax / F8_MAX).to(torch.float32)
-        x_scale = s.expand(x2.shape[0], 1).contiguous()
-        x8 = (x2.float() / s).to(F8)
-        out = torch._scaled_mm(
-            x8, self.weight_fp8.t(), scale_a=x_scale,
-            scale_b=self.weight_scale, out_dtype=x.dtype)
-        if pad:
-            out = out[:m]
+        x8, xs = ops.quant_fp8(x2)
+        out = self.forward_q(x8, xs, out_dtype=x.dtype)
         return out.reshape(*orig_shape[:-1], self.out_features)
 
 

@@ -150,6 +150,73 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate.dtype)
 
 
+# --------------------------------------------------------------------- #
+# fp8 (OCP e4m3) decode path: fused activation quant + skinny fp8 GEMM
+# (kernels: csrc/fp8_path.hip; replaces the round-1 un-fused quantization
+# that made fp8 slower than bf16 end to end)
+# --------------------------------------------------------------------- #
+F8_MAX = 448.0
+
+
+def _quant_fp8_ref(y: torch.Tensor):
+    """Reference per-row quantization (CPU path / numerics oracle)."""
+    y2 = y.float().reshape(-1, y.shape[-1])
+    amax = y2.abs().amax(dim=1).clamp(min=1e-8)
+    scale = (amax / F8_MAX).to(torch.float32)
+    q = (y2 / scale[:, None]).clamp(-F8_MAX, F8_MAX).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale
+
+
+def rmsnorm_fp8(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6,
+                residual: Optional[torch.Tensor] = None):
+    """rmsnorm (+fused residual update, same contract as rmsnorm()) emitting
+    fp8-e4m3 bytes + per-row scales in ONE pass over HBM."""
+    if x.is_cuda:
+        ext = _require_ext("rmsnorm_fp8")
+        if ext is not None:
+            out8, scales = ext.rmsnorm_fp8(x, weight, eps, residual)
+            return out8, scales
+    return _quant_fp8_ref(rmsnorm(x, weight, eps, residual=residual))
+
+
+def silu_mul_fp8(gate: torch.Tensor, up: torch.Tensor):
+    """silu(gate) * up -> fp8-e4m3 bytes + per-row scales, one fused pass."""
+    if gate.is_cuda:
+        ext = _require_ext("silu_mul_fp8")
+        if ext is not None:
+            out8, scales = ext.silu_mul_fp8(gate, up)
+            return out8, scales
+    return _quant_fp8_ref(silu_mul(gate, up))
+
+
+def quant_fp8(x: torch.Tensor):
+    """Per-row dynamic fp8 quantization (for activations whose producer
+    isn't one of the fused kernels, e.g. attention context)."""
+    if x.is_cuda:
+        ext = _require_ext("quant_fp8")
+        if ext is not None:
+            out8, scales = ext.quant_fp8(x)
+            return out8, scales
+    return _quant_fp8_ref(x)
+
+
+def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
+                    w8: torch.Tensor, w_scale: torch.Tensor) -> torch.Tensor:
+    """C[M,N] = (Aq*s_a) @ (Wq*s_w)^T -> bf16, decode-shaped M <= 64.
+
+    One workgroup per 16 W columns (fills the 256-CU chip at llama
+    projection N), fp8 operands halve the weight-streaming bytes vs bf16.
+    """
+    if a8.is_cuda:
+        ext = _require_ext("skinny_gemm_fp8")
+        if ext is not None:
+            return ext.skinny_gemm_fp8(
+                a8.view(torch.uint8), a_scale, w8.view(torch.uint8), w_scale)
+    a = a8.view(torch.float8_e4m3fn).float() * a_scale[:, None]
+    w = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None]
+    return (a @ w.t()).to(torch.bfloat16)
+
+
 def softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
     """Numerically-stable softmax over the last dim."""
     if x.is_cuda and dim in (-1, x.dim() - 1):

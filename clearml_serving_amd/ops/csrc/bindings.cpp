@@ -33,6 +33,13 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor k_cache,
                                       torch::Tensor q_lens, double scale);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   double theta);
+std::vector<torch::Tensor> rmsnorm_fp8(torch::Tensor x, torch::Tensor weight,
+                                       double eps,
+                                       c10::optional<torch::Tensor> residual);
+std::vector<torch::Tensor> silu_mul_fp8(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> quant_fp8(torch::Tensor x);
+torch::Tensor skinny_gemm_fp8(torch::Tensor a8, torch::Tensor a_scale,
+                              torch::Tensor w8, torch::Tensor w_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "clearml-serving-amd gfx950 kernel library";
@@ -57,4 +64,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, py::arg("a"), py::arg("w"),
         py::arg("variant") = 0);
   m.def("attention_prefill_paged", &attention_prefill_paged);
+  m.def("rmsnorm_fp8", &rmsnorm_fp8, py::arg("x"), py::arg("weight"),
+        py::arg("eps") = 1e-6, py::arg("residual") = py::none());
+  m.def("silu_mul_fp8", &silu_mul_fp8);
+  m.def("quant_fp8", &quant_fp8);
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8);
 }

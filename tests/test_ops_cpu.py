@@ -164,3 +164,85 @@ def test_bias_gelu_matches_torch():
     ref = torch.nn.functional.gelu(x + b)
     got = ops.bias_gelu(x, b)
     assert torch.allclose(got, ref, atol=1e-5)
+
+
+# ------------------------------------------------------------------ #
+# fp8 decode path (CPU reference implementations)
+# ------------------------------------------------------------------ #
+def test_quant_fp8_roundtrip():
+    from clearml_serving_amd import ops
+
+    x = torch.randn(4, 256) * 3
+    q8, s = ops.quant_fp8(x)
+    assert q8.shape == (4, 256) and q8.dtype == torch.uint8
+    deq = q8.view(torch.float8_e4m3fn).float() * s[:, None]
+    # e4m3 with per-row scale: <= ~4% relative error
+    rel = (deq - x).abs().max() / x.abs().max()
+    assert rel < 0.05, rel.item()
+
+
+def test_rmsnorm_fp8_matches_rmsnorm_plus_quant():
+    from clearml_serving_amd import ops
+
+    x = torch.randn(3, 128)
+    w = torch.rand(128) + 0.5
+    res = torch.randn(3, 128)
+    res2 = res.clone()
+    y = ops.rmsnorm(x, w, 1e-6, residual=res)
+    q_ref, s_ref = ops._quant_fp8_ref(y)
+    q, s = ops.rmsnorm_fp8(x, w, 1e-6, residual=res2)
+    torch.testing.assert_close(s, s_ref)
+    assert torch.equal(q, q_ref)
+    torch.testing.assert_close(res2, res)  # same fused residual update
+
+
+def test_silu_mul_fp8_matches_reference():
+    from clearml_serving_amd import ops
+
+    g = torch.randn(4, 256)
+    u = torch.randn(4, 256)
+    q, s = ops.silu_mul_fp8(g, u)
+    ref = torch.nn.functional.silu(g.float()) * u.float()
+    deq = q.view(torch.float8_e4m3fn).float() * s[:, None]
+    torch.testing.assert_close(deq, ref, atol=0.08, rtol=0.05)
+
+
+def test_skinny_gemm_fp8_fallback_matmul():
+    from clearml_serving_amd import ops
+
+    a = torch.randn(8, 128)
+    w = torch.randn(64, 128)
+    a8, as_ = ops.quant_fp8(a)
+    w8, ws = ops.quant_fp8(w)
+    got = ops.skinny_gemm_fp8(a8, as_, w8, ws).float()
+    ref = a @ w.t()
+    # fp8 quantization of both operands: loose tolerance, high correlation
+    corr = torch.corrcoef(torch.stack([got.flatten(), ref.flatten()]))[0, 1]
+    assert corr > 0.999, corr.item()
+
+
+def test_fp8_llama_close_to_fp32():
+    import copy
+
+    from clearml_serving_amd.models.llama import PRESETS, LlamaForCausalLM
+    from clearml_serving_amd.models.quant import Fp8Linear, quantize_llama_fp8
+
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(PRESETS["llama-tiny"]).eval().float()
+    mq = copy.deepcopy(m)
+    n = quantize_llama_fp8(mq)
+    assert n == 2 * 4  # 2 layers x 4 projections
+    assert isinstance(mq.layers[0].qkv, Fp8Linear)
+    tokens = torch.arange(10, 26, dtype=torch.long)
+    pos = torch.arange(16, dtype=torch.int32)
+    ctx = {"mode": "prefill", "batch": 1, "seq": 16,
+           "seq_lens": torch.tensor([16], dtype=torch.int32),
+           "slot_mapping": torch.full((16,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        ref = m(tokens, pos, kv_caches=None, attn_ctx=ctx)
+        got = mq(tokens, pos, kv_caches=None, attn_ctx=ctx)
+    corr = torch.corrcoef(torch.stack(
+        [ref.flatten().float(), got.flatten().float()]))[0, 1]
+    assert corr > 0.99, corr.item()
+    match = (ref.argmax(-1) == got.argmax(-1)).float().mean()
+    assert match > 0.8, match.item()

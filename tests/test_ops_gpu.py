@@ -347,3 +347,115 @@ def test_attention_causal_unequal_seq_lens():
         ri = ref[i, :n].float()
         assert torch.allclose(gi, ri, atol=2e-2, rtol=2e-2), \
             (i, (gi - ri).abs().max())
+
+
+# ------------------------------------------------------------------ #
+# fp8 decode path (kernels: csrc/fp8_path.hip)
+# ------------------------------------------------------------------ #
+def _dequant(q8, s):
+    return q8.view(torch.float8_e4m3fn).float() * s[:, None]
+
+
+def test_quant_fp8_gpu_matches_reference():
+    torch.manual_seed(0)
+    x = (torch.randn(5, 512, device=DEV) * 3).to(torch.bfloat16)
+    q, s = ops.quant_fp8(x)
+    q_ref, s_ref = ops._quant_fp8_ref(x.cpu())
+    torch.testing.assert_close(s.cpu(), s_ref, atol=1e-6, rtol=1e-4)
+    deq = _dequant(q.cpu(), s.cpu())
+    deq_ref = _dequant(q_ref, s_ref)
+    torch.testing.assert_close(deq, deq_ref, atol=0.05, rtol=0.05)
+
+
+def test_rmsnorm_fp8_gpu_matches_reference():
+    torch.manual_seed(1)
+    x = torch.randn(4, 4096, device=DEV).to(torch.bfloat16)
+    w = (torch.rand(4096, device=DEV) + 0.5).to(torch.bfloat16)
+    res = torch.randn(4, 4096, device=DEV).to(torch.bfloat16)
+    res_cpu = res.float().cpu()
+    q, s = ops.rmsnorm_fp8(x, w, 1e-5, residual=res)
+    q_ref, s_ref = ops.rmsnorm_fp8(x.float().cpu(), w.float().cpu(), 1e-5,
+                                   residual=res_cpu)
+    torch.testing.assert_close(s.cpu(), s_ref, atol=2e-3, rtol=2e-2)
+    torch.testing.assert_close(_dequant(q.cpu(), s.cpu()),
+                               _dequant(q_ref, s_ref), atol=0.08, rtol=0.08)
+    # fused residual update matches
+    torch.testing.assert_close(res.float().cpu(), res_cpu,
+                               atol=2e-2, rtol=2e-2)
+
+
+def test_silu_mul_fp8_gpu_strided_views():
+    torch.manual_seed(2)
+    inter = 1024
+    merged = torch.randn(6, 2 * inter, device=DEV).to(torch.bfloat16)
+    gate, up = merged.split([inter, inter], dim=-1)
+    q, s = ops.silu_mul_fp8(gate, up)
+    ref = (torch.nn.functional.silu(gate.float()) * up.float()).cpu()
+    deq = _dequant(q.cpu(), s.cpu())
+    torch.testing.assert_close(deq, ref, atol=0.08, rtol=0.08)
+
+
+@pytest.mark.parametrize("m", [1, 3, 16, 33, 64])
+def test_skinny_gemm_fp8_matches_dequant_matmul(m):
+    torch.manual_seed(m)
+    k, n = 256, 320
+    a = (torch.randn(m, k, device=DEV) / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=DEV) / 8).to(torch.bfloat16)
+    a8, as_ = ops.quant_fp8(a)
+    w8, ws = ops.quant_fp8(w)
+    got = ops.skinny_gemm_fp8(a8, as_, w8, ws).float().cpu()
+    ref = _dequant(a8.cpu(), as_.cpu()) @ _dequant(w8.cpu(), ws.cpu()).t()
+    assert torch.allclose(got, ref, atol=5e-2, rtol=5e-2), \
+        (got - ref).abs().max()
+
+
+def test_skinny_gemm_fp8_asymmetric_identity():
+    """A=I with asymmetric W catches any row/col-swapped fragment map."""
+    k = 64
+    a = torch.eye(16, k, device=DEV).to(torch.bfloat16)
+    w = torch.arange(32 * k, device=DEV, dtype=torch.float32) \
+        .reshape(32, k).to(torch.bfloat16) / (32 * k)
+    a8, as_ = ops.quant_fp8(a)
+    w8, ws = ops.quant_fp8(w)
+    got = ops.skinny_gemm_fp8(a8, as_, w8, ws).float().cpu()
+    ref = _dequant(a8.cpu(), as_.cpu()) @ _dequant(w8.cpu(), ws.cpu()).t()
+    assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), \
+        (got - ref).abs().max()
+
+
+def test_fp8_llama_gpu_generation():
+    """Full fp8 decode path on GPU: fused-quant kernels + fp8 skinny GEMM
+    through the engine; outputs correlate with the bf16 model."""
+    import asyncio
+
+    from clearml_serving_amd.engines.llm.engine import (
+        LlmEngine, LlmEngineConfig, SamplingParams)
+
+    def gen(quant):
+        cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
+                              block_size=16, max_model_len=128,
+                              device=DEV, quantization=quant,
+                              decode_graphs=False)
+        eng = LlmEngine(cfg)
+        eng.start()
+
+        async def go():
+            toks = []
+            async for item in eng.generate(
+                    "fp8 smoke", SamplingParams(temperature=0.0,
+                                                max_tokens=8,
+                                                ignore_eos=True)):
+                toks.extend(item["token_ids"])
+            return toks
+
+        loop = asyncio.new_event_loop()
+        out = loop.run_until_complete(go())
+        eng.stop()
+        return out
+
+    bf16 = gen(None)
+    fp8 = gen("fp8")
+    assert len(bf16) == 8 and len(fp8) == 8
+    # random-init tiny model: quantization noise may flip late tokens, but
+    # the first greedy steps should agree
+    assert bf16[0] == fp8[0]
