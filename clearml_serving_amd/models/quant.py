@@ -34,8 +34,9 @@ class Fp8Linear(nn.Module):
                               producer (rmsnorm_fp8 / silu_mul_fp8)
     """
 
-    # decode-shaped rows route to the in-tree fp8 MFMA kernel
-    SKINNY_MAX_M = 64
+    # decode-shaped rows route to the in-tree fp8 MFMA kernel; above this
+    # hipBLASLt scaled_mm wins (measured crossover, profiles/fp8_kernels)
+    SKINNY_MAX_M = int(__import__("os").environ.get("CMLS_FP8_SKINNY_MAX", 32))
 
     def __init__(self, linear: nn.Linear):
         super().__init__()

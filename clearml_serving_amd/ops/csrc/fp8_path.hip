@@ -309,35 +309,46 @@ __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
     arow[t] = a + (long)(m < M ? m : 0) * K + koff;
   }
 
-  // 2-deep unrolled: 2*(MT+1) 16-byte loads in flight, 4 MFMAs per tile
+  // 2-deep unrolled: 4*(MT+1) 8-byte loads in flight, 4 MFMAs per tile.
+  // NOTE the operand addressing: each MFMA consumes a 32-wide k-window in
+  // which THIS lane group holds bytes koff..koff+7, so the second window's
+  // operand lives at +32 bytes -- a single 16-byte load would hand the
+  // MFMA the NEXT LANE GROUP's bytes instead (measured wrong results;
+  // caught by the asymmetric-identity test).
   int k = k_lo;
   for (; k + 2 * F8_KSTEP <= k_hi; k += 2 * F8_KSTEP) {
-    longx2_t bw0 = *reinterpret_cast<const longx2_t*>(wrow + k);
-    longx2_t bw1 = *reinterpret_cast<const longx2_t*>(wrow + k + F8_KSTEP);
-    longx2_t ba0[MT], ba1[MT];
+    long bw0a = *reinterpret_cast<const long*>(wrow + k);
+    long bw0b = *reinterpret_cast<const long*>(wrow + k + 32);
+    long bw1a = *reinterpret_cast<const long*>(wrow + k + 64);
+    long bw1b = *reinterpret_cast<const long*>(wrow + k + 96);
+    long ba0a[MT], ba0b[MT], ba1a[MT], ba1b[MT];
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      ba0[t] = *reinterpret_cast<const longx2_t*>(arow[t] + k);
-      ba1[t] = *reinterpret_cast<const longx2_t*>(arow[t] + k + F8_KSTEP);
+      ba0a[t] = *reinterpret_cast<const long*>(arow[t] + k);
+      ba0b[t] = *reinterpret_cast<const long*>(arow[t] + k + 32);
+      ba1a[t] = *reinterpret_cast<const long*>(arow[t] + k + 64);
+      ba1b[t] = *reinterpret_cast<const long*>(arow[t] + k + 96);
     }
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      acc[t] = MFMA16F8(ba0[t][0], bw0[0], acc[t]);
-      acc[t] = MFMA16F8(ba0[t][1], bw0[1], acc[t]);
+      acc[t] = MFMA16F8(ba0a[t], bw0a, acc[t]);
+      acc[t] = MFMA16F8(ba0b[t], bw0b, acc[t]);
     }
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      acc[t] = MFMA16F8(ba1[t][0], bw1[0], acc[t]);
-      acc[t] = MFMA16F8(ba1[t][1], bw1[1], acc[t]);
+      acc[t] = MFMA16F8(ba1a[t], bw1a, acc[t]);
+      acc[t] = MFMA16F8(ba1b[t], bw1b, acc[t]);
     }
   }
   for (; k < k_hi; k += F8_KSTEP) {
-    longx2_t bw = *reinterpret_cast<const longx2_t*>(wrow + k);
+    long bwa = *reinterpret_cast<const long*>(wrow + k);
+    long bwb = *reinterpret_cast<const long*>(wrow + k + 32);
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      longx2_t ba = *reinterpret_cast<const longx2_t*>(arow[t] + k);
-      acc[t] = MFMA16F8(ba[0], bw[0], acc[t]);
-      acc[t] = MFMA16F8(ba[1], bw[1], acc[t]);
+      long baa = *reinterpret_cast<const long*>(arow[t] + k);
+      long bab = *reinterpret_cast<const long*>(arow[t] + k + 32);
+      acc[t] = MFMA16F8(baa, bwa, acc[t]);
+      acc[t] = MFMA16F8(bab, bwb, acc[t]);
     }
   }
 
