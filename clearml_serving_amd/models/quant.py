@@ -43,8 +43,12 @@ class Fp8Linear(nn.Module):
         w = linear.weight.data.float()
         row_max = w.abs().amax(dim=1, keepdim=True).clamp(min=1e-8)
         scale = row_max / F8_MAX
-        self.register_buffer("weight_fp8",
-                             (w / scale).to(F8).contiguous())
+        w8 = (w / scale).to(F8).contiguous()
+        self.register_buffer("weight_fp8", w8)
+        # decode kernel reads a byte-swizzled copy (one 16-B lane load =
+        # both MFMA operands of a 64-wide k pair); prefill scaled_mm needs
+        # the plain layout, so both live in HBM (fp8: still half of bf16)
+        self.register_buffer("weight_fp8_sw", ops.swizzle_fp8_weight(w8))
         self.register_buffer("weight_scale",
                              scale.to(torch.float32).reshape(-1)
                              .contiguous())  # [N]
@@ -59,8 +63,8 @@ class Fp8Linear(nn.Module):
         if (m <= self.SKINNY_MAX_M and x8.is_cuda
                 and self.in_features % 64 == 0
                 and self.out_features % 16 == 0):
-            out = ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8,
-                                      self.weight_scale)
+            out = ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8_sw,
+                                      self.weight_scale, swizzled=True)
             return out if out.dtype == out_dtype else out.to(out_dtype)
         if not x8.is_cuda:
             return ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8,

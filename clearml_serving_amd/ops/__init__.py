@@ -200,18 +200,41 @@ def quant_fp8(x: torch.Tensor):
     return _quant_fp8_ref(x)
 
 
+def swizzle_fp8_weight(w8: torch.Tensor) -> torch.Tensor:
+    """Reorder an fp8 [N, K] weight for skinny_gemm_fp8: within each
+    64-byte k window-pair, bytes become [lane-group][window][8] so one
+    16-byte lane load yields both MFMA operands of the pair (un-swizzled
+    weights need two 8-byte loads -- measured request-rate bound)."""
+    n, k = w8.shape
+    assert k % 64 == 0
+    v = w8.view(torch.uint8).reshape(n, k // 64, 2, 4, 8)
+    return v.permute(0, 1, 3, 2, 4).contiguous().reshape(n, k)
+
+
+def unswizzle_fp8_weight(w8s: torch.Tensor) -> torch.Tensor:
+    n, k = w8s.shape
+    v = w8s.view(torch.uint8).reshape(n, k // 64, 4, 2, 8)
+    return v.permute(0, 1, 3, 2, 4).contiguous().reshape(n, k)
+
+
 def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
-                    w8: torch.Tensor, w_scale: torch.Tensor) -> torch.Tensor:
+                    w8: torch.Tensor, w_scale: torch.Tensor,
+                    swizzled: bool = False) -> torch.Tensor:
     """C[M,N] = (Aq*s_a) @ (Wq*s_w)^T -> bf16, decode-shaped M <= 64.
 
     One workgroup per 16 W columns (fills the 256-CU chip at llama
     projection N), fp8 operands halve the weight-streaming bytes vs bf16.
+    ``swizzled``: w8 is already in swizzle_fp8_weight layout (Fp8Linear
+    prepares it once at quantization time).
     """
     if a8.is_cuda:
         ext = _require_ext("skinny_gemm_fp8")
         if ext is not None:
+            ws = w8 if swizzled else swizzle_fp8_weight(w8)
             return ext.skinny_gemm_fp8(
-                a8.view(torch.uint8), a_scale, w8.view(torch.uint8), w_scale)
+                a8.view(torch.uint8), a_scale, ws.view(torch.uint8), w_scale)
+    if swizzled:
+        w8 = unswizzle_fp8_weight(w8)
     a = a8.view(torch.float8_e4m3fn).float() * a_scale[:, None]
     w = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None]
     return (a @ w.t()).to(torch.bfloat16)

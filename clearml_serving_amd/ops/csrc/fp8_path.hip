@@ -283,7 +283,7 @@ template <int MT>  // 16-row m tiles: M <= 16*MT (MT 1..4 -> M <= 64)
 __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
     const unsigned char* __restrict__ a,  // [M, K] e4m3 row-major
     const float* __restrict__ a_scale,    // [M]
-    const unsigned char* __restrict__ w,  // [N, K] e4m3 row-major
+    const unsigned char* __restrict__ w,  // [N, K] e4m3 SWIZZLED (see note)
     const float* __restrict__ w_scale,    // [N]
     __hip_bfloat16* __restrict__ c,       // [M, N]
     int M, int N, int K) {
@@ -291,7 +291,7 @@ __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int row = lane & 15;
-  const int koff = (lane >> 4) * 8;
+  const int g = lane >> 4;  // lane group (k fragment)
 
   const int kchunk = ((K / F8_KSTEP + F8_NW - 1) / F8_NW) * F8_KSTEP;
   const int k_lo = wave * kchunk;
@@ -301,26 +301,27 @@ __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
 #pragma unroll
   for (int t = 0; t < MT; ++t) acc[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-  const unsigned char* wrow = w + (long)(n0 + row) * K + koff;
+  // W layout is PRE-SWIZZLED at quantization time (models/quant.py
+  // swizzle_fp8_weight): within each 64-byte k window-pair, the bytes are
+  // reordered [group][window][8] so ONE 16-byte lane load yields both MFMA
+  // operands of the pair. An un-swizzled row would force two 8-byte loads
+  // per pair (one per 32-wide k window: this lane group's bytes live at
+  // +8g and +32+8g), doubling the HBM request rate -- measured to erase
+  // the fp8 byte advantage entirely (67.7us vs bf16's 69.6us at M=1).
+  const unsigned char* wrow = w + (long)(n0 + row) * K + 16 * g;
+  // A stays row-major (L2-resident; every WG reads the same M*K tile):
+  // its two operands per pair load as 2x8 bytes at +8g and +32+8g.
   const unsigned char* arow[MT];
 #pragma unroll
   for (int t = 0; t < MT; ++t) {
     const int m = t * 16 + row;
-    arow[t] = a + (long)(m < M ? m : 0) * K + koff;
+    arow[t] = a + (long)(m < M ? m : 0) * K + 8 * g;
   }
 
-  // 2-deep unrolled: 4*(MT+1) 8-byte loads in flight, 4 MFMAs per tile.
-  // NOTE the operand addressing: each MFMA consumes a 32-wide k-window in
-  // which THIS lane group holds bytes koff..koff+7, so the second window's
-  // operand lives at +32 bytes -- a single 16-byte load would hand the
-  // MFMA the NEXT LANE GROUP's bytes instead (measured wrong results;
-  // caught by the asymmetric-identity test).
   int k = k_lo;
   for (; k + 2 * F8_KSTEP <= k_hi; k += 2 * F8_KSTEP) {
-    long bw0a = *reinterpret_cast<const long*>(wrow + k);
-    long bw0b = *reinterpret_cast<const long*>(wrow + k + 32);
-    long bw1a = *reinterpret_cast<const long*>(wrow + k + 64);
-    long bw1b = *reinterpret_cast<const long*>(wrow + k + 96);
+    longx2_t bw0 = *reinterpret_cast<const longx2_t*>(wrow + k);
+    longx2_t bw1 = *reinterpret_cast<const longx2_t*>(wrow + k + 64);
     long ba0a[MT], ba0b[MT], ba1a[MT], ba1b[MT];
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
@@ -331,24 +332,23 @@ __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
     }
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      acc[t] = MFMA16F8(ba0a[t], bw0a, acc[t]);
-      acc[t] = MFMA16F8(ba0b[t], bw0b, acc[t]);
+      acc[t] = MFMA16F8(ba0a[t], bw0[0], acc[t]);
+      acc[t] = MFMA16F8(ba0b[t], bw0[1], acc[t]);
     }
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      acc[t] = MFMA16F8(ba1a[t], bw1a, acc[t]);
-      acc[t] = MFMA16F8(ba1b[t], bw1b, acc[t]);
+      acc[t] = MFMA16F8(ba1a[t], bw1[0], acc[t]);
+      acc[t] = MFMA16F8(ba1b[t], bw1[1], acc[t]);
     }
   }
   for (; k < k_hi; k += F8_KSTEP) {
-    long bwa = *reinterpret_cast<const long*>(wrow + k);
-    long bwb = *reinterpret_cast<const long*>(wrow + k + 32);
+    longx2_t bw = *reinterpret_cast<const longx2_t*>(wrow + k);
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
       long baa = *reinterpret_cast<const long*>(arow[t] + k);
       long bab = *reinterpret_cast<const long*>(arow[t] + k + 32);
-      acc[t] = MFMA16F8(baa, bwa, acc[t]);
-      acc[t] = MFMA16F8(bab, bwb, acc[t]);
+      acc[t] = MFMA16F8(baa, bw[0], acc[t]);
+      acc[t] = MFMA16F8(bab, bw[1], acc[t]);
     }
   }
 

@@ -246,3 +246,18 @@ def test_fp8_llama_close_to_fp32():
     assert corr > 0.99, corr.item()
     match = (ref.argmax(-1) == got.argmax(-1)).float().mean()
     assert match > 0.8, match.item()
+
+
+def test_fp8_weight_swizzle_roundtrip():
+    from clearml_serving_amd import ops
+
+    w = torch.randint(0, 255, (3, 128), dtype=torch.uint8)
+    assert torch.equal(ops.unswizzle_fp8_weight(ops.swizzle_fp8_weight(w)), w)
+    # swizzled CPU fallback equals plain
+    a = torch.randn(2, 128)
+    a8, as_ = ops.quant_fp8(a)
+    wq, ws = ops.quant_fp8(torch.randn(32, 128))
+    plain = ops.skinny_gemm_fp8(a8, as_, wq, ws)
+    sw = ops.skinny_gemm_fp8(a8, as_, ops.swizzle_fp8_weight(wq), ws,
+                             swizzled=True)
+    torch.testing.assert_close(plain, sw)
