@@ -36,7 +36,7 @@ class Fp8Linear(nn.Module):
 
     # decode-shaped rows route to the in-tree fp8 MFMA kernel; above this
     # hipBLASLt scaled_mm wins (measured crossover, profiles/fp8_kernels)
-    SKINNY_MAX_M = int(__import__("os").environ.get("CMLS_FP8_SKINNY_MAX", 32))
+    SKINNY_MAX_M = int(__import__("os").environ.get("CMLS_FP8_SKINNY_MAX", 16))
 
     def __init__(self, linear: nn.Linear):
         super().__init__()

@@ -25,8 +25,9 @@ for M in (1, 16, 32, 64):
         a = (torch.randn(M, K, device="cuda") / 8).to(torch.bfloat16)
         w = (torch.randn(N, K, device="cuda") / 8).to(torch.bfloat16)
         a8, as_ = ops.quant_fp8(a); w8, ws = ops.quant_fp8(w)
+        wsw = ops.swizzle_fp8_weight(w8)
         ext = ops._require_ext("skinny_gemm_fp8")
-        us8 = t(lambda: ext.skinny_gemm_fp8(a8, as_, w8, ws))
+        us8 = t(lambda: ext.skinny_gemm_fp8(a8, as_, wsw, ws))
         usb = t(lambda: ops.skinny_linear(a, w))
         uslt = t(lambda: torch.nn.functional.linear(a, w))
         gb8 = (N*K + M*K) / us8 / 1e3   # fp8 bytes
