@@ -31,7 +31,16 @@ class FoldedConv(nn.Module):
         self.relu = relu
 
     def forward(self, x):
-        y = self.conv(x)
+        # 3x3 s1 p1 bottleneck convs at the ResNet-50 widths run on the
+        # in-tree implicit-GEMM MFMA kernel (ops/csrc/conv3x3.hip) with the
+        # bias+relu epilogue fused; everything else stays MIOpen (routed
+        # where each MEASURES faster -- same discipline as skinny_linear)
+        c = self.conv
+        if (c.kernel_size == (3, 3) and c.stride == (1, 1)
+                and c.padding == (1, 1)
+                and ops.conv3x3_supported(x, c.weight)):
+            return ops.conv3x3_nhwc(x, c.weight, c.bias, relu=self.relu)
+        y = c(x)
         if self.relu:
             y = F.relu(y)
         return y

@@ -240,6 +240,41 @@ def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
     return (a @ w.t()).to(torch.bfloat16)
 
 
+def conv3x3_supported(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """True when the in-tree conv kernel serves this shape (3x3 s1 p1,
+    ResNet-50 bottleneck widths, bf16 GPU)."""
+    return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4
+            and x.shape[2] == x.shape[3] and x.shape[2] in (56, 28, 14)
+            and x.shape[1] % 32 == 0 and w.shape[0] % 64 == 0
+            and w.shape[2] == 3 and w.shape[3] == 3
+            and os.environ.get("CMLS_CONV3", "1") != "0"
+            and _ext is not None)
+
+
+def conv3x3_nhwc(x: torch.Tensor, w: torch.Tensor,
+                 bias: Optional[torch.Tensor] = None, relu: bool = False,
+                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """3x3 stride-1 pad-1 NHWC bf16 conv (implicit GEMM, MFMA) for the
+    ResNet-50 bottleneck widths 56/28/14; optional fused bias+residual+relu
+    epilogue. Kernel: csrc/conv3x3.hip. CPU/odd shapes: torch reference."""
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and x.shape[2] == x.shape[3] and x.shape[2] in (56, 28, 14)
+            and x.shape[1] % 32 == 0 and w.shape[0] % 64 == 0):
+        ext = _require_ext("conv3x3_nhwc")
+        if ext is not None:
+            xc = x.contiguous(memory_format=torch.channels_last)
+            wc = w.contiguous(memory_format=torch.channels_last)
+            return ext.conv3x3_nhwc(xc, wc, bias, relu, residual)
+    y = torch.nn.functional.conv2d(x.float(), w.float(),
+                                   bias.float() if bias is not None else None,
+                                   stride=1, padding=1)
+    if residual is not None:
+        y = y + residual.float()
+    if relu:
+        y = torch.relu(y)
+    return y.to(x.dtype)
+
+
 def softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
     """Numerically-stable softmax over the last dim."""
     if x.is_cuda and dim in (-1, x.dim() - 1):

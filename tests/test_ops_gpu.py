@@ -459,3 +459,63 @@ def test_fp8_llama_gpu_generation():
     # random-init tiny model: quantization noise may flip late tokens, but
     # the first greedy steps should agree
     assert bf16[0] == fp8[0]
+
+
+# ------------------------------------------------------------------ #
+# hand-written 3x3 conv (csrc/conv3x3.hip)
+# ------------------------------------------------------------------ #
+@pytest.mark.parametrize("n,c,k,w", [
+    (2, 64, 64, 56),    # conv2-class
+    (2, 128, 128, 28),  # conv3-class
+    (3, 256, 256, 14),  # conv4-class
+    (2, 96, 128, 28),   # C not equal K
+])
+def test_conv3x3_matches_fp32_reference(n, c, k, w):
+    torch.manual_seed(n * 100 + w)
+    x = (torch.randn(n, c, w, w, device=DEV) / 4).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    wt = (torch.randn(k, c, 3, 3, device=DEV) / 8).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    b = torch.randn(k, device=DEV).to(torch.bfloat16)
+    ext = ops._require_ext("conv3x3_nhwc")
+    got = ext.conv3x3_nhwc(x, wt, b, False, None).float().cpu()
+    ref = torch.nn.functional.conv2d(
+        x.float().cpu(), wt.float().cpu(), b.float().cpu(),
+        stride=1, padding=1)
+    assert torch.allclose(got, ref, atol=6e-2, rtol=6e-2), \
+        (got - ref).abs().max()
+
+
+def test_conv3x3_fused_relu_residual():
+    torch.manual_seed(7)
+    n, c, k, w = 2, 64, 64, 56
+    x = (torch.randn(n, c, w, w, device=DEV) / 4).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    wt = (torch.randn(k, c, 3, 3, device=DEV) / 8).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    b = torch.randn(k, device=DEV).to(torch.bfloat16)
+    res = (torch.randn(n, k, w, w, device=DEV) / 4).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    ext = ops._require_ext("conv3x3_nhwc")
+    got = ext.conv3x3_nhwc(x, wt, b, True, res).float().cpu()
+    ref = torch.relu(torch.nn.functional.conv2d(
+        x.float().cpu(), wt.float().cpu(), b.float().cpu(), stride=1,
+        padding=1) + res.float().cpu())
+    assert torch.allclose(got, ref, atol=6e-2, rtol=6e-2), \
+        (got - ref).abs().max()
+
+
+def test_conv3x3_asymmetric_identity():
+    """Delta-filter with asymmetric weights: catches tap/row/col mixups."""
+    n, c, k, w = 1, 32, 64, 14
+    x = torch.zeros(n, c, w, w, device=DEV, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    x[0, 3, 5, 9] = 1.0  # single impulse
+    wt = torch.zeros(k, c, 3, 3, device=DEV, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    wt[7, 3, 0, 2] = 2.0  # tap (dy=0, dx=2)
+    ext = ops._require_ext("conv3x3_nhwc")
+    got = ext.conv3x3_nhwc(x, wt, None, False, None).float().cpu()
+    ref = torch.nn.functional.conv2d(x.float().cpu(), wt.float().cpu(),
+                                     stride=1, padding=1)
+    assert torch.allclose(got, ref, atol=1e-3), (got - ref).abs().max()
