@@ -46,6 +46,41 @@ class FoldedConv(nn.Module):
         return y
 
 
+class StemConv(nn.Module):
+    """ResNet stem (7x7 s2 p3, 3->64) via explicit im2col + hipBLASLt GEMM.
+
+    MIOpen has no tuned C=3 NHWC bf16 kernel on gfx950 and falls back to
+    ``naive_conv_ab_nonpacked_fwd_nhwc`` at ~2 ms per batch-64 call --
+    measured 52.9% of ALL serving-bench GPU time (profiles round 2 rocprof).
+    unfold -> [N*112*112, 147] @ [147, 64] runs the same math in ~0.1 ms,
+    and the [N, L, 64] GEMM output IS the NHWC layout, so no transpose."""
+
+    def __init__(self, cin, cout, k, stride, padding, relu=True):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=padding,
+                              bias=True)
+        self.relu = relu
+        self.k, self.stride, self.padding = k, stride, padding
+
+    def forward(self, x):
+        if not x.is_cuda:
+            y = self.conv(x)
+            return F.relu(y) if self.relu else y
+        n, c, h, w = x.shape
+        oh = (h + 2 * self.padding - self.k) // self.stride + 1
+        ow = (w + 2 * self.padding - self.k) // self.stride + 1
+        cols = F.unfold(x.contiguous(), self.k, stride=self.stride,
+                        padding=self.padding)          # [N, C*k*k, L]
+        flat = cols.transpose(1, 2).reshape(-1, c * self.k * self.k)
+        wmat = self.conv.weight.reshape(self.conv.out_channels, -1)
+        y = torch.nn.functional.linear(flat, wmat, self.conv.bias)
+        if self.relu:
+            y = torch.relu(y)
+        # [N, L, cout] == NHWC memory: expose as channels_last NCHW view
+        return y.view(n, oh, ow, self.conv.out_channels) \
+            .permute(0, 3, 1, 2)
+
+
 def fold_bn_into_conv(conv: nn.Conv2d, bn: nn.BatchNorm2d) -> None:
     """Fold BN(scale, shift, mean, var) into conv weight/bias in place --
     used when importing an externally trained conv+BN checkpoint."""
@@ -81,7 +116,7 @@ class Bottleneck(nn.Module):
 class ResNet(nn.Module):
     def __init__(self, layers=(3, 4, 6, 3), num_classes=1000):
         super().__init__()
-        self.stem = FoldedConv(3, 64, 7, stride=2, padding=3, relu=True)
+        self.stem = StemConv(3, 64, 7, stride=2, padding=3, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.inplanes = 64
         self.layer1 = self._make_layer(64, layers[0], stride=1)

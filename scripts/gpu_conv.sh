@@ -36,6 +36,17 @@ for name, n, c, k, w in SHAPES:
           f"MIOpen+relu {us_mi:7.1f}us {flops/us_mi/1e6:6.0f}TF | "
           f"{us_mi/us_ours:.2f}x", flush=True)
 
+# stem A/B: im2col+GEMM vs MIOpen (naive fallback at C=3 NHWC)
+from clearml_serving_amd.models.resnet import StemConv
+stem = StemConv(3, 64, 7, stride=2, padding=3).cuda().to(torch.bfloat16)
+xs = torch.randn(64, 3, 224, 224, device="cuda", dtype=torch.bfloat16) \
+    .contiguous(memory_format=torch.channels_last)
+with torch.inference_mode():
+    us_stem = t(lambda: stem(xs), iters=50)
+    us_mi_stem = t(lambda: torch.relu(torch.nn.functional.conv2d(
+        xs, stem.conv.weight, stem.conv.bias, stride=2, padding=3)), iters=50)
+print(f"stem 7x7s2 C=3 b64: im2col+GEMM {us_stem:7.1f}us | MIOpen {us_mi_stem:7.1f}us | {us_mi_stem/us_stem:.2f}x", flush=True)
+
 # end-to-end ResNet-50 b64
 from clearml_serving_amd.models import build_model
 m = build_model({"arch": "resnet50", "num_classes": 1000,

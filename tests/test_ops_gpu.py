@@ -519,3 +519,24 @@ def test_conv3x3_asymmetric_identity():
     ref = torch.nn.functional.conv2d(x.float().cpu(), wt.float().cpu(),
                                      stride=1, padding=1)
     assert torch.allclose(got, ref, atol=1e-3), (got - ref).abs().max()
+
+
+def test_resnet_stem_im2col_matches_conv2d():
+    """StemConv (im2col+GEMM; avoids MIOpen's naive C=3 NHWC fallback)
+    matches the conv2d reference."""
+    from clearml_serving_amd.models.resnet import StemConv
+
+    torch.manual_seed(0)
+    stem = StemConv(3, 64, 7, stride=2, padding=3, relu=True)
+    stem = stem.to(DEV).to(torch.bfloat16)
+    x = (torch.randn(4, 3, 224, 224, device=DEV) / 4).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    with torch.inference_mode():
+        got = stem(x).float()
+        ref = torch.relu(torch.nn.functional.conv2d(
+            x.float(), stem.conv.weight.float(), stem.conv.bias.float(),
+            stride=2, padding=3))
+    assert got.shape == ref.shape == (4, 64, 112, 112)
+    assert got.is_contiguous(memory_format=torch.channels_last)
+    assert torch.allclose(got, ref, atol=6e-2, rtol=6e-2), \
+        (got - ref).abs().max()
