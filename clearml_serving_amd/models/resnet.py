@@ -49,11 +49,12 @@ class FoldedConv(nn.Module):
 class StemConv(nn.Module):
     """ResNet stem (7x7 s2 p3, 3->64) via explicit im2col + hipBLASLt GEMM.
 
-    MIOpen has no tuned C=3 NHWC bf16 kernel on gfx950 and falls back to
-    ``naive_conv_ab_nonpacked_fwd_nhwc`` at ~2 ms per batch-64 call --
-    measured 52.9% of ALL serving-bench GPU time (profiles round 2 rocprof).
-    unfold -> [N*112*112, 147] @ [147, 64] runs the same math in ~0.1 ms,
-    and the [N, L, 64] GEMM output IS the NHWC layout, so no transpose."""
+    NOT ROUTED: measured 1457us vs MIOpen's 195us at batch 64 (the unfold
+    materializes 236 MB and the K=147 GEMM is inefficient). Kept as the
+    documented dead end: a short rocprof trace showed a naive MIOpen conv
+    at 52.9% of GPU time, but that was MIOpen's FIND phase enumerating
+    algorithms during warmup, not steady state -- e2e ResNet regressed
+    2.80 -> 4.07 ms with this stem, so FoldedConv/MIOpen stays."""
 
     def __init__(self, cin, cout, k, stride, padding, relu=True):
         super().__init__()
@@ -116,7 +117,7 @@ class Bottleneck(nn.Module):
 class ResNet(nn.Module):
     def __init__(self, layers=(3, 4, 6, 3), num_classes=1000):
         super().__init__()
-        self.stem = StemConv(3, 64, 7, stride=2, padding=3, relu=True)
+        self.stem = FoldedConv(3, 64, 7, stride=2, padding=3, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.inplanes = 64
         self.layer1 = self._make_layer(64, layers[0], stride=1)
