@@ -511,9 +511,14 @@ def gemm_bf16(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """C[M,N] = A[M,K] @ B[N,K]^T in bf16 with fp32 accumulation on MFMA
     (nn.Linear orientation: b is a [out, in] weight matrix).
 
-    The in-tree MFMA GEMM of the kernel library; library GEMMs in the
-    serving models go through torch.matmul (hipBLASLt) where that wins.
-    Shapes off the 128/128/32 tile grid are zero-padded here.
+    STATUS: test fixture / benchmark exhibit, NOT routed in any serving
+    model. Measured 867 TF vs hipBLASLt's 1362 TF on large dense shapes
+    (profiles/README.md §4), so production GEMMs go through torch.matmul
+    (hipBLASLt); the decode-shaped regime where hand-written kernels DO win
+    is covered by skinny_gemm / skinny_gemm_fp8 / conv3x3_nhwc, all routed.
+    Kept (with tests + kernel_bench entry) as the dense-tile reference the
+    conv kernel's tiling derives from. Shapes off the 128/128/32 tile grid
+    are zero-padded here.
     """
     if a.is_cuda:
         ext = _require_ext("gemm_bf16")

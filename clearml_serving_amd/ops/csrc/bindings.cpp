@@ -43,6 +43,14 @@ torch::Tensor skinny_gemm_fp8(torch::Tensor a8, torch::Tensor a_scale,
 torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w,
                            c10::optional<torch::Tensor> bias, bool relu,
                            c10::optional<torch::Tensor> residual);
+torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
+                                   torch::Tensor v, bool causal, double scale,
+                                   c10::optional<torch::Tensor> seq_lens,
+                                   bool bshd);
+torch::Tensor attention_prefill_paged_v2(
+    torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
+    torch::Tensor block_table, torch::Tensor kv_lens, torch::Tensor q_lens,
+    double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "clearml-serving-amd gfx950 kernel library";
@@ -75,4 +83,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_nhwc", &conv3x3_nhwc, py::arg("x"), py::arg("w"),
         py::arg("bias") = py::none(), py::arg("relu") = false,
         py::arg("residual") = py::none());
+  m.def("attention_prefill_v2", &attention_prefill_v2, py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
+        py::arg("seq_lens") = py::none(), py::arg("bshd") = false);
+  m.def("attention_prefill_paged_v2", &attention_prefill_paged_v2);
 }

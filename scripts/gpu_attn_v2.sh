@@ -1,0 +1,69 @@
+#!/bin/bash
+# Attention prefill v2 (swapped-QK^T in-register softmax) A/B vs v1:
+# numerics against the fp32 reference, then TF at BERT + llama shapes.
+set -x
+mkdir -p gpurun_out
+cat > /tmp/attn_v2_ab.py <<'PY'
+import sys, time, torch
+sys.path.insert(0, ".")
+from clearml_serving_amd import ops
+ext = ops._require_ext("attention_prefill_v2")
+
+def ref_attn(q, k, v, causal, seq_lens=None):
+    return ops.attention(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                         causal=causal,
+                         seq_lens=seq_lens.cpu() if seq_lens is not None
+                         else None)
+
+# ---- numerics ---- #
+torch.manual_seed(0)
+ok = True
+for (b, h, hkv, s, d, causal, use_sl) in [
+        (2, 4, 4, 128, 64, False, False),
+        (2, 4, 2, 200, 64, True, True),
+        (1, 8, 8, 384, 128, True, False),
+        (2, 2, 2, 64, 128, False, True)]:
+    q = (torch.randn(b, h, s, d, device="cuda") / 4).to(torch.bfloat16)
+    k = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
+    v = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
+    sl = (torch.tensor([s, s // 2][:b][:b], dtype=torch.int32, device="cuda")
+          if use_sl else None)
+    got = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False)
+    want = ref_attn(q, k, v, causal, sl)
+    # compare only valid rows (pad rows are garbage in both)
+    m = (got.float().cpu() - want).abs()
+    if sl is not None:
+        for i in range(b):
+            m[i, :, sl[i]:, :] = 0
+    bad = m.max().item()
+    print(f"numerics b{b} h{h}/{hkv} s{s} d{d} causal={causal} sl={use_sl}: "
+          f"maxerr {bad:.4f}", flush=True)
+    ok &= bad < 4e-2
+print("NUMERICS", "OK" if ok else "FAIL", flush=True)
+
+# ---- perf ---- #
+def t(fn, iters=50):
+    for _ in range(10): fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters): fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+ext1 = ops._require_ext("attention_prefill")
+for name, (b, h, hkv, s, d, causal) in [
+        ("bert-b64", (64, 12, 12, 128, 64, False)),
+        ("llama8b-s2048", (4, 32, 8, 2048, 128, True)),
+        ("llama8b-s4096", (1, 32, 8, 4096, 128, True)),
+        ("noncausal-s2048", (4, 32, 8, 2048, 128, False))]:
+    q = (torch.randn(b, h, s, d, device="cuda") / 4).to(torch.bfloat16)
+    k = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
+    v = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
+    t1 = t(lambda: ext1.attention_prefill(q, k, v, causal, d ** -0.5, None, False))
+    t2 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False))
+    fl = 4.0 * b * h * s * s * d * (0.5 if causal else 1.0)
+    print(f"{name}: v1 {t1*1e6:8.1f}us {fl/t1/1e12:6.0f}TF | "
+          f"v2 {t2*1e6:8.1f}us {fl/t2/1e12:6.0f}TF | {t1/t2:.2f}x", flush=True)
+PY
+timeout 600 python /tmp/attn_v2_ab.py 2>&1 | grep -v Warn | tee gpurun_out/attn_v2_ab.txt
+echo ATTNV2DONE
