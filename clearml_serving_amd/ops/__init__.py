@@ -306,6 +306,12 @@ def attention(
     if q.is_cuda:
         ext = _require_ext("attention")
         if ext is not None:
+            # v2 (swapped-QK^T in-register softmax) measures 1.10-1.18x v1
+            # across BERT/llama shapes (profiles/attn_v2_ab.txt);
+            # CMLS_ATTN_V2=0 falls back to v1
+            if os.environ.get("CMLS_ATTN_V2", "1") != "0":
+                return ext.attention_prefill_v2(q, k, v, bool(causal),
+                                                float(scale), seq_lens, bshd)
             return ext.attention_prefill(q, k, v, bool(causal), float(scale),
                                          seq_lens, bshd)
     if bshd:
@@ -347,6 +353,10 @@ def attention_prefill_paged(
     if q.is_cuda:
         ext = _require_ext("attention_prefill_paged")
         if ext is not None:
+            if os.environ.get("CMLS_ATTN_V2", "1") != "0":
+                return ext.attention_prefill_paged_v2(
+                    q, k_cache, v_cache, block_table, kv_lens, q_lens,
+                    float(scale))
             return ext.attention_prefill_paged(
                 q, k_cache, v_cache, block_table, kv_lens, q_lens,
                 float(scale))
