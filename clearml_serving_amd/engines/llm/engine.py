@@ -583,14 +583,40 @@ class LlmEngine:
     def _prefill_chunk(self, seqs: List["Sequence"], budget: int) -> None:
         """Advance several sequences' prefill by up to ``budget`` tokens
         total in ONE batched paged-attention forward; samples first tokens
-        for the sequences whose chunk completes the prompt."""
+        for the sequences whose chunk completes the prompt.
+
+        Fair-share: the budget splits evenly across chunking prompts (then
+        leftovers go FCFS), so a newly arrived long prompt advances every
+        step instead of starving behind an earlier prompt's remaining
+        chunks (round-1 packed strictly FCFS)."""
         batch = []
-        for s in seqs:
-            if budget <= 0:
-                break
-            chunk = min(budget, len(s.prompt_ids) - s.prefilled)
-            batch.append((s, chunk))
-            budget -= chunk
+        if len(seqs) > 1:
+            share = max(budget // len(seqs), 1)
+            remaining = budget
+            needs = []
+            for s in seqs:
+                if remaining <= 0:
+                    break
+                need = len(s.prompt_ids) - s.prefilled
+                take = min(share, need, remaining)
+                batch.append((s, take))
+                needs.append(need - take)
+                remaining -= take
+            # leftovers (some prompts needed < share) go FCFS
+            for i, (s, take) in enumerate(batch):
+                if remaining <= 0:
+                    break
+                extra = min(needs[i], remaining)
+                if extra:
+                    batch[i] = (s, take + extra)
+                    remaining -= extra
+        else:
+            for s in seqs:
+                if budget <= 0:
+                    break
+                chunk = min(budget, len(s.prompt_ids) - s.prefilled)
+                batch.append((s, chunk))
+                budget -= chunk
         if not batch:
             return
         self.stats["prefill_batches"] += 1

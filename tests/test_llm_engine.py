@@ -725,3 +725,40 @@ def test_engine_cache_key_distinguishes_aux_only_endpoints(tmp_path):
         LlmPreprocessRequest._engines.update(saved_engines)
         LlmPreprocessRequest._engine_refs.clear()
         LlmPreprocessRequest._engine_refs.update(saved_refs)
+
+
+def test_chunked_prefill_fair_share():
+    """Two long prompts chunking together each advance every step (the
+    budget splits; round-1 FCFS packing starved the later prompt until the
+    earlier finished)."""
+    eng = tiny_engine(prefill_chunk=32, max_prefill_tokens=32,
+                      max_num_seqs=8)
+
+    progress = []
+    orig = eng._exec_chunk
+
+    def spy(plan):
+        progress.append([len(t) for t in plan["tokens"]])
+        return orig(plan)
+
+    eng._exec_chunk = spy
+
+    async def main():
+        pa = [(7 * i + 1) % 500 for i in range(96)]
+        pb = [(11 * i + 3) % 500 for i in range(96)]
+        params = SamplingParams(temperature=0.0, max_tokens=2,
+                                ignore_eos=True)
+        sa = await eng.add_request(pa, params)
+        sb = await eng.add_request(pb, params)
+        for s in (sa, sb):
+            while True:
+                item = await asyncio.wait_for(s.stream.get(), timeout=30)
+                if item.get("finished"):
+                    break
+
+    run(main())
+    # steps where both prompts were mid-prefill must split the budget
+    both = [p for p in progress if len(p) == 2]
+    assert both, progress
+    for chunks in both:
+        assert chunks[0] == chunks[1] == 16, (chunks, progress)
