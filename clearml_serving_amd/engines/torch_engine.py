@@ -105,6 +105,7 @@ class HipPreprocessRequest(BasePreprocessRequest):
             use_graphs=bool(aux.get("use_graphs", True)),
             dtype=self.dtype,
             name=model_endpoint.serving_url,
+            out_convert=_to_numpy,  # whole-batch bf16 -> float numpy, once
         )
         if aux.get("warmup", True):
             sample = self._sample_from_spec()
@@ -199,8 +200,9 @@ class HipPreprocessRequest(BasePreprocessRequest):
             inputs = {k: self._to_tensor(v) for k, v in data.items()}
         else:
             inputs = self._to_tensor(data)
-        out = await self._batcher.submit(inputs)
-        return _to_numpy(out)
+        # the batcher's out_convert already produced numpy (whole batch at
+        # once); this request's slice is a view into that fresh array
+        return await self._batcher.submit(inputs)
 
 
 # alias for reference-CLI compatibility: `model add --engine triton` serves

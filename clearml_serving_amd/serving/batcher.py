@@ -95,6 +95,7 @@ class DynamicBatcher:
         use_graphs: bool = True,
         dtype: Optional[torch.dtype] = None,
         name: str = "endpoint",
+        out_convert: Optional[Callable] = None,
     ):
         self._model_fn = model_fn
         self.device = torch.device(device)
@@ -108,6 +109,11 @@ class DynamicBatcher:
         self.use_graphs = bool(use_graphs) and self.is_cuda
         self.dtype = dtype
         self.name = name
+        # applied ONCE to the whole batched output in the worker thread
+        # (e.g. bf16 -> float numpy); per-request conversion was 512 small
+        # torch ops per bench step on the event loop. When set, the split
+        # returns views into the converted (freshly materialized) batch.
+        self.out_convert = out_convert
 
         self._queue: "asyncio.Queue" = None  # created lazily on the loop
         self._loop: Optional[asyncio.AbstractEventLoop] = None
@@ -246,6 +252,8 @@ class DynamicBatcher:
 
             with self._exec_lock:
                 out_cpu = _to_cpu(self._run_model(_stage_cpu(), bucket))
+            if self.out_convert is not None:
+                out_cpu = self.out_convert(out_cpu)
             return [_slice(out_cpu, i) for i in range(n)]
 
         slot = self._slots.get()  # bounds in-flight batches (2)
@@ -272,6 +280,10 @@ class DynamicBatcher:
             t_done = time.monotonic()
             self.stats["stage_ms_sum"] += (t_enq - t_stage) * 1000
             self.stats["gpu_wait_ms_sum"] += (t_done - t_enq) * 1000
+            if self.out_convert is not None:
+                # conversion materializes a fresh batch: views are safe
+                conv = self.out_convert(out_cpu)
+                return [_slice(conv, i) for i in range(n)]
             # slices are views into the reusable pinned slab: clone them
             return [_slice_clone(out_cpu, i) for i in range(n)]
         finally:
