@@ -45,7 +45,13 @@ __device__ __forceinline__ unsigned int pack_bf16x2(float lo, float hi) {
          ((unsigned int)*reinterpret_cast<unsigned short*>(&b) << 16);
 }
 
-template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false>
+// KDIRECT: skip K LDS staging entirely -- QK^T A-fragments are 16-byte
+// contiguous per lane in the GLOBAL K layout already, and K tiles are
+// re-read by every m-tile workgroup of the same (b,h), so after the first
+// touch they come from L2. Halves the LDS footprint and the ds traffic.
+// Dense path only (the paged path needs per-row block-table lookups).
+template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false,
+          bool KDIRECT = false>
 __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* __restrict__ k,
@@ -62,7 +68,9 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
   constexpr int KSTRIDE = D + PAD;
   constexpr int VSTRIDE = BLOCK_N + PAD;
 
-  __shared__ short lds_k[2][BLOCK_N * KSTRIDE];
+  // KDIRECT uses a minimal K buffer (compiler elides unused LDS? no --
+  // size the array by the flag)
+  __shared__ short lds_k[KDIRECT ? 1 : 2][KDIRECT ? 1 : BLOCK_N * KSTRIDE];
   __shared__ short lds_vt[2][D * VSTRIDE];
 
   const int bh = blockIdx.x;
@@ -114,7 +122,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
   for (int dt = 0; dt < D / 16; ++dt) acc_o[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
   constexpr int PIECES = BLOCK_N * D / 8 / (NWAVES * 64);
-  bf16x8_t kreg[PIECES], vreg[PIECES];
+  bf16x8_t kreg[KDIRECT ? 1 : PIECES], vreg[PIECES];
 
   auto stage_load = [&](int n0) {
 #pragma unroll
@@ -131,10 +139,10 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
         } else {
           off = kv_base + (long)gkey * st.ks + d8;
         }
-        kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);
+        if (!KDIRECT) kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);
         vreg[i] = *reinterpret_cast<const bf16x8_t*>(v + off);
       } else {
-        kreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        if (!KDIRECT) kreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
         vreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
@@ -145,7 +153,9 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
       const int p = tid + i * NWAVES * 64;
       const int key = p / (D / 8);
       const int d8 = (p % (D / 8)) * 8;
-      *reinterpret_cast<bf16x8_t*>(&lds_k[buf][key * KSTRIDE + d8]) = kreg[i];
+      if (!KDIRECT)
+        *reinterpret_cast<bf16x8_t*>(&lds_k[buf][key * KSTRIDE + d8]) =
+            kreg[i];
       const int kswz = key ^ (((d8 >> 3) & 7) << 3);
 #pragma unroll
       for (int e = 0; e < 8; ++e)
@@ -166,13 +176,27 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
 #pragma unroll
     for (int t = 0; t < BLOCK_N / 16; ++t)
       acc_s[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-    for (int kc = 0; kc < D / 32; ++kc) {
+    if (KDIRECT) {
 #pragma unroll
       for (int t = 0; t < BLOCK_N / 16; ++t) {
-        const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
-            &lds_k[cur][(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
-        acc_s[t] = MFMA_16x16x32(kf, q_frag[kc], acc_s[t]);  // SWAPPED
+        const int gkey = min(n0 + t * 16 + frag_row, kv_len - 1);
+        const __hip_bfloat16* krow = k + kv_base + (long)gkey * st.ks;
+#pragma unroll
+        for (int kc = 0; kc < D / 32; ++kc) {
+          const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+              krow + kc * 32 + frag_ko);
+          acc_s[t] = MFMA_16x16x32(kf, q_frag[kc], acc_s[t]);  // SWAPPED
+        }
+      }
+    } else {
+#pragma unroll
+      for (int kc = 0; kc < D / 32; ++kc) {
+#pragma unroll
+        for (int t = 0; t < BLOCK_N / 16; ++t) {
+          const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+              &lds_k[cur][(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
+          acc_s[t] = MFMA_16x16x32(kf, q_frag[kc], acc_s[t]);  // SWAPPED
+        }
       }
     }
 
@@ -324,7 +348,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
 torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
                                    torch::Tensor v, bool causal, double scale,
                                    c10::optional<torch::Tensor> seq_lens,
-                                   bool bshd) {
+                                   bool bshd, bool kdirect) {
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1);
@@ -352,20 +376,25 @@ torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
   dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
   dim3 block(NWAVES * 64);
   hipStream_t stream_ = cmls::current_stream();
-#define LAUNCH_ATTN2(DD, CC, SS)                                             \
-  hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, CC, SS>), grid, block, 0,   \
+#define LAUNCH_ATTN2(DD, CC, SS, KD)                                         \
+  hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, CC, SS, false, KD>), grid,  \
+                     block, 0,                                               \
                      stream_, (const __hip_bfloat16*)q.data_ptr(),           \
                      (const __hip_bfloat16*)k.data_ptr(),                    \
                      (const __hip_bfloat16*)v.data_ptr(),                    \
                      (__hip_bfloat16*)out.data_ptr(), sl, nullptr, nullptr,  \
                      0, 0, st, B, H, Hkv, Sq, Sk, (float)scale)
+#define LAUNCH_ATTN2_KD(DD, CC, SS)                                          \
+  do { if (kdirect) LAUNCH_ATTN2(DD, CC, SS, true);                          \
+       else LAUNCH_ATTN2(DD, CC, SS, false); } while (0)
   if (D == 64) {
-    if (causal) { if (sl) LAUNCH_ATTN2(64, true, true); else LAUNCH_ATTN2(64, true, false); }
-    else        { if (sl) LAUNCH_ATTN2(64, false, true); else LAUNCH_ATTN2(64, false, false); }
+    if (causal) { if (sl) LAUNCH_ATTN2_KD(64, true, true); else LAUNCH_ATTN2_KD(64, true, false); }
+    else        { if (sl) LAUNCH_ATTN2_KD(64, false, true); else LAUNCH_ATTN2_KD(64, false, false); }
   } else {
-    if (causal) { if (sl) LAUNCH_ATTN2(128, true, true); else LAUNCH_ATTN2(128, true, false); }
-    else        { if (sl) LAUNCH_ATTN2(128, false, true); else LAUNCH_ATTN2(128, false, false); }
+    if (causal) { if (sl) LAUNCH_ATTN2_KD(128, true, true); else LAUNCH_ATTN2_KD(128, true, false); }
+    else        { if (sl) LAUNCH_ATTN2_KD(128, false, true); else LAUNCH_ATTN2_KD(128, false, false); }
   }
+#undef LAUNCH_ATTN2_KD
 #undef LAUNCH_ATTN2
   return out;
 }

@@ -29,7 +29,10 @@ for (b, h, hkv, s, d, causal, use_sl) in [
     sl = (torch.tensor([s, s // 2][:b][:b], dtype=torch.int32, device="cuda")
           if use_sl else None)
     got = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False)
+    gotkd = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False, True)
     want = ref_attn(q, k, v, causal, sl)
+    kd_delta = (gotkd.float() - got.float()).abs().max().item()
+    assert kd_delta < 1e-6, ("kdirect mismatch", kd_delta)
     # compare only valid rows (pad rows are garbage in both)
     m = (got.float().cpu() - want).abs()
     if sl is not None:
@@ -61,9 +64,11 @@ for name, (b, h, hkv, s, d, causal) in [
     v = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
     t1 = t(lambda: ext1.attention_prefill(q, k, v, causal, d ** -0.5, None, False))
     t2 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False))
+    t3 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False, True))
     fl = 4.0 * b * h * s * s * d * (0.5 if causal else 1.0)
     print(f"{name}: v1 {t1*1e6:8.1f}us {fl/t1/1e12:6.0f}TF | "
-          f"v2 {t2*1e6:8.1f}us {fl/t2/1e12:6.0f}TF | {t1/t2:.2f}x", flush=True)
+          f"v2 {t2*1e6:8.1f}us {fl/t2/1e12:6.0f}TF | "
+          f"v2kd {t3*1e6:8.1f}us {fl/t3/1e12:6.0f}TF", flush=True)
 PY
 timeout 600 python /tmp/attn_v2_ab.py 2>&1 | grep -v Warn | tee gpurun_out/attn_v2_ab.txt
 echo ATTNV2DONE
