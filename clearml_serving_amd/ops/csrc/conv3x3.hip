@@ -55,7 +55,10 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void conv3x3_kernel(
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  __shared__ __hip_bfloat16 lds[IROWS * ICOLS * PITCH];
+  // double-buffered input window, register-staged (T14 issue-early /
+  // write-late): chunk c+1's global loads are issued before computing on
+  // chunk c, so HBM latency hides under the 9-tap MFMA work
+  __shared__ __hip_bfloat16 lds[2][IROWS * ICOLS * PITCH];
 
   f32x4_t acc[MT][NTPW];
 #pragma unroll
@@ -67,33 +70,51 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void conv3x3_kernel(
   const int frow = lane & 15;          // fragment row (m or n within tile)
   const int fk = (lane >> 4) * 8;      // fragment k offset within CSTEP
 
-  for (int c0 = 0; c0 < C; c0 += CSTEP) {
-    // ---- stage input window (IROWS x ICOLS x CSTEP) into LDS ---------- //
-    // each thread covers 16-byte segments (8 channels); consecutive
-    // threads -> consecutive segments of one (iy, ix) site
-    __syncthreads();  // previous chunk's fragment reads done
-    constexpr int SITES = IROWS * ICOLS;
-    constexpr int SEGS_PER_SITE = CSTEP / 8;  // 4
-    constexpr int TOTAL = SITES * SEGS_PER_SITE;
-    for (int i = tid; i < TOTAL; i += NWAVES * 64) {
+  constexpr int SITES = IROWS * ICOLS;
+  constexpr int SEGS_PER_SITE = CSTEP / 8;  // 4
+  constexpr int TOTAL = SITES * SEGS_PER_SITE;
+  constexpr int SREGS = (TOTAL + NWAVES * 64 - 1) / (NWAVES * 64);
+  bf16x8_t sreg[SREGS];
+
+  // each thread covers 16-byte segments (8 channels); consecutive
+  // threads -> consecutive segments of one (iy, ix) site
+  auto load_chunk = [&](int c0) {
+#pragma unroll
+    for (int r = 0; r < SREGS; ++r) {
+      const int i = tid + r * NWAVES * 64;
+      if (i >= TOTAL) break;
       const int site = i / SEGS_PER_SITE;
       const int seg = (i % SEGS_PER_SITE) * 8;
       const int iy = site / ICOLS;         // 0..IROWS-1
       const int ix = site % ICOLS;         // 0..ICOLS-1
       const int gy = y0 + iy - 1;
       const int gx = ix - 1;
-      bf16x8_t v;
       if (gy >= 0 && gy < WIDTH && gx >= 0 && gx < WIDTH) {
-        v = *reinterpret_cast<const bf16x8_t*>(
+        sreg[r] = *reinterpret_cast<const bf16x8_t*>(
             x + in_base + ((long)gy * WIDTH + gx) * C + c0 + seg);
       } else {
-        v = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        sreg[r] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
-      *reinterpret_cast<bf16x8_t*>(
-          &lds[(long)site * PITCH + seg]) = v;
     }
-    __syncthreads();
+  };
+  auto write_chunk = [&](int buf) {
+#pragma unroll
+    for (int r = 0; r < SREGS; ++r) {
+      const int i = tid + r * NWAVES * 64;
+      if (i >= TOTAL) break;
+      const int site = i / SEGS_PER_SITE;
+      const int seg = (i % SEGS_PER_SITE) * 8;
+      *reinterpret_cast<bf16x8_t*>(
+          &lds[buf][(long)site * PITCH + seg]) = sreg[r];
+    }
+  };
 
+  load_chunk(0);
+  write_chunk(0);
+  if (CSTEP < C) load_chunk(CSTEP);
+  __syncthreads();
+  int cur = 0;
+  for (int c0 = 0; c0 < C; c0 += CSTEP) {
     // ---- 9 taps x 7 m-tiles x NTPW n-tiles ---------------------------- //
 #pragma unroll
     for (int dy = 0; dy < 3; ++dy) {
@@ -114,12 +135,19 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void conv3x3_kernel(
           const int ox = m % WIDTH;
           const int site = (oy + dy) * ICOLS + (ox + dx);
           const bf16x8_t ba = *reinterpret_cast<const bf16x8_t*>(
-              &lds[(long)site * PITCH + fk]);
+              &lds[cur][(long)site * PITCH + fk]);
 #pragma unroll
           for (int nt = 0; nt < NTPW; ++nt)
             acc[t][nt] = MFMA16(ba, bw[nt], acc[t][nt]);
         }
       }
+    }
+    if (c0 + CSTEP < C) {
+      __syncthreads();        // all waves done reading lds[cur ^ 1]
+      write_chunk(cur ^ 1);   // regs of chunk c0+CSTEP -> LDS
+      if (c0 + 2 * CSTEP < C) load_chunk(c0 + 2 * CSTEP);
+      __syncthreads();        // buf ready for the next iteration
+      cur ^= 1;
     }
   }
 
