@@ -51,7 +51,7 @@ __device__ __forceinline__ unsigned int pack_bf16x2(float lo, float hi) {
 // touch they come from L2. Halves the LDS footprint and the ds traffic.
 // Dense path only (the paged path needs per-row block-table lookups).
 template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false,
-          bool KDIRECT = false>
+          bool KDIRECT = false, bool PIPE = false>
 __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* __restrict__ k,
@@ -163,16 +163,8 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     }
   };
 
-  stage_load(0);
-  stage_write(0);
-  __syncthreads();
-  int cur = 0;
-  for (int n0 = 0; n0 < kv_hi; n0 += BLOCK_N) {
-    const bool has_next = n0 + BLOCK_N < kv_hi;
-    if (has_next) stage_load(n0 + BLOCK_N);
-
-    // ---- S^T = K Q^T: tile t covers keys [n0+16t, n0+16t+16) ---- //
-    f32x4_t acc_s[BLOCK_N / 16];
+  // ---- per-tile pieces as lambdas (shared by both loop schedules) ---- //
+  auto compute_qk = [&](int n0, int buf, f32x4_t (&acc_s)[BLOCK_N / 16]) {
 #pragma unroll
     for (int t = 0; t < BLOCK_N / 16; ++t)
       acc_s[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
@@ -194,13 +186,16 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
 #pragma unroll
         for (int t = 0; t < BLOCK_N / 16; ++t) {
           const bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
-              &lds_k[cur][(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
+              &lds_k[buf][(t * 16 + frag_row) * KSTRIDE + kc * 32 + frag_ko]);
           acc_s[t] = MFMA_16x16x32(kf, q_frag[kc], acc_s[t]);  // SWAPPED
         }
       }
     }
+  };
 
-    // ---- mask + in-register online softmax (per-lane q column) ---- //
+  // mask + in-register online softmax + P^T redistribution + PV, for the
+  // tile whose S^T sits in acc_s and whose V tile is in lds_vt[buf]
+  auto softmax_pv = [&](f32x4_t (&acc_s)[BLOCK_N / 16], int n0, int buf) {
     // lane's S values: key = n0 + 16t + 4g + r, q = wm0 + frag_row
     const int qrow = wm0 + frag_row;
     const bool tile_full =
@@ -263,7 +258,6 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     }
 
     // ---- P^T -> PV B-fragments via lane moves (no LDS round-trip) ---- //
-    // pack each tile's 4 p values as 2 u32 of bf16x2
     unsigned int pk01[BLOCK_N / 16], pk23[BLOCK_N / 16];
 #pragma unroll
     for (int t = 0; t < BLOCK_N / 16; ++t) {
@@ -271,16 +265,13 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
       pk23[t] = pack_bf16x2(pv[t][2], pv[t][3]);
     }
     // B[k=key][j=q] fragment for kc: this lane (group g) holds keys
-    // 32kc + 8g + [0..7]; element bytes come from source groups
-    // gA = 2*(g&1) wait -- keys 8g+e of the 32-key window: tile
-    // t_sel = (8g+e)/16 + 2kc = (g>>1) + 2kc (e<8 keeps it constant),
-    // source group = ((8g+e)%16)/4 = (2(g&1)) + (e>>2), r = e&3.
+    // 32kc + 8g + [0..7]; tile t_sel = (g>>1) + 2kc, source groups
+    // 2(g&1) and 2(g&1)+1 (r = e&3).
     const int col = frag_row;
 #pragma unroll
     for (int kc = 0; kc < BLOCK_N / 32; ++kc) {
-      // lane (group g) needs tile tsel = (g>>1) + 2kc -- but __shfl
-      // transports the SOURCE lane's register, and each source pair serves
-      // one tile-0 and one tile-1 target, so shuffle both tiles and select
+      // __shfl transports the SOURCE lane's register, and each source pair
+      // serves one tile-0 and one tile-1 target: shuffle both, select
       const int gA = 2 * (g & 1);
       const int srcA = (gA << 4) | col;
       const int srcB = ((gA + 1) << 4) | col;
@@ -312,15 +303,54 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
         const int vrow = dt * 16 + frag_row;
         const int vkey = (kc * 32 + frag_ko) ^ (((vrow >> 3) & 7) << 3);
         const bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-            &lds_vt[cur][vrow * VSTRIDE + vkey]);
+            &lds_vt[buf][vrow * VSTRIDE + vkey]);
         acc_o[dt] = MFMA_16x16x32(vf, pf, acc_o[dt]);
       }
     }
+  };
 
-    if (has_next) {
-      stage_write(cur ^ 1);
-      __syncthreads();
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (PIPE) {
+    // T15 double-pipeline: QK^T of tile n+1 (MFMA) issues BEFORE the
+    // softmax finish of tile n (VALU) -- independent instruction streams
+    // the scheduler interleaves, hiding ~1/3 of the softmax cost. Costs a
+    // second barrier per tile and one extra acc_s register set.
+    f32x4_t s_prev[BLOCK_N / 16], s_cur[BLOCK_N / 16];
+    compute_qk(0, 0, s_prev);
+    if (BLOCK_N < kv_hi) stage_load(BLOCK_N);
+    int n_prev = 0;
+    int cur = 0;
+    for (int n0 = BLOCK_N;; n0 += BLOCK_N) {
+      const bool has_cur = n0 < kv_hi;
+      if (has_cur) {
+        __syncthreads();         // all PV reads of lds[cur^1] done
+        stage_write(cur ^ 1);    // tile n0 -> other buffer
+        __syncthreads();
+        compute_qk(n0, cur ^ 1, s_cur);
+        if (n0 + BLOCK_N < kv_hi) stage_load(n0 + BLOCK_N);
+      }
+      softmax_pv(s_prev, n_prev, cur);
+      if (!has_cur) break;
+#pragma unroll
+      for (int t = 0; t < BLOCK_N / 16; ++t) s_prev[t] = s_cur[t];
+      n_prev = n0;
       cur ^= 1;
+    }
+  } else {
+    int cur = 0;
+    for (int n0 = 0; n0 < kv_hi; n0 += BLOCK_N) {
+      const bool has_next = n0 + BLOCK_N < kv_hi;
+      if (has_next) stage_load(n0 + BLOCK_N);
+      f32x4_t acc_s[BLOCK_N / 16];
+      compute_qk(n0, cur, acc_s);
+      softmax_pv(acc_s, n0, cur);
+      if (has_next) {
+        stage_write(cur ^ 1);
+        __syncthreads();
+        cur ^= 1;
+      }
     }
   }
 
@@ -348,7 +378,7 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
 torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
                                    torch::Tensor v, bool causal, double scale,
                                    c10::optional<torch::Tensor> seq_lens,
-                                   bool bshd, bool kdirect) {
+                                   bool bshd, bool kdirect, bool pipe) {
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1);
@@ -376,17 +406,18 @@ torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
   dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
   dim3 block(NWAVES * 64);
   hipStream_t stream_ = cmls::current_stream();
-#define LAUNCH_ATTN2(DD, CC, SS, KD)                                         \
-  hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, CC, SS, false, KD>), grid,  \
-                     block, 0,                                               \
+#define LAUNCH_ATTN2(DD, CC, SS, KD, PP)                                     \
+  hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, CC, SS, false, KD, PP>),    \
+                     grid, block, 0,                                         \
                      stream_, (const __hip_bfloat16*)q.data_ptr(),           \
                      (const __hip_bfloat16*)k.data_ptr(),                    \
                      (const __hip_bfloat16*)v.data_ptr(),                    \
                      (__hip_bfloat16*)out.data_ptr(), sl, nullptr, nullptr,  \
                      0, 0, st, B, H, Hkv, Sq, Sk, (float)scale)
 #define LAUNCH_ATTN2_KD(DD, CC, SS)                                          \
-  do { if (kdirect) LAUNCH_ATTN2(DD, CC, SS, true);                          \
-       else LAUNCH_ATTN2(DD, CC, SS, false); } while (0)
+  do { if (pipe) LAUNCH_ATTN2(DD, CC, SS, false, true);                      \
+       else if (kdirect) LAUNCH_ATTN2(DD, CC, SS, true, false);              \
+       else LAUNCH_ATTN2(DD, CC, SS, false, false); } while (0)
   if (D == 64) {
     if (causal) { if (sl) LAUNCH_ATTN2_KD(64, true, true); else LAUNCH_ATTN2_KD(64, true, false); }
     else        { if (sl) LAUNCH_ATTN2_KD(64, false, true); else LAUNCH_ATTN2_KD(64, false, false); }

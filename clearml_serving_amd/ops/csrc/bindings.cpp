@@ -46,7 +46,7 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w,
 torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
                                    torch::Tensor v, bool causal, double scale,
                                    c10::optional<torch::Tensor> seq_lens,
-                                   bool bshd, bool kdirect);
+                                   bool bshd, bool kdirect, bool pipe);
 torch::Tensor attention_prefill_paged_v2(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
     torch::Tensor block_table, torch::Tensor kv_lens, torch::Tensor q_lens,
@@ -86,6 +86,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_v2", &attention_prefill_v2, py::arg("q"),
         py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
         py::arg("seq_lens") = py::none(), py::arg("bshd") = false,
-        py::arg("kdirect") = false);
+        py::arg("kdirect") = false, py::arg("pipe") = false);
   m.def("attention_prefill_paged_v2", &attention_prefill_paged_v2);
 }

@@ -30,15 +30,19 @@ for (b, h, hkv, s, d, causal, use_sl) in [
           if use_sl else None)
     got = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False)
     gotkd = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False, True)
+    gotp = ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, sl, False, False, True)
     want = ref_attn(q, k, v, causal, sl)
     # compare only valid rows (rows beyond seq_len are unwritten garbage)
     kd = (gotkd.float() - got.float()).abs()
+    pp = (gotp.float() - got.float()).abs()
     m = (got.float().cpu() - want).abs()
     if sl is not None:
         for i in range(b):
             m[i, :, sl[i]:, :] = 0
             kd[i, :, sl[i]:, :] = 0
+            pp[i, :, sl[i]:, :] = 0
     assert kd.max().item() < 1e-6, ("kdirect mismatch", kd.max().item())
+    assert pp.max().item() < 1e-6, ("pipe mismatch", pp.max().item())
     bad = m.max().item()
     print(f"numerics b{b} h{h}/{hkv} s{s} d{d} causal={causal} sl={use_sl}: "
           f"maxerr {bad:.4f}", flush=True)
@@ -65,11 +69,11 @@ for name, (b, h, hkv, s, d, causal) in [
     v = (torch.randn(b, hkv, s, d, device="cuda") / 4).to(torch.bfloat16)
     t1 = t(lambda: ext1.attention_prefill(q, k, v, causal, d ** -0.5, None, False))
     t2 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False))
-    t3 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False, True))
+    t4 = t(lambda: ext.attention_prefill_v2(q, k, v, causal, d ** -0.5, None, False, False, True))
     fl = 4.0 * b * h * s * s * d * (0.5 if causal else 1.0)
     print(f"{name}: v1 {t1*1e6:8.1f}us {fl/t1/1e12:6.0f}TF | "
           f"v2 {t2*1e6:8.1f}us {fl/t2/1e12:6.0f}TF | "
-          f"v2kd {t3*1e6:8.1f}us {fl/t3/1e12:6.0f}TF", flush=True)
+          f"v2pipe {t4*1e6:8.1f}us {fl/t4/1e12:6.0f}TF", flush=True)
 PY
 timeout 600 python /tmp/attn_v2_ab.py 2>&1 | grep -v Warn | tee gpurun_out/attn_v2_ab.txt
 echo ATTNV2DONE
