@@ -101,6 +101,32 @@ class EngineOwner:
         self._engine_eps[url] = ep_dict
         return engine
 
+    async def _push(self, worker: int, record: bytes) -> None:
+        ring = self.resp_rings[worker]
+        while not ring.push(record):
+            await asyncio.sleep(0.001)
+
+    async def _relay_stream(self, req_id: int, worker: int, response) -> None:
+        """Relay a StreamingResponse body chunk-by-chunk over the response
+        ring (FIFO per worker, so chunk order is preserved); the front's
+        ShmClient.infer_stream reassembles it into an SSE response."""
+        from .shm_transport import (pack_response, pack_stream_chunk,
+                                    pack_stream_end)
+
+        try:
+            async for chunk in response.body_iterator:
+                if isinstance(chunk, str):
+                    chunk = chunk.encode()
+                if chunk:
+                    await self._push(worker, pack_stream_chunk(req_id, chunk))
+            await self._push(worker, pack_stream_end(req_id))
+            self.stats["requests"] += 1
+        except Exception as ex:
+            traceback.print_exc()
+            self.stats["errors"] += 1
+            await self._push(worker, pack_response(
+                req_id, error="{}: {}".format(type(ex).__name__, ex)))
+
     async def _handle(self, raw: bytes, worker: int) -> None:
         from .shm_transport import pack_response, unpack_request
 
@@ -124,6 +150,9 @@ class EngineOwner:
                         "engine for '{}' does not implement '{}'".format(
                             url, serve_type))
                 out = await method(data, {})
+            if hasattr(out, "body_iterator"):  # StreamingResponse (SSE)
+                await self._relay_stream(req_id, worker, out)
+                return
             resp = pack_response(req_id, out)
             self.stats["requests"] += 1
         except Exception as ex:
@@ -133,9 +162,7 @@ class EngineOwner:
                 return
             resp = pack_response(req_id, error="{}: {}".format(
                 type(ex).__name__, ex))
-        ring = self.resp_rings[worker]
-        while not ring.push(resp):
-            await asyncio.sleep(0.001)
+        await self._push(worker, resp)
 
     async def serve(self) -> None:
         idle_sleep = 0.0002

@@ -70,8 +70,8 @@ class ShmProxyRequest(BasePreprocessRequest):
 
 class ShmLlmProxyRequest(ShmProxyRequest):
     """LLM endpoints over the SHM transport: whole-request dispatch to the
-    owner's engine. Streaming (SSE) requires the single-process topology;
-    stream=True is rejected with a clear message."""
+    owner's engine; SSE streams relay chunk-by-chunk over the response
+    ring."""
 
     is_preprocess_async = True
     is_postprocess_async = True
@@ -101,14 +101,19 @@ class ShmLlmProxyRequest(ShmProxyRequest):
 
     async def _ship(self, serve_type: str, body: Any) -> Any:
         body = self._clean_body(body)
+        payload = {"__serve_type__": serve_type, "__body__": body}
         if isinstance(body, dict) and body.get("stream"):
-            raise ValueError(
-                "SSE streaming is not available through the multi-process "
-                "front; run the single-process topology (launch --workers 0) "
-                "for streaming LLM endpoints, or set stream=false")
-        return await self.client.infer(
-            self._url, {"__serve_type__": serve_type, "__body__": body},
-            owner=self._owner)
+            # SSE: the owner relays the engine's stream chunk-by-chunk over
+            # the response ring; re-expose it as a StreamingResponse here
+            from fastapi.responses import StreamingResponse
+
+            async def gen():
+                async for chunk in self.client.infer_stream(
+                        self._url, payload, owner=self._owner):
+                    yield chunk
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+        return await self.client.infer(self._url, payload, owner=self._owner)
 
     async def process(self, data, state, collect_custom_statistics_fn=None):
         return await self._ship("process", data)

@@ -174,6 +174,8 @@ KIND_PICKLE = 1
 STATUS_OK_TENSORS = 0
 STATUS_OK_PICKLE = 1
 STATUS_ERROR = 2
+STATUS_STREAM_CHUNK = 3  # SSE chunk bytes; more follow
+STATUS_STREAM_END = 4    # stream finished
 
 
 def _as_numpy(v) -> Optional[np.ndarray]:
@@ -288,10 +290,20 @@ def pack_response(req_id: int, result: Any = None,
     return head + bytes([STATUS_OK_PICKLE]) + pickle.dumps(result, protocol=4)
 
 
+def pack_stream_chunk(req_id: int, chunk: bytes) -> bytes:
+    return struct.pack("<Q", req_id) + bytes([STATUS_STREAM_CHUNK]) + chunk
+
+
+def pack_stream_end(req_id: int) -> bytes:
+    return struct.pack("<Q", req_id) + bytes([STATUS_STREAM_END])
+
+
 def unpack_response(buf: bytes) -> Tuple[int, int, Any]:
     (req_id,) = struct.unpack_from("<Q", buf, 0)
     status = buf[8]
     off = 9
+    if status in (STATUS_STREAM_CHUNK, STATUS_STREAM_END):
+        return req_id, status, bytes(buf[off:])
     if status == STATUS_OK_TENSORS:
         pairs = unpack_tensors(buf, off)
         if len(pairs) == 1 and pairs[0][0] == "":
@@ -326,6 +338,7 @@ class ShmClient:
             for o in range(n_owners)
         ]
         self._futures: Dict[int, asyncio.Future] = {}
+        self._streams: Dict[int, "asyncio.Queue"] = {}
         self._next_id = itertools.count(1)
         self._poller: Optional[asyncio.Task] = None
 
@@ -344,6 +357,15 @@ class ShmClient:
                 for raw in ring.drain(512):
                     got = True
                     req_id, status, payload = unpack_response(raw)
+                    if status in (STATUS_STREAM_CHUNK, STATUS_STREAM_END):
+                        q = self._streams.get(req_id)
+                        if q is not None:
+                            q.put_nowait((status, payload))
+                        continue
+                    if status == STATUS_ERROR and req_id in self._streams:
+                        # mid-stream failure routes to the stream consumer
+                        self._streams[req_id].put_nowait((status, payload))
+                        continue
                     fut = self._futures.pop(req_id, None)
                     if fut is not None and not fut.done():
                         fut.set_result((status, payload))
@@ -371,6 +393,32 @@ class ShmClient:
         if status == STATUS_ERROR:
             raise RuntimeError(result)
         return result
+
+    async def infer_stream(self, url: str, data: Any, owner: int = 0,
+                           timeout: float = 300.0):
+        """Streaming variant: yields raw SSE chunk bytes as the engine
+        owner relays them (STATUS_STREAM_CHUNK records on the response
+        ring, which is FIFO -- chunk order is preserved)."""
+        self._ensure_poller()
+        req_id = next(self._next_id)
+        q: "asyncio.Queue" = asyncio.Queue()
+        self._streams[req_id] = q
+        try:
+            payload = pack_request(req_id, url, data)
+            ring = self.req_rings[owner]
+            while not ring.push(payload):
+                await asyncio.sleep(0.001)
+            while True:
+                status, chunk = await asyncio.wait_for(q.get(),
+                                                       timeout=timeout)
+                if status == STATUS_STREAM_END:
+                    return
+                if status == STATUS_ERROR:
+                    raise RuntimeError(chunk if isinstance(chunk, str)
+                                       else chunk.decode())
+                yield chunk
+        finally:
+            self._streams.pop(req_id, None)
 
     def close(self):
         if self._poller is not None:

@@ -259,3 +259,87 @@ def test_multiproc_front_owner_http_e2e(serving_session):
             launcher.wait(timeout=20)
         except subprocess.TimeoutExpired:
             launcher.kill()
+
+
+@pytest.mark.timeout(240)
+def test_multiproc_llm_and_sse_streaming(tmp_path):
+    """LLM endpoints through the multi-process front: whole-request dispatch
+    AND SSE streaming relayed chunk-by-chunk over the response ring."""
+    import json as _json
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="shm-llm",
+                                 force_create=True)
+    card = tmp_path / "card.json"
+    card.write_text(_json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 128, "device": "cpu"}))
+    rec = store.register_model(name="tiny", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(engine_type="llm", serving_url="tinyllm",
+                                    model_id=rec.model_id))
+    proc.serialize()
+
+    port = _free_port()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    launcher = subprocess.Popen(
+        [sys.executable, "-m", "clearml_serving_amd.serving.launch",
+         "--store", store_root, "--session", proc.get_id(),
+         "--host", "127.0.0.1", "--port", str(port),
+         "--workers", "1", "--owners", "1", "--ring-mb", "8",
+         "--no-restart"],
+        env=env, cwd=REPO)
+    try:
+        import httpx
+
+        base = "http://127.0.0.1:{}".format(port)
+        with httpx.Client(base_url=base, timeout=60.0) as client:
+            for _ in range(240):
+                try:
+                    if client.get("/health").status_code == 200:
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.5)
+                assert launcher.poll() is None, "launcher died"
+            else:
+                pytest.fail("service did not come up")
+
+            # non-streaming chat completion through the SHM hop
+            r = client.post("/serve/openai/v1/chat/completions", json={
+                "model": "tinyllm",
+                "messages": [{"role": "user", "content": "hello"}],
+                "max_tokens": 4, "temperature": 0, "ignore_eos": True})
+            assert r.status_code == 200, r.text
+            body = r.json()
+            assert body["usage"]["completion_tokens"] == 4
+
+            # SSE streaming: chunks relayed over the ring
+            chunks = []
+            with client.stream("POST", "/serve/openai/v1/chat/completions",
+                               json={"model": "tinyllm", "stream": True,
+                                     "messages": [{"role": "user",
+                                                   "content": "hi"}],
+                                     "max_tokens": 5, "temperature": 0,
+                                     "ignore_eos": True}) as resp:
+                assert resp.status_code == 200
+                assert "text/event-stream" in resp.headers["content-type"]
+                for line in resp.iter_lines():
+                    if line.startswith("data: "):
+                        chunks.append(line[len("data: "):])
+            assert chunks[-1] == "[DONE]"
+            deltas = [_json.loads(c) for c in chunks[:-1]]
+            toks = [d["choices"][0]["delta"].get("content", "")
+                    for d in deltas]
+            assert len(toks) == 5
+    finally:
+        launcher.terminate()
+        try:
+            launcher.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            launcher.kill()
