@@ -111,8 +111,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     # defaults sized for a >=60 s steady-state timed region on one MI355X
-    # (~50 ms/step measured); still finishes within minutes
-    ap.add_argument("--steps", type=int, default=1200)
+    # (~40 ms/step measured); still finishes within minutes
+    ap.add_argument("--steps", type=int, default=1600)
     ap.add_argument("--warmup", type=int, default=50)
     ap.add_argument("--no-graphs", action="store_true")
     args = ap.parse_args()
