@@ -40,7 +40,6 @@ loop.run_until_complete(soak(65))   # the soak
 print("stats:", {k: v for k, v in eng.stats.items()}, flush=True)
 eng.stop()
 PY
-timeout 600 python /tmp/fp8_soak.py 2>&1 | grep -v "Task was\|Task pend" | tee gpurun_out/llm_soak_fp8.txt
 
 # kernel share of a short fp8 decode run (stats only; small CSV kept)
 cat > /tmp/fp8_short.py <<'PY'
@@ -57,8 +56,10 @@ async def one(i):
     while True:
         item = await seq.stream.get()
         if item["finished"]: return
+async def main():
+    await asyncio.gather(*[one(i) for i in range(32)])
 loop = asyncio.new_event_loop()
-loop.run_until_complete(asyncio.gather(*[one(i) for i in range(32)]))
+loop.run_until_complete(main())
 eng.stop()
 PY
 timeout 400 rocprofv3 --kernel-trace --stats --output-format csv \
