@@ -56,6 +56,9 @@ class Fp8Linear(nn.Module):
         self.in_features = linear.in_features
         self.out_dtype = linear.weight.dtype  # model compute dtype
 
+    # LDS-staged v2 covers M 17..V2_MAX_M (measured vs hipBLASLt scaled_mm)
+    V2_MAX_M = int(__import__("os").environ.get("CMLS_FP8_V2_MAX", 64))
+
     def forward_q(self, x8: torch.Tensor, x_scale: torch.Tensor,
                   out_dtype=None) -> torch.Tensor:
         out_dtype = out_dtype or self.out_dtype
@@ -65,6 +68,12 @@ class Fp8Linear(nn.Module):
                 and self.out_features % 16 == 0):
             out = ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8_sw,
                                       self.weight_scale, swizzled=True)
+            return out if out.dtype == out_dtype else out.to(out_dtype)
+        if (m <= self.V2_MAX_M and x8.is_cuda
+                and self.in_features % 128 == 0
+                and self.out_features % 16 == 0):
+            out = ops.skinny_gemm_fp8_v2(x8, x_scale, self.weight_fp8,
+                                         self.weight_scale)
             return out if out.dtype == out_dtype else out.to(out_dtype)
         if not x8.is_cuda:
             return ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8,

@@ -240,6 +240,23 @@ def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
     return (a @ w.t()).to(torch.bfloat16)
 
 
+def skinny_gemm_fp8_v2(a8: torch.Tensor, a_scale: torch.Tensor,
+                       w8: torch.Tensor, w_scale: torch.Tensor
+                       ) -> torch.Tensor:
+    """LDS-staged skinny fp8 GEMM for M 17..64 (plain fp8 weight layout):
+    4 waves share one k progression over register-pipelined 128-byte
+    windows -- covers the mid-batch decode regime where v1's direct
+    fragment loads collapse and hipBLASLt underfills the chip."""
+    if a8.is_cuda:
+        ext = _require_ext("skinny_gemm_fp8_v2")
+        if ext is not None:
+            return ext.skinny_gemm_fp8_v2(
+                a8.view(torch.uint8), a_scale, w8.view(torch.uint8), w_scale)
+    a = a8.view(torch.float8_e4m3fn).float() * a_scale[:, None]
+    w = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None]
+    return (a @ w.t()).to(torch.bfloat16)
+
+
 def conv3x3_supported(x: torch.Tensor, w: torch.Tensor) -> bool:
     """True when the in-tree conv kernel serves this shape (3x3 s1 p1,
     ResNet-50 bottleneck widths, bf16 GPU)."""

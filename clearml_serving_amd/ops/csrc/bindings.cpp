@@ -40,6 +40,8 @@ std::vector<torch::Tensor> silu_mul_fp8(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> quant_fp8(torch::Tensor x);
 torch::Tensor skinny_gemm_fp8(torch::Tensor a8, torch::Tensor a_scale,
                               torch::Tensor w8, torch::Tensor w_scale);
+torch::Tensor skinny_gemm_fp8_v2(torch::Tensor a8, torch::Tensor a_scale,
+                                 torch::Tensor w8, torch::Tensor w_scale);
 torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w,
                            c10::optional<torch::Tensor> bias, bool relu,
                            c10::optional<torch::Tensor> residual);
@@ -80,6 +82,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul_fp8", &silu_mul_fp8);
   m.def("quant_fp8", &quant_fp8);
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8);
+  m.def("skinny_gemm_fp8_v2", &skinny_gemm_fp8_v2);
   m.def("conv3x3_nhwc", &conv3x3_nhwc, py::arg("x"), py::arg("w"),
         py::arg("bias") = py::none(), py::arg("relu") = false,
         py::arg("residual") = py::none());

@@ -383,6 +383,130 @@ __global__ __launch_bounds__(F8_NW * 64, 2) void skinny_gemm_fp8_kernel(
 
 }  // namespace
 
+// ------------------------------------------------------------------- //
+// skinny_gemm_fp8 v2: LDS-staged variant for M 17..64 (v1's direct
+// fragment loads replay A rows per m-tile and collapse there, like the
+// bf16 v1 -> v2 story in skinny_gemm.hip). All 4 waves share one k
+// progression; a cooperative full-line stage brings W[16, 128B] and
+// A[M16, 128B] k-windows into LDS (pitch 144 B: 36-dword row stride lands
+// the 16 fragment rows in 16 distinct banks, rows stay 16-B aligned),
+// register-pipelined (issue window w+1's loads before computing w).
+// Weights here use the PLAIN fp8 layout (staging reads full lines, so the
+// v1 byte swizzle is unnecessary).
+// ------------------------------------------------------------------- //
+namespace {
+
+constexpr int F8V2_KW = 128;     // k per staged window (128 B/row)
+constexpr int F8V2_PITCH = 144;  // LDS row pitch (bytes): 36-dword row
+                                 // stride lands the 16 fragment rows in 16
+                                 // distinct banks AND keeps rows 16-byte
+                                 // aligned for the b128 staging stores
+                                 // (136 B would misalign odd rows)
+constexpr int F8V2_NW = 4;       // waves
+
+template <int MT>  // m tiles (2..4 -> M 17..64)
+__global__ __launch_bounds__(F8V2_NW * 64, 2) void skinny_gemm_fp8_v2_kernel(
+    const unsigned char* __restrict__ a,  // [M, K] e4m3 row-major (PLAIN)
+    const float* __restrict__ a_scale,    // [M]
+    const unsigned char* __restrict__ w,  // [N, K] e4m3 row-major (PLAIN)
+    const float* __restrict__ w_scale,    // [N]
+    __hip_bfloat16* __restrict__ c,       // [M, N]
+    int M, int N, int K) {
+  constexpr int M16 = MT * 16;
+  constexpr int ROWS = 16 + M16;  // W rows then A rows
+  const int n0 = blockIdx.x * F8_NT;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int frow = lane & 15;
+  const int fko = (lane >> 4) * 8;  // byte offset of this group's operand
+
+  __shared__ unsigned char lds[2][ROWS * F8V2_PITCH];
+
+  constexpr int TPW = (MT + F8V2_NW - 1) / F8V2_NW;
+  f32x4_t acc[TPW];
+#pragma unroll
+  for (int t = 0; t < TPW; ++t) acc[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  const int n_win = K / F8V2_KW;
+
+  // register-staged pipeline: each thread owns SEGS 16-byte segments of
+  // the (ROWS x 128 B) window; consecutive threads cover full lines
+  constexpr int SEGS = (ROWS * 8 + F8V2_NW * 64 - 1) / (F8V2_NW * 64);
+  longx2_t regs[SEGS];
+
+  auto load_win = [&](int win) {
+    const long kbase = (long)win * F8V2_KW;
+#pragma unroll
+    for (int sgi = 0; sgi < SEGS; ++sgi) {
+      const int i = tid + sgi * F8V2_NW * 64;
+      if (i >= ROWS * 8) break;
+      const int row = i >> 3;
+      const int seg = (i & 7) * 16;
+      const unsigned char* src =
+          row < 16 ? w + (long)(n0 + row) * K + kbase + seg
+                   : a + (long)min(row - 16, M - 1) * K + kbase + seg;
+      regs[sgi] = *reinterpret_cast<const longx2_t*>(src);
+    }
+  };
+  auto store_win = [&](int buf) {
+#pragma unroll
+    for (int sgi = 0; sgi < SEGS; ++sgi) {
+      const int i = tid + sgi * F8V2_NW * 64;
+      if (i >= ROWS * 8) break;
+      *reinterpret_cast<longx2_t*>(
+          &lds[buf][(i >> 3) * F8V2_PITCH + (i & 7) * 16]) = regs[sgi];
+    }
+  };
+
+  load_win(0);
+  store_win(0);
+  if (n_win > 1) load_win(1);
+  __syncthreads();
+  for (int win = 0; win < n_win; ++win) {
+    const int cur = win & 1;
+    // four 32-wide MFMA k-steps per window
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const long bw = *reinterpret_cast<const long*>(
+          &lds[cur][frow * F8V2_PITCH + s * 32 + fko]);
+#pragma unroll
+      for (int t = 0; t < TPW; ++t) {
+        const int tile = wave + F8V2_NW * t;
+        if (tile < MT) {
+          const long ba = *reinterpret_cast<const long*>(
+              &lds[cur][(16 + tile * 16 + frow) * F8V2_PITCH + s * 32 + fko]);
+          acc[t] = MFMA16F8(ba, bw, acc[t]);
+        }
+      }
+    }
+    if (win + 1 < n_win) {
+      __syncthreads();
+      store_win(cur ^ 1);
+      if (win + 2 < n_win) load_win(win + 2);
+      __syncthreads();
+    }
+  }
+
+  // store: each wave owns its tiles (waves shared k, no reduction)
+  const int crow = (lane >> 4) * 4;
+  const int n = n0 + (lane & 15);
+  const float ws = (n < N) ? w_scale[n] : 0.f;
+#pragma unroll
+  for (int t = 0; t < TPW; ++t) {
+    const int tile = wave + F8V2_NW * t;
+    if (tile >= MT) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = tile * 16 + crow + r;
+      if (m < M && n < N)
+        c[(long)m * N + n] = __float2bfloat16(acc[t][r] * a_scale[m] * ws);
+    }
+  }
+}
+
+}  // namespace
+
 #ifndef CMLS_KERNEL_ONLY
 #define DISPATCH_FLOAT_TYPES(TENSOR, NAME, ...)                              \
   [&] {                                                                     \
@@ -530,6 +654,41 @@ torch::Tensor skinny_gemm_fp8(torch::Tensor a8, torch::Tensor a_scale,
     default: LAUNCH_SKF8(4); break;
   }
 #undef LAUNCH_SKF8
+  return c;
+}
+
+torch::Tensor skinny_gemm_fp8_v2(torch::Tensor a8, torch::Tensor a_scale,
+                                 torch::Tensor w8, torch::Tensor w_scale) {
+  // LDS-staged variant for M 17..64; w8 is the PLAIN fp8 layout
+  TORCH_CHECK(a8.dim() == 2 && w8.dim() == 2);
+  TORCH_CHECK(a8.is_contiguous() && w8.is_contiguous());
+  TORCH_CHECK(a_scale.scalar_type() == at::kFloat &&
+              w_scale.scalar_type() == at::kFloat);
+  const int M = a8.size(0), K = a8.size(1), N = w8.size(0);
+  TORCH_CHECK(w8.size(1) == K);
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny_gemm_fp8_v2: M must be 1..64");
+  TORCH_CHECK(K % F8V2_KW == 0, "skinny_gemm_fp8_v2: K % 128 != 0");
+  TORCH_CHECK(N % F8_NT == 0);
+  TORCH_CHECK(a_scale.numel() == M && w_scale.numel() == N);
+  auto c = torch::empty({M, N}, a8.options().dtype(torch::kBFloat16));
+  hipStream_t stream_ = cmls::current_stream();
+  dim3 grid(N / F8_NT);
+  const int mt = (M + 15) / 16;
+#define LAUNCH_SKF8V2(T)                                                     \
+  hipLaunchKernelGGL((skinny_gemm_fp8_v2_kernel<T>), grid,                   \
+                     dim3(F8V2_NW * 64), 0, stream_,                         \
+                     (const unsigned char*)a8.data_ptr(),                    \
+                     (const float*)a_scale.data_ptr(),                       \
+                     (const unsigned char*)w8.data_ptr(),                    \
+                     (const float*)w_scale.data_ptr(),                       \
+                     (__hip_bfloat16*)c.data_ptr(), M, N, K)
+  switch (mt) {
+    case 1: LAUNCH_SKF8V2(1); break;
+    case 2: LAUNCH_SKF8V2(2); break;
+    case 3: LAUNCH_SKF8V2(3); break;
+    default: LAUNCH_SKF8V2(4); break;
+  }
+#undef LAUNCH_SKF8V2
   return c;
 }
 #endif  // CMLS_KERNEL_ONLY

@@ -28,13 +28,24 @@ for M in (1, 16, 32, 64):
         wsw = ops.swizzle_fp8_weight(w8)
         ext = ops._require_ext("skinny_gemm_fp8")
         us8 = t(lambda: ext.skinny_gemm_fp8(a8, as_, wsw, ws))
+        usv2 = t(lambda: ext.skinny_gemm_fp8_v2(a8, as_, w8, ws))
         usb = t(lambda: ops.skinny_linear(a, w))
         uslt = t(lambda: torch.nn.functional.linear(a, w))
+        # fp8 scaled_mm (hipBLASLt) comparison at the same precision
+        pad = (-M) % 16
+        a8p = torch.nn.functional.pad(a8, (0, 0, 0, pad)) if pad else a8
+        asp = torch.nn.functional.pad(as_.reshape(-1,1), (0, 0, 0, pad), value=1.0) if pad else as_.reshape(-1,1)
+        f8 = torch.float8_e4m3fn
+        uslt8 = t(lambda: torch._scaled_mm(a8p.view(f8), w8.view(f8).t(),
+                  scale_a=asp.contiguous(), scale_b=ws.reshape(1,-1),
+                  out_dtype=torch.bfloat16))
         gb8 = (N*K + M*K) / us8 / 1e3   # fp8 bytes
-        tot8 += us8; totb += usb; totlt += uslt
-        print(f"M={M:<3} {name:8} fp8 {us8:6.1f}us {gb8:5.0f}GB/s | "
+        gbv2 = (N*K + M*K) / usv2 / 1e3
+        tot8 += min(us8, usv2); totb += usb; totlt += uslt
+        print(f"M={M:<3} {name:8} fp8v1 {us8:6.1f}us {gb8:5.0f}GB/s | "
+              f"fp8v2 {usv2:6.1f}us {gbv2:5.0f}GB/s | lt8 {uslt8:6.1f}us | "
               f"bf16-auto {usb:6.1f}us | lt {uslt:6.1f}us", flush=True)
-    print(f"M={M:<3} ALL: fp8 {tot8:6.1f}us | bf16-auto {totb:6.1f}us | "
+    print(f"M={M:<3} ALL: fp8-best {tot8:6.1f}us | bf16-auto {totb:6.1f}us | "
           f"lt {totlt:6.1f}us  speedup vs lt {totlt/tot8:.2f}x", flush=True)
 PY
 timeout 300 python /tmp/fp8_kernels.py 2>&1 | grep -v Warn | tee gpurun_out/fp8_kernels.txt

@@ -540,3 +540,30 @@ def test_resnet_stem_im2col_matches_conv2d():
     assert got.is_contiguous(memory_format=torch.channels_last)
     assert torch.allclose(got, ref, atol=6e-2, rtol=6e-2), \
         (got - ref).abs().max()
+
+
+@pytest.mark.parametrize("m", [17, 32, 48, 64])
+def test_skinny_gemm_fp8_v2_matches_dequant_matmul(m):
+    torch.manual_seed(m)
+    k, n = 256, 320
+    a = (torch.randn(m, k, device=DEV) / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=DEV) / 8).to(torch.bfloat16)
+    a8, as_ = ops.quant_fp8(a)
+    w8, ws = ops.quant_fp8(w)
+    got = ops.skinny_gemm_fp8_v2(a8, as_, w8, ws).float().cpu()
+    ref = _dequant(a8.cpu(), as_.cpu()) @ _dequant(w8.cpu(), ws.cpu()).t()
+    assert torch.allclose(got, ref, atol=5e-2, rtol=5e-2), \
+        (got - ref).abs().max()
+
+
+def test_skinny_gemm_fp8_v2_asymmetric_identity():
+    k = 128
+    a = torch.eye(32, k, device=DEV).to(torch.bfloat16)
+    w = torch.arange(32 * k, device=DEV, dtype=torch.float32) \
+        .reshape(32, k).to(torch.bfloat16) / (32 * k)
+    a8, as_ = ops.quant_fp8(a)
+    w8, ws = ops.quant_fp8(w)
+    got = ops.skinny_gemm_fp8_v2(a8, as_, w8, ws).float().cpu()
+    ref = _dequant(a8.cpu(), as_.cpu()) @ _dequant(w8.cpu(), ws.cpu()).t()
+    assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), \
+        (got - ref).abs().max()
