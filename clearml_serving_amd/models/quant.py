@@ -59,19 +59,36 @@ class Fp8Linear(nn.Module):
     # LDS-staged v2 covers M 17..V2_MAX_M (measured vs hipBLASLt scaled_mm)
     V2_MAX_M = int(__import__("os").environ.get("CMLS_FP8_V2_MAX", 64))
 
+    def _route(self, m: int) -> str:
+        """Measured per-shape routing (profiles/fp8_kernels.txt):
+        - v1 (8-wave k-split, swizzled weights) owns M<=16 -- except the
+          huge-N gate_up shape above M=1, where hipBLASLt's ~22 us floor
+          wins
+        - v2 (LDS-staged, 4 waves) owns M 17..64 on the qkv/o_proj-class
+          shapes (N < 16k, K < 8k)
+        - hipBLASLt scaled_mm keeps huge-N / huge-K shapes at M >= 17 and
+          everything above 64."""
+        n, k = self.out_features, self.in_features
+        if not (k % 64 == 0 and n % 16 == 0):
+            return "lt"
+        if m <= self.SKINNY_MAX_M:
+            if n >= 16384 and m > 1:
+                return "lt"
+            return "v1"
+        if m <= self.V2_MAX_M and k % 128 == 0 and n < 16384 and k < 8192:
+            return "v2"
+        return "lt"
+
     def forward_q(self, x8: torch.Tensor, x_scale: torch.Tensor,
                   out_dtype=None) -> torch.Tensor:
         out_dtype = out_dtype or self.out_dtype
         m = x8.shape[0]
-        if (m <= self.SKINNY_MAX_M and x8.is_cuda
-                and self.in_features % 64 == 0
-                and self.out_features % 16 == 0):
+        route = self._route(m) if x8.is_cuda else "lt"
+        if route == "v1":
             out = ops.skinny_gemm_fp8(x8, x_scale, self.weight_fp8_sw,
                                       self.weight_scale, swizzled=True)
             return out if out.dtype == out_dtype else out.to(out_dtype)
-        if (m <= self.V2_MAX_M and x8.is_cuda
-                and self.in_features % 128 == 0
-                and self.out_features % 16 == 0):
+        if route == "v2":
             out = ops.skinny_gemm_fp8_v2(x8, x_scale, self.weight_fp8,
                                          self.weight_scale)
             return out if out.dtype == out_dtype else out.to(out_dtype)
