@@ -50,8 +50,9 @@ class EngineOwner:
         ]
         self._engines: Dict[str, Any] = {}
         self._engine_eps: Dict[str, dict] = {}
+        self._inflight: Dict[int, "asyncio.Task"] = {}
         self._stop = False
-        self.stats = {"requests": 0, "errors": 0}
+        self.stats = {"requests": 0, "errors": 0, "aborts": 0}
 
     # ------------------------------------------------------------------ #
     def _get_engine(self, url: str):
@@ -121,6 +122,19 @@ class EngineOwner:
                     await self._push(worker, pack_stream_chunk(req_id, chunk))
             await self._push(worker, pack_stream_end(req_id))
             self.stats["requests"] += 1
+        except asyncio.CancelledError:
+            # abort record: close the generator explicitly -- the cancel may
+            # land while suspended in _push, which unwinds the async-for
+            # WITHOUT finalizing the generator, so the engine's
+            # abort-on-disconnect (generate()'s finally) would otherwise
+            # wait for GC while generation runs to max_tokens
+            aclose = getattr(response.body_iterator, "aclose", None)
+            if aclose is not None:
+                try:
+                    await aclose()
+                except Exception:
+                    pass
+            raise
         except Exception as ex:
             traceback.print_exc()
             self.stats["errors"] += 1
@@ -133,6 +147,12 @@ class EngineOwner:
         req_id = None
         try:
             req_id, url, data = unpack_request(raw)
+            if url is None:  # abort record: cancel the in-flight request
+                task = self._inflight.get(req_id)
+                if task is not None and not task.done():
+                    task.cancel()
+                    self.stats["aborts"] += 1
+                return
             serve_type = "process"
             if isinstance(data, dict) and "__serve_type__" in data:
                 serve_type = data["__serve_type__"]
@@ -155,6 +175,8 @@ class EngineOwner:
                 return
             resp = pack_response(req_id, out)
             self.stats["requests"] += 1
+        except asyncio.CancelledError:
+            return  # aborted by the client: nothing to send back
         except Exception as ex:
             traceback.print_exc()
             self.stats["errors"] += 1
@@ -171,7 +193,19 @@ class EngineOwner:
             for w, ring in enumerate(self.req_rings):
                 for raw in ring.drain(512):
                     got = True
-                    asyncio.ensure_future(self._handle(raw, w))
+                    import struct as _struct
+
+                    rid, ulen = _struct.unpack_from("<QH", raw, 0)
+                    task = asyncio.ensure_future(self._handle(raw, w))
+                    # abort records carry the SAME req_id as the request
+                    # they cancel: registering them would clobber (and
+                    # self-cancel) the generation task's _inflight entry
+                    is_abort = (ulen == 0 and len(raw) > 10
+                                and raw[10] == 2)  # KIND_ABORT
+                    if not is_abort:
+                        self._inflight[rid] = task
+                        task.add_done_callback(
+                            lambda t, r=rid: self._inflight.pop(r, None))
             if got:
                 await asyncio.sleep(0)
             else:

@@ -127,6 +127,19 @@ def main(argv=None) -> None:
                     break
                 spawn = _spawn_owner if kind == "owner" else _spawn_front
                 children[key] = spawn(args, prefix, idx)
+                if kind == "owner":
+                    # an owner restart re-creates its rings: the fronts'
+                    # mappings of the old rings are orphaned, so restart
+                    # them too (they re-attach to the fresh rings)
+                    for (k2, i2), p2 in list(children.items()):
+                        if k2 == "front" and p2.poll() is None:
+                            p2.terminate()
+                            try:
+                                p2.wait(timeout=10)
+                            except subprocess.TimeoutExpired:
+                                p2.kill()
+                            children[("front", i2)] = _spawn_front(
+                                args, prefix, i2)
     finally:
         for proc in children.values():
             if proc.poll() is None:

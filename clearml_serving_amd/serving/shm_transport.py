@@ -171,6 +171,7 @@ _CODE_DTYPES = {v: k for k, v in _DTYPE_CODES.items()}
 
 KIND_TENSORS = 0
 KIND_PICKLE = 1
+KIND_ABORT = 2   # client gone: cancel the in-flight request (req_id only)
 STATUS_OK_TENSORS = 0
 STATUS_OK_PICKLE = 1
 STATUS_ERROR = 2
@@ -271,6 +272,8 @@ def unpack_request(buf: bytes) -> Tuple[int, str, Any]:
     off += ulen
     kind = buf[off]
     off += 1
+    if kind == KIND_ABORT:
+        return req_id, None, None
     if kind == KIND_TENSORS:
         pairs = unpack_tensors(buf, off)
         if len(pairs) == 1 and pairs[0][0] == "":
@@ -288,6 +291,10 @@ def pack_response(req_id: int, result: Any = None,
     if pairs is not None:
         return head + bytes([STATUS_OK_TENSORS]) + pack_tensors(pairs)
     return head + bytes([STATUS_OK_PICKLE]) + pickle.dumps(result, protocol=4)
+
+
+def pack_abort(req_id: int) -> bytes:
+    return struct.pack("<QH", req_id, 0) + bytes([KIND_ABORT])
 
 
 def pack_stream_chunk(req_id: int, chunk: bytes) -> bytes:
@@ -356,7 +363,13 @@ class ShmClient:
             for ring in self.resp_rings:
                 for raw in ring.drain(512):
                     got = True
-                    req_id, status, payload = unpack_response(raw)
+                    try:
+                        req_id, status, payload = unpack_response(raw)
+                    except Exception:  # malformed record must not kill the
+                        import traceback  # poller (every future would hang)
+
+                        traceback.print_exc()
+                        continue
                     if status in (STATUS_STREAM_CHUNK, STATUS_STREAM_END):
                         q = self._streams.get(req_id)
                         if q is not None:
@@ -419,6 +432,12 @@ class ShmClient:
                 yield chunk
         finally:
             self._streams.pop(req_id, None)
+            # client gone mid-stream: tell the owner to cancel generation
+            # (best effort; the owner otherwise runs to max_tokens)
+            try:
+                ring.push(pack_abort(req_id))
+            except Exception:
+                pass
 
     def close(self):
         if self._poller is not None:

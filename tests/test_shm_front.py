@@ -382,3 +382,76 @@ def test_shmring_asan_ubsan_tsan():
         r = subprocess.run([exe2], capture_output=True, text=True,
                            timeout=180)
         assert r.returncode == 0, r.stdout + r.stderr
+
+
+@pytest.mark.timeout(120)
+def test_owner_cancels_generation_on_stream_abort(tmp_path):
+    """Closing an SSE consumer mid-stream sends an abort record over the
+    request ring; the owner cancels the in-flight generation (the engine
+    does not run to max_tokens for a dead client)."""
+    import json as _json
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.engine_owner import EngineOwner
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.serving.shm_transport import ShmClient
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="abrt", force_create=True)
+    card = tmp_path / "card.json"
+    card.write_text(_json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 512, "device": "cpu"}))
+    rec = store.register_model(name="t", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(engine_type="llm", serving_url="tl",
+                                    model_id=rec.model_id))
+    proc.serialize()
+
+    prefix = "/cmls_abrt_{}".format(os.getpid())
+    owner = EngineOwner(store_root=store_root, session_id=proc.get_id(),
+                        prefix=prefix, owner_idx=0, n_workers=1,
+                        ring_bytes=1 << 20)
+    client = ShmClient(prefix, 0, 1, ring_bytes=1 << 20)
+
+    async def main():
+        serve_task = asyncio.get_running_loop().create_task(owner.serve())
+        payload = {"__serve_type__": "v1_chat_completions",
+                   "__body__": {"messages": [{"role": "user",
+                                              "content": "hi"}],
+                                "stream": True, "max_tokens": 400,
+                                "temperature": 0, "ignore_eos": True}}
+        agen = client.infer_stream("tl", payload)
+        chunks = 0
+        async for _chunk in agen:
+            chunks += 1
+            if chunks >= 2:
+                await agen.aclose()  # client disconnect -> abort record
+                break
+        # the owner should cancel; its llm engine records the abort
+        eng = owner._engines["tl"]._engine
+        for _ in range(200):
+            await asyncio.sleep(0.05)
+            if owner.stats["aborts"] >= 1 and eng.stats["aborts"] >= 1 \
+                    and not eng.running and not eng.waiting:
+                break
+        assert owner.stats["aborts"] >= 1
+        assert eng.stats["aborts"] >= 1
+        assert not eng.running and not eng.waiting
+        assert eng.stats["generated_tokens"] < 300  # nowhere near 400
+        serve_task.cancel()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
+        client.close()
+        owner.close()
