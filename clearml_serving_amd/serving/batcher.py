@@ -297,6 +297,14 @@ class DynamicBatcher:
         if ts[0].dtype != want:
             ts = [t.to(want) for t in ts]
         if self.is_cuda:
+            if ts[0].is_cuda:
+                # GPU-resident inputs (user preprocess ran on device): stack
+                # on device, skip the pinned host staging entirely
+                t = torch.stack(ts, dim=0)
+                if t.shape[0] < bucket:
+                    pad = t[:1].expand(bucket - t.shape[0], *t.shape[1:])
+                    t = torch.cat([t, pad], dim=0)
+                return t
             # fill this slot's reusable pinned host slab (the H2D copy is
             # enqueued later on the stream, inside the enqueue lock); one
             # max-batch slab per (slot, input) serves every bucket
