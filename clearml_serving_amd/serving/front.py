@@ -190,6 +190,13 @@ def main(argv=None) -> None:
     setup_front(args.prefix, args.worker, args.owners,
                 ring_bytes=args.ring_mb << 20)
 
+    # per-front Prometheus export: worker w serves metrics on
+    # STATS_PORT + w (a single shared port would only expose one front's
+    # sample; Kafka mode aggregates across processes like the reference)
+    base = int(os.environ.get("CLEARML_SERVING_STATS_PORT", 9999))
+    if base > 0:
+        os.environ["CLEARML_SERVING_STATS_PORT"] = str(base + args.worker)
+
     import uvicorn
 
     from .app import create_app
