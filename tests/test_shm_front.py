@@ -455,3 +455,83 @@ def test_owner_cancels_generation_on_stream_abort(tmp_path):
         loop.close()
         client.close()
         owner.close()
+
+
+@pytest.mark.timeout(120)
+def test_owner_engine_swaps_on_hot_reload(tmp_path):
+    """A config change (new model version for the same endpoint) propagates
+    to the engine owner's sync daemon and the owner rebuilds the engine --
+    the SHM-topology version of the zero-downtime hot reload."""
+    import json as _json
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.engine_owner import EngineOwner
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.serving.shm_transport import ShmClient
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="hotswap",
+                                 force_create=True)
+
+    def register_card(labels, name):
+        card = tmp_path / (name + ".json")
+        card.write_text(_json.dumps({
+            "arch": "bert-base", "num_labels": labels, "dtype": "float32",
+            "vocab_size": 300}))
+        return store.register_model(name=name, project="p", path=str(card))
+
+    rec1 = register_card(2, "v1")
+    proc.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="enc", model_id=rec1.model_id,
+        auxiliary_cfg={"use_graphs": False, "warmup": False,
+                       "max_queue_delay_us": 500}))
+    proc.serialize()
+
+    prefix = "/cmls_swap_{}".format(os.getpid())
+    owner = EngineOwner(store_root=store_root, session_id=proc.get_id(),
+                        prefix=prefix, owner_idx=0, n_workers=1,
+                        ring_bytes=1 << 20, poll_frequency_sec=0.5)
+    client = ShmClient(prefix, 0, 1, ring_bytes=1 << 20)
+
+    async def main():
+        serve_task = asyncio.get_running_loop().create_task(owner.serve())
+        body = {"input_ids": np.arange(1, 17, dtype=np.int64),
+                "attention_mask": np.ones(16, dtype=np.int64)}
+        out1 = await client.infer("enc", body)
+        assert np.asarray(out1).shape == (2,)
+        first_engine = owner._engines["enc"]
+
+        # hot swap: same endpoint, new model (3 labels)
+        rec2 = register_card(3, "v2")
+        proc.add_endpoint(ModelEndpoint(
+            engine_type="hip", serving_url="enc", model_id=rec2.model_id,
+            auxiliary_cfg={"use_graphs": False, "warmup": False,
+                          "max_queue_delay_us": 500}))
+        proc.serialize()
+
+        # owner's sync daemon polls every 0.5 s; the next request after the
+        # sync must serve the NEW model
+        for _ in range(60):
+            await asyncio.sleep(0.25)
+            out = await client.infer("enc", body)
+            if np.asarray(out).shape == (3,):
+                break
+        assert np.asarray(out).shape == (3,)
+        assert owner._engines["enc"] is not first_engine
+        serve_task.cancel()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
+        client.close()
+        owner.close()
