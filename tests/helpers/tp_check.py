@@ -117,6 +117,15 @@ def check_engine_protocol(rank, world):
             eng.embed_batch(["tp embed a", "a longer tp embed text b"]))
         assert len(vecs) == 2
         assert abs(sum(x * x for x in vecs[0]) - 1.0) < 1e-4
+
+        # oversize embed batch: exceeds the fixed plan buffer, so the
+        # codec's OBJECT fallback broadcast carries it (plan_codec mode 5)
+        assert eng._plan_codec.encode(
+            {"mode": "embed",
+             "prompts": [[1] * 100 for _ in range(400)]}) is None
+        big = ["x" * 100 for _ in range(400)]
+        vecs2 = loop.run_until_complete(eng.embed_batch(big))
+        assert len(vecs2) == 400
         eng.tp_shutdown()
         print("TP-ENGINE-OK", flush=True)
     else:
