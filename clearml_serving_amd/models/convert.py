@@ -42,13 +42,13 @@ def convert_hf_bert(hf: Dict[str, torch.Tensor],
         out[o + "attn.qkv.weight"] = torch.cat([q_w, k_w, v_w], dim=0)
         out[o + "attn.qkv.bias"] = torch.cat([q_b, k_b, v_b], dim=0)
         out[o + "attn.out.weight"] = g(p + "attention.output.dense.weight")
-        out[o + "attn.out_bias"] = g(p + "attention.output.dense.bias")
+        out[o + "attn.out.bias"] = g(p + "attention.output.dense.bias")
         out[o + "ln1_w"] = g(p + "attention.output.LayerNorm.weight")
         out[o + "ln1_b"] = g(p + "attention.output.LayerNorm.bias")
         out[o + "fc1.weight"] = g(p + "intermediate.dense.weight")
         out[o + "fc1_bias"] = g(p + "intermediate.dense.bias")
         out[o + "fc2.weight"] = g(p + "output.dense.weight")
-        out[o + "fc2_bias"] = g(p + "output.dense.bias")
+        out[o + "fc2.bias"] = g(p + "output.dense.bias")
         out[o + "ln2_w"] = g(p + "output.LayerNorm.weight")
         out[o + "ln2_b"] = g(p + "output.LayerNorm.bias")
     return out
