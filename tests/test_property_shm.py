@@ -1,0 +1,82 @@
+"""Property-based tests (hypothesis) for the SHM ring record format and the
+TP plan codec -- randomized sequences beyond the example-based tests."""
+
+import uuid
+
+import pytest
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from clearml_serving_amd.serving import shm_transport as tr
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.binary(min_size=0, max_size=700), min_size=1,
+                max_size=60))
+def test_ring_preserves_arbitrary_record_sequences(payloads):
+    name = "/cmls_prop_{}".format(uuid.uuid4().hex[:10])
+    w = tr.make_ring(name, 1 << 14, True)
+    r = tr.make_ring(name, 0, False)
+    try:
+        got = []
+        i = 0
+        while i < len(payloads):
+            # push a few, drain a few -- random interleave shapes
+            burst = payloads[i:i + 7]
+            for p in burst:
+                while not w.push(p):
+                    got.extend(r.drain(64))
+            i += len(burst)
+            got.extend(r.drain(3))
+        while True:
+            batch = r.drain(64)
+            if not batch:
+                break
+            got.extend(batch)
+        assert got == payloads
+    finally:
+        w.close()
+        r.close()
+        tr.unlink_ring(name)
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    b=st.integers(min_value=1, max_value=8),
+    seed=st.integers(min_value=0, max_value=2**31 - 1),
+    data=st.data(),
+)
+def test_plan_codec_decode_roundtrip_random(b, seed, data):
+    from clearml_serving_amd.engines.llm.engine import LlmEngineConfig
+    from clearml_serving_amd.engines.llm.plan_codec import PlanCodec
+
+    cfg = LlmEngineConfig(preset="llama-tiny", max_num_seqs=8,
+                          max_model_len=256, block_size=16,
+                          max_prefill_tokens=128, prefill_chunk=64)
+    codec = PlanCodec(cfg, torch.device("cpu"))
+    plan = {
+        "mode": "decode",
+        "tokens": [data.draw(st.integers(0, 511)) for _ in range(b)],
+        "positions": [data.draw(st.integers(0, 255)) for _ in range(b)],
+        "slots": [data.draw(st.integers(0, 4095)) for _ in range(b)],
+        "seq_lens": [data.draw(st.integers(1, 256)) for _ in range(b)],
+        "blocks": [[data.draw(st.integers(0, 255))
+                    for _ in range(data.draw(st.integers(1, 16)))]
+                   for _ in range(b)],
+        "sample": [(data.draw(st.floats(0, 4, allow_nan=False,
+                                        allow_subnormal=False, width=32)),
+                    data.draw(st.integers(0, 100)),
+                    data.draw(st.floats(0.01, 1.0, allow_nan=False,
+                                        allow_subnormal=False, width=32)),
+                    seed) for _ in range(b)],
+    }
+    buf = codec.encode(plan)
+    assert buf is not None
+    out = codec.decode(buf)
+    for k in ("tokens", "positions", "slots", "seq_lens", "blocks"):
+        assert out[k] == plan[k], k
+    for got, want in zip(out["sample"], plan["sample"]):
+        assert got[1] == want[1] and got[3] == want[3]
+        assert got[0] == pytest.approx(want[0], abs=1e-6)
+        assert got[2] == pytest.approx(want[2], abs=1e-6)
