@@ -67,7 +67,7 @@ def test_plan_codec_decode_roundtrip_random(b, seed, data):
         "sample": [(data.draw(st.floats(0, 4, allow_nan=False,
                                         allow_subnormal=False, width=32)),
                     data.draw(st.integers(0, 100)),
-                    data.draw(st.floats(0.01, 1.0, allow_nan=False,
+                    data.draw(st.floats(0.0625, 1.0, allow_nan=False,
                                         allow_subnormal=False, width=32)),
                     seed) for _ in range(b)],
     }
