@@ -48,6 +48,7 @@ class StatsRegistry:
     """Dynamic Prometheus metric creation per endpoint/variable."""
 
     def __init__(self, processor=None, registry=None):
+        self._observe_warned = set()
         self._processor = processor
         self._registry = registry or REGISTRY
         self._metrics: Dict[str, object] = {}
@@ -122,8 +123,14 @@ class StatsRegistry:
                     continue
                 try:
                     self._observe(metric, conf["type"], value)
-                except Exception:
-                    pass
+                except Exception as ex:
+                    # stats must never break serving, but a silently
+                    # mis-typed metric is an advisor trap: warn once per key
+                    key = (endpoint, variable)
+                    if key not in self._observe_warned:
+                        self._observe_warned.add(key)
+                        print("[stats] dropping metric {}:{} ({}: {})".format(
+                            endpoint, variable, type(ex).__name__, ex))
 
     @staticmethod
     def _observe(metric, mtype: str, value) -> None:
