@@ -153,6 +153,9 @@ class LlmEngineConfig:
     gpu_memory_fraction: float = 0.85
     num_kv_blocks: Optional[int] = None  # explicit override (CPU tests)
     quantization: Optional[str] = None   # "fp8": fp8 weights via hipBLASLt
+    kv_dtype: str = "bfloat16"  # "fp8": e4m3 KV cache + per-token-per-head
+                                # scales -- 2x cached tokens per HBM byte,
+                                # half the KV reads on long-context decode
     weights: Optional[str] = None
     tokenizer_path: Optional[str] = None
     device: Optional[str] = None
@@ -186,7 +189,7 @@ class LlmEngineConfig:
         for key in ("preset", "dtype", "block_size", "max_num_seqs",
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "quantization",
-                    "weights", "device", "decode_graphs"):
+                    "kv_dtype", "weights", "device", "decode_graphs"):
             for src in (card, aux):
                 if key in src and src[key] is not None:
                     setattr(cfg, key, src[key])
@@ -321,8 +324,11 @@ class LlmEngine:
         # KV cache sizing from free HBM (heads sharded under TP)
         bs = cfg.block_size
         kv_heads = mcfg.kv_heads // self.tp_size
-        per_block_bytes = (2 * kv_heads * bs * mcfg.head_dim *
-                           self.dtype.itemsize * mcfg.layers)
+        fp8_kv = str(cfg.kv_dtype).lower() in ("fp8", "float8", "e4m3")
+        kv_itemsize = 1 if fp8_kv else self.dtype.itemsize
+        per_block_bytes = (2 * kv_heads * bs *
+                           (mcfg.head_dim * kv_itemsize +
+                            (4 if fp8_kv else 0)) * mcfg.layers)
         if cfg.num_kv_blocks:
             num_blocks = int(cfg.num_kv_blocks)
         elif self.device.type == "cuda":
@@ -341,13 +347,29 @@ class LlmEngine:
             dist.all_reduce(t, op=dist.ReduceOp.MIN)
             num_blocks = int(t.item())
         self.allocator = BlockAllocator(num_blocks)
-        self.kv_caches = [
-            (torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
-                         dtype=self.dtype, device=self.device),
-             torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
-                         dtype=self.dtype, device=self.device))
-            for _ in range(mcfg.layers)
-        ]
+        if fp8_kv:
+            # 4-tuple per layer: e4m3 byte caches + f32 [NB, Hkv, BS]
+            # per-token-per-head scales (written by kv_cache_write's
+            # wave-per-head absmax quantizer, attention_decode.hip)
+            self.kv_caches = [
+                (torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
+                             dtype=torch.uint8, device=self.device),
+                 torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
+                             dtype=torch.uint8, device=self.device),
+                 torch.ones(num_blocks, kv_heads, bs,
+                            dtype=torch.float32, device=self.device),
+                 torch.ones(num_blocks, kv_heads, bs,
+                            dtype=torch.float32, device=self.device))
+                for _ in range(mcfg.layers)
+            ]
+        else:
+            self.kv_caches = [
+                (torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
+                             dtype=self.dtype, device=self.device),
+                 torch.zeros(num_blocks, kv_heads, bs, mcfg.head_dim,
+                             dtype=self.dtype, device=self.device))
+                for _ in range(mcfg.layers)
+            ]
         if self.tp_size > 1:
             from .plan_codec import PlanCodec
 

@@ -104,8 +104,14 @@ class LlamaLayer(nn.Module):
         q, k = ops.rope_inplace(q, k, positions, theta=cfg.rope_theta)
 
         if kv_cache is not None:
-            k_cache, v_cache = kv_cache
-            ops.kv_cache_write(k, v, k_cache, v_cache, attn_ctx["slot_mapping"])
+            # 2-tuple (k, v) bf16 cache, or 4-tuple (k8, v8, k_scale,
+            # v_scale) fp8 cache (engine cfg kv_dtype="fp8"): e4m3 bytes +
+            # per-token-per-head scales, halving KV reads on long decodes
+            k_cache, v_cache = kv_cache[0], kv_cache[1]
+            k_sc = kv_cache[2] if len(kv_cache) > 2 else None
+            v_sc = kv_cache[3] if len(kv_cache) > 2 else None
+            ops.kv_cache_write(k, v, k_cache, v_cache,
+                               attn_ctx["slot_mapping"], k_sc, v_sc)
 
         scale = 1.0 / math.sqrt(self.head_dim)
         if attn_ctx["mode"] == "prefill":
@@ -119,18 +125,16 @@ class LlamaLayer(nn.Module):
         elif attn_ctx["mode"] == "prefill_paged":
             # chunked prefill: this chunk's queries attend to the full paged
             # history (K/V of the chunk were just scattered into the cache)
-            k_cache, v_cache = kv_cache
             b, s = attn_ctx["batch"], attn_ctx["seq"]
             ctx = ops.attention_prefill_paged(
                 q.unflatten(0, (b, s)), k_cache, v_cache,
                 attn_ctx["block_table"], attn_ctx["kv_lens"],
-                attn_ctx["q_lens"], scale=scale)
+                attn_ctx["q_lens"], scale=scale, k_scale=k_sc, v_scale=v_sc)
             ctx = ctx.reshape(t, self.heads * self.head_dim)
         else:  # decode: one token per sequence
-            k_cache, v_cache = kv_cache
             ctx = ops.attention_decode(
                 q, k_cache, v_cache, attn_ctx["block_table"],
-                attn_ctx["seq_lens"], scale=scale)
+                attn_ctx["seq_lens"], scale=scale, k_scale=k_sc, v_scale=v_sc)
             ctx = ctx.view(t, self.heads * self.head_dim)
         if fp8:
             # attention context has no fused producer: one-pass dynamic quant

@@ -357,14 +357,27 @@ def attention(
     return torch.matmul(probs, vf).to(q.dtype)
 
 
+def _dequant_kv_cpu(cache: torch.Tensor, scale: torch.Tensor) -> torch.Tensor:
+    """uint8 e4m3 paged cache [NB, Hkv, BS, D] + scales [NB, Hkv, BS] ->
+    float cache (CPU reference path for the fp8 KV cache)."""
+    f = cache.view(torch.float8_e4m3fn).float()
+    return f * scale[..., None]
+
+
 def attention_prefill_paged(
     q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
     block_table: torch.Tensor, kv_lens: torch.Tensor, q_lens: torch.Tensor,
     scale: Optional[float] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Chunked-prefill attention: chunk queries [B, Sq, H, D] (bshd) attend
     causally to the full PAGED history (kv_lens keys per seq, already in the
-    cache); q_lens masks per-seq chunk padding. Output [B, Sq, H, D]."""
+    cache); q_lens masks per-seq chunk padding. Output [B, Sq, H, D].
+
+    fp8 KV: uint8 e4m3 caches + per-token-per-head ``k_scale``/``v_scale``
+    float32 [NB, Hkv, BS]; the gfx950 kernel dequantizes while staging tiles
+    to LDS (attention_v2.hip FP8KV)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
@@ -373,10 +386,18 @@ def attention_prefill_paged(
             if os.environ.get("CMLS_ATTN_V2", "1") != "0":
                 return ext.attention_prefill_paged_v2(
                     q, k_cache, v_cache, block_table, kv_lens, q_lens,
-                    float(scale))
+                    float(scale), k_scale, v_scale)
+            if k_scale is not None:
+                raise RuntimeError("fp8 KV prefill needs the v2 kernel "
+                                   "(CMLS_ATTN_V2=0 set?)")
             return ext.attention_prefill_paged(
                 q, k_cache, v_cache, block_table, kv_lens, q_lens,
                 float(scale))
+    if k_scale is not None:
+        return attention_prefill_paged(
+            q, _dequant_kv_cpu(k_cache, k_scale),
+            _dequant_kv_cpu(v_cache, v_scale), block_table, kv_lens, q_lens,
+            scale)
     # reference path: gather pages -> dense causal attention with history
     b, sq, h, d = q.shape
     hkv = k_cache.shape[1]
@@ -402,6 +423,8 @@ def attention_decode(
     q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
     block_table: torch.Tensor, seq_lens: torch.Tensor,
     scale: Optional[float] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Paged-KV decode attention: one new token per sequence.
 
@@ -410,6 +433,9 @@ def attention_decode(
     v_cache:  [num_blocks, H_kv, block_size, D]
     block_table: int32 [B, max_blocks] physical block ids per sequence
     seq_lens: int32 [B] total keys per sequence (including current token)
+
+    fp8 KV: uint8 e4m3 caches + float32 [NB, Hkv, BS] per-token scales;
+    halves the decode KV-read bytes (the long-context bandwidth bound).
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
@@ -417,7 +443,12 @@ def attention_decode(
         ext = _require_ext("attention_decode")
         if ext is not None:
             return ext.attention_decode(q, k_cache, v_cache, block_table,
-                                        seq_lens, float(scale))
+                                        seq_lens, float(scale),
+                                        k_scale, v_scale)
+    if k_scale is not None:
+        return attention_decode(
+            q, _dequant_kv_cpu(k_cache, k_scale),
+            _dequant_kv_cpu(v_cache, v_scale), block_table, seq_lens, scale)
     # reference path: gather pages then dense attention per sequence
     bsz, hq, d = q.shape
     hkv = k_cache.shape[1]
@@ -445,22 +476,39 @@ def attention_decode(
 def kv_cache_write(
     k_new: torch.Tensor, v_new: torch.Tensor, k_cache: torch.Tensor,
     v_cache: torch.Tensor, slot_mapping: torch.Tensor,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> None:
     """Scatter [T, H_kv, D] new keys/values into the paged caches at flat
-    slots (block_id * block_size + offset); slot -1 skips the token."""
+    slots (block_id * block_size + offset); slot -1 skips the token.
+
+    fp8 KV: pass uint8 caches + float32 [NB, Hkv, BS] scale tensors; each
+    (token, head) row is absmax-quantized to e4m3 and its scale recorded."""
     if k_new.is_cuda:
         ext = _require_ext("kv_cache_write")
         if ext is not None:
-            ext.kv_cache_write(k_new, v_new, k_cache, v_cache, slot_mapping)
+            ext.kv_cache_write(k_new, v_new, k_cache, v_cache, slot_mapping,
+                               k_scale, v_scale)
             return
     block_size = k_cache.shape[2]
+    fp8 = k_scale is not None
     for t in range(k_new.shape[0]):
         slot = int(slot_mapping[t])
         if slot < 0:
             continue
         blk, off = slot // block_size, slot % block_size
-        k_cache[blk, :, off] = k_new[t]
-        v_cache[blk, :, off] = v_new[t]
+        if fp8:
+            for cache, scales, new in ((k_cache, k_scale, k_new),
+                                       (v_cache, v_scale, v_new)):
+                row = new[t].float()  # [H_kv, D]
+                amax = row.abs().amax(dim=-1).clamp_min(1e-8)
+                sc = amax / 448.0
+                q8 = (row / sc[:, None]).to(torch.float8_e4m3fn)
+                cache[blk, :, off] = q8.view(torch.uint8)
+                scales[blk, :, off] = sc
+        else:
+            k_cache[blk, :, off] = k_new[t]
+            v_cache[blk, :, off] = v_new[t]
 
 
 # --------------------------------------------------------------------- #

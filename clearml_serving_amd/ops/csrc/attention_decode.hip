@@ -21,6 +21,14 @@
 
 namespace {
 
+__device__ __forceinline__ void fp8x4_to_f32(int w, float* o) {
+  // word-select of cvt_pk_f32_fp8 must be a literal constant
+  typedef __attribute__((ext_vector_type(2))) float adf32x2_t;
+  adf32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  adf32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  o[0] = lo[0]; o[1] = lo[1]; o[2] = hi[0]; o[3] = hi[1];
+}
+
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(2))) short bf16x2_t;
 
@@ -41,11 +49,20 @@ __device__ __forceinline__ float group_sum4(float v) {
 // unroll, 148 VGPR -> 3 waves/SIMD for GQ=4).  <.., 2, 4> is the
 // occupancy-first alternative (116 VGPR -> 4 waves/SIMD at half the loads
 // in flight per wave); scripts/decode_ab.hip A/Bs them.
-template <int D, int GQ, int UNROLL = 4, int MINB = 2>
+typedef __attribute__((ext_vector_type(2))) float f32x2_t;
+typedef __attribute__((ext_vector_type(2))) unsigned int uint32x2_t;
+
+// FP8KV: caches hold OCP-e4m3 bytes with per-token-per-head scales
+// ([NB, Hkv, BS] f32) -- halves the KV HBM stream AND doubles the cached
+// tokens per GB; scores/values dequantize inline (VALU headroom exists:
+// the kernel is HBM-bound at 3.4-4.9 TB/s).
+template <int D, int GQ, int UNROLL = 4, int MINB = 2, bool FP8KV = false>
 __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
     const __hip_bfloat16* __restrict__ q,        // [B, H, D]
-    const __hip_bfloat16* __restrict__ k_cache,  // [NB, Hkv, BS, D]
-    const __hip_bfloat16* __restrict__ v_cache,  // [NB, Hkv, BS, D]
+    const void* __restrict__ k_cache,            // [NB, Hkv, BS, D] bf16|e4m3
+    const void* __restrict__ v_cache,
+    const float* __restrict__ k_scale,           // [NB, Hkv, BS] (FP8KV)
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_table,         // [B, max_blocks]
     const int* __restrict__ seq_lens,            // [B]
     __hip_bfloat16* __restrict__ out,            // [B, H, D]
@@ -86,6 +103,7 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
   // q fragments: packed bf16 pairs feed v_dot2_f32_bf16 (2 MACs per
   // instruction in the K dot products; scale folds into the score later)
   bf16x2_t qp[GQ][EPL / 2];
+  float qf[FP8KV ? GQ : 1][FP8KV ? EPL : 1];  // plain-FMA path for fp8 K
 #pragma unroll
   for (int g = 0; g < GQ; ++g) {
     const int h = hkv * GQ + g;
@@ -94,6 +112,10 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
 #pragma unroll
     for (int e = 0; e < EPL / 2; ++e)
       qp[g][e] = *reinterpret_cast<const bf16x2_t*>(qrow + 2 * e);
+    if (FP8KV) {
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) qf[g][e] = to_f32(qrow[e]);
+    }
   }
 
   float m_run[GQ], l_run[GQ], o_acc[GQ][EPL];
@@ -116,6 +138,8 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
   const int n_iters = (n_keys - key_lo + per_iter - 1) / per_iter;
   for (int it0 = 0; it0 < n_iters; it0 += UNROLL) {
     bf16x8 kvec[UNROLL], vvec[UNROLL];
+    uint32x2_t k8v[FP8KV ? UNROLL : 1], v8v[FP8KV ? UNROLL : 1];
+    float ksc[FP8KV ? UNROLL : 1], vsc[FP8KV ? UNROLL : 1];
     bool valid[UNROLL];
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
@@ -125,23 +149,53 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
       const int blk = btab[kc / block_size];
       const long base = ((long)blk * Hkv) * block_size * D + hk_off +
                         (long)(kc % block_size) * D;
-      kvec[u].u = *reinterpret_cast<const uint32x4*>(
-          k_cache + base + sub * EPL);
-      vvec[u].u = *reinterpret_cast<const uint32x4*>(
-          v_cache + base + sub * EPL);
+      if (FP8KV) {
+        const unsigned char* kcp = (const unsigned char*)k_cache;
+        const unsigned char* vcp = (const unsigned char*)v_cache;
+        k8v[u] = *reinterpret_cast<const uint32x2_t*>(
+            kcp + base + sub * EPL);
+        v8v[u] = *reinterpret_cast<const uint32x2_t*>(
+            vcp + base + sub * EPL);
+        const long sidx = ((long)blk * Hkv + hkv) * block_size +
+                          (kc % block_size);
+        ksc[u] = k_scale[sidx];
+        vsc[u] = v_scale[sidx];
+      } else {
+        const __hip_bfloat16* kcp = (const __hip_bfloat16*)k_cache;
+        const __hip_bfloat16* vcp = (const __hip_bfloat16*)v_cache;
+        kvec[u].u = *reinterpret_cast<const uint32x4*>(
+            kcp + base + sub * EPL);
+        vvec[u].u = *reinterpret_cast<const uint32x4*>(
+            vcp + base + sub * EPL);
+      }
     }
     // scores for the whole 4-iteration chunk (4x16 = 64 wave keys)
     float score[UNROLL][GQ];
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
-      const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kvec[u]);
+      if (FP8KV) {
+        // dequantized plain-FMA dot (score scale absorbs k_scale later)
+        float kf[EPL];
+        const int* ki = reinterpret_cast<const int*>(&k8v[u]);
 #pragma unroll
-      for (int g = 0; g < GQ; ++g) {
-        float acc = 0.f;
+        for (int w = 0; w < EPL / 4; ++w) fp8x4_to_f32(ki[w], kf + 4 * w);
 #pragma unroll
-        for (int e = 0; e < EPL / 2; ++e)
-          acc = __builtin_amdgcn_fdot2_f32_bf16(qp[g][e], kp[e], acc, false);
-        score[u][g] = acc;
+        for (int g = 0; g < GQ; ++g) {
+          float acc = 0.f;
+#pragma unroll
+          for (int e = 0; e < EPL; ++e) acc = fmaf(qf[g][e], kf[e], acc);
+          score[u][g] = acc;
+        }
+      } else {
+        const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kvec[u]);
+#pragma unroll
+        for (int g = 0; g < GQ; ++g) {
+          float acc = 0.f;
+#pragma unroll
+          for (int e = 0; e < EPL / 2; ++e)
+            acc = __builtin_amdgcn_fdot2_f32_bf16(qp[g][e], kp[e], acc, false);
+          score[u][g] = acc;
+        }
       }
       // group-level dot reduction (16 lanes hold partial sums)
 #pragma unroll
@@ -149,7 +203,8 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
           score[u][g] += __shfl_xor(score[u][g], off, 64);
-        score[u][g] = valid[u] ? score[u][g] * scale2 : -INFINITY;
+        const float s2 = FP8KV ? scale2 * ksc[u] : scale2;
+        score[u][g] = valid[u] ? score[u][g] * s2 : -INFINITY;
       }
     }
     // ONE online-softmax update per chunk per head (rescaling O per
@@ -172,9 +227,19 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
       for (int u = 0; u < UNROLL; ++u) {
         const float p = valid[u] ? __builtin_amdgcn_exp2f(score[u][g] - m_new) : 0.f;
         psum += p;
+        if (FP8KV) {
+          const float pv = p * vsc[u];
+          const int* vi = reinterpret_cast<const int*>(&v8v[u]);
+          float vf[EPL];
 #pragma unroll
-        for (int e = 0; e < EPL; ++e)
-          o_acc[g][e] += p * to_f32(vvec[u].h[e]);
+          for (int w = 0; w < EPL / 4; ++w) fp8x4_to_f32(vi[w], vf + 4 * w);
+#pragma unroll
+          for (int e = 0; e < EPL; ++e) o_acc[g][e] += pv * vf[e];
+        } else {
+#pragma unroll
+          for (int e = 0; e < EPL; ++e)
+            o_acc[g][e] += p * to_f32(vvec[u].h[e]);
+        }
       }
       l_run[g] = l_run[g] * alpha + group_sum4(psum);
       m_run[g] = m_new;
@@ -308,18 +373,78 @@ __global__ void kv_cache_write_kernel(const T* __restrict__ knew,  // [T,Hkv,D]
   }
 }
 
+template <typename T>
+__global__ void kv_cache_write_fp8_kernel(
+    const T* __restrict__ knew,  // [T, Hkv, D] (token-strided views ok)
+    const T* __restrict__ vnew,
+    unsigned char* __restrict__ k8,  // [NB, Hkv, BS, D] e4m3
+    unsigned char* __restrict__ v8,
+    float* __restrict__ ks,          // [NB, Hkv, BS]
+    float* __restrict__ vs,
+    const int* __restrict__ slots, int hkv, int d, int block_size,
+    long k_tstride, long v_tstride) {
+  const int t = blockIdx.x;
+  const int slot = slots[t];
+  if (slot < 0) return;
+  const long blk = slot / block_size;
+  const int off = slot % block_size;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  // one wave per head: per-(token, head) absmax -> scale -> quantize
+  for (int h = wave; h < hkv; h += (int)blockDim.x / 64) {
+    const T* krow = knew + (long)t * k_tstride + (long)h * d;
+    const T* vrow = vnew + (long)t * v_tstride + (long)h * d;
+    float kamax = 0.f, vamax = 0.f;
+    for (int i = lane; i < d; i += 64) {
+      kamax = fmaxf(kamax, fabsf(to_f32(krow[i])));
+      vamax = fmaxf(vamax, fabsf(to_f32(vrow[i])));
+    }
+    kamax = wave_reduce_max(kamax);
+    vamax = wave_reduce_max(vamax);
+    const float kscale = fmaxf(kamax, 1e-8f) / 448.0f;
+    const float vscale = fmaxf(vamax, 1e-8f) / 448.0f;
+    const long sidx = (blk * hkv + h) * block_size + off;
+    if (lane == 0) {
+      ks[sidx] = kscale;
+      vs[sidx] = vscale;
+    }
+    const float kinv = 1.0f / kscale, vinv = 1.0f / vscale;
+    unsigned char* kd = k8 + ((blk * hkv + h) * block_size + off) * (long)d;
+    unsigned char* vd = v8 + ((blk * hkv + h) * block_size + off) * (long)d;
+    for (int i = lane; i < d; i += 64) {
+      const float kq = to_f32(krow[i]) * kinv;
+      const float vq = to_f32(vrow[i]) * vinv;
+      kd[i] = __builtin_amdgcn_cvt_pk_fp8_f32(kq, kq, 0, false) & 0xff;
+      vd[i] = __builtin_amdgcn_cvt_pk_fp8_f32(vq, vq, 0, false) & 0xff;
+    }
+  }
+}
+
 }  // namespace
 
 #ifndef CMLS_KERNEL_ONLY
 torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                                torch::Tensor v_cache,
                                torch::Tensor block_table,
-                               torch::Tensor seq_lens, double scale) {
+                               torch::Tensor seq_lens, double scale,
+                               c10::optional<torch::Tensor> k_scale,
+                               c10::optional<torch::Tensor> v_scale) {
   TORCH_CHECK(q.dim() == 3, "q must be [B, H, D]");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attention: bf16 only");
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
               "q heads must be dense");
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  const bool fp8kv = k_cache.scalar_type() == at::kByte ||
+                     k_cache.scalar_type() == at::kFloat8_e4m3fn;
+  const float* ksp = nullptr;
+  const float* vsp = nullptr;
+  if (fp8kv) {
+    TORCH_CHECK(k_scale.has_value() && v_scale.has_value(),
+                "fp8 KV cache needs k_scale/v_scale");
+    TORCH_CHECK(k_scale->is_contiguous() && v_scale->is_contiguous());
+    ksp = k_scale->data_ptr<float>();
+    vsp = v_scale->data_ptr<float>();
+  }
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int Hkv = k_cache.size(1), BS = k_cache.size(2);
   const int max_blocks = block_table.size(1);
@@ -351,37 +476,33 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
   }
   dim3 grid(B, Hkv, splits);
 
-#define LAUNCH_DEC(DD, GG)                                                   \
-  hipLaunchKernelGGL((attn_decode_kernel<DD, GG>), grid, dim3(256), 0,       \
+#define LAUNCH_DEC(DD, GG, UU, MM, F8)                                       \
+  hipLaunchKernelGGL((attn_decode_kernel<DD, GG, UU, MM, F8>), grid,         \
+                     dim3(256), 0,                                           \
                      stream_, (const __hip_bfloat16*)q.data_ptr(),           \
-                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
-                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
+                     (const void*)k_cache.data_ptr(),                        \
+                     (const void*)v_cache.data_ptr(), ksp, vsp,              \
                      bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
                      (__hip_bfloat16*)out.data_ptr(), po, pml, splits,       \
                      q.stride(0), H, Hkv, BS, max_blocks, (float)scale)
-#define LAUNCH_DEC4(DD, GG, UU, MM)                                          \
-  hipLaunchKernelGGL((attn_decode_kernel<DD, GG, UU, MM>), grid, dim3(256),  \
-                     0, stream_, (const __hip_bfloat16*)q.data_ptr(),        \
-                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
-                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
-                     bt.data_ptr<int>(), sl.data_ptr<int>(),                 \
-                     (__hip_bfloat16*)out.data_ptr(), po, pml, splits,       \
-                     q.stride(0), H, Hkv, BS, max_blocks, (float)scale)
+#define LAUNCH_DEC_F(DD, GG, UU, MM)                                         \
+  do { if (fp8kv) LAUNCH_DEC(DD, GG, UU, MM, true);                          \
+       else LAUNCH_DEC(DD, GG, UU, MM, false); } while (0)
   if (D == 128) {
-    if (GQ == 1) LAUNCH_DEC(128, 1);
-    else if (GQ == 2) LAUNCH_DEC(128, 2);
+    if (GQ == 1) LAUNCH_DEC_F(128, 1, 4, 2);
+    else if (GQ == 2) LAUNCH_DEC_F(128, 2, 4, 2);
     // GQ=4 (llama-3 shapes): occupancy-first variant measured 3-6% faster
     // at B=32-64, S=1-4k (scripts/decode_ab.hip, profiles/decode_ab.txt)
-    else if (GQ == 4) LAUNCH_DEC4(128, 4, 2, 4);
-    else LAUNCH_DEC(128, 8);
+    else if (GQ == 4) LAUNCH_DEC_F(128, 4, 2, 4);
+    else LAUNCH_DEC_F(128, 8, 4, 2);
   } else {
-    if (GQ == 1) LAUNCH_DEC(64, 1);
-    else if (GQ == 2) LAUNCH_DEC(64, 2);
-    else if (GQ == 4) LAUNCH_DEC(64, 4);
-    else LAUNCH_DEC(64, 8);
+    if (GQ == 1) LAUNCH_DEC_F(64, 1, 4, 2);
+    else if (GQ == 2) LAUNCH_DEC_F(64, 2, 4, 2);
+    else if (GQ == 4) LAUNCH_DEC_F(64, 4, 4, 2);
+    else LAUNCH_DEC_F(64, 8, 4, 2);
   }
+#undef LAUNCH_DEC_F
 #undef LAUNCH_DEC
-#undef LAUNCH_DEC4
   if (splits > 1) {
     dim3 cgrid(B, H);
     if (D == 128) {
@@ -399,7 +520,9 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
 
 void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
-                    torch::Tensor slot_mapping) {
+                    torch::Tensor slot_mapping,
+                    c10::optional<torch::Tensor> k_scale,
+                    c10::optional<torch::Tensor> v_scale) {
   TORCH_CHECK(knew.dim() == 3, "knew must be [T, Hkv, D]");
   // token-strided views of a merged QKV projection are accepted
   TORCH_CHECK(knew.stride(2) == 1 && knew.stride(1) == knew.size(2));
@@ -414,6 +537,23 @@ void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
   hipStream_t stream_ = cmls::current_stream();
   const auto st = knew.scalar_type();
   const int block = std::min(256, Hkv * D);
+  const bool fp8kv = k_cache.scalar_type() == at::kByte ||
+                     k_cache.scalar_type() == at::kFloat8_e4m3fn;
+  if (fp8kv) {
+    TORCH_CHECK(k_scale.has_value() && v_scale.has_value(),
+                "fp8 KV cache needs k_scale/v_scale");
+    TORCH_CHECK(st == at::kBFloat16, "fp8 KV write: bf16 source only");
+    hipLaunchKernelGGL(kv_cache_write_fp8_kernel<__hip_bfloat16>, dim3(T),
+                       dim3(std::min(256, 64 * Hkv)), 0, stream_,
+                       (const __hip_bfloat16*)knew.data_ptr(),
+                       (const __hip_bfloat16*)vnew.data_ptr(),
+                       (unsigned char*)k_cache.data_ptr(),
+                       (unsigned char*)v_cache.data_ptr(),
+                       k_scale->data_ptr<float>(),
+                       v_scale->data_ptr<float>(),
+                       slots.data_ptr<int>(), Hkv, D, BS, kts, vts);
+    return;
+  }
   if (st == at::kBFloat16) {
     hipLaunchKernelGGL(kv_cache_write_kernel<__hip_bfloat16>, dim3(T),
                        dim3(block), 0, stream_,

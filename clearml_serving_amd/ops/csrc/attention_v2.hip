@@ -50,12 +50,23 @@ __device__ __forceinline__ unsigned int pack_bf16x2(float lo, float hi) {
 // re-read by every m-tile workgroup of the same (b,h), so after the first
 // touch they come from L2. Halves the LDS footprint and the ds traffic.
 // Dense path only (the paged path needs per-row block-table lookups).
+typedef __attribute__((ext_vector_type(2))) float av2_f32x2_t;
+
+__device__ __forceinline__ void v2_fp8x4_to_f32(int w, float* o) {
+  // word-select of cvt_pk_f32_fp8 must be a literal constant
+  av2_f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  av2_f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  o[0] = lo[0]; o[1] = lo[1]; o[2] = hi[0]; o[3] = hi[1];
+}
+
 template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false,
-          bool KDIRECT = false, bool PIPE = false>
+          bool KDIRECT = false, bool PIPE = false, bool FP8KV = false>
 __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     const __hip_bfloat16* __restrict__ q,
-    const __hip_bfloat16* __restrict__ k,
-    const __hip_bfloat16* __restrict__ v,
+    const void* __restrict__ kvoid,   // bf16, or e4m3 bytes when FP8KV
+    const void* __restrict__ vvoid,
+    const float* __restrict__ k_scale,  // [NB, Hkv, BS] (FP8KV paged)
+    const float* __restrict__ v_scale,
     __hip_bfloat16* __restrict__ out,
     const int* __restrict__ seq_lens,
     const int* __restrict__ q_lens,
@@ -64,6 +75,8 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
     AttnStrides2 st,
     int B, int H, int Hkv, int Sq, int Sk, float scale) {
   const float scale2 = scale * 1.4426950408889634f;
+  const __hip_bfloat16* k = (const __hip_bfloat16*)kvoid;
+  const __hip_bfloat16* v = (const __hip_bfloat16*)vvoid;
   constexpr int D = HEAD_DIM;
   constexpr int KSTRIDE = D + PAD;
   constexpr int VSTRIDE = BLOCK_N + PAD;
@@ -132,15 +145,40 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
       const int d8 = (p % (D / 8)) * 8;
       if (gkey < kv_len) {
         long off;
+        long blk = 0;
         if (PAGED) {
-          const long blk = btab[gkey / block_size];
+          blk = btab[gkey / block_size];
           off = (blk * Hkv) * (long)block_size * D + kv_base +
                 (long)(gkey % block_size) * D + d8;
         } else {
           off = kv_base + (long)gkey * st.ks + d8;
         }
-        if (!KDIRECT) kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);
-        vreg[i] = *reinterpret_cast<const bf16x8_t*>(v + off);
+        if (FP8KV && PAGED) {
+          // e4m3 bytes + per-token scales -> bf16 fragments at stage time
+          const long sidx = (blk * Hkv + hkv) * block_size +
+                            (gkey % block_size);
+          const float ksc = k_scale[sidx], vsc = v_scale[sidx];
+          const int* k8 = reinterpret_cast<const int*>(
+              (const unsigned char*)kvoid + off);
+          const int* v8 = reinterpret_cast<const int*>(
+              (const unsigned char*)vvoid + off);
+          float kf[8], vf[8];
+#pragma unroll
+          for (int w = 0; w < 2; ++w) {
+            v2_fp8x4_to_f32(k8[w], kf + 4 * w);
+            v2_fp8x4_to_f32(v8[w], vf + 4 * w);
+          }
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            __hip_bfloat16 kb = __float2bfloat16(kf[e] * ksc);
+            __hip_bfloat16 vb = __float2bfloat16(vf[e] * vsc);
+            kreg[i][e] = *reinterpret_cast<short*>(&kb);
+            vreg[i][e] = *reinterpret_cast<short*>(&vb);
+          }
+        } else {
+          if (!KDIRECT) kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);
+          vreg[i] = *reinterpret_cast<const bf16x8_t*>(v + off);
+        }
       } else {
         if (!KDIRECT) kreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
         vreg[i] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
@@ -410,8 +448,8 @@ torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
   hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, CC, SS, false, KD, PP>),    \
                      grid, block, 0,                                         \
                      stream_, (const __hip_bfloat16*)q.data_ptr(),           \
-                     (const __hip_bfloat16*)k.data_ptr(),                    \
-                     (const __hip_bfloat16*)v.data_ptr(),                    \
+                     (const void*)k.data_ptr(),                              \
+                     (const void*)v.data_ptr(), nullptr, nullptr,            \
                      (__hip_bfloat16*)out.data_ptr(), sl, nullptr, nullptr,  \
                      0, 0, st, B, H, Hkv, Sq, Sk, (float)scale)
 #define LAUNCH_ATTN2_KD(DD, CC, SS)                                          \
@@ -433,10 +471,23 @@ torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
 torch::Tensor attention_prefill_paged_v2(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
     torch::Tensor block_table, torch::Tensor kv_lens, torch::Tensor q_lens,
-    double scale) {
+    double scale, c10::optional<torch::Tensor> k_scale,
+    c10::optional<torch::Tensor> v_scale) {
   TORCH_CHECK(q.dim() == 4 && q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(q.stride(3) == 1 && q.stride(2) == q.size(3));
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  const bool fp8kv = k_cache.scalar_type() == at::kByte ||
+                     k_cache.scalar_type() == at::kFloat8_e4m3fn;
+  const float* kscp = nullptr;
+  const float* vscp = nullptr;
+  if (fp8kv) {
+    TORCH_CHECK(k_scale.has_value() && v_scale.has_value(),
+                "fp8 kv cache needs k_scale/v_scale");
+    TORCH_CHECK(k_scale->scalar_type() == at::kFloat &&
+                k_scale->is_contiguous() && v_scale->is_contiguous());
+    kscp = k_scale->data_ptr<float>();
+    vscp = v_scale->data_ptr<float>();
+  }
   const int B = q.size(0), Sq = q.size(1), H = q.size(2), D = q.size(3);
   const int Hkv = k_cache.size(1), BS = k_cache.size(2);
   const int max_blocks = block_table.size(1);
@@ -454,17 +505,16 @@ torch::Tensor attention_prefill_paged_v2(
   dim3 grid(B * H, (Sq + BLOCK_M - 1) / BLOCK_M);
   dim3 block(NWAVES * 64);
   hipStream_t stream_ = cmls::current_stream();
-#define LAUNCH_PAGED2(DD)                                                    \
-  hipLaunchKernelGGL((attn_prefill_v2_kernel<DD, true, true, true>), grid,   \
-                     block, 0, stream_,                                      \
-                     (const __hip_bfloat16*)q.data_ptr(),                    \
-                     (const __hip_bfloat16*)k_cache.data_ptr(),              \
-                     (const __hip_bfloat16*)v_cache.data_ptr(),              \
-                     (__hip_bfloat16*)out.data_ptr(), kl.data_ptr<int>(),    \
-                     ql.data_ptr<int>(), bt.data_ptr<int>(), BS, max_blocks, \
-                     st, B, H, Hkv, Sq, Sk, (float)scale)
-  if (D == 64) LAUNCH_PAGED2(64);
-  else LAUNCH_PAGED2(128);
+#define LAUNCH_PAGED2(DD, F8)                                                \
+  hipLaunchKernelGGL(                                                        \
+      (attn_prefill_v2_kernel<DD, true, true, true, false, false, F8>),      \
+      grid, block, 0, stream_, (const __hip_bfloat16*)q.data_ptr(),          \
+      (const void*)k_cache.data_ptr(), (const void*)v_cache.data_ptr(),      \
+      kscp, vscp, (__hip_bfloat16*)out.data_ptr(), kl.data_ptr<int>(),       \
+      ql.data_ptr<int>(), bt.data_ptr<int>(), BS, max_blocks,                \
+      st, B, H, Hkv, Sq, Sk, (float)scale)
+  if (D == 64) { if (fp8kv) LAUNCH_PAGED2(64, true); else LAUNCH_PAGED2(64, false); }
+  else         { if (fp8kv) LAUNCH_PAGED2(128, true); else LAUNCH_PAGED2(128, false); }
 #undef LAUNCH_PAGED2
   return out;
 }

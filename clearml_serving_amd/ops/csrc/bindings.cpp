@@ -20,10 +20,14 @@ torch::Tensor sample_top_k_top_p(torch::Tensor logits, double temperature,
 torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
                                torch::Tensor v_cache,
                                torch::Tensor block_table,
-                               torch::Tensor seq_lens, double scale);
+                               torch::Tensor seq_lens, double scale,
+                               c10::optional<torch::Tensor> k_scale,
+                               c10::optional<torch::Tensor> v_scale);
 void kv_cache_write(torch::Tensor knew, torch::Tensor vnew,
                     torch::Tensor k_cache, torch::Tensor v_cache,
-                    torch::Tensor slot_mapping);
+                    torch::Tensor slot_mapping,
+                    c10::optional<torch::Tensor> k_scale,
+                    c10::optional<torch::Tensor> v_scale);
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b);
 torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor w, int64_t variant);
 torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor k_cache,
@@ -52,7 +56,8 @@ torch::Tensor attention_prefill_v2(torch::Tensor q, torch::Tensor k,
 torch::Tensor attention_prefill_paged_v2(
     torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
     torch::Tensor block_table, torch::Tensor kv_lens, torch::Tensor q_lens,
-    double scale);
+    double scale, c10::optional<torch::Tensor> k_scale,
+    c10::optional<torch::Tensor> v_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "clearml-serving-amd gfx950 kernel library";
@@ -71,8 +76,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("seq_lens") = py::none(), py::arg("bshd") = false);
   m.def("sample_top_k_top_p", &sample_top_k_top_p);
   m.def("rope_inplace", &rope_inplace);
-  m.def("attention_decode", &attention_decode);
-  m.def("kv_cache_write", &kv_cache_write);
+  m.def("attention_decode", &attention_decode, py::arg("q"),
+        py::arg("k_cache"), py::arg("v_cache"), py::arg("block_table"),
+        py::arg("seq_lens"), py::arg("scale"),
+        py::arg("k_scale") = py::none(), py::arg("v_scale") = py::none());
+  m.def("kv_cache_write", &kv_cache_write, py::arg("knew"), py::arg("vnew"),
+        py::arg("k_cache"), py::arg("v_cache"), py::arg("slot_mapping"),
+        py::arg("k_scale") = py::none(), py::arg("v_scale") = py::none());
   m.def("gemm_bf16", &gemm_bf16);
   m.def("skinny_gemm", &skinny_gemm, py::arg("a"), py::arg("w"),
         py::arg("variant") = 0);
@@ -90,5 +100,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
         py::arg("seq_lens") = py::none(), py::arg("bshd") = false,
         py::arg("kdirect") = false, py::arg("pipe") = false);
-  m.def("attention_prefill_paged_v2", &attention_prefill_paged_v2);
+  m.def("attention_prefill_paged_v2", &attention_prefill_paged_v2,
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("block_table"), py::arg("kv_lens"), py::arg("q_lens"),
+        py::arg("scale"), py::arg("k_scale") = py::none(),
+        py::arg("v_scale") = py::none());
 }

@@ -762,3 +762,73 @@ def test_chunked_prefill_fair_share():
     assert both, progress
     for chunks in both:
         assert chunks[0] == chunks[1] == 16, (chunks, progress)
+
+
+def test_fp8_kv_cache_engine_generates():
+    """kv_dtype="fp8": uint8 e4m3 caches + per-token scales; generation is
+    deterministic and the paged decode path runs on the quantized cache."""
+    eng = tiny_engine(kv_dtype="fp8")
+    assert len(eng.kv_caches[0]) == 4
+    assert eng.kv_caches[0][0].dtype == torch.uint8
+    assert eng.kv_caches[0][2].dtype == torch.float32
+
+    async def gen():
+        out = []
+        async for item in eng.generate(
+                "hello fp8", SamplingParams(temperature=0.0, max_tokens=8,
+                                            ignore_eos=True)):
+            out.extend(item["token_ids"])
+        return out
+
+    a = run(gen())
+    b = run(gen())
+    assert len(a) == 8 and a == b
+    # scales were actually written (not all still at init value 1.0)
+    assert (eng.kv_caches[0][2] != 1.0).any()
+
+
+def test_fp8_kv_close_to_bf16_kv_generation():
+    """Greedy tokens from the fp8-KV engine track the full-precision-KV
+    engine on a short continuation (e4m3 + per-token-per-head scales keep
+    logits within quantization noise)."""
+    eng16 = tiny_engine()
+    eng8 = tiny_engine(kv_dtype="fp8")
+    prompt = [3, 7, 11, 19, 23, 29, 31]
+
+    async def gen(eng):
+        seq = await eng.add_request(
+            list(prompt), SamplingParams(temperature=0.0, max_tokens=4,
+                                         ignore_eos=True))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    t16 = run(gen(eng16))
+    t8 = run(gen(eng8))
+    assert len(t8) == 4
+    # the first decoded token attends to a freshly-quantized prompt cache:
+    # argmax agreement here is the direct numerics check; later tokens may
+    # legitimately diverge once sequences differ
+    assert t8[0] == t16[0]
+
+
+def test_fp8_kv_chunked_prefill_generates():
+    """Chunked prefill reads the fp8 cache through the paged-prefill path."""
+    eng = tiny_engine(kv_dtype="fp8", prefill_chunk=16)
+    prompt = list(range(2, 50))
+
+    async def gen():
+        seq = await eng.add_request(
+            list(prompt), SamplingParams(temperature=0.0, max_tokens=4,
+                                         ignore_eos=True))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    assert len(run(gen())) == 4

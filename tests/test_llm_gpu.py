@@ -286,3 +286,135 @@ def test_decode_graphs_match_eager():
     eager = gen(False)
     graphed = gen(True)
     assert eager == graphed
+
+
+# --------------------------------------------------------------------- #
+# fp8 KV cache (attention_decode.hip FP8KV / kv_cache_write_fp8_kernel /
+# attention_v2.hip FP8KV paged prefill)
+# --------------------------------------------------------------------- #
+def test_kv_cache_write_fp8_matches_cpu_quantizer():
+    torch.manual_seed(4)
+    t, hkv, d, bs, nb = 9, 4, 128, 16, 4
+    k_new = torch.randn(t, hkv, d, device=DEV, dtype=torch.bfloat16) * 2
+    v_new = torch.randn(t, hkv, d, device=DEV, dtype=torch.bfloat16)
+    k8 = torch.zeros(nb, hkv, bs, d, device=DEV, dtype=torch.uint8)
+    v8 = torch.zeros(nb, hkv, bs, d, device=DEV, dtype=torch.uint8)
+    ks = torch.ones(nb, hkv, bs, device=DEV)
+    vs = torch.ones(nb, hkv, bs, device=DEV)
+    slots = torch.tensor([0, 1, 15, 16, 33, 63, -1, 40, 5],
+                         dtype=torch.int32, device=DEV)
+    ops.kv_cache_write(k_new, v_new, k8, v8, slots, ks, vs)
+    # CPU reference quantizer on the same inputs
+    k8c = torch.zeros(nb, hkv, bs, d, dtype=torch.uint8)
+    v8c = torch.zeros(nb, hkv, bs, d, dtype=torch.uint8)
+    ksc = torch.ones(nb, hkv, bs)
+    vsc = torch.ones(nb, hkv, bs)
+    ops.kv_cache_write(k_new.cpu(), v_new.cpu(), k8c, v8c, slots.cpu(),
+                       ksc, vsc)
+    gdq = ops._dequant_kv_cpu(k8.cpu(), ks.cpu())
+    cdq = ops._dequant_kv_cpu(k8c, ksc)
+    torch.testing.assert_close(gdq, cdq, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(ops._dequant_kv_cpu(v8.cpu(), vs.cpu()),
+                               ops._dequant_kv_cpu(v8c, vsc),
+                               atol=2e-2, rtol=2e-2)
+    assert (k8[3, :, 14] == 0).all()  # slot -1 skipped
+
+
+@pytest.mark.parametrize("b,h,hkv,d,seqs", [
+    (4, 32, 8, 128, [1, 17, 200, 1000]),
+    (2, 8, 8, 64, [33, 64]),
+])
+def test_attention_decode_fp8_kv_numerics(b, h, hkv, d, seqs):
+    torch.manual_seed(5)
+    block_size = 16
+    max_blocks = (max(seqs) + block_size - 1) // block_size
+    nblocks = b * max_blocks + 1
+    kf = torch.randn(nblocks, hkv, block_size, d, device=DEV,
+                     dtype=torch.bfloat16)
+    vf = torch.randn(nblocks, hkv, block_size, d, device=DEV,
+                     dtype=torch.bfloat16)
+    k8 = torch.zeros(nblocks, hkv, block_size, d, device=DEV,
+                     dtype=torch.uint8)
+    v8 = torch.zeros_like(k8)
+    ks = torch.ones(nblocks, hkv, block_size, device=DEV)
+    vs = torch.ones_like(ks)
+    slots = torch.arange(nblocks * block_size, dtype=torch.int32, device=DEV)
+    ops.kv_cache_write(
+        kf.permute(0, 2, 1, 3).reshape(-1, hkv, d).contiguous(),
+        vf.permute(0, 2, 1, 3).reshape(-1, hkv, d).contiguous(),
+        k8, v8, slots, ks, vs)
+    perm = torch.randperm(nblocks - 1) + 1
+    block_table = perm[:b * max_blocks].reshape(b, max_blocks).to(
+        torch.int32).to(DEV)
+    seq_lens = torch.tensor(seqs, dtype=torch.int32, device=DEV)
+    q = torch.randn(b, h, d, device=DEV, dtype=torch.bfloat16)
+    got = ops.attention_decode(q, k8, v8, block_table, seq_lens,
+                               k_scale=ks, v_scale=vs)
+    # fp32 CPU reference on the DEQUANTIZED cache isolates the kernel's
+    # fp8 read path from the quantization error itself
+    ref = ops.attention_decode(
+        q.float().cpu(), ops._dequant_kv_cpu(k8.cpu(), ks.cpu()),
+        ops._dequant_kv_cpu(v8.cpu(), vs.cpu()),
+        block_table.cpu(), seq_lens.cpu())
+    torch.testing.assert_close(got.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_prefill_paged_fp8_kv_numerics():
+    torch.manual_seed(6)
+    bsz, sq, h, hkv, d, bs = 2, 32, 8, 2, 128, 16
+    kv_lens = torch.tensor([200, 77], dtype=torch.int32)
+    q_lens = torch.tensor([32, 20], dtype=torch.int32)
+    max_blocks = (200 + bs - 1) // bs
+    nblocks = bsz * max_blocks + 1
+    kf = torch.randn(nblocks, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    vf = torch.randn(nblocks, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    k8 = torch.zeros(nblocks, hkv, bs, d, device=DEV, dtype=torch.uint8)
+    v8 = torch.zeros_like(k8)
+    ks = torch.ones(nblocks, hkv, bs, device=DEV)
+    vs = torch.ones_like(ks)
+    slots = torch.arange(nblocks * bs, dtype=torch.int32, device=DEV)
+    ops.kv_cache_write(
+        kf.permute(0, 2, 1, 3).reshape(-1, hkv, d).contiguous(),
+        vf.permute(0, 2, 1, 3).reshape(-1, hkv, d).contiguous(),
+        k8, v8, slots, ks, vs)
+    perm = torch.randperm(nblocks - 1) + 1
+    block_table = perm[:bsz * max_blocks].reshape(bsz, max_blocks).to(
+        torch.int32).to(DEV)
+    q = torch.randn(bsz, sq, h, d, device=DEV, dtype=torch.bfloat16)
+    got = ops.attention_prefill_paged(
+        q, k8, v8, block_table, kv_lens.to(DEV), q_lens.to(DEV),
+        k_scale=ks, v_scale=vs)
+    ref = ops.attention_prefill_paged(
+        q.float().cpu(), ops._dequant_kv_cpu(k8.cpu(), ks.cpu()),
+        ops._dequant_kv_cpu(v8.cpu(), vs.cpu()), block_table.cpu(),
+        kv_lens, q_lens)
+    for i in range(bsz):
+        n = int(q_lens[i])
+        torch.testing.assert_close(got[i, :n].float().cpu(), ref[i, :n],
+                                   atol=3e-2, rtol=3e-2)
+
+
+def test_llm_engine_gpu_fp8_kv_generation():
+    """End-to-end: fp8 weights + fp8 KV cache decode on the GPU."""
+    cfg = LlmEngineConfig(preset="llama-3-1b", num_kv_blocks=2048,
+                          block_size=16, max_model_len=1024, device=DEV,
+                          kv_dtype="fp8")
+    eng = LlmEngine(cfg)
+    eng.start()
+    assert eng.kv_caches[0][0].dtype == torch.uint8
+
+    async def main():
+        params = SamplingParams(temperature=0.0, max_tokens=16,
+                                ignore_eos=True)
+
+        async def one(text):
+            toks = []
+            async for item in eng.generate(text, params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return await asyncio.gather(*[one("prompt %d" % i) for i in range(4)])
+
+    outs = run(main())
+    eng.stop()
+    assert all(len(o) == 16 for o in outs)

@@ -261,3 +261,82 @@ def test_fp8_weight_swizzle_roundtrip():
     sw = ops.skinny_gemm_fp8(a8, as_, ops.swizzle_fp8_weight(wq), ws,
                              swizzled=True)
     torch.testing.assert_close(plain, sw)
+
+
+def test_kv_cache_write_fp8_roundtrip_cpu():
+    # absmax e4m3 quantization: dequantized rows within e4m3 precision
+    torch.manual_seed(1)
+    hkv, d, bs, nblocks, t = 2, 64, 16, 4, 10
+    k_cache = torch.zeros(nblocks, hkv, bs, d, dtype=torch.uint8)
+    v_cache = torch.zeros(nblocks, hkv, bs, d, dtype=torch.uint8)
+    k_scale = torch.ones(nblocks, hkv, bs)
+    v_scale = torch.ones(nblocks, hkv, bs)
+    k_new = torch.randn(t, hkv, d) * 3.0
+    v_new = torch.randn(t, hkv, d)
+    slots = torch.tensor([5, 17, 33, 48, 1, 60, 22, 9, 40, 55],
+                         dtype=torch.int32)
+    ops.kv_cache_write(k_new, v_new, k_cache, v_cache, slots,
+                       k_scale, v_scale)
+    kdq = ops._dequant_kv_cpu(k_cache, k_scale)
+    vdq = ops._dequant_kv_cpu(v_cache, v_scale)
+    for i, s in enumerate(slots.tolist()):
+        blk, off = s // bs, s % bs
+        # e4m3: 3 mantissa bits -> per-element relative error <= ~6.25%
+        # of the row absmax after per-row scaling
+        for dq, new in ((kdq, k_new), (vdq, v_new)):
+            row = new[i].float()
+            tol = row.abs().amax(dim=-1, keepdim=True) / 16.0 + 1e-6
+            assert ((dq[blk, :, off] - row).abs() <= tol).all()
+
+
+def test_attention_decode_fp8_kv_close_to_bf16_cpu():
+    torch.manual_seed(2)
+    b, h, hkv, d, bs, nblocks = 2, 8, 2, 64, 16, 8
+    kf = torch.randn(nblocks, hkv, bs, d)
+    vf = torch.randn(nblocks, hkv, bs, d)
+    k8 = torch.zeros_like(kf, dtype=torch.uint8)
+    v8 = torch.zeros_like(vf, dtype=torch.uint8)
+    ks = torch.ones(nblocks, hkv, bs)
+    vs = torch.ones(nblocks, hkv, bs)
+    # quantize whole cache through the op (flat slots cover every entry)
+    slots = torch.arange(nblocks * bs, dtype=torch.int32)
+    knew = kf.permute(0, 2, 1, 3).reshape(nblocks * bs, hkv, d)
+    vnew = vf.permute(0, 2, 1, 3).reshape(nblocks * bs, hkv, d)
+    ops.kv_cache_write(knew, vnew, k8, v8, slots, ks, vs)
+    block_table = torch.tensor([[0, 2, 4, 6], [1, 3, 5, 7]],
+                               dtype=torch.int32)
+    seq_lens = torch.tensor([50, 33], dtype=torch.int32)
+    q = torch.randn(b, h, d)
+    ref = ops.attention_decode(q, kf, vf, block_table, seq_lens)
+    out = ops.attention_decode(q, k8, v8, block_table, seq_lens,
+                               k_scale=ks, v_scale=vs)
+    # softmax-weighted averages tolerate e4m3 KV noise well
+    torch.testing.assert_close(out, ref, atol=0.08, rtol=0.05)
+
+
+def test_attention_prefill_paged_fp8_kv_close_to_bf16_cpu():
+    torch.manual_seed(3)
+    bsz, sq, h, hkv, d, bs, nblocks = 2, 8, 4, 2, 64, 16, 8
+    kf = torch.randn(nblocks, hkv, bs, d)
+    vf = torch.randn(nblocks, hkv, bs, d)
+    k8 = torch.zeros_like(kf, dtype=torch.uint8)
+    v8 = torch.zeros_like(vf, dtype=torch.uint8)
+    ks = torch.ones(nblocks, hkv, bs)
+    vs = torch.ones(nblocks, hkv, bs)
+    slots = torch.arange(nblocks * bs, dtype=torch.int32)
+    ops.kv_cache_write(kf.permute(0, 2, 1, 3).reshape(-1, hkv, d),
+                       vf.permute(0, 2, 1, 3).reshape(-1, hkv, d),
+                       k8, v8, slots, ks, vs)
+    block_table = torch.tensor([[0, 2, 4, 6], [1, 3, 5, 7]],
+                               dtype=torch.int32)
+    kv_lens = torch.tensor([50, 33], dtype=torch.int32)
+    q_lens = torch.tensor([8, 5], dtype=torch.int32)
+    q = torch.randn(bsz, sq, h, d)
+    ref = ops.attention_prefill_paged(q, kf, vf, block_table, kv_lens,
+                                      q_lens)
+    out = ops.attention_prefill_paged(q, k8, v8, block_table, kv_lens,
+                                      q_lens, k_scale=ks, v_scale=vs)
+    for i in range(bsz):
+        n = int(q_lens[i])
+        torch.testing.assert_close(out[i, :n], ref[i, :n], atol=0.08,
+                                   rtol=0.05)
