@@ -21,14 +21,6 @@
 
 namespace {
 
-__device__ __forceinline__ void fp8x4_to_f32(int w, float* o) {
-  // word-select of cvt_pk_f32_fp8 must be a literal constant
-  typedef __attribute__((ext_vector_type(2))) float adf32x2_t;
-  adf32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
-  adf32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
-  o[0] = lo[0]; o[1] = lo[1]; o[2] = hi[0]; o[3] = hi[1];
-}
-
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(2))) short bf16x2_t;
 
@@ -50,6 +42,25 @@ __device__ __forceinline__ float group_sum4(float v) {
 // occupancy-first alternative (116 VGPR -> 4 waves/SIMD at half the loads
 // in flight per wave); scripts/decode_ab.hip A/Bs them.
 typedef __attribute__((ext_vector_type(2))) float f32x2_t;
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16v2_t;
+
+// 4 e4m3 bytes -> 2 packed-bf16 pairs (v_cvt_scalef32_pk_bf16_fp8: the
+// word-select must be a literal constant, scale multiplies the result)
+__device__ __forceinline__ void fp8x4_to_bf16x2x2(int w, float sc,
+                                                  bf16x2_t* o) {
+  bf16v2_t lo = __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(w, sc, false);
+  bf16v2_t hi = __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(w, sc, true);
+  o[0] = *reinterpret_cast<bf16x2_t*>(&lo);
+  o[1] = *reinterpret_cast<bf16x2_t*>(&hi);
+}
+
+// 4 e4m3 bytes -> 4 f32, scaled (one v_cvt_scalef32_pk_f32_fp8 per pair)
+__device__ __forceinline__ void fp8x4_to_f32_scaled(int w, float sc,
+                                                    float* o) {
+  f32x2_t lo = __builtin_amdgcn_cvt_scalef32_pk_f32_fp8(w, sc, false);
+  f32x2_t hi = __builtin_amdgcn_cvt_scalef32_pk_f32_fp8(w, sc, true);
+  o[0] = lo[0]; o[1] = lo[1]; o[2] = hi[0]; o[3] = hi[1];
+}
 typedef __attribute__((ext_vector_type(2))) unsigned int uint32x2_t;
 
 // FP8KV: caches hold OCP-e4m3 bytes with per-token-per-head scales
@@ -103,7 +114,6 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
   // q fragments: packed bf16 pairs feed v_dot2_f32_bf16 (2 MACs per
   // instruction in the K dot products; scale folds into the score later)
   bf16x2_t qp[GQ][EPL / 2];
-  float qf[FP8KV ? GQ : 1][FP8KV ? EPL : 1];  // plain-FMA path for fp8 K
 #pragma unroll
   for (int g = 0; g < GQ; ++g) {
     const int h = hkv * GQ + g;
@@ -112,10 +122,6 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
 #pragma unroll
     for (int e = 0; e < EPL / 2; ++e)
       qp[g][e] = *reinterpret_cast<const bf16x2_t*>(qrow + 2 * e);
-    if (FP8KV) {
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) qf[g][e] = to_f32(qrow[e]);
-    }
   }
 
   float m_run[GQ], l_run[GQ], o_acc[GQ][EPL];
@@ -174,16 +180,20 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
       if (FP8KV) {
-        // dequantized plain-FMA dot (score scale absorbs k_scale later)
-        float kf[EPL];
+        // packed dequant (scale=1; k_scale folds into the score scale) ->
+        // the same v_dot2_f32_bf16 dot as the bf16 cache path
+        bf16x2_t kb[EPL / 2];
         const int* ki = reinterpret_cast<const int*>(&k8v[u]);
 #pragma unroll
-        for (int w = 0; w < EPL / 4; ++w) fp8x4_to_f32(ki[w], kf + 4 * w);
+        for (int w = 0; w < EPL / 4; ++w)
+          fp8x4_to_bf16x2x2(ki[w], 1.0f, kb + 2 * w);
 #pragma unroll
         for (int g = 0; g < GQ; ++g) {
           float acc = 0.f;
 #pragma unroll
-          for (int e = 0; e < EPL; ++e) acc = fmaf(qf[g][e], kf[e], acc);
+          for (int e = 0; e < EPL / 2; ++e)
+            acc = __builtin_amdgcn_fdot2_f32_bf16(qp[g][e], kb[e], acc,
+                                                  false);
           score[u][g] = acc;
         }
       } else {
@@ -209,6 +219,18 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
     }
     // ONE online-softmax update per chunk per head (rescaling O per
     // 4-key iteration was the serial VALU chain limiting bandwidth)
+    float vfq[FP8KV ? UNROLL : 1][FP8KV ? EPL : 1];
+    if (FP8KV) {
+      // dequant V ONCE per chunk (not per head): v_scale folds into the
+      // packed conversion, so the per-head loop below is plain FMA
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u) {
+        const int* vi = reinterpret_cast<const int*>(&v8v[u]);
+#pragma unroll
+        for (int w = 0; w < EPL / 4; ++w)
+          fp8x4_to_f32_scaled(vi[w], vsc[u], vfq[u] + 4 * w);
+      }
+    }
 #pragma unroll
     for (int g = 0; g < GQ; ++g) {
       float cmax = score[0][g];
@@ -228,13 +250,8 @@ __global__ __launch_bounds__(256, MINB) void attn_decode_kernel(
         const float p = valid[u] ? __builtin_amdgcn_exp2f(score[u][g] - m_new) : 0.f;
         psum += p;
         if (FP8KV) {
-          const float pv = p * vsc[u];
-          const int* vi = reinterpret_cast<const int*>(&v8v[u]);
-          float vf[EPL];
 #pragma unroll
-          for (int w = 0; w < EPL / 4; ++w) fp8x4_to_f32(vi[w], vf + 4 * w);
-#pragma unroll
-          for (int e = 0; e < EPL; ++e) o_acc[g][e] += pv * vf[e];
+          for (int e = 0; e < EPL; ++e) o_acc[g][e] += p * vfq[u][e];
         } else {
 #pragma unroll
           for (int e = 0; e < EPL; ++e)

@@ -52,11 +52,15 @@ __device__ __forceinline__ unsigned int pack_bf16x2(float lo, float hi) {
 // Dense path only (the paged path needs per-row block-table lookups).
 typedef __attribute__((ext_vector_type(2))) float av2_f32x2_t;
 
-__device__ __forceinline__ void v2_fp8x4_to_f32(int w, float* o) {
-  // word-select of cvt_pk_f32_fp8 must be a literal constant
-  av2_f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
-  av2_f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
-  o[0] = lo[0]; o[1] = lo[1]; o[2] = hi[0]; o[3] = hi[1];
+typedef __attribute__((ext_vector_type(2))) __bf16 av2_bf16x2_t;
+
+// 4 e4m3 bytes -> 2 packed-bf16 pairs with the scale folded into the
+// conversion (CDNA4 v_cvt_scalef32_pk_bf16_fp8; word-select is literal)
+__device__ __forceinline__ void v2_fp8x4_to_bf16(int w, float sc, short* o) {
+  av2_bf16x2_t lo = __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(w, sc, false);
+  av2_bf16x2_t hi = __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(w, sc, true);
+  *reinterpret_cast<av2_bf16x2_t*>(o) = lo;
+  *reinterpret_cast<av2_bf16x2_t*>(o + 2) = hi;
 }
 
 template <int HEAD_DIM, bool CAUSAL, bool HAS_SEQLENS, bool PAGED = false,
@@ -162,18 +166,16 @@ __global__ __launch_bounds__(NWAVES * 64, 2) void attn_prefill_v2_kernel(
               (const unsigned char*)kvoid + off);
           const int* v8 = reinterpret_cast<const int*>(
               (const unsigned char*)vvoid + off);
-          float kf[8], vf[8];
+          short kb[8], vb[8];
 #pragma unroll
           for (int w = 0; w < 2; ++w) {
-            v2_fp8x4_to_f32(k8[w], kf + 4 * w);
-            v2_fp8x4_to_f32(v8[w], vf + 4 * w);
+            v2_fp8x4_to_bf16(k8[w], ksc, kb + 4 * w);
+            v2_fp8x4_to_bf16(v8[w], vsc, vb + 4 * w);
           }
 #pragma unroll
           for (int e = 0; e < 8; ++e) {
-            __hip_bfloat16 kb = __float2bfloat16(kf[e] * ksc);
-            __hip_bfloat16 vb = __float2bfloat16(vf[e] * vsc);
-            kreg[i][e] = *reinterpret_cast<short*>(&kb);
-            vreg[i][e] = *reinterpret_cast<short*>(&vb);
+            kreg[i][e] = kb[e];
+            vreg[i][e] = vb[e];
           }
         } else {
           if (!KDIRECT) kreg[i] = *reinterpret_cast<const bf16x8_t*>(k + off);

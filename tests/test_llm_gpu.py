@@ -311,12 +311,21 @@ def test_kv_cache_write_fp8_matches_cpu_quantizer():
     vsc = torch.ones(nb, hkv, bs)
     ops.kv_cache_write(k_new.cpu(), v_new.cpu(), k8c, v8c, slots.cpu(),
                        ksc, vsc)
-    gdq = ops._dequant_kv_cpu(k8.cpu(), ks.cpu())
-    cdq = ops._dequant_kv_cpu(k8c, ksc)
-    torch.testing.assert_close(gdq, cdq, atol=2e-2, rtol=2e-2)
-    torch.testing.assert_close(ops._dequant_kv_cpu(v8.cpu(), vs.cpu()),
-                               ops._dequant_kv_cpu(v8c, vsc),
-                               atol=2e-2, rtol=2e-2)
+    # GPU and CPU quantizers may differ on round-to-nearest ties (different
+    # fp32 intermediate orderings), so compare each against the SOURCE rows
+    # within the e4m3 bound, and require near-total byte agreement
+    for g8, gsc, c8, csc, new in ((k8, ks, k8c, ksc, k_new),
+                                  (v8, vs, v8c, vsc, v_new)):
+        gdq = ops._dequant_kv_cpu(g8.cpu(), gsc.cpu())
+        for i, s in enumerate(slots.tolist()):
+            if s < 0:
+                continue
+            blk, off = s // bs, s % bs
+            row = new[i].float().cpu()
+            tol = row.abs().amax(dim=-1, keepdim=True) / 16.0 + 1e-6
+            assert ((gdq[blk, :, off] - row).abs() <= tol).all()
+        match = (g8.cpu() == c8).float().mean().item()
+        assert match > 0.999, "byte agreement {:.4f}".format(match)
     assert (k8[3, :, 14] == 0).all()  # slot -1 skipped
 
 
