@@ -502,7 +502,9 @@ def kv_cache_write(
                                        (v_cache, v_scale, v_new)):
                 row = new[t].float()  # [H_kv, D]
                 amax = row.abs().amax(dim=-1).clamp_min(1e-8)
-                sc = amax / 448.0
+                # pow2 scales match the GPU quantizer (scalef32 dequant
+                # applies only the scale exponent -- MX semantics)
+                sc = torch.exp2(torch.ceil(torch.log2(amax / 448.0)))
                 q8 = (row / sc[:, None]).to(torch.float8_e4m3fn)
                 cache[blk, :, off] = q8.view(torch.uint8)
                 scales[blk, :, off] = sc

@@ -418,8 +418,12 @@ __global__ void kv_cache_write_fp8_kernel(
     }
     kamax = wave_reduce_max(kamax);
     vamax = wave_reduce_max(vamax);
-    const float kscale = fmaxf(kamax, 1e-8f) / 448.0f;
-    const float vscale = fmaxf(vamax, 1e-8f) / 448.0f;
+    // POWER-OF-2 scales: v_cvt_scalef32_pk_* applies only the scale's
+    // exponent (MX semantics), so pow2 makes the hardware dequant exact;
+    // e4m3 is itself a float format, so the rounded-up scale costs no
+    // relative precision (values keep >= half the exponent range)
+    const float kscale = exp2f(ceilf(log2f(fmaxf(kamax, 1e-8f) / 448.0f)));
+    const float vscale = exp2f(ceilf(log2f(fmaxf(vamax, 1e-8f) / 448.0f)));
     const long sidx = (blk * hkv + h) * block_size + off;
     if (lane == 0) {
       ks[sidx] = kscale;
