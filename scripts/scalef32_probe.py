@@ -24,7 +24,8 @@ torch::Tensor run(torch::Tensor bytes, double sc) {
   return out;
 }
 """
-mod = load_inline(name="scalef32_probe", cpp_sources="", cuda_sources=SRC,
+CPP = "torch::Tensor run(torch::Tensor bytes, double sc);"
+mod = load_inline(name="scalef32_probe", cpp_sources=CPP, cuda_sources=SRC,
                   functions=["run"], with_cuda=True, verbose=False)
 b = torch.tensor([0x3C, 0x44], dtype=torch.uint8, device="cuda")  # e4m3 vals
 for sc in (1.0, 2.0, 3.0, 0.7, 448.0):
