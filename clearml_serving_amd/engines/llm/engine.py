@@ -21,6 +21,7 @@ import asyncio
 import json
 import os
 import random
+import threading
 import time
 import uuid
 from collections import deque
@@ -259,6 +260,12 @@ class LlmEngine:
         # aborts cross threads (event loop -> step worker): appended on the
         # loop, drained at the top of step() -- deque ops are GIL-atomic
         self._aborted: "deque[Sequence]" = deque()
+        # serializes GPU execution between the step worker thread and the
+        # non-generative paths (embed/pooling): both launch on the default
+        # stream, and a forward issued from a second thread while a decode
+        # hipGraph is being captured corrupts the capture; under TP it
+        # would also interleave two plan broadcasts (rank-order corruption)
+        self._exec_lock = threading.Lock()
         self._loop_task: Optional[asyncio.Task] = None
         self._wake: Optional[asyncio.Event] = None
         self._started = False
@@ -508,6 +515,10 @@ class LlmEngine:
         """One scheduler iteration: admit, prefill (chunked -- at most
         ``prefill_chunk`` prompt tokens per step so decode latency stays
         bounded under long-prompt load), then decode everything else."""
+        with self._exec_lock:
+            self._step_locked()
+
+    def _step_locked(self) -> None:
         self.stats["steps"] += 1
         self._drain_aborts()
         self._admit()
@@ -771,15 +782,21 @@ class LlmEngine:
             pooled = torch.nn.functional.normalize(pooled, dim=-1)
         return pooled
 
+    def _embed_sync(self, plan: Dict[str, Any]) -> List[List[float]]:
+        """Broadcast + dense forward under the exec lock: never concurrent
+        with step() (graph capture / TP plan ordering) and never on the
+        event-loop thread (a dense forward would stall every endpoint)."""
+        with self._exec_lock:
+            self._tp_broadcast(plan)
+            return self._exec_embed(plan).cpu().tolist()
+
     async def embed_batch(self, texts: List[str]) -> List[List[float]]:
         if self.model is None:
             raise RuntimeError("LLM engine not started")
         prompts = [self.tokenizer.encode(t)[:self.cfg.max_model_len]
                    for t in texts]
         plan = {"mode": "embed", "prompts": prompts}
-        self._tp_broadcast(plan)
-        out = self._exec_embed(plan)
-        return out.cpu().tolist()
+        return await asyncio.to_thread(self._embed_sync, plan)
 
     def _preempt_one(self) -> bool:
         """KV pressure relief: evict the newest running sequence back to the
@@ -817,6 +834,11 @@ class LlmEngine:
                     seqs.remove(s)
                     continue
                 s.blocks.extend(self.allocator.alloc(1))
+        # a preemption victim other than ``s`` stays in ``seqs`` until its
+        # own iteration, where blocks == [] forces it into the growth branch
+        # above and out of the batch -- make that invariant explicit so the
+        # plan below can never index a freed block list
+        seqs[:] = [s for s in seqs if s in self.running]
         if not seqs:
             return
         plan = {
@@ -1058,7 +1080,10 @@ class LlmEngine:
 
     def tp_shutdown(self) -> None:
         if getattr(self, "tp_size", 1) > 1 and self.tp_rank == 0:
-            self._tp_broadcast({"mode": "stop"})
+            # serialize with any in-flight step/embed broadcast so worker
+            # ranks never see "stop" interleaved inside another plan
+            with self._exec_lock:
+                self._tp_broadcast({"mode": "stop"})
 
     def _hit_stop_string(self, s: Sequence) -> bool:
         """OpenAI 'stop' strings: decode a bounded tail window of the output
@@ -1218,8 +1243,7 @@ class LlmEngine:
         prompts = [self.tokenizer.encode(str(t))[:self.cfg.max_model_len]
                    for t in texts]
         plan = {"mode": "embed", "prompts": prompts, "normalize": False}
-        self._tp_broadcast(plan)
-        out = self._exec_embed(plan).cpu().tolist()
+        out = await asyncio.to_thread(self._embed_sync, plan)
         return {"object": "list", "model": model_name,
                 "data": [{"object": "pooling", "index": i, "data": v}
                          for i, v in enumerate(out)]}

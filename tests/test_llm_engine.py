@@ -453,6 +453,44 @@ def test_embeddings_normalized_and_batch_invariant():
     assert max(abs(a - b) for a, b in zip(vb[0], vb[1])) > 1e-3
 
 
+def test_embeddings_concurrent_with_generation():
+    """embed_batch runs off-loop under the engine's exec lock: concurrent
+    generation + embedding requests both complete, embeddings match the
+    solo result (no interleaving with step()'s forward), and the event
+    loop is never blocked by the dense embed forward."""
+    torch.manual_seed(5)
+    eng = tiny_engine()
+    text = "concurrent embedding probe"
+
+    async def mixed():
+        params = SamplingParams(temperature=0.0, max_tokens=24,
+                                ignore_eos=True)
+
+        async def gen(i):
+            toks = []
+            async for item in eng.generate("gen prompt %d" % i, params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        gens = [asyncio.ensure_future(gen(i)) for i in range(3)]
+        # fire embeddings while decode steps are in flight
+        embeds = [asyncio.ensure_future(eng.embed_batch([text]))
+                  for _ in range(4)]
+        toks = await asyncio.gather(*gens)
+        vecs = await asyncio.gather(*embeds)
+        return toks, vecs
+
+    toks, vecs = run(mixed())
+    assert all(len(t) == 24 for t in toks)
+
+    async def solo():
+        return await eng.embed_batch([text])
+
+    want = run(solo())[0]
+    for v in vecs:
+        assert max(abs(a - b) for a, b in zip(v[0], want)) < 1e-4
+
+
 def test_tokenize_detokenize_roundtrip():
     eng = tiny_engine()
     out = eng.openai_tokenize({"prompt": "round trip!"})
