@@ -148,7 +148,7 @@ class EngineOwner:
         try:
             req_id, url, data = unpack_request(raw)
             if url is None:  # abort record: cancel the in-flight request
-                task = self._inflight.get(req_id)
+                task = self._inflight.get((worker, req_id))
                 if task is not None and not task.done():
                     task.cancel()
                     self.stats["aborts"] += 1
@@ -203,9 +203,14 @@ class EngineOwner:
                     is_abort = (ulen == 0 and len(raw) > 10
                                 and raw[10] == 2)  # KIND_ABORT
                     if not is_abort:
-                        self._inflight[rid] = task
+                        # keyed by (worker, id): every front worker counts
+                        # req_ids from 1, so bare ids collide across workers
+                        # and an abort from one front could cancel another
+                        # front's request
+                        self._inflight[(w, rid)] = task
                         task.add_done_callback(
-                            lambda t, r=rid: self._inflight.pop(r, None))
+                            lambda t, k=(w, rid):
+                            self._inflight.pop(k, None))
             if got:
                 await asyncio.sleep(0)
             else:

@@ -458,6 +458,101 @@ def test_owner_cancels_generation_on_stream_abort(tmp_path):
 
 
 @pytest.mark.timeout(120)
+def test_abort_from_one_front_never_cancels_another(tmp_path):
+    """Every front worker counts req_ids from 1, so the owner must track
+    in-flight requests by (worker, id): worker 0 aborting ITS request 1
+    must not cancel worker 1's request 1."""
+    import json as _json
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.engine_owner import EngineOwner
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.serving.shm_transport import ShmClient
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="xw", force_create=True)
+    card = tmp_path / "card.json"
+    card.write_text(_json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 512, "device": "cpu"}))
+    rec = store.register_model(name="t", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(engine_type="llm", serving_url="tl",
+                                    model_id=rec.model_id))
+    proc.serialize()
+
+    prefix = "/cmls_xw_{}".format(os.getpid())
+    owner = EngineOwner(store_root=store_root, session_id=proc.get_id(),
+                        prefix=prefix, owner_idx=0, n_workers=2,
+                        ring_bytes=1 << 20)
+    c0 = ShmClient(prefix, 0, 1, ring_bytes=1 << 20)
+    c1 = ShmClient(prefix, 1, 1, ring_bytes=1 << 20)
+
+    async def main():
+        serve_task = asyncio.get_running_loop().create_task(owner.serve())
+
+        def payload(n):
+            return {"__serve_type__": "v1_chat_completions",
+                    "__body__": {"messages": [{"role": "user",
+                                               "content": "hi"}],
+                                 "stream": True, "max_tokens": n,
+                                 "temperature": 0, "ignore_eos": True}}
+
+        # worker 1: full stream (40 tokens); worker 0: aborts after 2 chunks.
+        # Order matters to catch the collision: worker 0's request registers
+        # FIRST, worker 1's same-id request registers second (with a bare-id
+        # map it would clobber the key), then worker 0's abort arrives --
+        # a bare-id lookup would cancel worker 1's stream.
+        async def full_stream():
+            chunks = 0
+            async for _ in c1.infer_stream("tl", payload(40)):
+                chunks += 1
+            return chunks
+
+        async def aborting_stream():
+            agen = c0.infer_stream("tl", payload(400))
+            n = 0
+            async for _ in agen:
+                n += 1
+                if n == 1:
+                    # worker 0 in flight: now start worker 1's stream
+                    await asyncio.sleep(0.2)
+                if n >= 25:
+                    await agen.aclose()
+                    break
+
+        ab = asyncio.get_running_loop().create_task(aborting_stream())
+
+        async def delayed_full():
+            await asyncio.sleep(0.1)  # after worker 0 registered
+            return await full_stream()
+
+        full = asyncio.get_running_loop().create_task(delayed_full())
+        await asyncio.wait_for(ab, timeout=60)
+        chunks = await asyncio.wait_for(full, timeout=60)
+        # worker 1's stream survived worker 0's abort: 40 content chunks
+        # (+1 final empty-finish chunk depending on timing) and a clean end
+        assert chunks >= 40, chunks
+        assert owner.stats["aborts"] >= 1
+        serve_task.cancel()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
+        c0.close()
+        c1.close()
+        owner.close()
+
+
 def test_owner_engine_swaps_on_hot_reload(tmp_path):
     """A config change (new model version for the same endpoint) propagates
     to the engine owner's sync daemon and the owner rebuilds the engine --
