@@ -21,8 +21,20 @@ from clearml_serving_amd.models.llama import (  # noqa: E402
 from clearml_serving_amd.parallel import tp  # noqa: E402
 
 
-def check_tp_math(rank, world):
+def _tiny_cfg(world):
+    """llama-tiny, widened when the TP degree exceeds its kv_heads=2 (the
+    sharding contract needs heads and kv_heads divisible by world)."""
     cfg = PRESETS["llama-tiny"]
+    if cfg.kv_heads % world or cfg.heads % world:
+        from clearml_serving_amd.models.llama import LlamaConfig
+
+        cfg = LlamaConfig(**{**cfg.__dict__, "heads": 2 * world,
+                             "kv_heads": world})
+    return cfg
+
+
+def check_tp_math(rank, world):
+    cfg = _tiny_cfg(world)
     torch.manual_seed(7)
     full = LlamaForCausalLM(cfg).eval()
     shard_state = tp.shard_llama_weights(full.state_dict(), cfg, rank, world)
@@ -48,8 +60,9 @@ def check_sharded_sampling(rank, world):
     torch.manual_seed(3)
     full_logits = torch.randn(6, 64)
     # same full matrix on every rank (same seed); take this rank's shard
-    shard = full_logits[:, rank * 32:(rank + 1) * 32].contiguous()
-    got = tp.argmax_sharded(shard, rank * 32)
+    w = 64 // world
+    shard = full_logits[:, rank * w:(rank + 1) * w].contiguous()
+    got = tp.argmax_sharded(shard, rank * w)
     assert torch.equal(got, full_logits.argmax(dim=-1)), (got,)
 
     gathered = tp.gather_rows_to_rank0(shard)
@@ -60,11 +73,12 @@ def check_sharded_sampling(rank, world):
 
     # gumbel-max across shards: empirical dist ~ softmax on a 4-token vocab
     logits = torch.tensor([[1.5, 0.5, -0.5, -1.5]])
-    shard2 = logits[:, rank * 2:(rank + 1) * 2].contiguous()
+    w2 = 4 // world
+    shard2 = logits[:, rank * w2:(rank + 1) * w2].contiguous()
     counts = torch.zeros(4)
     n = 3000
     for i in range(n):
-        idx = tp.sample_gumbel_sharded(shard2, rank * 2, 1.0, [i])
+        idx = tp.sample_gumbel_sharded(shard2, rank * w2, 1.0, [i])
         counts[idx.item()] += 1
     probs = torch.softmax(logits[0], dim=-1)
     assert torch.allclose(counts / n, probs, atol=0.04), (counts / n, probs)
@@ -73,8 +87,11 @@ def check_sharded_sampling(rank, world):
 
 
 def check_engine_protocol(rank, world):
+    mcfg = _tiny_cfg(world)
     cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
-                          block_size=16, max_model_len=128, device="cpu")
+                          block_size=16, max_model_len=128, device="cpu",
+                          overrides={"heads": mcfg.heads,
+                                     "kv_heads": mcfg.kv_heads})
     eng = LlmEngine(cfg)
     eng.start()
     assert eng.tp_size == world

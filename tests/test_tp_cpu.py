@@ -26,6 +26,29 @@ def test_tp2_math_and_engine_protocol():
 
 
 @pytest.mark.timeout(300)
+def test_tp4_math_and_engine_protocol():
+    """Same checks at world=4 (widened llama-tiny: TP degree must divide
+    head counts): catches world>2-only bugs -- shard arithmetic, plan-codec
+    world independence, sampling vocab offsets -- before the first real
+    multi-GPU window."""
+    helper = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                          "helpers", "tp_check.py")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29532", helper],
+        capture_output=True, text=True, timeout=280, env=env,
+    )
+    assert out.returncode == 0, out.stdout + "\n" + out.stderr
+    assert "TP-MATH-OK" in out.stdout
+    assert "TP-SAMPLE-OK" in out.stdout
+    assert "TP-ENGINE-OK" in out.stdout
+
+
+@pytest.mark.timeout(300)
 def test_tp_serve_launcher_smoke(tmp_path):
     """The tp_serve LAUNCHER end to end under torchrun (gloo, world=2):
     store + endpoint prepared, rank 0 serves one chat completion through
