@@ -60,8 +60,16 @@ class HfTokenizer:
 
         self._tok = Tokenizer.from_file(path)
         self.vocab_size = self._tok.get_vocab_size()
+        # ChatML checkpoints (Qwen2 family) carry <|im_start|>/<|im_end|>;
+        # chat requests then format with ChatML and stop at <|im_end|>
+        self.is_chatml = (
+            self._tok.token_to_id("<|im_start|>") is not None
+            and self._tok.token_to_id("<|im_end|>") is not None)
         eos = None
-        for cand in ("</s>", "<|eot_id|>", "<|end_of_text|>", "<|endoftext|>"):
+        cands = ("</s>", "<|eot_id|>", "<|end_of_text|>", "<|endoftext|>")
+        if self.is_chatml:
+            cands = ("<|im_end|>",) + cands
+        for cand in cands:
             tid = self._tok.token_to_id(cand)
             if tid is not None:
                 eos = tid
@@ -1167,6 +1175,14 @@ class LlmEngine:
                              "<|eot_id|>".format(m.get("role", "user"),
                                                  m.get("content", "")))
             parts.append("<|start_header_id|>assistant<|end_header_id|>\n\n")
+            return "".join(parts)
+        if getattr(self.tokenizer, "is_chatml", False):
+            # ChatML (Qwen2 family): <|im_start|>role\ncontent<|im_end|>
+            parts = []
+            for m in messages:
+                parts.append("<|im_start|>{}\n{}<|im_end|>\n".format(
+                    m.get("role", "user"), m.get("content", "")))
+            parts.append("<|im_start|>assistant\n")
             return "".join(parts)
         # synthetic fallback (byte tokenizer / random-init serving)
         parts = []

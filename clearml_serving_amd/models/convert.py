@@ -56,7 +56,10 @@ def convert_hf_bert(hf: Dict[str, torch.Tensor],
 
 def convert_hf_llama(hf: Dict[str, torch.Tensor],
                      num_layers: int) -> Dict[str, torch.Tensor]:
-    """transformers LlamaForCausalLM -> models.llama.LlamaForCausalLM."""
+    """transformers LlamaForCausalLM -> models.llama.LlamaForCausalLM.
+
+    Also covers Qwen2ForCausalLM (identical key names; its QKV biases are
+    picked up when present -- build the native model with qkv_bias=True)."""
     out = {
         "embed.weight": hf["model.embed_tokens.weight"],
         "final_norm": hf["model.norm.weight"],
@@ -70,6 +73,11 @@ def convert_hf_llama(hf: Dict[str, torch.Tensor],
             [hf[p + "self_attn.q_proj.weight"],
              hf[p + "self_attn.k_proj.weight"],
              hf[p + "self_attn.v_proj.weight"]], dim=0)
+        if p + "self_attn.q_proj.bias" in hf:  # Qwen2
+            out[o + "qkv.bias"] = torch.cat(
+                [hf[p + "self_attn.q_proj.bias"],
+                 hf[p + "self_attn.k_proj.bias"],
+                 hf[p + "self_attn.v_proj.bias"]], dim=0)
         out[o + "o_proj.weight"] = hf[p + "self_attn.o_proj.weight"]
         out[o + "attn_norm"] = hf[p + "input_layernorm.weight"]
         out[o + "mlp_norm"] = hf[p + "post_attention_layernorm.weight"]
@@ -78,3 +86,7 @@ def convert_hf_llama(hf: Dict[str, torch.Tensor],
             dim=0)
         out[o + "down.weight"] = hf[p + "mlp.down_proj.weight"]
     return out
+
+
+# Qwen2 checkpoints use llama key names + QKV biases (handled above)
+convert_hf_qwen2 = convert_hf_llama

@@ -39,6 +39,7 @@ class LlamaConfig:
     rms_eps: float = 1e-5
     max_position: int = 8192
     tie_embeddings: bool = False
+    qkv_bias: bool = False  # Qwen2-family checkpoints carry QKV biases
 
     @property
     def head_dim(self) -> int:
@@ -61,7 +62,7 @@ class LlamaLayer(nn.Module):
         inter = cfg.intermediate // tp_size
 
         self.attn_norm = nn.Parameter(torch.ones(h))
-        self.qkv = nn.Linear(h, q_out + 2 * kv_out, bias=False)
+        self.qkv = nn.Linear(h, q_out + 2 * kv_out, bias=cfg.qkv_bias)
         self.o_proj = nn.Linear(q_out, h, bias=False)
         self.mlp_norm = nn.Parameter(torch.ones(h))
         self.gate_up = nn.Linear(h, 2 * inter, bias=False)
@@ -73,7 +74,10 @@ class LlamaLayer(nn.Module):
         # decode-shaped rows route to the in-tree skinny MFMA kernel;
         # prefill rows and non-plain modules (e.g. Fp8Linear) fall through
         if type(mod) is nn.Linear:
-            return ops.skinny_linear(x, mod.weight)
+            y = ops.skinny_linear(x, mod.weight)
+            if mod.bias is not None:  # Qwen2 QKV bias (skinny kernel is
+                y = y + mod.bias      # bias-free; one fused add per call)
+            return y
         return mod(x)
 
     def forward(self, x, residual, positions, kv_cache, attn_ctx):
@@ -236,6 +240,18 @@ PRESETS = {
     "llama-tiny": LlamaConfig(vocab_size=512, hidden=256, layers=2, heads=4,
                               kv_heads=2, intermediate=512, rope_theta=10000.0,
                               max_position=512),
+    # Qwen2 family: llama architecture + QKV biases + ChatML template
+    # (models/convert.py convert_hf_qwen2; numerics pinned vs transformers
+    # in tests/test_hf_convert.py)
+    "qwen2-7b": LlamaConfig(vocab_size=152064, hidden=3584, layers=28,
+                            heads=28, kv_heads=4, intermediate=18944,
+                            rope_theta=1e6, rms_eps=1e-6, max_position=32768,
+                            qkv_bias=True),
+    "qwen2-1.5b": LlamaConfig(vocab_size=151936, hidden=1536, layers=28,
+                              heads=12, kv_heads=2, intermediate=8960,
+                              rope_theta=1e6, rms_eps=1e-6,
+                              max_position=32768, qkv_bias=True,
+                              tie_embeddings=True),
 }
 
 

@@ -26,7 +26,8 @@ F8_MAX = 448.0
 
 
 class Fp8Linear(nn.Module):
-    """Bias-free nn.Linear replacement: fp8 weights with per-row scales.
+    """nn.Linear replacement: fp8 weights with per-row scales (an optional
+    bias -- Qwen2 QKV -- stays in the compute dtype, added after the GEMM).
 
     Two entry points:
       forward(x)           -- bf16 in: quantize (one fused kernel) then GEMM
@@ -55,6 +56,11 @@ class Fp8Linear(nn.Module):
         self.out_features = linear.out_features
         self.in_features = linear.in_features
         self.out_dtype = linear.weight.dtype  # model compute dtype
+        # Qwen2 QKV bias: added in the compute dtype after the fp8 GEMM
+        if linear.bias is not None:
+            self.register_buffer("bias", linear.bias.data.clone())
+        else:
+            self.bias = None
 
     # LDS-staged v2 (skinny_gemm_fp8_v2) beats STANDALONE hipBLASLt
     # scaled_mm at M 17..64 on qkv/o_proj shapes, but INSIDE the decode
@@ -87,6 +93,13 @@ class Fp8Linear(nn.Module):
     def forward_q(self, x8: torch.Tensor, x_scale: torch.Tensor,
                   out_dtype=None) -> torch.Tensor:
         out_dtype = out_dtype or self.out_dtype
+        out = self._gemm_q(x8, x_scale, out_dtype)
+        if self.bias is not None:
+            out = out + self.bias.to(out_dtype)
+        return out
+
+    def _gemm_q(self, x8: torch.Tensor, x_scale: torch.Tensor,
+                out_dtype) -> torch.Tensor:
         m = x8.shape[0]
         route = self._route(m) if x8.is_cuda else "lt"
         if route == "v1":

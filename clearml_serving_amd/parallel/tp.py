@@ -140,7 +140,8 @@ def shard_llama_weights(full_state: dict, cfg, rank: int, world: int) -> dict:
 
     out = {}
     for k, v in full_state.items():
-        if ".qkv.weight" in k:
+        if ".qkv.weight" in k or ".qkv.bias" in k:
+            # bias (Qwen2) shards exactly like the weight's output rows
             q = v[rank * hpr * hd:(rank + 1) * hpr * hd]
             kk = v[q_out + rank * kvpr * hd: q_out + (rank + 1) * kvpr * hd]
             vv = v[q_out + kv_out + rank * kvpr * hd:

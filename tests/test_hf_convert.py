@@ -96,3 +96,67 @@ def test_llama_matches_transformers():
         ref = hf(input_ids=ids[None]).logits[0]
         got = ours(ids, positions, kv_caches=None, attn_ctx=attn_ctx)
     torch.testing.assert_close(got, ref, atol=3e-4, rtol=3e-4)
+
+
+def test_qwen2_matches_transformers():
+    """Qwen2 = llama architecture + QKV biases + ChatML; the converter and
+    the native qkv_bias path must reproduce transformers' logits."""
+    from transformers import Qwen2Config as HfQwen2Config
+    from transformers import Qwen2ForCausalLM as HfQwen2
+
+    torch.manual_seed(4)
+    hf_cfg = HfQwen2Config(
+        vocab_size=160, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attention_dropout=0.0, tie_word_embeddings=False, use_cache=False,
+        attn_implementation="eager")
+    hf = HfQwen2(hf_cfg).eval()
+    # Qwen2's defining deviation from llama: q/k/v projections carry biases
+    assert hf.state_dict()["model.layers.0.self_attn.q_proj.bias"] is not None
+
+    cfg = LlamaConfig(vocab_size=160, hidden=64, layers=2, heads=4,
+                      kv_heads=2, intermediate=96, rope_theta=10000.0,
+                      rms_eps=1e-6, max_position=64, qkv_bias=True)
+    ours = LlamaForCausalLM(cfg).eval()
+    from clearml_serving_amd.models.convert import convert_hf_qwen2
+
+    ours.load_state_dict(convert_hf_qwen2(hf.state_dict(), num_layers=2))
+
+    t = 12
+    ids = torch.randint(0, 160, (t,))
+    positions = torch.arange(t, dtype=torch.int32)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": t,
+                "seq_lens": torch.tensor([t], dtype=torch.int32),
+                "slot_mapping": torch.full((t,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        ref = hf(input_ids=ids[None]).logits[0]
+        got = ours(ids, positions, kv_caches=None, attn_ctx=attn_ctx)
+    torch.testing.assert_close(got, ref, atol=3e-4, rtol=3e-4)
+
+
+def test_qwen2_tp_shard_roundtrip():
+    """shard_llama_weights handles the Qwen2 qkv bias: rank shards
+    reassemble to the full tensors for every world size that divides the
+    head counts."""
+    from clearml_serving_amd.parallel.tp import shard_llama_weights
+
+    torch.manual_seed(5)
+    cfg = LlamaConfig(vocab_size=64, hidden=64, layers=1, heads=4,
+                      kv_heads=2, intermediate=64, qkv_bias=True)
+    full = LlamaForCausalLM(cfg)
+    state = full.state_dict()
+    world = 2
+    shards = [shard_llama_weights(state, cfg, r, world)
+              for r in range(world)]
+    hd = cfg.head_dim
+    hpr, kvpr = cfg.heads // world, cfg.kv_heads // world
+    for key in ("layers.0.qkv.weight", "layers.0.qkv.bias"):
+        qs, ks, vs = [], [], []
+        for r in range(world):
+            s = shards[r][key]
+            qs.append(s[:hpr * hd])
+            ks.append(s[hpr * hd:hpr * hd + kvpr * hd])
+            vs.append(s[hpr * hd + kvpr * hd:])
+        rebuilt = torch.cat(qs + ks + vs, dim=0)
+        torch.testing.assert_close(rebuilt, state[key])

@@ -340,3 +340,25 @@ def test_attention_prefill_paged_fp8_kv_close_to_bf16_cpu():
         n = int(q_lens[i])
         torch.testing.assert_close(out[i, :n], ref[i, :n], atol=0.08,
                                    rtol=0.05)
+
+
+def test_fp8_linear_carries_qwen2_bias():
+    """Fp8Linear keeps an nn.Linear's bias (Qwen2 QKV): output matches the
+    bf16 linear to fp8 tolerance, and dropping the bias would not."""
+    from clearml_serving_amd.models.quant import Fp8Linear
+
+    torch.manual_seed(6)
+    lin = torch.nn.Linear(64, 96, bias=True).float()
+    with torch.no_grad():
+        lin.bias.mul_(5.0)  # make the bias contribution dominate
+    q = Fp8Linear(lin)
+    x = torch.randn(4, 64)
+    with torch.inference_mode():
+        ref = lin(x)
+        got = q(x)
+    corr = torch.corrcoef(torch.stack(
+        [ref.flatten(), got.flatten()]))[0, 1]
+    assert corr > 0.999, corr.item()
+    # the biasless GEMM alone is far off: proves the bias is applied
+    biasless = torch.nn.functional.linear(x, lin.weight)
+    assert (got - ref).abs().mean() < 0.1 * (biasless - ref).abs().mean()
