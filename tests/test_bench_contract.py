@@ -60,3 +60,24 @@ def test_bench_dp2_gloo():
     assert out["config"]["parallelism"] == "dp2"
     # whole-job aggregate: world * steps * req_per_step
     assert out["value"] > 0
+
+
+@pytest.mark.timeout(900)
+def test_bench_dp4_gloo():
+    """World=4 on CPU/gloo: half the driver's 8-GPU ladder; catches any
+    rendezvous/barrier/all-reduce issue that only appears beyond 2 ranks
+    (each rank builds its own serving replica, so this is the heaviest
+    CPU test -- tiny step counts keep it bounded)."""
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29562", os.path.join(REPO, "bench.py"),
+         "--gpus", "4", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=880, env=_env(), cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.strip().splitlines()
+            if ln.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 4
+    assert out["config"]["parallelism"] == "dp4"
+    assert out["value"] > 0
