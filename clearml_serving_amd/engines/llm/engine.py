@@ -168,6 +168,12 @@ class LlmEngineConfig:
     weights: Optional[str] = None
     tokenizer_path: Optional[str] = None
     device: Optional[str] = None
+    # speculative decoding (vLLM parity): {"method": "ngram",
+    # "num_spec_tokens": K, "ngram": N}. Prompt-lookup proposals verified
+    # in ONE multi-token forward; EXACT under greedy sampling (accepted
+    # tokens equal what step-by-step decode would emit). Off by default;
+    # greedy sequences only; TP=1 (sampled/TP requests use plain decode).
+    speculative: Optional[Dict[str, Any]] = None
     decode_graphs: bool = True  # capture decode steps into hipGraphs per
                                 # batch bucket (GPU, TP=1): collapses the
                                 # ~350 Python-dispatched launches of a
@@ -198,7 +204,8 @@ class LlmEngineConfig:
         for key in ("preset", "dtype", "block_size", "max_num_seqs",
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "quantization",
-                    "kv_dtype", "weights", "device", "decode_graphs"):
+                    "kv_dtype", "weights", "device", "decode_graphs",
+                    "speculative"):
             for src in (card, aux):
                 if key in src and src[key] is not None:
                     setattr(cfg, key, src[key])
@@ -290,7 +297,8 @@ class LlmEngine:
         self.stats = {"prompt_tokens": 0, "generated_tokens": 0, "steps": 0,
                       "prefill_batches": 0, "decode_batches": 0,
                       "preemptions": 0, "aborts": 0,
-                      "graph_captures": 0, "graph_replays": 0}
+                      "graph_captures": 0, "graph_replays": 0,
+                      "spec_proposed": 0, "spec_accepted": 0}
 
     # ------------------------------------------------------------------ #
     def start(self) -> None:
@@ -552,7 +560,18 @@ class LlmEngine:
         decoding = [s for s in self.running if not s.finished
                     and s.prefilled >= len(s.prompt_ids) and s.output_ids]
         if decoding:
-            self._decode(decoding)
+            spec_k = self._spec_tokens()
+            if spec_k and getattr(self, "tp_size", 1) <= 1:
+                # speculative path: greedy sequences only (acceptance is
+                # exact for argmax); sampled sequences take plain decode
+                spec = [s for s in decoding if s.params.temperature == 0.0]
+                rest = [s for s in decoding if s.params.temperature != 0.0]
+            else:
+                spec, rest = [], decoding
+            if spec:
+                self._decode_spec(spec, spec_k)
+            if rest:
+                self._decode(rest)
         for s in list(self.running):
             if s.finished:
                 self.running.remove(s)
@@ -887,6 +906,157 @@ class LlmEngine:
                           attn_ctx=attn_ctx, last_token_idx=None,
                           gather_logits=False)
 
+    # -------------------- speculative decoding (ngram) ---------------- #
+    # Prompt-lookup speculation (vLLM's "ngram" method): propose the K
+    # tokens that followed the most recent earlier occurrence of the
+    # context's trailing N-gram, verify all of them in ONE multi-token
+    # forward (the chunked-prefill paged-attention path), and accept the
+    # longest prefix that matches greedy argmax -- exact greedy output, up
+    # to K+1 tokens per step. Rollback is free: KV slots are addressed by
+    # sequence position, so a rejected position's K/V is overwritten when
+    # that position is really generated (and never attended before then,
+    # kv_lens only counts committed tokens).
+    def _spec_tokens(self) -> int:
+        spec = self.cfg.speculative
+        if not spec:
+            return 0
+        if str(spec.get("method", "ngram")) != "ngram":
+            raise ValueError("unsupported speculative method '{}' (have: "
+                             "ngram)".format(spec.get("method")))
+        return max(int(spec.get("num_spec_tokens", 4)), 0)
+
+    def _ngram_propose(self, s: Sequence, k: int) -> List[int]:
+        n = max(int((self.cfg.speculative or {}).get("ngram", 2)), 1)
+        ctx = s.prompt_ids + s.output_ids
+        if len(ctx) <= n:
+            return []
+        tail = ctx[-n:]
+        # most recent earlier occurrence of the trailing n-gram
+        for m in range(len(ctx) - n - 1, -1, -1):
+            if ctx[m:m + n] == tail:
+                return list(ctx[m + n:m + n + k])
+        return []
+
+    def _decode_spec(self, seqs: List[Sequence], k: int) -> None:
+        assert getattr(self, "tp_size", 1) <= 1
+        bs = self.cfg.block_size
+        # base-token block growth: identical contract to _decode (preempts
+        # under KV pressure; evicted victims leave the batch)
+        for s in list(seqs):
+            pos = len(s) - 1
+            if pos // bs >= len(s.blocks):
+                while self.allocator.available < 1:
+                    if not self._preempt_one():
+                        raise RuntimeError(
+                            "KV cache exhausted with a single sequence -- "
+                            "raise num_kv_blocks / gpu_memory_fraction")
+                    if s not in self.running:
+                        break
+                if s not in self.running:
+                    seqs.remove(s)
+                    continue
+                s.blocks.extend(self.allocator.alloc(1))
+        seqs[:] = [s for s in seqs if s in self.running]
+        if not seqs:
+            return
+
+        proposals: List[List[int]] = []
+        for s in seqs:
+            cap = min(k,
+                      s.params.max_tokens - s.generated - 1,
+                      self.cfg.max_model_len - len(s) - 1)
+            prop = self._ngram_propose(s, cap) if cap > 0 else []
+            # proposal tokens occupy positions len(s)..len(s)+p-1: extend
+            # the block table opportunistically, trimming to what the
+            # allocator has (never preempt for speculation)
+            while prop:
+                need = (len(s) + len(prop) + bs - 1) // bs - len(s.blocks)
+                if need <= 0:
+                    break
+                if need > self.allocator.available:
+                    prop = prop[:-1]
+                    continue
+                s.blocks.extend(self.allocator.alloc(need))
+                break
+            proposals.append(prop)
+
+        if not any(proposals):
+            # nothing to verify: the plain decode path (hipGraph-replayed
+            # on GPU) is strictly faster than a 1-token chunk forward
+            self._decode(seqs)
+            return
+
+        self.stats["decode_batches"] += 1
+        plan = {
+            "mode": "spec",
+            "tokens": [[s.output_ids[-1]] + prop
+                       for s, prop in zip(seqs, proposals)],
+            "starts": [len(s) - 1 for s in seqs],
+            "kv_lens": [len(s) + len(prop)
+                        for s, prop in zip(seqs, proposals)],
+            "slots": [[self._slot(s, p)
+                       for p in range(len(s) - 1, len(s) + len(prop))]
+                      for s, prop in zip(seqs, proposals)],
+            "blocks": [list(s.blocks) for s in seqs],
+        }
+        logits = self._exec_spec(plan)
+
+        off = 0
+        argmax = logits.argmax(dim=-1).cpu()
+        for s, prop in zip(seqs, proposals):
+            n = 1 + len(prop)
+            rows = argmax[off:off + n]
+            off += n
+            toks: List[int] = []
+            for j in range(n):
+                toks.append(int(rows[j]))
+                if not (j < len(prop) and toks[-1] == prop[j]):
+                    break
+            self.stats["spec_proposed"] = (
+                self.stats.get("spec_proposed", 0) + len(prop))
+            self.stats["spec_accepted"] = (
+                self.stats.get("spec_accepted", 0) + len(toks) - 1)
+            self._emit_tokens(s, toks)
+
+    @torch.inference_mode()
+    def _exec_spec(self, plan: Dict[str, Any]) -> torch.Tensor:
+        """Multi-token verification forward: the chunk layout (padded
+        [b, smax] batch, paged attention over each sequence's committed
+        history + in-flight proposal) but gathering logits at EVERY real
+        row, concatenated in batch order."""
+        dev = self.device
+        b = len(plan["tokens"])
+        lens = [len(t) for t in plan["tokens"]]
+        smax = max(lens)
+        tokens = torch.zeros(b, smax, dtype=torch.long)
+        positions = torch.zeros(b, smax, dtype=torch.int32)
+        slot_map = torch.full((b, smax), -1, dtype=torch.int32)
+        for i in range(b):
+            n = lens[i]
+            tokens[i, :n] = torch.tensor(plan["tokens"][i], dtype=torch.long)
+            positions[i, :n] = torch.arange(
+                plan["starts"][i], plan["starts"][i] + n, dtype=torch.int32)
+            slot_map[i, :n] = torch.tensor(plan["slots"][i],
+                                           dtype=torch.int32)
+        max_blocks = max(len(bl) for bl in plan["blocks"])
+        btab = torch.zeros(b, max_blocks, dtype=torch.int32)
+        for i, bl in enumerate(plan["blocks"]):
+            btab[i, :len(bl)] = torch.tensor(bl, dtype=torch.int32)
+        attn_ctx = {
+            "mode": "prefill_paged", "batch": b, "seq": smax,
+            "kv_lens": torch.tensor(plan["kv_lens"], dtype=torch.int32,
+                                    device=dev),
+            "q_lens": torch.tensor(lens, dtype=torch.int32, device=dev),
+            "block_table": btab.to(dev),
+            "slot_mapping": slot_map.view(-1).to(dev),
+        }
+        rows = [i * smax + j for i in range(b) for j in range(lens[i])]
+        last_idx = torch.tensor(rows, dtype=torch.long, device=dev)
+        return self.model(tokens.view(-1).to(dev),
+                          positions.view(-1).to(dev),
+                          kv_caches=self.kv_caches, attn_ctx=attn_ctx,
+                          last_token_idx=last_idx, gather_logits=False)
+
     # -------------------- decode hipGraph capture --------------------- #
     # A llama-8B decode step dispatches ~350 kernels from Python; at B=64
     # the weight streaming itself is only ~2.5 ms, so the step is
@@ -1133,10 +1303,18 @@ class LlmEngine:
                 sampled = ops.sample_top_k_top_p(
                     rows, temperature=temp, top_k=top_k, top_p=top_p)
                 next_ids[idxs] = sampled.cpu()
+        for i, s in enumerate(seqs):
+            self._emit_tokens(s, [int(next_ids[i])])
+
+    def _emit_tokens(self, s: Sequence, toks: List[int]) -> None:
+        """Append generated tokens to a sequence, applying the stop/eos/
+        length rules per token (emission halts at the first finish -- extra
+        speculative tokens past a stop are dropped)."""
         eos = self.tokenizer.eos_id
         now = time.time()
-        for i, s in enumerate(seqs):
-            tok = int(next_ids[i])
+        for tok in toks:
+            if s.finished:
+                break
             s.output_ids.append(tok)
             s.generated += 1
             if s.first_token_time is None:
