@@ -135,6 +135,29 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     async def detokenize(self, body, state, collect_fn=None):
         return self._engine.openai_detokenize(body)
 
+    # ---- reference serve types that need a model class this stack serves
+    # elsewhere or not at all (the reference's vLLM returns the same class
+    # of error when the model's task/runner_type does not match the
+    # handler, preprocess_service.py:775-808, 987-1068) ---------------- #
+    async def classify(self, body, state, collect_fn=None):
+        raise ValueError(
+            "classify needs a sequence-classification model; decoder-only "
+            "LLM endpoints have no classification head. Serve "
+            "classification natively through the hip engine instead (e.g. "
+            "a bert-base model card with num_labels -- examples/bert).")
+
+    async def v1_audio_transcriptions(self, body, state, collect_fn=None):
+        raise ValueError(
+            "audio transcription needs a speech model (the reference "
+            "delegates to vLLM Whisper runners); no audio model family or "
+            "audio dependencies exist in this stack's target image.")
+
+    async def v1_audio_translations(self, body, state, collect_fn=None):
+        raise ValueError(
+            "audio translation needs a speech model (the reference "
+            "delegates to vLLM Whisper runners); no audio model family or "
+            "audio dependencies exist in this stack's target image.")
+
 
 # reference-CLI compatibility: `--engine vllm` runs the native LLM engine
 BasePreprocessRequest.register_engine("vllm")(LlmPreprocessRequest)

@@ -122,7 +122,8 @@ class ShmLlmProxyRequest(ShmProxyRequest):
         # v1_chat_completions / v1_completions / v1_embeddings / ... all
         # ship the same way; generated on demand
         if name.startswith("v1_") or name in (
-                "pooling", "tokenize", "detokenize", "v2_rerank"):
+                "pooling", "tokenize", "detokenize", "v2_rerank",
+                "classify"):
             async def _method(body, state, collect_fn=None, _st=name):
                 # shipped as the method name; the owner resolves it the same
                 # way the processor does (serve_type.replace('/', '_'))

@@ -221,3 +221,20 @@ def test_removed_llm_endpoint_frees_engine(processor, store, tmp_path):
         LlmPreprocessRequest._engine_singleton = None
         LlmPreprocessRequest._engines = {}
         LlmPreprocessRequest._engine_refs = {}
+
+
+def test_unsupported_vllm_serve_types_explain_themselves(llm_client):
+    """classify / audio transcription / translation: the reference's vLLM
+    errors when the model's task does not match the handler; here the
+    routes 422 with a message naming the alternative (hip-engine
+    classification) or the missing model family (audio)."""
+    r = llm_client.post("/serve/openai/classify", json={
+        "model": "test_llm", "input": "some text"})
+    assert r.status_code == 422, r.text
+    assert "classification" in r.json()["detail"]
+
+    for route in ("v1/audio/transcriptions", "v1/audio/translations"):
+        r = llm_client.post("/serve/openai/" + route, json={
+            "model": "test_llm"})
+        assert r.status_code == 422, r.text
+        assert "audio" in r.json()["detail"]
