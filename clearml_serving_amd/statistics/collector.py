@@ -147,7 +147,12 @@ class StatsRegistry:
 
 
 class GpuStatsExporter:
-    """HBM/utilization gauges from the serving process itself."""
+    """HBM + utilization gauges from the serving process itself.
+
+    The reference scrapes these from Triton's own Prometheus exporter
+    (triton_helper.py:45-89 -- nv_gpu_utilization / nv_gpu_memory_*);
+    here HBM comes from torch.cuda and GPU/memory-controller activity
+    from amdsmi (amdsmi_get_gpu_activity), when available."""
 
     def __init__(self, registry=None, interval_sec: float = 10.0):
         self._registry = registry or REGISTRY
@@ -160,15 +165,33 @@ class GpuStatsExporter:
             self._hbm_total = Gauge(
                 "gpu_hbm_total_bytes", "HBM3E bytes total", ("gpu",),
                 registry=self._registry)
+            self._gfx_util = Gauge(
+                "gpu_utilization_percent", "GFX engine activity", ("gpu",),
+                registry=self._registry)
+            self._umc_util = Gauge(
+                "gpu_memory_activity_percent",
+                "memory-controller (UMC) activity", ("gpu",),
+                registry=self._registry)
         except ValueError:
             self._hbm_used = self._hbm_total = None
+            self._gfx_util = self._umc_util = None
         self._thread: Optional[threading.Thread] = None
+        self._amdsmi = None
+        self._amdsmi_devs = []
 
     def start(self):
         import torch
 
         if not torch.cuda.is_available() or self._hbm_used is None:
             return
+        try:
+            import amdsmi
+
+            amdsmi.amdsmi_init()
+            self._amdsmi = amdsmi
+            self._amdsmi_devs = amdsmi.amdsmi_get_processor_handles()
+        except Exception:
+            self._amdsmi = None  # HBM-only export
         self._thread = threading.Thread(target=self._loop, daemon=True)
         self._thread.start()
 
@@ -183,10 +206,28 @@ class GpuStatsExporter:
                     self._hbm_total.labels(gpu=str(i)).set(total)
             except Exception:
                 pass
+            if self._amdsmi is not None:
+                try:
+                    for i, dev in enumerate(self._amdsmi_devs):
+                        act = self._amdsmi.amdsmi_get_gpu_activity(dev)
+                        gfx = act.get("gfx_activity")
+                        umc = act.get("umc_activity")
+                        if gfx is not None:
+                            self._gfx_util.labels(gpu=str(i)).set(gfx)
+                        if umc is not None:
+                            self._umc_util.labels(gpu=str(i)).set(umc)
+                except Exception:
+                    pass
             time.sleep(self._interval)
 
     def stop(self):
         self._stop = True
+        if self._amdsmi is not None:
+            try:
+                self._amdsmi.amdsmi_shut_down()
+            except Exception:
+                pass
+            self._amdsmi = None
 
 
 class BatcherStagesCollector:
