@@ -471,8 +471,8 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
   const int max_blocks = block_table.size(1);
   const int GQ = H / Hkv;
   TORCH_CHECK(H % Hkv == 0 && (D == 64 || D == 128));
-  TORCH_CHECK(GQ == 1 || GQ == 2 || GQ == 4 || GQ == 8,
-              "GQA group must be 1/2/4/8, got ", GQ);
+  TORCH_CHECK(GQ >= 1 && GQ <= 8,
+              "GQA group must be 1..8, got ", GQ);
   auto bt = block_table.to(q.device(), at::kInt).contiguous();
   auto sl = seq_lens.to(q.device(), at::kInt).contiguous();
   auto out = torch::empty_like(q);
@@ -509,17 +509,28 @@ torch::Tensor attention_decode(torch::Tensor q, torch::Tensor k_cache,
 #define LAUNCH_DEC_F(DD, GG, UU, MM)                                         \
   do { if (fp8kv) LAUNCH_DEC(DD, GG, UU, MM, true);                          \
        else LAUNCH_DEC(DD, GG, UU, MM, false); } while (0)
+  // GQ 3/5/6/7 (qwen2-7b is 28q/4kv = 7, qwen2-1.5b 12q/2kv = 6) run the
+  // same kernel with shallower UNROLL: per-group accumulator arrays
+  // (o_acc[GQ][8] + score[UNROLL][GQ]) scale VGPRs linearly with GQ
   if (D == 128) {
     if (GQ == 1) LAUNCH_DEC_F(128, 1, 4, 2);
     else if (GQ == 2) LAUNCH_DEC_F(128, 2, 4, 2);
+    else if (GQ == 3) LAUNCH_DEC_F(128, 3, 4, 2);
     // GQ=4 (llama-3 shapes): occupancy-first variant measured 3-6% faster
     // at B=32-64, S=1-4k (scripts/decode_ab.hip, profiles/decode_ab.txt)
     else if (GQ == 4) LAUNCH_DEC_F(128, 4, 2, 4);
+    else if (GQ == 5) LAUNCH_DEC_F(128, 5, 2, 2);
+    else if (GQ == 6) LAUNCH_DEC_F(128, 6, 2, 2);
+    else if (GQ == 7) LAUNCH_DEC_F(128, 7, 2, 2);
     else LAUNCH_DEC_F(128, 8, 4, 2);
   } else {
     if (GQ == 1) LAUNCH_DEC_F(64, 1, 4, 2);
     else if (GQ == 2) LAUNCH_DEC_F(64, 2, 4, 2);
+    else if (GQ == 3) LAUNCH_DEC_F(64, 3, 4, 2);
     else if (GQ == 4) LAUNCH_DEC_F(64, 4, 4, 2);
+    else if (GQ == 5) LAUNCH_DEC_F(64, 5, 4, 2);
+    else if (GQ == 6) LAUNCH_DEC_F(64, 6, 4, 2);
+    else if (GQ == 7) LAUNCH_DEC_F(64, 7, 4, 2);
     else LAUNCH_DEC_F(64, 8, 4, 2);
   }
 #undef LAUNCH_DEC_F
