@@ -99,6 +99,14 @@ class SamplingParams:
     stop_token_ids: List[int] = field(default_factory=list)
     stop: List[str] = field(default_factory=list)
     ignore_eos: bool = False
+    seed: Optional[int] = None           # reproducible sampling per request
+    presence_penalty: float = 0.0        # OpenAI: [-2, 2], output tokens
+    frequency_penalty: float = 0.0       # OpenAI: [-2, 2], output counts
+    repetition_penalty: float = 1.0      # CTRL-style, prompt+output tokens
+
+    def has_penalties(self) -> bool:
+        return bool(self.presence_penalty or self.frequency_penalty
+                    or self.repetition_penalty != 1.0)
 
     @classmethod
     def from_request(cls, body: Dict[str, Any], default_max: int = 128):
@@ -109,11 +117,25 @@ class SamplingParams:
         try:
             temperature = float(body.get("temperature", 1.0))
             top_k = int(body.get("top_k", 0) or 0)
+            # max_completion_tokens: the newer OpenAI name for max_tokens
             top_p = float(body.get("top_p", 1.0))
-            max_tokens = int(body.get("max_tokens", default_max))
+            mt = body.get("max_tokens")
+            if mt is None:
+                mt = body.get("max_completion_tokens")
+            max_tokens = int(mt) if mt is not None else int(default_max)
+            def _num(key, default):
+                v = body.get(key)
+                return float(v) if v is not None else default
+
+            presence = _num("presence_penalty", 0.0)
+            frequency = _num("frequency_penalty", 0.0)
+            repetition = _num("repetition_penalty", 1.0)
+            seed = body.get("seed")
+            seed = int(seed) if seed is not None else None
         except (TypeError, ValueError):
             raise ValueError("sampling parameters must be numeric "
-                             "(temperature/top_k/top_p/max_tokens)")
+                             "(temperature/top_k/top_p/max_tokens/"
+                             "penalties/seed)")
         if temperature < 0.0:
             raise ValueError(
                 "temperature must be >= 0 (0 = greedy), got {}".format(
@@ -126,6 +148,15 @@ class SamplingParams:
         if max_tokens < 1:
             raise ValueError("max_tokens must be >= 1, got {}".format(
                 max_tokens))
+        if not (-2.0 <= presence <= 2.0):
+            raise ValueError("presence_penalty must be in [-2, 2], got {}"
+                             .format(presence))
+        if not (-2.0 <= frequency <= 2.0):
+            raise ValueError("frequency_penalty must be in [-2, 2], got {}"
+                             .format(frequency))
+        if repetition <= 0.0:
+            raise ValueError("repetition_penalty must be > 0, got {}"
+                             .format(repetition))
         stop = body.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
@@ -144,6 +175,10 @@ class SamplingParams:
             stop_token_ids=stop_ids,
             stop=[s for s in stop if s],
             ignore_eos=bool(body.get("ignore_eos", False)),
+            seed=seed,
+            presence_penalty=presence,
+            frequency_penalty=frequency,
+            repetition_penalty=repetition,
         )
 
 
@@ -410,6 +445,12 @@ class LlmEngine:
             raise ValueError(
                 "prompt length {} exceeds max_model_len {}".format(
                     len(prompt_ids), self.cfg.max_model_len))
+        if params.has_penalties() and getattr(self, "tp_size", 1) > 1:
+            # worker ranks hold vocab shards but not the token history the
+            # penalties need; fail loudly instead of silently ignoring
+            raise ValueError(
+                "presence/frequency/repetition penalties are not supported "
+                "under tensor parallelism yet")
         seq = Sequence(uuid.uuid4().hex, prompt_ids, params)
         self.waiting.append(seq)
         self._ensure_loop()
@@ -563,9 +604,11 @@ class LlmEngine:
             spec_k = self._spec_tokens()
             if spec_k and getattr(self, "tp_size", 1) <= 1:
                 # speculative path: greedy sequences only (acceptance is
-                # exact for argmax); sampled sequences take plain decode
-                spec = [s for s in decoding if s.params.temperature == 0.0]
-                rest = [s for s in decoding if s.params.temperature != 0.0]
+                # exact for argmax); sampled or penalized sequences take
+                # plain decode (penalties change the argmax per position)
+                spec = [s for s in decoding if s.params.temperature == 0.0
+                        and not s.params.has_penalties()]
+                rest = [s for s in decoding if s not in spec]
             else:
                 spec, rest = [], decoding
             if spec:
@@ -622,8 +665,12 @@ class LlmEngine:
         return seq.blocks[pos // bs] * bs + pos % bs
 
     def _sample_spec(self, seqs: List[Sequence]) -> List[tuple]:
-        seed = random.getrandbits(31)
-        return [(s.params.temperature, s.params.top_k, s.params.top_p, seed)
+        step_seed = random.getrandbits(31)
+        # a request-level seed makes the sequence reproducible: vary it per
+        # emitted token (seed + generated) so steps draw fresh noise
+        return [(s.params.temperature, s.params.top_k, s.params.top_p,
+                 (s.params.seed + s.generated) & 0x7FFFFFFF
+                 if s.params.seed is not None else step_seed)
                 for s in seqs]
 
     def _prefill(self, seqs: List[Sequence]) -> None:
@@ -1263,6 +1310,41 @@ class LlmEngine:
             with self._exec_lock:
                 self._tp_broadcast({"mode": "stop"})
 
+    def _apply_penalties(self, logits: torch.Tensor,
+                         seqs: List[Sequence]) -> torch.Tensor:
+        """OpenAI presence/frequency penalties (output tokens) + CTRL-style
+        repetition penalty (prompt+output). Applied per requesting row only;
+        the tensor is cloned first because decode-graph logits are static
+        replay buffers that must never be mutated."""
+        need = [i for i, s in enumerate(seqs) if s.params.has_penalties()]
+        if not need:
+            return logits
+        V = logits.shape[-1]
+        dev = logits.device
+        logits = logits.clone()
+        for i in need:
+            s = seqs[i]
+            p = s.params
+            row = logits[i].float()
+            if p.repetition_penalty != 1.0:
+                ctx = [t for t in dict.fromkeys(s.prompt_ids + s.output_ids)
+                       if 0 <= t < V]
+                if ctx:
+                    idx = torch.tensor(ctx, dtype=torch.long, device=dev)
+                    vals = row[idx]
+                    row[idx] = torch.where(
+                        vals > 0, vals / p.repetition_penalty,
+                        vals * p.repetition_penalty)
+            if s.output_ids and (p.presence_penalty or p.frequency_penalty):
+                out = torch.tensor(
+                    [t for t in s.output_ids if 0 <= t < V],
+                    dtype=torch.long, device=dev)
+                counts = torch.bincount(out, minlength=V).to(row.dtype)
+                row -= p.frequency_penalty * counts
+                row -= p.presence_penalty * (counts > 0).to(row.dtype)
+            logits[i] = row.to(logits.dtype)
+        return logits
+
     def _hit_stop_string(self, s: Sequence) -> bool:
         """OpenAI 'stop' strings: decode a bounded tail window of the output
         after each token and look for any stop sequence (the reference's
@@ -1292,16 +1374,25 @@ class LlmEngine:
             next_ids = self._tp_sample_rows(
                 logits, sample or self._sample_spec(seqs))
         else:
+            logits = self._apply_penalties(logits, seqs)
             # group rows by identical sampling params for batched kernels
+            # (a request-level seed gets its own group + generator)
             groups: Dict[tuple, List[int]] = {}
             for i, s in enumerate(seqs):
-                key = (s.params.temperature, s.params.top_k, s.params.top_p)
+                key = (s.params.temperature, s.params.top_k, s.params.top_p,
+                       s.params.seed)
                 groups.setdefault(key, []).append(i)
             next_ids = torch.empty(len(seqs), dtype=torch.long)
-            for (temp, top_k, top_p), idxs in groups.items():
+            for (temp, top_k, top_p, seed), idxs in groups.items():
                 rows = logits[idxs] if len(idxs) < len(seqs) else logits
+                gen = None
+                if seed is not None and temp != 0.0:
+                    s0 = seqs[idxs[0]]
+                    gen = torch.Generator(device=logits.device)
+                    gen.manual_seed((seed + s0.generated) & 0x7FFFFFFF)
                 sampled = ops.sample_top_k_top_p(
-                    rows, temperature=temp, top_k=top_k, top_p=top_p)
+                    rows, temperature=temp, top_k=top_k, top_p=top_p,
+                    generator=gen)
                 next_ids[idxs] = sampled.cpu()
         for i, s in enumerate(seqs):
             self._emit_tokens(s, [int(next_ids[i])])
@@ -1370,22 +1461,54 @@ class LlmEngine:
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
 
+    @staticmethod
+    def _n_choices(body: Dict[str, Any]) -> int:
+        n = body.get("n", 1)
+        try:
+            n = int(n) if n is not None else 1
+        except (TypeError, ValueError):
+            raise ValueError("'n' must be an integer")
+        if not (1 <= n <= 16):
+            raise ValueError("'n' must be in [1, 16], got {}".format(n))
+        if n > 1 and body.get("stream"):
+            raise ValueError("streaming with n > 1 is not supported; "
+                             "request the choices without 'stream'")
+        return n
+
+    def _choice_params(self, body: Dict[str, Any], n: int
+                       ) -> List[SamplingParams]:
+        """One SamplingParams per choice; an explicit request seed is
+        offset per choice so the n completions differ (same OpenAI
+        semantics: n identical greedy choices are expected)."""
+        out = []
+        for i in range(n):
+            p = SamplingParams.from_request(body)
+            if p.seed is not None:
+                p.seed = (p.seed + 7919 * i) & 0x7FFFFFFF
+            out.append(p)
+        return out
+
     async def openai_chat_completions(self, body: Dict[str, Any],
                                       model_name: str):
         messages = body.get("messages") or []
         prompt = self._chat_prompt(messages)
-        params = SamplingParams.from_request(body)
+        n = self._n_choices(body)
+        params = self._choice_params(body, n)
         rid = "chatcmpl-" + uuid.uuid4().hex[:24]
         if body.get("stream"):
-            return self._sse_stream(prompt, params, rid, model_name,
+            return self._sse_stream(prompt, params[0], rid, model_name,
                                     chat=True)
-        text, reason, ntok, nprompt = await self._collect(prompt, params)
+        results = await asyncio.gather(
+            *[self._collect(prompt, p) for p in params])
+        nprompt = results[0][3]
+        ntok = sum(r[2] for r in results)
         return {
             "id": rid, "object": "chat.completion", "created": int(time.time()),
             "model": model_name,
-            "choices": [{"index": 0,
-                         "message": {"role": "assistant", "content": text},
-                         "finish_reason": reason}],
+            "choices": [{"index": i,
+                         "message": {"role": "assistant", "content": r[0]},
+                         "finish_reason": r[1]}
+                        for i, r in enumerate(results)],
             "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
                       "total_tokens": nprompt + ntok},
         }
@@ -1394,16 +1517,21 @@ class LlmEngine:
         prompt = body.get("prompt") or ""
         if isinstance(prompt, list):
             prompt = prompt[0] if prompt else ""
-        params = SamplingParams.from_request(body)
+        n = self._n_choices(body)
+        params = self._choice_params(body, n)
         rid = "cmpl-" + uuid.uuid4().hex[:24]
         if body.get("stream"):
-            return self._sse_stream(prompt, params, rid, model_name,
+            return self._sse_stream(prompt, params[0], rid, model_name,
                                     chat=False)
-        text, reason, ntok, nprompt = await self._collect(prompt, params)
+        results = await asyncio.gather(
+            *[self._collect(prompt, p) for p in params])
+        nprompt = results[0][3]
+        ntok = sum(r[2] for r in results)
         return {
             "id": rid, "object": "text_completion", "created": int(time.time()),
             "model": model_name,
-            "choices": [{"index": 0, "text": text, "finish_reason": reason}],
+            "choices": [{"index": i, "text": r[0], "finish_reason": r[1]}
+                        for i, r in enumerate(results)],
             "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
                       "total_tokens": nprompt + ntok},
         }

@@ -1061,3 +1061,87 @@ def test_spec_decode_respects_max_tokens_and_eos():
     toks, reason = run(gen())
     assert toks[-1] == first and reason == "stop"
     assert len(toks) <= 20
+
+
+# ------------------------------------------------------------------ #
+# OpenAI sampling controls: n, seed, penalties
+# ------------------------------------------------------------------ #
+def test_openai_n_choices():
+    eng = tiny_engine()
+    resp = run(eng.openai_chat_completions(
+        {"messages": [{"role": "user", "content": "hi"}], "max_tokens": 4,
+         "temperature": 0.0, "ignore_eos": True, "n": 3}, "m"))
+    assert len(resp["choices"]) == 3
+    assert [c["index"] for c in resp["choices"]] == [0, 1, 2]
+    # greedy: all n choices identical (OpenAI semantics)
+    assert len({c["message"]["content"] for c in resp["choices"]}) == 1
+    assert resp["usage"]["completion_tokens"] == 12
+
+    with pytest.raises(ValueError):
+        run(eng.openai_completions({"prompt": "x", "n": 0}, "m"))
+    with pytest.raises(ValueError):
+        run(eng.openai_completions(
+            {"prompt": "x", "n": 2, "stream": True}, "m"))
+
+
+def test_seed_reproducible_sampling():
+    """Same request seed -> same sampled tokens; different seed differs
+    (vLLM/OpenAI 'seed' semantics)."""
+    eng = tiny_engine()
+
+    def gen(seed):
+        async def go():
+            toks = []
+            async for item in eng.generate("seed probe", SamplingParams(
+                    temperature=1.0, max_tokens=12, ignore_eos=True,
+                    seed=seed)):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return run(go())
+
+    a, b, c = gen(1234), gen(1234), gen(99)
+    assert a == b
+    assert a != c
+
+
+def test_frequency_penalty_reduces_repetition():
+    """Random-init tiny models loop on a few tokens under greedy; a strong
+    frequency penalty must strictly increase the distinct-token count."""
+    eng = tiny_engine()
+
+    def gen(**kw):
+        async def go():
+            toks = []
+            async for item in eng.generate("rep probe", SamplingParams(
+                    temperature=0.0, max_tokens=24, ignore_eos=True, **kw)):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return run(go())
+
+    plain = gen()
+    penalized = gen(frequency_penalty=2.0, presence_penalty=2.0)
+    assert len(set(penalized)) > len(set(plain)), (plain, penalized)
+
+    rep = gen(repetition_penalty=5.0)
+    assert len(set(rep)) > len(set(plain))
+
+
+def test_penalty_validation_and_tp_guard():
+    eng = tiny_engine()
+    with pytest.raises(ValueError):
+        SamplingParams.from_request({"presence_penalty": 3.0})
+    with pytest.raises(ValueError):
+        SamplingParams.from_request({"frequency_penalty": -2.5})
+    with pytest.raises(ValueError):
+        SamplingParams.from_request({"repetition_penalty": 0.0})
+    # max_completion_tokens is honored as the newer max_tokens name
+    p = SamplingParams.from_request({"max_completion_tokens": 7})
+    assert p.max_tokens == 7
+
+    # TP guard: penalties refuse loudly instead of silently diverging
+    eng.tp_size = 2
+    with pytest.raises(ValueError):
+        run(eng.add_request([1, 2], SamplingParams(presence_penalty=1.0)))
+    eng.tp_size = 1
