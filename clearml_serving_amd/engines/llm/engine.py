@@ -103,6 +103,8 @@ class SamplingParams:
     presence_penalty: float = 0.0        # OpenAI: [-2, 2], output tokens
     frequency_penalty: float = 0.0       # OpenAI: [-2, 2], output counts
     repetition_penalty: float = 1.0      # CTRL-style, prompt+output tokens
+    logprobs: Optional[int] = None       # None = off; N = N top_logprobs
+                                         # alongside each chosen token
 
     def has_penalties(self) -> bool:
         return bool(self.presence_penalty or self.frequency_penalty
@@ -157,6 +159,18 @@ class SamplingParams:
         if repetition <= 0.0:
             raise ValueError("repetition_penalty must be > 0, got {}"
                              .format(repetition))
+        # logprobs: completions style ("logprobs": N) or chat style
+        # ("logprobs": true + "top_logprobs": N)
+        lp = body.get("logprobs")
+        if isinstance(lp, bool):
+            lp = (int(body.get("top_logprobs", 0) or 0)) if lp else None
+        elif lp is not None:
+            try:
+                lp = int(lp)
+            except (TypeError, ValueError):
+                raise ValueError("'logprobs' must be an int or bool")
+        if lp is not None and not (0 <= lp <= 20):
+            raise ValueError("logprobs must be in [0, 20], got {}".format(lp))
         stop = body.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
@@ -179,6 +193,7 @@ class SamplingParams:
             presence_penalty=presence,
             frequency_penalty=frequency,
             repetition_penalty=repetition,
+            logprobs=lp,
         )
 
 
@@ -451,6 +466,10 @@ class LlmEngine:
             raise ValueError(
                 "presence/frequency/repetition penalties are not supported "
                 "under tensor parallelism yet")
+        if params.logprobs is not None and getattr(self, "tp_size", 1) > 1:
+            raise ValueError(
+                "logprobs need a full-vocab softmax and are not supported "
+                "under tensor parallelism yet")
         seq = Sequence(uuid.uuid4().hex, prompt_ids, params)
         self.waiting.append(seq)
         self._ensure_loop()
@@ -607,7 +626,8 @@ class LlmEngine:
                 # exact for argmax); sampled or penalized sequences take
                 # plain decode (penalties change the argmax per position)
                 spec = [s for s in decoding if s.params.temperature == 0.0
-                        and not s.params.has_penalties()]
+                        and not s.params.has_penalties()
+                        and s.params.logprobs is None]
                 rest = [s for s in decoding if s not in spec]
             else:
                 spec, rest = [], decoding
@@ -1395,15 +1415,31 @@ class LlmEngine:
                     generator=gen)
                 next_ids[idxs] = sampled.cpu()
         for i, s in enumerate(seqs):
-            self._emit_tokens(s, [int(next_ids[i])])
+            tok = int(next_ids[i])
+            extra = None
+            if s.params.logprobs is not None \
+                    and getattr(self, "tp_size", 1) <= 1:
+                # logprobs of the ACTUAL sampling distribution (post-
+                # penalty logits), like the reference's vLLM path
+                row = torch.log_softmax(logits[i].float(), dim=-1)
+                e = {"logprob": float(row[tok])}
+                k = s.params.logprobs
+                if k:
+                    topv, topi = row.topk(k)
+                    e["top_logprobs"] = {
+                        int(t): float(v)
+                        for v, t in zip(topv.tolist(), topi.tolist())}
+                extra = [e]
+            self._emit_tokens(s, [tok], extras=extra)
 
-    def _emit_tokens(self, s: Sequence, toks: List[int]) -> None:
+    def _emit_tokens(self, s: Sequence, toks: List[int],
+                     extras: Optional[List[Optional[dict]]] = None) -> None:
         """Append generated tokens to a sequence, applying the stop/eos/
         length rules per token (emission halts at the first finish -- extra
         speculative tokens past a stop are dropped)."""
         eos = self.tokenizer.eos_id
         now = time.time()
-        for tok in toks:
+        for ti, tok in enumerate(toks):
             if s.finished:
                 break
             s.output_ids.append(tok)
@@ -1424,12 +1460,15 @@ class LlmEngine:
                 finished, reason = True, "length"
             s.finished = finished
             s.finish_reason = reason
-            s.stream.put_nowait({
+            item = {
                 "token_ids": [tok],
                 "text": self.tokenizer.decode([tok]),
                 "finished": finished,
                 "finish_reason": reason,
-            })
+            }
+            if extras is not None and ti < len(extras) and extras[ti]:
+                item.update(extras[ti])
+            s.stream.put_nowait(item)
 
     # ------------------------------------------------------------------ #
     # OpenAI-compatible handlers (route /serve/openai/v1/*)
@@ -1502,13 +1541,18 @@ class LlmEngine:
             *[self._collect(prompt, p) for p in params])
         nprompt = results[0][3]
         ntok = sum(r[2] for r in results)
+        choices = []
+        for i, r in enumerate(results):
+            choice = {"index": i,
+                      "message": {"role": "assistant", "content": r[0]},
+                      "finish_reason": r[1]}
+            if params[i].logprobs is not None:
+                choice["logprobs"] = self._chat_logprobs(r[4])
+            choices.append(choice)
         return {
             "id": rid, "object": "chat.completion", "created": int(time.time()),
             "model": model_name,
-            "choices": [{"index": i,
-                         "message": {"role": "assistant", "content": r[0]},
-                         "finish_reason": r[1]}
-                        for i, r in enumerate(results)],
+            "choices": choices,
             "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
                       "total_tokens": nprompt + ntok},
         }
@@ -1527,11 +1571,16 @@ class LlmEngine:
             *[self._collect(prompt, p) for p in params])
         nprompt = results[0][3]
         ntok = sum(r[2] for r in results)
+        choices = []
+        for i, r in enumerate(results):
+            choice = {"index": i, "text": r[0], "finish_reason": r[1]}
+            if params[i].logprobs is not None:
+                choice["logprobs"] = self._completion_logprobs(r[4])
+            choices.append(choice)
         return {
             "id": rid, "object": "text_completion", "created": int(time.time()),
             "model": model_name,
-            "choices": [{"index": i, "text": r[0], "finish_reason": r[1]}
-                        for i, r in enumerate(results)],
+            "choices": choices,
             "usage": {"prompt_tokens": nprompt, "completion_tokens": ntok,
                       "total_tokens": nprompt + ntok},
         }
@@ -1617,6 +1666,7 @@ class LlmEngine:
         ids = self.tokenizer.encode(prompt)
         tokens: List[int] = []
         reason = None
+        lps: List[dict] = []
         async for item in self.generate(prompt, params):
             if item.get("error"):
                 # engine-side failure: surface it (the route maps to 500)
@@ -1624,10 +1674,50 @@ class LlmEngine:
                 raise RuntimeError("generation failed: {}".format(
                     item["error"]))
             tokens.extend(item.get("token_ids", []))
+            if "logprob" in item:
+                lps.append({"token_ids": item["token_ids"],
+                            "logprob": item["logprob"],
+                            "top_logprobs": item.get("top_logprobs") or {}})
             reason = item.get("finish_reason") or reason
         text = self._truncate_at_stop(
             self.tokenizer.decode(tokens), params, reason)
-        return text, reason, len(tokens), len(ids)
+        return text, reason, len(tokens), len(ids), lps
+
+    def _lp_entry(self, tok_id: int, logprob: float,
+                  top: Dict[int, float]) -> dict:
+        tokstr = self.tokenizer.decode([tok_id])
+        return {"token": tokstr, "logprob": logprob,
+                "bytes": list(tokstr.encode()),
+                "top_logprobs": [
+                    {"token": self.tokenizer.decode([t]), "logprob": v}
+                    for t, v in top.items()]}
+
+    def _chat_logprobs(self, lps: List[dict]) -> Optional[dict]:
+        if not lps:
+            return None
+        return {"content": [
+            self._lp_entry(e["token_ids"][0], e["logprob"],
+                           e["top_logprobs"]) for e in lps]}
+
+    def _completion_logprobs(self, lps: List[dict]) -> Optional[dict]:
+        if not lps:
+            return None
+        tops = []
+        for e in lps:
+            d: Dict[str, float] = {}
+            for t, v in e["top_logprobs"].items():
+                # this format keys by decoded STRING; ids that decode to
+                # the same text keep the max (never shadow the top-1)
+                k = self.tokenizer.decode([t])
+                if k not in d or v > d[k]:
+                    d[k] = v
+            tops.append(d)
+        return {
+            "tokens": [self.tokenizer.decode(e["token_ids"]) for e in lps],
+            "token_logprobs": [e["logprob"] for e in lps],
+            "top_logprobs": tops,
+            "text_offset": [],
+        }
 
     def _sse_stream(self, prompt: str, params: SamplingParams, rid: str,
                     model_name: str, chat: bool):
@@ -1649,6 +1739,14 @@ class LlmEngine:
                     choice = {"index": 0, "text": item.get("text", ""),
                               "finish_reason": item.get("finish_reason")}
                     obj = "text_completion"
+                if "logprob" in item:
+                    entry = self._lp_entry(item["token_ids"][0],
+                                           item["logprob"],
+                                           item.get("top_logprobs") or {})
+                    choice["logprobs"] = ({"content": [entry]} if chat
+                                          else {"tokens": [entry["token"]],
+                                                "token_logprobs":
+                                                    [entry["logprob"]]})
                 chunk = {"id": rid, "object": obj,
                          "created": int(time.time()), "model": model_name,
                          "choices": [choice]}
