@@ -67,3 +67,67 @@ def test_block_allocator_slot_math(prompt, block):
         assert s // block == blocks[p // block]
     alloc.free(blocks)
     assert alloc.available == 64
+
+
+# --------------------------------------------------------------------- #
+# speculative decoding: exactness as a PROPERTY over random prompts
+# --------------------------------------------------------------------- #
+@settings(max_examples=12, deadline=None)
+@given(prompt=st.lists(st.integers(min_value=1, max_value=500),
+                       min_size=2, max_size=40),
+       k=st.integers(min_value=1, max_value=6),
+       ngram=st.integers(min_value=1, max_value=3),
+       max_tokens=st.integers(min_value=1, max_value=20))
+def test_spec_decode_exactness_property(prompt, k, ngram, max_tokens):
+    """For ANY prompt/config, ngram speculative decoding emits exactly the
+    tokens plain greedy decode would (the defining acceptance invariant)."""
+    import asyncio
+
+    import torch
+
+    from clearml_serving_amd.engines.llm.engine import (
+        LlmEngine, LlmEngineConfig, SamplingParams)
+
+    global _SPEC_ENGINES
+    try:
+        plain, spec = _SPEC_ENGINES
+    except NameError:
+        def mk(speculative):
+            torch.manual_seed(7)
+            cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=128,
+                                  block_size=16, max_model_len=128,
+                                  device="cpu", speculative=speculative)
+            e = LlmEngine(cfg)
+            e.start()
+            return e
+
+        plain = mk(None)
+        spec = mk({"method": "ngram", "num_spec_tokens": 6, "ngram": 2})
+        _SPEC_ENGINES = (plain, spec)
+    spec.cfg.speculative = {"method": "ngram", "num_spec_tokens": k,
+                            "ngram": ngram}
+
+    def gen(eng):
+        async def go():
+            seq = await eng.add_request(list(prompt), SamplingParams(
+                temperature=0.0, max_tokens=max_tokens, ignore_eos=True))
+            toks = []
+            while True:
+                item = await seq.stream.get()
+                toks.extend(item["token_ids"])
+                if item["finished"]:
+                    return toks
+
+        loop = asyncio.new_event_loop()
+        try:
+            return loop.run_until_complete(go())
+        finally:
+            pending = asyncio.all_tasks(loop)
+            for t in pending:
+                t.cancel()
+            if pending:
+                loop.run_until_complete(
+                    asyncio.gather(*pending, return_exceptions=True))
+            loop.close()
+
+    assert gen(spec) == gen(plain)
