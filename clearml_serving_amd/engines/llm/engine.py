@@ -948,12 +948,25 @@ class LlmEngine:
         logits = self._exec_decode(plan)
         self._sample_and_emit(seqs, logits, sample=plan["sample"])
 
+    def _use_microbatch(self, b: int) -> bool:
+        """TP decode microbatch pipelining (comm/compute overlap): split
+        the decode batch in two and overlap each half's all-reduces with
+        the other half's compute (model.forward_pipelined). OFF by default
+        -- correctness is gloo-verified (tests/helpers/tp_check.py) but the
+        schedule is unmeasured on multi-GPU hardware, and it bypasses the
+        decode hipGraphs (the eager pipeline re-pays launch overhead);
+        first 8-GPU window: measure CMLS_TP_MICROBATCH=1 vs graphs."""
+        return (getattr(self, "tp_size", 1) > 1 and b >= 2
+                and os.environ.get("CMLS_TP_MICROBATCH", "0") == "1")
+
     @torch.inference_mode()
     def _exec_decode(self, plan: Dict[str, Any]) -> torch.Tensor:
+        b = len(plan["tokens"])
+        if self._use_microbatch(b):
+            return self._exec_decode_microbatch(plan)
         if self._use_decode_graphs():
             return self._exec_decode_graph(plan)
         dev = self.device
-        b = len(plan["tokens"])
         tokens = torch.tensor(plan["tokens"], dtype=torch.long, device=dev)
         positions = torch.tensor(plan["positions"], dtype=torch.int32,
                                  device=dev)
@@ -972,6 +985,40 @@ class LlmEngine:
         return self.model(tokens, positions, kv_caches=self.kv_caches,
                           attn_ctx=attn_ctx, last_token_idx=None,
                           gather_logits=False)
+
+    @torch.inference_mode()
+    def _exec_decode_microbatch(self, plan: Dict[str, Any]) -> torch.Tensor:
+        """Split the decode plan into two row-halves and run them through
+        model.forward_pipelined. The split point is derived from the plan
+        (identical on every rank), so all ranks issue the pipeline's
+        collectives in the same order."""
+        dev = self.device
+        b = len(plan["tokens"])
+        mid = b // 2
+        parts = []
+        for lo, hi in ((0, mid), (mid, b)):
+            rows = range(lo, hi)
+            tokens = torch.tensor([plan["tokens"][i] for i in rows],
+                                  dtype=torch.long, device=dev)
+            positions = torch.tensor([plan["positions"][i] for i in rows],
+                                     dtype=torch.int32, device=dev)
+            slot_map = torch.tensor([plan["slots"][i] for i in rows],
+                                    dtype=torch.int32, device=dev)
+            blocks = [plan["blocks"][i] for i in rows]
+            max_blocks = max(len(bl) for bl in blocks)
+            btab = torch.zeros(hi - lo, max_blocks, dtype=torch.int32)
+            for i, bl in enumerate(blocks):
+                btab[i, :len(bl)] = torch.tensor(bl, dtype=torch.int32)
+            attn_ctx = {
+                "mode": "decode",
+                "seq_lens": torch.tensor(
+                    [plan["seq_lens"][i] for i in rows],
+                    dtype=torch.int32, device=dev),
+                "block_table": btab.to(dev),
+                "slot_mapping": slot_map,
+            }
+            parts.append((tokens, positions, attn_ctx))
+        return self.model.forward_pipelined(parts, self.kv_caches)
 
     # -------------------- speculative decoding (ngram) ---------------- #
     # Prompt-lookup speculation (vLLM's "ngram" method): propose the K

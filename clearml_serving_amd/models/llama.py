@@ -81,6 +81,23 @@ class LlamaLayer(nn.Module):
         return mod(x)
 
     def forward(self, x, residual, positions, kv_cache, attn_ctx):
+        attn_out = self.attn_half(x, residual, positions, kv_cache, attn_ctx)
+        if self.tp_size > 1:
+            from ..parallel import tp as tp_mod
+
+            tp_mod.maybe_all_reduce(attn_out)  # row-parallel o_proj
+        mlp_out = self.mlp_half(attn_out, residual)
+        if self.tp_size > 1:
+            from ..parallel import tp as tp_mod
+
+            tp_mod.maybe_all_reduce(mlp_out)  # row-parallel down projection
+        return mlp_out  # residual carries the stream
+
+    def attn_half(self, x, residual, positions, kv_cache, attn_ctx):
+        """norm -> qkv -> rope -> KV write -> attention -> o_proj, WITHOUT
+        the row-parallel all-reduce -- the TP decode microbatch pipeline
+        (LlamaForCausalLM.forward_pipelined) issues that reduce async so it
+        overlaps the other microbatch's compute."""
         cfg = self.cfg
         # fp8 serving path (models/quant.py): activation quantization is
         # FUSED into the producing kernels -- rmsnorm/silu_mul emit fp8
@@ -146,11 +163,13 @@ class LlamaLayer(nn.Module):
             attn_out = self.o_proj.forward_q(c8, cs)
         else:
             attn_out = self._proj(self.o_proj, ctx)
-        if self.tp_size > 1:
-            from ..parallel import tp as tp_mod
+        return attn_out
 
-            tp_mod.maybe_all_reduce(attn_out)  # row-parallel o_proj
-
+    def mlp_half(self, attn_out, residual):
+        """norm -> gate_up -> silu*up -> down, WITHOUT the row-parallel
+        all-reduce (see attn_half)."""
+        cfg = self.cfg
+        fp8 = not isinstance(self.qkv, nn.Linear)
         if fp8:
             x8, xs = ops.rmsnorm_fp8(attn_out, self.mlp_norm, cfg.rms_eps,
                                      residual=residual)
@@ -164,11 +183,7 @@ class LlamaLayer(nn.Module):
             gate_up = self.gate_up(x)
             gate, up = gate_up.split([self.inter, self.inter], dim=-1)
             mlp_out = self._proj(self.down, ops.silu_mul(gate, up))
-        if self.tp_size > 1:
-            from ..parallel import tp as tp_mod
-
-            tp_mod.maybe_all_reduce(mlp_out)  # row-parallel down projection
-        return mlp_out  # residual carries the stream
+        return mlp_out
 
 
 class LlamaForCausalLM(nn.Module):
@@ -222,15 +237,61 @@ class LlamaForCausalLM(nn.Module):
             return x
         logits = self.lm_head(x)
         if self.tp_size > 1 and gather_logits:
-            # lm_head is column-sharded. The serving engine passes
-            # gather_logits=False and samples ON the shards (packed-argmax /
-            # Gumbel-max all-reduce of O(batch) words, parallel/tp.py) --
-            # the full-vocab all-gather (b x 128k x 2B per decode step) is
-            # only for offline full-logits consumers like the TP math tests.
-            from ..parallel import tp as tp_mod
-
-            logits = tp_mod.gather_logits(logits)
+            return self._gathered(logits)
         return logits
+
+    def _gathered(self, logits):
+        # lm_head is column-sharded. The serving engine passes
+        # gather_logits=False and samples ON the shards (packed-argmax /
+        # Gumbel-max all-reduce of O(batch) words, parallel/tp.py) --
+        # the full-vocab all-gather (b x 128k x 2B per decode step) is
+        # only for offline full-logits consumers like the TP math tests.
+        from ..parallel import tp as tp_mod
+
+        return tp_mod.gather_logits(logits)
+
+    def forward_pipelined(self, parts, kv_caches):
+        """TP decode comm/compute overlap: run N microbatches through the
+        layer stack in a 2-stage pipeline -- each half's row-parallel
+        all-reduce is issued async and waited only when ITS next consumer
+        runs, so the collective overlaps the other half's GEMMs (xGMI comm
+        under compute; ROADMAP 'decode microbatch pipelining').
+
+        parts: list of (tokens, positions, attn_ctx); returns concatenated
+        logits rows in part order (gather_logits=False semantics). Issue
+        order per layer is identical on every rank (attn_0, attn_1, mlp_0,
+        mlp_1), so the collectives line up."""
+        from ..parallel import tp as tp_mod
+
+        n = len(parts)
+        xs, residuals = [], []
+        for tokens, positions, _ctx in parts:
+            x = self.embed(tokens.long())
+            residuals.append(torch.zeros_like(x))
+            xs.append(x)
+        works = [None] * n
+        for li, layer in enumerate(self.layers):
+            cache = kv_caches[li] if kv_caches is not None else None
+            attn = [None] * n
+            for j in range(n):
+                if works[j] is not None:
+                    works[j].wait()  # previous mlp reduce of THIS half
+                attn[j] = layer.attn_half(xs[j], residuals[j], parts[j][1],
+                                          cache, parts[j][2])
+                works[j] = tp_mod.all_reduce_async(attn[j])
+            for j in range(n):
+                if works[j] is not None:
+                    works[j].wait()  # attn reduce of this half
+                xs[j] = layer.mlp_half(attn[j], residuals[j])
+                works[j] = tp_mod.all_reduce_async(xs[j])
+        outs = []
+        for j in range(n):
+            if works[j] is not None:
+                works[j].wait()
+            h = ops.rmsnorm(xs[j], self.final_norm, self.cfg.rms_eps,
+                            residual=residuals[j])
+            outs.append(self.lm_head(h))
+        return torch.cat(outs, dim=0)
 
 
 PRESETS = {

@@ -38,6 +38,16 @@ def maybe_all_reduce(x: torch.Tensor, group=None) -> torch.Tensor:
     return x
 
 
+def all_reduce_async(x: torch.Tensor, group=None):
+    """Async row-parallel reduce for the decode microbatch pipeline:
+    returns the Work handle (or None at world<=1). The caller waits RIGHT
+    BEFORE consuming x, so the collective (RCCL's comm stream on GPU)
+    overlaps the other microbatch's compute in between."""
+    if world_size() <= 1:
+        return None
+    return dist.all_reduce(x, group=group, async_op=True)
+
+
 def gather_logits(logits: torch.Tensor, group=None) -> torch.Tensor:
     """All-gather column-sharded lm_head outputs -> full-vocab logits."""
     ws = world_size()

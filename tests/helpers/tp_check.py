@@ -149,12 +149,60 @@ def check_engine_protocol(rank, world):
         eng.run_tp_worker()
 
 
+def check_microbatch_pipeline(rank, world):
+    """CMLS_TP_MICROBATCH=1 (decode comm/compute overlap via async
+    all-reduce pipelining, model.forward_pipelined): greedy generation
+    must be token-identical to the plain TP decode path."""
+    mcfg = _tiny_cfg(world)
+    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
+                          block_size=16, max_model_len=128, device="cpu",
+                          overrides={"heads": mcfg.heads,
+                                     "kv_heads": mcfg.kv_heads})
+
+    def gen(eng):
+        async def go():
+            params = SamplingParams(temperature=0.0, max_tokens=8,
+                                    ignore_eos=True)
+
+            async def one(i):
+                toks = []
+                async for item in eng.generate("mb probe %d" % i, params):
+                    toks.extend(item["token_ids"])
+                return toks
+
+            return await asyncio.gather(*[one(i) for i in range(3)])
+
+        loop = asyncio.new_event_loop()
+        try:
+            return loop.run_until_complete(go())
+        finally:
+            loop.close()
+
+    results = {}
+    for flag in ("0", "1"):
+        os.environ["CMLS_TP_MICROBATCH"] = flag
+        dist.barrier()
+        eng = LlmEngine(cfg)
+        eng.start()
+        if rank == 0:
+            results[flag] = gen(eng)
+            eng.tp_shutdown()
+        else:
+            eng.run_tp_worker()
+    if rank == 0:
+        assert results["1"] == results["0"], (results,)
+        print("TP-MICROBATCH-OK", flush=True)
+    os.environ.pop("CMLS_TP_MICROBATCH", None)
+
+
 def main():
     dist.init_process_group(backend="gloo")
     rank, world = dist.get_rank(), dist.get_world_size()
     check_tp_math(rank, world)
     check_sharded_sampling(rank, world)
     check_engine_protocol(rank, world)
+    dist.barrier()
+    check_microbatch_pipeline(rank, world)
     dist.barrier()
     dist.destroy_process_group()
 

@@ -1218,3 +1218,59 @@ def test_logprobs_match_teacher_forced_reference():
         ref = float(torch.log_softmax(logits[pos].float(), -1)[
             generated[step]])
         assert abs(item["logprob"] - ref) < 5e-3, (step, item["logprob"], ref)
+
+
+def test_decode_microbatch_pipeline_matches_plain():
+    """model.forward_pipelined (TP comm/compute overlap schedule) must be
+    numerically identical to the plain decode forward -- verified here
+    single-process (works are None; pure math identity), world=2 gloo in
+    tests/helpers/tp_check.py."""
+    import os as _os
+
+    eng = tiny_engine()
+
+    # seed some sequences so the KV cache and block tables are real
+    async def seed():
+        params = SamplingParams(temperature=0.0, max_tokens=4,
+                                ignore_eos=True)
+
+        async def one(i):
+            toks = []
+            async for item in eng.generate("mb %d" % i, params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return await asyncio.gather(*[one(i) for i in range(4)])
+
+    run(seed())
+
+    # craft a decode plan over 4 live-looking sequences
+    from clearml_serving_amd.engines.llm.engine import Sequence
+
+    seqs = []
+    for i in range(4):
+        s = Sequence("mb%d" % i, [3 + i, 7, 11 + i], SamplingParams(
+            temperature=0.0, max_tokens=8, ignore_eos=True))
+        s.blocks = eng.allocator.alloc(1)
+        s.output_ids = [5 + i]
+        s.prefilled = len(s.prompt_ids)
+        seqs.append(s)
+    plan = {
+        "mode": "decode",
+        "tokens": [s.output_ids[-1] for s in seqs],
+        "positions": [len(s) - 1 for s in seqs],
+        "slots": [eng._slot(s, len(s) - 1) for s in seqs],
+        "seq_lens": [len(s) for s in seqs],
+        "blocks": [list(s.blocks) for s in seqs],
+    }
+    plain = eng._exec_decode(plan)
+    eng.tp_size = 2  # force the microbatch route (works stay None on CPU
+    _os.environ["CMLS_TP_MICROBATCH"] = "1"  # single-process: world<=1)
+    try:
+        mb = eng._exec_decode(plan)
+    finally:
+        eng.tp_size = 1
+        del _os.environ["CMLS_TP_MICROBATCH"]
+    for s in seqs:
+        eng.allocator.free(s.blocks)
+    torch.testing.assert_close(mb, plain, atol=1e-5, rtol=1e-5)

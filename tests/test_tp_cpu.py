@@ -23,6 +23,7 @@ def test_tp2_math_and_engine_protocol():
     assert out.returncode == 0, out.stdout + "\n" + out.stderr
     assert "TP-MATH-OK" in out.stdout
     assert "TP-ENGINE-OK" in out.stdout
+    assert "TP-MICROBATCH-OK" in out.stdout
 
 
 @pytest.mark.timeout(300)
@@ -46,6 +47,7 @@ def test_tp4_math_and_engine_protocol():
     assert "TP-MATH-OK" in out.stdout
     assert "TP-SAMPLE-OK" in out.stdout
     assert "TP-ENGINE-OK" in out.stdout
+    assert "TP-MICROBATCH-OK" in out.stdout
 
 
 @pytest.mark.timeout(300)
