@@ -36,10 +36,6 @@ def run(coro):
     (4, 32, 8, 128, [1, 17, 200, 1000]),
     (2, 8, 8, 64, [33, 64]),
     (3, 16, 2, 128, [5, 130, 257]),
-    (2, 28, 4, 128, [9, 210]),   # GQ=7: qwen2-7b decode shape
-    (2, 12, 2, 128, [9, 130]),   # GQ=6: qwen2-1.5b
-    (2, 6, 2, 128, [21, 64]),    # GQ=3
-    (2, 10, 2, 128, [21, 64]),   # GQ=5
 ])
 def test_attention_decode_numerics(b, h, hkv, d, seqs):
     torch.manual_seed(0)
@@ -490,3 +486,16 @@ def test_qwen2_qkv_bias_gpu_generation():
             layer.qkv.bias.zero_()
     c = run(go())
     assert c != a
+
+
+# placed at file END deliberately: these GQA groups (3/5/6/7 -- the qwen2
+# shapes) were instantiated after the last GPU window of the round, so an
+# unexpected failure here must not halt (-x) the proven suite above
+@pytest.mark.parametrize("b,h,hkv,d,seqs", [
+    (2, 28, 4, 128, [9, 210]),   # GQ=7: qwen2-7b decode shape
+    (2, 12, 2, 128, [9, 130]),   # GQ=6: qwen2-1.5b
+    (2, 6, 2, 128, [21, 64]),    # GQ=3
+    (2, 10, 2, 128, [21, 64]),   # GQ=5
+])
+def test_attention_decode_numerics_extended_gqa(b, h, hkv, d, seqs):
+    test_attention_decode_numerics(b, h, hkv, d, seqs)
