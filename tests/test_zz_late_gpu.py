@@ -1,0 +1,96 @@
+"""GPU tests added AFTER the round's last GPU window (speculative
+decoding, Qwen2 qkv_bias, extended GQA groups): kept in a file that sorts
+LAST so a surprise on fresh hardware cannot halt (-x) the proven suite."""
+
+import asyncio
+
+import pytest
+import torch
+
+import clearml_serving_amd.ops as ops
+from clearml_serving_amd.engines.llm.engine import (
+    LlmEngine,
+    LlmEngineConfig,
+    SamplingParams,
+)
+from tests.test_llm_gpu import run  # noqa: F401
+from tests.test_llm_gpu import \
+    test_attention_decode_numerics as _decode_numerics  # not re-collected
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+
+
+def test_spec_decode_gpu_matches_plain_greedy():
+    """ngram speculative decoding on the GPU kernel path (multi-token
+    verify through attention_prefill_paged_v2): token-identical to plain
+    greedy decode, and speculation actually fires."""
+    def gen(speculative):
+        torch.manual_seed(11)
+        cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=128,
+                              block_size=16, max_model_len=256, device=DEV,
+                              max_num_seqs=8, speculative=speculative)
+        eng = LlmEngine(cfg)
+        eng.start()
+
+        async def go():
+            prompts = ["abcabcabcabcabc", "the quick brown fox", "zq!7#"]
+            outs = []
+            for p in prompts:
+                toks = []
+                async for item in eng.generate(p, SamplingParams(
+                        temperature=0.0, max_tokens=24, ignore_eos=True)):
+                    toks.extend(item["token_ids"])
+                outs.append(toks)
+            return outs
+
+        return run(go()), eng.stats
+
+    plain, _ = gen(None)
+    spec, stats = gen({"method": "ngram", "num_spec_tokens": 4, "ngram": 2})
+    assert plain == spec
+    assert stats["spec_proposed"] > 0 and stats["spec_accepted"] > 0
+
+
+def test_qwen2_qkv_bias_gpu_generation():
+    """qkv_bias (Qwen2 family) on the GPU path: skinny projections plus the
+    bias add; greedy generation deterministic and bias-sensitive."""
+    torch.manual_seed(12)
+    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=128,
+                          block_size=16, max_model_len=256, device=DEV,
+                          overrides={"qkv_bias": True})
+    eng = LlmEngine(cfg)
+    eng.start()
+    assert eng.model.layers[0].qkv.bias is not None
+
+    async def go():
+        toks = []
+        async for item in eng.generate("qwen gpu probe", SamplingParams(
+                temperature=0.0, max_tokens=12, ignore_eos=True)):
+            toks.extend(item["token_ids"])
+        return toks
+
+    a = run(go())
+    b = run(go())
+    assert len(a) == 12 and a == b
+    with torch.no_grad():
+        for layer in eng.model.layers:
+            layer.qkv.bias.zero_()
+    c = run(go())
+    assert c != a
+
+
+# placed at file END deliberately: these GQA groups (3/5/6/7 -- the qwen2
+# shapes) were instantiated after the last GPU window of the round, so an
+# unexpected failure here must not halt (-x) the proven suite above
+@pytest.mark.parametrize("b,h,hkv,d,seqs", [
+    (2, 28, 4, 128, [9, 210]),   # GQ=7: qwen2-7b decode shape
+    (2, 12, 2, 128, [9, 130]),   # GQ=6: qwen2-1.5b
+    (2, 6, 2, 128, [21, 64]),    # GQ=3
+    (2, 10, 2, 128, [21, 64]),   # GQ=5
+])
+def test_attention_decode_numerics_extended_gqa(b, h, hkv, d, seqs):
+    _decode_numerics(b, h, hkv, d, seqs)
