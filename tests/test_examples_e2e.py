@@ -265,3 +265,72 @@ def test_bert_example_readme_flow(tmp_path):
         assert r.status_code == 200, r.text
         out = r.json()
         assert out["label"] in (0, 1) and len(out["logits"]) == 2
+
+
+@pytest.mark.timeout(240)
+def test_llm_speculative_example_flow(tmp_path):
+    """The readme's speculative-decoding recipe end to end via CLI + HTTP:
+    a card with {'speculative': {...}} serves greedy chat identically to a
+    plain card, and the engine reports speculation activity."""
+    import json as _json
+
+    from clearml_serving_amd.engines.llm.adapter import LlmPreprocessRequest
+
+    LlmPreprocessRequest._engine_singleton = None
+    LlmPreprocessRequest._engines = {}
+    LlmPreprocessRequest._engine_refs = {}
+
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    base = {"arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+            "block_size": 16, "max_model_len": 128, "device": "cpu"}
+    plain_card = tmp_path / "plain.json"
+    plain_card.write_text(_json.dumps(base))
+    spec_card = tmp_path / "spec.json"
+    spec_card.write_text(_json.dumps({
+        **base, "speculative": {"method": "ngram", "num_spec_tokens": 4,
+                                "ngram": 2}}))
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "spec ex"])
+    for name, card in (("plain card", plain_card), ("spec card", spec_card)):
+        run_py(["-m", "clearml_serving_amd", "model", "upload", "--name",
+                name, "--project", "examples", "--path", str(card)])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine", "llm",
+            "--endpoint", "plain_llm", "--name", "plain card", "--project",
+            "examples"])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine", "vllm",
+            "--endpoint", "spec_llm", "--name", "spec card", "--project",
+            "examples"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    try:
+        with TestClient(app) as client:
+            body = {"max_tokens": 24, "temperature": 0.0, "ignore_eos": True,
+                    "messages": [{"role": "user", "content": "abcabcabc"}]}
+            outs = {}
+            for model in ("plain_llm", "spec_llm"):
+                r = client.post("/serve/openai/v1/chat/completions",
+                                json={**body, "model": model})
+                assert r.status_code == 200, r.text
+                outs[model] = r.json()["choices"][0]["message"]["content"]
+            # exactness: speculation must not change greedy output
+            assert outs["plain_llm"] == outs["spec_llm"]
+            # speculation fired on the spec endpoint
+            specs = [e for e in LlmPreprocessRequest._engines.values()
+                     if e.cfg.speculative]
+            assert specs and specs[0].stats["spec_proposed"] > 0
+    finally:
+        LlmPreprocessRequest._engine_singleton = None
+        LlmPreprocessRequest._engines = {}
+        LlmPreprocessRequest._engine_refs = {}
