@@ -105,6 +105,8 @@ class SamplingParams:
     repetition_penalty: float = 1.0      # CTRL-style, prompt+output tokens
     logprobs: Optional[int] = None       # None = off; N = N top_logprobs
                                          # alongside each chosen token
+    min_tokens: int = 0                  # ban eos/stop-token finishes until
+                                         # this many tokens are generated
 
     def has_penalties(self) -> bool:
         return bool(self.presence_penalty or self.frequency_penalty
@@ -150,6 +152,14 @@ class SamplingParams:
         if max_tokens < 1:
             raise ValueError("max_tokens must be >= 1, got {}".format(
                 max_tokens))
+        try:
+            min_tokens = int(body.get("min_tokens", 0) or 0)
+        except (TypeError, ValueError):
+            raise ValueError("'min_tokens' must be an int")
+        if not (0 <= min_tokens <= max_tokens):
+            raise ValueError(
+                "min_tokens must be in [0, max_tokens], got {}".format(
+                    min_tokens))
         if not (-2.0 <= presence <= 2.0):
             raise ValueError("presence_penalty must be in [-2, 2], got {}"
                              .format(presence))
@@ -194,6 +204,7 @@ class SamplingParams:
             frequency_penalty=frequency,
             repetition_penalty=repetition,
             logprobs=lp,
+            min_tokens=min_tokens,
         )
 
 
@@ -470,6 +481,9 @@ class LlmEngine:
             raise ValueError(
                 "logprobs need a full-vocab softmax and are not supported "
                 "under tensor parallelism yet")
+        if params.min_tokens > 0 and getattr(self, "tp_size", 1) > 1:
+            raise ValueError(
+                "min_tokens is not supported under tensor parallelism yet")
         seq = Sequence(uuid.uuid4().hex, prompt_ids, params)
         self.waiting.append(seq)
         self._ensure_loop()
@@ -627,7 +641,8 @@ class LlmEngine:
                 # plain decode (penalties change the argmax per position)
                 spec = [s for s in decoding if s.params.temperature == 0.0
                         and not s.params.has_penalties()
-                        and s.params.logprobs is None]
+                        and s.params.logprobs is None
+                        and s.params.min_tokens <= s.generated]
                 rest = [s for s in decoding if s not in spec]
             else:
                 spec, rest = [], decoding
@@ -1380,10 +1395,14 @@ class LlmEngine:
     def _apply_penalties(self, logits: torch.Tensor,
                          seqs: List[Sequence]) -> torch.Tensor:
         """OpenAI presence/frequency penalties (output tokens) + CTRL-style
-        repetition penalty (prompt+output). Applied per requesting row only;
-        the tensor is cloned first because decode-graph logits are static
-        replay buffers that must never be mutated."""
-        need = [i for i, s in enumerate(seqs) if s.params.has_penalties()]
+        repetition penalty (prompt+output) + min_tokens eos masking (like
+        vLLM: the eos/stop ids are banned from sampling until min_tokens,
+        not post-filtered). Applied per requesting row only; the tensor is
+        cloned first because decode-graph logits are static replay buffers
+        that must never be mutated."""
+        need = [i for i, s in enumerate(seqs)
+                if s.params.has_penalties()
+                or s.params.min_tokens > s.generated]
         if not need:
             return logits
         V = logits.shape[-1]
@@ -1393,6 +1412,14 @@ class LlmEngine:
             s = seqs[i]
             p = s.params
             row = logits[i].float()
+            if p.min_tokens > s.generated:
+                ban = list(p.stop_token_ids)
+                if not p.ignore_eos:
+                    ban.append(self.tokenizer.eos_id)
+                ban = [t for t in ban if 0 <= t < V]
+                if ban:
+                    row[torch.tensor(ban, dtype=torch.long,
+                                     device=dev)] = float("-inf")
             if p.repetition_penalty != 1.0:
                 ctx = [t for t in dict.fromkeys(s.prompt_ids + s.output_ids)
                        if 0 <= t < V]
@@ -1618,9 +1645,12 @@ class LlmEngine:
             *[self._collect(prompt, p) for p in params])
         nprompt = results[0][3]
         ntok = sum(r[2] for r in results)
+        echo = bool(body.get("echo"))
         choices = []
         for i, r in enumerate(results):
-            choice = {"index": i, "text": r[0], "finish_reason": r[1]}
+            choice = {"index": i,
+                      "text": (prompt + r[0]) if echo else r[0],
+                      "finish_reason": r[1]}
             if params[i].logprobs is not None:
                 choice["logprobs"] = self._completion_logprobs(r[4])
             choices.append(choice)

@@ -1274,3 +1274,58 @@ def test_decode_microbatch_pipeline_matches_plain():
     for s in seqs:
         eng.allocator.free(s.blocks)
     torch.testing.assert_close(mb, plain, atol=1e-5, rtol=1e-5)
+
+
+def test_min_tokens_and_echo():
+    """min_tokens bans eos/stop ids from sampling until the floor (vLLM
+    semantics: masked, not post-filtered); echo prefixes the prompt."""
+    eng = tiny_engine()
+
+    # find what greedy emits first, then use it as a stop id with a floor
+    first = run(eng.generate_simple(
+        {"prompt": "mt probe", "max_tokens": 1, "temperature": 0.0,
+         "ignore_eos": True}))
+
+    async def gen(min_tokens):
+        seq = await eng.add_request(
+            eng.tokenizer.encode("mt probe"),
+            SamplingParams(temperature=0.0, max_tokens=12,
+                           min_tokens=min_tokens,
+                           stop_token_ids=[]))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    toks = run(gen(0))
+    stop_id = toks[0]
+
+    async def gen_stop(min_tokens):
+        seq = await eng.add_request(
+            eng.tokenizer.encode("mt probe"),
+            SamplingParams(temperature=0.0, max_tokens=12,
+                           min_tokens=min_tokens,
+                           stop_token_ids=[stop_id]))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    # without the floor the first token stops generation immediately
+    assert len(run(gen_stop(0))) == 1
+    # with the floor, the banned id cannot be sampled before 6 tokens
+    out = run(gen_stop(6))
+    assert len(out) >= 6
+    assert stop_id not in out[:6]
+
+    with pytest.raises(ValueError):
+        SamplingParams.from_request({"min_tokens": 200, "max_tokens": 10})
+
+    resp = run(eng.openai_completions(
+        {"prompt": "hello", "max_tokens": 2, "temperature": 0.0,
+         "ignore_eos": True, "echo": True}, "m"))
+    assert resp["choices"][0]["text"].startswith("hello")
