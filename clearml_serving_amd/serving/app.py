@@ -162,6 +162,14 @@ def create_app(
     @router.post("/openai/{endpoint_type:path}")
     @router.get("/openai/{endpoint_type:path}")
     async def openai_serve_model(endpoint_type: str, request: Request):
+        # reference parity (main.py:207-215): OpenAI routes are JSON-only
+        content_type = (request.headers.get("content-type") or "").lower()
+        media_type = content_type.split(";", 1)[0].strip()
+        if media_type and media_type != "application/json":
+            raise HTTPException(
+                status_code=415,
+                detail="Unsupported Media Type: Only 'application/json' "
+                       "is allowed")
         body = await _read_body(request)
         combined = dict(body) if isinstance(body, dict) else {"body": body}
         combined["request"] = request
@@ -183,6 +191,7 @@ def create_app(
         return _jsonable(out)
 
     @router.post("/{model_id}/{version}")
+    @router.post("/{model_id}/")
     @router.post("/{model_id}")
     async def base_serve_model(
         model_id: str, request: Request, version: Optional[str] = None

@@ -238,3 +238,13 @@ def test_unsupported_vllm_serve_types_explain_themselves(llm_client):
             "model": "test_llm"})
         assert r.status_code == 422, r.text
         assert "audio" in r.json()["detail"]
+
+
+def test_openai_routes_are_json_only(llm_client):
+    """Reference parity (main.py:207-215): non-JSON content types 415 on
+    the OpenAI routes."""
+    r = llm_client.post("/serve/openai/v1/chat/completions",
+                        content=b"model=x",
+                        headers={"Content-Type":
+                                 "application/x-www-form-urlencoded"})
+    assert r.status_code == 415, r.text
