@@ -246,6 +246,7 @@ def create_app(
         return {
             "session": proc.get_id(),
             "revision": proc._last_revision,
+            "endpoint_telemetry": proc.endpoint_telemetry_snapshot(),
             "instances": proc.store.list_instances(proc.get_id(),
                                                    max_age_sec=600),
             "endpoints": endpoints,

@@ -163,6 +163,12 @@ class ModelRequestProcessor:
         self._stats_send_thread: Optional[threading.Thread] = None
         self._stop = False
         self._instance_messages: List[str] = []
+        # per-endpoint request/response counters (reference endpoint
+        # telemetry, model_request_processor.py:162-187 -- there it feeds
+        # the ClearML router UI; here GET /status + Prometheus read it)
+        self._endpoint_telemetry: Dict[str, dict] = {}
+        self._enable_endpoint_telemetry = os.environ.get(
+            "CLEARML_ENABLE_ENDPOINT_TELEMETRY", "1") != "0"
 
     # ------------------------------------------------------------------ #
     # identity / config
@@ -603,6 +609,27 @@ class ModelRequestProcessor:
     # ------------------------------------------------------------------ #
     # request dispatch (hot path)
     # ------------------------------------------------------------------ #
+    def _telemetry_counters(self, url: str) -> Optional[dict]:
+        if not self._enable_endpoint_telemetry:
+            return None
+        t = self._endpoint_telemetry.get(url)
+        if t is None:
+            t = self._endpoint_telemetry.setdefault(
+                url, {"requests": FastWriteCounter(),
+                      "responses": FastWriteCounter()})
+        return t
+
+    def endpoint_telemetry_snapshot(self) -> Dict[str, dict]:
+        """{url: {requests, responses, in_flight}} -- reading advances the
+        lock-free counters equally, so snapshots stay exact."""
+        out = {}
+        for url, t in list(self._endpoint_telemetry.items()):
+            req = t["requests"].value()
+            resp = t["responses"].value()
+            out[url] = {"requests": req, "responses": resp,
+                        "in_flight": max(req - resp, 0)}
+        return out
+
     async def process_request(
         self, base_url: str, version: str, request_body: Any,
         serve_type: str = "process",
@@ -614,11 +641,18 @@ class ModelRequestProcessor:
             await asyncio.sleep(0.5 + random.random())
             self._request_processing_state.inc()
 
+        telemetry = self._telemetry_counters(
+            self._normalize_endpoint_url(base_url, version))
+        if telemetry:
+            telemetry["requests"].inc()
         try:
-            return await self._process_request_inner(
+            result = await self._process_request_inner(
                 base_url=base_url, version=version, request_body=request_body,
                 serve_type=serve_type,
             )
+            if telemetry:
+                telemetry["responses"].inc()
+            return result
         finally:
             self._request_processing_state.dec()
 

@@ -140,3 +140,23 @@ def test_engine_crash_maps_to_500(processor, store, tmp_path):
     with TestClient(app) as c:
         r = c.post("/serve/crash_ep", json={})
         assert r.status_code == 500
+
+
+def test_endpoint_telemetry_counters(client):
+    """Reference endpoint telemetry parity (model_request_processor.py:
+    162-187): per-endpoint request/response counters, surfaced on
+    GET /status."""
+    for _ in range(3):
+        r = client.post("/serve/test_model_sklearn",
+                        json={"x0": 1.0, "x1": 0.0})
+        assert r.status_code == 200
+    r = client.post("/serve/nope_model", json=[[1.0]])
+    assert r.status_code == 404
+    tel = client.get("/status").json()["endpoint_telemetry"]
+    assert tel["test_model_sklearn"]["requests"] == 3
+    assert tel["test_model_sklearn"]["responses"] == 3
+    assert tel["test_model_sklearn"]["in_flight"] == 0
+    # failed request: counted in, never counted out (reference increments
+    # on_response only after success)
+    assert tel["nope_model"]["requests"] == 1
+    assert tel["nope_model"]["responses"] == 0
