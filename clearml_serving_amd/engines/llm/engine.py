@@ -1609,8 +1609,11 @@ class LlmEngine:
         params = self._choice_params(body, n)
         rid = "chatcmpl-" + uuid.uuid4().hex[:24]
         if body.get("stream"):
+            opts = body.get("stream_options") or {}
             return self._sse_stream(prompt, params[0], rid, model_name,
-                                    chat=True)
+                                    chat=True,
+                                    include_usage=bool(
+                                        opts.get("include_usage")))
         results = await asyncio.gather(
             *[self._collect(prompt, p) for p in params])
         nprompt = results[0][3]
@@ -1639,8 +1642,11 @@ class LlmEngine:
         params = self._choice_params(body, n)
         rid = "cmpl-" + uuid.uuid4().hex[:24]
         if body.get("stream"):
+            opts = body.get("stream_options") or {}
             return self._sse_stream(prompt, params[0], rid, model_name,
-                                    chat=False)
+                                    chat=False,
+                                    include_usage=bool(
+                                        opts.get("include_usage")))
         results = await asyncio.gather(
             *[self._collect(prompt, p) for p in params])
         nprompt = results[0][3]
@@ -1797,11 +1803,14 @@ class LlmEngine:
         }
 
     def _sse_stream(self, prompt: str, params: SamplingParams, rid: str,
-                    model_name: str, chat: bool):
+                    model_name: str, chat: bool,
+                    include_usage: bool = False):
         from fastapi.responses import StreamingResponse
 
         async def gen():
+            ntok = 0
             async for item in self.generate(prompt, params):
+                ntok += len(item.get("token_ids", []))
                 if item.get("error"):
                     yield "data: {}\n\n".format(json.dumps(
                         {"error": {"message": str(item["error"]),
@@ -1828,6 +1837,19 @@ class LlmEngine:
                          "created": int(time.time()), "model": model_name,
                          "choices": [choice]}
                 yield "data: {}\n\n".format(json.dumps(chunk))
+            if include_usage:
+                # OpenAI stream_options.include_usage: one final chunk
+                # with empty choices and the usage totals
+                nprompt = len(self.tokenizer.encode(prompt))
+                yield "data: {}\n\n".format(json.dumps({
+                    "id": rid,
+                    "object": ("chat.completion.chunk" if chat
+                               else "text_completion"),
+                    "created": int(time.time()), "model": model_name,
+                    "choices": [],
+                    "usage": {"prompt_tokens": nprompt,
+                              "completion_tokens": ntok,
+                              "total_tokens": nprompt + ntok}}))
             yield "data: [DONE]\n\n"
 
         return StreamingResponse(gen(), media_type="text/event-stream")

@@ -248,3 +248,20 @@ def test_openai_routes_are_json_only(llm_client):
                         headers={"Content-Type":
                                  "application/x-www-form-urlencoded"})
     assert r.status_code == 415, r.text
+
+
+def test_sse_stream_options_include_usage(llm_client):
+    """OpenAI stream_options.include_usage: a final pre-[DONE] chunk with
+    empty choices and the usage totals."""
+    r = llm_client.post("/serve/openai/v1/chat/completions", json={
+        "model": "test_llm", "max_tokens": 5, "temperature": 0.0,
+        "ignore_eos": True, "stream": True,
+        "stream_options": {"include_usage": True},
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 200
+    lines = [ln for ln in r.text.splitlines() if ln.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
+    final = json.loads(lines[-2][len("data: "):])
+    assert final["choices"] == []
+    assert final["usage"]["completion_tokens"] == 5
+    assert final["usage"]["total_tokens"] > 5
