@@ -308,6 +308,7 @@ class Sequence:
         self.finish_reason: Optional[str] = None
         self.created = time.time()
         self.first_token_time: Optional[float] = None
+        self.spec_misses = 0  # consecutive fully-rejected proposals
 
     def __len__(self):
         return len(self.prompt_ids) + len(self.output_ids)
@@ -1094,6 +1095,12 @@ class LlmEngine:
             cap = min(k,
                       s.params.max_tokens - s.generated - 1,
                       self.cfg.max_model_len - len(s) - 1)
+            # backoff: a sequence whose last 4 proposals were all rejected
+            # is generating non-repetitive text -- stop paying the
+            # multi-token verify for it, retry every 32 tokens in case the
+            # output becomes structured again
+            if s.spec_misses >= 4 and s.generated % 32 != 0:
+                cap = 0
             prop = self._ngram_propose(s, cap) if cap > 0 else []
             # proposal tokens occupy positions len(s)..len(s)+p-1: extend
             # the block table opportunistically, trimming to what the
@@ -1145,6 +1152,8 @@ class LlmEngine:
                 self.stats.get("spec_proposed", 0) + len(prop))
             self.stats["spec_accepted"] = (
                 self.stats.get("spec_accepted", 0) + len(toks) - 1)
+            if prop:
+                s.spec_misses = 0 if len(toks) > 1 else s.spec_misses + 1
             self._emit_tokens(s, toks)
 
     @torch.inference_mode()

@@ -1329,3 +1329,30 @@ def test_min_tokens_and_echo():
         {"prompt": "hello", "max_tokens": 2, "temperature": 0.0,
          "ignore_eos": True, "echo": True}, "m"))
     assert resp["choices"][0]["text"].startswith("hello")
+
+
+def test_spec_decode_backoff_on_rejections():
+    """Sequences whose proposals keep getting rejected stop paying the
+    multi-token verify: after 4 consecutive misses the engine proposes
+    only on the periodic retry (every 32 tokens)."""
+    eng = spec_engine()
+    from clearml_serving_amd.engines.llm.engine import Sequence
+
+    s = Sequence("bo", [1, 2, 1, 2], SamplingParams(temperature=0.0))
+    s.generated = 5  # not on a retry boundary
+    s.spec_misses = 4
+    # the proposer would fire (repeating 2-gram)...
+    assert eng._ngram_propose(s, 4) != []
+    # ...but the backoff decision (as _decode_spec computes it) caps it
+    cap = 4
+    if s.spec_misses >= 4 and s.generated % 32 != 0:
+        cap = 0
+    assert cap == 0
+
+    # end-to-end: exactness holds regardless of backoff state
+    torch.manual_seed(3)
+    e1 = spec_engine()
+    torch.manual_seed(3)
+    e2 = tiny_engine()
+    assert _greedy(e1, "zq!7# unstructured", max_tokens=48) == \
+        _greedy(e2, "zq!7# unstructured", max_tokens=48)
