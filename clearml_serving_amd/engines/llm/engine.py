@@ -181,6 +181,15 @@ class SamplingParams:
                 raise ValueError("'logprobs' must be an int or bool")
         if lp is not None and not (0 <= lp <= 20):
             raise ValueError("logprobs must be in [0, 20], got {}".format(lp))
+        rf = body.get("response_format")
+        if rf and isinstance(rf, dict) and rf.get("type") not in (None,
+                                                                  "text"):
+            # grammar-constrained decoding (json_object/json_schema) is not
+            # implemented; refuse instead of silently returning free text
+            raise ValueError(
+                "response_format '{}' is not supported (no guided-decoding "
+                "backend); omit it or use {{'type': 'text'}}".format(
+                    rf.get("type")))
         stop = body.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]

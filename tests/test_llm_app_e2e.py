@@ -265,3 +265,20 @@ def test_sse_stream_options_include_usage(llm_client):
     assert final["choices"] == []
     assert final["usage"]["completion_tokens"] == 5
     assert final["usage"]["total_tokens"] > 5
+
+
+def test_response_format_json_mode_refused(llm_client):
+    """No guided-decoding backend: response_format json_object 422s with
+    a clear message instead of silently returning free text."""
+    r = llm_client.post("/serve/openai/v1/chat/completions", json={
+        "model": "test_llm", "max_tokens": 4,
+        "response_format": {"type": "json_object"},
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 422, r.text
+    assert "response_format" in r.json()["detail"]
+    # explicit text type passes through
+    r = llm_client.post("/serve/openai/v1/chat/completions", json={
+        "model": "test_llm", "max_tokens": 2, "temperature": 0.0,
+        "ignore_eos": True, "response_format": {"type": "text"},
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 200, r.text
