@@ -334,3 +334,42 @@ def test_llm_speculative_example_flow(tmp_path):
         LlmPreprocessRequest._engine_singleton = None
         LlmPreprocessRequest._engines = {}
         LlmPreprocessRequest._engine_refs = {}
+
+
+@pytest.mark.timeout(240)
+def test_ensemble_example_readme_flow(tmp_path):
+    """The ensemble readme flow verbatim: train VotingRegressor, upload,
+    add sklearn endpoint with the example preprocess, query over HTTP."""
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    pkl = tmp_path / "ensemble-vr.pkl"
+    run_py(["examples/ensemble/train_ensemble.py", "--out", str(pkl)])
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "ens ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "upload", "--name",
+            "train model ensemble", "--project", "serving examples",
+            "--path", str(pkl)])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "sklearn", "--endpoint", "test_model_ensemble",
+            "--preprocess", os.path.join(ROOT, "examples/ensemble/preprocess.py"),
+            "--name", "train model ensemble", "--project",
+            "serving examples"])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/test_model_ensemble",
+                        json={"x0": -5.0, "x1": -2.0})
+        assert r.status_code == 200, r.text
+        assert "y" in r.json()
+        assert isinstance(r.json()["y"], (int, float, list))
