@@ -373,3 +373,36 @@ def test_ensemble_example_readme_flow(tmp_path):
         assert r.status_code == 200, r.text
         assert "y" in r.json()
         assert isinstance(r.json()["y"], (int, float, list))
+
+
+@pytest.mark.timeout(240)
+def test_preprocess_template_is_a_valid_preprocess(tmp_path):
+    """The shipped template is directly servable: register it verbatim on
+    a custom endpoint whose load()/process() are inherited defaults."""
+    env = dict(os.environ)
+    env["CLEARML_SERVING_AMD_STORE"] = str(tmp_path / "store")
+    env["PYTHONPATH"] = ROOT
+
+    def run_py(args, cwd=ROOT):
+        out = subprocess.run([sys.executable] + args, cwd=cwd, env=env,
+                             capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stdout + "\n" + out.stderr
+        return out.stdout
+
+    run_py(["-m", "clearml_serving_amd", "create", "--name", "tpl ex"])
+    run_py(["-m", "clearml_serving_amd", "model", "add", "--engine",
+            "custom", "--endpoint", "tpl", "--preprocess",
+            os.path.join(ROOT,
+                         "examples/preprocess_template/"
+                         "preprocess_template.py")])
+    from fastapi.testclient import TestClient
+
+    from clearml_serving_amd.serving.app import create_app
+
+    app = create_app(store_root=str(tmp_path / "store"),
+                     poll_frequency_sec=3600)
+    with TestClient(app) as client:
+        r = client.post("/serve/tpl", json={"anything": 1})
+        # template's process() returns None -> served as null, 200
+        assert r.status_code == 200, r.text
+        assert r.json() is None
