@@ -107,8 +107,13 @@ class ShmRingCore {
   bool push(const void* src_v, uint64_t len) {
     const char* src = static_cast<const char*>(src_v);
     uint64_t need = ring_align8(4 + len);
-    if (need + 8 > cap_) {
-      throw std::runtime_error("record larger than ring capacity");
+    // > cap/2 records can DEADLOCK, not just stall: when the write head
+    // sits so that at_end < need, the wrapped footprint (at_end + need)
+    // can exceed the whole capacity -- no amount of draining frees
+    // enough. Reject loudly; callers raise --ring-mb instead.
+    if (need > cap_ / 2) {
+      throw std::runtime_error(
+          "record larger than half the ring capacity (raise --ring-mb)");
     }
     uint64_t head = hdr_->head.load(std::memory_order_relaxed);
     uint64_t tail = hdr_->tail.load(std::memory_order_acquire);

@@ -82,8 +82,11 @@ class PyShmRing:
     def push(self, data: bytes) -> bool:
         ln = len(data)
         need = (4 + ln + 7) & ~7
-        if need + 8 > self._cap:
-            raise RuntimeError("record larger than ring capacity")
+        # same guard as the C++ ring: a record over cap/2 can deadlock at
+        # an unlucky wrap position (at_end + need > cap forever)
+        if need > self._cap // 2:
+            raise RuntimeError("record larger than half the ring capacity "
+                               "(raise --ring-mb)")
         head, tail = self._head(), self._tail()
         pos = head % self._cap
         at_end = self._cap - pos

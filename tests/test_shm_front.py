@@ -630,3 +630,32 @@ def test_owner_engine_swaps_on_hot_reload(tmp_path):
         loop.close()
         client.close()
         owner.close()
+
+
+def test_oversize_record_rejected_not_deadlocked():
+    """Records over cap/2 can deadlock a wrap position (at_end + need >
+    cap forever): both ring implementations must reject them loudly."""
+    from clearml_serving_amd.serving.shm_transport import (
+        HAVE_NATIVE_RING, PyShmRing, make_ring, unlink_ring)
+
+    name = "/cmls_big_{}".format(os.getpid())
+    ring = make_ring(name, 1 << 16, True)
+    try:
+        with pytest.raises(RuntimeError):
+            ring.push(b"x" * ((1 << 15) + 64))
+        # half-capacity records still flow
+        assert ring.push(b"y" * ((1 << 14)))
+        assert len(ring.drain(4)) == 1
+    finally:
+        ring.close()
+        unlink_ring(name)
+
+    # the pure-python ring enforces the same bound
+    name2 = "/cmls_big_py_{}".format(os.getpid())
+    ring2 = PyShmRing(name2, 1 << 16, True)
+    try:
+        with pytest.raises(RuntimeError):
+            ring2.push(b"x" * ((1 << 15) + 64))
+    finally:
+        ring2.close()
+        PyShmRing.unlink(name2)
