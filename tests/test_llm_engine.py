@@ -1356,3 +1356,18 @@ def test_spec_decode_backoff_on_rejections():
     e2 = tiny_engine()
     assert _greedy(e1, "zq!7# unstructured", max_tokens=48) == \
         _greedy(e2, "zq!7# unstructured", max_tokens=48)
+
+
+def test_spec_decode_with_fp8_kv_cache():
+    """Feature interaction: ngram speculation over the fp8 KV cache --
+    the multi-token verify reads quantized pages; output must equal the
+    plain fp8-KV decode path exactly (same quantization noise)."""
+    torch.manual_seed(9)
+    plain = tiny_engine(kv_dtype="fp8")
+    torch.manual_seed(9)
+    spec = tiny_engine(kv_dtype="fp8",
+                       speculative={"method": "ngram",
+                                    "num_spec_tokens": 4, "ngram": 2})
+    for p in ("abcabcabcabc", "mixed 123 text"):
+        assert _greedy(plain, p, 20) == _greedy(spec, p, 20)
+    assert spec.stats["spec_proposed"] > 0
