@@ -1670,13 +1670,24 @@ class LlmEngine:
         nprompt = results[0][3]
         ntok = sum(r[2] for r in results)
         echo = bool(body.get("echo"))
+        prompt_lps: List[dict] = []
+        if echo and params[0].logprobs is not None:
+            ids = self.tokenizer.encode(prompt)
+            if len(ids) > 1:
+                prompt_lps = await asyncio.to_thread(
+                    self._prompt_logprobs_sync, ids, params[0].logprobs)
+            if ids:
+                # the first prompt token has no predecessor: null logprob
+                prompt_lps = [{"token_ids": [ids[0]], "logprob": None,
+                               "top_logprobs": {}}] + prompt_lps
         choices = []
         for i, r in enumerate(results):
             choice = {"index": i,
                       "text": (prompt + r[0]) if echo else r[0],
                       "finish_reason": r[1]}
             if params[i].logprobs is not None:
-                choice["logprobs"] = self._completion_logprobs(r[4])
+                choice["logprobs"] = self._completion_logprobs(
+                    prompt_lps + r[4])
             choices.append(choice)
         return {
             "id": rid, "object": "text_completion", "created": int(time.time()),
@@ -1783,6 +1794,41 @@ class LlmEngine:
         text = self._truncate_at_stop(
             self.tokenizer.decode(tokens), params, reason)
         return text, reason, len(tokens), len(ids), lps
+
+    def _prompt_logprobs_sync(self, ids: List[int], k: int) -> List[dict]:
+        """Teacher-forced logprobs of the prompt tokens themselves (the
+        completions echo+logprobs contract): row i-1's distribution scores
+        token i; the first token has no predecessor (None upstream).
+        Runs under the exec lock like the embed path."""
+        with self._exec_lock:
+            with torch.inference_mode():
+                dev = self.device
+                n = len(ids)
+                tokens = torch.tensor(ids, dtype=torch.long, device=dev)
+                positions = torch.arange(n, dtype=torch.int32, device=dev)
+                attn_ctx = {
+                    "mode": "prefill", "batch": 1, "seq": n,
+                    "seq_lens": torch.tensor([n], dtype=torch.int32,
+                                             device=dev),
+                    "slot_mapping": torch.full((n,), -1, dtype=torch.int32,
+                                               device=dev),
+                }
+                logits = self.model(tokens, positions, kv_caches=None,
+                                    attn_ctx=attn_ctx)
+                lsm = torch.log_softmax(logits.float(), dim=-1)
+        out = []
+        for i in range(1, n):
+            row = lsm[i - 1]
+            e = {"token_ids": [ids[i]], "logprob": float(row[ids[i]])}
+            if k:
+                topv, topi = row.topk(k)
+                e["top_logprobs"] = {
+                    int(t): float(v)
+                    for v, t in zip(topv.tolist(), topi.tolist())}
+            else:
+                e["top_logprobs"] = {}
+            out.append(e)
+        return out
 
     def _lp_entry(self, tok_id: int, logprob: float,
                   top: Dict[int, float]) -> dict:

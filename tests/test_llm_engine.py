@@ -1371,3 +1371,36 @@ def test_spec_decode_with_fp8_kv_cache():
     for p in ("abcabcabcabc", "mixed 123 text"):
         assert _greedy(plain, p, 20) == _greedy(spec, p, 20)
     assert spec.stats["spec_proposed"] > 0
+
+
+def test_echo_with_prompt_logprobs():
+    """completions echo+logprobs: the logprobs arrays cover prompt AND
+    completion tokens; the first prompt token's logprob is null; prompt
+    entries match a teacher-forced log_softmax."""
+    eng = tiny_engine()
+    prompt = "prompt lp"
+    resp = run(eng.openai_completions(
+        {"prompt": prompt, "max_tokens": 3, "temperature": 0.0,
+         "ignore_eos": True, "echo": True, "logprobs": 2}, "m"))
+    choice = resp["choices"][0]
+    assert choice["text"].startswith(prompt)
+    lp = choice["logprobs"]
+    ids = eng.tokenizer.encode(prompt)
+    assert len(lp["tokens"]) == len(ids) + 3
+    assert lp["token_logprobs"][0] is None
+    assert all(v is not None for v in lp["token_logprobs"][1:])
+
+    # teacher-forced reference over the prompt
+    n = len(ids)
+    tokens = torch.tensor(ids, dtype=torch.long)
+    positions = torch.arange(n, dtype=torch.int32)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": n,
+                "seq_lens": torch.tensor([n], dtype=torch.int32),
+                "slot_mapping": torch.full((n,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        logits = eng.model(tokens, positions, kv_caches=None,
+                           attn_ctx=attn_ctx)
+    lsm = torch.log_softmax(logits.float(), -1)
+    for i in range(1, n):
+        assert abs(lp["token_logprobs"][i] - float(lsm[i - 1][ids[i]])) \
+            < 5e-3, i
