@@ -270,8 +270,7 @@ class LlmEngineConfig:
             elif model_path.endswith(".json"):
                 with open(model_path) as f:
                     card = json.load(f)
-        card.pop("arch", None)
-        for key in ("preset", "dtype", "block_size", "max_num_seqs",
+        for key in ("arch", "preset", "dtype", "block_size", "max_num_seqs",
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "quantization",
                     "kv_dtype", "weights", "device", "decode_graphs",
@@ -376,9 +375,6 @@ class LlmEngine:
         if self._started:
             return
         cfg = self.cfg
-        mcfg = LlamaConfig(**{**PRESETS[cfg.preset].__dict__,
-                              **cfg.overrides})
-        self.model_config = mcfg
         # TP over RCCL/xGMI: one process per GPU (torchrun); every rank
         # builds its shard, rank 0 owns scheduling + the HTTP front
         from ...parallel import tp as tp_mod
@@ -386,8 +382,29 @@ class LlmEngine:
         self.tp_size = tp_mod.world_size()
         self.tp_rank = tp_mod.rank()
         torch.manual_seed(1234)  # identical replicated params across ranks
-        model = LlamaForCausalLM(mcfg, tp_rank=self.tp_rank,
-                                 tp_size=self.tp_size)
+        gpt2 = (str(cfg.arch).lower() == "gpt2"
+                or str(cfg.preset).startswith("gpt2"))
+        if gpt2:
+            # GPT-2 family: same engine-facing forward interface as llama
+            # (paged kv_caches + attn_ctx modes), so scheduling, chunked
+            # prefill, decode graphs and speculation serve it unchanged
+            from ...models.gpt2 import PRESETS as GPT2_PRESETS
+            from ...models.gpt2 import GPT2Config, GPT2ForCausalLM
+
+            if self.tp_size > 1:
+                raise ValueError(
+                    "tensor parallelism is implemented for the llama "
+                    "family only (arch=gpt2 serves at tp=1)")
+            mcfg = GPT2Config(**{**GPT2_PRESETS[cfg.preset].__dict__,
+                                 **cfg.overrides})
+            self.model_config = mcfg
+            model = GPT2ForCausalLM(mcfg)
+        else:
+            mcfg = LlamaConfig(**{**PRESETS[cfg.preset].__dict__,
+                                  **cfg.overrides})
+            self.model_config = mcfg
+            model = LlamaForCausalLM(mcfg, tp_rank=self.tp_rank,
+                                     tp_size=self.tp_size)
         if cfg.weights:
             if self.tp_size > 1:
                 # full checkpoint -> this rank's shard
@@ -410,6 +427,10 @@ class LlmEngine:
             from ...models.quant import quantize_llama_fp8
 
             n = quantize_llama_fp8(self.model)
+            if n == 0:
+                raise ValueError(
+                    "fp8 quantization covers the llama-family projections "
+                    "only (arch '{}' has none)".format(cfg.arch))
             print("[llm] fp8-quantized {} projection layers".format(n))
 
         self.tokenizer = (HfTokenizer(cfg.tokenizer_path)

@@ -90,3 +90,38 @@ def convert_hf_llama(hf: Dict[str, torch.Tensor],
 
 # Qwen2 checkpoints use llama key names + QKV biases (handled above)
 convert_hf_qwen2 = convert_hf_llama
+
+
+def convert_hf_gpt2(hf: Dict[str, torch.Tensor],
+                    num_layers: int) -> Dict[str, torch.Tensor]:
+    """transformers GPT2LMHeadModel -> models.gpt2.GPT2ForCausalLM.
+
+    HF GPT-2 stores projections as Conv1D ([in, out]); nn.Linear wants
+    [out, in], so every projection weight transposes."""
+
+    def g(key):
+        return hf.get("transformer." + key, hf.get(key))
+
+    out = {
+        "embed.weight": g("wte.weight"),
+        "wpe.weight": g("wpe.weight"),
+        "lnf_w": g("ln_f.weight"),
+        "lnf_b": g("ln_f.bias"),
+        "lm_head.weight": hf.get("lm_head.weight", g("wte.weight")),
+    }
+    for i in range(num_layers):
+        p = "h.{}.".format(i)
+        o = "layers.{}.".format(i)
+        out[o + "ln1_w"] = g(p + "ln_1.weight")
+        out[o + "ln1_b"] = g(p + "ln_1.bias")
+        out[o + "attn_qkv.weight"] = g(p + "attn.c_attn.weight").t().contiguous()
+        out[o + "attn_qkv.bias"] = g(p + "attn.c_attn.bias")
+        out[o + "attn_out.weight"] = g(p + "attn.c_proj.weight").t().contiguous()
+        out[o + "attn_out.bias"] = g(p + "attn.c_proj.bias")
+        out[o + "ln2_w"] = g(p + "ln_2.weight")
+        out[o + "ln2_b"] = g(p + "ln_2.bias")
+        out[o + "fc_in.weight"] = g(p + "mlp.c_fc.weight").t().contiguous()
+        out[o + "fc_in_bias"] = g(p + "mlp.c_fc.bias")
+        out[o + "fc_out.weight"] = g(p + "mlp.c_proj.weight").t().contiguous()
+        out[o + "fc_out.bias"] = g(p + "mlp.c_proj.bias")
+    return out

@@ -160,3 +160,37 @@ def test_qwen2_tp_shard_roundtrip():
             vs.append(s[hpr * hd + kvpr * hd:])
         rebuilt = torch.cat(qs + ks + vs, dim=0)
         torch.testing.assert_close(rebuilt, state[key])
+
+
+def test_gpt2_matches_transformers():
+    """GPT-2 through the native pre-LN decoder + converter (HF Conv1D
+    weights transpose): logits must match transformers.GPT2LMHeadModel."""
+    from transformers import GPT2Config as HfGPT2Config
+    from transformers import GPT2LMHeadModel
+
+    from clearml_serving_amd.models.convert import convert_hf_gpt2
+    from clearml_serving_amd.models.gpt2 import (GPT2Config,
+                                                 GPT2ForCausalLM)
+
+    torch.manual_seed(6)
+    hf_cfg = HfGPT2Config(
+        vocab_size=160, n_embd=64, n_layer=2, n_head=4, n_inner=96,
+        n_positions=64, resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0,
+        attn_implementation="eager")
+    hf = GPT2LMHeadModel(hf_cfg).eval()
+
+    cfg = GPT2Config(vocab_size=160, hidden=64, layers=2, heads=4,
+                     intermediate=96, max_position=64)
+    ours = GPT2ForCausalLM(cfg).eval()
+    ours.load_state_dict(convert_hf_gpt2(hf.state_dict(), num_layers=2))
+
+    t = 12
+    ids = torch.randint(0, 160, (t,))
+    positions = torch.arange(t, dtype=torch.int32)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": t,
+                "seq_lens": torch.tensor([t], dtype=torch.int32),
+                "slot_mapping": torch.full((t,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        ref = hf(input_ids=ids[None]).logits[0]
+        got = ours(ids, positions, kv_caches=None, attn_ctx=attn_ctx)
+    torch.testing.assert_close(got, ref, atol=3e-4, rtol=3e-4)
