@@ -94,3 +94,25 @@ def test_qwen2_qkv_bias_gpu_generation():
 ])
 def test_attention_decode_numerics_extended_gqa(b, h, hkv, d, seqs):
     _decode_numerics(b, h, hkv, d, seqs)
+
+
+def test_gpt2_engine_gpu_generation():
+    """GPT-2 through the engine on the GPU kernel path (layernorm /
+    bias_gelu / attention / paged decode): deterministic greedy."""
+    torch.manual_seed(13)
+    cfg = LlmEngineConfig(preset="gpt2-tiny", num_kv_blocks=128,
+                          block_size=16, max_model_len=256, device=DEV)
+    eng = LlmEngine(cfg)
+    eng.start()
+
+    async def go():
+        toks = []
+        async for item in eng.generate("gpt2 gpu probe", SamplingParams(
+                temperature=0.0, max_tokens=12, ignore_eos=True)):
+            toks.extend(item["token_ids"])
+        return toks
+
+    a = run(go())
+    b = run(go())
+    assert len(a) == 12 and a == b
+    assert eng.allocator.available == eng.allocator.num_blocks
