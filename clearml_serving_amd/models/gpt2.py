@@ -164,7 +164,9 @@ PRESETS = {
                               intermediate=4096),
     "gpt2-large": GPT2Config(hidden=1280, layers=36, heads=20,
                              intermediate=5120),
-    "gpt2-tiny": GPT2Config(vocab_size=512, hidden=128, layers=2, heads=4,
+    # head_dim 64: the GPU attention kernels support D in {64, 128}
+    # (every real GPT-2 size is 64)
+    "gpt2-tiny": GPT2Config(vocab_size=512, hidden=128, layers=2, heads=2,
                             intermediate=256, max_position=512),
 }
 
