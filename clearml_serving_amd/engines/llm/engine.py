@@ -398,6 +398,13 @@ class LlmEngine:
             mcfg = GPT2Config(**{**GPT2_PRESETS[cfg.preset].__dict__,
                                  **cfg.overrides})
             self.model_config = mcfg
+            if cfg.max_model_len > mcfg.max_position:
+                # learned absolute positions END at max_position: serving
+                # beyond it would silently clamp position ids
+                print("[llm] gpt2 max_model_len capped to max_position "
+                      "({} -> {})".format(cfg.max_model_len,
+                                          mcfg.max_position))
+                cfg.max_model_len = mcfg.max_position
             model = GPT2ForCausalLM(mcfg)
         else:
             mcfg = LlamaConfig(**{**PRESETS[cfg.preset].__dict__,
