@@ -65,7 +65,12 @@ def load_weights(model: torch.nn.Module, path: str) -> None:
         state = load_file(path)
     else:
         state = torch.load(path, map_location="cpu", weights_only=True)
-    model.load_state_dict(state)
+    # raw HuggingFace checkpoints convert on the fly (key-layout
+    # detection; native layouts pass through) -- a hub download drops in
+    # without a manual conversion step
+    from .convert import convert_hf_auto
+
+    model.load_state_dict(convert_hf_auto(state))
 
 
 from . import bert, llama, resnet  # noqa: E402,F401  (register architectures)

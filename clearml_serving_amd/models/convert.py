@@ -125,3 +125,32 @@ def convert_hf_gpt2(hf: Dict[str, torch.Tensor],
         out[o + "fc_out.weight"] = g(p + "mlp.c_proj.weight").t().contiguous()
         out[o + "fc_out.bias"] = g(p + "mlp.c_proj.bias")
     return out
+
+
+def convert_hf_auto(hf: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    """Detect a raw HuggingFace state dict by its key layout and convert it
+    to the matching native layout (llama/qwen2, gpt2, or bert). Native-
+    layout dicts pass through unchanged -- callers can always route
+    through this."""
+    def n_layers(prefix, probe):
+        i = 0
+        while any(k.startswith("{}{}{}".format(prefix, i, probe))
+                  for k in hf):
+            i += 1
+        return i
+
+    keys = hf.keys()
+    if any(k.startswith("model.embed_tokens.") for k in keys):
+        return convert_hf_llama(hf, n_layers("model.layers.", "."))
+    if any("wte.weight" in k for k in keys) and \
+            any(".attn.c_attn." in k for k in keys):
+        pref = ("transformer.h."
+                if any(k.startswith("transformer.h.") for k in keys)
+                else "h.")
+        return convert_hf_gpt2(hf, n_layers(pref, "."))
+    if any("embeddings.word_embeddings." in k for k in keys):
+        pref = ("bert.encoder.layer."
+                if any(k.startswith("bert.encoder.layer.") for k in keys)
+                else "encoder.layer.")
+        return convert_hf_bert(hf, n_layers(pref, "."))
+    return hf  # already native

@@ -194,3 +194,61 @@ def test_gpt2_matches_transformers():
         ref = hf(input_ids=ids[None]).logits[0]
         got = ours(ids, positions, kv_caches=None, attn_ctx=attn_ctx)
     torch.testing.assert_close(got, ref, atol=3e-4, rtol=3e-4)
+
+
+def test_convert_hf_auto_detection(tmp_path):
+    """convert_hf_auto detects llama / qwen2 / gpt2 / bert key layouts
+    (and passes native dicts through), so raw HF checkpoints load via
+    models.load_weights without a manual conversion step."""
+    from transformers import GPT2Config as HfGPT2Config
+    from transformers import GPT2LMHeadModel
+    from transformers import LlamaConfig as HfLlamaConfig
+    from transformers import LlamaForCausalLM as HfLlama
+
+    from clearml_serving_amd.models import load_weights
+    from clearml_serving_amd.models.convert import convert_hf_auto
+    from clearml_serving_amd.models.gpt2 import (GPT2Config,
+                                                 GPT2ForCausalLM)
+
+    torch.manual_seed(8)
+    hf_l = HfLlama(HfLlamaConfig(
+        vocab_size=96, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=32, tie_word_embeddings=False,
+        attn_implementation="eager")).eval()
+    ours_l = LlamaForCausalLM(LlamaConfig(
+        vocab_size=96, hidden=32, layers=2, heads=4, kv_heads=2,
+        intermediate=48, max_position=32, rope_theta=10000.0,
+        rms_eps=1e-6)).eval()
+    ours_l.load_state_dict(convert_hf_auto(hf_l.state_dict()))
+
+    hf_g = GPT2LMHeadModel(HfGPT2Config(
+        vocab_size=96, n_embd=32, n_layer=2, n_head=4, n_inner=48,
+        n_positions=32, resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0,
+        attn_implementation="eager")).eval()
+    ours_g = GPT2ForCausalLM(GPT2Config(
+        vocab_size=96, hidden=32, layers=2, heads=4, intermediate=48,
+        max_position=32)).eval()
+
+    # end to end through load_weights on a saved HF state dict
+    pt = tmp_path / "hf_gpt2.pt"
+    torch.save(hf_g.state_dict(), str(pt))
+    load_weights(ours_g, str(pt))
+
+    t = 8
+    ids = torch.randint(0, 96, (t,))
+    positions = torch.arange(t, dtype=torch.int32)
+    attn_ctx = {"mode": "prefill", "batch": 1, "seq": t,
+                "seq_lens": torch.tensor([t], dtype=torch.int32),
+                "slot_mapping": torch.full((t,), -1, dtype=torch.int32)}
+    with torch.inference_mode():
+        torch.testing.assert_close(
+            ours_g(ids, positions, kv_caches=None, attn_ctx=attn_ctx),
+            hf_g(input_ids=ids[None]).logits[0], atol=3e-4, rtol=3e-4)
+        torch.testing.assert_close(
+            ours_l(ids, positions, kv_caches=None, attn_ctx=attn_ctx),
+            hf_l(input_ids=ids[None]).logits[0], atol=3e-4, rtol=3e-4)
+
+    # native dict passes through unchanged
+    native = ours_g.state_dict()
+    assert set(convert_hf_auto(native).keys()) == set(native.keys())
