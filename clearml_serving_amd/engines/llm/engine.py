@@ -421,8 +421,11 @@ class LlmEngine:
                         if cfg.weights.endswith(".safetensors")
                         else torch.load(cfg.weights, map_location="cpu",
                                         weights_only=True))
+                from ...models.convert import convert_hf_auto
+
                 model.load_state_dict(tp_mod.shard_llama_weights(
-                    full, mcfg, self.tp_rank, self.tp_size))
+                    convert_hf_auto(full), mcfg, self.tp_rank,
+                    self.tp_size))
             else:
                 from ...models import load_weights
 
