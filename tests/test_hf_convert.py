@@ -252,3 +252,65 @@ def test_convert_hf_auto_detection(tmp_path):
     # native dict passes through unchanged
     native = ours_g.state_dict()
     assert set(convert_hf_auto(native).keys()) == set(native.keys())
+
+
+def test_gpt2_hf_checkpoint_served_end_to_end(tmp_path):
+    """A saved HF GPT-2 checkpoint (raw key layout) drops into an engine
+    model card: weights auto-convert at load, and greedy continuation
+    matches transformers.generate token for token."""
+    import asyncio
+    import json
+
+    from safetensors.torch import save_file
+    from transformers import GPT2Config as HfGPT2Config
+    from transformers import GPT2LMHeadModel
+
+    from clearml_serving_amd.engines.llm.engine import (LlmEngine,
+                                                        LlmEngineConfig,
+                                                        SamplingParams)
+
+    torch.manual_seed(10)
+    hf = GPT2LMHeadModel(HfGPT2Config(
+        vocab_size=128, n_embd=64, n_layer=2, n_head=2, n_inner=128,
+        n_positions=128, resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0,
+        attn_implementation="eager")).eval()
+
+    mdir = tmp_path / "gpt2_ckpt"
+    mdir.mkdir()
+    # tied lm_head: save_file refuses shared tensors -- materialize a copy
+    sd = {k: v.clone() for k, v in hf.state_dict().items()}
+    save_file(sd, str(mdir / "model.safetensors"))
+    (mdir / "model_card.json").write_text(json.dumps({
+        "arch": "gpt2", "preset": "gpt2-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 128, "device": "cpu",
+        "overrides": {"vocab_size": 128, "hidden": 64, "layers": 2,
+                      "heads": 2, "intermediate": 128,
+                      "max_position": 128}}))
+
+    cfg = LlmEngineConfig.from_aux(str(mdir), {})
+    assert cfg.arch == "gpt2" and cfg.weights
+    eng = LlmEngine(cfg)
+    eng.start()
+
+    prompt_ids = [5, 17, 31, 44]
+
+    async def gen():
+        seq = await eng.add_request(list(prompt_ids), SamplingParams(
+            temperature=0.0, max_tokens=8, ignore_eos=True))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    loop = asyncio.new_event_loop()
+    try:
+        got = loop.run_until_complete(gen())
+    finally:
+        loop.close()
+
+    with torch.inference_mode():
+        ref = hf.generate(torch.tensor([prompt_ids]), max_new_tokens=8,
+                          do_sample=False, pad_token_id=0)[0].tolist()
+    assert got == ref[len(prompt_ids):], (got, ref)
