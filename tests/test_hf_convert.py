@@ -314,3 +314,66 @@ def test_gpt2_hf_checkpoint_served_end_to_end(tmp_path):
         ref = hf.generate(torch.tensor([prompt_ids]), max_new_tokens=8,
                           do_sample=False, pad_token_id=0)[0].tolist()
     assert got == ref[len(prompt_ids):], (got, ref)
+
+
+def test_qwen2_hf_checkpoint_served_end_to_end(tmp_path):
+    """A saved HF Qwen2 checkpoint (QKV biases) through the engine:
+    auto-converted at load, greedy continuation == transformers.generate."""
+    import asyncio
+    import json
+
+    from safetensors.torch import save_file
+    from transformers import Qwen2Config as HfQwen2Config
+    from transformers import Qwen2ForCausalLM
+
+    from clearml_serving_amd.engines.llm.engine import (LlmEngine,
+                                                        LlmEngineConfig,
+                                                        SamplingParams)
+
+    torch.manual_seed(14)
+    hf = Qwen2ForCausalLM(HfQwen2Config(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attention_dropout=0.0, tie_word_embeddings=False, use_cache=False,
+        attn_implementation="eager")).eval()
+
+    mdir = tmp_path / "qwen2_ckpt"
+    mdir.mkdir()
+    save_file({k: v.clone() for k, v in hf.state_dict().items()},
+              str(mdir / "model.safetensors"))
+    (mdir / "model_card.json").write_text(json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 128, "device": "cpu",
+        "overrides": {"vocab_size": 128, "hidden": 64, "layers": 2,
+                      "heads": 4, "kv_heads": 2, "intermediate": 96,
+                      "rope_theta": 10000.0, "rms_eps": 1e-6,
+                      "max_position": 128, "qkv_bias": True}}))
+
+    cfg = LlmEngineConfig.from_aux(str(mdir), {})
+    eng = LlmEngine(cfg)
+    eng.start()
+    assert eng.model.layers[0].qkv.bias is not None
+
+    prompt_ids = [7, 21, 42, 63, 11]
+
+    async def gen():
+        seq = await eng.add_request(list(prompt_ids), SamplingParams(
+            temperature=0.0, max_tokens=8, ignore_eos=True))
+        toks = []
+        while True:
+            item = await seq.stream.get()
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return toks
+
+    loop = asyncio.new_event_loop()
+    try:
+        got = loop.run_until_complete(gen())
+    finally:
+        loop.close()
+
+    with torch.inference_mode():
+        ref = hf.generate(torch.tensor([prompt_ids]), max_new_tokens=8,
+                          do_sample=False, pad_token_id=0)[0].tolist()
+    assert got == ref[len(prompt_ids):], (got, ref)
