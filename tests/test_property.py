@@ -131,3 +131,51 @@ def test_spec_decode_exactness_property(prompt, k, ngram, max_tokens):
             loop.close()
 
     assert gen(spec) == gen(plain)
+
+
+# --------------------------------------------------------------------- #
+# dynamic batcher: response integrity as a PROPERTY over random streams
+# --------------------------------------------------------------------- #
+@settings(max_examples=10, deadline=None)
+@given(sizes=st.lists(st.integers(min_value=1, max_value=9),
+                      min_size=1, max_size=40),
+       feat=st.integers(min_value=1, max_value=6))
+def test_batcher_every_response_matches_direct_call(sizes, feat):
+    """Whatever the arrival pattern (random burst sizes, padding, bucket
+    boundaries), each request's batched response must equal model(x) run
+    directly on its own input -- no cross-request leakage, no pad rows in
+    responses."""
+    import asyncio
+
+    import torch
+
+    from clearml_serving_amd.serving.batcher import DynamicBatcher
+
+    def model(x):
+        # row-wise, input-dependent (leakage across rows would show)
+        return x * 2.0 + x.sum(dim=-1, keepdim=True)
+
+    b = DynamicBatcher(model_fn=model, device="cpu", max_batch_size=8,
+                       max_queue_delay_us=500, use_graphs=False)
+
+    reqs = []
+    for burst in sizes:
+        for _ in range(burst):
+            reqs.append(torch.randn(feat))
+
+    async def main():
+        outs = await asyncio.gather(*[b.submit(r) for r in reqs])
+        return outs
+
+    loop = asyncio.new_event_loop()
+    try:
+        outs = loop.run_until_complete(main())
+    finally:
+        for t in asyncio.all_tasks(loop):
+            t.cancel()
+        loop.run_until_complete(asyncio.sleep(0))
+        loop.close()
+
+    for r, o in zip(reqs, outs):
+        want = model(r[None])[0]
+        torch.testing.assert_close(o, want, atol=1e-6, rtol=1e-6)
