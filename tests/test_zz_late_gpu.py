@@ -116,3 +116,33 @@ def test_gpt2_engine_gpu_generation():
     b = run(go())
     assert len(a) == 12 and a == b
     assert eng.allocator.available == eng.allocator.num_blocks
+
+
+def test_attention_prefill_paged_tiny_chunks():
+    """Spec decode feeds the paged-prefill kernel 1..K+1-row chunks;
+    exercise the q_len=1 and mixed tiny-q edge directly vs the CPU
+    reference."""
+    torch.manual_seed(21)
+    b, h, hkv, d, bs = 3, 8, 2, 128, 16
+    hist = [40, 7, 0]
+    chunk = [1, 3, 5]
+    sq = max(chunk)
+    nb = 12
+    k_cache = torch.randn(nb, hkv, bs, d, device=DEV, dtype=torch.bfloat16)
+    v_cache = torch.randn_like(k_cache)
+    block_table = torch.tensor([[0, 3, 6, 9], [1, 4, 7, 10], [2, 5, 8, 11]],
+                               dtype=torch.int32, device=DEV)
+    kv_lens = torch.tensor([hist[i] + chunk[i] for i in range(b)],
+                           dtype=torch.int32, device=DEV)
+    q_lens = torch.tensor(chunk, dtype=torch.int32, device=DEV)
+    q = torch.randn(b, sq, h, d, device=DEV, dtype=torch.bfloat16)
+
+    got = ops.attention_prefill_paged(q, k_cache, v_cache, block_table,
+                                      kv_lens, q_lens)
+    ref = ops.attention_prefill_paged(
+        q.float().cpu(), k_cache.float().cpu(), v_cache.float().cpu(),
+        block_table.cpu(), kv_lens.cpu(), q_lens.cpu())
+    for i in range(b):
+        torch.testing.assert_close(got[i, :chunk[i]].float().cpu(),
+                                   ref[i, :chunk[i]].float(),
+                                   atol=3e-2, rtol=3e-2)
