@@ -352,6 +352,61 @@ class ModelRequestProcessor:
     def list_endpoint_logging(self) -> Dict[str, EndpointMetricLogging]:
         return dict(self._metric_logging)
 
+    # reference API parity (model_request_processor.py:592-599): same
+    # content as list_endpoint_logging against the local store
+    def list_metric_logging(self) -> Dict[str, EndpointMetricLogging]:
+        return dict(self._metric_logging)
+
+    def get_configuration(self) -> Dict[str, Any]:
+        """Session parameters (reference :371-372: the General/* section)."""
+        return self._store.get_params(self._session_id)
+
+    def get_version(self) -> str:
+        """Serving package version recorded on the session at create time
+        (reference :762-770)."""
+        return self._store.get_params(self._session_id).get(
+            "serving_version") or "1.0.0"
+
+    def reload(self) -> None:
+        """Force a full state reload from the store (reference :734-739),
+        bypassing the revision no-op check."""
+        self._last_revision = None
+        self.deserialize(skip_sync=False)
+
+    @classmethod
+    def list_control_plane_tasks(
+        cls, name: Optional[str] = None, project: Optional[str] = None,
+        tags: Optional[List[str]] = None,
+    ) -> List[dict]:
+        """List serving sessions ("control plane tasks"), optionally
+        filtered (reference :1372-1395)."""
+        out = []
+        for s in ServingStore().list_sessions():
+            if name and name not in s["name"]:
+                continue
+            if project and project != s["project"]:
+                continue
+            if tags and not set(tags).issubset(set(s["tags"])):
+                continue
+            out.append(s)
+        return out
+
+    # public telemetry hooks (reference :165-187); process_request calls
+    # the counters directly, these exist for API compatibility
+    def on_request_endpoint_telemetry(self, base_url=None,
+                                      version=None) -> None:
+        t = self._telemetry_counters(
+            self._normalize_endpoint_url(base_url or "", version))
+        if t:
+            t["requests"].inc()
+
+    def on_response_endpoint_telemetry(self, base_url=None,
+                                       version=None) -> None:
+        t = self._telemetry_counters(
+            self._normalize_endpoint_url(base_url or "", version))
+        if t:
+            t["responses"].inc()
+
     def _upload_preprocess(self, url: str, preprocess_code: str) -> str:
         if not os.path.exists(preprocess_code):
             raise ValueError(

@@ -569,3 +569,40 @@ def test_lightgbm_engine_predict_path(processor, store, tmp_path, monkeypatch):
         engine_type="lightgbm", serving_url="lgbm_ep", model_id=rec.model_id))
     out = run(processor.process_request("lgbm_ep", "", [[3.0, 6.0, 9.0]]))
     assert np.allclose(np.asarray(out), [6.0])
+
+
+def test_reference_api_parity_methods(processor, store):
+    """Public reference-API methods exist with matching semantics:
+    get_configuration / get_version / reload / list_control_plane_tasks /
+    list_metric_logging / telemetry hooks."""
+    from clearml_serving_amd import __version__
+
+    processor.configure(external_serving_base_url="http://x:8080/serve")
+    assert processor.get_configuration()["serving_base_url"] \
+        == "http://x:8080/serve"
+    assert processor.get_version() == __version__
+
+    processor.serialize()
+    processor.deserialize(skip_sync=True)
+    rev = processor._last_revision
+    processor.reload()  # forces a full re-deserialize
+    assert processor._last_revision == rev
+
+    import os as _os
+
+    _os.environ["CLEARML_SERVING_AMD_STORE"] = store.root
+    try:
+        sessions = type(processor).list_control_plane_tasks()
+        assert any(s["session_id"] == processor.get_id() for s in sessions)
+        assert type(processor).list_control_plane_tasks(
+            name="no-such-name-xyz") == []
+    finally:
+        _os.environ.pop("CLEARML_SERVING_AMD_STORE", None)
+
+    assert processor.list_metric_logging() \
+        == processor.list_endpoint_logging()
+
+    processor.on_request_endpoint_telemetry(base_url="tele_ep")
+    processor.on_response_endpoint_telemetry(base_url="tele_ep")
+    snap = processor.endpoint_telemetry_snapshot()["tele_ep"]
+    assert snap["requests"] == 1 and snap["responses"] == 1
