@@ -124,9 +124,22 @@ class LlmPreprocessRequest(BasePreprocessRequest):
     async def v1_rerank(self, body, state, collect_fn=None):
         return await self._engine.openai_rerank(body, self._served_name)
 
-    # vLLM also mounts rerank at /v2/rerank for cohere-client compat
+    # vLLM also mounts rerank at /v2/rerank for cohere-client compat,
+    # plus the bare /score and /rerank routes (reference serve types
+    # preprocess_service.py:1290-1336 use the bare names)
     async def v2_rerank(self, body, state, collect_fn=None):
         return await self._engine.openai_rerank(body, self._served_name)
+
+    async def rerank(self, body, state, collect_fn=None):
+        return await self._engine.openai_rerank(body, self._served_name)
+
+    async def score(self, body, state, collect_fn=None):
+        return await self._engine.openai_score(body, self._served_name)
+
+    async def version(self, body, state, collect_fn=None):
+        from ... import __version__
+
+        return {"version": __version__}
 
     # vLLM-compatible token utility routes (/serve/openai/tokenize)
     async def tokenize(self, body, state, collect_fn=None):

@@ -123,7 +123,7 @@ class ShmLlmProxyRequest(ShmProxyRequest):
         # ship the same way; generated on demand
         if name.startswith("v1_") or name in (
                 "pooling", "tokenize", "detokenize", "v2_rerank",
-                "classify"):
+                "classify", "score", "rerank", "version"):
             async def _method(body, state, collect_fn=None, _st=name):
                 # shipped as the method name; the owner resolves it the same
                 # way the processor does (serve_type.replace('/', '_'))

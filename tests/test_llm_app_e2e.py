@@ -282,3 +282,23 @@ def test_response_format_json_mode_refused(llm_client):
         "ignore_eos": True, "response_format": {"type": "text"},
         "messages": [{"role": "user", "content": "hi"}]})
     assert r.status_code == 200, r.text
+
+
+def test_bare_score_rerank_version_routes(llm_client):
+    """Reference serve-type parity: vLLM also mounts bare /score and
+    /rerank (preprocess_service.py:1290-1336) and a /version route."""
+    r = llm_client.post("/serve/openai/score", json={
+        "model": "test_llm", "text_1": "q", "text_2": ["a", "b"]})
+    assert r.status_code == 200, r.text
+    assert len(r.json()["data"]) == 2
+
+    r = llm_client.post("/serve/openai/rerank", json={
+        "model": "test_llm", "query": "q", "documents": ["a", "b"],
+        "top_n": 1})
+    assert r.status_code == 200, r.text
+    assert len(r.json()["results"]) == 1
+
+    r = llm_client.post("/serve/openai/version", json={"model": "test_llm"})
+    assert r.status_code == 200, r.text
+    from clearml_serving_amd import __version__
+    assert r.json()["version"] == __version__
