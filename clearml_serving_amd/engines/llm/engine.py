@@ -24,7 +24,7 @@ import random
 import threading
 import time
 import uuid
-from collections import deque
+from collections import OrderedDict, deque
 from dataclasses import dataclass, field
 from typing import Any, AsyncGenerator, Dict, List, Optional
 
@@ -235,6 +235,9 @@ class LlmEngineConfig:
     kv_dtype: str = "bfloat16"  # "fp8": e4m3 KV cache + per-token-per-head
                                 # scales -- 2x cached tokens per HBM byte,
                                 # half the KV reads on long-context decode
+    enable_prefix_caching: bool = False  # vLLM parity: content-addressed
+                                # reuse of full prompt KV blocks across
+                                # requests sharing a prefix
     weights: Optional[str] = None
     tokenizer_path: Optional[str] = None
     device: Optional[str] = None
@@ -274,7 +277,7 @@ class LlmEngineConfig:
                     "max_model_len", "max_prefill_tokens", "prefill_chunk",
                     "gpu_memory_fraction", "num_kv_blocks", "quantization",
                     "kv_dtype", "weights", "device", "decode_graphs",
-                    "speculative"):
+                    "speculative", "enable_prefix_caching"):
             for src in (card, aux):
                 if key in src and src[key] is not None:
                     setattr(cfg, key, src[key])
@@ -299,6 +302,110 @@ class BlockAllocator:
     @property
     def available(self) -> int:
         return len(self._free)
+
+
+class PrefixCacheAllocator(BlockAllocator):
+    """Automatic prefix caching (vLLM parity, ``enable_prefix_caching``):
+    FULL prompt blocks are content-addressed by the chain hash of their
+    token ids, so a request sharing a prompt prefix with an earlier one
+    reuses the cached K/V pages and skips their prefill compute.
+
+    Invariants that keep this exact:
+    - only FULL blocks are ever shared, and a block's position range is
+      determined by its index in the chain, so reuse is always
+      position-consistent (RoPE/learned positions bake absolute positions
+      into K/V);
+    - shared blocks are never written: a sequence's own writes start at
+      its first un-cached position (partial final blocks are always
+      owned), and decode/speculative writes happen at positions past the
+      prompt;
+    - at least one prompt token is always left to prefill so the
+      first-token logits exist (ncached <= len(prompt) - 1).
+
+    Freed cached blocks keep their contents and move to an LRU of
+    evictable blocks; allocation takes never-cached free blocks first and
+    evicts LRU entries (dropping their hash) under pressure."""
+
+    def __init__(self, num_blocks: int, block_size: int):
+        super().__init__(num_blocks)
+        self.block_size = block_size
+        self._hash2block: Dict[int, int] = {}
+        self._meta: Dict[int, list] = {}   # block -> [chain_hash, refcount]
+        self._lru: "OrderedDict[int, None]" = OrderedDict()  # refcount==0
+        self.hits = 0        # cached blocks reused
+        self.hit_tokens = 0  # prompt tokens skipped
+
+    def _chain_hashes(self, prompt_ids: List[int]) -> List[int]:
+        bs = self.block_size
+        out = []
+        h = 0
+        for i in range(len(prompt_ids) // bs):
+            h = hash((h, tuple(prompt_ids[i * bs:(i + 1) * bs])))
+            out.append(h)
+        return out
+
+    def match(self, prompt_ids: List[int]):
+        """Longest cached chain for this prompt -> (block_ids, n_tokens);
+        matched blocks are ref'd (protected from eviction) immediately."""
+        blocks: List[int] = []
+        for h in self._chain_hashes(prompt_ids):
+            b = self._hash2block.get(h)
+            if b is None:
+                break
+            blocks.append(b)
+        # never cache-hit the WHOLE prompt: the last token must prefill so
+        # its logits exist for first-token sampling
+        while blocks and len(blocks) * self.block_size >= len(prompt_ids):
+            blocks.pop()
+        for b in blocks:
+            meta = self._meta[b]
+            if meta[1] == 0:
+                self._lru.pop(b, None)
+            meta[1] += 1
+        # hit stats are recorded by the caller on successful admission
+        # (a match may be released when the admission budget refuses it)
+        return blocks, len(blocks) * self.block_size
+
+    def register(self, prompt_ids: List[int], blocks: List[int]) -> None:
+        """Publish a sequence's FULL prompt blocks for reuse (called once
+        its prompt K/V is in the cache). First writer wins: a hash already
+        published keeps its existing block."""
+        for h, b in zip(self._chain_hashes(prompt_ids), blocks):
+            if b in self._meta:
+                continue  # already cached (a reused block re-registers)
+            if h in self._hash2block:
+                continue  # another sequence published this content first
+            self._hash2block[h] = b
+            self._meta[b] = [h, 1]  # owned by the registering sequence
+
+    def alloc(self, n: int) -> List[int]:
+        out: List[int] = []
+        for _ in range(n):
+            if self._free:
+                out.append(self._free.pop())
+            elif self._lru:
+                b, _ = self._lru.popitem(last=False)  # evict oldest
+                h = self._meta.pop(b)[0]
+                self._hash2block.pop(h, None)
+                out.append(b)
+            else:
+                self._free.extend(out)
+                raise RuntimeError("KV cache exhausted")
+        return out
+
+    def free(self, blocks: List[int]) -> None:
+        for b in blocks:
+            meta = self._meta.get(b)
+            if meta is None:
+                self._free.append(b)  # never cached: plain free
+            else:
+                meta[1] -= 1
+                if meta[1] <= 0:
+                    self._lru[b] = None  # evictable, contents kept
+
+    @property
+    def available(self) -> int:
+        return len(self._free) + len(self._lru)
 
 
 class Sequence:
@@ -471,7 +578,9 @@ class LlmEngine:
                              else "cpu")
             dist.all_reduce(t, op=dist.ReduceOp.MIN)
             num_blocks = int(t.item())
-        self.allocator = BlockAllocator(num_blocks)
+        self.allocator = (PrefixCacheAllocator(num_blocks, bs)
+                          if cfg.enable_prefix_caching
+                          else BlockAllocator(num_blocks))
         if fp8_kv:
             # 4-tuple per layer: e4m3 byte caches + f32 [NB, Hkv, BS]
             # per-token-per-head scales (written by kv_cache_write's
@@ -667,6 +776,7 @@ class LlmEngine:
             self._prefill(fresh)
             for s in fresh:
                 s.prefilled = len(s.prompt_ids)
+                self._register_prefix(s)
         # long/continuing prompts advance by chunks (batched paged
         # attention against their cached histories)
         cont = [s for s in pending if s not in fresh]
@@ -717,24 +827,46 @@ class LlmEngine:
         admitted: List[Sequence] = []
         tokens = 0
         bs = self.cfg.block_size
+        prefix_cache = isinstance(self.allocator, PrefixCacheAllocator)
         while self.waiting and len(self.running) < self.cfg.max_num_seqs:
             seq = self.waiting[0]
             if seq.finished:  # aborted after the drain at step() top
                 self.waiting.pop(0)
                 continue
-            need = len(seq.prompt_ids)
+            cached: List[int] = []
+            ncached = 0
+            if prefix_cache:
+                cached, ncached = self.allocator.match(seq.prompt_ids)
+            need = len(seq.prompt_ids) - ncached
             if admitted and tokens + need > self.cfg.max_prefill_tokens:
+                if cached:
+                    self.allocator.free(cached)  # un-ref the match
                 break
-            need_blocks = (need + bs - 1) // bs
+            total_blocks = (len(seq.prompt_ids) + bs - 1) // bs
+            need_blocks = total_blocks - len(cached)
             # keep one spare block per running seq for decode growth
             if need_blocks + len(self.running) + 1 > self.allocator.available:
+                if cached:
+                    self.allocator.free(cached)
                 break
-            seq.blocks = self.allocator.alloc(need_blocks)
+            seq.blocks = cached + self.allocator.alloc(need_blocks)
+            seq.prefilled = ncached  # cached prefix skips its prefill
+            if prefix_cache and ncached:
+                self.allocator.hits += len(cached)
+                self.allocator.hit_tokens += ncached
+                self.stats["prefix_cache_hit_tokens"] = (
+                    self.stats.get("prefix_cache_hit_tokens", 0) + ncached)
             self.waiting.pop(0)
             self.running.append(seq)
             admitted.append(seq)
             tokens += need
         return admitted
+
+    def _register_prefix(self, s: Sequence) -> None:
+        """Publish a fully-prefilled sequence's full prompt blocks for
+        reuse (first writer wins; reused blocks re-register as no-ops)."""
+        if isinstance(self.allocator, PrefixCacheAllocator):
+            self.allocator.register(s.prompt_ids, s.blocks)
 
     def _slot(self, seq: Sequence, pos: int) -> int:
         bs = self.cfg.block_size
@@ -828,6 +960,7 @@ class LlmEngine:
             self.stats["prompt_tokens"] += c
             if complete:
                 done.append(s)
+                self._register_prefix(s)
         if done:
             self._sample_and_emit(done, logits, sample=plan["sample"])
 

@@ -31,9 +31,9 @@ def run(coro):
 
 
 def tiny_engine(**kw) -> LlmEngine:
-    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=128,
-                          block_size=16, max_model_len=256,
-                          device="cpu", **kw)
+    defaults = dict(preset="llama-tiny", num_kv_blocks=128,
+                    block_size=16, max_model_len=256, device="cpu")
+    cfg = LlmEngineConfig(**{**defaults, **kw})
     eng = LlmEngine(cfg)
     eng.start()
     return eng
@@ -1496,3 +1496,126 @@ def test_gpt2_fp8_refused_loudly():
 
     eng = gpt2_engine()
     assert quantize_llama_fp8(eng.model) == 0
+
+
+# ------------------------------------------------------------------ #
+# automatic prefix caching
+# ------------------------------------------------------------------ #
+def pc_engine(**kw):
+    return tiny_engine(enable_prefix_caching=True, **kw)
+
+
+def test_prefix_cache_exact_and_hits():
+    """Identical prompts: the second request reuses the first's full
+    prompt blocks (hit stats move) and produces IDENTICAL greedy output;
+    divergent-suffix prompts share only the common full blocks."""
+    torch.manual_seed(4)
+    plain = tiny_engine()
+    torch.manual_seed(4)
+    pc = pc_engine()
+    prompt = "shared prefix " * 4  # > 2 full blocks of byte tokens
+
+    a1 = _greedy(plain, prompt, 12)
+    b1 = _greedy(pc, prompt, 12)
+    assert a1 == b1
+    assert pc.allocator.hit_tokens == 0  # first request: nothing cached
+
+    a2 = _greedy(plain, prompt, 12)
+    b2 = _greedy(pc, prompt, 12)
+    assert a2 == b2 == a1
+    assert pc.allocator.hit_tokens >= 16  # >= one full block reused
+    assert pc.stats["prefix_cache_hit_tokens"] >= 16
+
+    # divergent suffix: still shares the common prefix blocks
+    before = pc.allocator.hit_tokens
+    c = _greedy(pc, prompt + "DIFFERENT TAIL", 12)
+    assert pc.allocator.hit_tokens > before
+    d = _greedy(plain, prompt + "DIFFERENT TAIL", 12)
+    assert c == d
+
+
+def test_prefix_cache_under_eviction_pressure():
+    """A small pool forces evictions: requests keep completing correctly
+    and every block is accounted for at the end."""
+    torch.manual_seed(5)
+    pc = pc_engine(num_kv_blocks=24)
+    torch.manual_seed(5)
+    plain = tiny_engine(num_kv_blocks=24)
+    prompts = ["aaaa " * 8, "bbbb " * 8, "cccc " * 8, "aaaa " * 8,
+               "dddd " * 8, "aaaa " * 8]
+    for p in prompts:
+        assert _greedy(pc, p, 8) == _greedy(plain, p, 8)
+    # full accounting: every block is free or evictable
+    assert pc.allocator.available == pc.allocator.num_blocks
+
+
+def test_prefix_cache_concurrent_and_aborted():
+    eng = pc_engine()
+
+    async def main():
+        params = SamplingParams(temperature=0.0, max_tokens=8,
+                                ignore_eos=True)
+
+        async def one(i):
+            toks = []
+            async for item in eng.generate("common root " * 3 + str(i),
+                                           params):
+                toks.extend(item["token_ids"])
+            return toks
+
+        async def aborted():
+            task = asyncio.ensure_future(eng.generate_simple(
+                {"prompt": "common root " * 3 + "x", "max_tokens": 60,
+                 "temperature": 0.0, "ignore_eos": True}))
+            await asyncio.sleep(0.05)
+            task.cancel()
+            try:
+                await task
+            except asyncio.CancelledError:
+                pass
+
+        outs = await asyncio.gather(*[one(i) for i in range(4)], aborted())
+        return outs[:4]
+
+    outs = run(main())
+    assert all(len(o) == 8 for o in outs)
+    # drain cleanup
+    import time as _time
+
+    for _ in range(100):
+        if eng.allocator.available == eng.allocator.num_blocks:
+            break
+        _time.sleep(0.02)
+    assert eng.allocator.available == eng.allocator.num_blocks
+
+
+def test_prefix_cache_with_chunked_prefill_and_spec():
+    """Interactions: a cached prefix admits with prefilled>0 and continues
+    through the chunked path; speculation still produces exact output."""
+    torch.manual_seed(6)
+    pc = pc_engine(prefill_chunk=32,
+                   speculative={"method": "ngram", "num_spec_tokens": 4,
+                                "ngram": 2})
+    torch.manual_seed(6)
+    plain = tiny_engine(prefill_chunk=32)
+    prompt = "abc abc abc " * 6  # long enough to chunk
+    a = _greedy(plain, prompt, 16)
+    assert _greedy(pc, prompt, 16) == a
+    assert _greedy(pc, prompt, 16) == a  # cache-hit pass
+    assert pc.allocator.hit_tokens > 0
+
+
+def test_prefix_cache_never_full_prompt():
+    """ncached always leaves >= 1 token to prefill (the last token's
+    logits feed first-token sampling)."""
+    from clearml_serving_amd.engines.llm.engine import PrefixCacheAllocator
+
+    a = PrefixCacheAllocator(16, 4)
+    ids = list(range(8))  # exactly 2 full blocks
+    blocks = a.alloc(2)
+    a.register(ids, blocks)
+    got, n = a.match(list(ids))
+    assert len(got) == 1 and n == 4  # NOT both blocks: last token prefills
+    a.free(got)
+    got2, n2 = a.match(ids + [99])  # longer prompt: both blocks reusable
+    assert len(got2) == 2 and n2 == 8

@@ -146,3 +146,35 @@ def test_attention_prefill_paged_tiny_chunks():
         torch.testing.assert_close(got[i, :chunk[i]].float().cpu(),
                                    ref[i, :chunk[i]].float(),
                                    atol=3e-2, rtol=3e-2)
+
+
+def test_prefix_caching_gpu_exact():
+    """Automatic prefix caching on the GPU path: the cache-hit request
+    (paged-prefill over reused pages + decode) emits identical greedy
+    tokens, and blocks are actually reused."""
+    def mk(pc):
+        torch.manual_seed(17)
+        cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=128,
+                              block_size=16, max_model_len=256, device=DEV,
+                              enable_prefix_caching=pc)
+        eng = LlmEngine(cfg)
+        eng.start()
+        return eng
+
+    def gen(eng, prompt):
+        async def go():
+            toks = []
+            async for item in eng.generate(prompt, SamplingParams(
+                    temperature=0.0, max_tokens=16, ignore_eos=True)):
+                toks.extend(item["token_ids"])
+            return toks
+
+        return run(go())
+
+    plain = mk(False)
+    pc = mk(True)
+    prompt = "gpu shared prefix " * 4
+    a = gen(plain, prompt)
+    assert gen(pc, prompt) == a
+    assert gen(pc, prompt) == a  # hit pass
+    assert pc.allocator.hit_tokens >= 16
