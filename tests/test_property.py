@@ -179,3 +179,67 @@ def test_batcher_every_response_matches_direct_call(sizes, feat):
     for r, o in zip(reqs, outs):
         want = model(r[None])[0]
         torch.testing.assert_close(o, want, atol=1e-6, rtol=1e-6)
+
+
+# --------------------------------------------------------------------- #
+# prefix-cache allocator invariants under random op sequences
+# --------------------------------------------------------------------- #
+@settings(max_examples=30, deadline=None)
+@given(ops_seq=st.lists(
+    st.tuples(st.sampled_from(["admit", "finish"]),
+              st.integers(min_value=0, max_value=5),   # prompt family
+              st.integers(min_value=1, max_value=40)), # prompt length
+    min_size=1, max_size=60))
+def test_prefix_allocator_invariants(ops_seq):
+    """Random admit/finish interleavings: no block is ever handed to two
+    live sequences, accounting always balances, and matches never exceed
+    len(prompt)-1 tokens."""
+    from clearml_serving_amd.engines.llm.engine import PrefixCacheAllocator
+
+    BS = 4
+    N = 16
+    a = PrefixCacheAllocator(N, BS)
+    live = []  # (blocks, prompt)
+    in_use = {}  # block -> count of live holders (shared cached blocks ok)
+
+    def track(blocks):
+        for b in blocks:
+            in_use[b] = in_use.get(b, 0) + 1
+
+    def untrack(blocks):
+        for b in blocks:
+            in_use[b] -= 1
+            if not in_use[b]:
+                del in_use[b]
+
+    for op, fam, ln in ops_seq:
+        if op == "admit":
+            prompt = [(fam * 97 + i) % 50 for i in range(ln)]
+            cached, ncached = a.match(prompt)
+            assert ncached <= max(len(prompt) - 1, 0)
+            need = (len(prompt) + BS - 1) // BS - len(cached)
+            if need > a.available:
+                a.free(cached)
+                continue
+            try:
+                fresh = a.alloc(need)
+            except RuntimeError:
+                a.free(cached)
+                continue
+            # a fresh block must not be LIVE anywhere (shared cached
+            # blocks may be held by several sequences; owned may not)
+            for b in fresh:
+                assert b not in in_use, (b, in_use)
+            blocks = cached + fresh
+            track(blocks)
+            a.register(prompt, blocks)
+            live.append((blocks, prompt))
+        elif live:
+            blocks, _ = live.pop(fam % len(live))
+            untrack(blocks)
+            a.free(blocks)
+
+    for blocks, _ in live:
+        a.free(blocks)
+    # every block ends up free or evictable
+    assert a.available == N
