@@ -345,10 +345,12 @@ class PrefixCacheAllocator(BlockAllocator):
         return out
 
     def match(self, prompt_ids: List[int]):
-        """Longest cached chain for this prompt -> (block_ids, n_tokens);
-        matched blocks are ref'd (protected from eviction) immediately."""
+        """Longest cached chain for this prompt -> (block_ids, n_tokens,
+        chain_hash at the match point); matched blocks are ref'd
+        (protected from eviction) immediately."""
         blocks: List[int] = []
-        for h in self._chain_hashes(prompt_ids):
+        hashes = self._chain_hashes(prompt_ids)
+        for h in hashes:
             b = self._hash2block.get(h)
             if b is None:
                 break
@@ -364,19 +366,19 @@ class PrefixCacheAllocator(BlockAllocator):
             meta[1] += 1
         # hit stats are recorded by the caller on successful admission
         # (a match may be released when the admission budget refuses it)
-        return blocks, len(blocks) * self.block_size
+        chain = hashes[len(blocks) - 1] if blocks else 0
+        return blocks, len(blocks) * self.block_size, chain
 
-    def register(self, prompt_ids: List[int], blocks: List[int]) -> None:
-        """Publish a sequence's FULL prompt blocks for reuse (called once
-        its prompt K/V is in the cache). First writer wins: a hash already
-        published keeps its existing block."""
-        for h, b in zip(self._chain_hashes(prompt_ids), blocks):
-            if b in self._meta:
-                continue  # already cached (a reused block re-registers)
-            if h in self._hash2block:
-                continue  # another sequence published this content first
-            self._hash2block[h] = b
-            self._meta[b] = [h, 1]  # owned by the registering sequence
+    def register_block(self, prev_hash: int, tokens: List[int],
+                       block_id: int) -> int:
+        """Publish ONE full block (prompt or generated) for reuse; returns
+        the advanced chain hash. First writer wins: content already
+        published elsewhere leaves this copy private."""
+        h = hash((prev_hash, tuple(tokens)))
+        if h not in self._hash2block and block_id not in self._meta:
+            self._hash2block[h] = block_id
+            self._meta[block_id] = [h, 1]  # the registering holder's ref
+        return h
 
     def alloc(self, n: int) -> List[int]:
         out: List[int] = []
@@ -424,6 +426,10 @@ class Sequence:
         self.created = time.time()
         self.first_token_time: Optional[float] = None
         self.spec_misses = 0  # consecutive fully-rejected proposals
+        # prefix-cache registration progress: tokens covered by published
+        # blocks, and the chain hash at that point
+        self.cached_upto = 0
+        self.cache_hash = 0
 
     def __len__(self):
         return len(self.prompt_ids) + len(self.output_ids)
@@ -802,6 +808,7 @@ class LlmEngine:
             if rest:
                 self._decode(rest)
         for s in list(self.running):
+            self._register_prefix(s)  # publish newly-FILLED blocks
             if s.finished:
                 self.running.remove(s)
                 self.allocator.free(s.blocks)
@@ -835,8 +842,10 @@ class LlmEngine:
                 continue
             cached: List[int] = []
             ncached = 0
+            chain = 0
             if prefix_cache:
-                cached, ncached = self.allocator.match(seq.prompt_ids)
+                cached, ncached, chain = self.allocator.match(
+                    seq.prompt_ids)
             need = len(seq.prompt_ids) - ncached
             if admitted and tokens + need > self.cfg.max_prefill_tokens:
                 if cached:
@@ -851,6 +860,10 @@ class LlmEngine:
                 break
             seq.blocks = cached + self.allocator.alloc(need_blocks)
             seq.prefilled = ncached  # cached prefix skips its prefill
+            # registration chain restarts at the match point (also resets
+            # stale state on preemption re-admission)
+            seq.cached_upto = ncached
+            seq.cache_hash = chain
             if prefix_cache and ncached:
                 self.allocator.hits += len(cached)
                 self.allocator.hit_tokens += ncached
@@ -863,10 +876,29 @@ class LlmEngine:
         return admitted
 
     def _register_prefix(self, s: Sequence) -> None:
-        """Publish a fully-prefilled sequence's full prompt blocks for
-        reuse (first writer wins; reused blocks re-register as no-ops)."""
-        if isinstance(self.allocator, PrefixCacheAllocator):
-            self.allocator.register(s.prompt_ids, s.blocks)
+        """Publish every full block whose K/V is in the cache -- prompt
+        AND generated tokens (multi-turn chat resends assistant output as
+        the next prompt, so generated blocks are reusable prefixes too).
+        KV availability: mid-prefill only the first ``prefilled`` tokens
+        are written; once the prompt is done, everything except the
+        newest token (written on its NEXT step) is."""
+        alloc = self.allocator
+        if not isinstance(alloc, PrefixCacheAllocator):
+            return
+        bs = alloc.block_size
+        if s.prefilled < len(s.prompt_ids):
+            avail = s.prefilled
+        else:
+            avail = len(s.prompt_ids) + max(len(s.output_ids) - 1, 0)
+        if s.cached_upto + bs > avail:
+            return
+        toks = s.prompt_ids + s.output_ids
+        while s.cached_upto + bs <= avail:
+            blk = s.cached_upto // bs
+            s.cache_hash = alloc.register_block(
+                s.cache_hash, toks[s.cached_upto:s.cached_upto + bs],
+                s.blocks[blk])
+            s.cached_upto += bs
 
     def _slot(self, seq: Sequence, pos: int) -> int:
         bs = self.cfg.block_size

@@ -1613,9 +1613,54 @@ def test_prefix_cache_never_full_prompt():
     a = PrefixCacheAllocator(16, 4)
     ids = list(range(8))  # exactly 2 full blocks
     blocks = a.alloc(2)
-    a.register(ids, blocks)
-    got, n = a.match(list(ids))
+    h = a.register_block(0, ids[:4], blocks[0])
+    a.register_block(h, ids[4:], blocks[1])
+    got, n, _ = a.match(list(ids))
     assert len(got) == 1 and n == 4  # NOT both blocks: last token prefills
     a.free(got)
-    got2, n2 = a.match(ids + [99])  # longer prompt: both blocks reusable
+    got2, n2, _ = a.match(ids + [99])  # longer prompt: both reusable
     assert len(got2) == 2 and n2 == 8
+
+
+def test_prefix_cache_covers_generated_tokens():
+    """Multi-turn pattern: turn 2's prompt = turn 1's prompt + the
+    assistant's GENERATED reply + new text. Generated full blocks were
+    registered as they filled, so the turn-2 hit extends past the
+    original prompt into the generated region -- and output stays exact
+    vs the plain engine."""
+    torch.manual_seed(11)
+    pc = pc_engine()
+    torch.manual_seed(11)
+    plain = tiny_engine()
+
+    turn1_ids = pc.tokenizer.encode("multi turn root " * 3)
+
+    def gen_ids(eng, ids, n):
+        async def go():
+            seq = await eng.add_request(list(ids), SamplingParams(
+                temperature=0.0, max_tokens=n, ignore_eos=True))
+            toks = []
+            while True:
+                item = await seq.stream.get()
+                toks.extend(item["token_ids"])
+                if item["finished"]:
+                    return toks
+
+        return run(go())
+
+    out1_pc = gen_ids(pc, turn1_ids, 40)
+    out1_pl = gen_ids(plain, turn1_ids, 40)
+    assert out1_pc == out1_pl
+
+    # turn 2: context = turn1 prompt + reply + follow-up
+    follow = pc.tokenizer.encode(" and then?")
+    turn2_ids = turn1_ids + out1_pc + follow
+    before = pc.allocator.hit_tokens
+    out2_pc = gen_ids(pc, turn2_ids, 8)
+    out2_pl = gen_ids(plain, turn2_ids, 8)
+    assert out2_pc == out2_pl
+    hit = pc.allocator.hit_tokens - before
+    bs = pc.cfg.block_size
+    # the hit must reach INTO the generated region: more than the full
+    # blocks of the original prompt alone
+    assert hit > (len(turn1_ids) // bs) * bs, hit

@@ -215,7 +215,7 @@ def test_prefix_allocator_invariants(ops_seq):
     for op, fam, ln in ops_seq:
         if op == "admit":
             prompt = [(fam * 97 + i) % 50 for i in range(ln)]
-            cached, ncached = a.match(prompt)
+            cached, ncached, chain = a.match(prompt)
             assert ncached <= max(len(prompt) - 1, 0)
             need = (len(prompt) + BS - 1) // BS - len(cached)
             if need > a.available:
@@ -232,7 +232,10 @@ def test_prefix_allocator_invariants(ops_seq):
                 assert b not in in_use, (b, in_use)
             blocks = cached + fresh
             track(blocks)
-            a.register(prompt, blocks)
+            h = chain
+            for i in range(ncached // BS, len(prompt) // BS):
+                h = a.register_block(h, prompt[i * BS:(i + 1) * BS],
+                                     blocks[i])
             live.append((blocks, prompt))
         elif live:
             blocks, _ = live.pop(fam % len(live))
