@@ -13,8 +13,14 @@ from clearml_serving_amd.serving.processor import ModelRequestProcessor
 from clearml_serving_amd.store import ServingStore
 
 
-@pytest.mark.timeout(180)
-def test_mixed_traffic_survives_hot_reloads(tmp_path):
+@pytest.mark.timeout(360)
+@pytest.mark.parametrize("llm_extras", [
+    {},
+    # every opt-in engine feature at once: speculation + prefix caching
+    {"speculative": {"method": "ngram", "num_spec_tokens": 4, "ngram": 2},
+     "enable_prefix_caching": True},
+], ids=["plain", "spec+prefixcache"])
+def test_mixed_traffic_survives_hot_reloads(tmp_path, llm_extras):
     import joblib
     import numpy as np
     from sklearn.linear_model import LinearRegression
@@ -41,7 +47,8 @@ def test_mixed_traffic_survives_hot_reloads(tmp_path):
     card = tmp_path / "card.json"
     card.write_text(json.dumps({
         "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 96,
-        "block_size": 16, "max_model_len": 128, "device": "cpu"}))
+        "block_size": 16, "max_model_len": 128, "device": "cpu",
+        **llm_extras}))
     lrec = store.register_model(name="tl", project="p", path=str(card))
     proc.add_endpoint(ModelEndpoint(engine_type="llm", serving_url="tl",
                                     model_id=lrec.model_id))
