@@ -1664,3 +1664,17 @@ def test_prefix_cache_covers_generated_tokens():
     # the hit must reach INTO the generated region: more than the full
     # blocks of the original prompt alone
     assert hit > (len(turn1_ids) // bs) * bs, hit
+
+
+def test_prefix_cache_with_fp8_kv():
+    """Cached blocks carry their per-token fp8 scales with them: reuse
+    over the e4m3 cache stays exact vs the plain fp8-KV engine."""
+    torch.manual_seed(12)
+    pc = pc_engine(kv_dtype="fp8")
+    torch.manual_seed(12)
+    plain = tiny_engine(kv_dtype="fp8")
+    prompt = "fp8 cached prefix " * 4
+    a = _greedy(plain, prompt, 12)
+    assert _greedy(pc, prompt, 12) == a
+    assert _greedy(pc, prompt, 12) == a
+    assert pc.allocator.hit_tokens >= 16
