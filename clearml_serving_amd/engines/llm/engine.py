@@ -1800,6 +1800,14 @@ class LlmEngine:
         if n > 1 and body.get("stream"):
             raise ValueError("streaming with n > 1 is not supported; "
                              "request the choices without 'stream'")
+        bo = body.get("best_of")
+        if bo is not None and int(bo) != n:
+            # OpenAI deprecated best_of; silently ignoring it would change
+            # semantics (it implies server-side reranking)
+            raise ValueError(
+                "'best_of' != n is not supported (best_of reranking is "
+                "deprecated by OpenAI); request n choices and pick "
+                "client-side")
         return n
 
     def _choice_params(self, body: Dict[str, Any], n: int

@@ -1678,3 +1678,15 @@ def test_prefix_cache_with_fp8_kv():
     assert _greedy(pc, prompt, 12) == a
     assert _greedy(pc, prompt, 12) == a
     assert pc.allocator.hit_tokens >= 16
+
+
+def test_best_of_guard():
+    eng = tiny_engine()
+    with pytest.raises(ValueError):
+        run(eng.openai_completions(
+            {"prompt": "x", "n": 1, "best_of": 4}, "m"))
+    # best_of == n passes through
+    resp = run(eng.openai_completions(
+        {"prompt": "x", "n": 2, "best_of": 2, "max_tokens": 2,
+         "temperature": 0.0, "ignore_eos": True}, "m"))
+    assert len(resp["choices"]) == 2
