@@ -6,12 +6,24 @@ tokens through asyncio queues.
 
 Engine step (continuous batching):
 1. admit waiting prompts while the prefill-token budget and free KV blocks
-   allow (prompts padded to one batch; per-seq lengths mask attention)
+   allow (prompts padded to one batch; per-seq lengths mask attention);
+   with prefix caching on, a cached prompt prefix admits pre-filled
 2. run ONE prefill forward for the admitted batch, scatter K/V into pages,
-   sample each sequence's first token
+   sample each sequence's first token (long prompts advance in chunks,
+   fair-shared, interleaving with decode)
 3. run ONE decode forward for every running sequence (paged decode
-   attention), sample next tokens
+   attention, hipGraph-replayed per batch bucket), sample next tokens;
+   greedy sequences may take the ngram speculative path instead (several
+   tokens verified in one forward, exact output)
 4. emit tokens to per-request streams; finished sequences free their pages
+
+Model families: llama / qwen2 (models/llama.py) and gpt2 (models/gpt2.py)
+share one forward interface, so every scheduler feature serves both.
+Opt-in subsystems, each exact by test: fp8 weights (fused activation
+quant), fp8 KV cache, ngram speculation, automatic prefix caching
+(content-addressed blocks, prompt + generated), TP over RCCL/xGMI with
+single-tensor plan broadcast + on-shard sampling (+ opt-in microbatch
+comm/compute overlap). Sampling surface: docs/API.md.
 
 KV sizing: pages are allocated once at startup from a fraction of free HBM
 (288 GB/GPU -> tens of thousands of 16-token pages for an 8B model).
