@@ -1869,6 +1869,11 @@ class LlmEngine:
         }
 
     async def openai_completions(self, body: Dict[str, Any], model_name: str):
+        if body.get("suffix"):
+            # fill-in-middle needs a FIM-trained model + template; honest
+            # 422 instead of silently generating without the suffix
+            raise ValueError(
+                "'suffix' (fill-in-middle) is not supported; omit it")
         prompt = body.get("prompt") or ""
         if isinstance(prompt, list):
             prompt = prompt[0] if prompt else ""

@@ -1690,3 +1690,10 @@ def test_best_of_guard():
         {"prompt": "x", "n": 2, "best_of": 2, "max_tokens": 2,
          "temperature": 0.0, "ignore_eos": True}, "m"))
     assert len(resp["choices"]) == 2
+
+
+def test_suffix_fim_guard():
+    eng = tiny_engine()
+    with pytest.raises(ValueError):
+        run(eng.openai_completions(
+            {"prompt": "a", "suffix": "z", "max_tokens": 2}, "m"))
