@@ -73,4 +73,4 @@ def load_weights(model: torch.nn.Module, path: str) -> None:
     model.load_state_dict(convert_hf_auto(state))
 
 
-from . import bert, llama, resnet  # noqa: E402,F401  (register architectures)
+from . import bert, gpt2, llama, resnet  # noqa: E402,F401  (register architectures)
