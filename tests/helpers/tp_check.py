@@ -195,6 +195,43 @@ def check_microbatch_pipeline(rank, world):
     os.environ.pop("CMLS_TP_MICROBATCH", None)
 
 
+def check_prefix_cache_under_tp(rank, world):
+    """Prefix caching under TP: rank 0's allocator schedules, cached
+    admission routes the remainder through the CHUNK plan (prefilled>0),
+    so workers see a different broadcast mode -- outputs must stay
+    identical across the cold and cache-hit passes."""
+    mcfg = _tiny_cfg(world)
+    cfg = LlmEngineConfig(preset="llama-tiny", num_kv_blocks=64,
+                          block_size=16, max_model_len=128, device="cpu",
+                          enable_prefix_caching=True,
+                          overrides={"heads": mcfg.heads,
+                                     "kv_heads": mcfg.kv_heads})
+    eng = LlmEngine(cfg)
+    eng.start()
+    if rank == 0:
+        async def gen():
+            toks = []
+            async for item in eng.generate(
+                    "tp cached prefix " * 4,
+                    SamplingParams(temperature=0.0, max_tokens=8,
+                                   ignore_eos=True)):
+                toks.extend(item["token_ids"])
+            return toks
+
+        loop = asyncio.new_event_loop()
+        try:
+            a = loop.run_until_complete(gen())
+            b = loop.run_until_complete(gen())  # cache-hit pass
+        finally:
+            loop.close()
+        assert a == b, (a, b)
+        assert eng.allocator.hit_tokens >= 16, eng.allocator.hit_tokens
+        eng.tp_shutdown()
+        print("TP-PREFIXCACHE-OK", flush=True)
+    else:
+        eng.run_tp_worker()
+
+
 def main():
     dist.init_process_group(backend="gloo")
     rank, world = dist.get_rank(), dist.get_world_size()
@@ -203,6 +240,8 @@ def main():
     check_engine_protocol(rank, world)
     dist.barrier()
     check_microbatch_pipeline(rank, world)
+    dist.barrier()
+    check_prefix_cache_under_tp(rank, world)
     dist.barrier()
     dist.destroy_process_group()
 

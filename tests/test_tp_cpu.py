@@ -24,6 +24,7 @@ def test_tp2_math_and_engine_protocol():
     assert "TP-MATH-OK" in out.stdout
     assert "TP-ENGINE-OK" in out.stdout
     assert "TP-MICROBATCH-OK" in out.stdout
+    assert "TP-PREFIXCACHE-OK" in out.stdout
 
 
 @pytest.mark.timeout(300)
@@ -48,6 +49,7 @@ def test_tp4_math_and_engine_protocol():
     assert "TP-SAMPLE-OK" in out.stdout
     assert "TP-ENGINE-OK" in out.stdout
     assert "TP-MICROBATCH-OK" in out.stdout
+    assert "TP-PREFIXCACHE-OK" in out.stdout
 
 
 @pytest.mark.timeout(300)
