@@ -148,14 +148,16 @@ class SamplingParams:
             repetition = _num("repetition_penalty", 1.0)
             seed = body.get("seed")
             seed = int(seed) if seed is not None else None
-        except (TypeError, ValueError):
+        except (TypeError, ValueError, OverflowError):
             raise ValueError("sampling parameters must be numeric "
                              "(temperature/top_k/top_p/max_tokens/"
                              "penalties/seed)")
-        if temperature < 0.0:
+        import math as _math
+
+        if not (_math.isfinite(temperature) and temperature >= 0.0):
             raise ValueError(
-                "temperature must be >= 0 (0 = greedy), got {}".format(
-                    temperature))
+                "temperature must be finite and >= 0 (0 = greedy), got "
+                "{}".format(temperature))
         if top_k < 0:
             raise ValueError("top_k must be >= 0 (0 = disabled), got {}"
                              .format(top_k))
@@ -166,7 +168,7 @@ class SamplingParams:
                 max_tokens))
         try:
             min_tokens = int(body.get("min_tokens", 0) or 0)
-        except (TypeError, ValueError):
+        except (TypeError, ValueError, OverflowError):
             raise ValueError("'min_tokens' must be an int")
         if not (0 <= min_tokens <= max_tokens):
             raise ValueError(
@@ -178,18 +180,21 @@ class SamplingParams:
         if not (-2.0 <= frequency <= 2.0):
             raise ValueError("frequency_penalty must be in [-2, 2], got {}"
                              .format(frequency))
-        if repetition <= 0.0:
-            raise ValueError("repetition_penalty must be > 0, got {}"
-                             .format(repetition))
+        if not (_math.isfinite(repetition) and repetition > 0.0):
+            raise ValueError("repetition_penalty must be finite and > 0, "
+                             "got {}".format(repetition))
         # logprobs: completions style ("logprobs": N) or chat style
         # ("logprobs": true + "top_logprobs": N)
         lp = body.get("logprobs")
         if isinstance(lp, bool):
-            lp = (int(body.get("top_logprobs", 0) or 0)) if lp else None
+            try:
+                lp = (int(body.get("top_logprobs", 0) or 0)) if lp else None
+            except (TypeError, ValueError, OverflowError):
+                raise ValueError("'top_logprobs' must be an int")
         elif lp is not None:
             try:
                 lp = int(lp)
-            except (TypeError, ValueError):
+            except (TypeError, ValueError, OverflowError):
                 raise ValueError("'logprobs' must be an int or bool")
         if lp is not None and not (0 <= lp <= 20):
             raise ValueError("logprobs must be in [0, 20], got {}".format(lp))
@@ -205,12 +210,15 @@ class SamplingParams:
         stop = body.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
-        if not all(isinstance(s, str) for s in stop):
+        if not isinstance(stop, (list, tuple)) \
+                or not all(isinstance(s, str) for s in stop):
             raise ValueError("'stop' must be a string or list of strings")
         stop_ids = body.get("stop_token_ids") or []
+        if not isinstance(stop_ids, (list, tuple)):
+            raise ValueError("'stop_token_ids' must be a list of ints")
         try:
             stop_ids = [int(t) for t in stop_ids]
-        except (TypeError, ValueError):
+        except (TypeError, ValueError, OverflowError):
             raise ValueError("'stop_token_ids' must be a list of ints")
         return cls(
             temperature=temperature,
@@ -1805,7 +1813,7 @@ class LlmEngine:
         n = body.get("n", 1)
         try:
             n = int(n) if n is not None else 1
-        except (TypeError, ValueError):
+        except (TypeError, ValueError, OverflowError):
             raise ValueError("'n' must be an integer")
         if not (1 <= n <= 16):
             raise ValueError("'n' must be in [1, 16], got {}".format(n))
@@ -1813,7 +1821,11 @@ class LlmEngine:
             raise ValueError("streaming with n > 1 is not supported; "
                              "request the choices without 'stream'")
         bo = body.get("best_of")
-        if bo is not None and int(bo) != n:
+        try:
+            bo = int(bo) if bo is not None else None
+        except (TypeError, ValueError, OverflowError):
+            raise ValueError("'best_of' must be an integer")
+        if bo is not None and bo != n:
             # OpenAI deprecated best_of; silently ignoring it would change
             # semantics (it implies server-side reranking)
             raise ValueError(

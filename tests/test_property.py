@@ -246,3 +246,44 @@ def test_prefix_allocator_invariants(ops_seq):
         a.free(blocks)
     # every block ends up free or evictable
     assert a.available == N
+
+
+# --------------------------------------------------------------------- #
+# request-body fuzz: from_request never escapes ValueError
+# --------------------------------------------------------------------- #
+_scalar = st.one_of(st.none(), st.booleans(), st.integers(-10**6, 10**6),
+                    st.floats(allow_nan=True, allow_infinity=True),
+                    st.text(max_size=8))
+_value = st.one_of(_scalar, st.lists(_scalar, max_size=4),
+                   st.dictionaries(st.text(max_size=6), _scalar, max_size=3))
+
+
+@settings(max_examples=200, deadline=None)
+@given(body=st.dictionaries(
+    st.sampled_from(["temperature", "top_k", "top_p", "max_tokens",
+                     "max_completion_tokens", "min_tokens", "seed", "n",
+                     "presence_penalty", "frequency_penalty",
+                     "repetition_penalty", "logprobs", "top_logprobs",
+                     "stop", "stop_token_ids", "ignore_eos", "best_of",
+                     "response_format", "stream", "echo", "garbage"]),
+    _value, max_size=8))
+def test_sampling_params_fuzz_never_crashes(body):
+    """Arbitrary request bodies either validate (SamplingParams) or raise
+    ValueError (-> 422); any other exception class would 500 and is a
+    bug in the validation layer."""
+    from clearml_serving_amd.engines.llm.engine import SamplingParams
+
+    try:
+        p = SamplingParams.from_request(dict(body))
+    except ValueError:
+        return
+    # accepted: the invariants the sampling kernels rely on hold
+    assert p.temperature >= 0.0
+    assert p.top_k >= 0
+    assert 0.0 < p.top_p <= 1.0
+    assert 1 <= p.max_tokens
+    assert 0 <= p.min_tokens <= p.max_tokens
+    assert -2.0 <= p.presence_penalty <= 2.0
+    assert -2.0 <= p.frequency_penalty <= 2.0
+    assert p.repetition_penalty > 0.0
+    assert p.logprobs is None or 0 <= p.logprobs <= 20
