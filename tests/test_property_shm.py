@@ -80,3 +80,62 @@ def test_plan_codec_decode_roundtrip_random(b, seed, data):
         assert got[1] == want[1] and got[3] == want[3]
         assert got[0] == pytest.approx(want[0], abs=1e-6)
         assert got[2] == pytest.approx(want[2], abs=1e-6)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    url=st.text(min_size=0, max_size=40),
+    req_id=st.integers(min_value=0, max_value=2**63 - 1),
+    data=st.one_of(
+        # tensor payloads: dict of arrays with random dtypes/shapes
+        st.dictionaries(
+            st.text(min_size=1, max_size=10),
+            st.tuples(
+                st.sampled_from(["float32", "int64", "int32", "uint8",
+                                 "bool", "float16"]),
+                st.lists(st.integers(1, 5), min_size=0, max_size=3)),
+            min_size=1, max_size=4),
+        # object payloads (pickle fallback)
+        st.recursive(
+            st.one_of(st.none(), st.booleans(), st.integers(),
+                      st.floats(allow_nan=False), st.text(max_size=10)),
+            lambda c: st.lists(c, max_size=3)
+            | st.dictionaries(st.text(max_size=5), c, max_size=3),
+            max_leaves=10),
+    ))
+def test_request_framing_roundtrip_fuzz(url, req_id, data):
+    """pack_request/unpack_request roundtrip over arbitrary tensor maps
+    and JSON-ish objects: ids, urls, names, dtypes, shapes and values all
+    survive."""
+    import numpy as np
+
+    from clearml_serving_amd.serving import shm_transport as st_mod
+
+    if isinstance(data, dict) and data and all(
+            isinstance(v, tuple) for v in data.values()):
+        arrays = {k: (np.zeros(shape, dtype=dt) + 1).astype(dt)
+                  for k, (dt, shape) in data.items()}
+        buf = st_mod.pack_request(req_id, url, arrays)
+        rid, u, out = st_mod.unpack_request(buf)
+        assert rid == req_id and u == url
+        if len(arrays) == 1 and "" in arrays:
+            out = {"": out}
+        assert set(out) == set(arrays)
+        for k in arrays:
+            assert out[k].dtype == arrays[k].dtype
+            assert out[k].shape == arrays[k].shape
+            np.testing.assert_array_equal(out[k], arrays[k])
+    else:
+        buf = st_mod.pack_request(req_id, url, data)
+        rid, u, out = st_mod.unpack_request(buf)
+        assert rid == req_id and u == url
+        import numpy as _np
+
+        if isinstance(out, _np.ndarray):
+            # numeric (possibly empty) lists intentionally convert to
+            # arrays on the tensor path
+            _np.testing.assert_allclose(
+                out.astype(_np.float64),
+                _np.asarray(data, dtype=_np.float64))
+        else:
+            assert out == data or (out is None and data is None)
