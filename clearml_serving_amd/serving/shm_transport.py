@@ -199,7 +199,13 @@ def _as_numpy(v) -> Optional[np.ndarray]:
         try:
             a = np.asarray(v)
             if a.dtype == np.float64:
-                a = a.astype(np.float32)
+                with np.errstate(over="ignore"):
+                    a32 = a.astype(np.float32)
+                # keep float64 when the downcast would overflow finite
+                # values to inf (serving inputs are float32-scale, but a
+                # silent inf would corrupt the request)
+                if np.all(np.isfinite(a32) | ~np.isfinite(a)):
+                    a = a32
             return a if a.dtype in _DTYPE_CODES else None
         except Exception:
             return None
