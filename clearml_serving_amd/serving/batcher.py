@@ -202,15 +202,35 @@ class DynamicBatcher:
             now = time.monotonic()
             self.stats["queue_wait_ms_sum"] += sum(
                 (now - b[2]) * 1000 for b in batch)
-            inputs = [b[0] for b in batch]
-            futures = [b[1] for b in batch]
-            # fire-and-continue: the next batch forms while this one stages
-            # and executes (in-flight count bounded by the slot queue);
-            # hold a reference so the task isn't garbage-collected mid-run
-            task = asyncio.get_running_loop().create_task(
-                self._dispatch(inputs, futures))
-            self._inflight.add(task)
-            task.add_done_callback(self._inflight.discard)
+            # group by input signature (keys/shapes/dtypes): a window can
+            # coalesce heterogeneous requests, and stacking them together
+            # would crash the WHOLE batch -- innocent co-batched requests
+            # included. Homogeneous traffic stays one group.
+            groups: Dict = {}
+            for inp, fut, t0 in batch:
+                groups.setdefault(self._signature(inp), []).append(
+                    (inp, fut))
+            for items in groups.values():
+                inputs = [it[0] for it in items]
+                futures = [it[1] for it in items]
+                # fire-and-continue: the next batch forms while this one
+                # stages and executes (in-flight bounded by the slot
+                # queue); hold a reference so the task isn't GC'd mid-run
+                task = asyncio.get_running_loop().create_task(
+                    self._dispatch(inputs, futures))
+                self._inflight.add(task)
+                task.add_done_callback(self._inflight.discard)
+
+    @staticmethod
+    def _signature(x):
+        try:
+            if isinstance(x, dict):
+                return tuple(sorted(
+                    (k, tuple(v.shape), str(v.dtype))
+                    for k, v in x.items()))
+            return (tuple(x.shape), str(x.dtype))
+        except Exception:
+            return ("opaque", id(type(x)))
 
     async def _dispatch(self, inputs, futures) -> None:
         try:

@@ -350,3 +350,47 @@ def test_fp16_torchscript_endpoint_allowed(processor, store, tmp_path):
     ))
     out = run(processor.process_request("dbl_ep", "", [1.0, 2.0]))
     assert np.allclose(np.asarray(out), [2.0, 4.0])
+
+
+def test_batcher_heterogeneous_requests_do_not_poison_neighbors():
+    """A coalescing window can catch requests with different shapes or
+    key sets; each signature group dispatches separately, so every
+    request gets its own correct result instead of one stack failure
+    500ing the whole window."""
+    import asyncio
+
+    import torch
+
+    from clearml_serving_amd.serving.batcher import DynamicBatcher
+
+    def model(x):
+        if isinstance(x, dict):
+            return x["a"] * 2
+        return x + 1
+
+    b = DynamicBatcher(model_fn=model, device="cpu", max_batch_size=16,
+                       max_queue_delay_us=30000, use_graphs=False)
+
+    async def main():
+        reqs = [
+            torch.ones(3),                 # shape (3,)
+            torch.ones(5),                 # shape (5,) -- different!
+            {"a": torch.ones(2)},          # dict input
+            torch.ones(3) * 2,
+            {"a": torch.ones(2) * 3},
+        ]
+        return await asyncio.gather(*[b.submit(r) for r in reqs])
+
+    loop = asyncio.new_event_loop()
+    try:
+        outs = loop.run_until_complete(main())
+    finally:
+        for t in asyncio.all_tasks(loop):
+            t.cancel()
+        loop.run_until_complete(asyncio.sleep(0))
+        loop.close()
+    torch.testing.assert_close(outs[0], torch.ones(3) + 1)
+    torch.testing.assert_close(outs[1], torch.ones(5) + 1)
+    torch.testing.assert_close(outs[2], torch.ones(2) * 2)
+    torch.testing.assert_close(outs[3], torch.ones(3) * 2 + 1)
+    torch.testing.assert_close(outs[4], torch.ones(2) * 6)
