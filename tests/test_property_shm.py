@@ -133,9 +133,21 @@ def test_request_framing_roundtrip_fuzz(url, req_id, data):
 
         if isinstance(out, _np.ndarray):
             # numeric (possibly empty) lists intentionally convert to
-            # arrays on the tensor path
+            # arrays on the tensor path; a single ""-keyed dict entry is
+            # the positional-array convention
+            ref = (data[""] if isinstance(data, dict)
+                   and set(data) == {""} else data)
             _np.testing.assert_allclose(
                 out.astype(_np.float64),
-                _np.asarray(data, dtype=_np.float64))
+                _np.asarray(ref, dtype=_np.float64))
+        elif isinstance(out, dict) and any(
+                isinstance(v, _np.ndarray) for v in out.values()):
+            # dicts whose values are all numeric lists ride the tensor
+            # path: keys preserved, values as arrays
+            assert set(out) == set(data)
+            for k in out:
+                _np.testing.assert_allclose(
+                    _np.asarray(out[k], dtype=_np.float64),
+                    _np.asarray(data[k], dtype=_np.float64))
         else:
             assert out == data or (out is None and data is None)
