@@ -302,3 +302,31 @@ def test_bare_score_rerank_version_routes(llm_client):
     assert r.status_code == 200, r.text
     from clearml_serving_amd import __version__
     assert r.json()["version"] == __version__
+
+
+def test_full_sampling_surface_through_http(llm_client):
+    """Client-visible contract: n + seed + penalties + logprobs in one
+    chat request; response carries every structure."""
+    body = {
+        "model": "test_llm", "messages": [{"role": "user", "content": "x"}],
+        "max_tokens": 6, "temperature": 0.9, "top_p": 0.9, "seed": 42,
+        "n": 2, "presence_penalty": 0.5, "frequency_penalty": 0.5,
+        "repetition_penalty": 1.1, "logprobs": True, "top_logprobs": 2,
+        "min_tokens": 2, "ignore_eos": True,
+    }
+    r1 = llm_client.post("/serve/openai/v1/chat/completions", json=body)
+    assert r1.status_code == 200, r1.text
+    out1 = r1.json()
+    assert len(out1["choices"]) == 2
+    for c in out1["choices"]:
+        assert len(c["logprobs"]["content"]) == 6
+        assert all(len(e["top_logprobs"]) == 2
+                   for e in c["logprobs"]["content"])
+    # seeded: the same request reproduces both choices exactly
+    r2 = llm_client.post("/serve/openai/v1/chat/completions", json=body)
+    out2 = r2.json()
+    assert [c["message"]["content"] for c in out1["choices"]] == \
+        [c["message"]["content"] for c in out2["choices"]]
+    # and the two seeded choices differ from each other (offset seeds)
+    assert out1["choices"][0]["message"]["content"] != \
+        out1["choices"][1]["message"]["content"]
