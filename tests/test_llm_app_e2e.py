@@ -330,3 +330,12 @@ def test_full_sampling_surface_through_http(llm_client):
     # and the two seeded choices differ from each other (offset seeds)
     assert out1["choices"][0]["message"]["content"] != \
         out1["choices"][1]["message"]["content"]
+
+
+def test_models_listing_via_get(llm_client):
+    """GET /serve/openai/v1/models with no body (the openai SDK's list
+    call) returns the llm endpoints."""
+    r = llm_client.get("/serve/openai/v1/models")
+    assert r.status_code == 200, r.text
+    ids = [m["id"] for m in r.json()["data"]]
+    assert "test_llm" in ids
