@@ -450,6 +450,7 @@ class Sequence:
         # blocks, and the chain hash at that point
         self.cached_upto = 0
         self.cache_hash = 0
+        self.text_sent = 0  # chars of decoded output already streamed
 
     def __len__(self):
         return len(self.prompt_ids) + len(self.output_ids)
@@ -1768,9 +1769,21 @@ class LlmEngine:
                 finished, reason = True, "length"
             s.finished = finished
             s.finish_reason = reason
+            # streamed text = the DELTA of the cumulative decode, not a
+            # per-token decode: a multi-byte character split across BPE
+            # tokens would otherwise stream as replacement chars (vLLM's
+            # incremental detokenization does the same). An incomplete
+            # UTF-8 tail is held back until the next token completes it.
+            full = self.tokenizer.decode(s.output_ids)
+            if len(full) < s.text_sent:  # non-prefix-stable decode: resync
+                s.text_sent = len(full)
+            delta = full[s.text_sent:]
+            if not finished and delta.endswith("�"):
+                delta = delta[:-1]
+            s.text_sent += len(delta)
             item = {
                 "token_ids": [tok],
-                "text": self.tokenizer.decode([tok]),
+                "text": delta,
                 "finished": finished,
                 "finish_reason": reason,
             }

@@ -1697,3 +1697,37 @@ def test_suffix_fim_guard():
     with pytest.raises(ValueError):
         run(eng.openai_completions(
             {"prompt": "a", "suffix": "z", "max_tokens": 2}, "m"))
+
+
+def test_streamed_text_deltas_reassemble_and_handle_split_utf8():
+    """Streamed 'text' fields are cumulative-decode deltas: concatenated
+    they equal the full decode, and a multi-byte character split across
+    byte tokens streams intact (held back until complete) instead of as
+    replacement chars."""
+    eng = tiny_engine()
+
+    async def gen(prompt_ids, n):
+        seq = await eng.add_request(list(prompt_ids), SamplingParams(
+            temperature=0.0, max_tokens=n, ignore_eos=True))
+        texts, toks = [], []
+        while True:
+            item = await seq.stream.get()
+            texts.append(item["text"])
+            toks.extend(item["token_ids"])
+            if item["finished"]:
+                return texts, toks
+
+    texts, toks = run(gen([5, 9, 13], 12))
+    assert "".join(texts) == eng.tokenizer.decode(toks)
+
+    # force a split multi-byte char through the emission path directly:
+    # "é" = 0xC3 0xA9 -> byte-tokenizer ids 0xC3+1, 0xA9+1
+    from clearml_serving_amd.engines.llm.engine import Sequence
+
+    s = Sequence("utf", [1], SamplingParams(max_tokens=10, ignore_eos=True))
+    eng._emit_tokens(s, [0xC3 + 1])
+    first = s.stream.get_nowait()
+    assert first["text"] == ""  # incomplete tail held back
+    eng._emit_tokens(s, [0xA9 + 1])
+    second = s.stream.get_nowait()
+    assert second["text"] == "é"
