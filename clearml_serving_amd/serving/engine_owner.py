@@ -23,6 +23,20 @@ from typing import Any, Dict
 GPU_ENGINE_TYPES = ("hip", "triton", "pytorch", "llm", "vllm")
 
 
+class _OwnerTelemetryShim:
+    """Processor-shaped view over the owner's engine map for the
+    Prometheus collectors (live dict reference; the owner mutates it)."""
+
+    def __init__(self, engines):
+        self._engine_processor_lookup = engines
+
+    def set_stats_sink(self, sink):
+        pass  # the owner emits no per-request stat dicts
+
+    def list_endpoint_logging(self):
+        return {}
+
+
 class EngineOwner:
     def __init__(self, store_root: str, session_id: str, prefix: str,
                  owner_idx: int, n_workers: int, device: int = 0,
@@ -53,6 +67,22 @@ class EngineOwner:
         self._inflight: Dict[int, "asyncio.Task"] = {}
         self._stop = False
         self.stats = {"requests": 0, "errors": 0, "aborts": 0}
+        # the owner has no HTTP server, so its batcher/LLM/GPU telemetry
+        # exports straight to Prometheus (fronts take STATS_PORT+worker;
+        # owners take STATS_PORT+100+owner so scrape configs can list
+        # both). A shim over the owner's OWN engine map feeds the
+        # per-stage collector -- inserting engines into the processor's
+        # lookup would let its hot-reload flush tear them down mid-request
+        # (the owner manages engine lifecycle itself in _get_engine).
+        try:
+            from ..statistics.collector import install_stats_sink
+
+            base = int(os.environ.get("CLEARML_SERVING_STATS_PORT", 9999))
+            if base > 0:
+                install_stats_sink(_OwnerTelemetryShim(self._engines),
+                                   port=base + 100 + owner_idx)
+        except Exception as ex:
+            print("[engine-owner] stats export unavailable: {}".format(ex))
 
     # ------------------------------------------------------------------ #
     def _get_engine(self, url: str):

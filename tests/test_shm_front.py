@@ -659,3 +659,72 @@ def test_oversize_record_rejected_not_deadlocked():
     finally:
         ring2.close()
         PyShmRing.unlink(name2)
+
+
+def test_owner_exports_batcher_telemetry(tmp_path):
+    """The engine owner (no HTTP server) exports its per-stage batcher
+    telemetry to Prometheus via a shim over its OWN engine map -- the
+    processor's reload flush must never tear owner engines down, so they
+    stay out of its lookup."""
+    import json as _json
+
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.engine_owner import (
+        EngineOwner, _OwnerTelemetryShim)
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.serving.shm_transport import ShmClient
+    from clearml_serving_amd.statistics.collector import (
+        BatcherStagesCollector)
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="tel", force_create=True)
+    card = tmp_path / "bert.json"
+    card.write_text(_json.dumps({"arch": "bert-base", "num_labels": 2,
+                                 "dtype": "float32", "vocab_size": 200}))
+    rec = store.register_model(name="b", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(
+        engine_type="hip", serving_url="telmodel", model_id=rec.model_id,
+        auxiliary_cfg={"use_graphs": False, "warmup": False,
+                       "max_queue_delay_us": 1000}))
+    proc.serialize()
+
+    prefix = "/cmls_tel_{}".format(os.getpid())
+    owner = EngineOwner(store_root=store_root, session_id=proc.get_id(),
+                        prefix=prefix, owner_idx=0, n_workers=1,
+                        ring_bytes=1 << 20)
+    client = ShmClient(prefix, 0, 1, ring_bytes=1 << 20)
+
+    async def main():
+        serve = asyncio.get_running_loop().create_task(owner.serve())
+        ids = np.arange(16, dtype=np.int64) % 100
+        out = await client.infer("telmodel", {
+            "input_ids": ids, "attention_mask": np.ones(16, np.int32)})
+        assert np.asarray(out).shape[-1] == 2
+        serve.cancel()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
+        client.close()
+
+    # the owner's engines stay OUT of the processor lookup (reload-flush
+    # isolation) but the shim-backed collector sees their batcher stats
+    assert "telmodel" not in owner.proc._engine_processor_lookup
+    reg = CollectorRegistry()
+    reg.register(BatcherStagesCollector(_OwnerTelemetryShim(owner._engines)))
+    text = generate_latest(reg).decode()
+    assert "serving_batches_total" in text
+    assert 'endpoint="telmodel"' in text
+    owner.close()
