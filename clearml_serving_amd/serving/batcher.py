@@ -223,12 +223,14 @@ class DynamicBatcher:
 
     @staticmethod
     def _signature(x):
+        # hot loop (once per request): torch.dtype/shape tuples are
+        # hashable as-is -- no string formatting
         try:
             if isinstance(x, dict):
                 return tuple(sorted(
-                    (k, tuple(v.shape), str(v.dtype))
+                    (k, tuple(v.shape), v.dtype)
                     for k, v in x.items()))
-            return (tuple(x.shape), str(x.dtype))
+            return (tuple(x.shape), x.dtype)
         except Exception:
             return ("opaque", id(type(x)))
 
