@@ -412,6 +412,16 @@ class ShmClient:
             self._futures.pop(req_id, None)
             raise RuntimeError(
                 "engine owner did not answer within {}s".format(timeout))
+        except asyncio.CancelledError:
+            # client disconnected mid-request: tell the owner to stop
+            # computing (LLM generations would otherwise run to
+            # max_tokens for a dead client)
+            self._futures.pop(req_id, None)
+            try:
+                ring.push(pack_abort(req_id))
+            except Exception:
+                pass
+            raise
         if status == STATUS_ERROR:
             raise RuntimeError(result)
         return result

@@ -728,3 +728,74 @@ def test_owner_exports_batcher_telemetry(tmp_path):
     assert "serving_batches_total" in text
     assert 'endpoint="telmodel"' in text
     owner.close()
+
+
+def test_nonstream_abort_crosses_the_ring(tmp_path):
+    """Cancelling an awaiting NON-stream infer (client disconnect) sends
+    an abort record: the owner cancels the in-flight generation instead
+    of running to max_tokens for a dead client."""
+    import json as _json
+
+    from clearml_serving_amd.schemas import ModelEndpoint
+    from clearml_serving_amd.serving.engine_owner import EngineOwner
+    from clearml_serving_amd.serving.processor import ModelRequestProcessor
+    from clearml_serving_amd.serving.shm_transport import ShmClient
+    from clearml_serving_amd.store import ServingStore
+
+    store_root = str(tmp_path / "store")
+    store = ServingStore(store_root)
+    proc = ModelRequestProcessor(store=store, name="nsab",
+                                 force_create=True)
+    card = tmp_path / "card.json"
+    card.write_text(_json.dumps({
+        "arch": "llama", "preset": "llama-tiny", "num_kv_blocks": 64,
+        "block_size": 16, "max_model_len": 512, "device": "cpu"}))
+    rec = store.register_model(name="t", project="p", path=str(card))
+    proc.add_endpoint(ModelEndpoint(engine_type="llm", serving_url="tl",
+                                    model_id=rec.model_id))
+    proc.serialize()
+
+    prefix = "/cmls_nsab_{}".format(os.getpid())
+    owner = EngineOwner(store_root=store_root, session_id=proc.get_id(),
+                        prefix=prefix, owner_idx=0, n_workers=1,
+                        ring_bytes=1 << 20)
+    client = ShmClient(prefix, 0, 1, ring_bytes=1 << 20)
+
+    async def main():
+        serve = asyncio.get_running_loop().create_task(owner.serve())
+        payload = {"__serve_type__": "v1_chat_completions",
+                   "__body__": {"messages": [{"role": "user",
+                                              "content": "hi"}],
+                                "max_tokens": 400, "temperature": 0,
+                                "ignore_eos": True}}
+        task = asyncio.ensure_future(client.infer("tl", payload))
+        await asyncio.sleep(0.3)  # generation well in flight
+        task.cancel()
+        try:
+            await task
+        except asyncio.CancelledError:
+            pass
+        eng = owner._engines["tl"]._engine
+        for _ in range(200):
+            await asyncio.sleep(0.05)
+            if owner.stats["aborts"] >= 1 and not eng.running \
+                    and not eng.waiting:
+                break
+        assert owner.stats["aborts"] >= 1
+        assert eng.stats["generated_tokens"] < 300
+        assert not eng.running and not eng.waiting
+        serve.cancel()
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        pending = asyncio.all_tasks(loop)
+        for t in pending:
+            t.cancel()
+        if pending:
+            loop.run_until_complete(
+                asyncio.gather(*pending, return_exceptions=True))
+        loop.close()
+        client.close()
+        owner.close()
