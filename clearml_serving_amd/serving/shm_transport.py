@@ -404,8 +404,13 @@ class ShmClient:
         self._futures[req_id] = fut
         payload = pack_request(req_id, url, data)
         ring = self.req_rings[owner]
-        while not ring.push(payload):
-            await asyncio.sleep(0.001)  # ring full: backpressure
+        try:
+            while not ring.push(payload):
+                await asyncio.sleep(0.001)  # ring full: backpressure
+        except asyncio.CancelledError:
+            # cancelled during backpressure: nothing reached the owner
+            self._futures.pop(req_id, None)
+            raise
         try:
             status, result = await asyncio.wait_for(fut, timeout=timeout)
         except asyncio.TimeoutError:
